@@ -1,0 +1,378 @@
+"""`smg launch` CLI (reference: model_gateway/src/main.rs — Cli :103,
+CliArgs :161, ~130 flags in 26 help groups, to_router_config :1296; the
+--prefill/--encode URL [bootstrap_port] positional pairs parse :1692-1711).
+
+Flag names match the reference so launch scripts port unchanged.
+"""
+from __future__ import annotations
+
+import argparse
+import asyncio
+import logging
+import sys
+from typing import List, Optional, Tuple
+
+from .config import (
+    ConnectionMode,
+    PolicyConfig,
+    RouterConfig,
+    RoutingMode,
+)
+
+
+def _extract_url_port_pairs(argv: List[str], flag: str) -> Tuple[List[Tuple[str, Optional[int]]], List[str]]:
+    """--prefill URL [bootstrap_port] may repeat (reference main.rs:1692-1711)."""
+    out: List[Tuple[str, Optional[int]]] = []
+    rest: List[str] = []
+    i = 0
+    while i < len(argv):
+        if argv[i] == flag and i + 1 < len(argv):
+            url = argv[i + 1]
+            port: Optional[int] = None
+            skip = 2
+            if i + 2 < len(argv) and not argv[i + 2].startswith("--"):
+                try:
+                    port = int(argv[i + 2])
+                    skip = 3
+                except ValueError:
+                    pass
+            out.append((url, port))
+            i += skip
+        else:
+            rest.append(argv[i])
+            i += 1
+    return out, rest
+
+
+def build_parser() -> argparse.ArgumentParser:
+    p = argparse.ArgumentParser(
+        prog="smg",
+        description="smg — MI355X-native model-routing gateway (SMG-compatible CLI)",
+    )
+    p.add_argument("command", nargs="?", default="launch", choices=["launch", "start", "serve"])
+    g = p.add_argument_group("Network")
+    g.add_argument("--host", default="0.0.0.0")
+    g.add_argument("--port", type=int, default=30000)
+    g.add_argument("--health-check-port", type=int, default=None)
+
+    g = p.add_argument_group("Worker Configuration")
+    g.add_argument("--worker-urls", nargs="*", default=[])
+    g.add_argument("--decode", action="append", default=[], dest="decode_urls")
+    g.add_argument("--worker-startup-timeout-secs", type=int, default=600)
+    g.add_argument("--worker-startup-delay", type=float, default=0.0)
+    g.add_argument("--worker-startup-check-interval", type=int, default=30)
+    g.add_argument("--load-monitor-interval", type=int, default=5)
+    g.add_argument("--engine-metrics", action="store_true")
+
+    g = p.add_argument_group("Routing Policy")
+    g.add_argument("--policy", default="cache_aware", choices=[
+        "random", "round_robin", "passthrough", "cache_aware", "power_of_two",
+        "least_load", "prefix_hash", "consistent_hashing", "manual", "bucket"])
+    g.add_argument("--cache-threshold", type=float, default=0.3)
+    g.add_argument("--balance-abs-threshold", type=int, default=64)
+    g.add_argument("--balance-rel-threshold", type=float, default=1.5)
+    g.add_argument("--balance-token-usage-threshold", type=float, default=1.0)
+    g.add_argument("--overload-token-usage-threshold", type=float, default=1.0)
+    g.add_argument("--eviction-interval", type=int, default=120)
+    g.add_argument("--max-tree-size", type=int, default=67_108_864)
+    g.add_argument("--block-size", type=int, default=16)
+    g.add_argument("--max-idle-secs", type=int, default=14_400)
+    g.add_argument("--assignment-mode", default="random", choices=["random", "min_load", "min_group"])
+    g.add_argument("--prefix-token-count", type=int, default=256)
+    g.add_argument("--prefix-hash-load-factor", type=float, default=1.25)
+    g.add_argument("--least-load-kv-pressure-weight", type=float, default=0.15)
+    g.add_argument("--least-load-default-throughput", type=float, default=2000.0)
+    g.add_argument("--least-load-mean-prefill-tokens", type=int, default=1024)
+    g.add_argument("--dp-aware", action="store_true")
+    g.add_argument("--routing-key-override", default=None)
+    g.add_argument("--enable-igw", action="store_true")
+    g.add_argument("--dp-minimum-tokens-scheduler", action="store_true")
+    g.add_argument("--no-gpu-tree", action="store_true", help="MI355X: disable the GPU-resident prefix tree")
+
+    g = p.add_argument_group("PD Disaggregation")
+    g.add_argument("--pd-disaggregation", action="store_true")
+    g.add_argument("--epd-disaggregation", action="store_true")
+    g.add_argument("--prefill-policy", default=None)
+    g.add_argument("--decode-policy", default=None)
+    g.add_argument("--encode-policy", default=None)
+
+    g = p.add_argument_group("Connection")
+    g.add_argument("--connection-mode", default="http", choices=["http", "grpc", "rccl"])
+    g.add_argument("--multimodal-tensor-transport", default="inline", choices=["inline", "shm", "rdma", "xgmi"])
+    g.add_argument("--multimodal-shm-min-bytes", type=int, default=65_536)
+
+    g = p.add_argument_group("Model / Tokenizer")
+    g.add_argument("--model-path", default=None)
+    g.add_argument("--tokenizer-path", default=None)
+    g.add_argument("--chat-template", default=None)
+    g.add_argument("--disable-tokenizer-autoload", action="store_true")
+    g.add_argument("--tokenizer-cache-enable-l0", action="store_true", default=True)
+    g.add_argument("--tokenizer-cache-l0-max-entries", type=int, default=8192)
+    g.add_argument("--tokenizer-cache-enable-l1", action="store_true")
+    g.add_argument("--tokenizer-cache-l1-max-memory", type=int, default=64 << 20)
+    g.add_argument("--reasoning-parser", default=None)
+    g.add_argument("--tool-call-parser", default=None)
+    g.add_argument("--model-alias", action="append", default=[], help="alias=model_id, repeatable")
+
+    g = p.add_argument_group("Rate Limiting / Admission")
+    g.add_argument("--max-concurrent-requests", type=int, default=-1)
+    g.add_argument("--queue-size", type=int, default=100)
+    g.add_argument("--queue-timeout-secs", type=int, default=60)
+    g.add_argument("--rate-limit-tokens-per-second", type=int, default=None)
+    g.add_argument("--priority-scheduler-enabled", action="store_true")
+    g.add_argument("--priority-scheduler-default-max-class", default="default")
+    g.add_argument("--priority-scheduler-config", default=None)
+    g.add_argument("--priority-scheduler-tenant-metric-top-n", type=int, default=0)
+    g.add_argument("--tenant-rate-limit-enabled", action="store_true")
+    g.add_argument("--tenant-rate-limit-config", default=None)
+
+    g = p.add_argument_group("Retries")
+    g.add_argument("--retry-max-retries", type=int, default=5)
+    g.add_argument("--retry-initial-backoff-ms", type=int, default=50)
+    g.add_argument("--retry-max-backoff-ms", type=int, default=30_000)
+    g.add_argument("--retry-backoff-multiplier", type=float, default=1.5)
+    g.add_argument("--retry-jitter-factor", type=float, default=0.2)
+    g.add_argument("--disable-retries", action="store_true")
+
+    g = p.add_argument_group("Circuit Breaker")
+    g.add_argument("--cb-failure-threshold", type=int, default=10)
+    g.add_argument("--cb-success-threshold", type=int, default=3)
+    g.add_argument("--cb-timeout-duration-secs", type=int, default=60)
+    g.add_argument("--cb-window-duration-secs", type=int, default=120)
+    g.add_argument("--disable-circuit-breaker", action="store_true")
+
+    g = p.add_argument_group("Health Checks")
+    g.add_argument("--health-failure-threshold", type=int, default=3)
+    g.add_argument("--health-success-threshold", type=int, default=2)
+    g.add_argument("--health-check-timeout-secs", type=int, default=5)
+    g.add_argument("--health-check-interval-secs", type=int, default=60)
+    g.add_argument("--health-check-endpoint", default="/health")
+    g.add_argument("--disable-health-check", action="store_true")
+    g.add_argument("--remove-unhealthy-workers", action="store_true")
+    g.add_argument("--drain-settle-secs", type=int, default=0)
+
+    g = p.add_argument_group("Authentication")
+    g.add_argument("--api-key", default=None)
+    g.add_argument("--tenant-api-key", action="append", default=[], dest="tenant_api_keys",
+                   help="key=tenant, repeatable")
+    g.add_argument("--control-plane-api-keys", nargs="*", default=[])
+    g.add_argument("--jwt-issuer", default=None)
+    g.add_argument("--jwt-audience", default=None)
+    g.add_argument("--jwt-jwks-uri", default=None)
+    g.add_argument("--disable-audit-logging", action="store_true")
+    g.add_argument("--trust-tenant-header", action="store_true")
+    g.add_argument("--tenant-header-name", default="x-smg-tenant")
+
+    g = p.add_argument_group("Storage")
+    g.add_argument("--backend", default="memory", choices=["memory", "none", "postgres", "redis", "oracle"])
+    g.add_argument("--history-backend", default=None)
+    g.add_argument("--postgres-db-url", default=None)
+    g.add_argument("--redis-url", default=None)
+
+    g = p.add_argument_group("Observability")
+    g.add_argument("--log-dir", default=None)
+    g.add_argument("--log-level", default="info")
+    g.add_argument("--log-json", action="store_true")
+    g.add_argument("--prometheus-port", type=int, default=29000)
+    g.add_argument("--prometheus-host", default="0.0.0.0")
+    g.add_argument("--request-id-headers", nargs="*",
+                   default=["x-request-id", "x-correlation-id", "x-trace-id", "request-id"])
+    g.add_argument("--enable-trace", action="store_true")
+    g.add_argument("--otlp-traces-endpoint", default=None)
+
+    g = p.add_argument_group("HA Mesh")
+    g.add_argument("--enable-mesh", action="store_true")
+    g.add_argument("--mesh-server-name", default=None)
+    g.add_argument("--mesh-host", default="0.0.0.0")
+    g.add_argument("--mesh-advertise-host", default=None)
+    g.add_argument("--mesh-port", type=int, default=32300)
+    g.add_argument("--mesh-peer-urls", nargs="*", default=[])
+
+    g = p.add_argument_group("Service Discovery (Kubernetes)")
+    g.add_argument("--service-discovery", action="store_true")
+    g.add_argument("--selector", nargs="*", default=[])
+    g.add_argument("--service-discovery-port", type=int, default=8000)
+    g.add_argument("--service-discovery-namespace", default=None)
+
+    g = p.add_argument_group("MI355X Data Plane (RCCL over xGMI)")
+    g.add_argument("--rccl-world-size", type=int, default=0, help=">0 enables the RCCL tick plane")
+    g.add_argument("--rccl-tick-us", type=int, default=200)
+
+    g = p.add_argument_group("Misc")
+    g.add_argument("--request-timeout-secs", type=int, default=1800)
+    g.add_argument("--max-payload-size", type=int, default=512 << 20)
+    g.add_argument("--cors-allowed-origins", nargs="*", default=[])
+    g.add_argument("--mcp-config-path", default=None)
+    g.add_argument("--tls-cert-path", default=None)
+    g.add_argument("--tls-key-path", default=None)
+    g.add_argument("--version", "-V", action="version", version="smg-amd 0.1.0")
+    return p
+
+
+def _policy_cfg(args, name: Optional[str] = None) -> PolicyConfig:
+    return PolicyConfig(
+        name=name or args.policy,
+        cache_threshold=args.cache_threshold,
+        balance_abs_threshold=args.balance_abs_threshold,
+        balance_rel_threshold=args.balance_rel_threshold,
+        balance_token_usage_threshold=args.balance_token_usage_threshold,
+        overload_token_usage_threshold=args.overload_token_usage_threshold,
+        eviction_interval_secs=args.eviction_interval,
+        max_tree_size=args.max_tree_size,
+        block_size=args.block_size,
+        max_idle_secs=args.max_idle_secs,
+        assignment_mode=args.assignment_mode,
+        prefix_token_count=args.prefix_token_count,
+        prefix_hash_load_factor=args.prefix_hash_load_factor,
+        least_load_kv_pressure_weight=args.least_load_kv_pressure_weight,
+        least_load_default_throughput=args.least_load_default_throughput,
+        least_load_mean_prefill_tokens=args.least_load_mean_prefill_tokens,
+        gpu_tree=not args.no_gpu_tree,
+    )
+
+
+def to_router_config(argv: Optional[List[str]] = None) -> RouterConfig:
+    """CLI args -> RouterConfig (reference main.rs:1296 to_router_config)."""
+    argv = list(sys.argv[1:] if argv is None else argv)
+    prefill_urls, argv = _extract_url_port_pairs(argv, "--prefill")
+    encode_urls, argv = _extract_url_port_pairs(argv, "--encode")
+    args = build_parser().parse_args(argv)
+
+    mode = RoutingMode.REGULAR
+    if args.epd_disaggregation or encode_urls:
+        mode = RoutingMode.ENCODE_PREFILL_DECODE
+    elif args.pd_disaggregation or prefill_urls:
+        mode = RoutingMode.PREFILL_DECODE
+
+    cfg = RouterConfig(
+        host=args.host,
+        port=args.port,
+        health_check_port=args.health_check_port,
+        worker_urls=args.worker_urls,
+        prefill_urls=prefill_urls,
+        decode_urls=args.decode_urls,
+        encode_urls=encode_urls,
+        mode=mode,
+        connection_mode=ConnectionMode(args.connection_mode),
+        enable_igw=args.enable_igw,
+        policy=_policy_cfg(args),
+        prefill_policy=_policy_cfg(args, args.prefill_policy) if args.prefill_policy else None,
+        decode_policy=_policy_cfg(args, args.decode_policy) if args.decode_policy else None,
+        encode_policy=_policy_cfg(args, args.encode_policy) if args.encode_policy else None,
+        dp_aware=args.dp_aware,
+        dp_minimum_tokens_scheduler=args.dp_minimum_tokens_scheduler,
+        routing_key_override=args.routing_key_override,
+        model_path=args.model_path,
+        tokenizer_path=args.tokenizer_path,
+        chat_template=args.chat_template,
+        disable_tokenizer_autoload=args.disable_tokenizer_autoload,
+        reasoning_parser=args.reasoning_parser,
+        tool_call_parser=args.tool_call_parser,
+        request_timeout_secs=args.request_timeout_secs,
+        max_payload_size=args.max_payload_size,
+        cors_allowed_origins=args.cors_allowed_origins,
+        request_id_headers=args.request_id_headers,
+        trust_tenant_header=args.trust_tenant_header,
+        tenant_header_name=args.tenant_header_name,
+        log_dir=args.log_dir,
+        log_level=args.log_level,
+        log_json=args.log_json,
+        prometheus_port=args.prometheus_port,
+        prometheus_host=args.prometheus_host,
+        mcp_config_path=args.mcp_config_path,
+        tls_cert_path=args.tls_cert_path,
+        tls_key_path=args.tls_key_path,
+        worker_startup_timeout_secs=args.worker_startup_timeout_secs,
+        worker_startup_delay=args.worker_startup_delay,
+        load_monitor_interval=args.load_monitor_interval,
+        engine_metrics=args.engine_metrics,
+        multimodal_tensor_transport=args.multimodal_tensor_transport,
+        multimodal_shm_min_bytes=args.multimodal_shm_min_bytes,
+    )
+    cfg.model_aliases = dict(a.split("=", 1) for a in args.model_alias if "=" in a)
+    cfg.retry.max_retries = args.retry_max_retries
+    cfg.retry.initial_backoff_ms = args.retry_initial_backoff_ms
+    cfg.retry.max_backoff_ms = args.retry_max_backoff_ms
+    cfg.retry.backoff_multiplier = args.retry_backoff_multiplier
+    cfg.retry.jitter_factor = args.retry_jitter_factor
+    cfg.retry.disable = args.disable_retries
+    cfg.circuit_breaker.failure_threshold = args.cb_failure_threshold
+    cfg.circuit_breaker.success_threshold = args.cb_success_threshold
+    cfg.circuit_breaker.timeout_duration_secs = args.cb_timeout_duration_secs
+    cfg.circuit_breaker.window_duration_secs = args.cb_window_duration_secs
+    cfg.circuit_breaker.disable = args.disable_circuit_breaker
+    cfg.health_check.failure_threshold = args.health_failure_threshold
+    cfg.health_check.success_threshold = args.health_success_threshold
+    cfg.health_check.timeout_secs = args.health_check_timeout_secs
+    cfg.health_check.check_interval_secs = args.health_check_interval_secs
+    cfg.health_check.endpoint = args.health_check_endpoint
+    cfg.health_check.disable = args.disable_health_check
+    cfg.health_check.remove_unhealthy_workers = args.remove_unhealthy_workers
+    cfg.health_check.drain_settle_secs = args.drain_settle_secs
+    cfg.rate_limit.max_concurrent_requests = args.max_concurrent_requests
+    cfg.rate_limit.queue_size = args.queue_size
+    cfg.rate_limit.queue_timeout_secs = args.queue_timeout_secs
+    cfg.rate_limit.tokens_per_second = args.rate_limit_tokens_per_second
+    cfg.priority_scheduler.enabled = args.priority_scheduler_enabled
+    cfg.priority_scheduler.default_max_class = args.priority_scheduler_default_max_class
+    cfg.priority_scheduler.config_path = args.priority_scheduler_config
+    cfg.priority_scheduler.tenant_metric_top_n = args.priority_scheduler_tenant_metric_top_n
+    cfg.tenant_rate_limit.enabled = args.tenant_rate_limit_enabled
+    cfg.tenant_rate_limit.config_path = args.tenant_rate_limit_config
+    cfg.auth.api_key = args.api_key
+    cfg.auth.tenant_api_keys = dict(a.split("=", 1) for a in args.tenant_api_keys if "=" in a)
+    cfg.auth.control_plane_api_keys = args.control_plane_api_keys
+    cfg.auth.jwt_issuer = args.jwt_issuer
+    cfg.auth.jwt_audience = args.jwt_audience
+    cfg.auth.jwt_jwks_uri = args.jwt_jwks_uri
+    cfg.auth.disable_audit_logging = args.disable_audit_logging
+    cfg.storage.backend = args.backend
+    cfg.storage.history_backend = args.history_backend
+    cfg.storage.postgres_db_url = args.postgres_db_url
+    cfg.storage.redis_url = args.redis_url
+    cfg.mesh.enabled = args.enable_mesh
+    cfg.mesh.server_name = args.mesh_server_name
+    cfg.mesh.host = args.mesh_host
+    cfg.mesh.advertise_host = args.mesh_advertise_host
+    cfg.mesh.port = args.mesh_port
+    cfg.mesh.peer_urls = args.mesh_peer_urls
+    cfg.discovery.enabled = args.service_discovery
+    cfg.discovery.selector = dict(s.split("=", 1) for s in args.selector if "=" in s)
+    cfg.discovery.port = args.service_discovery_port
+    cfg.discovery.namespace = args.service_discovery_namespace
+    cfg.trace.enabled = args.enable_trace
+    cfg.trace.otlp_endpoint = args.otlp_traces_endpoint
+    cfg.tokenizer_cache.enable_l0 = args.tokenizer_cache_enable_l0
+    cfg.tokenizer_cache.l0_max_entries = args.tokenizer_cache_l0_max_entries
+    cfg.tokenizer_cache.enable_l1 = args.tokenizer_cache_enable_l1
+    cfg.tokenizer_cache.l1_max_memory = args.tokenizer_cache_l1_max_memory
+    if args.rccl_world_size > 0:
+        cfg.rccl.enabled = True
+        cfg.rccl.world_size = args.rccl_world_size
+        cfg.rccl.tick_interval_us = args.rccl_tick_us
+    return cfg
+
+
+def main(argv: Optional[List[str]] = None) -> None:
+    cfg = to_router_config(argv)
+    logging.basicConfig(
+        level=getattr(logging, cfg.log_level.upper(), logging.INFO),
+        format="%(asctime)s %(levelname)s %(name)s %(message)s",
+    )
+    from .server.app import startup
+
+    async def _run():
+        ctx = await startup(cfg)
+        try:
+            await asyncio.Event().wait()
+        finally:
+            await ctx.shutdown()
+
+    try:
+        asyncio.run(_run())
+    except KeyboardInterrupt:
+        pass
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,404 @@
+"""Gateway HTTP server (reference: model_gateway/src/server.rs — build_app :751,
+startup :984, route table :781-935).
+
+aiohttp.web application (C-accelerated HTTP parsing; uvicorn here has no
+httptools/uvloop) with the reference's route groups:
+  protected (inference), public (health/model info), admin (cache/profiling/
+  parsers/tokenizers), worker CRUD.  Middleware outside-in: request-id ->
+  metrics -> auth -> tenant resolution -> admission (concurrency limiter).
+"""
+from __future__ import annotations
+
+import asyncio
+import json
+import logging
+import time
+import uuid
+from typing import Optional
+
+from aiohttp import web
+
+from ..config import RouterConfig
+from ..protocols.openai import error_body
+from ..routers.base import RouteRequest
+from ..routers.factory import RouterManager
+from ..workers.worker import Worker, WorkerType
+from .app_context import AppContext
+
+log = logging.getLogger("smg.server")
+
+CTX_KEY = web.AppKey("smg_ctx", AppContext)
+
+
+# --------------------------------------------------------------------------
+# middleware
+# --------------------------------------------------------------------------
+@web.middleware
+async def request_id_middleware(request: web.Request, handler):
+    ctx: AppContext = request.app[CTX_KEY]
+    rid = None
+    for h in ctx.config.request_id_headers:
+        rid = request.headers.get(h)
+        if rid:
+            break
+    request["request_id"] = rid or uuid.uuid4().hex
+    resp = await handler(request)
+    try:
+        resp.headers["x-request-id"] = request["request_id"]
+    except Exception:
+        pass
+    return resp
+
+
+@web.middleware
+async def metrics_middleware(request: web.Request, handler):
+    ctx: AppContext = request.app[CTX_KEY]
+    t0 = time.perf_counter()
+    try:
+        resp = await handler(request)
+        return resp
+    finally:
+        status = getattr(locals().get("resp"), "status", 500)
+        ctx.metrics.observe_http(request.path, request.method, status, time.perf_counter() - t0)
+
+
+@web.middleware
+async def auth_middleware(request: web.Request, handler):
+    ctx: AppContext = request.app[CTX_KEY]
+    auth = ctx.config.auth
+    if request.path in PUBLIC_PATHS or (not auth.api_key and not auth.tenant_api_keys):
+        return await handler(request)
+    header = request.headers.get("authorization", "")
+    token = header[7:] if header.lower().startswith("bearer ") else request.headers.get("x-api-key", "")
+    if auth.api_key and _ct_eq(token, auth.api_key):
+        return await handler(request)
+    tenant = auth.tenant_api_keys.get(token)
+    if tenant is not None:
+        request["tenant_id"] = tenant
+        return await handler(request)
+    return web.Response(
+        status=401, body=error_body("invalid API key", 401, "authentication_error"),
+        content_type="application/json",
+    )
+
+
+def _ct_eq(a: str, b: str) -> bool:
+    import hmac
+
+    return hmac.compare_digest(a.encode(), b.encode())
+
+
+@web.middleware
+async def tenant_middleware(request: web.Request, handler):
+    ctx: AppContext = request.app[CTX_KEY]
+    if "tenant_id" not in request and ctx.config.trust_tenant_header:
+        t = request.headers.get(ctx.config.tenant_header_name)
+        if t:
+            request["tenant_id"] = t
+    return await handler(request)
+
+
+@web.middleware
+async def admission_middleware(request: web.Request, handler):
+    """Concurrency limiter + bounded queue (reference middleware/concurrency.rs);
+    replaced by the priority scheduler when enabled (middleware/scheduler/)."""
+    ctx: AppContext = request.app[CTX_KEY]
+    if request.path in PUBLIC_PATHS or request.path in ADMIN_PATHS:
+        return await handler(request)
+    if ctx.scheduler is not None:
+        return await ctx.scheduler.admit(request, handler)
+    limiter = ctx.rate_limiter
+    if limiter is None:
+        return await handler(request)
+    return await limiter.admit(request, handler)
+
+
+PUBLIC_PATHS = {
+    "/liveness",
+    "/readiness",
+    "/health",
+    "/health_generate",
+    "/engine_metrics",
+    "/v1/models",
+    "/get_model_info",
+    "/get_server_info",
+    "/metrics",
+}
+ADMIN_PATHS = {
+    "/flush_cache",
+    "/start_profile",
+    "/stop_profile",
+    "/get_loads",
+    "/parse/function_call",
+    "/parse/reasoning",
+    "/workers",
+}
+
+
+# --------------------------------------------------------------------------
+# handlers
+# --------------------------------------------------------------------------
+async def _read_json(request: web.Request) -> Optional[dict]:
+    raw = await request.read()
+    if not raw:
+        return None, b""
+    try:
+        return json.loads(raw), raw
+    except json.JSONDecodeError:
+        raise web.HTTPBadRequest(body=error_body("invalid JSON body"), content_type="application/json")
+
+
+async def _proxy_endpoint(request: web.Request) -> web.StreamResponse:
+    """Shared handler for all inference endpoints: parse -> RouterManager ->
+    stream or unary response."""
+    ctx: AppContext = request.app[CTX_KEY]
+    body, raw = await _read_json(request)
+    route_req = RouteRequest(
+        path=request.path,
+        method=request.method,
+        body=body,
+        raw_body=raw,
+        headers=dict(request.headers),
+        request_id=request.get("request_id", ""),
+        tenant_id=request.get("tenant_id"),
+        routing_key=request.headers.get("x-smg-routing-key") or (ctx.config.routing_key_override and request.headers.get(ctx.config.routing_key_override)),
+    )
+    ctx.inflight += 1
+    try:
+        resp = await ctx.router_manager.route(route_req)
+        if resp.is_stream:
+            out = web.StreamResponse(status=resp.status, headers={"content-type": "text/event-stream", **resp.headers})
+            out.enable_chunked_encoding()
+            await out.prepare(request)
+            try:
+                async for chunk in resp.stream:
+                    await out.write(chunk)
+            except (ConnectionResetError, asyncio.CancelledError):
+                pass
+            await out.write_eof()
+            return out
+        hdrs = dict(resp.headers)
+        ctype = hdrs.pop("content-type", "application/json").split(";")[0]
+        return web.Response(status=resp.status, body=resp.body, headers=hdrs, content_type=ctype)
+    finally:
+        ctx.inflight -= 1
+
+
+async def liveness(request):
+    return web.json_response({"status": "alive"})
+
+
+async def readiness(request):
+    ctx: AppContext = request.app[CTX_KEY]
+    healthy = ctx.worker_registry.healthy_count()
+    status = 200 if healthy > 0 or len(ctx.worker_registry) == 0 else 503
+    return web.json_response({"status": "ready" if status == 200 else "not ready", "healthy_workers": healthy}, status=status)
+
+
+async def health(request):
+    return web.json_response({"status": "healthy"})
+
+
+async def v1_models(request):
+    ctx: AppContext = request.app[CTX_KEY]
+    models = ctx.worker_registry.models()
+    data = [{"id": m, "object": "model", "created": 0, "owned_by": "smg"} for m in models]
+    return web.json_response({"object": "list", "data": data})
+
+
+async def get_model_info(request):
+    ctx: AppContext = request.app[CTX_KEY]
+    models = ctx.worker_registry.models()
+    return web.json_response({"model_path": models[0] if models else None, "is_generation": True})
+
+
+async def get_server_info(request):
+    ctx: AppContext = request.app[CTX_KEY]
+    from .. import __version__
+
+    return web.json_response(
+        {
+            "version": __version__,
+            "workers": [w.to_dict() for w in ctx.worker_registry.all()],
+            "policy": ctx.config.policy.name,
+            "mode": ctx.config.mode.value,
+            "connection_mode": ctx.config.connection_mode.value,
+        }
+    )
+
+
+async def metrics_endpoint(request):
+    ctx: AppContext = request.app[CTX_KEY]
+    return web.Response(body=ctx.metrics.export(), content_type="text/plain")
+
+
+async def engine_metrics(request):
+    ctx: AppContext = request.app[CTX_KEY]
+    out = {}
+    for w in ctx.worker_registry.all():
+        out[w.url] = {"token_usage": w.token_usage, "gen_throughput": w.gen_throughput, "load": w.active_requests}
+    return web.json_response(out)
+
+
+async def get_loads(request):
+    ctx: AppContext = request.app[CTX_KEY]
+    loads = await ctx.router_manager.default_router.get_loads()
+    return web.json_response(loads)
+
+
+async def flush_cache(request):
+    ctx: AppContext = request.app[CTX_KEY]
+    resp = await ctx.router_manager.default_router.flush_cache()
+    return web.Response(status=resp.status, body=resp.body, content_type="application/json")
+
+
+async def parse_function_call(request):
+    ctx: AppContext = request.app[CTX_KEY]
+    body, _ = await _read_json(request)
+    from ..parsers.tool import parse_complete
+
+    parser = (body or {}).get("tool_call_parser") or ctx.config.tool_call_parser or "json"
+    text = (body or {}).get("text", "")
+    tools = (body or {}).get("tools")
+    normal, calls = parse_complete(parser, text, tools)
+    return web.json_response({"normal_text": normal, "calls": calls})
+
+
+async def parse_reasoning(request):
+    ctx: AppContext = request.app[CTX_KEY]
+    body, _ = await _read_json(request)
+    from ..parsers.reasoning import parse_reasoning_complete
+
+    parser = (body or {}).get("reasoning_parser") or ctx.config.reasoning_parser or "deepseek_r1"
+    text = (body or {}).get("text", "")
+    reasoning, normal = parse_reasoning_complete(parser, text)
+    return web.json_response({"reasoning_text": reasoning, "text": normal})
+
+
+# ---- worker CRUD (reference worker/service.rs REST /workers) -------------
+async def create_worker(request):
+    ctx: AppContext = request.app[CTX_KEY]
+    body, _ = await _read_json(request)
+    if not body or "url" not in body:
+        return web.Response(status=400, body=error_body("'url' is required"), content_type="application/json")
+    w = Worker(
+        body["url"],
+        model_id=body.get("model_id", "default"),
+        worker_type=WorkerType(body.get("worker_type", "regular")),
+        labels=body.get("labels") or {},
+        priority=int(body.get("priority", 0)),
+        cost=float(body.get("cost", 1.0)),
+        api_key=body.get("api_key"),
+        bootstrap_port=body.get("bootstrap_port"),
+        model_aliases=body.get("model_aliases") or [],
+        circuit_breaker_config=ctx.config.circuit_breaker,
+    )
+    ctx.worker_registry.register(w)
+    return web.json_response(w.to_dict(), status=201)
+
+
+async def list_workers(request):
+    ctx: AppContext = request.app[CTX_KEY]
+    return web.json_response({"workers": [w.to_dict() for w in ctx.worker_registry.all()]})
+
+
+async def get_worker(request):
+    ctx: AppContext = request.app[CTX_KEY]
+    wid = request.match_info["worker_id"]
+    w = ctx.worker_registry.get_by_url(wid) or (
+        ctx.worker_registry.get(int(wid)) if wid.isdigit() else None
+    )
+    if w is None:
+        return web.Response(status=404, body=error_body("worker not found", 404), content_type="application/json")
+    return web.json_response(w.to_dict())
+
+
+async def delete_worker(request):
+    ctx: AppContext = request.app[CTX_KEY]
+    wid = request.match_info["worker_id"]
+    w = ctx.worker_registry.remove_by_url(wid) or (
+        ctx.worker_registry.remove(int(wid)) if wid.isdigit() else None
+    )
+    if w is None:
+        return web.Response(status=404, body=error_body("worker not found", 404), content_type="application/json")
+    return web.json_response({"status": "removed", "url": w.url})
+
+
+# --------------------------------------------------------------------------
+# app assembly
+# --------------------------------------------------------------------------
+def build_app(ctx: AppContext) -> web.Application:
+    app = web.Application(
+        middlewares=[
+            request_id_middleware,
+            metrics_middleware,
+            auth_middleware,
+            tenant_middleware,
+            admission_middleware,
+        ],
+        client_max_size=ctx.config.max_payload_size,
+    )
+    app[CTX_KEY] = ctx
+
+    inference_paths = [
+        "/generate",
+        "/v1/chat/completions",
+        "/v1/completions",
+        "/rerank",
+        "/v1/rerank",
+        "/v1/embeddings",
+        "/v1/messages",
+        "/v1/classify",
+        "/v1/responses",
+    ]
+    for p in inference_paths:
+        app.router.add_post(p, _proxy_endpoint)
+
+    app.router.add_get("/liveness", liveness)
+    app.router.add_get("/readiness", readiness)
+    app.router.add_get("/health", health)
+    app.router.add_get("/health_generate", health)
+    app.router.add_get("/engine_metrics", engine_metrics)
+    app.router.add_get("/v1/models", v1_models)
+    app.router.add_get("/get_model_info", get_model_info)
+    app.router.add_get("/get_server_info", get_server_info)
+    app.router.add_get("/metrics", metrics_endpoint)
+
+    app.router.add_post("/flush_cache", flush_cache)
+    app.router.add_get("/get_loads", get_loads)
+    app.router.add_post("/parse/function_call", parse_function_call)
+    app.router.add_post("/parse/reasoning", parse_reasoning)
+
+    app.router.add_post("/workers", create_worker)
+    app.router.add_get("/workers", list_workers)
+    app.router.add_get("/workers/{worker_id:.*}", get_worker)
+    app.router.add_delete("/workers/{worker_id:.*}", delete_worker)
+
+    # tokenize/detokenize wired once the tokenizer registry exists
+    from .tokenize_routes import add_tokenize_routes
+
+    add_tokenize_routes(app)
+    return app
+
+
+async def startup(config: RouterConfig, serve: bool = True) -> AppContext:
+    """Ordered boot (reference server.rs:984): validate -> metrics -> context ->
+    workers -> router manager -> monitor -> app -> serve."""
+    config.validate()
+    ctx = AppContext(config)
+    ctx.init_workers_from_config()
+    from ..rate_limit.limiter import ConcurrencyLimiter
+
+    if config.rate_limit.max_concurrent_requests > 0:
+        ctx.rate_limiter = ConcurrencyLimiter(config.rate_limit)
+    ctx.router_manager = RouterManager(ctx, config)
+    await ctx.start_background()
+    if serve:
+        app = build_app(ctx)
+        runner = web.AppRunner(app, access_log=None)
+        await runner.setup()
+        site = web.TCPSite(runner, config.host, config.port)
+        await site.start()
+        ctx._runner = runner
+        log.info("smg gateway listening on %s:%d", config.host, config.port)
+    return ctx

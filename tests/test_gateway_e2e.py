@@ -1,0 +1,273 @@
+"""Gateway end-to-end on CPU: in-process aiohttp app + mock workers over the
+sim:// transport (model: reference model_gateway/tests/common/test_app.rs +
+mock_worker.rs harness)."""
+import asyncio
+import json
+
+import pytest
+from aiohttp.test_utils import TestClient, TestServer
+
+from smg_amd.config import PolicyConfig, RouterConfig
+from smg_amd.mock.engine import MockWorkerEngine, SimConfig
+from smg_amd.routers.factory import RouterManager
+from smg_amd.server.app import build_app
+from smg_amd.server.app_context import AppContext
+from smg_amd.workers.worker import Worker, WorkerType
+
+
+def make_ctx(n_workers=2, policy="round_robin", speedup=50.0, auth_key=None, **sim_kw):
+    cfg = RouterConfig(policy=PolicyConfig(name=policy, gpu_tree=False))
+    cfg.health_check.disable = True
+    if auth_key:
+        cfg.auth.api_key = auth_key
+    ctx = AppContext(cfg)
+    engines = []
+    for i in range(n_workers):
+        engine = MockWorkerEngine(SimConfig(speedup=speedup, model_id="mock-model", **sim_kw))
+        w = Worker(f"sim://worker-{i}", model_id="mock-model")
+        w.extra["engine"] = engine
+        ctx.worker_registry.register(w)
+        engines.append(engine)
+    ctx.router_manager = RouterManager(ctx, cfg)
+    return ctx, engines
+
+
+async def start_client(ctx, engines):
+    for e in engines:
+        await e.start()
+    app = build_app(ctx)
+    client = TestClient(TestServer(app))
+    await client.start_server()
+    return client
+
+
+async def stop_all(client, engines):
+    await client.close()
+    for e in engines:
+        await e.stop()
+
+
+CHAT_BODY = {
+    "model": "mock-model",
+    "messages": [{"role": "user", "content": "hello there, what is the answer?"}],
+    "max_tokens": 4,
+}
+
+
+def test_health_endpoints(runner):
+    async def run():
+        ctx, engines = make_ctx()
+        client = await start_client(ctx, engines)
+        try:
+            for path in ("/liveness", "/health", "/readiness"):
+                resp = await client.get(path)
+                assert resp.status == 200, path
+            resp = await client.get("/v1/models")
+            data = await resp.json()
+            assert data["data"][0]["id"] == "mock-model"
+            resp = await client.get("/get_server_info")
+            info = await resp.json()
+            assert len(info["workers"]) == 2
+        finally:
+            await stop_all(client, engines)
+
+    runner(run())
+
+
+def test_chat_completion_unary(runner):
+    async def run():
+        ctx, engines = make_ctx()
+        client = await start_client(ctx, engines)
+        try:
+            resp = await client.post("/v1/chat/completions", json=CHAT_BODY)
+            assert resp.status == 200
+            data = await resp.json()
+            assert data["object"] == "chat.completion"
+            assert data["choices"][0]["message"]["content"]
+            assert data["usage"]["completion_tokens"] == 4
+        finally:
+            await stop_all(client, engines)
+
+    runner(run())
+
+
+def test_chat_completion_streaming(runner):
+    async def run():
+        ctx, engines = make_ctx()
+        client = await start_client(ctx, engines)
+        try:
+            body = dict(CHAT_BODY, stream=True)
+            resp = await client.post("/v1/chat/completions", json=body)
+            assert resp.status == 200
+            assert "text/event-stream" in resp.headers["content-type"]
+            chunks = []
+            async for line in resp.content:
+                line = line.decode().strip()
+                if line.startswith("data: ") and line != "data: [DONE]":
+                    chunks.append(json.loads(line[6:]))
+                elif line == "data: [DONE]":
+                    break
+            assert len(chunks) >= 4
+            assert chunks[0]["object"] == "chat.completion.chunk"
+        finally:
+            await stop_all(client, engines)
+
+    runner(run())
+
+
+def test_round_robin_distributes(runner):
+    async def run():
+        ctx, engines = make_ctx(n_workers=2)
+        client = await start_client(ctx, engines)
+        try:
+            for _ in range(6):
+                resp = await client.post("/v1/completions", json={"model": "mock-model", "prompt": "hi", "max_tokens": 1})
+                assert resp.status == 200
+            done = [w.processed_requests for w in ctx.worker_registry.all()]
+            assert done == [3, 3]
+        finally:
+            await stop_all(client, engines)
+
+    runner(run())
+
+
+def test_cache_aware_e2e(runner):
+    async def run():
+        ctx, engines = make_ctx(n_workers=2, policy="cache_aware")
+        client = await start_client(ctx, engines)
+        try:
+            prompt = "a very long shared prefix used for cache routing " * 10
+            picks = []
+            for i in range(4):
+                resp = await client.post(
+                    "/v1/completions",
+                    json={"model": "mock-model", "prompt": prompt, "max_tokens": 1},
+                )
+                assert resp.status == 200
+            loads = [w.processed_requests for w in ctx.worker_registry.all()]
+            assert max(loads) == 4  # all to the cached worker
+        finally:
+            await stop_all(client, engines)
+
+    runner(run())
+
+
+def test_worker_crud(runner):
+    async def run():
+        ctx, engines = make_ctx(n_workers=1)
+        client = await start_client(ctx, engines)
+        try:
+            resp = await client.post("/workers", json={"url": "http://new-worker:9000", "model_id": "other"})
+            assert resp.status == 201
+            resp = await client.get("/workers")
+            data = await resp.json()
+            assert len(data["workers"]) == 2
+            resp = await client.delete("/workers/http://new-worker:9000")
+            assert resp.status == 200
+            resp = await client.get("/workers")
+            assert len((await resp.json())["workers"]) == 1
+        finally:
+            await stop_all(client, engines)
+
+    runner(run())
+
+
+def test_auth_rejects_bad_key(runner):
+    async def run():
+        ctx, engines = make_ctx(auth_key="secret-key")
+        client = await start_client(ctx, engines)
+        try:
+            resp = await client.post("/v1/chat/completions", json=CHAT_BODY)
+            assert resp.status == 401
+            resp = await client.post(
+                "/v1/chat/completions", json=CHAT_BODY, headers={"authorization": "Bearer secret-key"}
+            )
+            assert resp.status == 200
+            resp = await client.get("/health")  # public
+            assert resp.status == 200
+        finally:
+            await stop_all(client, engines)
+
+    runner(run())
+
+
+def test_failure_retries_to_other_worker(runner):
+    async def run():
+        ctx, engines = make_ctx(n_workers=2)
+        engines[0].config.failure_rate = 1.0  # worker 0 always 500s
+        ctx.config.retry.initial_backoff_ms = 1
+        client = await start_client(ctx, engines)
+        try:
+            oks = 0
+            for _ in range(4):
+                resp = await client.post("/v1/completions", json={"model": "mock-model", "prompt": "x", "max_tokens": 1})
+                oks += resp.status == 200
+            assert oks == 4  # retry moved them to the healthy worker
+        finally:
+            await stop_all(client, engines)
+
+    runner(run())
+
+
+def test_circuit_breaker_opens(runner):
+    async def run():
+        ctx, engines = make_ctx(n_workers=2)
+        ctx.config.circuit_breaker.failure_threshold = 2
+        engines[0].config.failure_rate = 1.0
+        for w in ctx.worker_registry.all():
+            w.circuit_breaker.config.failure_threshold = 2
+        ctx.config.retry.initial_backoff_ms = 1
+        client = await start_client(ctx, engines)
+        try:
+            for _ in range(6):
+                await client.post("/v1/completions", json={"model": "mock-model", "prompt": "x", "max_tokens": 1})
+            bad = ctx.worker_registry.get_by_url("sim://worker-0")
+            assert bad.circuit_breaker.state.value == "open"
+        finally:
+            await stop_all(client, engines)
+
+    runner(run())
+
+
+def test_invalid_request_400(runner):
+    async def run():
+        ctx, engines = make_ctx()
+        client = await start_client(ctx, engines)
+        try:
+            resp = await client.post("/v1/chat/completions", json={"model": "m", "messages": []})
+            assert resp.status == 400
+            err = await resp.json()
+            assert err["error"]["type"] == "invalid_request_error"
+        finally:
+            await stop_all(client, engines)
+
+    runner(run())
+
+
+def test_no_worker_503(runner):
+    async def run():
+        ctx, engines = make_ctx(n_workers=0)
+        ctx.config.retry.disable = True
+        client = await start_client(ctx, engines)
+        try:
+            resp = await client.post("/v1/chat/completions", json=CHAT_BODY)
+            assert resp.status == 503
+        finally:
+            await stop_all(client, engines)
+
+    runner(run())
+
+
+def test_metrics_exported(runner):
+    async def run():
+        ctx, engines = make_ctx()
+        client = await start_client(ctx, engines)
+        try:
+            await client.post("/v1/chat/completions", json=CHAT_BODY)
+            resp = await client.get("/metrics")
+            text = await resp.text()
+            assert "smg_http_requests_total" in text
+        finally:
+            await stop_all(client, engines)
+
+    runner(run())
