@@ -1,0 +1,165 @@
+// pybind11 module smg_amd._core — host C++ tree + gfx950 GPU tree + hashing.
+//
+// Deliberately torch-free: device pointers/streams cross the boundary as
+// integers, so the .so builds with plain hipcc and loads in any process.
+#include <pybind11/numpy.h>
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include <cstdint>
+#include <stdexcept>
+#include <vector>
+
+#include "host_tree.cpp"
+
+namespace py = pybind11;
+
+// ---- GPU tree C API (gpu_tree.hip) ---------------------------------------
+extern "C" {
+void* smg_gpu_tree_create(int device, uint32_t node_cap, uint32_t table_size, uint32_t page_size,
+                          uint32_t max_pages, uint32_t max_batch_reqs, uint32_t max_batch_tokens);
+void smg_gpu_tree_destroy(void* p);
+int smg_gpu_tree_run(void* p, int n_reqs, unsigned long long healthy_mask, const int* loads,
+                     const int* processed, int n_workers, float cache_threshold, int imbalanced,
+                     int do_insert, int forced_tenant, int* out_selected, uint32_t* out_matched,
+                     uint32_t* out_tenant);
+void smg_gpu_tree_staging(void* p, uint32_t** tokens, uint32_t** offsets);
+int smg_gpu_tree_remove_tenant(void* p, int slot);
+int smg_gpu_tree_evict_older(void* p, uint32_t cutoff);
+int smg_gpu_tree_stats(void* p, unsigned long long* out);
+int smg_gpu_tree_clear(void* p);
+int smg_hip_device_count();
+}
+
+namespace {
+
+class PyHostTree {
+   public:
+    explicit PyHostTree(uint32_t page_size) : tree_(page_size) {}
+
+    py::tuple match(py::array_t<uint32_t, py::array::c_style | py::array::forcecast> toks, bool touch) {
+        auto r = tree_.match(toks.data(), (uint32_t)toks.size(), touch);
+        return py::make_tuple(r.tenant, r.matched, r.input);
+    }
+    uint32_t insert(py::array_t<uint32_t, py::array::c_style | py::array::forcecast> toks, int tenant) {
+        return tree_.insert(toks.data(), (uint32_t)toks.size(), tenant);
+    }
+    void remove_tenant(int t) { tree_.remove_tenant(t); }
+    uint32_t evict(size_t max_nodes) { return tree_.evict(max_nodes); }
+    void clear() { tree_.clear(); }
+    size_t size() const { return tree_.live_count(); }
+    uint64_t tenant_tokens(int t) { return tree_.tenant_tokens(t); }
+    uint32_t page_size() const { return tree_.page_size(); }
+
+   private:
+    smg::HostTree tree_;
+};
+
+class PyGpuTree {
+   public:
+    PyGpuTree(int device, uint32_t node_cap, uint32_t table_size, uint32_t page_size,
+              uint32_t max_pages, uint32_t max_batch_reqs, uint32_t max_batch_tokens)
+        : max_batch_reqs_(max_batch_reqs), max_batch_tokens_(max_batch_tokens) {
+        h_ = smg_gpu_tree_create(device, node_cap, table_size, page_size, max_pages, max_batch_reqs,
+                                 max_batch_tokens);
+        if (!h_) throw std::runtime_error("smg_gpu_tree_create failed (no GPU / OOM)");
+    }
+    ~PyGpuTree() { smg_gpu_tree_destroy(h_); }
+
+    // Batched match/decide/insert.  `tokens_flat` int32/uint32, `offsets`
+    // int32 (n+1).  Returns (selected[int32], matched[uint32], tenant[uint32]).
+    py::tuple run(py::array_t<uint32_t, py::array::c_style | py::array::forcecast> tokens_flat,
+                  py::array_t<uint32_t, py::array::c_style | py::array::forcecast> offsets,
+                  unsigned long long healthy_mask, std::vector<int> loads, std::vector<int> processed,
+                  int n_workers, float cache_threshold, bool imbalanced, bool do_insert,
+                  int forced_tenant) {
+        int n_reqs = (int)offsets.size() - 1;
+        if (n_reqs < 1 || (uint32_t)n_reqs > max_batch_reqs_)
+            throw std::runtime_error("batch size out of range");
+        if ((uint32_t)tokens_flat.size() > max_batch_tokens_)
+            throw std::runtime_error("batch token count out of range");
+        uint32_t *stg_tokens, *stg_offsets;
+        smg_gpu_tree_staging(h_, &stg_tokens, &stg_offsets);
+        std::memcpy(stg_tokens, tokens_flat.data(), tokens_flat.size() * 4);
+        std::memcpy(stg_offsets, offsets.data(), ((size_t)n_reqs + 1) * 4);
+        loads.resize(64, 0);
+        processed.resize(64, 0);
+        py::array_t<int> sel(n_reqs);
+        py::array_t<uint32_t> matched(n_reqs);
+        py::array_t<uint32_t> tenant(n_reqs);
+        int rc;
+        {
+            py::gil_scoped_release nogil;
+            rc = smg_gpu_tree_run(h_, n_reqs, healthy_mask, loads.data(), processed.data(),
+                                  n_workers, cache_threshold, imbalanced ? 1 : 0, do_insert ? 1 : 0,
+                                  forced_tenant, sel.mutable_data(), matched.mutable_data(),
+                                  tenant.mutable_data());
+        }
+        if (rc != 0) throw std::runtime_error("smg_gpu_tree_run failed rc=" + std::to_string(rc));
+        return py::make_tuple(sel, matched, tenant);
+    }
+
+    void remove_tenant(int slot) {
+        if (smg_gpu_tree_remove_tenant(h_, slot)) throw std::runtime_error("remove_tenant failed");
+    }
+    void evict_older(uint32_t cutoff) {
+        if (smg_gpu_tree_evict_older(h_, cutoff)) throw std::runtime_error("evict failed");
+    }
+    py::dict stats() {
+        unsigned long long out[67];
+        if (smg_gpu_tree_stats(h_, out)) throw std::runtime_error("stats failed");
+        py::dict d;
+        py::list per_tenant;
+        for (int i = 0; i < 64; ++i) per_tenant.append(out[i]);
+        d["tenant_nodes"] = per_tenant;
+        d["live_nodes"] = out[64];
+        d["allocated_nodes"] = out[65];
+        d["clock"] = out[66];
+        return d;
+    }
+    void clear() {
+        if (smg_gpu_tree_clear(h_)) throw std::runtime_error("clear failed");
+    }
+
+   private:
+    void* h_;
+    uint32_t max_batch_reqs_, max_batch_tokens_;
+};
+
+uint64_t py_page_hash(py::array_t<uint32_t, py::array::c_style | py::array::forcecast> toks) {
+    smg::HostTree t(1);
+    return t.page_hash(toks.data(), (uint32_t)toks.size());
+}
+
+}  // namespace
+
+PYBIND11_MODULE(_core, m) {
+    m.doc() = "smg_amd native core: host C++ radix tree + gfx950 GPU radix tree";
+    m.def("hip_device_count", &smg_hip_device_count);
+    m.def("page_hash", &py_page_hash);
+
+    py::class_<PyHostTree>(m, "HostTokenTree")
+        .def(py::init<uint32_t>(), py::arg("page_size") = 16)
+        .def("match", &PyHostTree::match, py::arg("tokens"), py::arg("touch") = true)
+        .def("insert", &PyHostTree::insert)
+        .def("remove_tenant", &PyHostTree::remove_tenant)
+        .def("evict", &PyHostTree::evict)
+        .def("clear", &PyHostTree::clear)
+        .def("tenant_tokens", &PyHostTree::tenant_tokens)
+        .def("page_size", &PyHostTree::page_size)
+        .def("__len__", &PyHostTree::size);
+
+    py::class_<PyGpuTree>(m, "GpuTree")
+        .def(py::init<int, uint32_t, uint32_t, uint32_t, uint32_t, uint32_t, uint32_t>(),
+             py::arg("device") = 0, py::arg("node_cap") = 1u << 22, py::arg("table_size") = 1u << 23,
+             py::arg("page_size") = 16, py::arg("max_pages") = 4096, py::arg("max_batch_reqs") = 4096,
+             py::arg("max_batch_tokens") = 1u << 22)
+        .def("run", &PyGpuTree::run, py::arg("tokens_flat"), py::arg("offsets"),
+             py::arg("healthy_mask"), py::arg("loads"), py::arg("processed"), py::arg("n_workers"),
+             py::arg("cache_threshold") = 0.3f, py::arg("imbalanced") = false,
+             py::arg("do_insert") = true, py::arg("forced_tenant") = -1)
+        .def("remove_tenant", &PyGpuTree::remove_tenant)
+        .def("evict_older", &PyGpuTree::evict_older)
+        .def("stats", &PyGpuTree::stats)
+        .def("clear", &PyGpuTree::clear);
+}

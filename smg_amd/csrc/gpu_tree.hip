@@ -1,0 +1,540 @@
+// GPU-resident paged radix prefix tree for cache-aware routing — gfx950 (MI355X).
+//
+// Re-designs the reference's host radix tree (crates/kv_index/src/token_tree.rs:303,
+// match_prefix_with_counts :620, match_and_insert :754) as a device-resident
+// structure sized for 288 GB HBM3E:
+//
+//   * nodes are one page (page_size tokens) each; the child relation is ONE
+//     global open-addressed hash table keyed by mix64(parent_id, page_hash)
+//     — a match walk is a chain of hash probes, no pointer chasing;
+//   * a request is serviced by one wave64: the page hash is a lane-parallel
+//     polynomial reduction, the probe checks 64 slots per step in parallel
+//     (ballot), and the tenant/min-load decision is a wave reduction;
+//   * no stored page tokens are verified: the 64-bit keyed identity makes a
+//     false match ~2^-64 per pair, and a radix-tree hit is a routing HINT —
+//     a collision costs one sub-optimally routed request, never correctness.
+//     This removes every cross-CU plain-memory read; all shared state
+//     (table keys/vals, tenant bitmasks, LRU stamps, load counters) is
+//     accessed with agent-scope atomics which bypass the non-coherent L1s
+//     (MI355X_MICROARCH.md §Workgroup dispatch: atomics/sc1 are L2-served).
+//   * per-(node, tenant) LRU stamps from a device logical clock drive the
+//     eviction sweep; tenants are worker slots 0..63 (bitmask u64).
+//
+// Kernels: match-decide-insert (the whole cache_aware decision per batch in
+// one launch), match-only, insert-only, tenant removal, LRU sweep, stats.
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+#include <cstdio>
+
+#define WAVE 64
+#define EMPTY_KEY 0ull
+#define TOMBSTONE_KEY 1ull
+
+// ---------------------------------------------------------------------------
+// device-side structure (all pointers into one device allocation)
+// ---------------------------------------------------------------------------
+struct GpuTreeDev {
+    // hash table: capacity `table_size` (power of two)
+    unsigned long long* table_keys;  // 0 empty, 1 tombstone, else key
+    uint32_t* table_vals;            // node id
+    // node pool: capacity `node_cap`
+    unsigned long long* node_tenants;  // bitmask of worker slots
+    uint32_t* node_ts;                 // [node_cap * 64] per-tenant LRU stamp
+    uint32_t* node_parent;             // parent node id (eviction bookkeeping)
+    uint32_t* node_slot;               // this node's table slot (for removal)
+    // counters
+    uint32_t* next_node;  // node allocator (node 0 = root, never allocated)
+    uint32_t* clock_;     // logical LRU clock
+    uint32_t node_cap;
+    uint32_t table_mask;  // table_size - 1
+    uint32_t page_size;
+    uint32_t max_pages;   // walk depth cap
+};
+
+__device__ __forceinline__ unsigned long long mix64(unsigned long long x) {
+    // splitmix64 finalizer
+    x += 0x9E3779B97F4A7C15ull;
+    x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
+    x = (x ^ (x >> 27)) * 0x94D049BB133111EBull;
+    return x ^ (x >> 31);
+}
+
+__device__ __forceinline__ unsigned long long child_key(uint32_t parent, unsigned long long page_hash) {
+    unsigned long long k = mix64(((unsigned long long)parent << 32) ^ page_hash * 0x9E3779B97F4A7C15ull);
+    // reserve 0/1
+    return k < 2 ? k + 2 : k;
+}
+
+// Order-sensitive page hash: lane i contributes mix64(token_i) * PRIME^i,
+// wave-summed.  POW table in constant memory, page_size <= 64.
+__constant__ unsigned long long c_pow[WAVE];
+
+__device__ __forceinline__ unsigned long long wave_sum_u64(unsigned long long v) {
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, WAVE);
+    return v;
+}
+
+__device__ __forceinline__ unsigned long long page_hash_wave(const uint32_t* toks, int n, int lane) {
+    unsigned long long contrib = 0;
+    if (lane < n) contrib = mix64((unsigned long long)toks[lane] + 0x5851F42D4C957F2Dull) * c_pow[lane];
+    return mix64(wave_sum_u64(contrib) ^ (unsigned long long)n);
+}
+
+__device__ __forceinline__ unsigned long long atomic_load_key(const unsigned long long* p) {
+    return __hip_atomic_load(p, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+}
+
+// Probe the table for `key`.  64 lanes scan 64 slots per step.  Returns the
+// slot index if found, or ~0u if hit an empty slot (absent), scanning at most
+// 16 * 64 slots.  `out_val` = node id when found.
+__device__ int probe_find(const GpuTreeDev& T, unsigned long long key, uint32_t* out_val, int lane) {
+    uint32_t base = (uint32_t)(key & T.table_mask);
+    for (int step = 0; step < 16; ++step) {
+        uint32_t slot = (base + step * WAVE + lane) & T.table_mask;
+        unsigned long long k = atomic_load_key(&T.table_keys[slot]);
+        unsigned long long found = __ballot(k == key);
+        if (found) {
+            int src = __ffsll((long long)found) - 1;
+            uint32_t fslot = (base + step * WAVE + src) & T.table_mask;
+            *out_val = __hip_atomic_load(&T.table_vals[fslot], __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+            return (int)fslot;
+        }
+        unsigned long long empt = __ballot(k == EMPTY_KEY);
+        if (empt) {
+            // key would have been placed at/before the first empty slot
+            return -1;
+        }
+    }
+    return -1;  // table pathologically full in this neighborhood
+}
+
+// Insert `key`->`val` (lane-0 driven).  Two-step publication so a losing
+// wave can never clobber the winner's val: CAS the key slot EMPTY->BUSY,
+// store val, then release-store the real key.  A racing same-key insert that
+// sees BUSY simply probes on and may create an unreachable duplicate node —
+// bounded memory noise, never a wrong lookup.  Returns the node id placed in
+// the slot (ours, or an existing same-key entry's), -1 when the table
+// neighborhood is full.
+#define BUSY_KEY 2ull
+__device__ int probe_insert(const GpuTreeDev& T, unsigned long long key, uint32_t val, uint32_t* out_slot) {
+    uint32_t base = (uint32_t)(key & T.table_mask);
+    for (uint32_t i = 0; i < 1024u; ++i) {
+        uint32_t slot = (base + i) & T.table_mask;
+        unsigned long long k =
+            __hip_atomic_load(&T.table_keys[slot], __ATOMIC_ACQUIRE, __HIP_MEMORY_SCOPE_AGENT);
+        if (k == key) {
+            *out_slot = slot;
+            return (int)__hip_atomic_load(&T.table_vals[slot], __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        }
+        if (k == EMPTY_KEY || k == TOMBSTONE_KEY) {
+            unsigned long long expected = k;
+            if (__hip_atomic_compare_exchange_strong(&T.table_keys[slot], &expected, BUSY_KEY,
+                                                     __ATOMIC_ACQ_REL, __ATOMIC_RELAXED,
+                                                     __HIP_MEMORY_SCOPE_AGENT)) {
+                __hip_atomic_store(&T.table_vals[slot], val, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                __hip_atomic_store(&T.table_keys[slot], key, __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_AGENT);
+                *out_slot = slot;
+                return (int)val;
+            }
+            // lost the reservation; fall through and keep probing
+        }
+        // BUSY or other key: probe on
+    }
+    return -1;
+}
+
+// ---------------------------------------------------------------------------
+// match + decide + insert, one wave per request
+// ---------------------------------------------------------------------------
+struct BatchArgs {
+    const uint32_t* tokens;   // flattened
+    const uint32_t* offsets;  // n_reqs + 1
+    int n_reqs;
+    // decision inputs
+    unsigned long long healthy_mask;  // candidate worker slots
+    int* loads;                       // [64] live load counters (device, atomically bumped)
+    const int* processed;             // [64] processed_requests tie-break
+    float cache_threshold;
+    int imbalanced;  // batch-level trigger: force min-load
+    int n_workers;
+    int do_insert;
+    int forced_tenant;  // >=0: skip decision, insert for this slot (insert-only)
+    // outputs per request
+    int* out_selected;       // worker slot (or -1)
+    uint32_t* out_matched;   // matched token count
+    uint32_t* out_tenant;    // matched (pre-decision) tenant slot or 0xffffffff
+};
+
+extern "C" __global__ void __launch_bounds__(WAVE)
+smg_tree_match_insert(GpuTreeDev T, BatchArgs A) {
+    int req = blockIdx.x;
+    if (req >= A.n_reqs) return;
+    int lane = threadIdx.x;
+
+    uint32_t beg = A.offsets[req], end = A.offsets[req + 1];
+    uint32_t n_tokens = end - beg;
+    uint32_t n_pages = n_tokens / T.page_size;
+    if (n_pages > T.max_pages) n_pages = T.max_pages;
+
+    // ---- phase A: match walk ------------------------------------------------
+    uint32_t cur = 0;  // root
+    uint32_t matched_pages = 0;
+    uint32_t deepest_tenanted = 0xffffffffu;
+    unsigned long long deepest_mask = 0;
+    for (uint32_t p = 0; p < n_pages; ++p) {
+        unsigned long long h = page_hash_wave(A.tokens + beg + p * T.page_size,
+                                              (int)T.page_size, lane);
+        unsigned long long key = child_key(cur, h);
+        uint32_t nid;
+        int slot = probe_find(T, key, &nid, lane);
+        if (slot < 0) break;
+        unsigned long long tenants =
+            __hip_atomic_load(&T.node_tenants[nid], __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        if (tenants == 0ull) break;  // fully evicted: prefix not cached anywhere
+        matched_pages = p + 1;
+        deepest_tenanted = nid;
+        deepest_mask = tenants;
+        cur = nid;
+    }
+    uint32_t matched_tokens = matched_pages * T.page_size;
+
+    // MRU healthy tenant of the deepest matched node (lane-parallel argmax ts)
+    int matched_tenant = -1;
+    if (deepest_tenanted != 0xffffffffu) {
+        unsigned long long m = deepest_mask & A.healthy_mask;
+        uint32_t my_ts = 0;
+        int my_slot = -1;
+        if (lane < 64 && ((m >> lane) & 1ull)) {
+            my_ts = __hip_atomic_load(&T.node_ts[(size_t)deepest_tenanted * WAVE + lane],
+                                      __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+            my_slot = lane;
+        }
+        // wave argmax (ts, slot)
+        unsigned long long packed = ((unsigned long long)my_ts << 32) | (unsigned)(my_slot + 1);
+        if (my_slot < 0) packed = 0;
+#pragma unroll
+        for (int off = 32; off > 0; off >>= 1) {
+            unsigned long long other = __shfl_xor(packed, off, WAVE);
+            if (other > packed) packed = other;
+        }
+        if ((packed & 0xffffffffu) != 0) matched_tenant = (int)(packed & 0xffffffffu) - 1;
+        // touch the matched tenant's stamp along the deepest node (mirrors
+        // token_tree.rs:693 touch_tenant on match)
+        if (lane == 0 && matched_tenant >= 0) {
+            uint32_t now = atomicAdd(T.clock_, 1u) + 1u;
+            atomicMax(&T.node_ts[(size_t)deepest_tenanted * WAVE + matched_tenant], now);
+        }
+    }
+
+    // ---- decision (cache_aware.rs:986 semantics) ---------------------------
+    int selected;
+    if (A.forced_tenant >= 0) {
+        selected = A.forced_tenant;
+    } else {
+        float rate = n_tokens ? (float)matched_tokens / (float)n_tokens : 0.f;
+        bool hit = !A.imbalanced && rate > A.cache_threshold && matched_tenant >= 0;
+        if (hit) {
+            selected = matched_tenant;
+        } else {
+            // min-load over healthy slots, tie-break (load, processed, slot)
+            int my_slot = -1;
+            long long my_key = 0x7fffffffffffffffll;
+            if (lane < A.n_workers && ((A.healthy_mask >> lane) & 1ull)) {
+                int ld = __hip_atomic_load(&A.loads[lane], __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                int pr = A.processed[lane];
+                my_key = ((long long)ld << 40) | ((long long)(pr & 0xffffff) << 16) | lane;
+                my_slot = lane;
+            }
+#pragma unroll
+            for (int off = 32; off > 0; off >>= 1) {
+                long long ok = __shfl_xor(my_key, off, WAVE);
+                int os = __shfl_xor(my_slot, off, WAVE);
+                if (ok < my_key) { my_key = ok; my_slot = os; }
+            }
+            selected = my_slot;
+        }
+        if (lane == 0 && selected >= 0)
+            atomicAdd(&A.loads[selected], 1);  // intra-batch load guard
+    }
+
+    if (lane == 0) {
+        A.out_selected[req] = selected;
+        A.out_matched[req] = matched_tokens;
+        A.out_tenant[req] = (matched_tenant >= 0) ? (uint32_t)matched_tenant : 0xffffffffu;
+    }
+    if (!A.do_insert || selected < 0) return;
+
+    // ---- phase B: insert path for `selected` --------------------------------
+    uint32_t now = 0;
+    if (lane == 0) now = atomicAdd(T.clock_, 1u) + 1u;
+    now = __shfl(now, 0, WAVE);
+    unsigned long long tbit = 1ull << selected;
+    cur = 0;
+    for (uint32_t p = 0; p < n_pages; ++p) {
+        unsigned long long h = page_hash_wave(A.tokens + beg + p * T.page_size,
+                                              (int)T.page_size, lane);
+        unsigned long long key = child_key(cur, h);
+        uint32_t nid;
+        int slot = probe_find(T, key, &nid, lane);  // slot uniform across the wave
+        int winner;
+        if (slot >= 0) {
+            winner = (int)nid;  // all lanes hold nid from probe_find
+        } else {
+            winner = -1;
+            if (lane == 0) {
+                uint32_t fresh = atomicAdd(T.next_node, 1u);
+                if (fresh < T.node_cap) {
+                    T.node_parent[fresh] = cur;
+                    uint32_t oslot;
+                    winner = probe_insert(T, key, fresh, &oslot);
+                    if (winner == (int)fresh) T.node_slot[fresh] = oslot;
+                }
+            }
+            winner = __shfl(winner, 0, WAVE);
+        }
+        if (winner < 0) return;  // pool exhausted: stop attributing
+        if (lane == 0) {
+            atomicOr(&T.node_tenants[winner], tbit);
+            atomicMax(&T.node_ts[(size_t)winner * WAVE + selected], now);
+        }
+        cur = (uint32_t)winner;
+    }
+}
+
+// ---------------------------------------------------------------------------
+// maintenance kernels
+// ---------------------------------------------------------------------------
+extern "C" __global__ void smg_tree_remove_tenant(GpuTreeDev T, int slot_idx) {
+    uint32_t n = *T.next_node;
+    unsigned long long mask = ~(1ull << slot_idx);
+    for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n; i += gridDim.x * blockDim.x) {
+        atomicAnd(&T.node_tenants[i], mask);
+        T.node_ts[(size_t)i * WAVE + slot_idx] = 0;
+    }
+}
+
+// Clear tenant bits whose stamp is older than `cutoff`; nodes left with no
+// tenants block matches through them (eviction semantics of token_tree.rs).
+extern "C" __global__ void smg_tree_evict_older(GpuTreeDev T, uint32_t cutoff) {
+    uint32_t n = *T.next_node;
+    uint32_t node = blockIdx.x;  // one wave per node, lane per tenant slot
+    int lane = threadIdx.x;
+    for (uint32_t i = node; i < n; i += gridDim.x) {
+        unsigned long long tenants =
+            __hip_atomic_load(&T.node_tenants[i], __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        if (!tenants) continue;
+        if (((tenants >> lane) & 1ull) && T.node_ts[(size_t)i * WAVE + lane] < cutoff) {
+            atomicAnd(&T.node_tenants[i], ~(1ull << lane));
+        }
+    }
+}
+
+// Count live nodes / per-tenant attributed nodes (stats + tie-breaks).
+extern "C" __global__ void smg_tree_stats(GpuTreeDev T, unsigned long long* out_counts /*65*/) {
+    uint32_t n = *T.next_node;
+    int lane = threadIdx.x;
+    unsigned long long live = 0, mine = 0;
+    // grid-stride over nodes; one wave per node, lane = tenant slot
+    for (uint32_t i = blockIdx.x; i < n; i += gridDim.x) {
+        unsigned long long tenants =
+            __hip_atomic_load(&T.node_tenants[i], __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        if (lane == 0 && tenants) live++;
+        if ((tenants >> lane) & 1ull) mine++;
+    }
+    if (lane == 0 && live) atomicAdd(&out_counts[64], live);
+    if (mine) atomicAdd(&out_counts[lane], mine);
+}
+
+// ---------------------------------------------------------------------------
+// host-side management (exported C API; pybind in bindings.cpp)
+// ---------------------------------------------------------------------------
+#define HIP_CHECK(x)                                                                   \
+    do {                                                                               \
+        hipError_t _e = (x);                                                           \
+        if (_e != hipSuccess) {                                                        \
+            fprintf(stderr, "HIP error %s at %s:%d\n", hipGetErrorString(_e), __FILE__, __LINE__); \
+            return nullptr;                                                            \
+        }                                                                              \
+    } while (0)
+
+struct GpuTreeHost {
+    GpuTreeDev dev;
+    hipStream_t stream;
+    uint32_t table_size;
+    // pinned staging
+    uint32_t* h_tokens;
+    uint32_t* h_offsets;
+    int* h_selected;
+    uint32_t* h_matched;
+    uint32_t* h_tenant;
+    uint32_t max_batch_tokens;
+    uint32_t max_batch_reqs;
+    // device batch buffers
+    uint32_t* d_tokens;
+    uint32_t* d_offsets;
+    int* d_selected;
+    uint32_t* d_matched;
+    uint32_t* d_tenant;
+    int* d_loads;
+    int* d_processed;
+    unsigned long long* d_counts;
+};
+
+extern "C" void* smg_gpu_tree_create(int device, uint32_t node_cap, uint32_t table_size,
+                                     uint32_t page_size, uint32_t max_pages,
+                                     uint32_t max_batch_reqs, uint32_t max_batch_tokens) {
+    if (hipSetDevice(device) != hipSuccess) return nullptr;
+    GpuTreeHost* t = new GpuTreeHost{};
+    t->table_size = table_size;
+    t->max_batch_reqs = max_batch_reqs;
+    t->max_batch_tokens = max_batch_tokens;
+    GpuTreeDev& D = t->dev;
+    D.node_cap = node_cap;
+    D.table_mask = table_size - 1;
+    D.page_size = page_size;
+    D.max_pages = max_pages;
+    HIP_CHECK(hipStreamCreate(&t->stream));
+    HIP_CHECK(hipMalloc(&D.table_keys, sizeof(unsigned long long) * table_size));
+    HIP_CHECK(hipMalloc(&D.table_vals, sizeof(uint32_t) * table_size));
+    HIP_CHECK(hipMalloc(&D.node_tenants, sizeof(unsigned long long) * node_cap));
+    HIP_CHECK(hipMalloc(&D.node_ts, sizeof(uint32_t) * (size_t)node_cap * WAVE));
+    HIP_CHECK(hipMalloc(&D.node_parent, sizeof(uint32_t) * node_cap));
+    HIP_CHECK(hipMalloc(&D.node_slot, sizeof(uint32_t) * node_cap));
+    HIP_CHECK(hipMalloc(&D.next_node, sizeof(uint32_t) * 2));
+    D.clock_ = D.next_node + 1;
+    HIP_CHECK(hipMemset(D.table_keys, 0, sizeof(unsigned long long) * table_size));
+    HIP_CHECK(hipMemset(D.node_tenants, 0, sizeof(unsigned long long) * node_cap));
+    HIP_CHECK(hipMemset(D.node_ts, 0, sizeof(uint32_t) * (size_t)node_cap * WAVE));
+    uint32_t init[2] = {1u, 0u};  // node 0 = root
+    HIP_CHECK(hipMemcpy(D.next_node, init, sizeof(init), hipMemcpyHostToDevice));
+    // polynomial powers for the page hash
+    unsigned long long pows[WAVE];
+    pows[0] = 1ull;
+    for (int i = 1; i < WAVE; ++i) pows[i] = pows[i - 1] * 0x100000001B3ull;
+    HIP_CHECK(hipMemcpyToSymbol(HIP_SYMBOL(c_pow), pows, sizeof(pows)));
+    // batch buffers
+    HIP_CHECK(hipHostMalloc((void**)&t->h_tokens, sizeof(uint32_t) * max_batch_tokens));
+    HIP_CHECK(hipHostMalloc((void**)&t->h_offsets, sizeof(uint32_t) * (max_batch_reqs + 1)));
+    HIP_CHECK(hipHostMalloc((void**)&t->h_selected, sizeof(int) * max_batch_reqs));
+    HIP_CHECK(hipHostMalloc((void**)&t->h_matched, sizeof(uint32_t) * max_batch_reqs));
+    HIP_CHECK(hipHostMalloc((void**)&t->h_tenant, sizeof(uint32_t) * max_batch_reqs));
+    HIP_CHECK(hipMalloc(&t->d_tokens, sizeof(uint32_t) * max_batch_tokens));
+    HIP_CHECK(hipMalloc(&t->d_offsets, sizeof(uint32_t) * (max_batch_reqs + 1)));
+    HIP_CHECK(hipMalloc(&t->d_selected, sizeof(int) * max_batch_reqs));
+    HIP_CHECK(hipMalloc(&t->d_matched, sizeof(uint32_t) * max_batch_reqs));
+    HIP_CHECK(hipMalloc(&t->d_tenant, sizeof(uint32_t) * max_batch_reqs));
+    HIP_CHECK(hipMalloc(&t->d_loads, sizeof(int) * WAVE));
+    HIP_CHECK(hipMalloc(&t->d_processed, sizeof(int) * WAVE));
+    HIP_CHECK(hipMalloc(&t->d_counts, sizeof(unsigned long long) * 65));
+    return t;
+}
+
+extern "C" void smg_gpu_tree_destroy(void* p) {
+    if (!p) return;
+    GpuTreeHost* t = (GpuTreeHost*)p;
+    hipFree(t->dev.table_keys); hipFree(t->dev.table_vals);
+    hipFree(t->dev.node_tenants); hipFree(t->dev.node_ts);
+    hipFree(t->dev.node_parent); hipFree(t->dev.node_slot);
+    hipFree(t->dev.next_node);
+    hipHostFree(t->h_tokens); hipHostFree(t->h_offsets); hipHostFree(t->h_selected);
+    hipHostFree(t->h_matched); hipHostFree(t->h_tenant);
+    hipFree(t->d_tokens); hipFree(t->d_offsets); hipFree(t->d_selected);
+    hipFree(t->d_matched); hipFree(t->d_tenant); hipFree(t->d_loads);
+    hipFree(t->d_processed); hipFree(t->d_counts);
+    hipStreamDestroy(t->stream);
+    delete t;
+}
+
+// Batched match/decide/insert.  tokens/offsets are caller-filled into the
+// pinned buffers via smg_gpu_tree_staging().  Returns 0 on success.
+extern "C" int smg_gpu_tree_run(void* p, int n_reqs, unsigned long long healthy_mask,
+                                const int* loads, const int* processed, int n_workers,
+                                float cache_threshold, int imbalanced, int do_insert,
+                                int forced_tenant,
+                                int* out_selected, uint32_t* out_matched, uint32_t* out_tenant) {
+    GpuTreeHost* t = (GpuTreeHost*)p;
+    if (n_reqs <= 0 || (uint32_t)n_reqs > t->max_batch_reqs) return -1;
+    uint32_t n_tokens = t->h_offsets[n_reqs];
+    if (n_tokens > t->max_batch_tokens) return -2;
+    hipStream_t s = t->stream;
+    hipMemcpyAsync(t->d_tokens, t->h_tokens, sizeof(uint32_t) * n_tokens, hipMemcpyHostToDevice, s);
+    hipMemcpyAsync(t->d_offsets, t->h_offsets, sizeof(uint32_t) * (n_reqs + 1), hipMemcpyHostToDevice, s);
+    hipMemcpyAsync(t->d_loads, loads, sizeof(int) * WAVE, hipMemcpyHostToDevice, s);
+    hipMemcpyAsync(t->d_processed, processed, sizeof(int) * WAVE, hipMemcpyHostToDevice, s);
+    BatchArgs A{};
+    A.tokens = t->d_tokens;
+    A.offsets = t->d_offsets;
+    A.n_reqs = n_reqs;
+    A.healthy_mask = healthy_mask;
+    A.loads = t->d_loads;
+    A.processed = t->d_processed;
+    A.cache_threshold = cache_threshold;
+    A.imbalanced = imbalanced;
+    A.n_workers = n_workers;
+    A.do_insert = do_insert;
+    A.forced_tenant = forced_tenant;
+    A.out_selected = t->d_selected;
+    A.out_matched = t->d_matched;
+    A.out_tenant = t->d_tenant;
+    hipLaunchKernelGGL(smg_tree_match_insert, dim3(n_reqs), dim3(WAVE), 0, s, t->dev, A);
+    hipMemcpyAsync(t->h_selected, t->d_selected, sizeof(int) * n_reqs, hipMemcpyDeviceToHost, s);
+    hipMemcpyAsync(t->h_matched, t->d_matched, sizeof(uint32_t) * n_reqs, hipMemcpyDeviceToHost, s);
+    hipMemcpyAsync(t->h_tenant, t->d_tenant, sizeof(uint32_t) * n_reqs, hipMemcpyDeviceToHost, s);
+    if (hipStreamSynchronize(s) != hipSuccess) return -3;
+    for (int i = 0; i < n_reqs; ++i) {
+        out_selected[i] = t->h_selected[i];
+        out_matched[i] = t->h_matched[i];
+        out_tenant[i] = t->h_tenant[i];
+    }
+    return 0;
+}
+
+extern "C" void smg_gpu_tree_staging(void* p, uint32_t** tokens, uint32_t** offsets) {
+    GpuTreeHost* t = (GpuTreeHost*)p;
+    *tokens = t->h_tokens;
+    *offsets = t->h_offsets;
+}
+
+extern "C" int smg_gpu_tree_remove_tenant(void* p, int slot) {
+    GpuTreeHost* t = (GpuTreeHost*)p;
+    hipLaunchKernelGGL(smg_tree_remove_tenant, dim3(256), dim3(256), 0, t->stream, t->dev, slot);
+    return hipStreamSynchronize(t->stream) == hipSuccess ? 0 : -1;
+}
+
+extern "C" int smg_gpu_tree_evict_older(void* p, uint32_t cutoff) {
+    GpuTreeHost* t = (GpuTreeHost*)p;
+    hipLaunchKernelGGL(smg_tree_evict_older, dim3(2048), dim3(WAVE), 0, t->stream, t->dev, cutoff);
+    return hipStreamSynchronize(t->stream) == hipSuccess ? 0 : -1;
+}
+
+// out[0..63] per-tenant node counts, out[64] live nodes, out[65] allocated, out[66] clock
+extern "C" int smg_gpu_tree_stats(void* p, unsigned long long* out /*67*/) {
+    GpuTreeHost* t = (GpuTreeHost*)p;
+    hipMemsetAsync(t->d_counts, 0, sizeof(unsigned long long) * 65, t->stream);
+    hipLaunchKernelGGL(smg_tree_stats, dim3(1024), dim3(WAVE), 0, t->stream, t->dev, t->d_counts);
+    hipMemcpyAsync(out, t->d_counts, sizeof(unsigned long long) * 65, hipMemcpyDeviceToHost, t->stream);
+    uint32_t counters[2];
+    hipMemcpyAsync(counters, t->dev.next_node, sizeof(counters), hipMemcpyDeviceToHost, t->stream);
+    if (hipStreamSynchronize(t->stream) != hipSuccess) return -1;
+    out[65] = counters[0];
+    out[66] = counters[1];
+    return 0;
+}
+
+extern "C" int smg_gpu_tree_clear(void* p) {
+    GpuTreeHost* t = (GpuTreeHost*)p;
+    hipMemsetAsync(t->dev.table_keys, 0, sizeof(unsigned long long) * t->table_size, t->stream);
+    hipMemsetAsync(t->dev.node_tenants, 0, sizeof(unsigned long long) * t->dev.node_cap, t->stream);
+    hipMemsetAsync(t->dev.node_ts, 0, sizeof(uint32_t) * (size_t)t->dev.node_cap * WAVE, t->stream);
+    uint32_t init[2] = {1u, 0u};
+    hipMemcpyAsync(t->dev.next_node, init, sizeof(init), hipMemcpyHostToDevice, t->stream);
+    return hipStreamSynchronize(t->stream) == hipSuccess ? 0 : -1;
+}
+
+extern "C" int smg_hip_device_count() {
+    int n = 0;
+    if (hipGetDeviceCount(&n) != hipSuccess) return 0;
+    return n;
+}

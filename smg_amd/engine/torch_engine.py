@@ -1,0 +1,294 @@
+"""GPU worker engine: a continuous-batching transformer decode engine on
+PyTorch-ROCm (bf16, random-init weights).
+
+Role: the MI355X stand-in for the real inference engines the gateway fronts
+(vLLM/SGLang in the reference).  The bench runs one TorchEngine
+per GPU rank behind the gateway's RCCL data plane, so the measured req/s is a
+full gateway+engine system on real hardware.  Tests use the CPU mock engine
+(mock/engine.py); this class is the GPU path and fails loudly without one.
+
+Design: slot-based continuous batching with a preallocated KV cache
+[slots, layers, 2, heads, max_seq, head_dim] in bf16, chunked prefill
+(one chunk per step alongside the batched single-token decode), greedy
+sampling.  The decode forward is shape-stable per active-slot-count, so HIP
+graphs can capture it (enable with graphs=True).
+"""
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass
+from typing import Dict, List, Optional
+
+import torch
+import torch.nn.functional as F
+
+
+@dataclass
+class TorchEngineConfig:
+    vocab_size: int = 32768
+    n_layers: int = 16
+    d_model: int = 2048
+    n_heads: int = 16
+    ffn_mult: float = 2.6875  # 5504/2048, llama-style
+    max_slots: int = 64
+    max_seq: int = 2048
+    prefill_chunk: int = 2048
+    dtype: str = "bfloat16"
+    seed: int = 1234
+
+    @classmethod
+    def tiny(cls) -> "TorchEngineConfig":
+        return cls(vocab_size=2048, n_layers=2, d_model=256, n_heads=4, max_slots=8, max_seq=256, prefill_chunk=256)
+
+    @classmethod
+    def bench_1b(cls) -> "TorchEngineConfig":
+        """The flagship bench model: ~1.1B params bf16 (≈2.3 GB weights)."""
+        return cls()
+
+    @property
+    def head_dim(self) -> int:
+        return self.d_model // self.n_heads
+
+    @property
+    def d_ffn(self) -> int:
+        return int(self.d_model * self.ffn_mult) // 64 * 64
+
+
+class _Layer:
+    def __init__(self, cfg: TorchEngineConfig, device, dtype, gen):
+        d, f = cfg.d_model, cfg.d_ffn
+        k = 1.0 / math.sqrt(d)
+
+        def w(*shape):
+            return (torch.rand(*shape, generator=gen, device=device, dtype=torch.float32) * 2 - 1).to(dtype) * k
+
+        self.wqkv = w(d, 3 * d)
+        self.wo = w(d, d)
+        self.w1 = w(d, f)
+        self.w3 = w(d, f)
+        self.w2 = w(f, d)
+        self.ln1 = torch.ones(d, device=device, dtype=dtype)
+        self.ln2 = torch.ones(d, device=device, dtype=dtype)
+
+
+def _rms(x, weight):
+    v = x.float()
+    v = v * torch.rsqrt(v.pow(2).mean(-1, keepdim=True) + 1e-5)
+    return (v * weight.float()).to(x.dtype)
+
+
+class _Request:
+    __slots__ = ("rid", "tokens", "max_new", "slot", "prefilled", "generated", "done")
+
+    def __init__(self, rid, tokens, max_new):
+        self.rid = rid
+        self.tokens = tokens
+        self.max_new = max_new
+        self.slot = -1
+        self.prefilled = 0
+        self.generated: List[int] = []
+        self.done = False
+
+
+class TorchEngine:
+    def __init__(self, cfg: Optional[TorchEngineConfig] = None, device: str = "cuda:0", graphs: bool = False):
+        if device.startswith("cuda") and not torch.cuda.is_available():
+            raise RuntimeError("TorchEngine requires a GPU (use mock/engine.py on CPU)")
+        self.cfg = cfg or TorchEngineConfig()
+        self.device = torch.device(device)
+        # cpu device is for logic tests only; compute dtype stays bf16 on GPU
+        self.dtype = getattr(torch, self.cfg.dtype) if device.startswith("cuda") else torch.float32
+        gen = torch.Generator(device=self.device).manual_seed(self.cfg.seed)
+        c = self.cfg
+        with torch.device(self.device):
+            self.embed = (torch.rand(c.vocab_size, c.d_model, generator=gen, device=self.device,
+                                     dtype=torch.float32) * 2 - 1).to(self.dtype) / math.sqrt(c.d_model)
+            self.layers = [_Layer(c, self.device, self.dtype, gen) for _ in range(c.n_layers)]
+            self.ln_f = torch.ones(c.d_model, device=self.device, dtype=self.dtype)
+            # KV cache: [layers, 2, slots, heads, max_seq, head_dim]
+            self.kv = torch.zeros(
+                c.n_layers, 2, c.max_slots, c.n_heads, c.max_seq, c.head_dim,
+                device=self.device, dtype=self.dtype,
+            )
+            # rotary tables
+            inv = 1.0 / (10000.0 ** (torch.arange(0, c.head_dim, 2, device=self.device).float() / c.head_dim))
+            t = torch.arange(c.max_seq, device=self.device).float()
+            freqs = torch.outer(t, inv)
+            self.cos = freqs.cos().to(self.dtype)
+            self.sin = freqs.sin().to(self.dtype)
+        self.seq_len = torch.zeros(c.max_slots, dtype=torch.long, device=self.device)
+        self._seq_len_host = [0] * c.max_slots
+        self._free_slots = list(range(c.max_slots - 1, -1, -1))
+        self.waiting: List[_Request] = []
+        self.running: Dict[int, _Request] = {}  # slot -> request
+        self._requests: Dict[str, _Request] = {}
+        self._rid_counter = 0
+        self.total_generated = 0
+        self.graphs = graphs
+        self._graph_cache: Dict[int, tuple] = {}
+
+    # ---- API -------------------------------------------------------------
+    def submit(self, tokens: List[int], max_new_tokens: int, rid: Optional[str] = None) -> str:
+        if rid is None:
+            self._rid_counter += 1
+            rid = f"req-{self._rid_counter}"
+        c = self.cfg
+        toks = [t % c.vocab_size for t in tokens][-(c.max_seq - max_new_tokens - 1):]
+        req = _Request(rid, toks, max_new_tokens)
+        self._requests[rid] = req
+        self.waiting.append(req)
+        return rid
+
+    def finished(self, rid: str) -> bool:
+        r = self._requests.get(rid)
+        return r is None or r.done
+
+    def collect(self, rid: str) -> List[int]:
+        r = self._requests.pop(rid, None)
+        return r.generated if r else []
+
+    def n_active(self) -> int:
+        return len(self.running) + len(self.waiting)
+
+    def load_snapshot(self) -> Dict:
+        c = self.cfg
+        used = sum(self._seq_len_host[s] for s in self.running)
+        return {
+            "num_queue_tokens": sum(len(r.tokens) for r in self.waiting),
+            "num_inflight_tokens": used,
+            "num_running_reqs": len(self.running),
+            "num_queue_reqs": len(self.waiting),
+            "token_usage": used / (c.max_slots * c.max_seq),
+            "gen_throughput": None,
+        }
+
+    # ---- engine step -----------------------------------------------------
+    def step(self) -> int:
+        """One engine iteration: admit + one prefill chunk + one decode token
+        for every running slot.  Returns tokens produced."""
+        c = self.cfg
+        # admission
+        while self.waiting and self._free_slots:
+            req = self.waiting.pop(0)
+            req.slot = self._free_slots.pop()
+            self.running[req.slot] = req
+            self._seq_len_host[req.slot] = 0
+
+        produced = 0
+        # prefill: one chunk for the first unprefilled request
+        for slot, req in self.running.items():
+            if req.prefilled < len(req.tokens):
+                chunk = req.tokens[req.prefilled: req.prefilled + c.prefill_chunk]
+                self._prefill(slot, chunk, req.prefilled)
+                req.prefilled += len(chunk)
+                break
+
+        # decode: all slots whose prompt is fully prefilled
+        decode_slots = [s for s, r in self.running.items() if r.prefilled >= len(r.tokens)]
+        if decode_slots:
+            last_tokens = []
+            for s in decode_slots:
+                r = self.running[s]
+                last_tokens.append(r.generated[-1] if r.generated else r.tokens[-1])
+            new_tokens = self._decode(decode_slots, last_tokens)
+            finished = []
+            for s, tok in zip(decode_slots, new_tokens):
+                r = self.running[s]
+                r.generated.append(int(tok))
+                produced += 1
+                self.total_generated += 1
+                if len(r.generated) >= r.max_new or self._seq_len_host[s] >= c.max_seq - 2:
+                    r.done = True
+                    finished.append(s)
+            for s in finished:
+                del self.running[s]
+                self._free_slots.append(s)
+                self._seq_len_host[s] = 0
+        return produced
+
+    # ---- forwards ----------------------------------------------------------
+    @staticmethod
+    def _apply_rope(x, cos, sin):
+        # x: [B, H, T, D]; cos/sin broadcastable to [B, H, T, D/2]
+        x1, x2 = x[..., ::2], x[..., 1::2]
+        out = torch.empty_like(x)
+        out[..., ::2] = x1 * cos - x2 * sin
+        out[..., 1::2] = x1 * sin + x2 * cos
+        return out
+
+    def _forward_block(self, h, layer, slot_idx, pos, is_prefill):
+        c = self.cfg
+        B, T, _ = h.shape
+        qkv = _rms(h, layer.ln1) @ layer.wqkv
+        q, k, v = qkv.split(c.d_model, dim=-1)
+        q = q.view(B, T, c.n_heads, c.head_dim).transpose(1, 2)
+        k = k.view(B, T, c.n_heads, c.head_dim).transpose(1, 2)
+        v = v.view(B, T, c.n_heads, c.head_dim).transpose(1, 2)
+        if is_prefill:
+            cos = self.cos[pos].view(1, 1, T, -1)  # pos: [T]
+            sin = self.sin[pos].view(1, 1, T, -1)
+        else:
+            cos = self.cos[pos].view(B, 1, 1, -1)  # pos: [B]
+            sin = self.sin[pos].view(B, 1, 1, -1)
+        q = self._apply_rope(q, cos, sin)
+        k = self._apply_rope(k, cos, sin)
+        li = layer._idx
+        if is_prefill:
+            s = slot_idx[0]
+            start = int(pos[0])
+            self.kv[li, 0, s, :, start: start + T] = k[0]
+            self.kv[li, 1, s, :, start: start + T] = v[0]
+            kk = self.kv[li, 0, s: s + 1, :, : start + T]
+            vv = self.kv[li, 1, s: s + 1, :, : start + T]
+            if start == 0:
+                attn = F.scaled_dot_product_attention(q, kk, vv, is_causal=True)
+            else:  # chunked continuation: causal mask with history offset
+                qpos = torch.arange(start, start + T, device=self.device)
+                kpos = torch.arange(0, start + T, device=self.device)
+                mask = kpos.unsqueeze(0) <= qpos.unsqueeze(1)
+                attn = F.scaled_dot_product_attention(q, kk, vv, attn_mask=mask.unsqueeze(0).unsqueeze(0))
+        else:
+            # decode: T == 1; write this step's k/v, then attend over windows
+            self.kv[li, 0, slot_idx, :, pos] = k[:, :, 0]
+            self.kv[li, 1, slot_idx, :, pos] = v[:, :, 0]
+            maxlen = int(pos.max().item()) + 1
+            kk = self.kv[li, 0].index_select(0, slot_idx)[:, :, :maxlen]
+            vv = self.kv[li, 1].index_select(0, slot_idx)[:, :, :maxlen]
+            kpos = torch.arange(maxlen, device=self.device)
+            mask = kpos.unsqueeze(0) <= pos.unsqueeze(1)  # [B, maxlen]
+            attn = F.scaled_dot_product_attention(q, kk, vv, attn_mask=mask.unsqueeze(1).unsqueeze(1))
+        attn = attn.transpose(1, 2).reshape(B, T, c.d_model)
+        h = h + attn @ layer.wo
+        x = _rms(h, layer.ln2)
+        h = h + (F.silu(x @ layer.w1) * (x @ layer.w3)) @ layer.w2
+        return h
+
+    def _prefill(self, slot: int, tokens: List[int], start: int) -> None:
+        c = self.cfg
+        t = torch.tensor([tokens], device=self.device, dtype=torch.long)
+        h = self.embed[t]
+        pos = torch.arange(start, start + len(tokens), device=self.device)
+        slot_idx = torch.tensor([slot], device=self.device)
+        for i, layer in enumerate(self.layers):
+            layer._idx = i
+            h = self._forward_block(h, layer, slot_idx, pos, is_prefill=True)
+        self._seq_len_host[slot] = start + len(tokens)
+        self.seq_len[slot] = start + len(tokens)
+
+    @torch.no_grad()
+    def _decode(self, slots: List[int], last_tokens: List[int]) -> List[int]:
+        c = self.cfg
+        slot_idx = torch.tensor(slots, device=self.device)
+        pos = self.seq_len[slot_idx]  # position to write = current length
+        t = torch.tensor(last_tokens, device=self.device, dtype=torch.long).unsqueeze(1)
+        h = self.embed[t]
+        for i, layer in enumerate(self.layers):
+            layer._idx = i
+            h = self._forward_block(h, layer, slot_idx, pos, is_prefill=False)
+        h = _rms(h, self.ln_f)
+        logits = h[:, 0] @ self.embed.t()
+        out = logits.argmax(-1)
+        self.seq_len[slot_idx] = pos + 1
+        for s in slots:
+            self._seq_len_host[s] += 1
+        return out.tolist()

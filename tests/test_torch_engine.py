@@ -1,0 +1,109 @@
+"""TorchEngine logic tests on CPU (tiny config, fp32); the same engine runs
+bf16 on MI355X (gpu-marked numerics test compares decode against a CPU fp32
+run of identical weights)."""
+import pytest
+import torch
+
+from smg_amd.engine.torch_engine import TorchEngine, TorchEngineConfig
+
+
+def make_engine(**kw):
+    return TorchEngine(TorchEngineConfig.tiny(), device="cpu", **kw)
+
+
+class TestEngineLogic:
+    def test_single_request_decodes(self):
+        eng = make_engine()
+        rid = eng.submit(list(range(10)), max_new_tokens=5)
+        for _ in range(32):
+            if eng.finished(rid):
+                break
+            eng.step()
+        out = eng.collect(rid)
+        assert len(out) == 5
+        assert all(0 <= t < eng.cfg.vocab_size for t in out)
+
+    def test_deterministic(self):
+        out = []
+        for _ in range(2):
+            eng = make_engine()
+            rid = eng.submit(list(range(16)), max_new_tokens=4)
+            while not eng.finished(rid):
+                eng.step()
+            out.append(eng.collect(rid))
+        assert out[0] == out[1]
+
+    def test_continuous_batching(self):
+        eng = make_engine()
+        rids = [eng.submit([i * 7 + j for j in range(8)], max_new_tokens=3) for i in range(4)]
+        for _ in range(64):
+            if all(eng.finished(r) for r in rids):
+                break
+            eng.step()
+        outs = [eng.collect(r) for r in rids]
+        assert all(len(o) == 3 for o in outs)
+
+    def test_decode_matches_unchunked_prefill(self):
+        # prefill in 2 chunks must produce the same next token as one chunk
+        cfg = TorchEngineConfig.tiny()
+        cfg.prefill_chunk = 8
+        eng1 = TorchEngine(cfg, device="cpu")
+        rid1 = eng1.submit(list(range(20)), max_new_tokens=1)
+        while not eng1.finished(rid1):
+            eng1.step()
+        cfg2 = TorchEngineConfig.tiny()  # prefill_chunk = 256 (single chunk)
+        eng2 = TorchEngine(cfg2, device="cpu")
+        rid2 = eng2.submit(list(range(20)), max_new_tokens=1)
+        while not eng2.finished(rid2):
+            eng2.step()
+        assert eng1.collect(rid1) == eng2.collect(rid2)
+
+    def test_slot_reuse(self):
+        eng = make_engine()
+        for round_ in range(3):
+            rids = [eng.submit(list(range(8)), max_new_tokens=2) for _ in range(eng.cfg.max_slots)]
+            for _ in range(64):
+                if all(eng.finished(r) for r in rids):
+                    break
+                eng.step()
+            assert all(len(eng.collect(r)) == 2 for r in rids)
+        assert len(eng._free_slots) == eng.cfg.max_slots
+
+    def test_load_snapshot(self):
+        eng = make_engine()
+        eng.submit(list(range(12)), max_new_tokens=3)
+        snap = eng.load_snapshot()
+        assert snap["num_queue_reqs"] + snap["num_running_reqs"] >= 1
+
+
+@pytest.mark.gpu
+class TestEngineGpu:
+    def test_gpu_decode_matches_cpu_fp32(self):
+        """bf16 GPU decode vs fp32 CPU decode of the same random weights:
+        greedy argmax token streams must agree for a short horizon."""
+        cfg = TorchEngineConfig.tiny()
+        gpu = TorchEngine(cfg, device="cuda:0")
+        cpu = TorchEngine(cfg, device="cpu")
+        prompt = list(range(32))
+        rg = gpu.submit(prompt, max_new_tokens=8)
+        rc = cpu.submit(prompt, max_new_tokens=8)
+        while not gpu.finished(rg):
+            gpu.step()
+        while not cpu.finished(rc):
+            cpu.step()
+        out_g, out_c = gpu.collect(rg), cpu.collect(rc)
+        # bf16 vs fp32 can diverge after the first argmax ties; require the
+        # first tokens to agree (deterministic weight init)
+        assert out_g[0] == out_c[0]
+        assert len(out_g) == 8
+
+    def test_gpu_throughput_smoke(self):
+        cfg = TorchEngineConfig.tiny()
+        eng = TorchEngine(cfg, device="cuda:0")
+        rids = [eng.submit(list(range(16)), max_new_tokens=8) for _ in range(8)]
+        for _ in range(128):
+            if all(eng.finished(r) for r in rids):
+                break
+            eng.step()
+        torch.cuda.synchronize()
+        assert all(len(eng.collect(r)) == 8 for r in rids)
