@@ -1,0 +1,300 @@
+#!/usr/bin/env python3
+"""Flagship serving benchmark — BASELINE.json metric: sustained req/s +
+p50 routing latency, cache_aware policy, at 1/2/4/8 workers.
+
+One process per GPU (torchrun for N>1).  Rank 0 runs the gateway: synthetic
+shared-prefix chat traffic (sim_load.py-style), cache-aware routing over the
+gfx950 GPU radix tree (one batched kernel launch per tick), RCCL-over-xGMI
+lockstep data plane to the other ranks.  Every rank runs a continuous-batching
+bf16 transformer worker engine (~1.1B params, random init) on its GPU.
+
+A "step" = reqs_per_step completed requests PER GPU (weak scaling).  W warmup
+steps untimed, then exactly K timed steps bracketed by barrier +
+torch.cuda.synchronize() on both sides; elapsed is MAX over ranks.  Rank 0
+prints one JSON line.
+
+Reference rig being reproduced: scripts/sim_ab.sh + scripts/sim_load.py +
+crates/mock_worker — here with real GPU engines instead of CPU
+simulators.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import random
+import statistics
+import sys
+import time
+from typing import Dict, List, Optional
+
+import torch
+
+from smg_amd.comm.plane import DONE, GatewayPlane, PlaneConfig, WorkerPlane
+from smg_amd.config import PolicyConfig
+from smg_amd.engine.torch_engine import TorchEngine, TorchEngineConfig
+from smg_amd.policies import CacheAwarePolicy, SelectWorkerInfo
+from smg_amd.workers.worker import Worker
+
+BASELINE_REQ_S = 2.65  # Rust SMG request throughput (other hardware)
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=8)
+    p.add_argument("--warmup", type=int, default=2)
+    p.add_argument("--reqs-per-step", type=int, default=32)
+    p.add_argument("--concurrency", type=int, default=48, help="in-flight requests per worker")
+    p.add_argument("--prefix-pool", type=int, default=8)
+    p.add_argument("--prefix-len", type=int, default=512)
+    p.add_argument("--suffix-len", type=int, default=64)
+    p.add_argument("--max-new", type=int, default=32)
+    p.add_argument("--tiny", action="store_true", help="tiny model (CPU smoke / CI)")
+    p.add_argument("--seed", type=int, default=1234)
+    return p.parse_args()
+
+
+class LoadGen:
+    """Shared-prefix synthetic chat traffic (reference scripts/sim_load.py)."""
+
+    def __init__(self, args, vocab: int):
+        self.rng = random.Random(args.seed)
+        self.vocab = vocab
+        self.prefixes = [
+            [self.rng.randrange(vocab) for _ in range(args.prefix_len)] for _ in range(args.prefix_pool)
+        ]
+        self.suffix_len = args.suffix_len
+        self.max_new = args.max_new
+        self._rid = 0
+
+    def make(self):
+        self._rid += 1
+        prompt = list(self.rng.choice(self.prefixes)) + [
+            self.rng.randrange(self.vocab) for _ in range(self.suffix_len)
+        ]
+        return self._rid, prompt, self.max_new
+
+
+def engine_config(args) -> TorchEngineConfig:
+    if args.tiny:
+        cfg = TorchEngineConfig.tiny()
+        cfg.max_slots = max(8, args.concurrency)
+        cfg.max_seq = 256
+        return cfg
+    cfg = TorchEngineConfig.bench_1b()
+    cfg.max_slots = args.concurrency + 8
+    cfg.max_seq = args.prefix_len + args.suffix_len + args.max_new + 16
+    return cfg
+
+
+def worker_main(rank: int, world: int, args, device: str, backend: str):
+    eng = TorchEngine(engine_config(args), device=device)
+    plane = WorkerPlane(PlaneConfig(max_prompt=args.prefix_len + args.suffix_len + 8, device=device if backend == "nccl" else "cpu"))
+    events: List[tuple] = []
+    t0 = t1 = None
+    while True:
+        reqs, stop = plane.tick(events)
+        if plane.barrier_requested:
+            torch.distributed.barrier()
+            if device.startswith("cuda"):
+                torch.cuda.synchronize()
+            if t0 is None:
+                t0 = time.perf_counter()
+            else:
+                t1 = time.perf_counter()
+        if stop:
+            break
+        for rid, max_new, prompt in reqs:
+            eng.submit(prompt, max_new, rid=rid)
+        eng.step()
+        events = eng.drain_events()
+    elapsed = (t1 - t0) if (t0 is not None and t1 is not None) else 0.0
+    el = torch.tensor([elapsed])
+    torch.distributed.all_reduce(el, op=torch.distributed.ReduceOp.MAX)
+
+
+class LocalWorker:
+    """Rank-0's own engine, driven inline in the tick loop."""
+
+    def __init__(self, eng: TorchEngine):
+        self.eng = eng
+        self.pending: List[tuple] = []
+
+    def enqueue(self, rid, max_new, prompt):
+        self.pending.append((rid, max_new, prompt))
+
+    def tick(self) -> List[tuple]:
+        for rid, max_new, prompt in self.pending:
+            self.eng.submit(prompt, max_new, rid=rid)
+        self.pending.clear()
+        self.eng.step()
+        return self.eng.drain_events()
+
+
+def gateway_main(rank: int, world: int, args, device: str, backend: str):
+    use_gpu = device.startswith("cuda")
+    eng = TorchEngine(engine_config(args), device=device)
+    local = LocalWorker(eng)
+    remote_ranks = list(range(1, world))
+    plane = (
+        GatewayPlane(
+            PlaneConfig(max_prompt=args.prefix_len + args.suffix_len + 8, device=device if backend == "nccl" else "cpu"),
+            remote_ranks,
+        )
+        if remote_ranks
+        else None
+    )
+
+    # gateway state: one Worker per rank; cache_aware over the GPU tree
+    workers = []
+    for r in range(world):
+        w = Worker(f"rccl://rank-{r}", model_id="bench-1b", rccl_rank=r)
+        workers.append(w)
+    pol_cfg = PolicyConfig(name="cache_aware", block_size=16, gpu_tree=use_gpu, gpu_tree_device=0)
+    policy = CacheAwarePolicy(pol_cfg)
+    vocab = engine_config(args).vocab_size
+    gen = LoadGen(args, vocab)
+    inflight: Dict[int, int] = {}  # rid -> worker rank
+    target_inflight = args.concurrency * world
+
+    completed_total = 0
+    routing_lat: List[float] = []
+
+    def one_tick():
+        nonlocal completed_total
+        # 1) new arrivals to hold steady-state concurrency
+        new_reqs = []
+        while len(inflight) + len(new_reqs) < target_inflight:
+            new_reqs.append(gen.make())
+            if len(new_reqs) >= 64:
+                break
+        # 2) route the batch (GPU kernel: one launch)
+        if new_reqs:
+            infos = [
+                SelectWorkerInfo(request_id=str(rid), model_id="bench-1b", tokens=prompt, est_tokens=len(prompt))
+                for rid, prompt, _ in new_reqs
+            ]
+            t0 = time.perf_counter()
+            sels = policy.select_worker_batch(workers, infos)
+            dt = time.perf_counter() - t0
+            routing_lat.extend([dt / max(1, len(new_reqs))] * len(new_reqs))
+            for (rid, prompt, max_new), sel in zip(new_reqs, sels):
+                sel = 0 if sel is None else sel
+                workers[sel].incr_load()
+                inflight[rid] = sel
+                if sel == 0:
+                    local.enqueue(rid, max_new, prompt)
+                else:
+                    plane.enqueue(sel, rid, max_new, prompt)
+        # 3) lockstep exchange + local step
+        events = list(local.tick())
+        if plane is not None:
+            for w, evs in plane.tick().items():
+                events.extend(evs)
+        # 4) completions
+        done_now = 0
+        for rid, _token, flags in events:
+            if flags & DONE:
+                wrk = inflight.pop(rid, None)
+                if wrk is not None:
+                    workers[wrk].decr_load()
+                    workers[wrk].processed_requests += 1
+                    done_now += 1
+        completed_total += done_now
+        return done_now
+
+    def run_until(n_completions: int, max_ticks: int = 1_000_000):
+        base = completed_total
+        ticks = 0
+        while completed_total - base < n_completions and ticks < max_ticks:
+            one_tick()
+            ticks += 1
+
+    def sync_point(first: bool):
+        if plane is not None:
+            plane.tick(barrier=True)
+            torch.distributed.barrier()
+        if use_gpu:
+            torch.cuda.synchronize()
+
+    # ---- fill + warmup ----------------------------------------------------
+    run_until(args.warmup * args.reqs_per_step * world + target_inflight // 2)
+    sync_point(first=True)
+    t0 = time.perf_counter()
+    run_until(args.steps * args.reqs_per_step * world)
+    timed_completions = args.steps * args.reqs_per_step * world
+    sync_point(first=False)
+    t1 = time.perf_counter()
+    elapsed = t1 - t0
+    # stop workers and fold in their elapsed (max over ranks)
+    if plane is not None:
+        plane.tick(stop=True)
+        el = torch.tensor([elapsed])
+        torch.distributed.all_reduce(el, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(el[0])
+
+    req_s = timed_completions / elapsed
+    p50_route = statistics.median(routing_lat) * 1e3 if routing_lat else None
+    tokens_done = timed_completions * args.max_new
+    result = {
+        "metric": "sustained req/s + p50 routing latency, cache_aware policy at 1/2/4/8 workers",
+        "value": round(req_s, 3),
+        "unit": "req/s",
+        "n_gpus": world,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": round(elapsed * 1e3 / args.steps, 3),
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": round(req_s / BASELINE_REQ_S, 2),
+        "dtype": "bf16" if use_gpu else "fp32",
+        "data": "synthetic",
+        "config": {
+            "model": "smg-bench-1b (16L d2048 h16, bf16, random-init)" if not args.tiny else "tiny",
+            "global_batch": args.concurrency * world,
+            "seq_len": args.prefix_len + args.suffix_len,
+            "parallelism": f"gateway+{world}workers (dp{world}, cache_aware, rccl-xgmi)",
+            "policy": "cache_aware",
+            "gpu_tree": use_gpu,
+            "p50_routing_latency_ms": round(p50_route, 4) if p50_route is not None else None,
+            "output_tokens_per_req": args.max_new,
+            "total_output_tokens_per_s": round(tokens_done / elapsed, 1),
+            "reqs_per_step": args.reqs_per_step,
+        },
+    }
+    print(json.dumps(result), flush=True)
+
+
+def main():
+    args = parse_args()
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    use_gpu = torch.cuda.is_available()
+    if not use_gpu:
+        args.tiny = True
+    device = f"cuda:{local_rank}" if use_gpu else "cpu"
+    if use_gpu:
+        torch.cuda.set_device(local_rank)
+    if world > 1:
+        backend = "nccl" if use_gpu else "gloo"
+        torch.distributed.init_process_group(
+            backend=backend,
+            rank=rank,
+            world_size=world,
+        )
+    else:
+        backend = "local"
+    try:
+        if rank == 0:
+            gateway_main(rank, world, args, device, backend)
+        else:
+            worker_main(rank, world, args, device, backend)
+    finally:
+        if world > 1:
+            torch.distributed.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -124,6 +124,7 @@ class TorchEngine:
         self._requests: Dict[str, _Request] = {}
         self._rid_counter = 0
         self.total_generated = 0
+        self._step_events: List[tuple] = []  # (rid, token, done) since last drain
         self.graphs = graphs
         self._graph_cache: Dict[int, tuple] = {}
 
@@ -149,6 +150,12 @@ class TorchEngine:
 
     def n_active(self) -> int:
         return len(self.running) + len(self.waiting)
+
+    def drain_events(self) -> List[tuple]:
+        """(rid, token, done) tuples produced since the last drain — the
+        event stream the RCCL plane ships back to the gateway each tick."""
+        out, self._step_events = self._step_events, []
+        return out
 
     def load_snapshot(self) -> Dict:
         c = self.cfg
@@ -200,6 +207,7 @@ class TorchEngine:
                 if len(r.generated) >= r.max_new or self._seq_len_host[s] >= c.max_seq - 2:
                     r.done = True
                     finished.append(s)
+                self._step_events.append((r.rid, int(tok), 1 if r.done else 0))
             for s in finished:
                 del self.running[s]
                 self._free_slots.append(s)

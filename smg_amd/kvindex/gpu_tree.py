@@ -29,6 +29,10 @@ class GpuTokenTree:
         max_batch_reqs: int = 4096,
         max_batch_tokens: int = 1 << 22,
     ):
+        try:
+            import torch  # noqa: F401 — HIP runtime must load before _core.so
+        except ImportError:
+            pass
         from .. import _core
 
         if _core.hip_device_count() <= 0:

@@ -7,6 +7,7 @@ import pytest
 
 from smg_amd.kvindex.pytree import PagedRadixTree
 
+pytest.importorskip("torch")  # torch's HIP runtime must load before _core.so
 core = pytest.importorskip("smg_amd._core")
 
 
@@ -126,7 +127,7 @@ class TestGpuTreeDifferential:
             gpu.insert([i * 100 + j for j in range(16)], "http://a")
         before = len(gpu)
         assert before >= 50
-        gpu._tree.evict_older(int(gpu.stats()["clock"]))  # everything is old
+        gpu._tree.evict_older(int(gpu.stats()["clock"]) + 1)  # everything is older than cutoff
         assert len(gpu) == 0
 
     def test_gpu_large_batch(self):
