@@ -51,6 +51,7 @@ def parse_args():
     p.add_argument("--suffix-len", type=int, default=64)
     p.add_argument("--max-new", type=int, default=32)
     p.add_argument("--tiny", action="store_true", help="tiny model (CPU smoke / CI)")
+    p.add_argument("--no-graphs", action="store_true", help="disable hipGraph decode capture")
     p.add_argument("--seed", type=int, default=1234)
     return p.parse_args()
 
@@ -89,7 +90,7 @@ def engine_config(args) -> TorchEngineConfig:
 
 
 def worker_main(rank: int, world: int, args, device: str, backend: str):
-    eng = TorchEngine(engine_config(args), device=device)
+    eng = TorchEngine(engine_config(args), device=device, graphs=device.startswith("cuda") and not args.no_graphs)
     plane = WorkerPlane(PlaneConfig(max_prompt=args.prefix_len + args.suffix_len + 8, device=device if backend == "nccl" else "cpu"))
     events: List[tuple] = []
     t0 = t1 = None
@@ -134,7 +135,7 @@ class LocalWorker:
 
 def gateway_main(rank: int, world: int, args, device: str, backend: str):
     use_gpu = device.startswith("cuda")
-    eng = TorchEngine(engine_config(args), device=device)
+    eng = TorchEngine(engine_config(args), device=device, graphs=use_gpu and not args.no_graphs)
     local = LocalWorker(eng)
     remote_ranks = list(range(1, world))
     plane = (

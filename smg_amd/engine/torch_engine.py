@@ -72,9 +72,8 @@ class _Layer:
 
 
 def _rms(x, weight):
-    v = x.float()
-    v = v * torch.rsqrt(v.pow(2).mean(-1, keepdim=True) + 1e-5)
-    return (v * weight.float()).to(x.dtype)
+    # fused on ROCm (single kernel); computes in fp32 internally
+    return F.rms_norm(x, (x.shape[-1],), weight=weight, eps=1e-5)
 
 
 class _Request:
@@ -118,6 +117,8 @@ class TorchEngine:
             self.sin = freqs.sin().to(self.dtype)
         self.seq_len = torch.zeros(c.max_slots, dtype=torch.long, device=self.device)
         self._seq_len_host = [0] * c.max_slots
+        self._last_tok = torch.zeros(c.max_slots, dtype=torch.long, device=self.device)
+        self._arange_slots = torch.arange(c.max_slots, device=self.device)
         self._free_slots = list(range(c.max_slots - 1, -1, -1))
         self.waiting: List[_Request] = []
         self.running: Dict[int, _Request] = {}  # slot -> request
@@ -193,11 +194,7 @@ class TorchEngine:
         # decode: all slots whose prompt is fully prefilled
         decode_slots = [s for s, r in self.running.items() if r.prefilled >= len(r.tokens)]
         if decode_slots:
-            last_tokens = []
-            for s in decode_slots:
-                r = self.running[s]
-                last_tokens.append(r.generated[-1] if r.generated else r.tokens[-1])
-            new_tokens = self._decode(decode_slots, last_tokens)
+            new_tokens = self._decode(decode_slots)
             finished = []
             for s, tok in zip(decode_slots, new_tokens):
                 r = self.running[s]
@@ -224,7 +221,11 @@ class TorchEngine:
         out[..., 1::2] = x1 * sin + x2 * cos
         return out
 
-    def _forward_block(self, h, layer, slot_idx, pos, is_prefill):
+    def _mlp(self, h, layer):
+        x = _rms(h, layer.ln2)
+        return h + (F.silu(x @ layer.w1) * (x @ layer.w3)) @ layer.w2
+
+    def _qkv(self, h, layer, cos, sin):
         c = self.cfg
         B, T, _ = h.shape
         qkv = _rms(h, layer.ln1) @ layer.wqkv
@@ -232,71 +233,103 @@ class TorchEngine:
         q = q.view(B, T, c.n_heads, c.head_dim).transpose(1, 2)
         k = k.view(B, T, c.n_heads, c.head_dim).transpose(1, 2)
         v = v.view(B, T, c.n_heads, c.head_dim).transpose(1, 2)
-        if is_prefill:
-            cos = self.cos[pos].view(1, 1, T, -1)  # pos: [T]
-            sin = self.sin[pos].view(1, 1, T, -1)
-        else:
-            cos = self.cos[pos].view(B, 1, 1, -1)  # pos: [B]
-            sin = self.sin[pos].view(B, 1, 1, -1)
-        q = self._apply_rope(q, cos, sin)
-        k = self._apply_rope(k, cos, sin)
-        li = layer._idx
-        if is_prefill:
-            s = slot_idx[0]
-            start = int(pos[0])
-            self.kv[li, 0, s, :, start: start + T] = k[0]
-            self.kv[li, 1, s, :, start: start + T] = v[0]
-            kk = self.kv[li, 0, s: s + 1, :, : start + T]
-            vv = self.kv[li, 1, s: s + 1, :, : start + T]
-            if start == 0:
-                attn = F.scaled_dot_product_attention(q, kk, vv, is_causal=True)
-            else:  # chunked continuation: causal mask with history offset
-                qpos = torch.arange(start, start + T, device=self.device)
-                kpos = torch.arange(0, start + T, device=self.device)
-                mask = kpos.unsqueeze(0) <= qpos.unsqueeze(1)
-                attn = F.scaled_dot_product_attention(q, kk, vv, attn_mask=mask.unsqueeze(0).unsqueeze(0))
-        else:
-            # decode: T == 1; write this step's k/v, then attend over windows
-            self.kv[li, 0, slot_idx, :, pos] = k[:, :, 0]
-            self.kv[li, 1, slot_idx, :, pos] = v[:, :, 0]
-            maxlen = int(pos.max().item()) + 1
-            kk = self.kv[li, 0].index_select(0, slot_idx)[:, :, :maxlen]
-            vv = self.kv[li, 1].index_select(0, slot_idx)[:, :, :maxlen]
-            kpos = torch.arange(maxlen, device=self.device)
-            mask = kpos.unsqueeze(0) <= pos.unsqueeze(1)  # [B, maxlen]
-            attn = F.scaled_dot_product_attention(q, kk, vv, attn_mask=mask.unsqueeze(1).unsqueeze(1))
-        attn = attn.transpose(1, 2).reshape(B, T, c.d_model)
-        h = h + attn @ layer.wo
-        x = _rms(h, layer.ln2)
-        h = h + (F.silu(x @ layer.w1) * (x @ layer.w3)) @ layer.w2
-        return h
-
-    def _prefill(self, slot: int, tokens: List[int], start: int) -> None:
-        c = self.cfg
-        t = torch.tensor([tokens], device=self.device, dtype=torch.long)
-        h = self.embed[t]
-        pos = torch.arange(start, start + len(tokens), device=self.device)
-        slot_idx = torch.tensor([slot], device=self.device)
-        for i, layer in enumerate(self.layers):
-            layer._idx = i
-            h = self._forward_block(h, layer, slot_idx, pos, is_prefill=True)
-        self._seq_len_host[slot] = start + len(tokens)
-        self.seq_len[slot] = start + len(tokens)
+        return self._apply_rope(q, cos, sin), self._apply_rope(k, cos, sin), v
 
     @torch.no_grad()
-    def _decode(self, slots: List[int], last_tokens: List[int]) -> List[int]:
+    def _prefill(self, slot: int, tokens: List[int], start: int) -> None:
         c = self.cfg
-        slot_idx = torch.tensor(slots, device=self.device)
-        pos = self.seq_len[slot_idx]  # position to write = current length
-        t = torch.tensor(last_tokens, device=self.device, dtype=torch.long).unsqueeze(1)
+        T = len(tokens)
+        t = torch.tensor([tokens], device=self.device, dtype=torch.long)
         h = self.embed[t]
-        for i, layer in enumerate(self.layers):
-            layer._idx = i
-            h = self._forward_block(h, layer, slot_idx, pos, is_prefill=False)
+        pos = torch.arange(start, start + T, device=self.device)
+        cos = self.cos[pos].view(1, 1, T, -1)
+        sin = self.sin[pos].view(1, 1, T, -1)
+        mask = None
+        if start:  # chunked continuation: causal mask with history offset
+            kpos = torch.arange(0, start + T, device=self.device)
+            mask = (kpos.unsqueeze(0) <= pos.unsqueeze(1)).unsqueeze(0).unsqueeze(0)
+        for li, layer in enumerate(self.layers):
+            q, k, v = self._qkv(h, layer, cos, sin)
+            self.kv[li, 0, slot, :, start: start + T] = k[0]
+            self.kv[li, 1, slot, :, start: start + T] = v[0]
+            kk = self.kv[li, 0, slot: slot + 1, :, : start + T]
+            vv = self.kv[li, 1, slot: slot + 1, :, : start + T]
+            if mask is None:
+                attn = F.scaled_dot_product_attention(q, kk, vv, is_causal=True)
+            else:
+                attn = F.scaled_dot_product_attention(q, kk, vv, attn_mask=mask)
+            h = h + attn.transpose(1, 2).reshape(1, T, c.d_model) @ layer.wo
+            h = self._mlp(h, layer)
+        self._seq_len_host[slot] = start + T
+        self.seq_len[slot] = start + T
+        self._last_tok[slot] = int(tokens[-1])
+
+    def _decode_core(self, maxlen: int) -> torch.Tensor:
+        """Full-arena decode forward: every slot participates with a static
+        shape (no KV gather — the per-slot windows are views); inactive slots
+        compute garbage that is masked out of the logical state.  A free
+        slot's stray K/V write lands at its stale position and is always
+        overwritten by the next prefill before it can be attended.  Static
+        shapes per `maxlen` make this hipGraph-capturable."""
+        c = self.cfg
+        S = c.max_slots
+        pos = self.seq_len  # [S] current length == write position
+        cos = self.cos[pos].view(S, 1, 1, -1)
+        sin = self.sin[pos].view(S, 1, 1, -1)
+        kpos = torch.arange(maxlen, device=self.device)
+        mask = (kpos.unsqueeze(0) <= pos.unsqueeze(1)).unsqueeze(1).unsqueeze(1)
+        h = self.embed[self._last_tok.unsqueeze(1)]  # [S, 1, D]
+        for li, layer in enumerate(self.layers):
+            q, k, v = self._qkv(h, layer, cos, sin)
+            self.kv[li, 0, self._arange_slots, :, pos] = k[:, :, 0]
+            self.kv[li, 1, self._arange_slots, :, pos] = v[:, :, 0]
+            kk = self.kv[li, 0][:, :, :maxlen]
+            vv = self.kv[li, 1][:, :, :maxlen]
+            attn = F.scaled_dot_product_attention(q, kk, vv, attn_mask=mask)
+            h = h + attn.transpose(1, 2).reshape(S, 1, c.d_model) @ layer.wo
+            h = self._mlp(h, layer)
         h = _rms(h, self.ln_f)
         logits = h[:, 0] @ self.embed.t()
-        out = logits.argmax(-1)
-        self.seq_len[slot_idx] = pos + 1
+        return logits.argmax(-1)  # [S]
+
+    GRAPH_BUCKET = 256
+
+    def _decode_graphed(self, maxlen: int) -> torch.Tensor:
+        """hipGraph-captured decode keyed by the maxlen bucket: one replay
+        instead of ~8 kernel launches per layer.  pos/_last_tok are read
+        inside the graph from their persistent device tensors."""
+        bucket = min(((maxlen + self.GRAPH_BUCKET - 1) // self.GRAPH_BUCKET) * self.GRAPH_BUCKET,
+                     self.cfg.max_seq)
+        entry = self._graph_cache.get(bucket)
+        if entry is None:
+            torch.cuda.synchronize()
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                for _ in range(2):  # warmup allocations outside capture
+                    self._decode_core(bucket)
+            torch.cuda.current_stream().wait_stream(s)
+            torch.cuda.synchronize()
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                out = self._decode_core(bucket)
+            entry = (g, out)
+            self._graph_cache[bucket] = entry
+        g, out = entry
+        g.replay()
+        return out
+
+    @torch.no_grad()
+    def _decode(self, slots: List[int]) -> List[int]:
+        maxlen = max(self._seq_len_host) + 1
+        if self.graphs and self.device.type == "cuda":
+            nxt = self._decode_graphed(maxlen)
+        else:
+            nxt = self._decode_core(maxlen)
+        act = torch.tensor(slots, device=self.device)
+        self._last_tok.index_copy_(0, act, nxt.index_select(0, act))
+        self.seq_len.index_add_(0, act, torch.ones_like(act))
+        toks = nxt.index_select(0, act).tolist()
         for s in slots:
             self._seq_len_host[s] += 1
-        return out.tolist()
+        return toks

@@ -78,24 +78,39 @@ class TestEngineLogic:
 
 @pytest.mark.gpu
 class TestEngineGpu:
-    def test_gpu_decode_matches_cpu_fp32(self):
-        """bf16 GPU decode vs fp32 CPU decode of the same random weights:
-        greedy argmax token streams must agree for a short horizon."""
+    def test_gpu_numerics_vs_cpu_fp32(self):
+        """Numerics check against a plain PyTorch fp32 reference: run the GPU
+        engine in fp32 and compare the prefill KV cache + first decode logits
+        path (via the KV written at the first decode position) against the
+        identical-weights CPU fp32 engine.  bf16 argmax on random-init logits
+        is not a meaningful comparison (near-uniform logits); hidden-state
+        closeness is."""
         cfg = TorchEngineConfig.tiny()
+        cfg.dtype = "float32"
         gpu = TorchEngine(cfg, device="cuda:0")
-        cpu = TorchEngine(cfg, device="cpu")
+        cpu = TorchEngine(TorchEngineConfig.tiny(), device="cpu")
+        # identical weights: copy CPU weights to GPU engine
+        gpu.embed.copy_(cpu.embed.to(gpu.device))
+        for lg, lc in zip(gpu.layers, cpu.layers):
+            for attr in ("wqkv", "wo", "w1", "w2", "w3", "ln1", "ln2"):
+                getattr(lg, attr).copy_(getattr(lc, attr).to(gpu.device))
+        gpu.ln_f.copy_(cpu.ln_f.to(gpu.device))
         prompt = list(range(32))
-        rg = gpu.submit(prompt, max_new_tokens=8)
-        rc = cpu.submit(prompt, max_new_tokens=8)
+        rg = gpu.submit(prompt, max_new_tokens=2)
+        rc = cpu.submit(prompt, max_new_tokens=2)
         while not gpu.finished(rg):
             gpu.step()
         while not cpu.finished(rc):
             cpu.step()
-        out_g, out_c = gpu.collect(rg), cpu.collect(rc)
-        # bf16 vs fp32 can diverge after the first argmax ties; require the
-        # first tokens to agree (deterministic weight init)
-        assert out_g[0] == out_c[0]
-        assert len(out_g) == 8
+        torch.cuda.synchronize()
+        # prefill + decode KV of slot 0 must agree to fp32 GEMM tolerance
+        n = len(prompt) + 2
+        kv_g = gpu.kv[:, :, 0, :, :n].float().cpu()
+        kv_c = cpu.kv[:, :, 0, :, :n].float()
+        assert torch.allclose(kv_g, kv_c, rtol=1e-3, atol=1e-4), (
+            (kv_g - kv_c).abs().max().item()
+        )
+        assert len(gpu.collect(rg)) == 2
 
     def test_gpu_throughput_smoke(self):
         cfg = TorchEngineConfig.tiny()
