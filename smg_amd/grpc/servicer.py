@@ -1,0 +1,212 @@
+"""Engine-side gRPC servicer (reference: grpc_servicer/ Python package —
+vllm/servicer.py, sglang/servicer.py).
+
+Bridges the smg.Scheduler RPC surface onto an engine: the CPU mock simulator
+(tests / CPU fleets) or the GPU TorchEngine (MI355X workers).  grpc.aio with
+generic bytes handlers + msgpack (see api.py).
+"""
+from __future__ import annotations
+
+import asyncio
+import logging
+from typing import AsyncIterator, Optional
+
+import grpc
+
+from ..mock.engine import MockWorkerEngine, SimConfig
+from . import api
+
+log = logging.getLogger("smg.grpc.servicer")
+
+
+class EngineAdapter:
+    """Uniform async facade over MockWorkerEngine / TorchEngine."""
+
+    def __init__(self, engine):
+        self.engine = engine
+        self.is_mock = isinstance(engine, MockWorkerEngine)
+        self._torch_task: Optional[asyncio.Task] = None
+        self._streams = {}
+        self._kv_event_subs = []
+
+    async def start(self):
+        if self.is_mock:
+            await self.engine.start()
+        else:
+            self._torch_task = asyncio.ensure_future(self._torch_loop())
+
+    async def stop(self):
+        if self.is_mock:
+            await self.engine.stop()
+        elif self._torch_task:
+            self._torch_task.cancel()
+
+    async def _torch_loop(self):
+        """Drive the GPU engine; fan out per-request events."""
+        loop = asyncio.get_event_loop()
+        while True:
+            if self.engine.n_active() == 0:
+                await asyncio.sleep(0.002)
+                continue
+            await loop.run_in_executor(None, self.engine.step)
+            for rid, token, done in self.engine.drain_events():
+                q = self._streams.get(rid)
+                if q is not None:
+                    q.put_nowait((token, bool(done)))
+            await asyncio.sleep(0)
+
+    async def generate(self, req: api.GenerateRequest) -> AsyncIterator[api.GenerateChunk]:
+        if self.is_mock:
+            sim_req = self.engine.sim.submit(req.input_ids, req.sampling.max_new_tokens, rid=req.request_id)
+            n = 0
+            while True:
+                kind, tok = await sim_req.stream_q.get()
+                if kind == "token":
+                    n += 1
+                    yield api.GenerateChunk(req.request_id, [tok], False, None, len(req.input_ids), n)
+                elif kind == "done":
+                    yield api.GenerateChunk(req.request_id, [], True, "length", len(req.input_ids), n)
+                    return
+                elif kind == "aborted":
+                    yield api.GenerateChunk(req.request_id, [], True, "abort", len(req.input_ids), n)
+                    return
+        else:
+            q: asyncio.Queue = asyncio.Queue()
+            self._streams[req.request_id] = q
+            self.engine.submit(req.input_ids, req.sampling.max_new_tokens, rid=req.request_id)
+            n = 0
+            try:
+                while True:
+                    token, done = await q.get()
+                    n += 1
+                    yield api.GenerateChunk(req.request_id, [token], False, None, len(req.input_ids), n)
+                    if done:
+                        yield api.GenerateChunk(req.request_id, [], True, "length", len(req.input_ids), n)
+                        return
+            finally:
+                self._streams.pop(req.request_id, None)
+
+    def abort(self, request_id: str) -> None:
+        if self.is_mock:
+            self.engine.sim.abort(request_id)
+
+    def loads(self) -> dict:
+        if self.is_mock:
+            return self.engine.sim.load_snapshot()
+        return self.engine.load_snapshot()
+
+    def drain_kv_events(self):
+        if self.is_mock:
+            return self.engine.sim.drain_kv_events()
+        return []
+
+
+class SchedulerServicer(grpc.GenericRpcHandler):
+    """Generic bytes-level handler for the smg.Scheduler service."""
+
+    def __init__(self, adapter: EngineAdapter, model_id: str = "mock-model"):
+        self.adapter = adapter
+        self.model_id = model_id
+
+    def service(self, handler_call_details):
+        name = handler_call_details.method.rsplit("/", 1)[-1]
+        if handler_call_details.method != api.method(name) or name not in api.METHODS:
+            return None
+        kind = api.METHODS[name]
+        fn = getattr(self, f"_h_{name.lower()}", None)
+        if fn is None:
+            return None
+        if kind == "server_stream":
+            return grpc.unary_stream_rpc_method_handler(fn)
+        return grpc.unary_unary_rpc_method_handler(fn)
+
+    # ---- handlers (sync grpc server; executed in thread pool) ------------
+    def _h_generate(self, request: bytes, context):
+        req = api.GenerateRequest.from_dict(api.loads(request))
+        loop = self._loop()
+        agen = self.adapter.generate(req)
+        try:
+            while True:
+                chunk = asyncio.run_coroutine_threadsafe(agen.__anext__(), loop).result()
+                yield api.dumps(chunk)
+                if chunk.finished:
+                    break
+        except StopAsyncIteration:
+            pass
+        except Exception as exc:  # abort on client cancel / engine error
+            log.warning("generate stream error: %s", exc)
+            context.abort(grpc.StatusCode.INTERNAL, str(exc))
+
+    def _h_embed(self, request: bytes, context):
+        d = api.loads(request)
+        import hashlib
+
+        ids = d.get("input_ids") or []
+        vec = [int.from_bytes(hashlib.blake2b(bytes(str(i), "utf8"), digest_size=4).digest(), "little") % 1000 / 1000.0 for i in ids[:16]]
+        vec += [0.0] * (16 - len(vec))
+        return api.dumps({"request_id": d.get("request_id"), "embedding": vec})
+
+    def _h_healthcheck(self, request: bytes, context):
+        return api.dumps({"healthy": True})
+
+    def _h_abort(self, request: bytes, context):
+        d = api.loads(request)
+        self.adapter.abort(d.get("request_id", ""))
+        return api.dumps({"status": "aborted"})
+
+    def _h_getmodelinfo(self, request: bytes, context):
+        return api.dumps({"model_path": self.model_id, "is_generation": True, "max_context_length": 131072})
+
+    def _h_getserverinfo(self, request: bytes, context):
+        return api.dumps({"version": "0.1.0", "engine": "mock" if self.adapter.is_mock else "torch"})
+
+    def _h_getloads(self, request: bytes, context):
+        return api.dumps({"loads": self.adapter.loads()})
+
+    def _h_flushcache(self, request: bytes, context):
+        if self.adapter.is_mock:
+            self.adapter.engine.sim.prefix_cache.clear()
+        return api.dumps({"status": "ok"})
+
+    def _h_subscribekvevents(self, request: bytes, context):
+        import time
+
+        while context.is_active():
+            events = self.adapter.drain_kv_events()
+            if events:
+                yield api.dumps({"events": events, "block_size": getattr(self.adapter.engine.config, "block_size", 16)})
+            time.sleep(0.05)
+
+    def _h_startprofile(self, request: bytes, context):
+        return api.dumps({"status": "profiling not active on this engine"})
+
+    _h_stopprofile = _h_startprofile
+
+    def _loop(self):
+        return self._event_loop
+
+    def bind_loop(self, loop):
+        self._event_loop = loop
+
+
+async def serve_grpc_worker(
+    host: str = "127.0.0.1",
+    port: int = 50051,
+    engine=None,
+    model_id: str = "mock-model",
+    sim_config: Optional[SimConfig] = None,
+):
+    """Start a gRPC engine worker; returns (server, adapter, bound_port)."""
+    from concurrent.futures import ThreadPoolExecutor
+
+    if engine is None:
+        engine = MockWorkerEngine(sim_config or SimConfig(model_id=model_id))
+    adapter = EngineAdapter(engine)
+    await adapter.start()
+    servicer = SchedulerServicer(adapter, model_id)
+    servicer.bind_loop(asyncio.get_event_loop())
+    server = grpc.server(ThreadPoolExecutor(max_workers=32))
+    server.add_generic_rpc_handlers((servicer,))
+    bound = server.add_insecure_port(f"{host}:{port}")
+    server.start()
+    return server, adapter, bound

@@ -1,0 +1,245 @@
+"""gRPC router pipeline (reference: model_gateway/src/routers/grpc/pipeline.rs:192
+— stage lists :258: Preparation -> WorkerSelection -> ClientAcquisition ->
+[Encode (EPD)] -> RequestBuilding -> DispatchMetadata -> RequestExecution ->
+ResponseProcessing; endpoints pipeline.rs:64-71).
+
+Each stage is an async callable over a shared PipelineContext; the response
+processing stage runs the streaming tail: incremental detok ->
+StopSequenceDecoder -> streaming reasoning parser -> streaming tool parser ->
+SSE encode (reference regular/streaming.rs).
+"""
+from __future__ import annotations
+
+import asyncio
+import json
+import time
+import uuid
+from dataclasses import dataclass, field
+from typing import Any, AsyncIterator, Dict, List, Optional
+
+from ...grpc import api
+from ...grpc.client import ClientPool
+from ...parsers.reasoning import get_reasoning_parser
+from ...parsers.tool import get_parser
+from ...parsers.tool.stream import StreamingToolParser
+from ...policies import SelectWorkerInfo
+from ...protocols.openai import ProtocolError, error_body
+from ...tokenizer.chat_template import ChatTemplate
+from ...tokenizer.stop import DecodeStream, StopSequenceDecoder
+from ..base import RouteRequest, RouteResponse
+
+
+@dataclass
+class PipelineContext:
+    req: RouteRequest
+    endpoint: str  # chat | completion | generate | embedding
+    body: Dict[str, Any] = field(default_factory=dict)
+    model_id: Optional[str] = None
+    text: str = ""
+    input_ids: List[int] = field(default_factory=list)
+    sampling: api.SamplingParams = field(default_factory=api.SamplingParams)
+    worker: Any = None
+    dp_rank: Optional[int] = None
+    client: Any = None
+    gen_request: Optional[api.GenerateRequest] = None
+    request_id: str = ""
+    stream: bool = False
+    tokenizer: Any = None
+    error: Optional[RouteResponse] = None
+
+
+class Stage:
+    name = "stage"
+
+    async def run(self, ctx: PipelineContext) -> bool:
+        """Returns False to short-circuit (ctx.error set)."""
+        raise NotImplementedError
+
+
+class PreparationStage(Stage):
+    """Tokenize + chat-template render + sampling extraction
+    (reference regular/stages prep + utils/chat_utils.rs)."""
+
+    name = "preparation"
+
+    def __init__(self, app_ctx):
+        self.app = app_ctx
+        self._template: Optional[ChatTemplate] = None
+
+    def template(self) -> ChatTemplate:
+        if self._template is None:
+            from ...tokenizer.chat_template import load_chat_template
+
+            self._template = load_chat_template(self.app.config.chat_template)
+        return self._template
+
+    async def run(self, ctx: PipelineContext) -> bool:
+        body = ctx.req.body or {}
+        ctx.body = body
+        ctx.model_id = self.app.worker_registry.resolve_model(
+            ctx.req.model_override or (body.get("model") if isinstance(body.get("model"), str) else None)
+        )
+        ctx.stream = bool(body.get("stream", False))
+        ctx.request_id = ctx.req.request_id or uuid.uuid4().hex
+
+        # sampling params (OpenAI surface -> engine sampling)
+        sp = api.SamplingParams()
+        sp.max_new_tokens = int(
+            body.get("max_completion_tokens") or body.get("max_tokens") or body.get("max_new_tokens") or 128
+        )
+        sp.temperature = float(body.get("temperature", 1.0))
+        sp.top_p = float(body.get("top_p", 1.0))
+        stop = body.get("stop")
+        if isinstance(stop, str):
+            sp.stop = [stop]
+        elif isinstance(stop, list):
+            sp.stop = [s for s in stop if isinstance(s, str)]
+        ctx.sampling = sp
+
+        # text assembly
+        if ctx.endpoint == "chat":
+            msgs = body.get("messages")
+            if not isinstance(msgs, list) or not msgs:
+                ctx.error = RouteResponse(status=400, body=error_body("'messages' must be a non-empty array"))
+                return False
+            ctx.text = self.template().render(msgs, tools=body.get("tools"))
+        elif ctx.endpoint == "completion":
+            p = body.get("prompt")
+            if isinstance(p, list) and p and all(isinstance(x, int) for x in p):
+                ctx.input_ids = p
+            else:
+                ctx.text = p if isinstance(p, str) else "\n".join(p or [])
+        elif ctx.endpoint == "generate":
+            ids = body.get("input_ids")
+            if isinstance(ids, list) and ids:
+                ctx.input_ids = ids
+            else:
+                ctx.text = body.get("text") or ""
+        elif ctx.endpoint == "embedding":
+            i = body.get("input")
+            ctx.text = i if isinstance(i, str) else "\n".join(x for x in (i or []) if isinstance(x, str))
+
+        # tokenize
+        if not ctx.input_ids:
+            reg = self.app.tokenizer_registry
+            tok = reg.get(ctx.model_id) if reg else None
+            if tok is None:
+                from ...tokenizer.registry import MockTokenizer
+
+                tok = MockTokenizer()
+            ctx.tokenizer = tok
+            ctx.input_ids = tok.encode(ctx.text)
+        else:
+            reg = self.app.tokenizer_registry
+            ctx.tokenizer = reg.get(ctx.model_id) if reg else None
+            if ctx.tokenizer is None:
+                from ...tokenizer.registry import MockTokenizer
+
+                ctx.tokenizer = MockTokenizer()
+        return True
+
+
+class WorkerSelectionStage(Stage):
+    """Policy select on token ids (reference common/stages/worker_selection.rs:41)."""
+
+    name = "worker_selection"
+
+    def __init__(self, app_ctx, metrics=None):
+        self.app = app_ctx
+
+    async def run(self, ctx: PipelineContext) -> bool:
+        workers = self.app.worker_registry.for_model(ctx.model_id)
+        if not workers:
+            ctx.error = RouteResponse(
+                status=503, body=error_body(f"no available worker for model {ctx.model_id or 'any'}", 503)
+            )
+            return False
+        policy = self.app.policy_registry.get(ctx.model_id)
+        info = SelectWorkerInfo(
+            request_id=ctx.request_id,
+            model_id=ctx.model_id,
+            tokens=ctx.input_ids,
+            routing_key=ctx.req.routing_key,
+            tenant_id=ctx.req.tenant_id,
+            est_tokens=len(ctx.input_ids),
+        )
+        t0 = time.perf_counter_ns()
+        idx = policy.select_worker(workers, info)
+        self.app.metrics.observe_routing_latency((time.perf_counter_ns() - t0) / 1e9)
+        if idx is None:
+            ctx.error = RouteResponse(status=503, body=error_body("selection failed", 503))
+            return False
+        ctx.worker = workers[idx]
+        # DP-aware rank routing (reference dp_min_token.rs via monitor.rs:164)
+        if self.app.config.dp_aware and ctx.worker.dp_size:
+            ctx.dp_rank = self.app.policy_registry.dp_policy.select_dp_rank(ctx.worker)
+        return True
+
+
+class ClientAcquisitionStage(Stage):
+    name = "client_acquisition"
+
+    def __init__(self, pool: ClientPool):
+        self.pool = pool
+
+    async def run(self, ctx: PipelineContext) -> bool:
+        ctx.client = self.pool.get(ctx.worker.url)
+        return True
+
+
+class RequestBuildingStage(Stage):
+    name = "request_building"
+
+    async def run(self, ctx: PipelineContext) -> bool:
+        ctx.gen_request = api.GenerateRequest(
+            request_id=ctx.request_id,
+            input_ids=ctx.input_ids,
+            sampling=ctx.sampling,
+            stream=True,
+            dp_rank=ctx.dp_rank,
+        )
+        return True
+
+
+class DispatchMetadataStage(Stage):
+    """Request-id stamping incl. the `_dp{rank}` suffix
+    (reference request_execution.rs:110)."""
+
+    name = "dispatch_metadata"
+
+    async def run(self, ctx: PipelineContext) -> bool:
+        if ctx.dp_rank is not None:
+            ctx.gen_request.request_id = f"{ctx.request_id}_dp{ctx.dp_rank}"
+        return True
+
+
+class RequestExecutionStage(Stage):
+    """Single dispatch (reference request_execution.rs:410; PD dual :253)."""
+
+    name = "request_execution"
+
+    async def run(self, ctx: PipelineContext) -> bool:
+        ctx.worker.incr_load(len(ctx.input_ids))
+        return True
+
+
+class ResponseProcessingStage(Stage):
+    """The per-token streaming tail (reference regular/streaming.rs)."""
+
+    name = "response_processing"
+
+    def __init__(self, app_ctx):
+        self.app = app_ctx
+
+    async def run(self, ctx: PipelineContext) -> bool:
+        return True
+
+    # ---- helpers shared by GrpcRouter -------------------------------------
+    def make_processors(self, ctx: PipelineContext):
+        stop = StopSequenceDecoder(ctx.sampling.stop, ctx.sampling.stop_token_ids)
+        detok = DecodeStream(ctx.tokenizer)
+        rp_name = self.app.config.reasoning_parser
+        reasoning = get_reasoning_parser(rp_name) if rp_name else None
+        tp_name = self.app.config.tool_call_parser
+        tool_stream = StreamingToolParser(get_parser(tp_name)) if tp_name and (ctx.body or {}).get("tools") else None
+        return detok, stop, reasoning, tool_stream
