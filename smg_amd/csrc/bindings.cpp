@@ -29,7 +29,19 @@ int smg_gpu_tree_evict_older(void* p, uint32_t cutoff);
 int smg_gpu_tree_stats(void* p, unsigned long long* out);
 int smg_gpu_tree_clear(void* p);
 int smg_hip_device_count();
+// ---- BPE (bpe.hip) ----
+void* smg_bpe_create(const unsigned long long* pair_keys, const unsigned long long* pair_vals,
+                     uint32_t n_pairs, const uint32_t* byte_to_tok, uint32_t max_pieces,
+                     uint32_t max_bytes, int use_gpu);
+void smg_bpe_destroy(void* p);
+int smg_bpe_on_gpu(void* p);
+int smg_bpe_encode(void* p, const uint8_t* bytes, const uint32_t* offsets, uint32_t n_pieces,
+                   uint32_t* out_tokens, uint32_t* out_counts);
+uint32_t smg_bpe_pretokenize(const uint8_t* s, uint32_t n, uint32_t* piece_offsets, uint32_t max_pieces);
+uint32_t smg_bpe_encode_long(void* p, const uint8_t* bytes, uint32_t len, uint32_t* out_tokens, uint32_t cap);
 }
+
+#define BPE_MAX_PIECE 64
 
 namespace {
 
@@ -131,6 +143,103 @@ uint64_t py_page_hash(py::array_t<uint32_t, py::array::c_style | py::array::forc
     return t.page_hash(toks.data(), (uint32_t)toks.size());
 }
 
+class PyBpe {
+   public:
+    PyBpe(py::array_t<uint64_t, py::array::c_style | py::array::forcecast> pair_keys,
+          py::array_t<uint64_t, py::array::c_style | py::array::forcecast> pair_vals,
+          py::array_t<uint32_t, py::array::c_style | py::array::forcecast> byte_to_tok,
+          uint32_t max_pieces, uint32_t max_bytes, bool use_gpu) {
+        if (byte_to_tok.size() != 256) throw std::runtime_error("byte_to_tok must have 256 entries");
+        h_ = smg_bpe_create((const unsigned long long*)pair_keys.data(),
+                            (const unsigned long long*)pair_vals.data(), (uint32_t)pair_keys.size(),
+                            byte_to_tok.data(), max_pieces, max_bytes, use_gpu ? 1 : 0);
+        if (!h_) throw std::runtime_error("smg_bpe_create failed");
+        max_pieces_ = max_pieces;
+    }
+    ~PyBpe() { smg_bpe_destroy(h_); }
+
+    bool on_gpu() const { return smg_bpe_on_gpu(h_) != 0; }
+
+    // pieces: flattened bytes + offsets -> per-piece token lists.  Pieces
+    // longer than BPE_MAX_PIECE take the uncapped host path; the rest run in
+    // one GPU batch (or the identical host loop without a GPU).
+    py::tuple encode_pieces(py::bytes data,
+                            py::array_t<uint32_t, py::array::c_style | py::array::forcecast> offsets) {
+        std::string buf = data;
+        const uint8_t* bytes = (const uint8_t*)buf.data();
+        uint32_t n_pieces = (uint32_t)offsets.size() - 1;
+        const uint32_t* offs = offsets.data();
+        // split short/long
+        std::vector<uint32_t> short_offsets{0};
+        std::vector<uint8_t> short_bytes;
+        std::vector<int> kind(n_pieces);  // index into short list, or -1-longIdx
+        std::vector<std::vector<uint32_t>> long_tokens;
+        for (uint32_t i = 0; i < n_pieces; ++i) {
+            uint32_t len = offs[i + 1] - offs[i];
+            if (len > BPE_MAX_PIECE) {
+                std::vector<uint32_t> out(len);
+                uint32_t n;
+                {
+                    py::gil_scoped_release nogil;
+                    n = smg_bpe_encode_long(h_, bytes + offs[i], len, out.data(), len);
+                }
+                out.resize(n);
+                kind[i] = -1 - (int)long_tokens.size();
+                long_tokens.push_back(std::move(out));
+            } else {
+                kind[i] = (int)(short_offsets.size() - 1);
+                short_bytes.insert(short_bytes.end(), bytes + offs[i], bytes + offs[i + 1]);
+                short_offsets.push_back((uint32_t)short_bytes.size());
+            }
+        }
+        uint32_t n_short = (uint32_t)short_offsets.size() - 1;
+        std::vector<uint32_t> out_tokens((size_t)n_short * BPE_MAX_PIECE);
+        std::vector<uint32_t> short_counts(n_short);
+        if (n_short) {
+            int rc;
+            {
+                py::gil_scoped_release nogil;
+                rc = smg_bpe_encode(h_, short_bytes.data(), short_offsets.data(), n_short,
+                                    out_tokens.data(), short_counts.data());
+            }
+            if (rc != 0) throw std::runtime_error("smg_bpe_encode rc=" + std::to_string(rc));
+        }
+        // reassemble in original piece order
+        py::array_t<uint32_t> counts(n_pieces);
+        uint32_t total = 0;
+        for (uint32_t i = 0; i < n_pieces; ++i) {
+            uint32_t c = kind[i] >= 0 ? short_counts[kind[i]] : (uint32_t)long_tokens[-1 - kind[i]].size();
+            counts.mutable_data()[i] = c;
+            total += c;
+        }
+        py::array_t<uint32_t> flat(total);
+        uint32_t w = 0;
+        for (uint32_t i = 0; i < n_pieces; ++i) {
+            if (kind[i] >= 0) {
+                uint32_t* src = out_tokens.data() + (size_t)kind[i] * BPE_MAX_PIECE;
+                for (uint32_t j = 0; j < short_counts[kind[i]]; ++j) flat.mutable_data()[w++] = src[j];
+            } else {
+                for (uint32_t v : long_tokens[-1 - kind[i]]) flat.mutable_data()[w++] = v;
+            }
+        }
+        return py::make_tuple(flat, counts);
+    }
+
+    py::array_t<uint32_t> pretokenize(py::bytes data) {
+        std::string buf = data;
+        std::vector<uint32_t> offs(max_pieces_ + 2);
+        uint32_t np = smg_bpe_pretokenize((const uint8_t*)buf.data(), (uint32_t)buf.size(),
+                                          offs.data(), max_pieces_);
+        py::array_t<uint32_t> out(np + 1);
+        std::memcpy(out.mutable_data(), offs.data(), sizeof(uint32_t) * (np + 1));
+        return out;
+    }
+
+   private:
+    void* h_;
+    uint32_t max_pieces_;
+};
+
 }  // namespace
 
 PYBIND11_MODULE(_core, m) {
@@ -162,4 +271,16 @@ PYBIND11_MODULE(_core, m) {
         .def("evict_older", &PyGpuTree::evict_older)
         .def("stats", &PyGpuTree::stats)
         .def("clear", &PyGpuTree::clear);
+
+    py::class_<PyBpe>(m, "Bpe")
+        .def(py::init<py::array_t<uint64_t, py::array::c_style | py::array::forcecast>,
+                      py::array_t<uint64_t, py::array::c_style | py::array::forcecast>,
+                      py::array_t<uint32_t, py::array::c_style | py::array::forcecast>, uint32_t,
+                      uint32_t, bool>(),
+             py::arg("pair_keys"), py::arg("pair_vals"), py::arg("byte_to_tok"),
+             py::arg("max_pieces") = 1u << 20, py::arg("max_bytes") = 1u << 24,
+             py::arg("use_gpu") = true)
+        .def("on_gpu", &PyBpe::on_gpu)
+        .def("encode_pieces", &PyBpe::encode_pieces)
+        .def("pretokenize", &PyBpe::pretokenize);
 }
