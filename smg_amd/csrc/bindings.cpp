@@ -39,6 +39,13 @@ int smg_bpe_encode(void* p, const uint8_t* bytes, const uint32_t* offsets, uint3
                    uint32_t* out_tokens, uint32_t* out_counts);
 uint32_t smg_bpe_pretokenize(const uint8_t* s, uint32_t n, uint32_t* piece_offsets, uint32_t max_pieces);
 uint32_t smg_bpe_encode_long(void* p, const uint8_t* bytes, uint32_t len, uint32_t* out_tokens, uint32_t cap);
+// ---- image (image.hip) ----
+void* smg_img_create(int use_gpu);
+void smg_img_destroy(void* p);
+int smg_img_on_gpu(void* p);
+int smg_img_resize_normalize(void* p, const uint8_t* in, int in_w, int in_h, int channels,
+                             int out_w, int out_h, const float* mean, const float* stddev,
+                             uint8_t* out_u8, float* out_f32);
 }
 
 #define BPE_MAX_PIECE 64
@@ -240,6 +247,56 @@ class PyBpe {
     uint32_t max_pieces_;
 };
 
+class PyImg {
+   public:
+    explicit PyImg(bool use_gpu) { h_ = smg_img_create(use_gpu ? 1 : 0); }
+    ~PyImg() { smg_img_destroy(h_); }
+    bool on_gpu() const { return smg_img_on_gpu(h_) != 0; }
+
+    // image: u8 HWC array (H, W, C); returns (u8 HWC resized | None,
+    // f32 CHW normalized | None)
+    py::tuple resize_normalize(py::array_t<uint8_t, py::array::c_style | py::array::forcecast> image,
+                               int out_w, int out_h, py::object mean, py::object std,
+                               bool want_u8, bool want_f32) {
+        if (image.ndim() != 3) throw std::runtime_error("image must be HWC u8");
+        int in_h = (int)image.shape(0), in_w = (int)image.shape(1), ch = (int)image.shape(2);
+        if (ch < 1 || ch > 4) throw std::runtime_error("1..4 channels supported");
+        std::vector<float> mv(ch, 0.f), sv(ch, 1.f);
+        if (!mean.is_none()) {
+            auto m = mean.cast<std::vector<float>>();
+            for (int i = 0; i < ch && i < (int)m.size(); ++i) mv[i] = m[i];
+        }
+        if (!std.is_none()) {
+            auto s = std.cast<std::vector<float>>();
+            for (int i = 0; i < ch && i < (int)s.size(); ++i) sv[i] = s[i];
+        }
+        py::array_t<uint8_t> out_u8;
+        py::array_t<float> out_f32;
+        uint8_t* pu8 = nullptr;
+        float* pf32 = nullptr;
+        if (want_u8) {
+            out_u8 = py::array_t<uint8_t>({out_h, out_w, ch});
+            pu8 = out_u8.mutable_data();
+        }
+        if (want_f32) {
+            out_f32 = py::array_t<float>({ch, out_h, out_w});
+            pf32 = out_f32.mutable_data();
+        }
+        int rc;
+        {
+            py::gil_scoped_release nogil;
+            rc = smg_img_resize_normalize(h_, image.data(), in_w, in_h, ch, out_w, out_h,
+                                          mv.data(), sv.data(), pu8, pf32);
+        }
+        if (rc != 0) throw std::runtime_error("resize failed rc=" + std::to_string(rc));
+        return py::make_tuple(want_u8 ? py::object(out_u8) : py::none(),
+                              want_f32 ? py::object(out_f32) : py::none());
+    }
+
+   private:
+    void* h_;
+};
+
 }  // namespace
 
 PYBIND11_MODULE(_core, m) {
@@ -283,4 +340,11 @@ PYBIND11_MODULE(_core, m) {
         .def("on_gpu", &PyBpe::on_gpu)
         .def("encode_pieces", &PyBpe::encode_pieces)
         .def("pretokenize", &PyBpe::pretokenize);
+
+    py::class_<PyImg>(m, "ImageProcessor")
+        .def(py::init<bool>(), py::arg("use_gpu") = true)
+        .def("on_gpu", &PyImg::on_gpu)
+        .def("resize_normalize", &PyImg::resize_normalize, py::arg("image"), py::arg("out_w"),
+             py::arg("out_h"), py::arg("mean") = py::none(), py::arg("std") = py::none(),
+             py::arg("want_u8") = true, py::arg("want_f32") = true);
 }

@@ -1,0 +1,283 @@
+// Multimodal image preprocessing on gfx950: Pillow-exact bicubic resize +
+// fused to-tensor/normalize.
+//
+// Re-designs the reference's SIMD image path (crates/multimodal/src/vision/
+// transforms.rs — Pillow-exact bicubic :432-882, fused
+// to_tensor_and_normalize :315) as HIP kernels:
+//
+//   pass 1: horizontal resample  u8 HWC -> u8 temp   (out_w x in_h)
+//   pass 2: vertical resample + normalize  -> f32 CHW (and u8 HWC view)
+//
+// "Pillow-exact" means bit-identical u8 output to PIL Image.resize(BICUBIC):
+// the same separable convolution with a=-0.5 bicubic weights, coefficients
+// quantized to 1<<PRECISION_BITS fixed point, int32 accumulation and
+// round-half-away >> shift with clip-to-u8 — the documented/observed PIL
+// algorithm, required for model-accuracy parity with HF processors.
+// Normalization then applies (x/255 - mean)/std per channel INTO the same
+// kernel so the resized u8 never round-trips through HBM twice.
+//
+// Host precomputes per-output coefficient tables (double precision); the
+// identical resample code runs host-side (HOSTDEV) for the CPU fallback and
+// the PIL differential tests.
+#include <hip/hip_runtime.h>
+
+#include <cmath>
+#include <cstdint>
+#include <cstring>
+#include <vector>
+
+#define IMG_PRECISION_BITS (32 - 8 - 2)  // PIL's 8bpc fixed-point scale
+
+// ---------------------------------------------------------------------------
+// coefficient precompute (host)
+// ---------------------------------------------------------------------------
+static inline double bicubic_weight(double x) {
+    // PIL bicubic filter, a = -0.5, support 2.0
+    const double a = -0.5;
+    if (x < 0) x = -x;
+    if (x < 1.0) return ((a + 2.0) * x - (a + 3.0)) * x * x + 1.0;
+    if (x < 2.0) return (((x - 5.0) * x + 8.0) * x - 4.0) * a;
+    return 0.0;
+}
+
+struct ResampleCoeffs {
+    std::vector<int> bounds;  // per out pixel: xmin, xsize
+    std::vector<int> coeffs;  // per out pixel: ksize ints (1<<PRECISION_BITS fixed point)
+    int ksize = 0;
+};
+
+static ResampleCoeffs precompute_coeffs(int in_size, int out_size) {
+    ResampleCoeffs rc;
+    double scale = (double)in_size / out_size;
+    double filterscale = scale < 1.0 ? 1.0 : scale;
+    double support = 2.0 * filterscale;  // bicubic support
+    int ksize = (int)ceil(support) * 2 + 1;
+    rc.ksize = ksize;
+    rc.bounds.resize(out_size * 2);
+    rc.coeffs.resize((size_t)out_size * ksize, 0);
+    std::vector<double> k(ksize);
+    for (int xx = 0; xx < out_size; ++xx) {
+        double center = (xx + 0.5) * scale;
+        double ww = 0.0;
+        double ss = 1.0 / filterscale;
+        int xmin = (int)(center - support + 0.5);
+        if (xmin < 0) xmin = 0;
+        int xmax = (int)(center + support + 0.5);
+        if (xmax > in_size) xmax = in_size;
+        xmax -= xmin;
+        for (int x = 0; x < xmax; ++x) {
+            double w = bicubic_weight((x + xmin - center + 0.5) * ss);
+            k[x] = w;
+            ww += w;
+        }
+        for (int x = 0; x < xmax; ++x)
+            if (ww != 0.0) k[x] /= ww;
+        rc.bounds[xx * 2] = xmin;
+        rc.bounds[xx * 2 + 1] = xmax;
+        for (int x = 0; x < xmax; ++x) {
+            double v = k[x] * (1 << IMG_PRECISION_BITS);
+            rc.coeffs[(size_t)xx * ksize + x] = (int)(v < 0 ? v - 0.5 : v + 0.5);
+        }
+    }
+    return rc;
+}
+
+__host__ __device__ static inline uint8_t clip8(int v) {
+    v >>= IMG_PRECISION_BITS;
+    return v < 0 ? 0 : (v > 255 ? 255 : (uint8_t)v);
+}
+
+// ---------------------------------------------------------------------------
+// kernels
+// ---------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(256) smg_img_resize_h(
+    const uint8_t* in, int in_w, int in_h, int channels,
+    uint8_t* out, int out_w,
+    const int* bounds, const int* coeffs, int ksize) {
+    int idx = blockIdx.x * blockDim.x + threadIdx.x;
+    int total = out_w * in_h;
+    if (idx >= total) return;
+    int xx = idx % out_w;
+    int y = idx / out_w;
+    int xmin = bounds[xx * 2], xmax = bounds[xx * 2 + 1];
+    const int* k = coeffs + (size_t)xx * ksize;
+    const int half = 1 << (IMG_PRECISION_BITS - 1);
+    for (int c = 0; c < channels; ++c) {
+        int ss = half;
+        for (int x = 0; x < xmax; ++x)
+            ss += (int)in[((size_t)y * in_w + (xmin + x)) * channels + c] * k[x];
+        out[((size_t)y * out_w + xx) * channels + c] = clip8(ss);
+    }
+}
+
+extern "C" __global__ void __launch_bounds__(256) smg_img_resize_v_norm(
+    const uint8_t* in, int in_w, int in_h, int channels,
+    uint8_t* out_u8, int out_h,
+    float* out_f32,  // CHW normalized (nullable)
+    const float* mean, const float* stdinv,
+    const int* bounds, const int* coeffs, int ksize) {
+    int idx = blockIdx.x * blockDim.x + threadIdx.x;
+    int total = in_w * out_h;
+    if (idx >= total) return;
+    int x = idx % in_w;
+    int yy = idx / in_w;
+    int ymin = bounds[yy * 2], ymax = bounds[yy * 2 + 1];
+    const int* k = coeffs + (size_t)yy * ksize;
+    const int half = 1 << (IMG_PRECISION_BITS - 1);
+    for (int c = 0; c < channels; ++c) {
+        int ss = half;
+        for (int y = 0; y < ymax; ++y)
+            ss += (int)in[((size_t)(ymin + y) * in_w + x) * channels + c] * k[y];
+        uint8_t px = clip8(ss);
+        if (out_u8) out_u8[((size_t)yy * in_w + x) * channels + c] = px;
+        if (out_f32)
+            out_f32[(size_t)c * out_h * in_w + (size_t)yy * in_w + x] =
+                ((float)px / 255.0f - mean[c]) * stdinv[c];
+    }
+}
+
+// ---------------------------------------------------------------------------
+// host orchestration (+ identical CPU path)
+// ---------------------------------------------------------------------------
+struct ImgHost {
+    bool on_gpu = false;
+    hipStream_t stream{};
+    uint8_t* d_in = nullptr;
+    uint8_t* d_tmp = nullptr;
+    uint8_t* d_out = nullptr;
+    float* d_outf = nullptr;
+    int* d_coef = nullptr;
+    float* d_norm = nullptr;  // mean[4] + stdinv[4]
+    size_t cap_in = 0, cap_tmp = 0, cap_out = 0, cap_coef = 0;
+};
+
+static bool ensure(void** p, size_t* cap, size_t need, hipStream_t) {
+    if (*cap >= need) return true;
+    if (*p) hipFree(*p);
+    if (hipMalloc(p, need) != hipSuccess) { *p = nullptr; *cap = 0; return false; }
+    *cap = need;
+    return true;
+}
+
+extern "C" void* smg_img_create(int use_gpu) {
+    ImgHost* h = new ImgHost();
+    if (use_gpu) {
+        int n = 0;
+        if (hipGetDeviceCount(&n) == hipSuccess && n > 0) {
+            if (hipStreamCreate(&h->stream) == hipSuccess) {
+                if (hipMalloc(&h->d_norm, sizeof(float) * 8) == hipSuccess) h->on_gpu = true;
+            }
+        }
+    }
+    return h;
+}
+
+extern "C" void smg_img_destroy(void* p) {
+    ImgHost* h = (ImgHost*)p;
+    if (!h) return;
+    if (h->on_gpu) {
+        hipFree(h->d_in); hipFree(h->d_tmp); hipFree(h->d_out); hipFree(h->d_outf);
+        hipFree(h->d_coef); hipFree(h->d_norm);
+        hipStreamDestroy(h->stream);
+    }
+    delete h;
+}
+
+extern "C" int smg_img_on_gpu(void* p) { return ((ImgHost*)p)->on_gpu ? 1 : 0; }
+
+// Pillow-exact bicubic resize of an RGB(A) u8 HWC image + fused normalize.
+// out_u8 (HWC) and/or out_f32 (CHW) may be null.  Returns 0 on success.
+extern "C" int smg_img_resize_normalize(void* p, const uint8_t* in, int in_w, int in_h,
+                                        int channels, int out_w, int out_h,
+                                        const float* mean, const float* stddev,
+                                        uint8_t* out_u8, float* out_f32) {
+    ImgHost* h = (ImgHost*)p;
+    ResampleCoeffs ch = precompute_coeffs(in_w, out_w);
+    ResampleCoeffs cv = precompute_coeffs(in_h, out_h);
+    float stdinv[4] = {1, 1, 1, 1};
+    float mean4[4] = {0, 0, 0, 0};
+    for (int c = 0; c < channels && c < 4; ++c) {
+        stdinv[c] = 1.0f / (stddev ? stddev[c] : 1.0f);
+        mean4[c] = mean ? mean[c] : 0.0f;
+    }
+    size_t in_bytes = (size_t)in_w * in_h * channels;
+    size_t tmp_bytes = (size_t)out_w * in_h * channels;
+    size_t out_bytes = (size_t)out_w * out_h * channels;
+
+    if (h->on_gpu) {
+        // pack coefficient tables: [h bounds][h coeffs][v bounds][v coeffs]
+        size_t nh_b = ch.bounds.size(), nh_c = ch.coeffs.size();
+        size_t nv_b = cv.bounds.size(), nv_c = cv.coeffs.size();
+        size_t coef_bytes = (nh_b + nh_c + nv_b + nv_c) * sizeof(int);
+        bool ok = ensure((void**)&h->d_in, &h->cap_in, in_bytes, h->stream) &&
+                  ensure((void**)&h->d_tmp, &h->cap_tmp, tmp_bytes, h->stream) &&
+                  ensure((void**)&h->d_out, &h->cap_out, out_bytes * (1 + 4 * sizeof(float)), h->stream) &&
+                  ensure((void**)&h->d_coef, &h->cap_coef, coef_bytes, h->stream);
+        if (!ok) return -1;
+        h->d_outf = (float*)(h->d_out + out_bytes);
+        std::vector<int> packed;
+        packed.reserve(nh_b + nh_c + nv_b + nv_c);
+        packed.insert(packed.end(), ch.bounds.begin(), ch.bounds.end());
+        packed.insert(packed.end(), ch.coeffs.begin(), ch.coeffs.end());
+        packed.insert(packed.end(), cv.bounds.begin(), cv.bounds.end());
+        packed.insert(packed.end(), cv.coeffs.begin(), cv.coeffs.end());
+        hipMemcpyAsync(h->d_coef, packed.data(), coef_bytes, hipMemcpyHostToDevice, h->stream);
+        hipMemcpyAsync(h->d_in, in, in_bytes, hipMemcpyHostToDevice, h->stream);
+        float norm[8];
+        memcpy(norm, mean4, sizeof(mean4));
+        memcpy(norm + 4, stdinv, sizeof(stdinv));
+        hipMemcpyAsync(h->d_norm, norm, sizeof(norm), hipMemcpyHostToDevice, h->stream);
+        int* db_h = h->d_coef;
+        int* dc_h = db_h + nh_b;
+        int* db_v = dc_h + nh_c;
+        int* dc_v = db_v + nv_b;
+        int threads = 256;
+        int total1 = out_w * in_h;
+        hipLaunchKernelGGL(smg_img_resize_h, dim3((total1 + threads - 1) / threads), dim3(threads),
+                           0, h->stream, h->d_in, in_w, in_h, channels, h->d_tmp, out_w, db_h, dc_h,
+                           ch.ksize);
+        int total2 = out_w * out_h;
+        hipLaunchKernelGGL(smg_img_resize_v_norm, dim3((total2 + threads - 1) / threads),
+                           dim3(threads), 0, h->stream, h->d_tmp, out_w, in_h, channels,
+                           out_u8 ? h->d_out : nullptr, out_h, out_f32 ? h->d_outf : nullptr,
+                           h->d_norm, h->d_norm + 4, db_v, dc_v, cv.ksize);
+        if (out_u8)
+            hipMemcpyAsync(out_u8, h->d_out, out_bytes, hipMemcpyDeviceToHost, h->stream);
+        if (out_f32)
+            hipMemcpyAsync(out_f32, h->d_outf, out_bytes * sizeof(float), hipMemcpyDeviceToHost, h->stream);
+        return hipStreamSynchronize(h->stream) == hipSuccess ? 0 : -2;
+    }
+
+    // CPU path: identical arithmetic
+    std::vector<uint8_t> tmp(tmp_bytes);
+    const int half = 1 << (IMG_PRECISION_BITS - 1);
+    for (int y = 0; y < in_h; ++y) {
+        for (int xx = 0; xx < out_w; ++xx) {
+            int xmin = ch.bounds[xx * 2], xmax = ch.bounds[xx * 2 + 1];
+            const int* k = ch.coeffs.data() + (size_t)xx * ch.ksize;
+            for (int c = 0; c < channels; ++c) {
+                int ss = half;
+                for (int x = 0; x < xmax; ++x)
+                    ss += (int)in[((size_t)y * in_w + (xmin + x)) * channels + c] * k[x];
+                tmp[((size_t)y * out_w + xx) * channels + c] = clip8(ss);
+            }
+        }
+    }
+    for (int yy = 0; yy < out_h; ++yy) {
+        int ymin = cv.bounds[yy * 2], ymax = cv.bounds[yy * 2 + 1];
+        const int* k = cv.coeffs.data() + (size_t)yy * cv.ksize;
+        for (int x = 0; x < out_w; ++x) {
+            for (int c = 0; c < channels; ++c) {
+                int ss = half;
+                for (int y = 0; y < ymax; ++y)
+                    ss += (int)tmp[((size_t)(ymin + y) * out_w + x) * channels + c] * k[y];
+                uint8_t px = clip8(ss);
+                if (out_u8) out_u8[((size_t)yy * out_w + x) * channels + c] = px;
+                if (out_f32)
+                    out_f32[(size_t)c * out_h * out_w + (size_t)yy * out_w + x] =
+                        ((float)px / 255.0f - mean4[c]) * stdinv[c];
+            }
+        }
+    }
+    return 0;
+}
