@@ -209,12 +209,13 @@ extern "C" int smg_img_resize_normalize(void* p, const uint8_t* in, int in_w, in
         size_t nh_b = ch.bounds.size(), nh_c = ch.coeffs.size();
         size_t nv_b = cv.bounds.size(), nv_c = cv.coeffs.size();
         size_t coef_bytes = (nh_b + nh_c + nv_b + nv_c) * sizeof(int);
+        size_t out_u8_pad = (out_bytes + 255) & ~(size_t)255;  // keep the f32 view aligned
         bool ok = ensure((void**)&h->d_in, &h->cap_in, in_bytes, h->stream) &&
                   ensure((void**)&h->d_tmp, &h->cap_tmp, tmp_bytes, h->stream) &&
-                  ensure((void**)&h->d_out, &h->cap_out, out_bytes * (1 + 4 * sizeof(float)), h->stream) &&
+                  ensure((void**)&h->d_out, &h->cap_out, out_u8_pad + out_bytes * sizeof(float), h->stream) &&
                   ensure((void**)&h->d_coef, &h->cap_coef, coef_bytes, h->stream);
         if (!ok) return -1;
-        h->d_outf = (float*)(h->d_out + out_bytes);
+        h->d_outf = (float*)(h->d_out + out_u8_pad);
         std::vector<int> packed;
         packed.reserve(nh_b + nh_c + nv_b + nv_c);
         packed.insert(packed.end(), ch.bounds.begin(), ch.bounds.end());
