@@ -19,11 +19,20 @@ from ..workers.worker import Worker, WorkerType
 
 class AppContext:
     def __init__(self, config: RouterConfig, metrics: Optional[GatewayMetrics] = None):
+        from ..config import ConnectionMode
+        from ..kvindex.event_index import PositionalIndexer
+
         self.config = config
         self.metrics = metrics or GatewayMetrics()
         self.worker_registry = WorkerRegistry()
+        indexer = (
+            PositionalIndexer(block_size=config.policy.block_size)
+            if config.connection_mode == ConnectionMode.GRPC
+            else None
+        )
         self.policy_registry = PolicyRegistry(
             config.policy,
+            indexer=indexer,
             prefill_cfg=config.prefill_policy,
             decode_cfg=config.decode_policy,
             encode_cfg=config.encode_policy,
@@ -90,9 +99,16 @@ class AppContext:
 
     async def start_background(self) -> None:
         await self.worker_monitor.start()
+        if self.policy_registry.indexer is not None:
+            from ..workers.kv_event_monitor import KvEventMonitor
+
+            self.kv_event_monitor = KvEventMonitor(self.worker_registry, self.policy_registry.indexer)
+            await self.kv_event_monitor.start()
 
     async def shutdown(self) -> None:
         await self.worker_monitor.stop()
+        if self.kv_event_monitor is not None:
+            await self.kv_event_monitor.stop()
         for t in self._background:
             t.cancel()
         if self.router_manager is not None:
