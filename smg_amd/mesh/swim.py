@@ -1,0 +1,252 @@
+"""SWIM-style gossip membership + sync rounds (reference: crates/mesh —
+GossipController::event_loop 1 Hz (gossip_controller.rs:1-26): SWIM probe
+Ping -> indirect PingReq -> suspect -> dead; round collection drains local
+CRDT ops into a RoundBatch shipped to every live peer; partition detection
+(partition.rs)).
+
+Transport: HTTP POST between gateways (aiohttp) — the mesh is the low-rate
+inter-gateway control plane (stays TCP).
+"""
+from __future__ import annotations
+
+import asyncio
+import logging
+import random
+import time
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional
+
+import aiohttp
+
+from .crdt import MeshKV, Op
+
+log = logging.getLogger("smg.mesh")
+
+ALIVE, SUSPECT, DEAD = "alive", "suspect", "dead"
+
+
+@dataclass
+class Member:
+    node_id: str
+    url: str  # base http url of the peer's mesh endpoint
+    state: str = ALIVE
+    incarnation: int = 0
+    last_ack: float = field(default_factory=time.monotonic)
+    suspect_since: Optional[float] = None
+
+
+class MeshNode:
+    def __init__(
+        self,
+        node_id: str,
+        advertise_url: str,
+        kv: Optional[MeshKV] = None,
+        probe_interval: float = 1.0,
+        probe_timeout: float = 0.5,
+        suspect_timeout: float = 3.0,
+        indirect_k: int = 2,
+    ):
+        self.node_id = node_id
+        self.advertise_url = advertise_url.rstrip("/")
+        self.kv = kv or MeshKV(node_id)
+        self.members: Dict[str, Member] = {}
+        self.probe_interval = probe_interval
+        self.probe_timeout = probe_timeout
+        self.suspect_timeout = suspect_timeout
+        self.indirect_k = indirect_k
+        self._session: Optional[aiohttp.ClientSession] = None
+        self._task: Optional[asyncio.Task] = None
+        self._stopped = asyncio.Event()
+        self._sent_watermarks: Dict[str, int] = {}  # peer -> last local seq sent
+        self.incarnation = 0
+
+    # ---- lifecycle ---------------------------------------------------------
+    async def start(self, peer_urls: List[str]) -> None:
+        self._stopped.clear()
+        self._session = aiohttp.ClientSession(timeout=aiohttp.ClientTimeout(total=5))
+        for url in peer_urls:
+            await self.join(url)
+        self._task = asyncio.ensure_future(self._event_loop())
+
+    async def stop(self) -> None:
+        self._stopped.set()
+        if self._task:
+            self._task.cancel()
+            try:
+                await self._task
+            except (asyncio.CancelledError, Exception):
+                pass
+        if self._session:
+            await self._session.close()
+
+    async def join(self, peer_url: str) -> bool:
+        """Contact a seed peer: exchange membership + full CRDT snapshot."""
+        peer_url = peer_url.rstrip("/")
+        try:
+            async with self._session.post(
+                peer_url + "/mesh/join",
+                json={
+                    "node_id": self.node_id,
+                    "url": self.advertise_url,
+                    "members": self._member_dicts(),
+                    "snapshot": [op.to_dict() for op in self.kv.snapshot_ops()],
+                },
+            ) as resp:
+                if resp.status != 200:
+                    return False
+                data = await resp.json()
+        except Exception as exc:
+            log.warning("mesh join %s failed: %s", peer_url, exc)
+            return False
+        self._merge_members(data.get("members", []))
+        for opd in data.get("snapshot", []):
+            op = Op.from_dict(opd)
+            self.kv.apply_remote(op, op.hlc[2])
+        return True
+
+    # ---- event loop (1 Hz rounds) ------------------------------------------
+    async def _event_loop(self) -> None:
+        while not self._stopped.is_set():
+            try:
+                await self._round()
+            except asyncio.CancelledError:
+                return
+            except Exception as exc:
+                log.warning("mesh round error: %s", exc)
+            try:
+                await asyncio.wait_for(self._stopped.wait(), self.probe_interval)
+                return
+            except asyncio.TimeoutError:
+                pass
+
+    async def _round(self) -> None:
+        now = time.monotonic()
+        live = [m for m in self.members.values() if m.state != DEAD]
+        # SWIM probe: one random member per round
+        if live:
+            target = random.choice(live)
+            ok = await self._ping(target)
+            if not ok:
+                ok = await self._indirect_ping(target, live)
+            if ok:
+                target.state = ALIVE
+                target.last_ack = now
+                target.suspect_since = None
+            else:
+                if target.state == ALIVE:
+                    target.state = SUSPECT
+                    target.suspect_since = now
+                elif target.state == SUSPECT and now - (target.suspect_since or now) > self.suspect_timeout:
+                    target.state = DEAD
+                    log.info("mesh member %s declared dead", target.node_id)
+        # sync round: ship new ops to every live peer
+        await asyncio.gather(*(self._sync_peer(m) for m in self.members.values() if m.state != DEAD))
+
+    async def _ping(self, m: Member) -> bool:
+        try:
+            async with self._session.post(
+                m.url + "/mesh/ping",
+                json={"from": self.node_id, "members": self._member_dicts()},
+                timeout=aiohttp.ClientTimeout(total=self.probe_timeout),
+            ) as resp:
+                if resp.status == 200:
+                    data = await resp.json()
+                    self._merge_members(data.get("members", []))
+                    return True
+        except Exception:
+            pass
+        return False
+
+    async def _indirect_ping(self, target: Member, live: List[Member]) -> bool:
+        helpers = [m for m in live if m.node_id != target.node_id]
+        random.shuffle(helpers)
+        for h in helpers[: self.indirect_k]:
+            try:
+                async with self._session.post(
+                    h.url + "/mesh/ping_req",
+                    json={"from": self.node_id, "target": target.url},
+                    timeout=aiohttp.ClientTimeout(total=self.probe_timeout * 2),
+                ) as resp:
+                    if resp.status == 200 and (await resp.json()).get("ack"):
+                        return True
+            except Exception:
+                continue
+        return False
+
+    async def _sync_peer(self, m: Member) -> None:
+        since = self._sent_watermarks.get(m.node_id, 0)
+        ops = self.kv.ops_since(since)
+        if not ops:
+            return
+        try:
+            async with self._session.post(
+                m.url + "/mesh/sync",
+                json={"from": self.node_id, "ops": [op.to_dict() for op in ops]},
+            ) as resp:
+                if resp.status == 200:
+                    self._sent_watermarks[m.node_id] = ops[-1].seq
+        except Exception:
+            pass
+
+    # ---- membership helpers -------------------------------------------------
+    def _member_dicts(self) -> List[dict]:
+        out = [{"node_id": self.node_id, "url": self.advertise_url, "state": ALIVE, "incarnation": self.incarnation}]
+        for m in self.members.values():
+            out.append({"node_id": m.node_id, "url": m.url, "state": m.state, "incarnation": m.incarnation})
+        return out
+
+    def _merge_members(self, dicts: List[dict]) -> None:
+        for d in dicts:
+            nid = d.get("node_id")
+            if not nid or nid == self.node_id:
+                continue
+            m = self.members.get(nid)
+            if m is None:
+                self.members[nid] = Member(nid, d["url"], d.get("state", ALIVE), d.get("incarnation", 0))
+            else:
+                if d.get("incarnation", 0) > m.incarnation:
+                    m.incarnation = d["incarnation"]
+                    m.state = d.get("state", ALIVE)
+                elif d.get("state") == DEAD and m.state != DEAD:
+                    pass  # rumors of death need a newer incarnation; keep probing
+
+    # ---- inbound handlers (wired by mesh/server.py) -------------------------
+    def handle_ping(self, payload: dict) -> dict:
+        self._merge_members(payload.get("members", []))
+        return {"ack": True, "members": self._member_dicts()}
+
+    async def handle_ping_req(self, payload: dict) -> dict:
+        target_url = payload.get("target", "")
+        try:
+            async with self._session.post(
+                target_url + "/mesh/ping",
+                json={"from": self.node_id, "members": []},
+                timeout=aiohttp.ClientTimeout(total=self.probe_timeout),
+            ) as resp:
+                return {"ack": resp.status == 200}
+        except Exception:
+            return {"ack": False}
+
+    def handle_sync(self, payload: dict) -> dict:
+        from_node = payload.get("from", "?")
+        applied = 0
+        for opd in payload.get("ops", []):
+            op = Op.from_dict(opd)
+            if self.kv.apply_remote(op, from_node):
+                applied += 1
+        return {"applied": applied, "watermark": self.kv.watermarks.get(from_node, 0)}
+
+    def handle_join(self, payload: dict) -> dict:
+        self._merge_members(
+            payload.get("members", []) + [{"node_id": payload["node_id"], "url": payload["url"], "state": ALIVE}]
+        )
+        for opd in payload.get("snapshot", []):
+            op = Op.from_dict(opd)
+            self.kv.apply_remote(op, op.hlc[2])
+        return {
+            "members": self._member_dicts(),
+            "snapshot": [op.to_dict() for op in self.kv.snapshot_ops()],
+        }
+
+    def live_members(self) -> List[Member]:
+        return [m for m in self.members.values() if m.state == ALIVE]

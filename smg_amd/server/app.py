@@ -392,6 +392,18 @@ async def startup(config: RouterConfig, serve: bool = True) -> AppContext:
     if config.rate_limit.max_concurrent_requests > 0:
         ctx.rate_limiter = ConcurrencyLimiter(config.rate_limit)
     ctx.router_manager = RouterManager(ctx, config)
+    if config.mesh.enabled:
+        from ..mesh.adapters import MeshAdapters
+        from ..mesh.server import start_mesh_server
+        from ..mesh.swim import MeshNode
+
+        node_id = config.mesh.server_name or f"{config.host}:{config.port}"
+        advertise = f"http://{config.mesh.advertise_host or '127.0.0.1'}:{config.mesh.port}"
+        ctx.mesh = MeshNode(node_id, advertise)
+        ctx.mesh_adapters = MeshAdapters(ctx.mesh, ctx)
+        if serve:
+            ctx._mesh_runner = await start_mesh_server(ctx.mesh, config.mesh.host, config.mesh.port)
+        await ctx.mesh.start(config.mesh.peer_urls)
     await ctx.start_background()
     if serve:
         app = build_app(ctx)

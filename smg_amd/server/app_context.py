@@ -109,6 +109,11 @@ class AppContext:
         await self.worker_monitor.stop()
         if self.kv_event_monitor is not None:
             await self.kv_event_monitor.stop()
+        if self.mesh is not None:
+            await self.mesh.stop()
+            runner = getattr(self, "_mesh_runner", None)
+            if runner is not None:
+                await runner.cleanup()
         for t in self._background:
             t.cancel()
         if self.router_manager is not None:

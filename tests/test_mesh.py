@@ -1,0 +1,178 @@
+"""HA mesh tests: SWIM membership, CRDT sync, worker/rate-limit adapters
+(reference: crates/mesh + src/mesh/adapters e2e behavior)."""
+import asyncio
+
+import pytest
+from aiohttp import web
+from aiohttp.test_utils import TestServer
+
+from smg_amd.config import PolicyConfig, RouterConfig
+from smg_amd.mesh.adapters import MeshAdapters, RateLimitSyncAdapter, WorkerSyncAdapter
+from smg_amd.mesh.crdt import MeshKV, Op, epoch_max_wins_merge, lww_merge
+from smg_amd.mesh.server import add_mesh_routes
+from smg_amd.mesh.swim import ALIVE, DEAD, MeshNode
+from smg_amd.server.app_context import AppContext
+from smg_amd.workers.worker import Worker
+
+
+class TestCrdt:
+    def test_lww(self):
+        kv = MeshKV("a")
+        kv.register_namespace("x")
+        op0 = kv.put("x", "k", 1)
+        newer = Op("x", "k", 2, (op0.hlc[0] + 10, 0, "b"), 1)
+        assert kv.apply_remote(newer, "b")
+        assert kv.get("x", "k") == 2
+        older = Op("x", "k", 3, (1, 0, "c"), 1)
+        assert not kv.apply_remote(older, "c")
+        assert kv.get("x", "k") == 2
+
+    def test_epoch_max_wins(self):
+        kv = MeshKV("a")
+        kv.register_namespace("rl", epoch_max_wins_merge)
+        kv.put("rl", "t1:a", {"epoch": 1, "used": 5})
+        assert kv.apply_remote(Op("rl", "t1:a", {"epoch": 1, "used": 9}, (0, 0, "b"), 1), "b")
+        assert kv.get("rl", "t1:a")["used"] == 9
+        assert not kv.apply_remote(Op("rl", "t1:a", {"epoch": 1, "used": 3}, (99, 0, "c"), 2), "c")
+        assert kv.apply_remote(Op("rl", "t1:a", {"epoch": 2, "used": 1}, (0, 0, "d"), 3), "d")
+        assert kv.get("rl", "t1:a")["epoch"] == 2
+
+    def test_ops_since_watermark(self):
+        kv = MeshKV("a")
+        kv.register_namespace("x")
+        kv.put("x", "k1", 1)
+        kv.put("x", "k2", 2)
+        ops = kv.ops_since(0)
+        assert len(ops) == 2
+        assert kv.ops_since(ops[-1].seq) == []
+
+
+async def make_node(name, peers=()):
+    mesh = MeshNode(name, "placeholder", probe_interval=0.1, probe_timeout=0.3, suspect_timeout=0.5)
+    app = web.Application()
+    add_mesh_routes(app, mesh)
+    server = TestServer(app)
+    await server.start_server()
+    mesh.advertise_url = f"http://127.0.0.1:{server.port}"
+    await mesh.start([p.advertise_url for p in peers])
+    return mesh, server
+
+
+def test_membership_and_sync(runner):
+    async def run():
+        a, sa = await make_node("a")
+        b, sb = await make_node("b", peers=[a])
+        try:
+            await asyncio.sleep(0.3)
+            assert "b" in a.members and a.members["b"].state == ALIVE
+            assert "a" in b.members
+            # CRDT propagation a -> b
+            a.kv.register_namespace("x")
+            b.kv.register_namespace("x")
+            a.kv.put("x", "hello", {"v": 42})
+            for _ in range(30):
+                if b.kv.get("x", "hello"):
+                    break
+                await asyncio.sleep(0.1)
+            assert b.kv.get("x", "hello") == {"v": 42}
+        finally:
+            await a.stop()
+            await b.stop()
+            await sa.close()
+            await sb.close()
+
+    runner(run())
+
+
+def test_failure_detection(runner):
+    async def run():
+        a, sa = await make_node("a")
+        b, sb = await make_node("b", peers=[a])
+        try:
+            await asyncio.sleep(0.3)
+            await b.stop()
+            await sb.close()
+            for _ in range(60):
+                if a.members.get("b") and a.members["b"].state == DEAD:
+                    break
+                await asyncio.sleep(0.1)
+            assert a.members["b"].state == DEAD
+        finally:
+            await a.stop()
+            await sa.close()
+
+    runner(run())
+
+
+def test_worker_sync_adapter(runner):
+    async def run():
+        a, sa = await make_node("a")
+        b, sb = await make_node("b", peers=[a])
+        cfg = RouterConfig(policy=PolicyConfig(name="round_robin", gpu_tree=False))
+        ctx_a, ctx_b = AppContext(cfg), AppContext(cfg)
+        MeshAdapters(a, ctx_a)
+        MeshAdapters(b, ctx_b)
+        try:
+            ctx_a.worker_registry.register(Worker("http://wX:9000", model_id="m"))
+            for _ in range(30):
+                if ctx_b.worker_registry.get_by_url("http://wX:9000"):
+                    break
+                await asyncio.sleep(0.1)
+            imported = ctx_b.worker_registry.get_by_url("http://wX:9000")
+            assert imported is not None
+            assert imported.labels.get("mesh_origin") == "a"
+            # removal propagates
+            ctx_a.worker_registry.remove_by_url("http://wX:9000")
+            for _ in range(30):
+                if not ctx_b.worker_registry.get_by_url("http://wX:9000"):
+                    break
+                await asyncio.sleep(0.1)
+            assert ctx_b.worker_registry.get_by_url("http://wX:9000") is None
+        finally:
+            await a.stop()
+            await b.stop()
+            await sa.close()
+            await sb.close()
+
+    runner(run())
+
+
+def test_join_snapshot_bootstraps_late_node(runner):
+    async def run():
+        a, sa = await make_node("a")
+        a.kv.register_namespace("worker")
+        a.kv.put("worker", "http://w1", {"url": "http://w1", "model_id": "m", "worker_type": "regular", "origin": "a"})
+        b, sb = await make_node("b", peers=[a])
+        try:
+            assert b.kv.get("worker", "http://w1") is not None  # via join snapshot
+        finally:
+            await a.stop()
+            await b.stop()
+            await sa.close()
+            await sb.close()
+
+    runner(run())
+
+
+def test_rate_limit_shards(runner):
+    async def run():
+        a, sa = await make_node("a")
+        b, sb = await make_node("b", peers=[a])
+        ra = RateLimitSyncAdapter(a)
+        rb = RateLimitSyncAdapter(b)
+        try:
+            ra.publish_usage("acme", epoch=7, used=40)
+            rb.publish_usage("acme", epoch=7, used=25)
+            for _ in range(40):
+                if ra.cluster_usage("acme", 7) >= 65 and rb.cluster_usage("acme", 7) >= 65:
+                    break
+                await asyncio.sleep(0.1)
+            assert ra.cluster_usage("acme", 7) == 65
+            assert rb.cluster_usage("acme", 7) == 65
+        finally:
+            await a.stop()
+            await b.stop()
+            await sa.close()
+            await sb.close()
+
+    runner(run())
