@@ -105,12 +105,29 @@ async def admission_middleware(request: web.Request, handler):
     ctx: AppContext = request.app[CTX_KEY]
     if request.path in PUBLIC_PATHS or request.path in ADMIN_PATHS:
         return await handler(request)
-    if ctx.scheduler is not None:
-        return await ctx.scheduler.admit(request, handler)
-    limiter = ctx.rate_limiter
-    if limiter is None:
-        return await handler(request)
-    return await limiter.admit(request, handler)
+    # per-tenant rate limit (reference rate_limit/manager.rs reserve -> settle)
+    trl = getattr(ctx, "tenant_rate_limiter", None)
+    reservation = None
+    if trl is not None:
+        tenant = request.get("tenant_id") or "default"
+        reservation = trl.reserve(tenant, est_tokens=max(0, request.content_length or 0) // 4)
+        if reservation is None:
+            ctx.metrics.rate_limited.labels(tenant).inc() if not ctx.metrics._null else None
+            return web.Response(
+                status=429,
+                body=error_body(f"tenant {tenant} rate limit exceeded", 429, "rate_limit_error"),
+                content_type="application/json",
+            )
+    try:
+        if ctx.scheduler is not None:
+            return await ctx.scheduler.admit(request, handler)
+        limiter = ctx.rate_limiter
+        if limiter is None:
+            return await handler(request)
+        return await limiter.admit(request, handler)
+    finally:
+        if reservation is not None:
+            trl.settle(reservation)
 
 
 PUBLIC_PATHS = {
@@ -391,6 +408,18 @@ async def startup(config: RouterConfig, serve: bool = True) -> AppContext:
 
     if config.rate_limit.max_concurrent_requests > 0:
         ctx.rate_limiter = ConcurrencyLimiter(config.rate_limit)
+    if config.priority_scheduler.enabled:
+        from ..scheduler.engine import PriorityScheduler, SchedulerConfig
+
+        sched_cfg = SchedulerConfig.from_yaml(config.priority_scheduler.config_path)
+        sched_cfg.default_class = config.priority_scheduler.default_max_class
+        ctx.scheduler = PriorityScheduler(sched_cfg, ctx.worker_registry, ctx.metrics)
+    if config.tenant_rate_limit.enabled:
+        from ..rate_limit.tenant import RateLimitManager, TenantRateLimitSettings
+
+        ctx.tenant_rate_limiter = RateLimitManager(
+            TenantRateLimitSettings.from_yaml(config.tenant_rate_limit.config_path)
+        )
     ctx.router_manager = RouterManager(ctx, config)
     if config.mesh.enabled:
         from ..mesh.adapters import MeshAdapters
