@@ -364,12 +364,19 @@ def build_app(ctx: AppContext) -> web.Application:
         "/rerank",
         "/v1/rerank",
         "/v1/embeddings",
-        "/v1/messages",
         "/v1/classify",
-        "/v1/responses",
     ]
     for p in inference_paths:
         app.router.add_post(p, _proxy_endpoint)
+
+    from ..routers.anthropic import v1_messages_handler
+
+    app.router.add_post("/v1/messages", v1_messages_handler)
+
+    from .responses_routes import add_responses_routes, v1_responses
+
+    app.router.add_post("/v1/responses", v1_responses)
+    add_responses_routes(app)
 
     app.router.add_get("/liveness", liveness)
     app.router.add_get("/readiness", readiness)
