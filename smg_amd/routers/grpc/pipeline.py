@@ -45,6 +45,7 @@ class PipelineContext:
     request_id: str = ""
     stream: bool = False
     tokenizer: Any = None
+    multimodal: Optional[Dict[str, Any]] = None
     error: Optional[RouteResponse] = None
 
 
@@ -72,6 +73,33 @@ class PreparationStage(Stage):
 
             self._template = load_chat_template(self.app.config.chat_template)
         return self._template
+
+    async def _process_images(self, ctx: PipelineContext, urls) -> bool:
+        from ...multimodal.media import MediaError, decode_image, fetch_image_bytes
+        from ...multimodal.processors import processor_for_model
+        from ...multimodal.transport import encode_tensor
+
+        proc = processor_for_model(ctx.model_id)
+        mode = self.app.config.multimodal_tensor_transport
+        images = []
+        for url in urls:
+            try:
+                data = await fetch_image_bytes(url)
+                arr = decode_image(data)
+            except MediaError as e:
+                ctx.error = RouteResponse(status=400, body=error_body(str(e)))
+                return False
+            out = proc.process(arr)
+            desc = encode_tensor(
+                out["pixel_values"], mode=mode, min_shm_bytes=self.app.config.multimodal_shm_min_bytes
+            )
+            desc["height"] = out["height"]
+            desc["width"] = out["width"]
+            if "grid_thw" in out:
+                desc["grid_thw"] = list(out["grid_thw"])
+            images.append(desc)
+        ctx.multimodal = {"images": images}
+        return True
 
     async def run(self, ctx: PipelineContext) -> bool:
         body = ctx.req.body or {}
@@ -103,6 +131,19 @@ class PreparationStage(Stage):
                 ctx.error = RouteResponse(status=400, body=error_body("'messages' must be a non-empty array"))
                 return False
             ctx.text = self.template().render(msgs, tools=body.get("tools"))
+            # multimodal: fetch + preprocess image parts on the gfx950 kernel
+            # (reference grpc/multimodal/process.rs + assemble.rs)
+            image_urls = [
+                p.get("image_url", {}).get("url") if isinstance(p.get("image_url"), dict) else p.get("image_url")
+                for m in msgs
+                if isinstance(m.get("content"), list)
+                for p in m["content"]
+                if isinstance(p, dict) and p.get("type") == "image_url"
+            ]
+            if image_urls:
+                ok = await self._process_images(ctx, [u for u in image_urls if u])
+                if not ok:
+                    return False
         elif ctx.endpoint == "completion":
             p = body.get("prompt")
             if isinstance(p, list) and p and all(isinstance(x, int) for x in p):
@@ -196,6 +237,7 @@ class RequestBuildingStage(Stage):
             input_ids=ctx.input_ids,
             sampling=ctx.sampling,
             stream=True,
+            multimodal=ctx.multimodal,
             dp_rank=ctx.dp_rank,
         )
         return True
