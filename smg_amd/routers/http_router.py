@@ -15,6 +15,7 @@ Transports:
 from __future__ import annotations
 
 import asyncio
+import json
 import logging
 import time
 import uuid
@@ -146,8 +147,15 @@ class HttpRouter(Router):
 
         try:
             engine = worker.extra.get("engine")
+            provider = worker.labels.get("provider")
             if engine is not None:
                 resp = await self._dispatch_sim(engine, worker, req)
+            elif provider in ("anthropic", "gemini") and req.path == "/v1/chat/completions" and not (req.body or {}).get("stream"):
+                # vendor translation (reference openai/provider registry)
+                from .providers import dispatch_to_provider
+
+                body = await dispatch_to_provider(await self.session(), worker, req.body or {})
+                resp = RouteResponse(status=200, body=json.dumps(body).encode())
             else:
                 resp = await self._dispatch_http(worker, req)
         except asyncio.CancelledError:
