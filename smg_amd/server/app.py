@@ -63,6 +63,43 @@ async def metrics_middleware(request: web.Request, handler):
 
 
 @web.middleware
+async def plugin_middleware(request: web.Request, handler):
+    """WASM-hook equivalent: OnRequest / OnResponse phases
+    (reference middleware/wasm.rs)."""
+    ctx: AppContext = request.app[CTX_KEY]
+    plugins = getattr(ctx, "plugins", None)
+    if plugins is None or len(plugins) == 0 or request.path in PUBLIC_PATHS:
+        return await handler(request)
+    from ..plugins import ShortCircuit
+
+    pctx = {
+        "path": request.path,
+        "method": request.method,
+        "headers": dict(request.headers),
+        "request_id": request.get("request_id"),
+        "tenant_id": request.get("tenant_id"),
+    }
+    try:
+        plugins.run_phase("on_request", pctx)
+    except ShortCircuit as sc:
+        return web.Response(status=sc.status, body=sc.body, headers=sc.headers,
+                            content_type="application/json")
+    if pctx.get("tenant_id"):
+        request["tenant_id"] = pctx["tenant_id"]
+    resp = await handler(request)
+    try:
+        out = {"status": resp.status, "headers": dict(resp.headers), "path": request.path}
+        plugins.run_phase("on_response", out)
+        for k, v in out.get("extra_headers", {}).items():
+            resp.headers[k] = v
+    except ShortCircuit as sc:
+        return web.Response(status=sc.status, body=sc.body, content_type="application/json")
+    except Exception:
+        pass
+    return resp
+
+
+@web.middleware
 async def auth_middleware(request: web.Request, handler):
     ctx: AppContext = request.app[CTX_KEY]
     auth = ctx.config.auth
@@ -351,11 +388,16 @@ def build_app(ctx: AppContext) -> web.Application:
             metrics_middleware,
             auth_middleware,
             tenant_middleware,
+            plugin_middleware,
             admission_middleware,
         ],
         client_max_size=ctx.config.max_payload_size,
     )
     app[CTX_KEY] = ctx
+    if ctx.plugins is None:
+        from ..plugins import PluginManager
+
+        ctx.plugins = PluginManager()
 
     inference_paths = [
         "/generate",
@@ -402,6 +444,34 @@ def build_app(ctx: AppContext) -> web.Application:
     from .tokenize_routes import add_tokenize_routes
 
     add_tokenize_routes(app)
+
+    from ..routers.realtime import add_realtime_routes
+
+    add_realtime_routes(app)
+
+    # plugin (WASM-equivalent) module management — route names kept /wasm
+    # for CLI/API compatibility with the reference
+    async def add_plugin(request):
+        body, _ = await _read_json(request)
+        if not body or "path" not in body:
+            return web.Response(status=400, body=error_body("'path' is required"), content_type="application/json")
+        try:
+            mod_id = ctx.plugins.add_module(body["path"], body.get("name"))
+        except Exception as exc:
+            return web.Response(status=400, body=error_body(str(exc)), content_type="application/json")
+        return web.json_response({"module_uuid": mod_id}, status=201)
+
+    async def remove_plugin(request):
+        if not ctx.plugins.remove_module(request.match_info["module_uuid"]):
+            return web.Response(status=404, body=error_body("module not found", 404), content_type="application/json")
+        return web.json_response({"status": "removed"})
+
+    async def list_plugins(request):
+        return web.json_response({"modules": ctx.plugins.list_modules()})
+
+    app.router.add_post("/wasm", add_plugin)
+    app.router.add_delete("/wasm/{module_uuid}", remove_plugin)
+    app.router.add_get("/wasm", list_plugins)
     return app
 
 
@@ -440,6 +510,20 @@ async def startup(config: RouterConfig, serve: bool = True) -> AppContext:
         if serve:
             ctx._mesh_runner = await start_mesh_server(ctx.mesh, config.mesh.host, config.mesh.port)
         await ctx.mesh.start(config.mesh.peer_urls)
+    if config.discovery.enabled:
+        from ..discovery import FileDiscovery, start_discovery
+        from ..discovery.source import KubernetesDiscovery
+
+        file_path = config.discovery.selector.get("file")
+        if file_path:
+            source = FileDiscovery(file_path)
+        else:
+            source = KubernetesDiscovery(
+                config.discovery.selector, config.discovery.port, config.discovery.namespace
+            )
+        ctx._background.append(
+            await start_discovery(ctx.worker_registry, source, circuit_breaker_config=config.circuit_breaker)
+        )
     await ctx.start_background()
     if serve:
         app = build_app(ctx)

@@ -50,6 +50,7 @@ class AppContext:
         self.mesh = None
         self.storage = None
         self.mcp = None
+        self.plugins = None
         self.kv_event_monitor = None
         self.inflight = 0
         self._background: list = []
