@@ -198,6 +198,11 @@ def build_parser() -> argparse.ArgumentParser:
     g.add_argument("--rccl-world-size", type=int, default=0, help=">0 enables the RCCL tick plane")
     g.add_argument("--rccl-tick-us", type=int, default=200)
 
+    g = p.add_argument_group("Serve (worker spawn; reference bindings/python serve.py)")
+    g.add_argument("--local-workers", type=int, default=0,
+                   help="serve mode: spawn N in-process workers (GPU TorchEngine when available, mock simulator otherwise)")
+    g.add_argument("--local-worker-model", default="local-model")
+
     g = p.add_argument_group("Misc")
     g.add_argument("--request-timeout-secs", type=int, default=1800)
     g.add_argument("--max-payload-size", type=int, default=512 << 20)
@@ -354,24 +359,111 @@ def to_router_config(argv: Optional[List[str]] = None) -> RouterConfig:
 
 
 def main(argv: Optional[List[str]] = None) -> None:
+    raw_argv = list(sys.argv[1:] if argv is None else argv)
     cfg = to_router_config(argv)
     logging.basicConfig(
         level=getattr(logging, cfg.log_level.upper(), logging.INFO),
         format="%(asctime)s %(levelname)s %(name)s %(message)s",
     )
+    # `smg serve`: spawn local workers next to the router (reference
+    # bindings/python serve.py spawns engine workers + router together)
+    n_local = 0
+    model = "local-model"
+    if "--local-workers" in raw_argv:
+        i = raw_argv.index("--local-workers")
+        n_local = int(raw_argv[i + 1])
+    if "--local-worker-model" in raw_argv:
+        model = raw_argv[raw_argv.index("--local-worker-model") + 1]
+
     from .server.app import startup
 
     async def _run():
         ctx = await startup(cfg)
+        engines = []
+        if n_local:
+            from .workers.worker import Worker
+
+            try:
+                import torch
+
+                use_gpu = torch.cuda.is_available()
+            except ImportError:
+                use_gpu = False
+            for i in range(n_local):
+                if use_gpu:
+                    from .engine.torch_engine import TorchEngine, TorchEngineConfig
+                    from .grpc.servicer import EngineAdapter
+
+                    eng = TorchEngine(TorchEngineConfig.bench_1b(), device=f"cuda:{i % torch.cuda.device_count()}")
+                    adapter = EngineAdapter(eng)
+                    await adapter.start()
+                    engines.append(adapter)
+                    w = Worker(f"sim://gpu-{i}", model_id=model)
+                    w.extra["engine"] = _TorchHttpShim(adapter, model)
+                else:
+                    from .mock.engine import MockWorkerEngine
+
+                    eng = MockWorkerEngine()
+                    eng.config.model_id = model
+                    await eng.start()
+                    engines.append(eng)
+                    w = Worker(f"sim://local-{i}", model_id=model)
+                    w.extra["engine"] = eng
+                ctx.worker_registry.register(w)
         try:
             await asyncio.Event().wait()
         finally:
+            for e in engines:
+                await e.stop()
             await ctx.shutdown()
 
     try:
         asyncio.run(_run())
     except KeyboardInterrupt:
         pass
+
+
+class _TorchHttpShim:
+    """Minimal handle() shim presenting the GPU engine over the sim:// transport."""
+
+    def __init__(self, adapter, model):
+        self.adapter = adapter
+        self.model = model
+
+    async def handle(self, path, body, headers):
+        import json as _json
+
+        from .grpc import api as _api
+
+        if path in ("/health", "/health_generate"):
+            return 200, {}, b'{"status":"ok"}'
+        if path == "/get_loads":
+            return 200, {}, _json.dumps({"loads": self.adapter.loads()}).encode()
+        if path in ("/v1/chat/completions", "/v1/completions", "/generate"):
+            body = body or {}
+            text = ""
+            msgs = body.get("messages")
+            if msgs:
+                text = "\n".join(m.get("content", "") for m in msgs if isinstance(m.get("content"), str))
+            else:
+                text = body.get("prompt") or body.get("text") or ""
+            ids = [hash(text[i: i + 4]) & 0x7FFF for i in range(0, len(text), 4)] or [1]
+            req = _api.GenerateRequest("r", input_ids=ids,
+                                       sampling=_api.SamplingParams(max_new_tokens=int(body.get("max_tokens") or 16)))
+            toks = []
+            async for chunk in self.adapter.generate(req):
+                toks.extend(chunk.token_ids)
+                if chunk.finished:
+                    break
+            content = "".join(f" tok{t}" for t in toks)
+            return 200, {}, _json.dumps({
+                "id": "cmpl-local", "object": "chat.completion", "model": self.model,
+                "choices": [{"index": 0, "message": {"role": "assistant", "content": content},
+                             "text": content, "finish_reason": "stop"}],
+                "usage": {"prompt_tokens": len(ids), "completion_tokens": len(toks),
+                          "total_tokens": len(ids) + len(toks)},
+            }).encode()
+        return 404, {}, b'{"error":"not found"}'
 
 
 if __name__ == "__main__":

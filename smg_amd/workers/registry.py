@@ -3,7 +3,9 @@
 Central store of workers with secondary indexes: by model, by worker type, plus
 per-model consistent-hash rings and an event stream for policy/monitor rewiring.
 The reference uses DashMap + lock-free Arc<[T]> snapshots; here index rebuilds
-produce immutable tuples so policies iterate a stable snapshot without locks.
+produce immutable tuples so policies iterate a stable snapshot without locks,
+and rebuilds are LAZY (dirty flag) so bulk registration of thousands of
+workers is O(n), not O(n^2) over ring inserts.
 """
 from __future__ import annotations
 
@@ -23,6 +25,7 @@ class WorkerRegistry:
         self._alias_index: Dict[str, str] = {}  # alias -> model_id
         self._rings: Dict[str, HashRing] = {}
         self._all: Tuple[Worker, ...] = ()
+        self._dirty = False
         self._listeners: List[Callable[[str, Worker], None]] = []
         self._change_event: Optional[asyncio.Event] = None
 
@@ -36,7 +39,7 @@ class WorkerRegistry:
         self._by_url[worker.url] = worker.worker_id
         for alias in worker.model_aliases:
             self._alias_index[alias] = worker.model_id
-        self._rebuild()
+        self._dirty = True
         self._emit("add", worker)
         return worker
 
@@ -46,7 +49,7 @@ class WorkerRegistry:
             return None
         if self._by_url.get(worker.url) == worker_id:
             del self._by_url[worker.url]
-        self._rebuild()
+        self._dirty = True
         self._emit("remove", worker)
         return worker
 
@@ -62,7 +65,13 @@ class WorkerRegistry:
         wid = self._by_url.get(url.rstrip("/"))
         return self._workers.get(wid) if wid is not None else None
 
+    def _ensure(self) -> None:
+        if self._dirty:
+            self._rebuild()
+            self._dirty = False
+
     def all(self) -> Tuple[Worker, ...]:
+        self._ensure()
         return self._all
 
     def resolve_model(self, model_id: Optional[str]) -> Optional[str]:
@@ -76,6 +85,7 @@ class WorkerRegistry:
         worker_type: Optional[WorkerType] = None,
         available_only: bool = True,
     ) -> List[Worker]:
+        self._ensure()
         model_id = self.resolve_model(model_id)
         if model_id is not None and model_id in self._by_model:
             candidates = self._by_model[model_id]
@@ -93,16 +103,20 @@ class WorkerRegistry:
         return out
 
     def by_type(self, worker_type: WorkerType, available_only: bool = True) -> List[Worker]:
+        self._ensure()
         ws = self._by_type.get(worker_type, ())
         return [w for w in ws if not available_only or w.is_available()]
 
     def ring_for_model(self, model_id: Optional[str]) -> Optional[HashRing]:
+        self._ensure()
         return self._rings.get(self.resolve_model(model_id) or "*")
 
     def models(self) -> List[str]:
+        self._ensure()
         return sorted(self._by_model.keys())
 
     def healthy_count(self) -> int:
+        self._ensure()
         return sum(1 for w in self._all if w.is_available())
 
     def __len__(self) -> int:
