@@ -52,6 +52,7 @@ def parse_args():
     p.add_argument("--max-new", type=int, default=32)
     p.add_argument("--tiny", action="store_true", help="tiny model (CPU smoke / CI)")
     p.add_argument("--no-graphs", action="store_true", help="disable hipGraph decode capture")
+    p.add_argument("--decode-burst", type=int, default=2, help="decode iterations per tick")
     p.add_argument("--seed", type=int, default=1234)
     return p.parse_args()
 
@@ -108,7 +109,7 @@ def worker_main(rank: int, world: int, args, device: str, backend: str):
             break
         for rid, max_new, prompt in reqs:
             eng.submit(prompt, max_new, rid=rid)
-        eng.step()
+        eng.step(decode_burst=args.decode_burst)
         events = eng.drain_events()
     elapsed = (t1 - t0) if (t0 is not None and t1 is not None) else 0.0
     el = torch.tensor([elapsed])
@@ -118,8 +119,9 @@ def worker_main(rank: int, world: int, args, device: str, backend: str):
 class LocalWorker:
     """Rank-0's own engine, driven inline in the tick loop."""
 
-    def __init__(self, eng: TorchEngine):
+    def __init__(self, eng: TorchEngine, decode_burst: int = 4):
         self.eng = eng
+        self.decode_burst = decode_burst
         self.pending: List[tuple] = []
 
     def enqueue(self, rid, max_new, prompt):
@@ -129,14 +131,14 @@ class LocalWorker:
         for rid, max_new, prompt in self.pending:
             self.eng.submit(prompt, max_new, rid=rid)
         self.pending.clear()
-        self.eng.step()
+        self.eng.step(decode_burst=self.decode_burst)
         return self.eng.drain_events()
 
 
 def gateway_main(rank: int, world: int, args, device: str, backend: str):
     use_gpu = device.startswith("cuda")
     eng = TorchEngine(engine_config(args), device=device, graphs=use_gpu and not args.no_graphs)
-    local = LocalWorker(eng)
+    local = LocalWorker(eng, decode_burst=args.decode_burst)
     remote_ranks = list(range(1, world))
     plane = (
         GatewayPlane(

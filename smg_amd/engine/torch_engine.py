@@ -109,12 +109,13 @@ class TorchEngine:
                 c.n_layers, 2, c.max_slots, c.n_heads, c.max_seq, c.head_dim,
                 device=self.device, dtype=self.dtype,
             )
-            # rotary tables
+            # rotary tables (complex form: one complex mul applies the rotation)
             inv = 1.0 / (10000.0 ** (torch.arange(0, c.head_dim, 2, device=self.device).float() / c.head_dim))
             t = torch.arange(c.max_seq, device=self.device).float()
             freqs = torch.outer(t, inv)
             self.cos = freqs.cos().to(self.dtype)
             self.sin = freqs.sin().to(self.dtype)
+            self.freqs_cis = torch.polar(torch.ones_like(freqs), freqs)  # complex64 [T, D/2]
         self.seq_len = torch.zeros(c.max_slots, dtype=torch.long, device=self.device)
         self._seq_len_host = [0] * c.max_slots
         self._last_tok = torch.zeros(c.max_slots, dtype=torch.long, device=self.device)
@@ -171,9 +172,11 @@ class TorchEngine:
         }
 
     # ---- engine step -----------------------------------------------------
-    def step(self) -> int:
-        """One engine iteration: admit + one prefill chunk + one decode token
-        for every running slot.  Returns tokens produced."""
+    def step(self, decode_burst: int = 1) -> int:
+        """One engine iteration: admit + one prefill chunk + up to
+        `decode_burst` decode tokens for every running slot (bursting
+        amortizes the host-side tick/event overhead; the serving tick loop
+        uses burst 4).  Returns tokens produced."""
         c = self.cfg
         # admission
         while self.waiting and self._free_slots:
@@ -183,17 +186,21 @@ class TorchEngine:
             self._seq_len_host[req.slot] = 0
 
         produced = 0
-        # prefill: one chunk for the first unprefilled request
+        # prefill: chunked continuous batching under a per-step token budget
+        budget = c.prefill_chunk
         for slot, req in self.running.items():
+            if budget <= 0:
+                break
             if req.prefilled < len(req.tokens):
-                chunk = req.tokens[req.prefilled: req.prefilled + c.prefill_chunk]
+                chunk = req.tokens[req.prefilled: req.prefilled + budget]
                 self._prefill(slot, chunk, req.prefilled)
                 req.prefilled += len(chunk)
-                break
+                budget -= len(chunk)
 
-        # decode: all slots whose prompt is fully prefilled
-        decode_slots = [s for s, r in self.running.items() if r.prefilled >= len(r.tokens)]
-        if decode_slots:
+        for _ in range(max(1, decode_burst)):
+            decode_slots = [s for s, r in self.running.items() if r.prefilled >= len(r.tokens)]
+            if not decode_slots:
+                break
             new_tokens = self._decode(decode_slots)
             finished = []
             for s, tok in zip(decode_slots, new_tokens):
@@ -213,19 +220,18 @@ class TorchEngine:
 
     # ---- forwards ----------------------------------------------------------
     @staticmethod
-    def _apply_rope(x, cos, sin):
-        # x: [B, H, T, D]; cos/sin broadcastable to [B, H, T, D/2]
-        x1, x2 = x[..., ::2], x[..., 1::2]
-        out = torch.empty_like(x)
-        out[..., ::2] = x1 * cos - x2 * sin
-        out[..., 1::2] = x1 * sin + x2 * cos
-        return out
+    def _apply_rope(x, freqs):
+        # x: [B, H, T, D]; freqs complex broadcastable to [B, H, T, D/2].
+        # One complex multiply == the interleaved-pair rotation (3 kernels
+        # instead of the 8 slice/mul/add kernels of the real-domain form).
+        xc = torch.view_as_complex(x.float().reshape(*x.shape[:-1], -1, 2))
+        return torch.view_as_real(xc * freqs).flatten(-2).to(x.dtype)
 
     def _mlp(self, h, layer):
         x = _rms(h, layer.ln2)
         return h + (F.silu(x @ layer.w1) * (x @ layer.w3)) @ layer.w2
 
-    def _qkv(self, h, layer, cos, sin):
+    def _qkv(self, h, layer, freqs):
         c = self.cfg
         B, T, _ = h.shape
         qkv = _rms(h, layer.ln1) @ layer.wqkv
@@ -233,7 +239,7 @@ class TorchEngine:
         q = q.view(B, T, c.n_heads, c.head_dim).transpose(1, 2)
         k = k.view(B, T, c.n_heads, c.head_dim).transpose(1, 2)
         v = v.view(B, T, c.n_heads, c.head_dim).transpose(1, 2)
-        return self._apply_rope(q, cos, sin), self._apply_rope(k, cos, sin), v
+        return self._apply_rope(q, freqs), self._apply_rope(k, freqs), v
 
     @torch.no_grad()
     def _prefill(self, slot: int, tokens: List[int], start: int) -> None:
@@ -242,14 +248,13 @@ class TorchEngine:
         t = torch.tensor([tokens], device=self.device, dtype=torch.long)
         h = self.embed[t]
         pos = torch.arange(start, start + T, device=self.device)
-        cos = self.cos[pos].view(1, 1, T, -1)
-        sin = self.sin[pos].view(1, 1, T, -1)
+        freqs = self.freqs_cis[pos].view(1, 1, T, -1)
         mask = None
         if start:  # chunked continuation: causal mask with history offset
             kpos = torch.arange(0, start + T, device=self.device)
             mask = (kpos.unsqueeze(0) <= pos.unsqueeze(1)).unsqueeze(0).unsqueeze(0)
         for li, layer in enumerate(self.layers):
-            q, k, v = self._qkv(h, layer, cos, sin)
+            q, k, v = self._qkv(h, layer, freqs)
             self.kv[li, 0, slot, :, start: start + T] = k[0]
             self.kv[li, 1, slot, :, start: start + T] = v[0]
             kk = self.kv[li, 0, slot: slot + 1, :, : start + T]
@@ -274,13 +279,12 @@ class TorchEngine:
         c = self.cfg
         S = c.max_slots
         pos = self.seq_len  # [S] current length == write position
-        cos = self.cos[pos].view(S, 1, 1, -1)
-        sin = self.sin[pos].view(S, 1, 1, -1)
+        freqs = self.freqs_cis[pos].view(S, 1, 1, -1)
         kpos = torch.arange(maxlen, device=self.device)
         mask = (kpos.unsqueeze(0) <= pos.unsqueeze(1)).unsqueeze(1).unsqueeze(1)
         h = self.embed[self._last_tok.unsqueeze(1)]  # [S, 1, D]
         for li, layer in enumerate(self.layers):
-            q, k, v = self._qkv(h, layer, cos, sin)
+            q, k, v = self._qkv(h, layer, freqs)
             self.kv[li, 0, self._arange_slots, :, pos] = k[:, :, 0]
             self.kv[li, 1, self._arange_slots, :, pos] = v[:, :, 0]
             kk = self.kv[li, 0][:, :, :maxlen]
