@@ -176,7 +176,8 @@ extern "C" void smg_img_destroy(void* p) {
     ImgHost* h = (ImgHost*)p;
     if (!h) return;
     if (h->on_gpu) {
-        hipFree(h->d_in); hipFree(h->d_tmp); hipFree(h->d_out); hipFree(h->d_outf);
+        // d_outf is an interior pointer into d_out — never hipFree it
+        hipFree(h->d_in); hipFree(h->d_tmp); hipFree(h->d_out);
         hipFree(h->d_coef); hipFree(h->d_norm);
         hipStreamDestroy(h->stream);
     }
