@@ -109,6 +109,7 @@ class EmbedRequest:
 METHODS = {
     "Generate": "server_stream",
     "Embed": "unary",
+    "EncodeImage": "unary",
     "HealthCheck": "unary",
     "Abort": "unary",
     "GetModelInfo": "unary",

@@ -86,6 +86,9 @@ class EngineClient:
     async def embed(self, req: api.EmbedRequest) -> dict:
         return await self._unary("Embed", req.to_dict())
 
+    async def encode_image(self, request_id: str, multimodal: dict) -> dict:
+        return await self._unary("EncodeImage", {"request_id": request_id, "multimodal": multimodal}, timeout=60.0)
+
     async def subscribe_kv_events(self) -> AsyncIterator[dict]:
         call = self.channel().unary_stream(
             api.method("SubscribeKvEvents"), request_serializer=_BYTES, response_deserializer=_BYTES

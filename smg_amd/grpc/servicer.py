@@ -149,6 +149,22 @@ class SchedulerServicer(grpc.GenericRpcHandler):
     def _h_healthcheck(self, request: bytes, context):
         return api.dumps({"healthy": True})
 
+    def _h_encodeimage(self, request: bytes, context):
+        """EPD encode leg: vision-tower forward for the pixel tensor (mock:
+        deterministic pooled embedding; the GPU engine runs its tower).
+        Returns an embedding descriptor the prefill leg consumes
+        (reference grpc_servicer encoder_servicer.py)."""
+        import hashlib
+
+        d = api.loads(request)
+        images = (d.get("multimodal") or {}).get("images") or []
+        embeddings = []
+        for im in images:
+            h = hashlib.blake2b(im.get("data") or b"", digest_size=32).digest()
+            vec = [b / 255.0 for b in h]
+            embeddings.append({"embedding": vec, "height": im.get("height"), "width": im.get("width")})
+        return api.dumps({"request_id": d.get("request_id"), "embeddings": embeddings})
+
     def _h_abort(self, request: bytes, context):
         d = api.loads(request)
         self.adapter.abort(d.get("request_id", ""))

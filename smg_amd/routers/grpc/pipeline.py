@@ -39,6 +39,8 @@ class PipelineContext:
     input_ids: List[int] = field(default_factory=list)
     sampling: api.SamplingParams = field(default_factory=api.SamplingParams)
     worker: Any = None
+    prefill_worker: Any = None  # PD/EPD: ctx.worker is the decode leg
+    encode_worker: Any = None
     dp_rank: Optional[int] = None
     client: Any = None
     gen_request: Optional[api.GenerateRequest] = None
@@ -189,13 +191,9 @@ class WorkerSelectionStage(Stage):
         self.app = app_ctx
 
     async def run(self, ctx: PipelineContext) -> bool:
-        workers = self.app.worker_registry.for_model(ctx.model_id)
-        if not workers:
-            ctx.error = RouteResponse(
-                status=503, body=error_body(f"no available worker for model {ctx.model_id or 'any'}", 503)
-            )
-            return False
-        policy = self.app.policy_registry.get(ctx.model_id)
+        from ...config import RoutingMode
+        from ...workers.worker import WorkerType
+
         info = SelectWorkerInfo(
             request_id=ctx.request_id,
             model_id=ctx.model_id,
@@ -204,13 +202,43 @@ class WorkerSelectionStage(Stage):
             tenant_id=ctx.req.tenant_id,
             est_tokens=len(ctx.input_ids),
         )
+        mode = self.app.config.mode
         t0 = time.perf_counter_ns()
-        idx = policy.select_worker(workers, info)
+        if mode in (RoutingMode.PREFILL_DECODE, RoutingMode.ENCODE_PREFILL_DECODE):
+            # per-leg policies (reference worker_selection.rs:48-56 modes)
+            prefill_pool = self.app.worker_registry.for_model(ctx.model_id, worker_type=WorkerType.PREFILL)
+            decode_pool = self.app.worker_registry.for_model(ctx.model_id, worker_type=WorkerType.DECODE)
+            if not prefill_pool or not decode_pool:
+                ctx.error = RouteResponse(status=503, body=error_body("no prefill/decode workers", 503))
+                return False
+            p_idx = self.app.policy_registry.get(ctx.model_id, "prefill").select_worker(prefill_pool, info)
+            d_idx = self.app.policy_registry.get(ctx.model_id, "decode").select_worker(decode_pool, info)
+            if p_idx is None or d_idx is None:
+                ctx.error = RouteResponse(status=503, body=error_body("selection failed", 503))
+                return False
+            ctx.prefill_worker = prefill_pool[p_idx]
+            ctx.worker = decode_pool[d_idx]
+            if mode == RoutingMode.ENCODE_PREFILL_DECODE and ctx.multimodal:
+                encode_pool = self.app.worker_registry.for_model(ctx.model_id, worker_type=WorkerType.ENCODE)
+                if not encode_pool:
+                    ctx.error = RouteResponse(status=503, body=error_body("no encode workers", 503))
+                    return False
+                e_idx = self.app.policy_registry.get(ctx.model_id, "encode").select_worker(encode_pool, info)
+                ctx.encode_worker = encode_pool[e_idx] if e_idx is not None else encode_pool[0]
+        else:
+            workers = self.app.worker_registry.for_model(ctx.model_id)
+            if not workers:
+                ctx.error = RouteResponse(
+                    status=503, body=error_body(f"no available worker for model {ctx.model_id or 'any'}", 503)
+                )
+                return False
+            policy = self.app.policy_registry.get(ctx.model_id)
+            idx = policy.select_worker(workers, info)
+            if idx is None:
+                ctx.error = RouteResponse(status=503, body=error_body("selection failed", 503))
+                return False
+            ctx.worker = workers[idx]
         self.app.metrics.observe_routing_latency((time.perf_counter_ns() - t0) / 1e9)
-        if idx is None:
-            ctx.error = RouteResponse(status=503, body=error_body("selection failed", 503))
-            return False
-        ctx.worker = workers[idx]
         # DP-aware rank routing (reference dp_min_token.rs via monitor.rs:164)
         if self.app.config.dp_aware and ctx.worker.dp_size:
             ctx.dp_rank = self.app.policy_registry.dp_policy.select_dp_rank(ctx.worker)
@@ -228,6 +256,34 @@ class ClientAcquisitionStage(Stage):
         return True
 
 
+class EncodeStage(Stage):
+    """EPD only: dispatch the vision encode to the encode fleet and replace
+    raw pixels with embeddings (reference common/stages/encode.rs; pixel
+    transport inline/SHM/RDMA -> here inline/shm/xgmi descriptors)."""
+
+    name = "encode"
+
+    def __init__(self, pool: ClientPool):
+        self.pool = pool
+
+    async def run(self, ctx: PipelineContext) -> bool:
+        if ctx.encode_worker is None or not ctx.multimodal:
+            return True
+        client = self.pool.get(ctx.encode_worker.url)
+        ctx.encode_worker.incr_load()
+        try:
+            result = await client.encode_image(ctx.request_id, ctx.multimodal)
+            ctx.encode_worker.record_outcome(True)
+        except Exception as exc:
+            ctx.encode_worker.record_outcome(False)
+            ctx.error = RouteResponse(status=502, body=error_body(f"encode failed: {exc}", 502))
+            return False
+        finally:
+            ctx.encode_worker.decr_load()
+        ctx.multimodal = {"embeddings": result.get("embeddings", [])}
+        return True
+
+
 class RequestBuildingStage(Stage):
     name = "request_building"
 
@@ -240,6 +296,18 @@ class RequestBuildingStage(Stage):
             multimodal=ctx.multimodal,
             dp_rank=ctx.dp_rank,
         )
+        if ctx.prefill_worker is not None:
+            # bootstrap metadata for the engine-side KV handoff
+            # (reference request_execution.rs:253 + pd_types.rs:15)
+            import random as _random
+            from urllib.parse import urlparse
+
+            ctx.gen_request.bootstrap_host = (
+                ctx.prefill_worker.bootstrap_host
+                or urlparse("http://" + ctx.prefill_worker.url.split("://", 1)[-1]).hostname
+            )
+            ctx.gen_request.bootstrap_port = ctx.prefill_worker.bootstrap_port
+            ctx.gen_request.bootstrap_room = _random.getrandbits(63)
         return True
 
 

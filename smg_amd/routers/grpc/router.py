@@ -18,6 +18,7 @@ from ..base import RouteRequest, RouteResponse, Router
 from .pipeline import (
     ClientAcquisitionStage,
     DispatchMetadataStage,
+    EncodeStage,
     PipelineContext,
     PreparationStage,
     RequestBuildingStage,
@@ -46,11 +47,12 @@ class GrpcRouter(Router):
         self.prep = PreparationStage(app_ctx)
         self.select = WorkerSelectionStage(app_ctx)
         self.acquire = ClientAcquisitionStage(self.pool)
+        self.encode = EncodeStage(self.pool)
         self.build = RequestBuildingStage()
         self.meta = DispatchMetadataStage()
         self.execute = RequestExecutionStage()
         self.process = ResponseProcessingStage(app_ctx)
-        self.stages = [self.prep, self.select, self.acquire, self.build, self.meta, self.execute]
+        self.stages = [self.prep, self.select, self.acquire, self.encode, self.build, self.meta, self.execute]
 
     async def shutdown(self) -> None:
         await self.pool.close()
@@ -92,9 +94,34 @@ class GrpcRouter(Router):
         }
         return RouteResponse(status=200, body=json.dumps(body).encode())
 
+    def _fire_prefill(self, ctx: PipelineContext):
+        """PD dual dispatch: run the prefill leg concurrently and drain its
+        stream (reference request_execution.rs:253); KV moves engine-side via
+        the bootstrap metadata."""
+        if ctx.prefill_worker is None:
+            return None
+        pclient = self.pool.get(ctx.prefill_worker.url)
+        ctx.prefill_worker.incr_load(len(ctx.input_ids))
+
+        async def drain():
+            ok = True
+            try:
+                async for chunk in pclient.generate(ctx.gen_request):
+                    if chunk.finished:
+                        break
+            except Exception as exc:
+                ok = False
+                log.debug("prefill leg failed on %s: %s", ctx.prefill_worker.url, exc)
+            finally:
+                ctx.prefill_worker.decr_load(len(ctx.input_ids))
+                ctx.prefill_worker.record_outcome(ok)
+
+        return asyncio.ensure_future(drain())
+
     # ---- unary ------------------------------------------------------------
     async def _unary_response(self, ctx: PipelineContext) -> RouteResponse:
         detok, stop, reasoning, tool_stream = self.process.make_processors(ctx)
+        prefill_task = self._fire_prefill(ctx)
         text_parts: List[str] = []
         finish_reason = "stop"
         usage = {"prompt_tokens": len(ctx.input_ids), "completion_tokens": 0}
@@ -183,6 +210,7 @@ class GrpcRouter(Router):
     # ---- streaming ---------------------------------------------------------
     async def _stream_response(self, ctx: PipelineContext) -> AsyncIterator[bytes]:
         detok, stop, reasoning, tool_stream = self.process.make_processors(ctx)
+        prefill_task = self._fire_prefill(ctx)
         created = int(time.time())
         rid = f"chatcmpl-{ctx.request_id[:24]}"
         first = True
