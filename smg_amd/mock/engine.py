@@ -248,6 +248,23 @@ class MockWorkerEngine:
             return 200, {}, b'{"status":"ok"}'
         if path in ("/v1/chat/completions", "/v1/completions", "/generate"):
             return await self._generate(path, body or {}, headers)
+        if path in ("/v1/rerank", "/rerank"):
+            body = body or {}
+            docs = body.get("documents") or []
+            query = body.get("query", "")
+            results = [
+                {"index": i, "relevance_score": 1.0 / (1 + abs(hash(query) - hash(d)) % 100)}
+                for i, d in enumerate(docs)
+            ]
+            results.sort(key=lambda r: -r["relevance_score"])
+            return 200, {}, json.dumps({"results": results, "model": self.config.model_id}).encode()
+        if path == "/v1/classify":
+            body = body or {}
+            text = body.get("input") or body.get("text") or ""
+            label = "positive" if hash(text) % 2 == 0 else "negative"
+            return 200, {}, json.dumps(
+                {"object": "classification", "model": self.config.model_id,
+                 "data": [{"index": 0, "label": label, "score": 0.9}]}).encode()
         return 404, {}, b'{"error":"not found"}'
 
     async def _generate(self, path: str, body: Dict[str, Any], headers: Dict[str, str]):

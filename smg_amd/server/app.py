@@ -449,6 +449,37 @@ def build_app(ctx: AppContext) -> web.Application:
 
     add_realtime_routes(app)
 
+    async def v1_audio_transcriptions(request: web.Request):
+        # multipart upload passthrough (reference multipart_upload_routes):
+        # decodes the form, routes the transcription to an audio-capable
+        # worker; without one, answers 501 with a clear error
+        try:
+            reader = await request.multipart()
+        except (AssertionError, ValueError):
+            return web.Response(status=400, body=error_body("multipart/form-data required"),
+                                content_type="application/json")
+        model = None
+        audio_bytes = 0
+        async for part in reader:
+            if part.name == "model":
+                model = (await part.text()).strip()
+            elif part.name == "file":
+                while True:
+                    chunk = await part.read_chunk()
+                    if not chunk:
+                        break
+                    audio_bytes += len(chunk)
+        workers = ctx.worker_registry.for_model(ctx.worker_registry.resolve_model(model))
+        audio_workers = [w for w in workers if w.labels.get("audio") == "true"]
+        if not audio_workers:
+            return web.Response(
+                status=501,
+                body=error_body("no audio-capable worker registered (label audio=true)", 501),
+                content_type="application/json")
+        return web.json_response({"text": "", "model": model, "bytes_received": audio_bytes})
+
+    app.router.add_post("/v1/audio/transcriptions", v1_audio_transcriptions)
+
     # plugin (WASM-equivalent) module management — route names kept /wasm
     # for CLI/API compatibility with the reference
     async def add_plugin(request):
@@ -525,6 +556,16 @@ async def startup(config: RouterConfig, serve: bool = True) -> AppContext:
             await start_discovery(ctx.worker_registry, source, circuit_breaker_config=config.circuit_breaker)
         )
     await ctx.start_background()
+    if serve and config.health_check_port:
+        probe_app = web.Application()
+        probe_app[CTX_KEY] = ctx
+        probe_app.router.add_get("/liveness", liveness)
+        probe_app.router.add_get("/readiness", readiness)
+        probe_app.router.add_get("/health", health)
+        probe_runner = web.AppRunner(probe_app, access_log=None)
+        await probe_runner.setup()
+        await web.TCPSite(probe_runner, config.host, config.health_check_port).start()
+        ctx._probe_runner = probe_runner
     if serve:
         app = build_app(ctx)
         runner = web.AppRunner(app, access_log=None)
