@@ -163,15 +163,19 @@ def gateway_main(rank: int, world: int, args, device: str, backend: str):
 
     completed_total = 0
     routing_lat: List[float] = []
+    phase_t = {"gen": 0.0, "route": 0.0, "local": 0.0, "plane": 0.0, "events": 0.0, "ticks": 0}
 
     def one_tick():
         nonlocal completed_total
+        phase_t["ticks"] += 1
+        tg = time.perf_counter()
         # 1) new arrivals to hold steady-state concurrency
         new_reqs = []
         while len(inflight) + len(new_reqs) < target_inflight:
             new_reqs.append(gen.make())
             if len(new_reqs) >= 64:
                 break
+        phase_t["gen"] += time.perf_counter() - tg
         # 2) route the batch (GPU kernel: one launch)
         if new_reqs:
             infos = [
@@ -182,6 +186,7 @@ def gateway_main(rank: int, world: int, args, device: str, backend: str):
             sels = policy.select_worker_batch(workers, infos)
             dt = time.perf_counter() - t0
             routing_lat.extend([dt / max(1, len(new_reqs))] * len(new_reqs))
+            phase_t["route"] += dt
             for (rid, prompt, max_new), sel in zip(new_reqs, sels):
                 sel = 0 if sel is None else sel
                 workers[sel].incr_load()
@@ -191,11 +196,16 @@ def gateway_main(rank: int, world: int, args, device: str, backend: str):
                 else:
                     plane.enqueue(sel, rid, max_new, prompt)
         # 3) lockstep exchange + local step
+        tl = time.perf_counter()
         events = list(local.tick())
+        phase_t["local"] += time.perf_counter() - tl
         if plane is not None:
+            tp = time.perf_counter()
             for w, evs in plane.tick().items():
                 events.extend(evs)
+            phase_t["plane"] += time.perf_counter() - tp
         # 4) completions
+        tev = time.perf_counter()
         done_now = 0
         for rid, _token, flags in events:
             if flags & DONE:
@@ -205,6 +215,7 @@ def gateway_main(rank: int, world: int, args, device: str, backend: str):
                     workers[wrk].processed_requests += 1
                     done_now += 1
         completed_total += done_now
+        phase_t["events"] += time.perf_counter() - tev
         return done_now
 
     def run_until(n_completions: int, max_ticks: int = 1_000_000):
@@ -266,6 +277,11 @@ def gateway_main(rank: int, world: int, args, device: str, backend: str):
             "reqs_per_step": args.reqs_per_step,
         },
     }
+    import sys as _sys
+
+    ticks = max(1, phase_t.pop("ticks"))
+    print("# tick breakdown ms/tick: " + " ".join(f"{k}={v*1e3/ticks:.3f}" for k, v in phase_t.items())
+          + f" ticks={ticks}", file=_sys.stderr)
     print(json.dumps(result), flush=True)
 
 
