@@ -35,6 +35,11 @@ class TorchEngineConfig:
     prefill_chunk: int = 2048
     dtype: str = "bfloat16"
     seed: int = 1234
+    # prefix KV cache (vLLM/SGLang-style): shared prompt prefixes skip
+    # recompute — the engine-side half of cache-aware routing
+    prefix_cache_slots: int = 16
+    prefix_cache_page: int = 64
+    prefix_cache_max: int = 1024  # max cached prefix tokens
 
     @classmethod
     def tiny(cls) -> "TorchEngineConfig":
@@ -129,6 +134,22 @@ class TorchEngine:
         self._step_events: List[tuple] = []  # (rid, token, done) since last drain
         self.graphs = graphs
         self._graph_cache: Dict[int, tuple] = {}
+        # prefix KV cache arena: same [L, 2, slot, H, T, D] layout as self.kv
+        # so a hit is one strided device copy
+        pc = self.cfg
+        if pc.prefix_cache_slots > 0:
+            self._pc_arena = torch.zeros(
+                pc.n_layers, 2, pc.prefix_cache_slots, pc.n_heads,
+                min(pc.prefix_cache_max, pc.max_seq), pc.head_dim,
+                device=self.device, dtype=self.dtype,
+            )
+        else:
+            self._pc_arena = None
+        self._pc_keys: Dict[int, tuple] = {}  # hash(prefix pages) -> (slot, plen)
+        self._pc_lru: List[int] = list(range(pc.prefix_cache_slots))  # front = LRU victim
+        self._pc_slot_keys: Dict[int, List[int]] = {}  # slot -> its hash keys
+        self.prefix_cache_hits = 0
+        self.prefix_cache_miss = 0
 
     # ---- API -------------------------------------------------------------
     def submit(self, tokens: List[int], max_new_tokens: int, rid: Optional[str] = None) -> str:
@@ -186,16 +207,33 @@ class TorchEngine:
             self._seq_len_host[req.slot] = 0
 
         produced = 0
-        # prefill: chunked continuous batching under a per-step token budget
+        # prefill: chunked continuous batching under a per-step token budget,
+        # with prefix-cache restore for fresh slots
         budget = c.prefill_chunk
         for slot, req in self.running.items():
             if budget <= 0:
                 break
+            if req.prefilled == 0 and self._pc_arena is not None:
+                hit = self._prefix_lookup(req.tokens)
+                if hit is not None:
+                    pslot, plen = hit
+                    self.kv[:, :, slot, :, :plen] = self._pc_arena[:, :, pslot, :, :plen]
+                    self._seq_len_host[slot] = plen
+                    self.seq_len[slot] = plen
+                    self._last_tok[slot] = int(req.tokens[plen - 1])
+                    req.prefilled = plen
+                    self.prefix_cache_hits += 1
+                    if req.prefilled >= len(req.tokens):
+                        continue
+                else:
+                    self.prefix_cache_miss += 1
             if req.prefilled < len(req.tokens):
                 chunk = req.tokens[req.prefilled: req.prefilled + budget]
                 self._prefill(slot, chunk, req.prefilled)
                 req.prefilled += len(chunk)
                 budget -= len(chunk)
+                if req.prefilled >= len(req.tokens):
+                    self._prefix_store(slot, req.tokens)
 
         for _ in range(max(1, decode_burst)):
             decode_slots = [s for s, r in self.running.items() if r.prefilled >= len(r.tokens)]
@@ -226,6 +264,49 @@ class TorchEngine:
         # instead of the 8 slice/mul/add kernels of the real-domain form).
         xc = torch.view_as_complex(x.float().reshape(*x.shape[:-1], -1, 2))
         return torch.view_as_real(xc * freqs).flatten(-2).to(x.dtype)
+
+    # ---- prefix KV cache ---------------------------------------------------
+    def _prefix_lookup(self, tokens: List[int]):
+        """Longest page-aligned cached prefix of `tokens` -> (arena_slot, plen)."""
+        c = self.cfg
+        page = c.prefix_cache_page
+        pmax = min((len(tokens) // page) * page, c.prefix_cache_max, c.max_seq - 1)
+        for p in range(pmax, page - 1, -page):
+            entry = self._pc_keys.get(hash(tuple(tokens[:p])))
+            if entry is not None:
+                slot, stored = entry
+                if stored >= p:
+                    # LRU touch
+                    try:
+                        self._pc_lru.remove(slot)
+                        self._pc_lru.append(slot)
+                    except ValueError:
+                        pass
+                    return slot, p
+        return None
+
+    def _prefix_store(self, kv_slot: int, tokens: List[int]) -> None:
+        if self._pc_arena is None:
+            return
+        c = self.cfg
+        page = c.prefix_cache_page
+        plen = min((len(tokens) // page) * page, c.prefix_cache_max, c.max_seq - 1)
+        if plen < page:
+            return
+        top_key = hash(tuple(tokens[:plen]))
+        if top_key in self._pc_keys:
+            return  # already cached
+        victim = self._pc_lru.pop(0)
+        for k in self._pc_slot_keys.pop(victim, []):
+            self._pc_keys.pop(k, None)
+        self._pc_arena[:, :, victim, :, :plen] = self.kv[:, :, kv_slot, :, :plen]
+        keys = []
+        for p in range(page, plen + 1, page):
+            k = hash(tuple(tokens[:p]))
+            self._pc_keys[k] = (victim, p)
+            keys.append(k)
+        self._pc_slot_keys[victim] = keys
+        self._pc_lru.append(victim)
 
     def _mlp(self, h, layer):
         x = _rms(h, layer.ln2)
