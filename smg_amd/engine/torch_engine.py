@@ -157,7 +157,10 @@ class TorchEngine:
             self._rid_counter += 1
             rid = f"req-{self._rid_counter}"
         c = self.cfg
-        toks = [t % c.vocab_size for t in tokens][-(c.max_seq - max_new_tokens - 1):]
+        toks = [t % c.vocab_size for t in tokens]
+        if len(toks) > c.max_seq - 2:
+            toks = toks[-(c.max_seq - 2):]  # keep the prompt tail
+        max_new_tokens = max(1, min(max_new_tokens, c.max_seq - 1 - len(toks)))
         req = _Request(rid, toks, max_new_tokens)
         self._requests[rid] = req
         self.waiting.append(req)
