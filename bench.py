@@ -112,7 +112,7 @@ def worker_main(rank: int, world: int, args, device: str, backend: str):
         eng.step(decode_burst=args.decode_burst)
         events = eng.drain_events()
     elapsed = (t1 - t0) if (t0 is not None and t1 is not None) else 0.0
-    el = torch.tensor([elapsed])
+    el = torch.tensor([elapsed], device=device if backend == "nccl" else "cpu")
     torch.distributed.all_reduce(el, op=torch.distributed.ReduceOp.MAX)
 
 
@@ -244,7 +244,7 @@ def gateway_main(rank: int, world: int, args, device: str, backend: str):
     # stop workers and fold in their elapsed (max over ranks)
     if plane is not None:
         plane.tick(stop=True)
-        el = torch.tensor([elapsed])
+        el = torch.tensor([elapsed], device=device if backend == "nccl" else "cpu")
         torch.distributed.all_reduce(el, op=torch.distributed.ReduceOp.MAX)
         elapsed = float(el[0])
 
