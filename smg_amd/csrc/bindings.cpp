@@ -318,7 +318,7 @@ PYBIND11_MODULE(_core, m) {
     py::class_<PyGpuTree>(m, "GpuTree")
         .def(py::init<int, uint32_t, uint32_t, uint32_t, uint32_t, uint32_t, uint32_t>(),
              py::arg("device") = 0, py::arg("node_cap") = 1u << 22, py::arg("table_size") = 1u << 23,
-             py::arg("page_size") = 16, py::arg("max_pages") = 4096, py::arg("max_batch_reqs") = 4096,
+             py::arg("page_size") = 16, py::arg("max_pages") = 1024, py::arg("max_batch_reqs") = 4096,
              py::arg("max_batch_tokens") = 1u << 22)
         .def("run", &PyGpuTree::run, py::arg("tokens_flat"), py::arg("offsets"),
              py::arg("healthy_mask"), py::arg("loads"), py::arg("processed"), py::arg("n_workers"),

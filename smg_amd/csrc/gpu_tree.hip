@@ -1,27 +1,30 @@
-// GPU-resident paged radix prefix tree for cache-aware routing — gfx950 (MI355X).
+// GPU-resident paged prefix index for cache-aware routing — gfx950 (MI355X).
 //
 // Re-designs the reference's host radix tree (crates/kv_index/src/token_tree.rs:303,
 // match_prefix_with_counts :620, match_and_insert :754) as a device-resident
 // structure sized for 288 GB HBM3E:
 //
-//   * nodes are one page (page_size tokens) each; the child relation is ONE
-//     global open-addressed hash table keyed by mix64(parent_id, page_hash)
-//     — a match walk is a chain of hash probes, no pointer chasing;
-//   * a request is serviced by one wave64: the page hash is a lane-parallel
-//     polynomial reduction, the probe checks 64 slots per step in parallel
-//     (ballot), and the tenant/min-load decision is a wave reduction;
-//   * no stored page tokens are verified: the 64-bit keyed identity makes a
-//     false match ~2^-64 per pair, and a radix-tree hit is a routing HINT —
-//     a collision costs one sub-optimally routed request, never correctness.
-//     This removes every cross-CU plain-memory read; all shared state
-//     (table keys/vals, tenant bitmasks, LRU stamps, load counters) is
-//     accessed with agent-scope atomics which bypass the non-coherent L1s
-//     (MI355X_MICROARCH.md §Workgroup dispatch: atomics/sc1 are L2-served).
-//   * per-(node, tenant) LRU stamps from a device logical clock drive the
+//   * one entry per (prefix depth, page), keyed by a CHAIN HASH — a rolling
+//     polynomial over the whole token prefix (pages weighted by W^page).
+//     The key is a pure function of the tokens, NOT of tree state, so a
+//     request's 64 page lookups are INDEPENDENT: one wave matches 64 prefix
+//     depths in parallel (lane p probes depth p) instead of a serial
+//     parent->child walk with a ~900-cycle HBM dependency per level.
+//     matched length = count of trailing ones in the presence ballot;
+//   * no stored tokens are verified: the 64-bit chained key commits to the
+//     entire prefix; a collision is ~2^-64 and costs one sub-optimally
+//     routed request, never correctness (routing is a hint);
+//   * all shared state (table, tenant bitmasks, LRU stamps, load counters)
+//     is touched with agent-scope atomics which bypass the non-coherent L1s
+//     (MI355X_MICROARCH.md §Workgroup dispatch: atomics/sc1 are L2-served);
+//   * per-(entry, tenant) LRU stamps from a device logical clock drive the
 //     eviction sweep; tenants are worker slots 0..63 (bitmask u64).
 //
+// The host C++ twin (host_tree.cpp) computes the IDENTICAL chain keys
+// sequentially, so host and device indexes are differentially testable.
+//
 // Kernels: match-decide-insert (the whole cache_aware decision per batch in
-// one launch), match-only, insert-only, tenant removal, LRU sweep, stats.
+// one launch), tenant removal, LRU sweep, stats.
 #include <hip/hip_runtime.h>
 
 #include <cstdint>
@@ -60,54 +63,55 @@ __device__ __forceinline__ unsigned long long mix64(unsigned long long x) {
     return x ^ (x >> 31);
 }
 
-__device__ __forceinline__ unsigned long long child_key(uint32_t parent, unsigned long long page_hash) {
-    unsigned long long k = mix64(((unsigned long long)parent << 32) ^ page_hash * 0x9E3779B97F4A7C15ull);
-    // reserve 0/1
+// ---- chain-hash key schedule (host_tree.cpp computes the identical one) ----
+// page_hash(page) = mix(sum_j mix(tok_j + SALT) * C^j  ^  page_size)
+// chain(p)       = sum_{q<=p} page_hash(q) * W^q        (mod 2^64)
+// key(p)         = max(2, mix(chain(p) ^ (p+1)*GOLD))
+#define CHAIN_W 0xA24BAED4963EE407ull
+#define CHAIN_GOLD 0x9E3779B97F4A7C15ull
+#define CHAIN_SALT 0x5851F42D4C957F2Dull
+#define MAX_LDS_PAGES 1024
+
+__constant__ unsigned long long c_pow[WAVE];   // C^j, j < 64 (within-page)
+__constant__ unsigned long long c_wpow[WAVE];  // W^p, p < 64 (within-chunk)
+__constant__ unsigned long long c_w64[1];      // W^64 (chunk carry scale)
+
+__device__ __forceinline__ unsigned long long lane_page_hash(const uint32_t* toks, int n) {
+    unsigned long long h = 0;
+    for (int j = 0; j < n; ++j)
+        h += mix64((unsigned long long)toks[j] + CHAIN_SALT) * c_pow[j];
+    return mix64(h ^ (unsigned long long)n);
+}
+
+__device__ __forceinline__ unsigned long long chain_key(unsigned long long chain, uint32_t depth) {
+    unsigned long long k = mix64(chain ^ ((unsigned long long)(depth + 1) * CHAIN_GOLD));
     return k < 2 ? k + 2 : k;
 }
 
-// Order-sensitive page hash: lane i contributes mix64(token_i) * PRIME^i,
-// wave-summed.  POW table in constant memory, page_size <= 64.
-__constant__ unsigned long long c_pow[WAVE];
-
-__device__ __forceinline__ unsigned long long wave_sum_u64(unsigned long long v) {
+__device__ __forceinline__ unsigned long long wave_prefix_sum_u64(unsigned long long v, int lane) {
 #pragma unroll
-    for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, WAVE);
+    for (int off = 1; off < WAVE; off <<= 1) {
+        unsigned long long up = __shfl_up(v, off, WAVE);
+        if (lane >= off) v += up;
+    }
     return v;
-}
-
-__device__ __forceinline__ unsigned long long page_hash_wave(const uint32_t* toks, int n, int lane) {
-    unsigned long long contrib = 0;
-    if (lane < n) contrib = mix64((unsigned long long)toks[lane] + 0x5851F42D4C957F2Dull) * c_pow[lane];
-    return mix64(wave_sum_u64(contrib) ^ (unsigned long long)n);
 }
 
 __device__ __forceinline__ unsigned long long atomic_load_key(const unsigned long long* p) {
     return __hip_atomic_load(p, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
 }
 
-// Probe the table for `key`.  64 lanes scan 64 slots per step.  Returns the
-// slot index if found, or ~0u if hit an empty slot (absent), scanning at most
-// 16 * 64 slots.  `out_val` = node id when found.
-__device__ int probe_find(const GpuTreeDev& T, unsigned long long key, uint32_t* out_val, int lane) {
+// Per-lane linear probe: entry id if `key` is present, -1 at first empty.
+__device__ int probe_find_lane(const GpuTreeDev& T, unsigned long long key) {
     uint32_t base = (uint32_t)(key & T.table_mask);
-    for (int step = 0; step < 16; ++step) {
-        uint32_t slot = (base + step * WAVE + lane) & T.table_mask;
+    for (uint32_t i = 0; i < 256u; ++i) {
+        uint32_t slot = (base + i) & T.table_mask;
         unsigned long long k = atomic_load_key(&T.table_keys[slot]);
-        unsigned long long found = __ballot(k == key);
-        if (found) {
-            int src = __ffsll((long long)found) - 1;
-            uint32_t fslot = (base + step * WAVE + src) & T.table_mask;
-            *out_val = __hip_atomic_load(&T.table_vals[fslot], __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-            return (int)fslot;
-        }
-        unsigned long long empt = __ballot(k == EMPTY_KEY);
-        if (empt) {
-            // key would have been placed at/before the first empty slot
-            return -1;
-        }
+        if (k == key)
+            return (int)__hip_atomic_load(&T.table_vals[slot], __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        if (k == EMPTY_KEY) return -1;
     }
-    return -1;  // table pathologically full in this neighborhood
+    return -1;
 }
 
 // Insert `key`->`val` (lane-0 driven).  Two-step publication so a losing
@@ -177,28 +181,63 @@ smg_tree_match_insert(GpuTreeDev T, BatchArgs A) {
     uint32_t n_tokens = end - beg;
     uint32_t n_pages = n_tokens / T.page_size;
     if (n_pages > T.max_pages) n_pages = T.max_pages;
+    if (n_pages > MAX_LDS_PAGES) n_pages = MAX_LDS_PAGES;
 
-    // ---- phase A: match walk ------------------------------------------------
-    uint32_t cur = 0;  // root
+    __shared__ unsigned long long s_keys[MAX_LDS_PAGES];
+    __shared__ int s_ids[MAX_LDS_PAGES];
+
+    // ---- phase A: parallel prefix match -----------------------------------
+    // Lane p computes the chain key for depth (chunk*64 + p) and probes it;
+    // all depths in a chunk resolve in ONE memory round trip instead of a
+    // dependent walk.  matched = trailing-ones of the presence ballot.
     uint32_t matched_pages = 0;
-    uint32_t deepest_tenanted = 0xffffffffu;
-    unsigned long long deepest_mask = 0;
-    for (uint32_t p = 0; p < n_pages; ++p) {
-        unsigned long long h = page_hash_wave(A.tokens + beg + p * T.page_size,
-                                              (int)T.page_size, lane);
-        unsigned long long key = child_key(cur, h);
-        uint32_t nid;
-        int slot = probe_find(T, key, &nid, lane);
-        if (slot < 0) break;
-        unsigned long long tenants =
-            __hip_atomic_load(&T.node_tenants[nid], __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-        if (tenants == 0ull) break;  // fully evicted: prefix not cached anywhere
-        matched_pages = p + 1;
-        deepest_tenanted = nid;
-        deepest_mask = tenants;
-        cur = nid;
+    int deepest_id = -1;
+    unsigned long long carry = 0;       // chain sum over previous chunks
+    unsigned long long chunk_scale = 1; // W^(64*chunk)
+    bool open = true;                   // no gap seen yet
+    for (uint32_t base = 0; base < n_pages; base += WAVE) {
+        uint32_t page = base + (uint32_t)lane;
+        bool valid = page < n_pages;
+        unsigned long long contrib = 0;
+        if (valid)
+            contrib = lane_page_hash(A.tokens + beg + page * T.page_size, (int)T.page_size) *
+                      c_wpow[lane] * chunk_scale;
+        unsigned long long scan = wave_prefix_sum_u64(contrib, lane);
+        unsigned long long chain = carry + scan;
+        unsigned long long key = valid ? chain_key(chain, page) : 0;
+        int id = -1;
+        bool tenanted = false;
+        if (valid) {
+            id = probe_find_lane(T, key);
+            if (id >= 0)
+                tenanted = __hip_atomic_load(&T.node_tenants[id], __ATOMIC_RELAXED,
+                                             __HIP_MEMORY_SCOPE_AGENT) != 0ull;
+            s_keys[page] = key;
+            s_ids[page] = id;
+        }
+        unsigned long long ok = __ballot(valid && id >= 0 && tenanted);
+        uint32_t chunk_n = min(n_pages - base, (uint32_t)WAVE);
+        uint32_t run = (~ok == 0ull) ? 64u : (uint32_t)(__ffsll((long long)~ok) - 1);
+        if (run > chunk_n) run = chunk_n;
+        if (open) {
+            matched_pages += run;
+            if (run > 0) {
+                int last_id = __shfl(id, (int)(run - 1), WAVE);
+                deepest_id = last_id;
+            }
+            if (run < chunk_n) open = false;
+        }
+        // chunk carry: total of this chunk's contributions (lane 63's scan)
+        carry += __shfl(scan, WAVE - 1, WAVE);
+        chunk_scale *= c_w64[0];
     }
+    __syncthreads();
     uint32_t matched_tokens = matched_pages * T.page_size;
+    uint32_t deepest_tenanted = deepest_id >= 0 ? (uint32_t)deepest_id : 0xffffffffu;
+    unsigned long long deepest_mask =
+        deepest_id >= 0
+            ? __hip_atomic_load(&T.node_tenants[deepest_id], __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT)
+            : 0ull;
 
     // MRU healthy tenant of the deepest matched node (lane-parallel argmax ts)
     int matched_tenant = -1;
@@ -266,40 +305,27 @@ smg_tree_match_insert(GpuTreeDev T, BatchArgs A) {
     }
     if (!A.do_insert || selected < 0) return;
 
-    // ---- phase B: insert path for `selected` --------------------------------
+    // ---- phase B: parallel insert of the whole path for `selected` ---------
+    // Each lane owns one page depth: reuse the id probed in phase A, or
+    // allocate+publish a fresh entry, then attribute tenant bit + LRU stamp.
+    // Fully parallel — entries are independent under the chain-key scheme.
     uint32_t now = 0;
     if (lane == 0) now = atomicAdd(T.clock_, 1u) + 1u;
     now = __shfl(now, 0, WAVE);
     unsigned long long tbit = 1ull << selected;
-    cur = 0;
-    for (uint32_t p = 0; p < n_pages; ++p) {
-        unsigned long long h = page_hash_wave(A.tokens + beg + p * T.page_size,
-                                              (int)T.page_size, lane);
-        unsigned long long key = child_key(cur, h);
-        uint32_t nid;
-        int slot = probe_find(T, key, &nid, lane);  // slot uniform across the wave
-        int winner;
-        if (slot >= 0) {
-            winner = (int)nid;  // all lanes hold nid from probe_find
-        } else {
-            winner = -1;
-            if (lane == 0) {
-                uint32_t fresh = atomicAdd(T.next_node, 1u);
-                if (fresh < T.node_cap) {
-                    T.node_parent[fresh] = cur;
-                    uint32_t oslot;
-                    winner = probe_insert(T, key, fresh, &oslot);
-                    if (winner == (int)fresh) T.node_slot[fresh] = oslot;
-                }
-            }
-            winner = __shfl(winner, 0, WAVE);
+    for (uint32_t base = 0; base < n_pages; base += WAVE) {
+        uint32_t page = base + (uint32_t)lane;
+        if (page >= n_pages) continue;
+        int id = s_ids[page];
+        if (id < 0) {
+            uint32_t fresh = atomicAdd(T.next_node, 1u);
+            if (fresh >= T.node_cap) continue;  // pool exhausted
+            uint32_t oslot;
+            id = probe_insert(T, s_keys[page], fresh, &oslot);
+            if (id < 0) continue;
         }
-        if (winner < 0) return;  // pool exhausted: stop attributing
-        if (lane == 0) {
-            atomicOr(&T.node_tenants[winner], tbit);
-            atomicMax(&T.node_ts[(size_t)winner * WAVE + selected], now);
-        }
-        cur = (uint32_t)winner;
+        atomicOr(&T.node_tenants[id], tbit);
+        atomicMax(&T.node_ts[(size_t)id * WAVE + selected], now);
     }
 }
 
@@ -409,11 +435,16 @@ extern "C" void* smg_gpu_tree_create(int device, uint32_t node_cap, uint32_t tab
     HIP_CHECK(hipMemset(D.node_ts, 0, sizeof(uint32_t) * (size_t)node_cap * WAVE));
     uint32_t init[2] = {1u, 0u};  // node 0 = root
     HIP_CHECK(hipMemcpy(D.next_node, init, sizeof(init), hipMemcpyHostToDevice));
-    // polynomial powers for the page hash
-    unsigned long long pows[WAVE];
+    // polynomial powers: C^j for the in-page hash, W^p + W^64 for the chain
+    unsigned long long pows[WAVE], wpows[WAVE], w64;
     pows[0] = 1ull;
     for (int i = 1; i < WAVE; ++i) pows[i] = pows[i - 1] * 0x100000001B3ull;
+    wpows[0] = 1ull;
+    for (int i = 1; i < WAVE; ++i) wpows[i] = wpows[i - 1] * CHAIN_W;
+    w64 = wpows[WAVE - 1] * CHAIN_W;
     HIP_CHECK(hipMemcpyToSymbol(HIP_SYMBOL(c_pow), pows, sizeof(pows)));
+    HIP_CHECK(hipMemcpyToSymbol(HIP_SYMBOL(c_wpow), wpows, sizeof(wpows)));
+    HIP_CHECK(hipMemcpyToSymbol(HIP_SYMBOL(c_w64), &w64, sizeof(w64)));
     // batch buffers
     HIP_CHECK(hipHostMalloc((void**)&t->h_tokens, sizeof(uint32_t) * max_batch_tokens));
     HIP_CHECK(hipHostMalloc((void**)&t->h_offsets, sizeof(uint32_t) * (max_batch_reqs + 1)));

@@ -22,6 +22,15 @@ static inline uint64_t mix64(uint64_t x) {
     return x ^ (x >> 31);
 }
 
+// chain-key schedule — MUST match gpu_tree.hip exactly
+static const uint64_t CHAIN_W = 0xA24BAED4963EE407ull;
+static const uint64_t CHAIN_GOLD = 0x9E3779B97F4A7C15ull;
+
+static inline uint64_t chain_key_of(uint64_t chain, uint32_t depth) {
+    uint64_t k = mix64(chain ^ ((uint64_t)(depth + 1) * CHAIN_GOLD));
+    return k < 2 ? k + 2 : k;
+}
+
 struct MatchOut {
     int tenant = -1;  // tenant slot, -1 = none
     uint32_t matched = 0;
@@ -49,10 +58,12 @@ class HostTree {
         MatchOut out;
         out.input = n;
         uint32_t pages = n / page_size_;
-        uint32_t cur = 0;
         uint32_t now = ++clock_;
+        uint64_t chain = 0, wpow = 1;
         for (uint32_t p = 0; p < pages; ++p) {
-            uint64_t key = child_key(cur, page_hash(toks + p * page_size_, page_size_));
+            chain += page_hash(toks + p * page_size_, page_size_) * wpow;
+            wpow *= CHAIN_W;
+            uint64_t key = chain_key_of(chain, p);
             auto it = children_.find(key);
             if (it == children_.end()) break;
             Node& node = nodes_[it->second];
@@ -72,7 +83,6 @@ class HostTree {
             if (touch && best >= 0) node.ts[best] = now;
             out.tenant = best;
             out.matched = (p + 1) * page_size_;
-            cur = it->second;
         }
         return out;
     }
@@ -84,8 +94,11 @@ class HostTree {
         uint32_t added = 0;
         uint32_t now = ++clock_;
         uint64_t bit = 1ull << tenant;
+        uint64_t chain = 0, wpow = 1;
         for (uint32_t p = 0; p < pages; ++p) {
-            uint64_t key = child_key(cur, page_hash(toks + p * page_size_, page_size_));
+            chain += page_hash(toks + p * page_size_, page_size_) * wpow;
+            wpow *= CHAIN_W;
+            uint64_t key = chain_key_of(chain, p);
             auto it = children_.find(key);
             uint32_t nid;
             if (it == children_.end()) {

@@ -25,7 +25,7 @@ class GpuTokenTree:
         device: int = 0,
         capacity: int = 1 << 22,
         table_size: Optional[int] = None,
-        max_pages: int = 4096,
+        max_pages: int = 1024,
         max_batch_reqs: int = 4096,
         max_batch_tokens: int = 1 << 22,
     ):
