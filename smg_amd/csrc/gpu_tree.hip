@@ -126,8 +126,10 @@ __device__ int probe_insert(const GpuTreeDev& T, unsigned long long key, uint32_
     uint32_t base = (uint32_t)(key & T.table_mask);
     for (uint32_t i = 0; i < 1024u; ++i) {
         uint32_t slot = (base + i) & T.table_mask;
-        unsigned long long k =
-            __hip_atomic_load(&T.table_keys[slot], __ATOMIC_ACQUIRE, __HIP_MEMORY_SCOPE_AGENT);
+        // relaxed scan: acquire loads cost 2-3x per hop and invalidate the
+        // CU's L1 chip-wide (MI355X_MICROARCH.md §polling); ordering comes
+        // from the CAS below and the val-before-key publication
+        unsigned long long k = atomic_load_key(&T.table_keys[slot]);
         if (k == key) {
             *out_slot = slot;
             return (int)__hip_atomic_load(&T.table_vals[slot], __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
