@@ -361,10 +361,9 @@ def to_router_config(argv: Optional[List[str]] = None) -> RouterConfig:
 def main(argv: Optional[List[str]] = None) -> None:
     raw_argv = list(sys.argv[1:] if argv is None else argv)
     cfg = to_router_config(argv)
-    logging.basicConfig(
-        level=getattr(logging, cfg.log_level.upper(), logging.INFO),
-        format="%(asctime)s %(levelname)s %(name)s %(message)s",
-    )
+    from .observability.logging import setup_logging
+
+    setup_logging(cfg.log_level, cfg.log_json, cfg.log_dir)
     # `smg serve`: spawn local workers next to the router (reference
     # bindings/python serve.py spawns engine workers + router together)
     n_local = 0

@@ -430,6 +430,33 @@ def build_app(ctx: AppContext) -> web.Application:
     app.router.add_get("/get_server_info", get_server_info)
     app.router.add_get("/metrics", metrics_endpoint)
 
+    async def start_profile(request):
+        out = {}
+        for w in ctx.worker_registry.all():
+            if w.url.startswith("grpc"):
+                try:
+                    router = ctx.router_manager.default_router
+                    client = router.pool.get(w.url)
+                    out[w.url] = await client._unary("StartProfile", {})
+                except Exception as exc:
+                    out[w.url] = {"error": str(exc)}
+        # MI355X-native equivalent: rocprofv3 wraps the worker process; this
+        # passthrough asks engines to start their own profilers
+        return web.json_response({"status": "requested", "workers": out})
+
+    async def stop_profile(request):
+        out = {}
+        for w in ctx.worker_registry.all():
+            if w.url.startswith("grpc"):
+                try:
+                    client = ctx.router_manager.default_router.pool.get(w.url)
+                    out[w.url] = await client._unary("StopProfile", {})
+                except Exception as exc:
+                    out[w.url] = {"error": str(exc)}
+        return web.json_response({"status": "requested", "workers": out})
+
+    app.router.add_post("/start_profile", start_profile)
+    app.router.add_post("/stop_profile", stop_profile)
     app.router.add_post("/flush_cache", flush_cache)
     app.router.add_get("/get_loads", get_loads)
     app.router.add_post("/parse/function_call", parse_function_call)
@@ -528,6 +555,30 @@ async def startup(config: RouterConfig, serve: bool = True) -> AppContext:
         ctx.tenant_rate_limiter = RateLimitManager(
             TenantRateLimitSettings.from_yaml(config.tenant_rate_limit.config_path)
         )
+    if (config.tokenizer_path or config.model_path) and not config.disable_tokenizer_autoload:
+        from ..tokenizer.registry import TokenizerRegistry
+
+        ctx.tokenizer_registry = TokenizerRegistry()
+        path = config.tokenizer_path or config.model_path
+        try:
+            import os as _os
+
+            tok_json = path if path.endswith(".json") else _os.path.join(path, "tokenizer.json")
+            if _os.path.exists(tok_json):
+                try:
+                    # prefer the gfx950 batch-BPE tokenizer when the vocab is BPE
+                    from ..tokenizer.gpu_bpe import GpuBpeTokenizer
+
+                    tok = GpuBpeTokenizer(tok_json)
+                except Exception:
+                    tok = None
+                if tok is None:
+                    ctx.tokenizer_registry.load(config.model_path or "default", tok_json)
+                else:
+                    ctx.tokenizer_registry.add(config.model_path or "default", tok)
+                log.info("tokenizer loaded from %s", tok_json)
+        except Exception as exc:
+            log.warning("tokenizer autoload failed: %s", exc)
     ctx.router_manager = RouterManager(ctx, config)
     if config.mesh.enabled:
         from ..mesh.adapters import MeshAdapters
