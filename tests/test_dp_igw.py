@@ -31,10 +31,21 @@ def test_dp_aware_rank_suffix(runner):
         ctx.worker_registry.register(w)
         router = GrpcRouter(ctx, cfg)
         try:
+            # capture the request id the engine actually received
+            seen = []
+            orig = adapter.generate
+
+            def spy(req):
+                seen.append(req.request_id)
+                return orig(req)
+
+            adapter.generate = spy
             resp = await router.route(_req("/v1/completions", {"model": "m", "prompt": "x", "max_tokens": 1}))
             assert resp.status == 200
-            # rank 1 had the lowest load; its counter was bumped at selection
-            assert w.dp_loads[1] >= 1
+            # rank 1 had the lowest load: request stamped `_dp1`
+            # (reference request_execution.rs:110) and the rank guard released
+            assert seen and seen[0].endswith("_dp1")
+            assert w.dp_loads == [3, 0, 2, 5]
         finally:
             await router.shutdown()
             await adapter.stop()
