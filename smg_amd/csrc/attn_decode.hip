@@ -1,0 +1,114 @@
+// Fused single-token (decode) attention for the GPU worker engine — gfx950.
+//
+// One wave per (slot, head): online-softmax over the slot's KV window.
+// Replaces torch sdpa in the engine's full-arena decode, where sdpa must
+// read the rectangular [slots, maxlen] window for every slot; this kernel
+// reads only kv[slot][:pos+1], so idle/short slots cost nothing — the decode
+// step becomes KV-bandwidth bound on the ACTIVE tokens.
+//
+// Layout contract (the engine's KV arena, one layer):
+//   K, V:  [n_slots, n_heads, max_seq, head_dim]  bf16, contiguous
+//   q:     [n_slots, n_heads, head_dim]           bf16, contiguous
+//   pos:   [n_slots] int32 — attend kpos <= pos[slot] (the current token's
+//          K/V must already be written at pos[slot])
+//   out:   [n_slots, n_heads, head_dim]           bf16
+//
+// head_dim <= 128 (2 elements per lane).  Round of 64 timesteps per
+// iteration: lane t computes the full q·K[t] dot (K rows stream per lane),
+// then the P·V accumulation broadcasts each lane's probability with shfl
+// while V rows load coalesced (4 B per lane).
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#include <cstdint>
+
+#define WAVE 64
+
+extern "C" __global__ void __launch_bounds__(WAVE) smg_attn_decode(
+    const __hip_bfloat16* __restrict__ q,
+    const __hip_bfloat16* __restrict__ k,
+    const __hip_bfloat16* __restrict__ v,
+    const int* __restrict__ pos,
+    __hip_bfloat16* __restrict__ out,
+    int n_slots, int n_heads, int max_seq, int head_dim, float scale) {
+    int sh = blockIdx.x;
+    int slot = sh / n_heads;
+    int head = sh % n_heads;
+    if (slot >= n_slots) return;
+    int lane = threadIdx.x;
+    int T = pos[slot] + 1;  // inclusive current position
+    if (T > max_seq) T = max_seq;
+
+    const size_t head_base = ((size_t)slot * n_heads + head) * (size_t)max_seq * head_dim;
+    const __hip_bfloat16* kh = k + head_base;
+    const __hip_bfloat16* vh = v + head_base;
+    const __hip_bfloat16* qh = q + ((size_t)slot * n_heads + head) * head_dim;
+
+    // stage q in LDS (whole wave reads it every dot)
+    __shared__ float s_q[128];
+    for (int i = lane; i < head_dim; i += WAVE) s_q[i] = (float)qh[i];
+    __syncthreads();
+
+    // accumulator: 2 output elements per lane (head_dim <= 128)
+    float acc0 = 0.f, acc1 = 0.f;
+    float m = -1e30f, l = 0.f;
+    int e0 = lane * 2, e1 = lane * 2 + 1;
+
+    for (int base = 0; base < T; base += WAVE) {
+        int t = base + lane;
+        float score = -1e30f;
+        if (t < T) {
+            const __hip_bfloat16* row = kh + (size_t)t * head_dim;
+            float d = 0.f;
+            for (int i = 0; i < head_dim; i += 2) {
+                // 4-byte bf16x2 loads per step
+                d += s_q[i] * (float)row[i] + s_q[i + 1] * (float)row[i + 1];
+            }
+            score = d * scale;
+        }
+        // online softmax across the wave's 64 scores
+        float mr = score;
+#pragma unroll
+        for (int off = 32; off > 0; off >>= 1) mr = fmaxf(mr, __shfl_xor(mr, off, WAVE));
+        float m_new = fmaxf(m, mr);
+        float alpha = __expf(m - m_new);
+        float p = (t < T) ? __expf(score - m_new) : 0.f;
+        float pr = p;
+#pragma unroll
+        for (int off = 32; off > 0; off >>= 1) pr += __shfl_xor(pr, off, WAVE);
+        l = l * alpha + pr;
+        acc0 *= alpha;
+        acc1 *= alpha;
+        // P·V: broadcast each lane's p, V rows load coalesced (4B/lane)
+        int lim = min(WAVE, T - base);
+        for (int j = 0; j < lim; ++j) {
+            float pj = __shfl(p, j, WAVE);
+            if (pj != 0.f) {
+                const __hip_bfloat16* vrow = vh + (size_t)(base + j) * head_dim;
+                if (e0 < head_dim) {
+                    acc0 += pj * (float)vrow[e0];
+                    acc1 += pj * (float)vrow[e1];
+                }
+            }
+        }
+        m = m_new;
+    }
+    float inv = l > 0.f ? 1.f / l : 0.f;
+    __hip_bfloat16* orow = out + ((size_t)slot * n_heads + head) * head_dim;
+    if (e0 < head_dim) {
+        orow[e0] = (__hip_bfloat16)(acc0 * inv);
+        orow[e1] = (__hip_bfloat16)(acc1 * inv);
+    }
+}
+
+extern "C" int smg_attn_decode_launch(const void* q, const void* k, const void* v,
+                                      const void* pos, void* out, int n_slots, int n_heads,
+                                      int max_seq, int head_dim, float scale, void* stream) {
+    if (head_dim > 128 || (head_dim & 1)) return -1;
+    dim3 grid(n_slots * n_heads);
+    hipLaunchKernelGGL(smg_attn_decode, grid, dim3(WAVE), 0, (hipStream_t)stream,
+                       (const __hip_bfloat16*)q, (const __hip_bfloat16*)k,
+                       (const __hip_bfloat16*)v, (const int*)pos, (__hip_bfloat16*)out,
+                       n_slots, n_heads, max_seq, head_dim, scale);
+    return hipGetLastError() == hipSuccess ? 0 : -2;
+}

@@ -46,6 +46,10 @@ int smg_img_on_gpu(void* p);
 int smg_img_resize_normalize(void* p, const uint8_t* in, int in_w, int in_h, int channels,
                              int out_w, int out_h, const float* mean, const float* stddev,
                              uint8_t* out_u8, float* out_f32);
+// ---- decode attention (attn_decode.hip) ----
+int smg_attn_decode_launch(const void* q, const void* k, const void* v, const void* pos,
+                           void* out, int n_slots, int n_heads, int max_seq, int head_dim,
+                           float scale, void* stream);
 }
 
 #define BPE_MAX_PIECE 64
@@ -303,6 +307,18 @@ PYBIND11_MODULE(_core, m) {
     m.doc() = "smg_amd native core: host C++ radix tree + gfx950 GPU radix tree";
     m.def("hip_device_count", &smg_hip_device_count);
     m.def("page_hash", &py_page_hash);
+    // decode attention: raw device pointers + stream (ints from torch)
+    m.def("attn_decode",
+          [](uintptr_t q, uintptr_t k, uintptr_t v, uintptr_t pos, uintptr_t out, int n_slots,
+             int n_heads, int max_seq, int head_dim, float scale, uintptr_t stream) {
+              int rc = smg_attn_decode_launch((const void*)q, (const void*)k, (const void*)v,
+                                              (const void*)pos, (void*)out, n_slots, n_heads,
+                                              max_seq, head_dim, scale, (void*)stream);
+              if (rc != 0) throw std::runtime_error("attn_decode launch failed rc=" + std::to_string(rc));
+          },
+          py::arg("q"), py::arg("k"), py::arg("v"), py::arg("pos"), py::arg("out"),
+          py::arg("n_slots"), py::arg("n_heads"), py::arg("max_seq"), py::arg("head_dim"),
+          py::arg("scale"), py::arg("stream"));
 
     py::class_<PyHostTree>(m, "HostTokenTree")
         .def(py::init<uint32_t>(), py::arg("page_size") = 16)

@@ -150,6 +150,21 @@ class TorchEngine:
         self._pc_slot_keys: Dict[int, List[int]] = {}  # slot -> its hash keys
         self.prefix_cache_hits = 0
         self.prefix_cache_miss = 0
+        # fused gfx950 decode-attention kernel (csrc/attn_decode.hip): reads
+        # only kv[slot][:pos+1] per slot instead of sdpa's rectangular window
+        self._hip_attn = None
+        if self.device.type == "cuda" and self.dtype == torch.bfloat16 and c.head_dim <= 128:
+            try:
+                from .. import _core
+
+                if hasattr(_core, "attn_decode"):
+                    self._hip_attn = _core.attn_decode
+                    self._pos_i32 = torch.zeros(c.max_slots, dtype=torch.int32, device=self.device)
+                    self._attn_out = torch.zeros(
+                        c.max_slots, c.n_heads, c.head_dim, device=self.device, dtype=self.dtype
+                    )
+            except ImportError:
+                pass
 
     # ---- API -------------------------------------------------------------
     def submit(self, tokens: List[int], max_new_tokens: int, rid: Optional[str] = None) -> str:
@@ -364,17 +379,33 @@ class TorchEngine:
         S = c.max_slots
         pos = self.seq_len  # [S] current length == write position
         freqs = self.freqs_cis[pos].view(S, 1, 1, -1)
-        kpos = torch.arange(maxlen, device=self.device)
-        mask = (kpos.unsqueeze(0) <= pos.unsqueeze(1)).unsqueeze(1).unsqueeze(1)
+        use_hip = self._hip_attn is not None
+        if use_hip:
+            self._pos_i32.copy_(pos.to(torch.int32))
+            stream = torch.cuda.current_stream().cuda_stream
+            scale = 1.0 / math.sqrt(c.head_dim)
+        else:
+            kpos = torch.arange(maxlen, device=self.device)
+            mask = (kpos.unsqueeze(0) <= pos.unsqueeze(1)).unsqueeze(1).unsqueeze(1)
         h = self.embed[self._last_tok.unsqueeze(1)]  # [S, 1, D]
         for li, layer in enumerate(self.layers):
             q, k, v = self._qkv(h, layer, freqs)
             self.kv[li, 0, self._arange_slots, :, pos] = k[:, :, 0]
             self.kv[li, 1, self._arange_slots, :, pos] = v[:, :, 0]
-            kk = self.kv[li, 0][:, :, :maxlen]
-            vv = self.kv[li, 1][:, :, :maxlen]
-            attn = F.scaled_dot_product_attention(q, kk, vv, attn_mask=mask)
-            h = h + attn.transpose(1, 2).reshape(S, 1, c.d_model) @ layer.wo
+            if use_hip:
+                qc = q.reshape(S, c.n_heads, c.head_dim).contiguous()
+                self._hip_attn(
+                    qc.data_ptr(), self.kv[li, 0].data_ptr(), self.kv[li, 1].data_ptr(),
+                    self._pos_i32.data_ptr(), self._attn_out.data_ptr(),
+                    S, c.n_heads, c.max_seq, c.head_dim, scale, stream,
+                )
+                attn_flat = self._attn_out.view(S, 1, c.d_model)
+            else:
+                kk = self.kv[li, 0][:, :, :maxlen]
+                vv = self.kv[li, 1][:, :, :maxlen]
+                attn = F.scaled_dot_product_attention(q, kk, vv, attn_mask=mask)
+                attn_flat = attn.transpose(1, 2).reshape(S, 1, c.d_model)
+            h = h + attn_flat @ layer.wo
             h = self._mlp(h, layer)
         h = _rms(h, self.ln_f)
         logits = h[:, 0] @ self.embed.t()
