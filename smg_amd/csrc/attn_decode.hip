@@ -46,6 +46,9 @@ extern "C" __global__ void __launch_bounds__(WAVE) smg_attn_decode(
 
     // stage q in LDS (whole wave reads it every dot)
     __shared__ float s_q[128];
+    // K tile: 64 rows, padded stride so the per-lane row dots are
+    // bank-conflict-free (odd dword stride 67)
+    __shared__ __hip_bfloat16 s_k[WAVE][134];  // stride 67 dwords: odd -> conflict-free
     for (int i = lane; i < head_dim; i += WAVE) s_q[i] = (float)qh[i];
     __syncthreads();
 
@@ -55,17 +58,26 @@ extern "C" __global__ void __launch_bounds__(WAVE) smg_attn_decode(
     int e0 = lane * 2, e1 = lane * 2 + 1;
 
     for (int base = 0; base < T; base += WAVE) {
+        int lim0 = min(WAVE, T - base);
+        // coalesced tile load: all lanes fetch row r together (4 B/lane)
+        for (int r = 0; r < lim0; ++r) {
+            const __hip_bfloat16* row = kh + (size_t)(base + r) * head_dim;
+            if (e0 < head_dim) {
+                s_k[r][e0] = row[e0];
+                s_k[r][e1] = row[e1];
+            }
+        }
+        __syncthreads();
         int t = base + lane;
         float score = -1e30f;
         if (t < T) {
-            const __hip_bfloat16* row = kh + (size_t)t * head_dim;
             float d = 0.f;
-            for (int i = 0; i < head_dim; i += 2) {
-                // 4-byte bf16x2 loads per step
-                d += s_q[i] * (float)row[i] + s_q[i + 1] * (float)row[i + 1];
-            }
+#pragma unroll 8
+            for (int i = 0; i < head_dim; i += 2)
+                d += s_q[i] * (float)s_k[lane][i] + s_q[i + 1] * (float)s_k[lane][i + 1];
             score = d * scale;
         }
+        __syncthreads();
         // online softmax across the wave's 64 scores
         float mr = score;
 #pragma unroll
