@@ -57,14 +57,20 @@ extern "C" __global__ void __launch_bounds__(WAVE) smg_attn_decode(
     float m = -1e30f, l = 0.f;
     int e0 = lane * 2, e1 = lane * 2 + 1;
 
+    __shared__ __hip_bfloat16 s_v[WAVE][134];
+    __shared__ float s_p[WAVE];
     for (int base = 0; base < T; base += WAVE) {
-        int lim0 = min(WAVE, T - base);
-        // coalesced tile load: all lanes fetch row r together (4 B/lane)
-        for (int r = 0; r < lim0; ++r) {
-            const __hip_bfloat16* row = kh + (size_t)(base + r) * head_dim;
-            if (e0 < head_dim) {
-                s_k[r][e0] = row[e0];
-                s_k[r][e1] = row[e1];
+        int lim = min(WAVE, T - base);
+        // coalesced K+V tile load: all lanes fetch row r together (4 B/lane),
+        // many rows in flight
+        if (e0 < head_dim) {
+            for (int r = 0; r < lim; ++r) {
+                const __hip_bfloat16* krow = kh + (size_t)(base + r) * head_dim;
+                const __hip_bfloat16* vrow = vh + (size_t)(base + r) * head_dim;
+                s_k[r][e0] = krow[e0];
+                s_k[r][e1] = krow[e1];
+                s_v[r][e0] = vrow[e0];
+                s_v[r][e1] = vrow[e1];
             }
         }
         __syncthreads();
@@ -77,7 +83,6 @@ extern "C" __global__ void __launch_bounds__(WAVE) smg_attn_decode(
                 d += s_q[i] * (float)s_k[lane][i] + s_q[i + 1] * (float)s_k[lane][i + 1];
             score = d * scale;
         }
-        __syncthreads();
         // online softmax across the wave's 64 scores
         float mr = score;
 #pragma unroll
@@ -85,24 +90,24 @@ extern "C" __global__ void __launch_bounds__(WAVE) smg_attn_decode(
         float m_new = fmaxf(m, mr);
         float alpha = __expf(m - m_new);
         float p = (t < T) ? __expf(score - m_new) : 0.f;
+        s_p[lane] = p;
         float pr = p;
 #pragma unroll
         for (int off = 32; off > 0; off >>= 1) pr += __shfl_xor(pr, off, WAVE);
         l = l * alpha + pr;
         acc0 *= alpha;
         acc1 *= alpha;
-        // P·V: broadcast each lane's p, V rows load coalesced (4B/lane)
-        int lim = min(WAVE, T - base);
-        for (int j = 0; j < lim; ++j) {
-            float pj = __shfl(p, j, WAVE);
-            if (pj != 0.f) {
-                const __hip_bfloat16* vrow = vh + (size_t)(base + j) * head_dim;
-                if (e0 < head_dim) {
-                    acc0 += pj * (float)vrow[e0];
-                    acc1 += pj * (float)vrow[e1];
-                }
+        __syncthreads();
+        // P·V from LDS: lane owns output elements (e0, e1)
+        if (e0 < head_dim) {
+#pragma unroll 8
+            for (int j = 0; j < lim; ++j) {
+                float pj = s_p[j];
+                acc0 += pj * (float)s_v[j][e0];
+                acc1 += pj * (float)s_v[j][e1];
             }
         }
+        __syncthreads();
         m = m_new;
     }
     float inv = l > 0.f ? 1.f / l : 0.f;
