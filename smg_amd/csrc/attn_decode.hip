@@ -46,9 +46,9 @@ extern "C" __global__ void __launch_bounds__(WAVE) smg_attn_decode(
 
     // stage q in LDS (whole wave reads it every dot)
     __shared__ float s_q[128];
-    // K tile: 64 rows, padded stride so the per-lane row dots are
-    // bank-conflict-free (odd dword stride 67)
-    __shared__ __hip_bfloat16 s_k[WAVE][134];  // stride 67 dwords: odd -> conflict-free
+    // K/V tiles: 64 rows, stride 136 bf16 = 17 16-byte slots — rows stay
+    // 16 B aligned for dwordx4 loads and the odd slot stride staggers banks
+    __shared__ __hip_bfloat16 s_k[WAVE][136];
     for (int i = lane; i < head_dim; i += WAVE) s_q[i] = (float)qh[i];
     __syncthreads();
 
@@ -57,20 +57,22 @@ extern "C" __global__ void __launch_bounds__(WAVE) smg_attn_decode(
     float m = -1e30f, l = 0.f;
     int e0 = lane * 2, e1 = lane * 2 + 1;
 
-    __shared__ __hip_bfloat16 s_v[WAVE][134];
+    __shared__ __hip_bfloat16 s_v[WAVE][136];
     __shared__ float s_p[WAVE];
+    // vectorized tile fill: 16 B per lane, 4 rows per instruction
+    const int vec_per_row = head_dim / 8;  // 16-byte chunks per row (<=16)
+    const int rows_per_iter = WAVE / vec_per_row;
+    const int sub = lane % vec_per_row;
+    const int rofs = lane / vec_per_row;
     for (int base = 0; base < T; base += WAVE) {
         int lim = min(WAVE, T - base);
-        // coalesced K+V tile load: all lanes fetch row r together (4 B/lane),
-        // many rows in flight
-        if (e0 < head_dim) {
-            for (int r = 0; r < lim; ++r) {
-                const __hip_bfloat16* krow = kh + (size_t)(base + r) * head_dim;
-                const __hip_bfloat16* vrow = vh + (size_t)(base + r) * head_dim;
-                s_k[r][e0] = krow[e0];
-                s_k[r][e1] = krow[e1];
-                s_v[r][e0] = vrow[e0];
-                s_v[r][e1] = vrow[e1];
+        for (int r0 = 0; r0 < lim; r0 += rows_per_iter) {
+            int r = r0 + rofs;
+            if (r < lim) {
+                const uint4* krow = (const uint4*)(kh + (size_t)(base + r) * head_dim);
+                const uint4* vrow = (const uint4*)(vh + (size_t)(base + r) * head_dim);
+                ((uint4*)&s_k[r][0])[sub] = krow[sub];
+                ((uint4*)&s_v[r][0])[sub] = vrow[sub];
             }
         }
         __syncthreads();
@@ -121,7 +123,7 @@ extern "C" __global__ void __launch_bounds__(WAVE) smg_attn_decode(
 extern "C" int smg_attn_decode_launch(const void* q, const void* k, const void* v,
                                       const void* pos, void* out, int n_slots, int n_heads,
                                       int max_seq, int head_dim, float scale, void* stream) {
-    if (head_dim > 128 || (head_dim & 1)) return -1;
+    if (head_dim > 128 || (head_dim & 7)) return -1;
     dim3 grid(n_slots * n_heads);
     hipLaunchKernelGGL(smg_attn_decode, grid, dim3(WAVE), 0, (hipStream_t)stream,
                        (const __hip_bfloat16*)q, (const __hip_bfloat16*)k,
