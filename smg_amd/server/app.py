@@ -330,24 +330,25 @@ async def parse_reasoning(request):
 
 
 # ---- worker CRUD (reference worker/service.rs REST /workers) -------------
+def _job_queue(ctx):
+    if getattr(ctx, "job_queue", None) is None:
+        from .jobs import AuditLog, JobQueue
+
+        ctx.job_queue = JobQueue(ctx, AuditLog(enabled=not ctx.config.auth.disable_audit_logging))
+    return ctx.job_queue
+
+
 async def create_worker(request):
     ctx: AppContext = request.app[CTX_KEY]
     body, _ = await _read_json(request)
     if not body or "url" not in body:
         return web.Response(status=400, body=error_body("'url' is required"), content_type="application/json")
-    w = Worker(
-        body["url"],
-        model_id=body.get("model_id", "default"),
-        worker_type=WorkerType(body.get("worker_type", "regular")),
-        labels=body.get("labels") or {},
-        priority=int(body.get("priority", 0)),
-        cost=float(body.get("cost", 1.0)),
-        api_key=body.get("api_key"),
-        bootstrap_port=body.get("bootstrap_port"),
-        model_aliases=body.get("model_aliases") or [],
-        circuit_breaker_config=ctx.config.circuit_breaker,
-    )
-    ctx.worker_registry.register(w)
+    from .jobs import JobKind
+
+    try:
+        w = await _job_queue(ctx).submit(JobKind.ADD_WORKER, body, actor=request.get("tenant_id"))
+    except Exception as exc:
+        return web.Response(status=400, body=error_body(str(exc)), content_type="application/json")
     return web.json_response(w.to_dict(), status=201)
 
 
@@ -370,10 +371,13 @@ async def get_worker(request):
 async def delete_worker(request):
     ctx: AppContext = request.app[CTX_KEY]
     wid = request.match_info["worker_id"]
-    w = ctx.worker_registry.remove_by_url(wid) or (
-        ctx.worker_registry.remove(int(wid)) if wid.isdigit() else None
-    )
-    if w is None:
+    if wid.isdigit() and ctx.worker_registry.get(int(wid)) is not None:
+        wid = ctx.worker_registry.get(int(wid)).url
+    from .jobs import JobKind
+
+    try:
+        w = await _job_queue(ctx).submit(JobKind.REMOVE_WORKER, {"url": wid}, actor=request.get("tenant_id"))
+    except Exception:
         return web.Response(status=404, body=error_body("worker not found", 404), content_type="application/json")
     return web.json_response({"status": "removed", "url": w.url})
 
