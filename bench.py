@@ -173,7 +173,7 @@ def gateway_main(rank: int, world: int, args, device: str, backend: str):
         new_reqs = []
         while len(inflight) + len(new_reqs) < target_inflight:
             new_reqs.append(gen.make())
-            if len(new_reqs) >= 64:
+            if len(new_reqs) >= 128:
                 break
         phase_t["gen"] += time.perf_counter() - tg
         # 2) route the batch (GPU kernel: one launch)
@@ -195,13 +195,17 @@ def gateway_main(rank: int, world: int, args, device: str, backend: str):
                     local.enqueue(rid, max_new, prompt)
                 else:
                     plane.enqueue(sel, rid, max_new, prompt)
-        # 3) lockstep exchange + local step
+        # 3) ship remote work, overlap the local engine step, then collect
+        if plane is not None:
+            tp = time.perf_counter()
+            plane.tick_send()
+            phase_t["plane"] += time.perf_counter() - tp
         tl = time.perf_counter()
         events = list(local.tick())
         phase_t["local"] += time.perf_counter() - tl
         if plane is not None:
             tp = time.perf_counter()
-            for w, evs in plane.tick().items():
+            for w, evs in plane.tick_recv().items():
                 events.extend(evs)
             phase_t["plane"] += time.perf_counter() - tp
         # 4) completions
@@ -243,7 +247,8 @@ def gateway_main(rank: int, world: int, args, device: str, backend: str):
     elapsed = t1 - t0
     # stop workers and fold in their elapsed (max over ranks)
     if plane is not None:
-        plane.tick(stop=True)
+        plane.tick_send(stop=True)
+        plane.tick_recv()  # drain the workers' final event sends
         el = torch.tensor([elapsed], device=device if backend == "nccl" else "cpu")
         torch.distributed.all_reduce(el, op=torch.distributed.ReduceOp.MAX)
         elapsed = float(el[0])

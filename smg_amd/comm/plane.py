@@ -71,10 +71,10 @@ class GatewayPlane:
     def enqueue(self, worker_rank: int, rid: int, max_new: int, prompt: List[int]) -> None:
         self._pending[worker_rank].append((rid, max_new, prompt[-self.cfg.max_prompt:]))
 
-    def tick(self, stop: bool = False, barrier: bool = False) -> Dict[int, List[Tuple[int, int, int]]]:
-        """One lockstep exchange with every worker.  Returns per-rank events
-        [(rid, token, flags)].  stop=True signals workers to exit; barrier=True
-        asks them to join a dist.barrier() + device sync after this tick."""
+    def tick_send(self, stop: bool = False, barrier: bool = False) -> None:
+        """Phase 1: ship this tick's requests to every worker (the workers
+        start their engine step as soon as the send lands, overlapping the
+        gateway's own local work)."""
         cfg = self.cfg
         for w in self.worker_ranks:
             buf = self._send[w]
@@ -90,10 +90,15 @@ class GatewayPlane:
                 rows[base + 2] = len(prompt)
                 rows[base + 3: base + 3 + len(prompt)] = torch.tensor(prompt, dtype=torch.int32)
             buf.copy_(rows.to(buf.device))
-        ops = []
-        for w in self.worker_ranks:
-            ops.append(dist.P2POp(dist.isend, self._send[w], w))
-            ops.append(dist.P2POp(dist.irecv, self._recv[w], w))
+        ops = [dist.P2POp(dist.isend, self._send[w], w) for w in self.worker_ranks]
+        if ops:
+            for work in dist.batch_isend_irecv(ops):
+                work.wait()
+
+    def tick_recv(self) -> Dict[int, List[Tuple[int, int, int]]]:
+        """Phase 2: collect every worker's event tensor."""
+        cfg = self.cfg
+        ops = [dist.P2POp(dist.irecv, self._recv[w], w) for w in self.worker_ranks]
         if ops:
             for work in dist.batch_isend_irecv(ops):
                 work.wait()
@@ -108,6 +113,15 @@ class GatewayPlane:
             out[w] = events
         return out
 
+    def tick(self, stop: bool = False, barrier: bool = False) -> Dict[int, List[Tuple[int, int, int]]]:
+        """One lockstep exchange (send + recv back-to-back).  The bench's tick
+        loop uses tick_send()/tick_recv() split around the gateway's local
+        engine step so remote workers compute concurrently."""
+        self.tick_send(stop=stop, barrier=barrier)
+        if stop:
+            return {}
+        return self.tick_recv()
+
 
 class WorkerPlane:
     """Worker-rank side: mirror of the gateway's per-tick exchange."""
@@ -118,11 +132,20 @@ class WorkerPlane:
         dev = torch.device(cfg.device)
         self._recv = torch.zeros(cfg.req_len, dtype=torch.int32, device=dev)
         self._send = torch.zeros(cfg.ev_len, dtype=torch.int32, device=dev)
+        self._send_work = None
 
     def tick(self, events: List[Tuple[int, int, int]]) -> Tuple[List[Tuple[int, int, List[int]]], bool]:
         """One lockstep exchange: sends `events` [(rid, token, flags)], receives
-        new requests.  Returns (new_requests, stop)."""
+        new requests.  Returns (new_requests, stop).
+
+        Only the request RECV is awaited here: the event send drains while the
+        gateway runs its own local engine step (its recv is posted after), so
+        worker and gateway compute concurrently.  The send handle is awaited
+        at the next tick before the buffer is reused."""
         cfg = self.cfg
+        if self._send_work is not None:
+            self._send_work.wait()
+            self._send_work = None
         ev = torch.zeros(cfg.ev_len, dtype=torch.int32)
         n = min(len(events), cfg.max_events_per_tick)
         ev[0] = n
@@ -134,8 +157,9 @@ class WorkerPlane:
             dist.P2POp(dist.irecv, self._recv, self.gateway_rank),
             dist.P2POp(dist.isend, self._send, self.gateway_rank),
         ]
-        for work in dist.batch_isend_irecv(ops):
-            work.wait()
+        works = dist.batch_isend_irecv(ops)
+        works[0].wait()  # requests arrived; step can start
+        self._send_work = works[1]
         req = self._recv.cpu()
         hdr = int(req[0])
         stop = bool(hdr & STOP_FLAG)
