@@ -577,9 +577,16 @@ async def startup(config: RouterConfig, serve: bool = True) -> AppContext:
                 except Exception:
                     tok = None
                 if tok is None:
-                    ctx.tokenizer_registry.load(config.model_path or "default", tok_json)
+                    tok = ctx.tokenizer_registry.load(config.model_path or "default", tok_json)
                 else:
                     ctx.tokenizer_registry.add(config.model_path or "default", tok)
+                if config.tokenizer_cache.enable_l1:
+                    from ..tokenizer.l1_cache import L1CachedTokenizer
+
+                    ctx.tokenizer_registry.add(
+                        config.model_path or "default",
+                        L1CachedTokenizer(tok, config.tokenizer_cache.l1_max_memory),
+                    )
                 log.info("tokenizer loaded from %s", tok_json)
         except Exception as exc:
             log.warning("tokenizer autoload failed: %s", exc)
