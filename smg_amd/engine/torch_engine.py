@@ -69,8 +69,7 @@ class _Layer:
 
         self.wqkv = w(d, 3 * d)
         self.wo = w(d, d)
-        self.w1 = w(d, f)
-        self.w3 = w(d, f)
+        self.w13 = w(d, 2 * f)  # gate+up fused: one GEMM instead of two
         self.w2 = w(f, d)
         self.ln1 = torch.ones(d, device=device, dtype=dtype)
         self.ln2 = torch.ones(d, device=device, dtype=dtype)
@@ -328,7 +327,9 @@ class TorchEngine:
 
     def _mlp(self, h, layer):
         x = _rms(h, layer.ln2)
-        return h + (F.silu(x @ layer.w1) * (x @ layer.w3)) @ layer.w2
+        gu = x @ layer.w13
+        g, u = gu.chunk(2, dim=-1)
+        return h + (F.silu(g) * u) @ layer.w2
 
     def _qkv(self, h, layer, freqs):
         c = self.cfg

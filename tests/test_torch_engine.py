@@ -92,7 +92,7 @@ class TestEngineGpu:
         # identical weights: copy CPU weights to GPU engine
         gpu.embed.copy_(cpu.embed.to(gpu.device))
         for lg, lc in zip(gpu.layers, cpu.layers):
-            for attr in ("wqkv", "wo", "w1", "w2", "w3", "ln1", "ln2"):
+            for attr in ("wqkv", "wo", "w13", "w2", "ln1", "ln2"):
                 getattr(lg, attr).copy_(getattr(lc, attr).to(gpu.device))
         gpu.ln_f.copy_(cpu.ln_f.to(gpu.device))
         prompt = list(range(32))
