@@ -224,12 +224,10 @@ class TorchEngine:
             self._seq_len_host[req.slot] = 0
 
         produced = 0
-        # prefill: chunked continuous batching under a per-step token budget,
-        # with prefix-cache restore for fresh slots
-        budget = c.prefill_chunk
+        # prefill: prefix-cache restore for fresh slots, then BATCHED chunked
+        # prefill — equal-remaining requests share one forward (the common
+        # case: cache-hit suffixes of identical length)
         for slot, req in self.running.items():
-            if budget <= 0:
-                break
             if req.prefilled == 0 and self._pc_arena is not None:
                 hit = self._prefix_lookup(req.tokens)
                 if hit is not None:
@@ -240,15 +238,23 @@ class TorchEngine:
                     self._last_tok[slot] = int(req.tokens[plen - 1])
                     req.prefilled = plen
                     self.prefix_cache_hits += 1
-                    if req.prefilled >= len(req.tokens):
-                        continue
                 else:
                     self.prefix_cache_miss += 1
-            if req.prefilled < len(req.tokens):
-                chunk = req.tokens[req.prefilled: req.prefilled + budget]
-                self._prefill(slot, chunk, req.prefilled)
-                req.prefilled += len(chunk)
-                budget -= len(chunk)
+        pending = [(s, r) for s, r in self.running.items() if r.prefilled < len(r.tokens)]
+        if pending:
+            pending.sort(key=lambda sr: len(sr[1].tokens) - sr[1].prefilled)
+            min_rem = len(pending[0][1].tokens) - pending[0][1].prefilled
+            L = min(min_rem, c.prefill_chunk)
+            group = [sr for sr in pending if len(sr[1].tokens) - sr[1].prefilled >= L][:16]
+            # keep the total under ~2x budget
+            while len(group) > 1 and len(group) * L > 2 * c.prefill_chunk:
+                group.pop()
+            items = []
+            for slot, req in group:
+                items.append((slot, req.prefilled, req.tokens[req.prefilled: req.prefilled + L]))
+            self._prefill_batch(items)
+            for slot, req in group:
+                req.prefilled += L
                 if req.prefilled >= len(req.tokens):
                     self._prefix_store(slot, req.tokens)
 
@@ -343,31 +349,39 @@ class TorchEngine:
 
     @torch.no_grad()
     def _prefill(self, slot: int, tokens: List[int], start: int) -> None:
+        self._prefill_batch([(slot, start, list(tokens))])
+
+    @torch.no_grad()
+    def _prefill_batch(self, items: List[tuple]) -> None:
+        """Batched chunked prefill: `items` = [(slot, start, tokens)] with
+        equal chunk lengths.  Per-request absolute positions drive rope and a
+        per-request causal mask over each slot's own KV window."""
         c = self.cfg
-        T = len(tokens)
-        t = torch.tensor([tokens], device=self.device, dtype=torch.long)
-        h = self.embed[t]
-        pos = torch.arange(start, start + T, device=self.device)
-        freqs = self.freqs_cis[pos].view(1, 1, T, -1)
-        mask = None
-        if start:  # chunked continuation: causal mask with history offset
-            kpos = torch.arange(0, start + T, device=self.device)
-            mask = (kpos.unsqueeze(0) <= pos.unsqueeze(1)).unsqueeze(0).unsqueeze(0)
+        B = len(items)
+        L = len(items[0][2])
+        slots = torch.tensor([s for s, _, _ in items], device=self.device)
+        starts = torch.tensor([st for _, st, _ in items], device=self.device)
+        t = torch.tensor([toks for _, _, toks in items], device=self.device, dtype=torch.long)
+        h = self.embed[t]  # [B, L, D]
+        pos = starts.unsqueeze(1) + torch.arange(L, device=self.device)  # [B, L]
+        freqs = self.freqs_cis[pos].unsqueeze(1)  # [B, 1, L, D/2]
+        t_max = int(starts.max().item()) + L
+        kpos = torch.arange(t_max, device=self.device)
+        mask = kpos.view(1, 1, 1, -1) <= pos.view(B, 1, L, 1)
         for li, layer in enumerate(self.layers):
             q, k, v = self._qkv(h, layer, freqs)
-            self.kv[li, 0, slot, :, start: start + T] = k[0]
-            self.kv[li, 1, slot, :, start: start + T] = v[0]
-            kk = self.kv[li, 0, slot: slot + 1, :, : start + T]
-            vv = self.kv[li, 1, slot: slot + 1, :, : start + T]
-            if mask is None:
-                attn = F.scaled_dot_product_attention(q, kk, vv, is_causal=True)
-            else:
-                attn = F.scaled_dot_product_attention(q, kk, vv, attn_mask=mask)
-            h = h + attn.transpose(1, 2).reshape(1, T, c.d_model) @ layer.wo
+            # scatter this chunk's K/V into each request's slot window
+            self.kv[li, 0][slots[:, None], :, pos] = k.permute(0, 2, 1, 3)
+            self.kv[li, 1][slots[:, None], :, pos] = v.permute(0, 2, 1, 3)
+            kk = self.kv[li, 0].index_select(0, slots)[:, :, :t_max]
+            vv = self.kv[li, 1].index_select(0, slots)[:, :, :t_max]
+            attn = F.scaled_dot_product_attention(q, kk, vv, attn_mask=mask)
+            h = h + attn.transpose(1, 2).reshape(B, L, c.d_model) @ layer.wo
             h = self._mlp(h, layer)
-        self._seq_len_host[slot] = start + T
-        self.seq_len[slot] = start + T
-        self._last_tok[slot] = int(tokens[-1])
+        for slot, start, toks in items:
+            self._seq_len_host[slot] = start + L
+            self._last_tok[slot] = int(toks[-1])
+        self.seq_len[slots] = starts + L
 
     def _decode_core(self, maxlen: int) -> torch.Tensor:
         """Full-arena decode forward: every slot participates with a static
