@@ -21,8 +21,8 @@ void* smg_gpu_tree_create(int device, uint32_t node_cap, uint32_t table_size, ui
 void smg_gpu_tree_destroy(void* p);
 int smg_gpu_tree_run(void* p, int n_reqs, unsigned long long healthy_mask, const int* loads,
                      const int* processed, int n_workers, float cache_threshold, int imbalanced,
-                     int do_insert, int forced_tenant, int* out_selected, uint32_t* out_matched,
-                     uint32_t* out_tenant);
+                     int do_insert, int forced_tenant, int mode, int* out_selected,
+                     uint32_t* out_matched, uint32_t* out_tenant, uint32_t* out_depths);
 void smg_gpu_tree_staging(void* p, uint32_t** tokens, uint32_t** offsets);
 int smg_gpu_tree_remove_tenant(void* p, int slot);
 int smg_gpu_tree_evict_older(void* p, uint32_t cutoff);
@@ -95,7 +95,7 @@ class PyGpuTree {
                   py::array_t<uint32_t, py::array::c_style | py::array::forcecast> offsets,
                   unsigned long long healthy_mask, std::vector<int> loads, std::vector<int> processed,
                   int n_workers, float cache_threshold, bool imbalanced, bool do_insert,
-                  int forced_tenant) {
+                  int forced_tenant, int mode) {
         int n_reqs = (int)offsets.size() - 1;
         if (n_reqs < 1 || (uint32_t)n_reqs > max_batch_reqs_)
             throw std::runtime_error("batch size out of range");
@@ -110,16 +110,17 @@ class PyGpuTree {
         py::array_t<int> sel(n_reqs);
         py::array_t<uint32_t> matched(n_reqs);
         py::array_t<uint32_t> tenant(n_reqs);
+        py::array_t<uint32_t> depths(mode == 1 ? n_reqs * 64 : 0);
         int rc;
         {
             py::gil_scoped_release nogil;
             rc = smg_gpu_tree_run(h_, n_reqs, healthy_mask, loads.data(), processed.data(),
                                   n_workers, cache_threshold, imbalanced ? 1 : 0, do_insert ? 1 : 0,
-                                  forced_tenant, sel.mutable_data(), matched.mutable_data(),
-                                  tenant.mutable_data());
+                                  forced_tenant, mode, sel.mutable_data(), matched.mutable_data(),
+                                  tenant.mutable_data(), mode == 1 ? depths.mutable_data() : nullptr);
         }
         if (rc != 0) throw std::runtime_error("smg_gpu_tree_run failed rc=" + std::to_string(rc));
-        return py::make_tuple(sel, matched, tenant);
+        return py::make_tuple(sel, matched, tenant, depths);
     }
 
     void remove_tenant(int slot) {
@@ -339,7 +340,7 @@ PYBIND11_MODULE(_core, m) {
         .def("run", &PyGpuTree::run, py::arg("tokens_flat"), py::arg("offsets"),
              py::arg("healthy_mask"), py::arg("loads"), py::arg("processed"), py::arg("n_workers"),
              py::arg("cache_threshold") = 0.3f, py::arg("imbalanced") = false,
-             py::arg("do_insert") = true, py::arg("forced_tenant") = -1)
+             py::arg("do_insert") = true, py::arg("forced_tenant") = -1, py::arg("mode") = 0)
         .def("remove_tenant", &PyGpuTree::remove_tenant)
         .def("evict_older", &PyGpuTree::evict_older)
         .def("stats", &PyGpuTree::stats)

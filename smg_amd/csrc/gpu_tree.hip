@@ -167,6 +167,11 @@ struct BatchArgs {
     int n_workers;
     int do_insert;
     int forced_tenant;  // >=0: skip decision, insert for this slot (insert-only)
+    // mode 0: match+decide+insert (cache_aware).  mode 1: per-worker matched
+    // depths only (KV-event overlap scoring, event_tree.rs:571 find_matches).
+    // mode 2: clear forced_tenant's bit along the path (apply_removed).
+    int mode;
+    uint32_t* out_depths;  // mode 1: [n_reqs * 64] matched tokens per worker
     // outputs per request
     int* out_selected;       // worker slot (or -1)
     uint32_t* out_matched;   // matched token count
@@ -197,6 +202,9 @@ smg_tree_match_insert(GpuTreeDev T, BatchArgs A) {
     unsigned long long carry = 0;       // chain sum over previous chunks
     unsigned long long chunk_scale = 1; // W^(64*chunk)
     bool open = true;                   // no gap seen yet
+    // mode 1 state: this lane owns worker slot `lane`
+    uint32_t my_worker_pages = 0;
+    bool my_worker_open = true;
     for (uint32_t base = 0; base < n_pages; base += WAVE) {
         uint32_t page = base + (uint32_t)lane;
         bool valid = page < n_pages;
@@ -208,15 +216,16 @@ smg_tree_match_insert(GpuTreeDev T, BatchArgs A) {
         unsigned long long chain = carry + scan;
         unsigned long long key = valid ? chain_key(chain, page) : 0;
         int id = -1;
-        bool tenanted = false;
+        unsigned long long tmask = 0;
         if (valid) {
             id = probe_find_lane(T, key);
             if (id >= 0)
-                tenanted = __hip_atomic_load(&T.node_tenants[id], __ATOMIC_RELAXED,
-                                             __HIP_MEMORY_SCOPE_AGENT) != 0ull;
+                tmask = __hip_atomic_load(&T.node_tenants[id], __ATOMIC_RELAXED,
+                                          __HIP_MEMORY_SCOPE_AGENT);
             s_keys[page] = key;
             s_ids[page] = id;
         }
+        bool tenanted = tmask != 0ull;
         unsigned long long ok = __ballot(valid && id >= 0 && tenanted);
         uint32_t chunk_n = min(n_pages - base, (uint32_t)WAVE);
         uint32_t run = (~ok == 0ull) ? 64u : (uint32_t)(__ffsll((long long)~ok) - 1);
@@ -229,11 +238,50 @@ smg_tree_match_insert(GpuTreeDev T, BatchArgs A) {
             }
             if (run < chunk_n) open = false;
         }
+        if (A.mode == 1) {
+            // per-worker runs: 64 ballots transpose the lane-held tenant masks
+            // into per-worker presence vectors; lane w folds worker w's run
+            for (int w = 0; w < WAVE; ++w) {
+                unsigned long long okw = __ballot(valid && id >= 0 && ((tmask >> w) & 1ull));
+                if (lane == w && my_worker_open) {
+                    uint32_t runw = (~okw == 0ull) ? 64u : (uint32_t)(__ffsll((long long)~okw) - 1);
+                    if (runw > chunk_n) runw = chunk_n;
+                    my_worker_pages += runw;
+                    if (runw < chunk_n) my_worker_open = false;
+                }
+            }
+        }
         // chunk carry: total of this chunk's contributions (lane 63's scan)
         carry += __shfl(scan, WAVE - 1, WAVE);
         chunk_scale *= c_w64[0];
     }
     __syncthreads();
+    if (A.mode == 1) {
+        A.out_depths[(size_t)req * WAVE + lane] = my_worker_pages * T.page_size;
+        if (lane == 0) {
+            A.out_matched[req] = matched_pages * T.page_size;
+            A.out_selected[req] = -1;
+            A.out_tenant[req] = 0xffffffffu;
+        }
+        return;
+    }
+    if (A.mode == 2) {
+        // clear forced_tenant's attribution along the matched path
+        unsigned long long clear = ~(1ull << A.forced_tenant);
+        for (uint32_t page = lane; page < n_pages; page += WAVE) {
+            int id = s_ids[page];
+            if (id >= 0) {
+                atomicAnd(&T.node_tenants[id], clear);
+                T.node_ts[(size_t)id * WAVE + A.forced_tenant] = 0;
+            }
+        }
+        if (lane == 0) {
+            A.out_selected[req] = -1;
+            A.out_matched[req] = matched_pages * T.page_size;
+            A.out_tenant[req] = 0xffffffffu;
+        }
+        return;
+    }
     uint32_t matched_tokens = matched_pages * T.page_size;
     uint32_t deepest_tenanted = deepest_id >= 0 ? (uint32_t)deepest_id : 0xffffffffu;
     unsigned long long deepest_mask =
@@ -408,6 +456,7 @@ struct GpuTreeHost {
     int* d_loads;
     int* d_processed;
     unsigned long long* d_counts;
+    uint32_t* d_depths;
 };
 
 extern "C" void* smg_gpu_tree_create(int device, uint32_t node_cap, uint32_t table_size,
@@ -461,6 +510,7 @@ extern "C" void* smg_gpu_tree_create(int device, uint32_t node_cap, uint32_t tab
     HIP_CHECK(hipMalloc(&t->d_loads, sizeof(int) * WAVE));
     HIP_CHECK(hipMalloc(&t->d_processed, sizeof(int) * WAVE));
     HIP_CHECK(hipMalloc(&t->d_counts, sizeof(unsigned long long) * 65));
+    HIP_CHECK(hipMalloc(&t->d_depths, sizeof(uint32_t) * (size_t)max_batch_reqs * 64));
     return t;
 }
 
@@ -475,7 +525,7 @@ extern "C" void smg_gpu_tree_destroy(void* p) {
     hipHostFree(t->h_matched); hipHostFree(t->h_tenant);
     hipFree(t->d_tokens); hipFree(t->d_offsets); hipFree(t->d_selected);
     hipFree(t->d_matched); hipFree(t->d_tenant); hipFree(t->d_loads);
-    hipFree(t->d_processed); hipFree(t->d_counts);
+    hipFree(t->d_processed); hipFree(t->d_counts); hipFree(t->d_depths);
     hipStreamDestroy(t->stream);
     delete t;
 }
@@ -485,8 +535,9 @@ extern "C" void smg_gpu_tree_destroy(void* p) {
 extern "C" int smg_gpu_tree_run(void* p, int n_reqs, unsigned long long healthy_mask,
                                 const int* loads, const int* processed, int n_workers,
                                 float cache_threshold, int imbalanced, int do_insert,
-                                int forced_tenant,
-                                int* out_selected, uint32_t* out_matched, uint32_t* out_tenant) {
+                                int forced_tenant, int mode,
+                                int* out_selected, uint32_t* out_matched, uint32_t* out_tenant,
+                                uint32_t* out_depths) {
     GpuTreeHost* t = (GpuTreeHost*)p;
     if (n_reqs <= 0 || (uint32_t)n_reqs > t->max_batch_reqs) return -1;
     uint32_t n_tokens = t->h_offsets[n_reqs];
@@ -508,6 +559,8 @@ extern "C" int smg_gpu_tree_run(void* p, int n_reqs, unsigned long long healthy_
     A.n_workers = n_workers;
     A.do_insert = do_insert;
     A.forced_tenant = forced_tenant;
+    A.mode = mode;
+    A.out_depths = t->d_depths;
     A.out_selected = t->d_selected;
     A.out_matched = t->d_matched;
     A.out_tenant = t->d_tenant;
@@ -515,6 +568,9 @@ extern "C" int smg_gpu_tree_run(void* p, int n_reqs, unsigned long long healthy_
     hipMemcpyAsync(t->h_selected, t->d_selected, sizeof(int) * n_reqs, hipMemcpyDeviceToHost, s);
     hipMemcpyAsync(t->h_matched, t->d_matched, sizeof(uint32_t) * n_reqs, hipMemcpyDeviceToHost, s);
     hipMemcpyAsync(t->h_tenant, t->d_tenant, sizeof(uint32_t) * n_reqs, hipMemcpyDeviceToHost, s);
+    if (mode == 1 && out_depths)
+        hipMemcpyAsync(out_depths, t->d_depths, sizeof(uint32_t) * (size_t)n_reqs * 64,
+                       hipMemcpyDeviceToHost, s);
     if (hipStreamSynchronize(s) != hipSuccess) return -3;
     for (int i = 0; i < n_reqs; ++i) {
         out_selected[i] = t->h_selected[i];

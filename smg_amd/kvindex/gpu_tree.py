@@ -68,6 +68,7 @@ class GpuTokenTree:
         imbalanced: bool = False,
         do_insert: bool = True,
         forced_tenant: int = -1,
+        mode: int = 0,
     ):
         offsets = np.zeros(len(token_lists) + 1, dtype=np.uint32)
         for i, t in enumerate(token_lists):
@@ -75,7 +76,7 @@ class GpuTokenTree:
         flat = np.empty(int(offsets[-1]), dtype=np.uint32)
         for i, t in enumerate(token_lists):
             flat[offsets[i]: offsets[i + 1]] = np.asarray(t, dtype=np.uint32)
-        sel, matched, tenant = self._tree.run(
+        sel, matched, tenant, depths = self._tree.run(
             flat,
             offsets,
             healthy_mask,
@@ -86,8 +87,11 @@ class GpuTokenTree:
             imbalanced=imbalanced,
             do_insert=do_insert,
             forced_tenant=forced_tenant,
+            mode=mode,
         )
         self._approx_nodes += int(offsets[-1]) // self.page_size  # upper bound
+        if mode == 1:
+            return sel, matched, tenant, depths.reshape(len(token_lists), 64)
         return sel, matched, tenant
 
     # ---- tree API (parity with pytree) -----------------------------------
@@ -150,6 +154,24 @@ class GpuTokenTree:
             do_insert=True,
         )
         return [idx_of_slot.get(int(s)) if int(s) >= 0 else None for s in sel]
+
+    # ---- KV-event overlap scoring (mode 1/2 kernel paths) -----------------
+    def match_worker_depths(self, token_lists):
+        """[n, 64] matched token counts per worker slot (one kernel launch;
+        reference event_tree.rs:571 find_matches semantics)."""
+        _s, _m, _t, depths = self._run(
+            token_lists, healthy_mask=(1 << 64) - 1, loads=[0] * 64, processed=[0] * 64,
+            n_workers=64, do_insert=False, mode=1,
+        )
+        return depths
+
+    def remove_path(self, tokens, tenant: str) -> None:
+        """Clear `tenant`'s attribution along this token path (apply_removed)."""
+        slot = self.slots.existing_slot(tenant)
+        if slot is None:
+            return
+        self._run([tokens], healthy_mask=(1 << 64) - 1, loads=[0] * 64, processed=[0] * 64,
+                  n_workers=64, do_insert=False, forced_tenant=slot, mode=2)
 
     # ---- maintenance -----------------------------------------------------
     def remove_tenant(self, tenant: str) -> None:
