@@ -25,6 +25,7 @@ int smg_gpu_tree_run(void* p, int n_reqs, unsigned long long healthy_mask, const
                      uint32_t* out_matched, uint32_t* out_tenant, uint32_t* out_depths);
 void smg_gpu_tree_staging(void* p, uint32_t** tokens, uint32_t** offsets);
 int smg_gpu_tree_remove_tenant(void* p, int slot);
+int smg_gpu_tree_clear_entries(void* p, const unsigned long long* keys, int n, int slot);
 int smg_gpu_tree_evict_older(void* p, uint32_t cutoff);
 int smg_gpu_tree_stats(void* p, unsigned long long* out);
 int smg_gpu_tree_clear(void* p);
@@ -125,6 +126,10 @@ class PyGpuTree {
 
     void remove_tenant(int slot) {
         if (smg_gpu_tree_remove_tenant(h_, slot)) throw std::runtime_error("remove_tenant failed");
+    }
+    void clear_entries(py::array_t<uint64_t, py::array::c_style | py::array::forcecast> keys, int slot) {
+        if (smg_gpu_tree_clear_entries(h_, (const unsigned long long*)keys.data(), (int)keys.size(), slot))
+            throw std::runtime_error("clear_entries failed");
     }
     void evict_older(uint32_t cutoff) {
         if (smg_gpu_tree_evict_older(h_, cutoff)) throw std::runtime_error("evict failed");
@@ -342,6 +347,7 @@ PYBIND11_MODULE(_core, m) {
              py::arg("cache_threshold") = 0.3f, py::arg("imbalanced") = false,
              py::arg("do_insert") = true, py::arg("forced_tenant") = -1, py::arg("mode") = 0)
         .def("remove_tenant", &PyGpuTree::remove_tenant)
+        .def("clear_entries", &PyGpuTree::clear_entries)
         .def("evict_older", &PyGpuTree::evict_older)
         .def("stats", &PyGpuTree::stats)
         .def("clear", &PyGpuTree::clear);

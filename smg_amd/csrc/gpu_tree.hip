@@ -407,6 +407,19 @@ extern "C" __global__ void smg_tree_evict_older(GpuTreeDev T, uint32_t cutoff) {
     }
 }
 
+// Clear one tenant's bit on explicit entries (KV-event apply_removed: the
+// host resolves removed block hashes to their stored chain keys).
+extern "C" __global__ void smg_tree_clear_entries(GpuTreeDev T, const unsigned long long* keys,
+                                                  int n, int slot_idx) {
+    int i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    int id = probe_find_lane(T, keys[i]);
+    if (id >= 0) {
+        atomicAnd(&T.node_tenants[id], ~(1ull << slot_idx));
+        T.node_ts[(size_t)id * WAVE + slot_idx] = 0;
+    }
+}
+
 // Count live nodes / per-tenant attributed nodes (stats + tie-breaks).
 extern "C" __global__ void smg_tree_stats(GpuTreeDev T, unsigned long long* out_counts /*65*/) {
     uint32_t n = *T.next_node;
@@ -584,6 +597,19 @@ extern "C" void smg_gpu_tree_staging(void* p, uint32_t** tokens, uint32_t** offs
     GpuTreeHost* t = (GpuTreeHost*)p;
     *tokens = t->h_tokens;
     *offsets = t->h_offsets;
+}
+
+extern "C" int smg_gpu_tree_clear_entries(void* p, const unsigned long long* keys, int n, int slot) {
+    GpuTreeHost* t = (GpuTreeHost*)p;
+    if (n <= 0) return 0;
+    unsigned long long* d_keys = nullptr;
+    if (hipMalloc(&d_keys, sizeof(unsigned long long) * n) != hipSuccess) return -1;
+    hipMemcpyAsync(d_keys, keys, sizeof(unsigned long long) * n, hipMemcpyHostToDevice, t->stream);
+    hipLaunchKernelGGL(smg_tree_clear_entries, dim3((n + 255) / 256), dim3(256), 0, t->stream,
+                       t->dev, d_keys, n, slot);
+    int rc = hipStreamSynchronize(t->stream) == hipSuccess ? 0 : -2;
+    hipFree(d_keys);
+    return rc;
 }
 
 extern "C" int smg_gpu_tree_remove_tenant(void* p, int slot) {

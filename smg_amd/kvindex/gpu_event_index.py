@@ -11,6 +11,9 @@ from __future__ import annotations
 
 from typing import Dict, List, Optional, Sequence
 
+import numpy as np
+
+from .chain_keys import chain_keys
 from .event_index import compute_content_hashes
 from .gpu_tree import GpuTokenTree
 
@@ -31,6 +34,8 @@ class GpuPositionalIndexer:
         self._capacity = capacity
         self._has_events: Dict[str, bool] = {}
         self._block_counts: Dict[str, Dict[str, int]] = {}
+        # chained content hash -> table key (for removal by value)
+        self._hash_keys: Dict[str, Dict[int, int]] = {}
 
     def _tree(self, model: str) -> GpuTokenTree:
         t = self._trees.get(model)
@@ -47,7 +52,12 @@ class GpuPositionalIndexer:
             # offset appends need the preceding prefix; engines report full
             # prefixes here (the monitor always applies from 0)
             return
-        self._tree(model).insert(_hashes_to_tokens(hashes), url)
+        hash_tokens = _hashes_to_tokens(hashes)
+        self._tree(model).insert(hash_tokens, url)
+        keys = chain_keys(hash_tokens, 2)
+        hk = self._hash_keys.setdefault(model, {})
+        for h, k in zip(hashes, keys):
+            hk[h] = k
         self._has_events[model] = True
         bc = self._block_counts.setdefault(model, {})
         bc[url] = bc.get(url, 0) + len(hashes)
@@ -56,9 +66,15 @@ class GpuPositionalIndexer:
         tree = self._trees.get(model)
         if tree is None:
             return
-        tree.remove_path(_hashes_to_tokens(hashes), url)
+        slot = tree.slots.existing_slot(url)
+        if slot is None:
+            return
+        hk = self._hash_keys.get(model, {})
+        keys = [hk[h] for h in hashes if h in hk]
+        if keys:
+            tree._tree.clear_entries(np.asarray(keys, dtype=np.uint64), slot)
         bc = self._block_counts.setdefault(model, {})
-        bc[url] = max(0, bc.get(url, 0) - len(hashes))
+        bc[url] = max(0, bc.get(url, 0) - len(keys))
 
     def remove_worker(self, model: str, url: str) -> None:
         tree = self._trees.get(model)
