@@ -25,11 +25,18 @@ class AppContext:
         self.config = config
         self.metrics = metrics or GatewayMetrics()
         self.worker_registry = WorkerRegistry()
-        indexer = (
-            PositionalIndexer(block_size=config.policy.block_size)
-            if config.connection_mode == ConnectionMode.GRPC
-            else None
-        )
+        indexer = None
+        if config.connection_mode == ConnectionMode.GRPC:
+            indexer = PositionalIndexer(block_size=config.policy.block_size)
+            if config.policy.gpu_tree:
+                try:
+                    from ..kvindex import gpu_available
+                    from ..kvindex.gpu_event_index import GpuPositionalIndexer
+
+                    if gpu_available():
+                        indexer = GpuPositionalIndexer(block_size=config.policy.block_size)
+                except ImportError:
+                    pass
         self.policy_registry = PolicyRegistry(
             config.policy,
             indexer=indexer,
