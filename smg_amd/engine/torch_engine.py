@@ -117,8 +117,6 @@ class TorchEngine:
             inv = 1.0 / (10000.0 ** (torch.arange(0, c.head_dim, 2, device=self.device).float() / c.head_dim))
             t = torch.arange(c.max_seq, device=self.device).float()
             freqs = torch.outer(t, inv)
-            self.cos = freqs.cos().to(self.dtype)
-            self.sin = freqs.sin().to(self.dtype)
             self.freqs_cis = torch.polar(torch.ones_like(freqs), freqs)  # complex64 [T, D/2]
         self.seq_len = torch.zeros(c.max_slots, dtype=torch.long, device=self.device)
         self._seq_len_host = [0] * c.max_slots
@@ -366,16 +364,22 @@ class TorchEngine:
         pos = starts.unsqueeze(1) + torch.arange(L, device=self.device)  # [B, L]
         freqs = self.freqs_cis[pos].unsqueeze(1)  # [B, 1, L, D/2]
         t_max = int(starts.max().item()) + L
-        kpos = torch.arange(t_max, device=self.device)
-        mask = kpos.view(1, 1, 1, -1) <= pos.view(B, 1, L, 1)
+        fresh = int(starts.max().item()) == 0  # cold prefill: flash causal path
+        if not fresh:
+            kpos = torch.arange(t_max, device=self.device)
+            mask = kpos.view(1, 1, 1, -1) <= pos.view(B, 1, L, 1)
         for li, layer in enumerate(self.layers):
             q, k, v = self._qkv(h, layer, freqs)
             # scatter this chunk's K/V into each request's slot window
             self.kv[li, 0][slots[:, None], :, pos] = k.permute(0, 2, 1, 3)
             self.kv[li, 1][slots[:, None], :, pos] = v.permute(0, 2, 1, 3)
-            kk = self.kv[li, 0].index_select(0, slots)[:, :, :t_max]
-            vv = self.kv[li, 1].index_select(0, slots)[:, :, :t_max]
-            attn = F.scaled_dot_product_attention(q, kk, vv, attn_mask=mask)
+            if fresh:
+                # no history: attend within the chunk itself, flash kernel
+                attn = F.scaled_dot_product_attention(q, k, v, is_causal=True)
+            else:
+                kk = self.kv[li, 0].index_select(0, slots)[:, :, :t_max]
+                vv = self.kv[li, 1].index_select(0, slots)[:, :, :t_max]
+                attn = F.scaled_dot_product_attention(q, kk, vv, attn_mask=mask)
             h = h + attn.transpose(1, 2).reshape(B, L, c.d_model) @ layer.wo
             h = self._mlp(h, layer)
         for slot, start, toks in items:
