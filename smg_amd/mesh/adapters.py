@@ -83,35 +83,69 @@ class WorkerSyncAdapter:
 
 
 class TreeSyncAdapter:
-    """Cache-tree delta sync: local inserts publish (hash(prefix) -> matched
-    tenant/url) deltas so replicas pre-warm affinity for prefixes they have
-    not seen (reference tree_sync.rs:1-20; repair pages :38-67 are covered by
-    the join snapshot)."""
+    """Cache-tree delta sync (reference tree_sync.rs:1-20): local cache-aware
+    inserts publish the routed prefix's TOKENS + tenant, and replicas insert
+    them into their own tree — a request landing on any gateway replica routes
+    to the worker that already holds the prefix.  The join snapshot doubles
+    as the repair protocol (:38-67)."""
+
+    MAX_SYNC_TOKENS = 1024
 
     def __init__(self, mesh: MeshNode, policy_registry):
         self.mesh = mesh
         self.policies = policy_registry
+        self._importing = False
         mesh.kv.register_namespace(NS_TREE)
         mesh.kv.watch(NS_TREE, self._on_remote_op)
+        # hook every current + future cache_aware policy
+        self._wire_existing()
 
-    def publish_insert(self, model_id: str, prefix_hash: int, tenant_url: str, n_tokens: int) -> None:
+    def _wire_existing(self) -> None:
+        for policy in self.policies.all_policies():
+            self._wire(policy)
+        orig_get = self.policies.get
+        adapter = self
+
+        def hooked_get(model_id=None, role="regular"):
+            policy = orig_get(model_id, role)
+            adapter._wire(policy)
+            return policy
+
+        self.policies.get = hooked_get
+
+    def _wire(self, policy) -> None:
+        if hasattr(policy, "mesh_hook") and policy.mesh_hook is None:
+            policy.mesh_hook = self.publish_insert
+
+    def publish_insert(self, model_id: str, tokens, tenant_url: str) -> None:
+        if self._importing:
+            return
+        tokens = list(tokens)[: self.MAX_SYNC_TOKENS]
+        import hashlib
+
+        key = hashlib.blake2b(bytes(str((model_id, tenant_url, tokens[:64]))[:512], "utf8"),
+                              digest_size=12).hexdigest()
         self.mesh.kv.put(
-            NS_TREE,
-            f"{model_id}:{prefix_hash:x}",
-            {"model": model_id, "tenant": tenant_url, "tokens": n_tokens},
+            NS_TREE, f"{model_id}:{key}",
+            {"model": model_id, "tenant": tenant_url, "tokens": tokens},
         )
 
     def _on_remote_op(self, op: Op) -> None:
-        # replicas record the remote affinity in their policy's text/token tree
         if op.value is None:
             return
         v = op.value
-        policy = self.policies.get(v.get("model"))
-        if hasattr(policy, "token_trees"):
-            # the delta carries only the hash, not tokens; pre-warming the
-            # routing table happens through the worker-load path.  Full page
-            # repair arrives with the join snapshot.
-            pass
+        tokens = v.get("tokens")
+        if not tokens:
+            return
+        self._importing = True
+        try:
+            policy = self.policies.get(v.get("model"))
+            if hasattr(policy, "token_trees"):
+                policy._token_tree(v.get("model") or "default").insert(tokens, v.get("tenant"))
+        except Exception as exc:
+            log.debug("tree sync apply failed: %s", exc)
+        finally:
+            self._importing = False
 
 
 class RateLimitSyncAdapter:

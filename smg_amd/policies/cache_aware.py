@@ -48,6 +48,9 @@ class CacheAwarePolicy(LoadBalancingPolicy):
         self.text_trees: Dict[str, object] = {}
         self.indexer = indexer  # KV-event PositionalIndexer (event-driven mode)
         self._last_eviction = time.monotonic()
+        # HA mesh hook: on_local_insert(model_id, tokens, tenant_url) publishes
+        # tree deltas to replicas (reference mesh/adapters/tree_sync.rs:1-20)
+        self.mesh_hook = None
 
     # ---- tree plumbing ---------------------------------------------------
     def _token_tree(self, model_id: str):
@@ -117,7 +120,7 @@ class CacheAwarePolicy(LoadBalancingPolicy):
         self._maybe_evict()
         return selected if selected is not None else (candidates[0] if candidates else None)
 
-    def _choose(self, workers, candidates, min_idx):
+    def _choose(self, workers, candidates, min_idx, model_id=None, tokens=None):
         """Builds the choose_tenant closure shared by token and text paths."""
         picked: List[Optional[int]] = [None]
 
@@ -136,6 +139,8 @@ class CacheAwarePolicy(LoadBalancingPolicy):
         tree = self._token_tree(model_id)
         picked, choose = self._choose(workers, candidates, min_idx)
         tree.match_and_insert(list(info.tokens), choose)
+        if self.mesh_hook is not None and picked[0] is not None:
+            self.mesh_hook(model_id, list(info.tokens), workers[picked[0]].url)
         return picked[0]
 
     def _select_with_text(self, workers, candidates, info, model_id, min_idx):

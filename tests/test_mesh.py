@@ -176,3 +176,45 @@ def test_rate_limit_shards(runner):
             await sb.close()
 
     runner(run())
+
+
+def test_tree_delta_sync_prewarms_replica(runner):
+    """A prefix routed on gateway A routes to the SAME worker on gateway B
+    (reference tree_sync.rs behavior)."""
+    import asyncio
+
+    from smg_amd.policies import SelectWorkerInfo
+
+    async def run():
+        a, sa = await make_node("a")
+        b, sb = await make_node("b", peers=[a])
+        cfg = RouterConfig(policy=PolicyConfig(name="cache_aware", gpu_tree=False, block_size=4))
+        ctx_a, ctx_b = AppContext(cfg), AppContext(cfg)
+        MeshAdapters(a, ctx_a)
+        MeshAdapters(b, ctx_b)
+        try:
+            for ctx in (ctx_a, ctx_b):
+                ctx.worker_registry.register(Worker("http://w0:1", model_id="m"))
+                ctx.worker_registry.register(Worker("http://w1:1", model_id="m"))
+            toks = list(range(64))
+            pol_a = ctx_a.policy_registry.get("m")
+            workers_a = ctx_a.worker_registry.for_model("m")
+            sel_a = pol_a.select_worker(workers_a, SelectWorkerInfo(model_id="m", tokens=toks))
+            chosen_url = workers_a[sel_a].url
+            # wait for the delta to reach B
+            pol_b = ctx_b.policy_registry.get("m")
+            for _ in range(40):
+                tree = pol_b.token_trees.get("m")
+                if tree is not None and len(tree) > 0:
+                    break
+                await asyncio.sleep(0.1)
+            workers_b = ctx_b.worker_registry.for_model("m")
+            sel_b = pol_b.select_worker(workers_b, SelectWorkerInfo(model_id="m", tokens=toks))
+            assert workers_b[sel_b].url == chosen_url
+        finally:
+            await a.stop()
+            await b.stop()
+            await sa.close()
+            await sb.close()
+
+    runner(run())
