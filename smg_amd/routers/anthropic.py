@@ -31,13 +31,55 @@ def messages_to_chat(body: Dict[str, Any]) -> Dict[str, Any]:
             messages.append({"role": "system", "content": text})
     for m in body.get("messages") or []:
         content = m.get("content")
+        role = m.get("role", "user")
         if isinstance(content, list):
-            text = "".join(
-                b.get("text", "") for b in content if isinstance(b, dict) and b.get("type") == "text"
-            )
+            parts = []
+            tool_calls = []
+            tool_results = []
+            for b in content:
+                if not isinstance(b, dict):
+                    continue
+                btype = b.get("type")
+                if btype == "text":
+                    parts.append({"type": "text", "text": b.get("text", "")})
+                elif btype == "image":
+                    src = b.get("source", {})
+                    if src.get("type") == "base64":
+                        parts.append({
+                            "type": "image_url",
+                            "image_url": {"url": f"data:{src.get('media_type','image/png')};base64,{src.get('data','')}"},
+                        })
+                    elif src.get("type") == "url":
+                        parts.append({"type": "image_url", "image_url": {"url": src.get("url", "")}})
+                elif btype == "tool_use":
+                    tool_calls.append({
+                        "id": b.get("id"),
+                        "type": "function",
+                        "function": {"name": b.get("name"), "arguments": json.dumps(b.get("input") or {})},
+                    })
+                elif btype == "tool_result":
+                    rc = b.get("content")
+                    if isinstance(rc, list):
+                        rc = "".join(x.get("text", "") for x in rc if isinstance(x, dict))
+                    tool_results.append({"role": "tool", "tool_call_id": b.get("tool_use_id"),
+                                         "content": rc if isinstance(rc, str) else json.dumps(rc)})
+            if tool_results:
+                messages.extend(tool_results)
+                continue
+            msg = {"role": role}
+            if len(parts) == 1 and parts[0]["type"] == "text":
+                msg["content"] = parts[0]["text"]
+            elif parts:
+                msg["content"] = parts
+            else:
+                msg["content"] = ""
+            if tool_calls:
+                msg["tool_calls"] = tool_calls
+                if not parts:
+                    msg["content"] = None
+            messages.append(msg)
         else:
-            text = content or ""
-        messages.append({"role": m.get("role", "user"), "content": text})
+            messages.append({"role": role, "content": content or ""})
     chat: Dict[str, Any] = {
         "model": body.get("model"),
         "messages": messages,
@@ -71,6 +113,8 @@ def chat_to_message(body: Dict[str, Any], chat_resp: Dict[str, Any]) -> Dict[str
     choice = (chat_resp.get("choices") or [{}])[0]
     msg = choice.get("message", {})
     content: List[Dict[str, Any]] = []
+    if msg.get("reasoning_content"):
+        content.append({"type": "thinking", "thinking": msg["reasoning_content"]})
     if msg.get("content"):
         content.append({"type": "text", "text": msg["content"]})
     for tc in msg.get("tool_calls") or []:
