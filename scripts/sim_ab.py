@@ -41,6 +41,7 @@ def parse_args():
     p.add_argument("--suffix-chars", type=int, default=200)
     p.add_argument("--max-tokens", type=int, default=24)
     p.add_argument("--speedup", type=float, default=10.0)
+    p.add_argument("--prefill-tps", type=float, default=8000.0)
     p.add_argument("--seed", type=int, default=7)
     return p.parse_args()
 
@@ -51,7 +52,7 @@ async def run_policy(policy_name: str, args) -> dict:
     ctx = AppContext(cfg)
     engines = []
     for i in range(args.workers):
-        eng = MockWorkerEngine(SimConfig(speedup=args.speedup, model_id="sim-model"))
+        eng = MockWorkerEngine(SimConfig(speedup=args.speedup, model_id="sim-model", prefill_tokens_per_sec=args.prefill_tps))
         await eng.start()
         w = Worker(f"sim://w{i}", model_id="sim-model")
         w.extra["engine"] = eng

@@ -11,16 +11,18 @@ from __future__ import annotations
 
 from .pytree import MatchResult, PagedRadixTree, StringTree, TokenTree
 
+# torch must load its libamdhip64 BEFORE _core.so links the system one (same
+# soname, first load wins): a second runtime would leave torch.cuda
+# unavailable on a live GPU box.  Imported at module load so the first
+# routing decision never pays the ~2 s torch import.
+try:
+    import torch  # noqa: F401
+except ImportError:
+    pass
+
 
 def _core():
     try:
-        # torch must load its libamdhip64 BEFORE _core.so links the system
-        # one (same soname, first load wins): a second runtime would leave
-        # torch.cuda.is_available() false on a live GPU box.
-        try:
-            import torch  # noqa: F401
-        except ImportError:
-            pass
         from .. import _core as core  # built in-tree by __graft_entry__.build()
 
         return core

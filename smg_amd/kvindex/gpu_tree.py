@@ -29,11 +29,7 @@ class GpuTokenTree:
         max_batch_reqs: int = 4096,
         max_batch_tokens: int = 1 << 22,
     ):
-        try:
-            import torch  # noqa: F401 — HIP runtime must load before _core.so
-        except ImportError:
-            pass
-        from .. import _core
+        from .. import _core  # torch-first guard lives in kvindex.__init__
 
         if _core.hip_device_count() <= 0:
             raise RuntimeError("GpuTokenTree requires an AMD GPU (hip_device_count==0)")

@@ -13,11 +13,7 @@ from .slots import SlotMap
 
 class HostTokenTree:
     def __init__(self, page_size: int = 16):
-        try:
-            import torch  # noqa: F401 — HIP runtime must load before _core.so
-        except ImportError:
-            pass
-        from .. import _core
+        from .. import _core  # torch-first guard lives in kvindex.__init__
 
         self._tree = _core.HostTokenTree(page_size=page_size)
         self.page_size = page_size
