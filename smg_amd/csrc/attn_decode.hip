@@ -1,28 +1,39 @@
 // Fused single-token (decode) attention for the GPU worker engine — gfx950.
 //
-// One wave per (slot, head): online-softmax over the slot's KV window.
-// Replaces torch sdpa in the engine's full-arena decode, where sdpa must
-// read the rectangular [slots, maxlen] window for every slot; this kernel
-// reads only kv[slot][:pos+1], so idle/short slots cost nothing — the decode
-// step becomes KV-bandwidth bound on the ACTIVE tokens.
+// One wave per (slot, head): online-softmax over the slot's live KV window
+// (kv[slot][:pos+1]); idle/short slots cost nothing, unlike sdpa's
+// rectangular [slots, maxlen] window.
+//
+// v5 layout notes (measured on MI355X):
+//   * K dot phase: each lane streams ITS OWN timestep's K row with dwordx4
+//     loads — a row is 256 B = 4 consecutive cache lines, so per-lane
+//     streaming is line-efficient, and 64 lanes x 8 wide loads keep >500
+//     lines in flight with no LDS round-trip or barrier;
+//   * P·V phase: probabilities via LDS broadcast, V rows loaded 4 B/lane
+//     coalesced, no branch between shfl and load so loads pipeline;
+//   * zero LDS tiles -> no occupancy cap from shared memory.
 //
 // Layout contract (the engine's KV arena, one layer):
 //   K, V:  [n_slots, n_heads, max_seq, head_dim]  bf16, contiguous
 //   q:     [n_slots, n_heads, head_dim]           bf16, contiguous
-//   pos:   [n_slots] int32 — attend kpos <= pos[slot] (the current token's
-//          K/V must already be written at pos[slot])
+//   pos:   [n_slots] int32 — attend kpos <= pos[slot]
 //   out:   [n_slots, n_heads, head_dim]           bf16
-//
-// head_dim <= 128 (2 elements per lane).  Round of 64 timesteps per
-// iteration: lane t computes the full q·K[t] dot (K rows stream per lane),
-// then the P·V accumulation broadcasts each lane's probability with shfl
-// while V rows load coalesced (4 B per lane).
+// head_dim <= 128 and divisible by 8.
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
 
 #include <cstdint>
 
 #define WAVE 64
+
+__device__ __forceinline__ float bf16_to_f32(unsigned short u) {
+    union {
+        unsigned int i;
+        float f;
+    } c;
+    c.i = ((unsigned int)u) << 16;
+    return c.f;
+}
 
 extern "C" __global__ void __launch_bounds__(WAVE) smg_attn_decode(
     const __hip_bfloat16* __restrict__ q,
@@ -44,48 +55,35 @@ extern "C" __global__ void __launch_bounds__(WAVE) smg_attn_decode(
     const __hip_bfloat16* vh = v + head_base;
     const __hip_bfloat16* qh = q + ((size_t)slot * n_heads + head) * head_dim;
 
-    // stage q in LDS (whole wave reads it every dot)
+    // q staged once in LDS as f32 (read by every lane's dot)
     __shared__ float s_q[128];
-    // K/V tiles: 64 rows, stride 136 bf16 = 17 16-byte slots — rows stay
-    // 16 B aligned for dwordx4 loads and the odd slot stride staggers banks
-    __shared__ __hip_bfloat16 s_k[WAVE][136];
+    __shared__ float s_p[WAVE];
     for (int i = lane; i < head_dim; i += WAVE) s_q[i] = (float)qh[i];
     __syncthreads();
 
-    // accumulator: 2 output elements per lane (head_dim <= 128)
+    const int vec_n = head_dim / 8;  // dwordx4 chunks per row (<=16)
     float acc0 = 0.f, acc1 = 0.f;
     float m = -1e30f, l = 0.f;
-    int e0 = lane * 2, e1 = lane * 2 + 1;
+    const int e0 = lane * 2, e1 = lane * 2 + 1;
 
-    __shared__ __hip_bfloat16 s_v[WAVE][136];
-    __shared__ float s_p[WAVE];
-    // vectorized tile fill: 16 B per lane, 4 rows per instruction
-    const int vec_per_row = head_dim / 8;  // 16-byte chunks per row (<=16)
-    const int rows_per_iter = WAVE / vec_per_row;
-    const int sub = lane % vec_per_row;
-    const int rofs = lane / vec_per_row;
     for (int base = 0; base < T; base += WAVE) {
         int lim = min(WAVE, T - base);
-        for (int r0 = 0; r0 < lim; r0 += rows_per_iter) {
-            int r = r0 + rofs;
-            if (r < lim) {
-                const uint4* krow = (const uint4*)(kh + (size_t)(base + r) * head_dim);
-                const uint4* vrow = (const uint4*)(vh + (size_t)(base + r) * head_dim);
-                ((uint4*)&s_k[r][0])[sub] = krow[sub];
-                ((uint4*)&s_v[r][0])[sub] = vrow[sub];
-            }
-        }
-        __syncthreads();
         int t = base + lane;
         float score = -1e30f;
         if (t < T) {
+            const uint4* row = (const uint4*)(kh + (size_t)t * head_dim);
             float d = 0.f;
-#pragma unroll 8
-            for (int i = 0; i < head_dim; i += 2)
-                d += s_q[i] * (float)s_k[lane][i] + s_q[i + 1] * (float)s_k[lane][i + 1];
+#pragma unroll 4
+            for (int c = 0; c < vec_n; ++c) {
+                uint4 w = row[c];  // 8 bf16
+                const unsigned short* hsp = (const unsigned short*)&w;
+                int ib = c * 8;
+#pragma unroll
+                for (int j = 0; j < 8; ++j) d += s_q[ib + j] * bf16_to_f32(hsp[j]);
+            }
             score = d * scale;
         }
-        // online softmax across the wave's 64 scores
+        // online softmax across the wave's scores
         float mr = score;
 #pragma unroll
         for (int off = 32; off > 0; off >>= 1) mr = fmaxf(mr, __shfl_xor(mr, off, WAVE));
@@ -100,13 +98,14 @@ extern "C" __global__ void __launch_bounds__(WAVE) smg_attn_decode(
         acc0 *= alpha;
         acc1 *= alpha;
         __syncthreads();
-        // P·V from LDS: lane owns output elements (e0, e1)
+        // P·V: V rows stream 4 B/lane coalesced; no branch gates the loads
         if (e0 < head_dim) {
 #pragma unroll 8
             for (int j = 0; j < lim; ++j) {
+                const __hip_bfloat16* vrow = vh + (size_t)(base + j) * head_dim;
                 float pj = s_p[j];
-                acc0 += pj * (float)s_v[j][e0];
-                acc1 += pj * (float)s_v[j][e1];
+                acc0 += pj * (float)vrow[e0];
+                acc1 += pj * (float)vrow[e1];
             }
         }
         __syncthreads();
