@@ -110,6 +110,8 @@ METHODS = {
     "Generate": "server_stream",
     "Embed": "unary",
     "EncodeImage": "unary",
+    "Rerank": "unary",
+    "Classify": "unary",
     "HealthCheck": "unary",
     "Abort": "unary",
     "GetModelInfo": "unary",

@@ -89,6 +89,12 @@ class EngineClient:
     async def encode_image(self, request_id: str, multimodal: dict) -> dict:
         return await self._unary("EncodeImage", {"request_id": request_id, "multimodal": multimodal}, timeout=60.0)
 
+    async def rerank(self, query: str, documents) -> dict:
+        return await self._unary("Rerank", {"query": query, "documents": list(documents)}, timeout=30.0)
+
+    async def classify(self, text: str) -> dict:
+        return await self._unary("Classify", {"input": text}, timeout=30.0)
+
     async def subscribe_kv_events(self) -> AsyncIterator[dict]:
         call = self.channel().unary_stream(
             api.method("SubscribeKvEvents"), request_serializer=_BYTES, response_deserializer=_BYTES

@@ -149,6 +149,23 @@ class SchedulerServicer(grpc.GenericRpcHandler):
     def _h_healthcheck(self, request: bytes, context):
         return api.dumps({"healthy": True})
 
+    def _h_rerank(self, request: bytes, context):
+        d = api.loads(request)
+        query = d.get("query", "")
+        docs = d.get("documents") or []
+        results = [
+            {"index": i, "relevance_score": 1.0 / (1 + abs(hash(query) - hash(doc)) % 100)}
+            for i, doc in enumerate(docs)
+        ]
+        results.sort(key=lambda r: -r["relevance_score"])
+        return api.dumps({"results": results})
+
+    def _h_classify(self, request: bytes, context):
+        d = api.loads(request)
+        text = d.get("input") or d.get("text") or ""
+        label = "positive" if hash(text) % 2 == 0 else "negative"
+        return api.dumps({"data": [{"index": 0, "label": label, "score": 0.9}]})
+
     def _h_encodeimage(self, request: bytes, context):
         """EPD encode leg: vision-tower forward for the pixel tensor (mock:
         deterministic pooled embedding; the GPU engine runs its tower).

@@ -161,6 +161,11 @@ class PreparationStage(Stage):
         elif ctx.endpoint == "embedding":
             i = body.get("input")
             ctx.text = i if isinstance(i, str) else "\n".join(x for x in (i or []) if isinstance(x, str))
+        elif ctx.endpoint == "rerank":
+            ctx.text = str(body.get("query") or "")
+        elif ctx.endpoint == "classify":
+            i = body.get("input") or body.get("text") or ""
+            ctx.text = i if isinstance(i, str) else ""
 
         # tokenize
         if not ctx.input_ids:

@@ -34,6 +34,9 @@ ENDPOINT_BY_PATH = {
     "/v1/completions": "completion",
     "/generate": "generate",
     "/v1/embeddings": "embedding",
+    "/v1/rerank": "rerank",
+    "/rerank": "rerank",
+    "/v1/classify": "classify",
 }
 
 
@@ -68,6 +71,10 @@ class GrpcRouter(Router):
                 return ctx.error
         if endpoint == "embedding":
             return await self._embed(ctx)
+        if endpoint == "rerank":
+            return await self._rerank(ctx)
+        if endpoint == "classify":
+            return await self._classify(ctx)
         if ctx.stream:
             return RouteResponse(
                 status=200,
@@ -93,6 +100,32 @@ class GrpcRouter(Router):
             "usage": {"prompt_tokens": len(ctx.input_ids), "total_tokens": len(ctx.input_ids)},
         }
         return RouteResponse(status=200, body=json.dumps(body).encode())
+
+    async def _rerank(self, ctx: PipelineContext) -> RouteResponse:
+        body = ctx.body or {}
+        try:
+            d = await ctx.client.rerank(body.get("query", ""), body.get("documents") or [])
+            ctx.worker.record_outcome(True)
+        except Exception as exc:
+            ctx.worker.record_outcome(False)
+            return RouteResponse(status=502, body=error_body(f"engine error: {exc}", 502))
+        finally:
+            ctx.worker.decr_load(len(ctx.input_ids))
+        return RouteResponse(status=200, body=json.dumps(
+            {"results": d.get("results", []), "model": ctx.model_id}).encode())
+
+    async def _classify(self, ctx: PipelineContext) -> RouteResponse:
+        body = ctx.body or {}
+        try:
+            d = await ctx.client.classify(body.get("input") or body.get("text") or "")
+            ctx.worker.record_outcome(True)
+        except Exception as exc:
+            ctx.worker.record_outcome(False)
+            return RouteResponse(status=502, body=error_body(f"engine error: {exc}", 502))
+        finally:
+            ctx.worker.decr_load(len(ctx.input_ids))
+        return RouteResponse(status=200, body=json.dumps(
+            {"object": "classification", "model": ctx.model_id, "data": d.get("data", [])}).encode())
 
     def _fire_prefill(self, ctx: PipelineContext):
         """PD dual dispatch: run the prefill leg concurrently and drain its

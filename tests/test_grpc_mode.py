@@ -179,3 +179,23 @@ def test_grpc_worker_failure_is_502(runner):
             await router.shutdown()
 
     runner(run())
+
+
+def test_grpc_rerank_and_classify(runner):
+    async def run():
+        ctx, router, servers = await setup(n_workers=1)
+        try:
+            resp = await router.route(_req("/v1/rerank", {"model": "mock-model", "query": "q",
+                                                          "documents": ["a", "b", "c"]}))
+            assert resp.status == 200, resp.body
+            data = json.loads(resp.body)
+            assert len(data["results"]) == 3
+            assert all("relevance_score" in r for r in data["results"])
+            resp = await router.route(_req("/v1/classify", {"model": "mock-model", "input": "great stuff"}))
+            assert resp.status == 200
+            data = json.loads(resp.body)
+            assert data["data"][0]["label"] in ("positive", "negative")
+        finally:
+            await teardown(router, servers)
+
+    runner(run())
