@@ -8,7 +8,6 @@ from __future__ import annotations
 
 import argparse
 import asyncio
-import logging
 import sys
 from typing import List, Optional, Tuple
 

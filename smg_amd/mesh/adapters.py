@@ -5,10 +5,8 @@ wiring.rs)."""
 from __future__ import annotations
 
 import logging
-from typing import Optional
-
 from ..workers.worker import Worker, WorkerType
-from .crdt import MeshKV, Op, epoch_max_wins_merge
+from .crdt import Op, epoch_max_wins_merge
 from .swim import MeshNode
 
 log = logging.getLogger("smg.mesh.adapters")

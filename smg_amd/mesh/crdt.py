@@ -6,7 +6,7 @@ op-log + per-peer watermarks (crdt_kv/operation.rs, watermark.rs)).
 from __future__ import annotations
 
 import time
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Any, Callable, Dict, List, Optional, Tuple
 
 

@@ -8,8 +8,6 @@ from __future__ import annotations
 import base64
 import binascii
 import io
-from typing import Optional, Tuple
-
 import numpy as np
 
 

@@ -10,7 +10,7 @@ engine processes)."""
 from __future__ import annotations
 
 from multiprocessing import shared_memory
-from typing import Any, Dict, Tuple
+from typing import Any, Dict
 
 import numpy as np
 

@@ -4,8 +4,6 @@ from __future__ import annotations
 
 import asyncio
 import time
-from typing import Optional
-
 from aiohttp import web
 
 from ..config import RateLimitConfig

@@ -22,7 +22,6 @@ from ..config import RouterConfig
 from ..protocols.openai import error_body
 from ..routers.base import RouteRequest
 from ..routers.factory import RouterManager
-from ..workers.worker import Worker, WorkerType
 from .app_context import AppContext
 
 log = logging.getLogger("smg.server")

@@ -6,7 +6,6 @@ Constructed once at startup; handlers reach it through the aiohttp app.
 """
 from __future__ import annotations
 
-import asyncio
 from typing import Optional
 
 from ..config import RouterConfig

@@ -2,7 +2,6 @@
 server.rs:821-822 and the tokenizer management group :909-918)."""
 from __future__ import annotations
 
-import json
 
 from aiohttp import web
 
