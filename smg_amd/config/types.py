@@ -184,7 +184,11 @@ class AuthConfig:
     tenant_api_keys: Dict[str, str] = field(default_factory=dict)  # key -> tenant id
     jwt_issuer: Optional[str] = None
     jwt_audience: Optional[str] = None
-    jwt_jwks_uri: Optional[str] = None
+    jwt_jwks_uri: Optional[str] = None  # http(s) URI or local file path
+    jwt_jwks_inline: Optional[Dict] = None  # inline {"keys": [...]} set
+    jwt_leeway_secs: float = 60.0
+    jwt_jwks_cache_ttl_secs: float = 3600.0
+    jwt_enable_jti_check: bool = False
     jwt_role_claim: str = "roles"
     jwt_role_mapping: Dict[str, str] = field(default_factory=dict)
     control_plane_api_keys: List[str] = field(default_factory=list)

@@ -98,11 +98,39 @@ async def plugin_middleware(request: web.Request, handler):
     return resp
 
 
+def _jwt_enabled(auth) -> bool:
+    return bool(auth.jwt_jwks_uri or auth.jwt_jwks_inline)
+
+
+def _get_jwt_validator(ctx: "AppContext"):
+    """Lazily build the JwtValidator from AuthConfig (reference
+    crates/auth/src/jwt.rs:235-258 from_config)."""
+    v = getattr(ctx, "jwt_validator", None)
+    if v is None:
+        from .jwt_auth import JwksProvider, JwtValidator
+
+        auth = ctx.config.auth
+        jwks = JwksProvider(auth.jwt_jwks_inline or auth.jwt_jwks_uri, ttl_secs=auth.jwt_jwks_cache_ttl_secs)
+        v = JwtValidator(
+            jwks,
+            issuer=auth.jwt_issuer,
+            audience=auth.jwt_audience,
+            leeway_secs=auth.jwt_leeway_secs,
+            role_claim=auth.jwt_role_claim,
+            role_mapping=auth.jwt_role_mapping,
+            enable_jti_check=auth.jwt_enable_jti_check,
+        )
+        ctx.jwt_validator = v
+    return v
+
+
 @web.middleware
 async def auth_middleware(request: web.Request, handler):
     ctx: AppContext = request.app[CTX_KEY]
     auth = ctx.config.auth
-    if request.path in PUBLIC_PATHS or (not auth.api_key and not auth.tenant_api_keys):
+    if request.path in PUBLIC_PATHS or (
+        not auth.api_key and not auth.tenant_api_keys and not _jwt_enabled(auth)
+    ):
         return await handler(request)
     header = request.headers.get("authorization", "")
     token = header[7:] if header.lower().startswith("bearer ") else request.headers.get("x-api-key", "")
@@ -111,6 +139,21 @@ async def auth_middleware(request: web.Request, handler):
     tenant = auth.tenant_api_keys.get(token)
     if tenant is not None:
         request["tenant_id"] = tenant
+        return await handler(request)
+    if _jwt_enabled(auth) and token.count(".") == 2:
+        from .jwt_auth import JwtError
+
+        try:
+            validated = await _get_jwt_validator(ctx).validate(token, session=getattr(ctx, "client_session", None))
+        except JwtError as e:
+            return web.Response(
+                status=401, body=error_body(f"invalid token: {e}", 401, "authentication_error"),
+                content_type="application/json",
+            )
+        request["auth_user"] = validated.subject
+        request["auth_role"] = validated.role
+        if validated.claims.get("tenant_id"):
+            request["tenant_id"] = validated.claims["tenant_id"]
         return await handler(request)
     return web.Response(
         status=401, body=error_body("invalid API key", 401, "authentication_error"),

@@ -2,7 +2,6 @@
 memory.rs, noop.rs)."""
 from __future__ import annotations
 
-import itertools
 import time
 import uuid
 from typing import Any, Dict, List, Optional
@@ -153,6 +152,10 @@ def make_storage(backend: str):
         return MemoryResponseStorage(), MemoryConversationStorage()
     if backend in ("none", "noop"):
         return NoopResponseStorage(), MemoryConversationStorage()
+    if backend == "sqlite" or backend.startswith("sqlite:"):
+        from .sqlite import open_sqlite
+
+        return open_sqlite(backend)
     if backend == "redis":
         raise StorageError("redis backend requires a redis driver (not in this image)")
     if backend == "postgres":
