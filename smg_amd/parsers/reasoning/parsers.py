@@ -123,12 +123,64 @@ class InklingReasoningParser(ReasoningParser):
             think_start="<|content_thinking|>", think_end="<|content_text|>", name="inkling"
         )
 
+    def fresh(self) -> "InklingReasoningParser":
+        return InklingReasoningParser()
+
     def parse(self, text: str) -> Tuple[str, str]:
         reasoning, normal = super().parse(text)
         for tag in ("<|end_message|>", "<|message_model|>", "<|content_model_end_sampling|>"):
             normal = normal.replace(tag, "")
             reasoning = reasoning.replace(tag, "")
         return reasoning.strip(), normal.strip()
+
+
+class HarmonyReasoningParser(ReasoningParser):
+    """gpt-oss Harmony channel format (reference grpc/harmony/parser.rs +
+    streaming.rs): analysis/commentary channels -> reasoning, final ->
+    normal.  Tool-call segments (recipient functions.*) are re-emitted into
+    the normal stream in raw Harmony framing so the downstream `harmony`
+    tool parser can extract them (pipeline order: reasoning -> tool)."""
+
+    def __init__(self):
+        super().__init__(name="harmony")
+
+    def reset(self) -> None:
+        from ...protocols import harmony as _h
+
+        self._hs = _h.HarmonyStreamParser()
+        self._buffer = ""
+        self._done = False
+        self._in_reasoning = False
+
+    def fresh(self) -> "HarmonyReasoningParser":
+        return HarmonyReasoningParser()
+
+    def _events_to_deltas(self, events) -> Tuple[str, str]:
+        from ...protocols import harmony as _h
+
+        reasoning, normal = [], []
+        for ev in events:
+            t = ev["type"]
+            if t == "reasoning":
+                reasoning.append(ev["text"])
+            elif t == "content":
+                normal.append(ev["text"])
+            elif t == "tool_call_start":
+                normal.append(f"{_h.START}assistant{_h.CHANNEL}commentary to=functions.{ev['name']}{_h.MESSAGE}")
+            elif t == "tool_call_args":
+                normal.append(ev["arguments"])
+            elif t == "tool_call_end":
+                normal.append(_h.CALL)
+        return "".join(reasoning), "".join(normal)
+
+    def parse(self, text: str) -> Tuple[str, str]:
+        r1, n1 = self._events_to_deltas(self._hs.feed(text))
+        r2, n2 = self._events_to_deltas(self._hs.finalize())
+        self.reset()
+        return (r1 + r2).strip(), (n1 + n2).strip()
+
+    def parse_streaming(self, delta: str) -> Tuple[str, str]:
+        return self._events_to_deltas(self._hs.feed(delta))
 
 
 def _mk(name, start="<think>", end="</think>", always=False, stream=True):
@@ -152,9 +204,11 @@ PARSERS: Dict[str, ReasoningParser] = {
     "nano_v3": _mk("nano_v3"),
     "cohere_cmd": _mk("cohere_cmd", "<|START_THINKING|>", "<|END_THINKING|>"),
     "inkling": InklingReasoningParser(),
+    "harmony": HarmonyReasoningParser(),
 }
 
 MODEL_PATTERNS = [
+    (r"gpt[-_]?oss", "harmony"),
     (r"deepseek-r1", "deepseek_r1"),
     (r"deepseek-v3[.-]1", "deepseek_v31"),
     (r"qwen3?-thinking|qwen-thinking", "qwen3_thinking"),
@@ -172,7 +226,7 @@ def get_reasoning_parser(name_or_model: Optional[str]) -> Optional[ReasoningPars
     if not name_or_model:
         return None
     if name_or_model in PARSERS:
-        return PARSERS[name_or_model].fresh() if name_or_model != "inkling" else InklingReasoningParser()
+        return PARSERS[name_or_model].fresh()
     low = name_or_model.lower()
     for pat, pname in MODEL_PATTERNS:
         if re.search(pat, low):

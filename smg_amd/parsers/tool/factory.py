@@ -609,6 +609,28 @@ class KimiK3Parser(ToolParser):
         return normal.strip(), calls
 
 
+class HarmonyToolParser(ToolParser):
+    """gpt-oss Harmony commentary tool calls (reference grpc/harmony/parser.rs:
+    messages with recipient `functions.NAME` on the commentary — or, model
+    quirk, analysis — channel are tool calls; `final` text is content)."""
+
+    name = "harmony"
+
+    def has_tool_markers(self, text):
+        return "to=functions." in text
+
+    def parse(self, text, tools=None):
+        from ...protocols.harmony import parse_complete as _hp
+
+        res = _hp(text)
+        calls = [
+            _mk_call(tc["function"]["name"], tc["function"]["arguments"], i)
+            for i, tc in enumerate(res.tool_calls)
+        ]
+        normal = res.content or ""
+        return normal.strip(), calls
+
+
 class InklingParser(ToolParser):
     """inkling channel format (inkling.rs:14-20): `<|content_invoke_tool_json|>
     {"tool": .., ...}<|end_message|>`."""
@@ -688,6 +710,7 @@ def _build_registry() -> Dict[str, ToolParser]:
         "minimax_m2": MinimaxM2Parser(),
         "cohere": CohereParser(),
         "inkling": InklingParser(),
+        "harmony": HarmonyToolParser(),
     }
 
 
@@ -710,6 +733,7 @@ MODEL_MAPPING = [
     (r"step-?3", "step3"),
     (r"command|cohere", "cohere"),
     (r"sarashina", "sarashina"),
+    (r"gpt[-_]?oss", "harmony"),
 ]
 
 

@@ -147,6 +147,7 @@ class StreamingToolParser:
             "cohere": ["<|START_ACTION|>"],
             "sarashina": ["<|tool_calls|>"],
             "inkling": ["<|content_invoke_tool_json|>", "<|content_invoke_tool_text|>"],
+            "harmony": ["<|start|>"],
             "json": ["{", "["],
             "pythonic": ["["],
         }.get(parser.name, [])

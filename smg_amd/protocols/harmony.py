@@ -136,9 +136,24 @@ def split_messages(text: str) -> List[HarmonyMessage]:
         head, sep, body = chunk.partition(MESSAGE)
         if not sep:
             head, body = "", chunk
-        for sp in _SPECIALS:
-            body = body.replace(sp, "")
         channel, recipient = _parse_head(head)
+        # a terminator ends the message; trailing text (no <|start|> framing)
+        # is a new channel-less plain-content message
+        cut = None
+        for sp in _SPECIALS:
+            i = body.find(sp)
+            if i >= 0 and (cut is None or i < cut[0]):
+                cut = (i, sp)
+        if cut is not None:
+            tail = body[cut[0] + len(cut[1]):]
+            body = body[: cut[0]]
+            msgs.append(HarmonyMessage(channel, recipient, body))
+            if tail:
+                for sp in _SPECIALS:
+                    tail = tail.replace(sp, "")
+                if tail:
+                    msgs.append(HarmonyMessage(None, None, tail))
+            continue
         msgs.append(HarmonyMessage(channel, recipient, body))
     return msgs
 
@@ -221,6 +236,8 @@ class HarmonyStreamParser:
                     cut = (i, sp)
             if cut is not None:
                 events.extend(self._emit(self._buf[: cut[0]]))
+                if self._recipient and self._recipient.startswith("functions."):
+                    events.append({"type": "tool_call_end", "index": self._tool_idx})
                 self._buf = self._buf[cut[0] + len(cut[1]):]
                 self._in_message = False
                 self._channel = self._recipient = None

@@ -114,19 +114,16 @@ class JobQueue:
 
     # ---- handlers ----------------------------------------------------------
     async def _add_worker(self, p: Dict[str, Any]):
-        from ..workers.worker import Worker, WorkerType
+        """Routes through the worker-registration workflow (reference
+        src/workflow/steps/mod.rs:91 — classify → detect → discover →
+        create, with per-step retries)."""
+        from .worker_workflow import make_engine, register_worker_via_workflow
 
-        w = Worker(
-            p["url"],
-            model_id=p.get("model_id", "default"),
-            worker_type=WorkerType(p.get("worker_type", "regular")),
-            labels=p.get("labels") or {},
-            api_key=p.get("api_key"),
-            bootstrap_port=p.get("bootstrap_port"),
-            model_aliases=p.get("model_aliases") or [],
-            circuit_breaker_config=self.ctx.config.circuit_breaker,
-        )
-        return self.ctx.worker_registry.register(w)
+        engine = getattr(self.ctx, "workflow_engine", None)
+        if engine is None:
+            engine = make_engine(self.ctx)
+            self.ctx.workflow_engine = engine
+        return await register_worker_via_workflow(engine, p)
 
     async def _remove_worker(self, p: Dict[str, Any]):
         w = self.ctx.worker_registry.remove_by_url(p["url"])
