@@ -63,6 +63,7 @@ class GenerateRequest:
     bootstrap_port: Optional[int] = None
     bootstrap_room: Optional[int] = None
     dp_rank: Optional[int] = None
+    lora_id: Optional[str] = None  # adapter id minted at LoadLoraAdapter time
 
     def to_dict(self):
         d = asdict(self)
@@ -121,4 +122,8 @@ METHODS = {
     "SubscribeKvEvents": "server_stream",
     "StartProfile": "unary",
     "StopProfile": "unary",
+    # LoRA adapter management (reference sglang_scheduler.proto:385-420)
+    "LoadLoraAdapter": "unary",
+    "UnloadLoraAdapter": "unary",
+    "ListLoraAdapters": "unary",
 }

@@ -92,6 +92,18 @@ class EngineClient:
     async def rerank(self, query: str, documents) -> dict:
         return await self._unary("Rerank", {"query": query, "documents": list(documents)}, timeout=30.0)
 
+    async def load_lora_adapter(self, lora_name: str, lora_path: str, lora_id: str, pinned: bool = False) -> dict:
+        return await self._unary(
+            "LoadLoraAdapter",
+            {"lora_name": lora_name, "lora_path": lora_path, "lora_id": lora_id, "pinned": pinned},
+        )
+
+    async def unload_lora_adapter(self, lora_name: str, lora_id: str) -> dict:
+        return await self._unary("UnloadLoraAdapter", {"lora_name": lora_name, "lora_id": lora_id})
+
+    async def list_lora_adapters(self) -> dict:
+        return await self._unary("ListLoraAdapters", {})
+
     async def classify(self, text: str) -> dict:
         return await self._unary("Classify", {"input": text}, timeout=30.0)
 

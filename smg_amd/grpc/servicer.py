@@ -28,6 +28,30 @@ class EngineAdapter:
         self._torch_task: Optional[asyncio.Task] = None
         self._streams = {}
         self._kv_event_subs = []
+        # lora_id -> {"lora_name", "lora_path", "pinned"} (reference
+        # sglang_scheduler.proto:385-420 — the engine owns GPU residency;
+        # this registry is the management surface the gateway drives)
+        self.lora_adapters: dict = {}
+
+    # ---- LoRA management ---------------------------------------------------
+    def load_lora(self, lora_name: str, lora_path: str, lora_id: str, pinned: bool = False):
+        import os
+
+        if not lora_id:
+            raise ValueError("lora_id is required (caller-minted)")
+        if not self.is_mock and lora_path and not os.path.isdir(lora_path):
+            raise FileNotFoundError(f"adapter path not readable: {lora_path}")
+        self.lora_adapters[lora_id] = {"lora_name": lora_name, "lora_path": lora_path, "pinned": pinned}
+        return sorted(self.lora_adapters)
+
+    def unload_lora(self, lora_name: str, lora_id: str):
+        entry = self.lora_adapters.get(lora_id)
+        if entry is None or (lora_name and entry["lora_name"] != lora_name):
+            raise KeyError(f"adapter not loaded: name={lora_name!r} id={lora_id!r}")
+        if entry["pinned"]:
+            raise PermissionError(f"adapter {lora_id!r} is pinned")
+        del self.lora_adapters[lora_id]
+        return sorted(self.lora_adapters)
 
     async def start(self):
         if self.is_mock:
@@ -214,6 +238,33 @@ class SchedulerServicer(grpc.GenericRpcHandler):
         return api.dumps({"status": "profiling not active on this engine"})
 
     _h_stopprofile = _h_startprofile
+
+    def _h_loadloraadapter(self, request: bytes, context):
+        d = api.loads(request)
+        try:
+            ids = self.adapter.load_lora(
+                d.get("lora_name", ""), d.get("lora_path", ""), d.get("lora_id", ""), bool(d.get("pinned"))
+            )
+            return api.dumps({"success": True, "message": "", "loaded_lora_ids": ids})
+        except (ValueError, FileNotFoundError) as e:
+            return api.dumps({"success": False, "message": str(e), "loaded_lora_ids": sorted(self.adapter.lora_adapters)})
+
+    def _h_unloadloraadapter(self, request: bytes, context):
+        d = api.loads(request)
+        try:
+            ids = self.adapter.unload_lora(d.get("lora_name", ""), d.get("lora_id", ""))
+            return api.dumps({"success": True, "message": "", "loaded_lora_ids": ids})
+        except (KeyError, PermissionError) as e:
+            return api.dumps({"success": False, "message": str(e), "loaded_lora_ids": sorted(self.adapter.lora_adapters)})
+
+    def _h_listloraadapters(self, request: bytes, context):
+        return api.dumps(
+            {
+                "adapters": [
+                    {"lora_id": lid, **entry} for lid, entry in sorted(self.adapter.lora_adapters.items())
+                ]
+            }
+        )
 
     def _loop(self):
         return self._event_loop

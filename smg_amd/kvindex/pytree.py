@@ -183,6 +183,57 @@ class PagedRadixTree:
     def clear(self) -> None:
         self.__init__(self.page_size)
 
+    # ---- snapshot / replay (reference kv_index/src/snapshot.rs:18-59 —
+    # TreeSnapshot of SnapshotNodes serialized for mesh join / persistence) --
+    def snapshot(self) -> bytes:
+        """Serialize nodes in BFS order so parents precede children; each
+        entry is (local_idx, parent_idx, page, {tenant: stamp})."""
+        import msgpack
+
+        by_parent: Dict[int, List[Tuple[Tuple, int]]] = {}
+        for (parent, page), cid in self._children.items():
+            by_parent.setdefault(parent, []).append((page, cid))
+        order: Dict[int, int] = {0: 0}
+        entries = []
+        queue = [0]
+        while queue:
+            nid = queue.pop(0)
+            for page, cid in by_parent.get(nid, ()):
+                order[cid] = len(order)
+                child = self._nodes[cid]
+                entries.append(
+                    (order[nid], list(page), {t: ts for t, ts in child.tenants.items()})
+                )
+                queue.append(cid)
+        return msgpack.packb(
+            {"page_size": self.page_size, "clock": self._clock, "nodes": entries},
+            use_bin_type=True,
+        )
+
+    @classmethod
+    def from_snapshot(cls, blob: bytes) -> "PagedRadixTree":
+        import msgpack
+
+        d = msgpack.unpackb(blob, raw=False, strict_map_key=False)
+        tree = cls(d["page_size"])
+        idx_to_nid = {0: 0}
+        for i, (parent_idx, page, tenants) in enumerate(d["nodes"], start=1):
+            parent_nid = idx_to_nid[parent_idx]
+            nid = tree._next_id
+            tree._next_id += 1
+            node = _Node(nid, parent_nid, tree._nodes[parent_nid].depth + 1)
+            node.tenants = dict(tenants)
+            key = (parent_nid, tuple(page))
+            tree._nodes[nid] = node
+            tree._children[key] = nid
+            tree._keys[nid] = key
+            tree._nodes[parent_nid].child_count += 1
+            idx_to_nid[i] = nid
+            for t in tenants:
+                tree.tenant_token_count[t] = tree.tenant_token_count.get(t, 0) + tree.page_size
+        tree._clock = d.get("clock", 0)
+        return tree
+
     def _prune(self, nid: int) -> None:
         node = self._nodes.pop(nid)
         key = self._keys.pop(nid)
