@@ -103,6 +103,34 @@ class PreparationStage(Stage):
         ctx.multimodal = {"images": images}
         return True
 
+    async def _process_audio(self, ctx: PipelineContext, parts) -> bool:
+        """input_audio content parts -> whisper-style log-mel feature tensors
+        (reference audio/processors/qwen3_audio.rs preprocess)."""
+        import base64
+
+        from ...multimodal.audio import AudioError, preprocess_audio
+        from ...multimodal.transport import encode_tensor
+
+        mode = self.app.config.multimodal_tensor_transport
+        audios = []
+        for part in parts:
+            try:
+                data = base64.b64decode(part.get("data") or "")
+                out = preprocess_audio(data)
+            except (AudioError, ValueError) as e:
+                ctx.error = RouteResponse(status=400, body=error_body(f"audio decode failed: {e}"))
+                return False
+            desc = encode_tensor(
+                out["features"], mode=mode, min_shm_bytes=self.app.config.multimodal_shm_min_bytes
+            )
+            desc["feature_length"] = out["feature_length"]
+            desc["sample_rate"] = out["sample_rate"]
+            audios.append(desc)
+        if ctx.multimodal is None:
+            ctx.multimodal = {}
+        ctx.multimodal["audios"] = audios
+        return True
+
     async def run(self, ctx: PipelineContext) -> bool:
         body = ctx.req.body or {}
         ctx.body = body
@@ -144,6 +172,19 @@ class PreparationStage(Stage):
             ]
             if image_urls:
                 ok = await self._process_images(ctx, [u for u in image_urls if u])
+                if not ok:
+                    return False
+            # audio parts -> log-mel features (reference audio/ +
+            # processors/qwen3_audio.rs; OpenAI input_audio content part)
+            audio_parts = [
+                p.get("input_audio")
+                for m in msgs
+                if isinstance(m.get("content"), list)
+                for p in m["content"]
+                if isinstance(p, dict) and p.get("type") == "input_audio" and isinstance(p.get("input_audio"), dict)
+            ]
+            if audio_parts:
+                ok = await self._process_audio(ctx, audio_parts)
                 if not ok:
                     return False
         elif ctx.endpoint == "completion":
