@@ -51,6 +51,11 @@ int smg_img_resize_normalize(void* p, const uint8_t* in, int in_w, int in_h, int
 int smg_attn_decode_launch(const void* q, const void* k, const void* v, const void* pos,
                            void* out, int n_slots, int n_heads, int max_seq, int head_dim,
                            float scale, void* stream);
+// ---- fused decode elementwise (fused_decode.hip) ----
+int smg_rope_kv_store_launch(const void* qkv, const void* freqs, const void* pos,
+                             void* k_cache, void* v_cache, void* q_out,
+                             int n_slots, int n_heads, int max_seq, int head_dim, void* stream);
+int smg_silu_mul_launch(const void* gu, void* out, long long rows, long long inner, void* stream);
 }
 
 #define BPE_MAX_PIECE 64
@@ -325,6 +330,25 @@ PYBIND11_MODULE(_core, m) {
           py::arg("q"), py::arg("k"), py::arg("v"), py::arg("pos"), py::arg("out"),
           py::arg("n_slots"), py::arg("n_heads"), py::arg("max_seq"), py::arg("head_dim"),
           py::arg("scale"), py::arg("stream"));
+    // fused rope + KV-store + q-pack (fused_decode.hip)
+    m.def("rope_kv_store",
+          [](uintptr_t qkv, uintptr_t freqs, uintptr_t pos, uintptr_t k_cache, uintptr_t v_cache,
+             uintptr_t q_out, int n_slots, int n_heads, int max_seq, int head_dim, uintptr_t stream) {
+              int rc = smg_rope_kv_store_launch((const void*)qkv, (const void*)freqs,
+                                                (const void*)pos, (void*)k_cache, (void*)v_cache,
+                                                (void*)q_out, n_slots, n_heads, max_seq, head_dim,
+                                                (void*)stream);
+              if (rc != 0) throw std::runtime_error("rope_kv_store launch failed rc=" + std::to_string(rc));
+          },
+          py::arg("qkv"), py::arg("freqs"), py::arg("pos"), py::arg("k_cache"), py::arg("v_cache"),
+          py::arg("q_out"), py::arg("n_slots"), py::arg("n_heads"), py::arg("max_seq"),
+          py::arg("head_dim"), py::arg("stream"));
+    m.def("silu_mul",
+          [](uintptr_t gu, uintptr_t out, long long rows, long long inner, uintptr_t stream) {
+              int rc = smg_silu_mul_launch((const void*)gu, (void*)out, rows, inner, (void*)stream);
+              if (rc != 0) throw std::runtime_error("silu_mul launch failed rc=" + std::to_string(rc));
+          },
+          py::arg("gu"), py::arg("out"), py::arg("rows"), py::arg("inner"), py::arg("stream"));
 
     py::class_<PyHostTree>(m, "HostTokenTree")
         .def(py::init<uint32_t>(), py::arg("page_size") = 16)

@@ -1,0 +1,113 @@
+// Fused decode-step elementwise kernels — gfx950.
+//
+// The graphed decode loop spends ~20% of its device time in the elementwise
+// "soup" between GEMMs (rope complex-mul chains, KV index_put scatter, the
+// q-contiguous pack, silu+mul).  Two kernels remove ~10 launches per layer:
+//
+//   smg_rope_kv_store: consumes the fused QKV GEMM output [S, 3*D] directly —
+//     applies the interleaved-pair rotary rotation to q and k (reading
+//     cos/sin from the engine's complex64 freqs table, which is (re,im)
+//     float pairs in memory), stores rotated k and raw v into the KV arena
+//     at each slot's write position, and emits q packed [S, H, hd] for the
+//     decode-attention kernel.  Replaces: 2x rope chains (float cast,
+//     complex view/mul, real view, dtype cast), 2x index_put, freqs gather,
+//     and the q .contiguous() copy.
+//
+//   smg_silu_mul: silu(g) * u from the fused gate+up GEMM output [N, 2F]
+//     (one kernel instead of silu + mul; the producing GEMM stays hipBLASLt).
+//
+// Layer layout contract (TorchEngine):
+//   qkv:     [S, 3*D] bf16, D = n_heads * head_dim (q | k | v blocks)
+//   freqs:   complex64 [max_seq, head_dim/2] -> float2 (cos, sin)
+//   pos:     [S] int32 — the write position of each slot this step
+//   k,v:     [S, n_heads, max_seq, head_dim] bf16 (one layer's arena slice)
+//   q_out:   [S, n_heads, head_dim] bf16
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#include <cstdint>
+
+#define WAVE 64
+
+extern "C" __global__ void __launch_bounds__(WAVE) smg_rope_kv_store(
+    const __hip_bfloat16* __restrict__ qkv,
+    const float2* __restrict__ freqs,
+    const int* __restrict__ pos,
+    __hip_bfloat16* __restrict__ k_cache,
+    __hip_bfloat16* __restrict__ v_cache,
+    __hip_bfloat16* __restrict__ q_out,
+    int n_slots, int n_heads, int max_seq, int head_dim) {
+    const int sh = blockIdx.x;
+    const int slot = sh / n_heads;
+    const int head = sh % n_heads;
+    if (slot >= n_slots) return;
+    const int lane = threadIdx.x;
+    const int p = pos[slot];
+    const int pairs = head_dim >> 1;
+    const int D = n_heads * head_dim;
+
+    const __hip_bfloat16* qrow = qkv + (size_t)slot * 3 * D + (size_t)head * head_dim;
+    const __hip_bfloat16* krow = qrow + D;
+    const __hip_bfloat16* vrow = qrow + 2 * D;
+    const float2* f = freqs + (size_t)p * pairs;
+
+    __hip_bfloat16* kdst = k_cache + (((size_t)slot * n_heads + head) * max_seq + p) * head_dim;
+    __hip_bfloat16* vdst = v_cache + (((size_t)slot * n_heads + head) * max_seq + p) * head_dim;
+    __hip_bfloat16* qdst = q_out + ((size_t)slot * n_heads + head) * head_dim;
+
+    for (int i = lane; i < pairs; i += WAVE) {
+        const float2 cs = f[i];  // (cos, sin)
+        const int e = 2 * i;
+        const float q0 = __bfloat162float(qrow[e]);
+        const float q1 = __bfloat162float(qrow[e + 1]);
+        qdst[e] = __float2bfloat16(q0 * cs.x - q1 * cs.y);
+        qdst[e + 1] = __float2bfloat16(q0 * cs.y + q1 * cs.x);
+        const float k0 = __bfloat162float(krow[e]);
+        const float k1 = __bfloat162float(krow[e + 1]);
+        kdst[e] = __float2bfloat16(k0 * cs.x - k1 * cs.y);
+        kdst[e + 1] = __float2bfloat16(k0 * cs.y + k1 * cs.x);
+        // v passes through untouched (pair copy keeps the access pattern)
+        vdst[e] = vrow[e];
+        vdst[e + 1] = vrow[e + 1];
+    }
+}
+
+extern "C" __global__ void __launch_bounds__(256) smg_silu_mul(
+    const __hip_bfloat16* __restrict__ gu,
+    __hip_bfloat16* __restrict__ out,
+    long long rows, long long inner) {
+    const long long total = rows * inner;
+    for (long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+         idx += (long long)gridDim.x * blockDim.x) {
+        const long long row = idx / inner;
+        const long long col = idx - row * inner;
+        const float g = __bfloat162float(gu[row * 2 * inner + col]);
+        const float u = __bfloat162float(gu[row * 2 * inner + inner + col]);
+        const float s = g / (1.0f + __expf(-g));  // silu
+        out[idx] = __float2bfloat16(s * u);
+    }
+}
+
+extern "C" int smg_rope_kv_store_launch(
+    const void* qkv, const void* freqs, const void* pos,
+    void* k_cache, void* v_cache, void* q_out,
+    int n_slots, int n_heads, int max_seq, int head_dim, void* stream) {
+    if (head_dim % 2 != 0 || head_dim > 256) return 1;
+    dim3 grid(n_slots * n_heads);
+    hipLaunchKernelGGL(smg_rope_kv_store, grid, dim3(WAVE), 0, (hipStream_t)stream,
+                       (const __hip_bfloat16*)qkv, (const float2*)freqs, (const int*)pos,
+                       (__hip_bfloat16*)k_cache, (__hip_bfloat16*)v_cache,
+                       (__hip_bfloat16*)q_out, n_slots, n_heads, max_seq, head_dim);
+    return (int)hipGetLastError();
+}
+
+extern "C" int smg_silu_mul_launch(const void* gu, void* out, long long rows,
+                                   long long inner, void* stream) {
+    long long total = rows * inner;
+    int blocks = (int)((total + 2047) / 2048);  // 8 elems/thread target
+    if (blocks < 1) blocks = 1;
+    if (blocks > 65535) blocks = 65535;
+    hipLaunchKernelGGL(smg_silu_mul, dim3(blocks), dim3(256), 0, (hipStream_t)stream,
+                       (const __hip_bfloat16*)gu, (__hip_bfloat16*)out, rows, inner);
+    return (int)hipGetLastError();
+}

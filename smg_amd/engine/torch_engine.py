@@ -150,6 +150,8 @@ class TorchEngine:
         # fused gfx950 decode-attention kernel (csrc/attn_decode.hip): reads
         # only kv[slot][:pos+1] per slot instead of sdpa's rectangular window
         self._hip_attn = None
+        self._hip_fused = None
+        self._hip_silu_mul = None
         if self.device.type == "cuda" and self.dtype == torch.bfloat16 and c.head_dim <= 128:
             try:
                 from .. import _core
@@ -159,6 +161,17 @@ class TorchEngine:
                     self._pos_i32 = torch.zeros(c.max_slots, dtype=torch.int32, device=self.device)
                     self._attn_out = torch.zeros(
                         c.max_slots, c.n_heads, c.head_dim, device=self.device, dtype=self.dtype
+                    )
+                # fused rope+KV-store+q-pack and silu*mul (csrc/fused_decode.hip):
+                # removes ~10 elementwise launches per layer from the decode loop
+                if hasattr(_core, "rope_kv_store"):
+                    self._hip_fused = _core.rope_kv_store
+                    self._hip_silu_mul = _core.silu_mul
+                    self._q_buf = torch.zeros(
+                        c.max_slots, c.d_model, device=self.device, dtype=self.dtype
+                    )
+                    self._smul_buf = torch.zeros(
+                        c.max_slots, c.d_ffn, device=self.device, dtype=self.dtype
                     )
             except ImportError:
                 pass
@@ -396,6 +409,8 @@ class TorchEngine:
         shapes per `maxlen` make this hipGraph-capturable."""
         c = self.cfg
         S = c.max_slots
+        if self._hip_fused is not None and self._hip_attn is not None:
+            return self._decode_core_fused()
         pos = self.seq_len  # [S] current length == write position
         freqs = self.freqs_cis[pos].view(S, 1, 1, -1)
         use_hip = self._hip_attn is not None
@@ -430,14 +445,50 @@ class TorchEngine:
         logits = h[:, 0] @ self.embed.t()
         return logits.argmax(-1)  # [S]
 
+    @torch.no_grad()
+    def _decode_core_fused(self) -> torch.Tensor:
+        """GPU decode with the fused_decode.hip kernels: per layer the only
+        torch-launched ops are rms_norm and the four hipBLASLt GEMMs (two of
+        them addmm so the residual adds ride the GEMM epilogue); rope,
+        KV-arena store, q pack and silu*mul are one HIP kernel each.
+        hipGraph-capturable: every launch lands on the capturing stream."""
+        c = self.cfg
+        S = c.max_slots
+        self._pos_i32.copy_(self.seq_len.to(torch.int32))
+        stream = torch.cuda.current_stream().cuda_stream
+        scale = 1.0 / math.sqrt(c.head_dim)
+        h = self.embed[self._last_tok]  # [S, D]
+        freqs_ptr = self.freqs_cis.data_ptr()  # complex64 [T, hd/2] == float2
+        for li, layer in enumerate(self.layers):
+            qkv = _rms(h, layer.ln1) @ layer.wqkv  # [S, 3D]
+            self._hip_fused(
+                qkv.data_ptr(), freqs_ptr, self._pos_i32.data_ptr(),
+                self.kv[li, 0].data_ptr(), self.kv[li, 1].data_ptr(), self._q_buf.data_ptr(),
+                S, c.n_heads, c.max_seq, c.head_dim, stream,
+            )
+            self._hip_attn(
+                self._q_buf.data_ptr(), self.kv[li, 0].data_ptr(), self.kv[li, 1].data_ptr(),
+                self._pos_i32.data_ptr(), self._attn_out.data_ptr(),
+                S, c.n_heads, c.max_seq, c.head_dim, scale, stream,
+            )
+            h = torch.addmm(h, self._attn_out.view(S, c.d_model), layer.wo)
+            gu = _rms(h, layer.ln2) @ layer.w13  # [S, 2F]
+            self._hip_silu_mul(gu.data_ptr(), self._smul_buf.data_ptr(), S, c.d_ffn, stream)
+            h = torch.addmm(h, self._smul_buf, layer.w2)
+        h = _rms(h, self.ln_f)
+        return (h @ self.embed.t()).argmax(-1)  # [S]
+
     GRAPH_BUCKET = 256
 
     def _decode_graphed(self, maxlen: int) -> torch.Tensor:
         """hipGraph-captured decode keyed by the maxlen bucket: one replay
         instead of ~8 kernel launches per layer.  pos/_last_tok are read
         inside the graph from their persistent device tensors."""
-        bucket = min(((maxlen + self.GRAPH_BUCKET - 1) // self.GRAPH_BUCKET) * self.GRAPH_BUCKET,
-                     self.cfg.max_seq)
+        if self._hip_fused is not None and self._hip_attn is not None:
+            bucket = self.cfg.max_seq  # fused path reads per-slot pos: one graph fits all
+        else:
+            bucket = min(((maxlen + self.GRAPH_BUCKET - 1) // self.GRAPH_BUCKET) * self.GRAPH_BUCKET,
+                         self.cfg.max_seq)
         entry = self._graph_cache.get(bucket)
         if entry is None:
             torch.cuda.synchronize()
