@@ -219,6 +219,7 @@ PUBLIC_PATHS = {
     "/get_model_info",
     "/get_server_info",
     "/metrics",
+    "/openapi.json",
 }
 ADMIN_PATHS = {
     "/flush_cache",
@@ -475,6 +476,14 @@ def build_app(ctx: AppContext) -> web.Application:
     app.router.add_get("/get_model_info", get_model_info)
     app.router.add_get("/get_server_info", get_server_info)
     app.router.add_get("/metrics", metrics_endpoint)
+
+    async def openapi_json(request: web.Request):
+        # generated from the live route table (reference clients/openapi-gen)
+        from .openapi import build_openapi
+
+        return web.json_response(build_openapi(request.app))
+
+    app.router.add_get("/openapi.json", openapi_json)
 
     async def start_profile(request):
         out = {}

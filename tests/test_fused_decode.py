@@ -71,7 +71,7 @@ class TestFusedDecodeKernels:
 
         cfg = TorchEngineConfig.tiny()
         cfg.max_slots = 8
-        eng = TorchEngine(cfg, device="cuda:0", graphs=False, dtype="bfloat16")
+        eng = TorchEngine(cfg, device="cuda:0", graphs=False)
         assert eng._hip_fused is not None, "fused kernels must load on a GPU box"
         for i in range(4):
             eng.submit([7 + i, 11, 13, 17, 19, 23 + i], 4, rid=f"r{i}")

@@ -129,3 +129,36 @@ def test_anthropic_image_block():
     parts = chat["messages"][0]["content"]
     assert parts[1]["type"] == "image_url"
     assert parts[1]["image_url"]["url"].startswith("data:image/png;base64,")
+
+
+def test_openapi_schema(runner):
+    """GET /openapi.json covers the live route table with valid structure
+    (reference clients/openapi-gen)."""
+    async def run():
+        from tests.test_gateway_e2e import make_ctx, start_client, stop_all
+
+        ctx, engines = make_ctx(n_workers=1)
+        client = await start_client(ctx, engines)
+        try:
+            r = await client.get("/openapi.json")
+            assert r.status == 200
+            doc = await r.json()
+            assert doc["openapi"].startswith("3.")
+            for path in ("/v1/chat/completions", "/v1/completions", "/v1/embeddings",
+                         "/v1/responses", "/v1/messages", "/workers", "/health"):
+                assert path in doc["paths"], path
+            post = doc["paths"]["/v1/chat/completions"]["post"]
+            ref = post["requestBody"]["content"]["application/json"]["schema"]["$ref"]
+            assert ref.endswith("ChatCompletionRequest")
+            # every $ref resolves
+            import json as _json
+
+            blob = _json.dumps(doc)
+            import re as _re
+
+            for m in set(_re.findall(r'#/components/schemas/(\w+)', blob)):
+                assert m in doc["components"]["schemas"], m
+        finally:
+            await stop_all(client, engines)
+
+    runner(run())
