@@ -466,6 +466,9 @@ def build_app(ctx: AppContext) -> web.Application:
 
     app.router.add_post("/v1/responses", v1_responses)
     add_responses_routes(app)
+    from .interactions_routes import add_interactions_routes
+
+    add_interactions_routes(app)
 
     app.router.add_get("/liveness", liveness)
     app.router.add_get("/readiness", readiness)

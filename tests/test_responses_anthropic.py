@@ -140,3 +140,45 @@ def test_anthropic_requires_messages(runner):
             await stop_all(client, engines)
 
     runner(run())
+
+
+def test_interactions_api(runner):
+    """Gemini Interactions API (reference routers/gemini/, interactions.rs)."""
+    async def run():
+        ctx, engines = make_ctx()
+        client = await start_client(ctx, engines)
+        try:
+            r = await client.post("/v1/interactions", json={"input": "hello"})
+            assert r.status == 400  # model or agent required
+            r = await client.post(
+                "/v1/interactions",
+                json={"model": "mock-model", "input": "what is the answer to everything?",
+                      "generation_config": {"max_output_tokens": 8}},
+            )
+            assert r.status == 200
+            first = await r.json()
+            assert first["object"] == "interaction" and first["status"] == "completed"
+            assert first["outputs"] and first["outputs"][0]["type"] == "text"
+            assert first["usage"]["output_tokens"] > 0
+            # chained turn via previous_interaction_id
+            r = await client.post(
+                "/v1/interactions",
+                json={"model": "mock-model", "input": "and why?",
+                      "previous_interaction_id": first["id"],
+                      "generation_config": {"max_output_tokens": 8}},
+            )
+            assert r.status == 200
+            second = await r.json()
+            assert second["previous_interaction_id"] == first["id"]
+            # system_instruction + list input forms
+            r = await client.post(
+                "/v1/interactions",
+                json={"model": "mock-model", "system_instruction": "be brief",
+                      "input": [{"role": "user", "parts": [{"text": "hi"}]}],
+                      "generation_config": {"max_output_tokens": 4}},
+            )
+            assert r.status == 200
+        finally:
+            await stop_all(client, engines)
+
+    runner(run())
