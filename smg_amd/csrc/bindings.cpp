@@ -56,6 +56,10 @@ int smg_rope_kv_store_launch(const void* qkv, const void* freqs, const void* pos
                              void* k_cache, void* v_cache, void* q_out,
                              int n_slots, int n_heads, int max_seq, int head_dim, void* stream);
 int smg_silu_mul_launch(const void* gu, void* out, long long rows, long long inner, void* stream);
+int smg_rope_prefill_launch(const void* qkv, const void* freqs, const void* slots,
+                            const void* starts, void* k_cache, void* v_cache, void* q_out,
+                            void* k_out, void* v_out, int B, int L, int n_heads, int max_seq,
+                            int head_dim, void* stream);
 }
 
 #define BPE_MAX_PIECE 64
@@ -343,6 +347,20 @@ PYBIND11_MODULE(_core, m) {
           py::arg("qkv"), py::arg("freqs"), py::arg("pos"), py::arg("k_cache"), py::arg("v_cache"),
           py::arg("q_out"), py::arg("n_slots"), py::arg("n_heads"), py::arg("max_seq"),
           py::arg("head_dim"), py::arg("stream"));
+    m.def("rope_prefill",
+          [](uintptr_t qkv, uintptr_t freqs, uintptr_t slots, uintptr_t starts, uintptr_t k_cache,
+             uintptr_t v_cache, uintptr_t q_out, uintptr_t k_out, uintptr_t v_out, int B, int L,
+             int n_heads, int max_seq, int head_dim, uintptr_t stream) {
+              int rc = smg_rope_prefill_launch(
+                  (const void*)qkv, (const void*)freqs, (const void*)slots, (const void*)starts,
+                  (void*)k_cache, (void*)v_cache, (void*)q_out, (void*)k_out, (void*)v_out, B, L,
+                  n_heads, max_seq, head_dim, (void*)stream);
+              if (rc != 0) throw std::runtime_error("rope_prefill launch failed rc=" + std::to_string(rc));
+          },
+          py::arg("qkv"), py::arg("freqs"), py::arg("slots"), py::arg("starts"), py::arg("k_cache"),
+          py::arg("v_cache"), py::arg("q_out"), py::arg("k_out"), py::arg("v_out"), py::arg("B"),
+          py::arg("L"), py::arg("n_heads"), py::arg("max_seq"), py::arg("head_dim"),
+          py::arg("stream"));
     m.def("silu_mul",
           [](uintptr_t gu, uintptr_t out, long long rows, long long inner, uintptr_t stream) {
               int rc = smg_silu_mul_launch((const void*)gu, (void*)out, rows, inner, (void*)stream);

@@ -88,6 +88,85 @@ extern "C" __global__ void __launch_bounds__(256) smg_silu_mul(
     }
 }
 
+// Prefill variant: one workgroup-wave per (request, position, head).  Applies
+// the rotary rotation at absolute position start[b]+t, scatters K/V into each
+// request's slot window in the KV arena, and emits q/k/v in the [B, H, L, hd]
+// layout sdpa wants (so the flash prefill consumes them with zero transpose
+// copies).  Replaces the per-layer torch chain: 2x rope complex-mul chains,
+// 2x advanced-index KV scatter, and the implicit transpose-contiguous copies.
+extern "C" __global__ void __launch_bounds__(WAVE) smg_rope_prefill(
+    const __hip_bfloat16* __restrict__ qkv,   // [B, L, 3*D]
+    const float2* __restrict__ freqs,         // [max_seq, hd/2]
+    const int* __restrict__ slots,            // [B]
+    const int* __restrict__ starts,           // [B]
+    __hip_bfloat16* __restrict__ k_cache,     // [n_slots, H, max_seq, hd]
+    __hip_bfloat16* __restrict__ v_cache,
+    __hip_bfloat16* __restrict__ q_out,       // [B, H, L, hd]
+    __hip_bfloat16* __restrict__ k_out,
+    __hip_bfloat16* __restrict__ v_out,
+    int B, int L, int n_heads, int max_seq, int head_dim) {
+    const int idx = blockIdx.x;
+    const int h = idx % n_heads;
+    const int bt = idx / n_heads;
+    const int t = bt % L;
+    const int b = bt / L;
+    if (b >= B) return;
+    const int lane = threadIdx.x;
+    const int pairs = head_dim >> 1;
+    const int D = n_heads * head_dim;
+    const int p = starts[b] + t;
+    const int slot = slots[b];
+
+    const __hip_bfloat16* qrow = qkv + ((size_t)b * L + t) * 3 * D + (size_t)h * head_dim;
+    const __hip_bfloat16* krow = qrow + D;
+    const __hip_bfloat16* vrow = qrow + 2 * D;
+    const float2* f = freqs + (size_t)p * pairs;
+
+    __hip_bfloat16* kc = k_cache + (((size_t)slot * n_heads + h) * max_seq + p) * head_dim;
+    __hip_bfloat16* vc = v_cache + (((size_t)slot * n_heads + h) * max_seq + p) * head_dim;
+    const size_t out_off = (((size_t)b * n_heads + h) * L + t) * head_dim;
+    __hip_bfloat16* qo = q_out + out_off;
+    __hip_bfloat16* ko = k_out + out_off;
+    __hip_bfloat16* vo = v_out + out_off;
+
+    for (int i = lane; i < pairs; i += WAVE) {
+        const float2 cs = f[i];
+        const int e = 2 * i;
+        const float q0 = __bfloat162float(qrow[e]);
+        const float q1 = __bfloat162float(qrow[e + 1]);
+        qo[e] = __float2bfloat16(q0 * cs.x - q1 * cs.y);
+        qo[e + 1] = __float2bfloat16(q0 * cs.y + q1 * cs.x);
+        const float k0 = __bfloat162float(krow[e]);
+        const float k1 = __bfloat162float(krow[e + 1]);
+        const __hip_bfloat16 kr0 = __float2bfloat16(k0 * cs.x - k1 * cs.y);
+        const __hip_bfloat16 kr1 = __float2bfloat16(k0 * cs.y + k1 * cs.x);
+        ko[e] = kr0;
+        ko[e + 1] = kr1;
+        kc[e] = kr0;
+        kc[e + 1] = kr1;
+        const __hip_bfloat16 v0 = vrow[e];
+        const __hip_bfloat16 v1 = vrow[e + 1];
+        vo[e] = v0;
+        vo[e + 1] = v1;
+        vc[e] = v0;
+        vc[e + 1] = v1;
+    }
+}
+
+extern "C" int smg_rope_prefill_launch(
+    const void* qkv, const void* freqs, const void* slots, const void* starts,
+    void* k_cache, void* v_cache, void* q_out, void* k_out, void* v_out,
+    int B, int L, int n_heads, int max_seq, int head_dim, void* stream) {
+    if (head_dim % 2 != 0 || head_dim > 256) return 1;
+    dim3 grid((unsigned)B * L * n_heads);
+    hipLaunchKernelGGL(smg_rope_prefill, grid, dim3(WAVE), 0, (hipStream_t)stream,
+                       (const __hip_bfloat16*)qkv, (const float2*)freqs, (const int*)slots,
+                       (const int*)starts, (__hip_bfloat16*)k_cache, (__hip_bfloat16*)v_cache,
+                       (__hip_bfloat16*)q_out, (__hip_bfloat16*)k_out, (__hip_bfloat16*)v_out,
+                       B, L, n_heads, max_seq, head_dim);
+    return (int)hipGetLastError();
+}
+
 extern "C" int smg_rope_kv_store_launch(
     const void* qkv, const void* freqs, const void* pos,
     void* k_cache, void* v_cache, void* q_out,

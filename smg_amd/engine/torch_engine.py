@@ -167,6 +167,7 @@ class TorchEngine:
                 if hasattr(_core, "rope_kv_store"):
                     self._hip_fused = _core.rope_kv_store
                     self._hip_silu_mul = _core.silu_mul
+                    self._hip_rope_prefill = getattr(_core, "rope_prefill", None)
                     self._q_buf = torch.zeros(
                         c.max_slots, c.d_model, device=self.device, dtype=self.dtype
                     )
@@ -345,6 +346,15 @@ class TorchEngine:
     def _mlp(self, h, layer):
         x = _rms(h, layer.ln2)
         gu = x @ layer.w13
+        if self._hip_silu_mul is not None:
+            inner = gu.shape[-1] // 2
+            rows = gu.numel() // gu.shape[-1]
+            smul = torch.empty(*gu.shape[:-1], inner, device=gu.device, dtype=gu.dtype)
+            self._hip_silu_mul(
+                gu.data_ptr(), smul.data_ptr(), rows, inner,
+                torch.cuda.current_stream().cuda_stream,
+            )
+            return torch.addmm(h.reshape(rows, -1), smul.view(rows, inner), layer.w2).view(h.shape)
         g, u = gu.chunk(2, dim=-1)
         return h + (F.silu(g) * u) @ layer.w2
 
@@ -381,11 +391,33 @@ class TorchEngine:
         if not fresh:
             kpos = torch.arange(t_max, device=self.device)
             mask = kpos.view(1, 1, 1, -1) <= pos.view(B, 1, L, 1)
+        use_fused = self._hip_fused is not None and getattr(self, "_hip_rope_prefill", None) is not None
+        if use_fused:
+            # fused rope + KV scatter + [B,H,L,hd] emit (csrc/fused_decode.hip
+            # smg_rope_prefill): replaces the rope complex-mul chains, the two
+            # advanced-index scatters, and sdpa's transpose-contiguous copies
+            slots_i32 = slots.to(torch.int32)
+            starts_i32 = starts.to(torch.int32)
+            qb = torch.empty(B, c.n_heads, L, c.head_dim, device=self.device, dtype=self.dtype)
+            kb = torch.empty_like(qb)
+            vb = torch.empty_like(qb)
+            stream = torch.cuda.current_stream().cuda_stream
+            freqs_ptr = self.freqs_cis.data_ptr()
         for li, layer in enumerate(self.layers):
-            q, k, v = self._qkv(h, layer, freqs)
-            # scatter this chunk's K/V into each request's slot window
-            self.kv[li, 0][slots[:, None], :, pos] = k.permute(0, 2, 1, 3)
-            self.kv[li, 1][slots[:, None], :, pos] = v.permute(0, 2, 1, 3)
+            if use_fused:
+                qkv = (_rms(h, layer.ln1) @ layer.wqkv).view(B, L, 3 * c.d_model)
+                self._hip_rope_prefill(
+                    qkv.data_ptr(), freqs_ptr, slots_i32.data_ptr(), starts_i32.data_ptr(),
+                    self.kv[li, 0].data_ptr(), self.kv[li, 1].data_ptr(),
+                    qb.data_ptr(), kb.data_ptr(), vb.data_ptr(),
+                    B, L, c.n_heads, c.max_seq, c.head_dim, stream,
+                )
+                q, k, v = qb, kb, vb
+            else:
+                q, k, v = self._qkv(h, layer, freqs)
+                # scatter this chunk's K/V into each request's slot window
+                self.kv[li, 0][slots[:, None], :, pos] = k.permute(0, 2, 1, 3)
+                self.kv[li, 1][slots[:, None], :, pos] = v.permute(0, 2, 1, 3)
             if fresh:
                 # no history: attend within the chunk itself, flash kernel
                 attn = F.scaled_dot_product_attention(q, k, v, is_causal=True)
@@ -393,7 +425,8 @@ class TorchEngine:
                 kk = self.kv[li, 0].index_select(0, slots)[:, :, :t_max]
                 vv = self.kv[li, 1].index_select(0, slots)[:, :, :t_max]
                 attn = F.scaled_dot_product_attention(q, kk, vv, attn_mask=mask)
-            h = h + attn.transpose(1, 2).reshape(B, L, c.d_model) @ layer.wo
+            attn2 = attn.transpose(1, 2).reshape(B * L, c.d_model)
+            h = torch.addmm(h.view(B * L, c.d_model), attn2, layer.wo).view(B, L, c.d_model)
             h = self._mlp(h, layer)
         for slot, start, toks in items:
             self._seq_len_host[slot] = start + L
