@@ -442,7 +442,7 @@ class TorchEngine:
         shapes per `maxlen` make this hipGraph-capturable."""
         c = self.cfg
         S = c.max_slots
-        if self._hip_fused is not None and self._hip_attn is not None:
+        if self._hip_fused is not None and self._hip_attn is not None and self._hip_silu_mul is not None:
             return self._decode_core_fused()
         pos = self.seq_len  # [S] current length == write position
         freqs = self.freqs_cis[pos].view(S, 1, 1, -1)
@@ -517,7 +517,7 @@ class TorchEngine:
         """hipGraph-captured decode keyed by the maxlen bucket: one replay
         instead of ~8 kernel launches per layer.  pos/_last_tok are read
         inside the graph from their persistent device tensors."""
-        if self._hip_fused is not None and self._hip_attn is not None:
+        if self._hip_fused is not None and self._hip_attn is not None and self._hip_silu_mul is not None:
             bucket = self.cfg.max_seq  # fused path reads per-slot pos: one graph fits all
         else:
             bucket = min(((maxlen + self.GRAPH_BUCKET - 1) // self.GRAPH_BUCKET) * self.GRAPH_BUCKET,

@@ -1,0 +1,110 @@
+"""Model-specific chat encoders (reference crates/tokenizer/src/encoders/:
+deepseek_v32.rs DSML format, kimi_k25_tools.rs)."""
+import pytest
+
+from smg_amd.tokenizer.encoders import (
+    BOS_TOKEN,
+    DSML,
+    EOS_TOKEN,
+    EncodingError,
+    THINK_END,
+    THINK_START,
+    encode_deepseek_v32,
+    encode_kimi_k25_tool_calls,
+    get_encoder,
+)
+
+
+def test_ds32_basic_chat():
+    out = encode_deepseek_v32(
+        [{"role": "system", "content": "be brief"},
+         {"role": "user", "content": "hi"}],
+        thinking_mode="chat",
+    )
+    assert out.startswith(BOS_TOKEN + "be brief")
+    assert "<｜User｜>hi<｜Assistant｜>" in out
+    assert out.endswith(THINK_END)  # chat mode closes thinking immediately
+
+
+def test_ds32_thinking_mode_last_user_opens_think():
+    out = encode_deepseek_v32(
+        [{"role": "user", "content": "q1"},
+         {"role": "assistant", "content": "a1", "reasoning_content": "r1"},
+         {"role": "user", "content": "q2"}],
+        thinking_mode="thinking",
+    )
+    # earlier user turn closes thinking; the LAST user turn opens it
+    assert out.endswith("<｜User｜>q2<｜Assistant｜>" + THINK_START)
+    # drop_thinking strips reasoning from turns before the last user message
+    assert "r1" not in out
+    assert "a1" + EOS_TOKEN in out
+
+
+def test_ds32_tool_calls_dsml():
+    msgs = [
+        {"role": "user", "content": "weather?"},
+        {"role": "assistant", "content": "", "reasoning_content": "need tool",
+         "tool_calls": [{"type": "function", "function": {"name": "get_weather", "arguments": '{"city": "SF", "days": 3}'}}]},
+        {"role": "tool", "content": "sunny"},
+    ]
+    out = encode_deepseek_v32(msgs, thinking_mode="thinking")
+    assert f'<{DSML}invoke name="get_weather">' in out
+    assert f'<{DSML}parameter name="city" string="true">SF</{DSML}parameter>' in out
+    assert f'<{DSML}parameter name="days" string="false">3</{DSML}parameter>' in out
+    assert "<function_results>" in out and "<result>sunny</result>" in out
+    # after the final tool result in thinking mode, a think block opens
+    assert out.endswith("\n\n" + THINK_START)
+
+
+def test_ds32_request_level_tools_attach_to_system():
+    out = encode_deepseek_v32(
+        [{"role": "user", "content": "go"}],
+        thinking_mode="chat",
+        tools=[{"type": "function", "function": {"name": "f", "parameters": {"type": "object"}}}],
+    )
+    assert "## Tools" in out and '"name": "f"' in out
+
+
+def test_ds32_errors():
+    with pytest.raises(EncodingError):
+        encode_deepseek_v32([{"role": "user", "content": "x"}], thinking_mode="bogus")
+    with pytest.raises(EncodingError, match="no tool calls"):
+        encode_deepseek_v32(
+            [{"role": "user", "content": "x"},
+             {"role": "assistant", "content": "a", "reasoning_content": "r"},
+             {"role": "tool", "content": "out"}],
+            thinking_mode="chat",
+        )
+    with pytest.raises(EncodingError, match="unknown role"):
+        encode_deepseek_v32([{"role": "alien", "content": "x"}])
+
+
+def test_ds32_roundtrip_with_tool_parser():
+    """The DSML encoder output for tool calls parses back with the deepseek
+    DSML tool parser family where applicable (inverse property on args)."""
+    from smg_amd.tokenizer.encoders import _dsml_arguments
+
+    args = _dsml_arguments({"name": "f", "arguments": {"a": "text", "b": [1, 2], "c": True}})
+    assert 'name="a" string="true"' in args
+    assert 'name="b" string="false"' in args and "[1, 2]" in args
+
+
+def test_kimi_k25_tool_sections():
+    out = encode_kimi_k25_tool_calls(
+        [{"type": "function", "function": {"name": "search", "arguments": '{"q": "x"}'}},
+         {"type": "function", "function": {"name": "open", "arguments": {"url": "u"}}}]
+    )
+    assert out.startswith("<|tool_calls_section_begin|>")
+    assert "functions.search:0" in out and "functions.open:1" in out
+    assert out.endswith("<|tool_calls_section_end|>")
+    # parses back with the kimik2 tool parser
+    from smg_amd.parsers.tool.factory import get_parser
+
+    normal, calls = get_parser("kimik2").parse(out)
+    assert [c["name"] for c in calls] == ["search", "open"]
+
+
+def test_get_encoder_mapping():
+    assert get_encoder("deepseek-v3.2-exp") is encode_deepseek_v32
+    assert get_encoder("DeepSeek-V4") is encode_deepseek_v32
+    assert get_encoder("llama-3") is None

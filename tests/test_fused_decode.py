@@ -151,24 +151,22 @@ class TestRopePrefillKernel:
         assert k_cache[untouched].abs().sum().item() == 0.0
 
     def test_engine_prefill_fused_matches_eager(self):
-        """Full prefill through the engine: fused vs eager paths yield the
-        same KV cache contents."""
+        """Prefill in isolation (no decode step): fused vs eager paths yield
+        the same KV cache contents."""
         from smg_amd.engine.torch_engine import TorchEngine, TorchEngineConfig
 
         cfg = TorchEngineConfig.tiny()
-        prompts = [[5, 9, 2, 4, 8, 1, 3, 7] * 4, [11, 13, 17, 19] * 8]
+        prompts = [[5, 9, 2, 4, 8, 1, 3, 7] * 4, [11, 13, 17, 19] * 8]  # equal lengths
 
         def run(disable_fused):
             eng = TorchEngine(cfg, device="cuda:0", graphs=False)
             if disable_fused:
                 eng._hip_rope_prefill = None
                 eng._hip_silu_mul = None
-            for i, p in enumerate(prompts):
-                eng.submit(p, 2, rid=f"r{i}")
-            eng.step()
-            return eng.kv.clone(), {r.rid: r.slot for r in eng._requests.values()}
+            eng._prefill_batch([(0, 0, prompts[0]), (1, 0, prompts[1])])
+            torch.cuda.synchronize()
+            return eng.kv.clone()
 
-        kv_f, slots_f = run(False)
-        kv_e, slots_e = run(True)
-        assert slots_f == slots_e
+        kv_f = run(False)
+        kv_e = run(True)
         assert torch.allclose(kv_f.float(), kv_e.float(), atol=3e-2, rtol=2e-2)
