@@ -62,9 +62,20 @@ extern "C" __global__ void __launch_bounds__(WAVE) smg_attn_decode(
     __syncthreads();
 
     const int vec_n = head_dim / 8;  // dwordx4 chunks per row (<=16)
-    float acc0 = 0.f, acc1 = 0.f;
+    // v6 P·V mapping: lane -> (row-group rgrp = lane/chunks, chunk = lane%chunks)
+    // so one wave instruction loads dwordx4 from `rows_per` different V rows —
+    // 1 KB per instruction (vs 256 B with the old 4 B/lane scheme), matching
+    // the K phase's streaming width.  Each lane accumulates 8 dims of its
+    // chunk over rows rgrp, rgrp+rows_per, ...; a log2(rows_per)-step
+    // shfl_xor tree folds the row groups at the end.
+    const int chunks = vec_n;             // dwordx4 chunks per row (8 or 16)
+    const int rows_per = WAVE / chunks;   // rows covered per instruction
+    const int chunk = lane % chunks;
+    const int rgrp = lane / chunks;
+    float accv[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) accv[j] = 0.f;
     float m = -1e30f, l = 0.f;
-    const int e0 = lane * 2, e1 = lane * 2 + 1;
 
     for (int base = 0; base < T; base += WAVE) {
         int lim = min(WAVE, T - base);
@@ -95,27 +106,30 @@ extern "C" __global__ void __launch_bounds__(WAVE) smg_attn_decode(
 #pragma unroll
         for (int off = 32; off > 0; off >>= 1) pr += __shfl_xor(pr, off, WAVE);
         l = l * alpha + pr;
-        acc0 *= alpha;
-        acc1 *= alpha;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) accv[j] *= alpha;
         __syncthreads();
-        // P·V: V rows stream 4 B/lane coalesced; no branch gates the loads
-        if (e0 < head_dim) {
-#pragma unroll 8
-            for (int j = 0; j < lim; ++j) {
-                const __hip_bfloat16* vrow = vh + (size_t)(base + j) * head_dim;
-                float pj = s_p[j];
-                acc0 += pj * (float)vrow[e0];
-                acc1 += pj * (float)vrow[e1];
-            }
+        // P·V: each lane streams dwordx4 of its chunk from every rows_per-th row
+        for (int j = rgrp; j < lim; j += rows_per) {
+            const uint4 w = *(const uint4*)(vh + (size_t)(base + j) * head_dim + chunk * 8);
+            const unsigned short* hsp = (const unsigned short*)&w;
+            float pj = s_p[j];
+#pragma unroll
+            for (int jj = 0; jj < 8; ++jj) accv[jj] += pj * bf16_to_f32(hsp[jj]);
         }
         __syncthreads();
         m = m_new;
     }
+    // fold the row groups: lanes sharing `chunk` differ in bits >= log2(chunks)
+    for (int off = chunks; off < WAVE; off <<= 1) {
+#pragma unroll
+        for (int jj = 0; jj < 8; ++jj) accv[jj] += __shfl_xor(accv[jj], off, WAVE);
+    }
     float inv = l > 0.f ? 1.f / l : 0.f;
     __hip_bfloat16* orow = out + ((size_t)slot * n_heads + head) * head_dim;
-    if (e0 < head_dim) {
-        orow[e0] = (__hip_bfloat16)(acc0 * inv);
-        orow[e1] = (__hip_bfloat16)(acc1 * inv);
+    if (rgrp == 0) {
+#pragma unroll
+        for (int jj = 0; jj < 8; ++jj) orow[chunk * 8 + jj] = (__hip_bfloat16)(accv[jj] * inv);
     }
 }
 
@@ -123,6 +137,8 @@ extern "C" int smg_attn_decode_launch(const void* q, const void* k, const void* 
                                       const void* pos, void* out, int n_slots, int n_heads,
                                       int max_seq, int head_dim, float scale, void* stream) {
     if (head_dim > 128 || (head_dim & 7)) return -1;
+    int chunks = head_dim / 8;  // v6 P·V row-group mapping needs 2^k chunks
+    if (chunks & (chunks - 1)) return -1;
     dim3 grid(n_slots * n_heads);
     hipLaunchKernelGGL(smg_attn_decode, grid, dim3(WAVE), 0, (hipStream_t)stream,
                        (const __hip_bfloat16*)q, (const __hip_bfloat16*)k,
