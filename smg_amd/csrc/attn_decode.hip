@@ -109,13 +109,29 @@ extern "C" __global__ void __launch_bounds__(WAVE) smg_attn_decode(
 #pragma unroll
         for (int j = 0; j < 8; ++j) accv[j] *= alpha;
         __syncthreads();
-        // P·V: each lane streams dwordx4 of its chunk from every rows_per-th row
-        for (int j = rgrp; j < lim; j += rows_per) {
-            const uint4 w = *(const uint4*)(vh + (size_t)(base + j) * head_dim + chunk * 8);
-            const unsigned short* hsp = (const unsigned short*)&w;
-            float pj = s_p[j];
+        // P·V: each lane streams dwordx4 of its chunk from every rows_per-th
+        // row.  Full tiles take the unrolled fast path (compile-time trip
+        // count exposes all 16 loads to the scheduler); only the final
+        // partial tile pays the runtime loop.
+        const __hip_bfloat16* vtile = vh + (size_t)base * head_dim + chunk * 8;
+        if (lim == WAVE && head_dim == 128) {
 #pragma unroll
-            for (int jj = 0; jj < 8; ++jj) accv[jj] += pj * bf16_to_f32(hsp[jj]);
+            for (int it = 0; it < 16; ++it) {
+                const int j = rgrp + it * 4;  // rows_per == 4 when chunks == 16
+                const uint4 w = *(const uint4*)(vtile + (size_t)j * head_dim);
+                const unsigned short* hsp = (const unsigned short*)&w;
+                float pj = s_p[j];
+#pragma unroll
+                for (int jj = 0; jj < 8; ++jj) accv[jj] += pj * bf16_to_f32(hsp[jj]);
+            }
+        } else {
+            for (int j = rgrp; j < lim; j += rows_per) {
+                const uint4 w = *(const uint4*)(vtile + (size_t)j * head_dim);
+                const unsigned short* hsp = (const unsigned short*)&w;
+                float pj = s_p[j];
+#pragma unroll
+                for (int jj = 0; jj < 8; ++jj) accv[jj] += pj * bf16_to_f32(hsp[jj]);
+            }
         }
         __syncthreads();
         m = m_new;

@@ -122,8 +122,27 @@ async def v1_realtime_ws(request: web.Request):
     return ws_client
 
 
+async def v1_realtime_calls(request: web.Request):
+    """WebRTC call setup (reference realtime/webrtc_bridge.rs — str0m-based
+    media relay).  This build ships the WebSocket relay only; WebRTC needs a
+    media stack (ICE/DTLS/SRTP) that is out of scope here, so the endpoint
+    exists and says so instead of 404ing."""
+    from ..protocols.openai import error_body
+
+    return web.Response(
+        status=501,
+        body=error_body(
+            "WebRTC realtime calls are not supported by this build; connect via"
+            " the WebSocket endpoint GET /v1/realtime instead",
+            501,
+        ),
+        content_type="application/json",
+    )
+
+
 def add_realtime_routes(app: web.Application) -> None:
     app.router.add_get("/v1/realtime", v1_realtime_ws)
     app.router.add_post("/v1/realtime/sessions", v1_realtime_session)
     app.router.add_post("/v1/realtime/client_secrets", v1_realtime_session)
     app.router.add_post("/v1/realtime/transcription_sessions", v1_realtime_session)
+    app.router.add_post("/v1/realtime/calls", v1_realtime_calls)
