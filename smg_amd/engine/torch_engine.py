@@ -176,6 +176,7 @@ class TorchEngine:
                     )
             except ImportError:
                 pass
+        self._flash_lse = self._probe_flash_lse() if self.device.type == "cuda" else None
 
     # ---- API -------------------------------------------------------------
     def submit(self, tokens: List[int], max_new_tokens: int, rid: Optional[str] = None) -> str:
@@ -292,6 +293,25 @@ class TorchEngine:
         return produced
 
     # ---- forwards ----------------------------------------------------------
+    def _probe_flash_lse(self):
+        """(q,k,v,is_causal) -> (out, logsumexp) via the flash kernel, or None
+        when the aten op is unavailable.  The lse output lets the suffix-
+        prefill path merge a no-mask history pass with a square causal chunk
+        pass instead of taking the masked-sdpa math path."""
+
+        def call(q, k, v, causal):
+            r = torch.ops.aten._scaled_dot_product_flash_attention(q, k, v, 0.0, causal, False)
+            return r[0], r[1]
+
+        try:
+            probe = torch.randn(1, 2, 8, 64, device=self.device, dtype=self.dtype)
+            out, lse = call(probe, probe, probe, True)
+            if out.shape != probe.shape or lse.shape[-1] != 8:
+                return None
+            return call
+        except Exception:
+            return None
+
     @staticmethod
     def _apply_rope(x, freqs):
         # x: [B, H, T, D]; freqs complex broadcastable to [B, H, T, D/2].
@@ -392,6 +412,7 @@ class TorchEngine:
             kpos = torch.arange(t_max, device=self.device)
             mask = kpos.view(1, 1, 1, -1) <= pos.view(B, 1, L, 1)
         use_fused = self._hip_fused is not None and getattr(self, "_hip_rope_prefill", None) is not None
+        uniform_start = not fresh and all(int(st) == int(items[0][1]) for _, st, _ in items)
         if use_fused:
             # fused rope + KV scatter + [B,H,L,hd] emit (csrc/fused_decode.hip
             # smg_rope_prefill): replaces the rope complex-mul chains, the two
@@ -421,6 +442,20 @@ class TorchEngine:
             if fresh:
                 # no history: attend within the chunk itself, flash kernel
                 attn = F.scaled_dot_product_attention(q, k, v, is_causal=True)
+            elif use_fused and uniform_start and self._flash_lse:
+                # uniform-start suffix chunk (the prefix-cache-hit fast path):
+                # two FLASH passes — history cross-attention (all keys valid,
+                # no mask) + square causal chunk — merged by logsumexp.
+                # Replaces the masked-sdpa math path (bmm + 37 MB mask add +
+                # softmax per layer).
+                start0 = int(starts[0].item())
+                kk = self.kv[li, 0].index_select(0, slots)[:, :, :start0]
+                vv = self.kv[li, 1].index_select(0, slots)[:, :, :start0]
+                o1, lse1 = self._flash_lse(q, kk, vv, False)
+                o2, lse2 = self._flash_lse(q, k, v, True)
+                lse_tot = torch.logaddexp(lse1, lse2)
+                attn = o1 * (lse1 - lse_tot).exp().unsqueeze(-1) + o2 * (lse2 - lse_tot).exp().unsqueeze(-1)
+                attn = attn.to(q.dtype)
             else:
                 kk = self.kv[li, 0].index_select(0, slots)[:, :, :t_max]
                 vv = self.kv[li, 1].index_select(0, slots)[:, :, :t_max]

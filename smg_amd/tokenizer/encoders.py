@@ -227,9 +227,194 @@ def encode_kimi_k25_tool_calls(tool_calls: List[Dict[str, Any]], start_index: in
     return "".join(parts)
 
 
+# ---- Kimi K3 XTML chat template (kimi_k3_xtml.rs) ---------------------------
+XT_OPEN = "<|open|>"
+XT_CLOSE = "<|close|>"
+XT_SEP = "<|sep|>"
+XT_EOM = "<|end_of_msg|>"
+XT_IMAGE = "<|kimi_image_placeholder|>"
+K3_EFFORTS = ("low", "high", "max")
+
+
+def _xt_escape(v: str) -> str:
+    return v.replace("&", "&amp;").replace('"', "&quot;")
+
+
+def _xt_open(tag: str, attrs: Optional[List[tuple]] = None) -> str:
+    s = XT_OPEN + tag
+    for k, v in attrs or []:
+        s += f' {k}="{_xt_escape(str(v))}"'
+    return s + XT_SEP
+
+
+def _xt_close(tag: str) -> str:
+    return XT_CLOSE + tag + XT_SEP
+
+
+def _xt_content(content: Any) -> str:
+    if isinstance(content, str):
+        return content
+    if isinstance(content, list):
+        out = []
+        for p in content:
+            if not isinstance(p, dict):
+                continue
+            if p.get("type") in ("image", "image_url"):
+                out.append(XT_IMAGE)
+            elif isinstance(p.get("text"), str):
+                out.append(p["text"])
+        return "".join(out)
+    return ""
+
+
+def _xt_type(v: Any) -> str:
+    if isinstance(v, bool):
+        return "boolean"
+    if v is None:
+        return "null"
+    if isinstance(v, (int, float)):
+        return "number"
+    if isinstance(v, str):
+        return "string"
+    if isinstance(v, dict):
+        return "object"
+    return "array"
+
+
+def _xt_value(v: Any) -> str:
+    return v if isinstance(v, str) else _to_json(v)
+
+
+def _xt_internal_system(kind: str, body: str) -> str:
+    return (
+        _xt_open("message", [("role", "system"), ("type", kind)])
+        + body.strip()
+        + _xt_close("message")
+        + XT_EOM
+    )
+
+
+def _xt_assistant_segments(msg: Dict, thinking: bool) -> str:
+    out = []
+    if thinking:
+        # the think channel is structural: open/close even when empty
+        reasoning = msg.get("reasoning_content") or msg.get("reasoning") or ""
+        out.append(_xt_open("think") + (reasoning if str(reasoning).strip() else "") + _xt_close("think"))
+    out.append(_xt_open("response") + _xt_content(msg.get("content")) + _xt_close("response"))
+    tool_calls = msg.get("tool_calls") or []
+    if tool_calls:
+        out.append(_xt_open("tools"))
+        for i, tc in enumerate(tool_calls, start=1):
+            fn = tc.get("function", tc) if isinstance(tc.get("function"), dict) else tc
+            name = fn.get("name")
+            if not name:
+                raise EncodingError("Kimi K3 tool call is missing a function name")
+            out.append(_xt_open("call", [("tool", name), ("index", i)]))
+            args = fn.get("arguments")
+            if isinstance(args, str):
+                try:
+                    args = json.loads(args)
+                except json.JSONDecodeError:
+                    out.append(_xt_open("json", [("type", "object")]) + fn["arguments"] + _xt_close("json"))
+                    args = None
+            if isinstance(args, dict):
+                for k, v in args.items():
+                    out.append(
+                        _xt_open("argument", [("key", k), ("type", _xt_type(v))]) + _xt_value(v) + _xt_close("argument")
+                    )
+            out.append(_xt_close("call"))
+        out.append(_xt_close("tools"))
+    return "".join(out)
+
+
+def encode_kimi_k3_xtml(
+    messages: List[Dict[str, Any]],
+    tools: Optional[List[Dict]] = None,
+    thinking: bool = True,
+    thinking_effort: Optional[str] = None,
+    tool_choice: Optional[str] = None,
+    add_generation_prompt: bool = True,
+) -> str:
+    """OpenAI-style messages -> Kimi K3 XTML prompt (kimi_k3_xtml.rs
+    render_xtml): <|open|>tag attrs<|sep|>…<|close|>tag<|sep|> framing,
+    tool-declare + thinking-effort internal system messages, think/response/
+    tools assistant channels."""
+    if thinking_effort is not None and thinking_effort not in K3_EFFORTS:
+        raise EncodingError(f"unsupported thinking_effort {thinking_effort!r}; supported: {list(K3_EFFORTS)}")
+    out = []
+    if tools:
+        body = (
+            "# Tools\nHere are the available tools, described in JSONSchema.\n\n"
+            f"```json\n{json.dumps(tools, ensure_ascii=False, separators=(',', ':'), sort_keys=True)}\n```"
+        )
+        out.append(_xt_internal_system("tool-declare", body))
+    if thinking and thinking_effort:
+        out.append(
+            _xt_internal_system(
+                "thinking-effort",
+                "`thinking_effort` guides on how much to think in your thinking channel "
+                "(not including the response channel), supported values include `low`, "
+                f"`medium`, `high`, and `max`.\nNow the system is invoked with `thinking_effort={thinking_effort}`.",
+            )
+        )
+    current_tool_calls: List[Dict] = []
+    tool_index = 0
+    for msg in messages:
+        role = msg.get("role", "")
+        if role in ("user", "system"):
+            attrs = [("role", role)]
+            if msg.get("name"):
+                attrs.append(("name", msg["name"]))
+            out.append(_xt_open("message", attrs) + _xt_content(msg.get("content")) + _xt_close("message") + XT_EOM)
+        elif role == "tool":
+            tool_index += 1
+            name = msg.get("tool") or msg.get("name")
+            if not name and tool_index <= len(current_tool_calls):
+                tc = current_tool_calls[tool_index - 1]
+                fn = tc.get("function", tc) if isinstance(tc.get("function"), dict) else tc
+                name = fn.get("name")
+            if not name:
+                raise EncodingError("Kimi K3 tool messages need a resolvable tool name")
+            out.append(
+                _xt_open("message", [("role", "tool"), ("tool", name), ("index", tool_index)])
+                + _xt_content(msg.get("content"))
+                + _xt_close("message")
+                + XT_EOM
+            )
+        elif role == "assistant":
+            current_tool_calls = msg.get("tool_calls") or []
+            tool_index = 0
+            attrs = [("role", "assistant")]
+            if msg.get("name"):
+                attrs.append(("name", msg["name"]))
+            out.append(
+                _xt_open("message", attrs) + _xt_assistant_segments(msg, thinking) + _xt_close("message") + XT_EOM
+            )
+        # unknown roles render nothing (kimi_k3_xtml.rs loop)
+    if tool_choice == "required":
+        out.append(
+            _xt_internal_system(
+                "tool-choice",
+                "The system is invoked with `tool_choice=required`.\nYou MUST call tools in the next message.",
+            )
+        )
+    elif tool_choice == "none":
+        out.append(
+            _xt_internal_system(
+                "tool-choice",
+                "The system is invoked with `tool_choice=none`.\nYou MUST NOT call any tools in the next message.",
+            )
+        )
+    if add_generation_prompt:
+        out.append(_xt_open("message", [("role", "assistant")]))
+        out.append(_xt_open("think" if thinking else "response"))
+    return "".join(out)
+
+
 ENCODERS = {
     "deepseek_v32": encode_deepseek_v32,
     "deepseek_v4": encode_deepseek_v32,  # v4 shares the DSML surface (deepseek_v4.rs)
+    "kimi_k3_xtml": encode_kimi_k3_xtml,
 }
 
 
@@ -241,4 +426,6 @@ def get_encoder(name_or_model: Optional[str]):
         return ENCODERS[low]
     if "deepseek-v3.2" in low or "deepseek_v32" in low or "deepseek-v4" in low:
         return encode_deepseek_v32
+    if "kimi-k3" in low or "kimi_k3" in low:
+        return encode_kimi_k3_xtml
     return None

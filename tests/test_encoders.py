@@ -108,3 +108,57 @@ def test_get_encoder_mapping():
     assert get_encoder("deepseek-v3.2-exp") is encode_deepseek_v32
     assert get_encoder("DeepSeek-V4") is encode_deepseek_v32
     assert get_encoder("llama-3") is None
+
+
+def test_kimi_k3_xtml_basic():
+    from smg_amd.tokenizer.encoders import encode_kimi_k3_xtml
+
+    out = encode_kimi_k3_xtml(
+        [{"role": "system", "content": "be kind"},
+         {"role": "user", "content": "hi"}],
+        thinking=True, thinking_effort="max",
+    )
+    assert '<|open|>message role="system" type="thinking-effort"<|sep|>' in out
+    assert "thinking_effort=max" in out
+    assert '<|open|>message role="user"<|sep|>hi<|close|>message<|sep|><|end_of_msg|>' in out
+    # generation prompt opens an assistant message and the think channel
+    assert out.endswith('<|open|>message role="assistant"<|sep|><|open|>think<|sep|>')
+
+
+def test_kimi_k3_xtml_tools_and_calls():
+    from smg_amd.tokenizer.encoders import EncodingError, encode_kimi_k3_xtml
+
+    msgs = [
+        {"role": "user", "content": "weather?"},
+        {"role": "assistant", "content": "", "reasoning_content": "use tool",
+         "tool_calls": [{"type": "function", "function": {"name": "wx", "arguments": '{"city": "SF", "days": 3}'}}]},
+        {"role": "tool", "content": "sunny"},
+    ]
+    out = encode_kimi_k3_xtml(
+        msgs,
+        tools=[{"type": "function", "function": {"name": "wx", "parameters": {"type": "object"}}}],
+        tool_choice="required",
+    )
+    assert '<|open|>message role="system" type="tool-declare"<|sep|># Tools' in out
+    assert '<|open|>call tool="wx" index="1"<|sep|>' in out
+    assert '<|open|>argument key="city" type="string"<|sep|>SF<|close|>argument<|sep|>' in out
+    assert '<|open|>argument key="days" type="number"<|sep|>3<|close|>argument<|sep|>' in out
+    # tool result resolves its name from the preceding assistant call by order
+    assert '<|open|>message role="tool" tool="wx" index="1"<|sep|>sunny' in out
+    assert "tool_choice=required" in out
+    # think channel is structural: present even on plain assistant turns
+    assert "<|open|>think<|sep|>use tool<|close|>think<|sep|>" in out
+    with pytest.raises(EncodingError, match="thinking_effort"):
+        encode_kimi_k3_xtml(msgs, thinking_effort="medium")
+
+
+def test_kimi_k3_xtml_roundtrip_with_parsers():
+    """The K3 encoder's think channel parses back with the kimi_k3 reasoning
+    parser (inverse property)."""
+    from smg_amd.parsers.reasoning import get_reasoning_parser
+
+    raw = "<|open|>think<|sep|>pondering<|close|>think<|sep|>the answer"
+    rp = get_reasoning_parser("kimi_k3")
+    reasoning, normal = rp.parse(raw)
+    assert reasoning == "pondering"
+    assert "the answer" in normal

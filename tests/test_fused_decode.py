@@ -170,3 +170,27 @@ class TestRopePrefillKernel:
         kv_f = run(False)
         kv_e = run(True)
         assert torch.allclose(kv_f.float(), kv_e.float(), atol=3e-2, rtol=2e-2)
+
+    def test_suffix_prefill_flash_lse_matches_masked(self):
+        """Uniform-start suffix prefill: the two-pass flash+LSE merge must
+        match the masked-sdpa path."""
+        from smg_amd.engine.torch_engine import TorchEngine, TorchEngineConfig
+
+        cfg = TorchEngineConfig.tiny()
+        base = [3, 1, 4, 1, 5, 9, 2, 6] * 8  # 64-token shared prefix
+        sufs = [[i + 1, i + 2, i + 3, i + 4] * 4 for i in range(2)]  # 16 each
+
+        def run(force_masked):
+            eng = TorchEngine(cfg, device="cuda:0", graphs=False)
+            if force_masked:
+                eng._flash_lse = None
+            # cold prefill of the shared prefix into both slots
+            eng._prefill_batch([(0, 0, base), (1, 0, base)])
+            # uniform-start suffix chunks
+            eng._prefill_batch([(0, len(base), sufs[0]), (1, len(base), sufs[1])])
+            torch.cuda.synchronize()
+            return eng.kv.clone()
+
+        kv_flash = run(False)
+        kv_masked = run(True)
+        assert torch.allclose(kv_flash.float(), kv_masked.float(), atol=3e-2, rtol=2e-2)
