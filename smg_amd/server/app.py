@@ -330,11 +330,19 @@ async def metrics_endpoint(request):
 
 
 async def engine_metrics(request):
+    """Merged Prometheus exposition across the worker fleet (reference
+    worker/metrics_aggregator.rs via /engine_metrics); `?format=json` keeps
+    the compact summary."""
     ctx: AppContext = request.app[CTX_KEY]
-    out = {}
-    for w in ctx.worker_registry.all():
-        out[w.url] = {"token_usage": w.token_usage, "gen_throughput": w.gen_throughput, "load": w.active_requests}
-    return web.json_response(out)
+    if request.query.get("format") == "json":
+        out = {}
+        for w in ctx.worker_registry.all():
+            out[w.url] = {"token_usage": w.token_usage, "gen_throughput": w.gen_throughput, "load": w.active_requests}
+        return web.json_response(out)
+    from ..observability.aggregate import collect_engine_metrics
+
+    text = await collect_engine_metrics(ctx, session=getattr(ctx, "client_session", None))
+    return web.Response(body=text, content_type="text/plain")
 
 
 async def get_loads(request):
