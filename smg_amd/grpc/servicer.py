@@ -235,9 +235,43 @@ class SchedulerServicer(grpc.GenericRpcHandler):
             time.sleep(0.05)
 
     def _h_startprofile(self, request: bytes, context):
-        return api.dumps({"status": "profiling not active on this engine"})
+        """Engine profiling passthrough (reference /start_profile ->
+        StartProfile RPC, sglang_scheduler.proto:37-40).  On the GPU engine
+        this starts torch.profiler, which on ROCm records through
+        roctracer/kineto — the rocprofv3-family tooling."""
+        if self.adapter.is_mock:
+            return api.dumps({"status": "profiling not active on the mock engine"})
+        if getattr(self, "_profiler", None) is not None:
+            return api.dumps({"status": "already profiling"})
+        try:
+            import torch
+            from torch.profiler import ProfilerActivity, profile
 
-    _h_stopprofile = _h_startprofile
+            self._profiler = profile(activities=[ProfilerActivity.CPU, ProfilerActivity.CUDA])
+            self._profiler.__enter__()
+            return api.dumps({"status": "profiling started"})
+        except Exception as exc:
+            self._profiler = None
+            return api.dumps({"status": f"profiler unavailable: {exc}"})
+
+    def _h_stopprofile(self, request: bytes, context):
+        prof = getattr(self, "_profiler", None)
+        if prof is None:
+            return api.dumps({"status": "not profiling"})
+        self._profiler = None
+        try:
+            prof.__exit__(None, None, None)
+            import os
+            import tempfile
+
+            d = api.loads(request) if request else {}
+            out = (d.get("output_dir") or tempfile.gettempdir())
+            os.makedirs(out, exist_ok=True)
+            path = os.path.join(out, "smg_engine_trace.json")
+            prof.export_chrome_trace(path)
+            return api.dumps({"status": "profiling stopped", "trace": path})
+        except Exception as exc:
+            return api.dumps({"status": f"profiler stop failed: {exc}"})
 
     def _h_loadloraadapter(self, request: bytes, context):
         d = api.loads(request)

@@ -65,6 +65,71 @@ class GatewayMetrics:
         self.tree_size = Gauge("smg_policy_tree_nodes", "prefix-tree node count", ["model"], registry=r)
         self.queue_depth = Gauge("smg_scheduler_queue_depth", "admission queue depth", ["klass"], registry=r)
         self.rate_limited = Counter("smg_rate_limited_total", "429 rejections", ["tenant"], registry=r)
+        # ---- extended families (reference observability/metrics.rs groups) --
+        self.tpot = Histogram(
+            "smg_router_tpot_seconds", "time per output token", buckets=_ROUTING_BUCKETS, registry=r
+        )
+        self.request_input_tokens = Histogram(
+            "smg_router_input_tokens", "prompt tokens per request",
+            buckets=(16, 64, 256, 1024, 4096, 16384, 65536), registry=r,
+        )
+        self.request_output_tokens = Histogram(
+            "smg_router_output_tokens", "output tokens per request",
+            buckets=(1, 8, 32, 128, 512, 2048, 8192), registry=r,
+        )
+        # PD disaggregation (metrics.rs:207-227)
+        self.pd_prefill_duration = Histogram(
+            "smg_pd_prefill_duration_seconds", "prefill-leg duration", buckets=_LATENCY_BUCKETS, registry=r
+        )
+        self.pd_bootstrap_failures = Counter(
+            "smg_pd_bootstrap_failures_total", "PD bootstrap injection failures", registry=r
+        )
+        self.pd_dual_dispatch = Counter(
+            "smg_pd_dual_dispatch_total", "prefill+decode dual dispatches", registry=r
+        )
+        # circuit breaker (metrics.rs:272-288)
+        self.cb_state = Gauge(
+            "smg_worker_circuit_breaker_state", "0=closed 1=half-open 2=open", ["worker"], registry=r
+        )
+        self.cb_transitions = Counter(
+            "smg_worker_circuit_breaker_transitions_total", "state transitions", ["worker", "to"], registry=r
+        )
+        # scheduler (middleware/scheduler)
+        self.scheduler_admitted = Counter(
+            "smg_scheduler_admitted_total", "requests admitted", ["klass"], registry=r
+        )
+        self.scheduler_preempted = Counter(
+            "smg_scheduler_preempted_total", "requests preempted", ["klass"], registry=r
+        )
+        self.scheduler_timeout = Counter(
+            "smg_scheduler_queue_timeout_total", "admission queue timeouts", ["klass"], registry=r
+        )
+        # mesh (mesh/src/metrics.rs)
+        self.mesh_gossip_rounds = Counter("smg_mesh_gossip_rounds_total", "gossip rounds", registry=r)
+        self.mesh_ops_applied = Counter("smg_mesh_ops_applied_total", "CRDT ops applied", ["ns"], registry=r)
+        self.mesh_peers = Gauge("smg_mesh_alive_peers", "alive mesh peers", registry=r)
+        # tokenizer caches (tokenizer/src/cache)
+        self.tokenizer_l0_hits = Counter("smg_tokenizer_l0_hits_total", "L0 exact-match hits", registry=r)
+        self.tokenizer_l1_hits = Counter("smg_tokenizer_l1_hits_total", "L1 prefix hits", registry=r)
+        # MCP / plugins
+        self.mcp_tool_calls = Counter("smg_mcp_tool_calls_total", "MCP tool invocations", ["server"], registry=r)
+        self.plugin_short_circuits = Counter(
+            "smg_plugin_short_circuits_total", "plugin OnRequest short-circuits", registry=r
+        )
+        # KV-event index (worker/kv_event_monitor.rs)
+        self.kv_events_applied = Counter(
+            "smg_kv_events_applied_total", "KV cache events applied", ["kind"], registry=r
+        )
+        # runtime self-observability (observability/runtime_metrics.rs canary)
+        self.event_loop_lag = Histogram(
+            "smg_event_loop_lag_seconds", "asyncio scheduling lag (canary)",
+            buckets=(0.0005, 0.001, 0.005, 0.01, 0.05, 0.1, 0.5), registry=r,
+        )
+        # inflight age buckets (inflight_tracker.rs:22)
+        self.inflight_age = Gauge(
+            "smg_inflight_request_age_bucket", "in-flight requests older than the bucket", ["ge_seconds"],
+            registry=r,
+        )
 
     @classmethod
     def null(cls) -> "GatewayMetrics":

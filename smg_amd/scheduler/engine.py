@@ -190,12 +190,16 @@ class PriorityScheduler:
         klass = self.classify(request.get("tenant_id"), request.headers.get("x-smg-priority"))
         ok = await self.acquire(klass)
         if not ok:
+            if self.metrics is not None and not self.metrics._null:
+                self.metrics.scheduler_timeout.labels(klass).inc()
             return web.Response(
                 status=429,
                 body=error_body(f"admission rejected (class={klass})", 429, "rate_limit_error"),
                 content_type="application/json",
                 headers={"x-smg-class": klass},
             )
+        if self.metrics is not None and not self.metrics._null:
+            self.metrics.scheduler_admitted.labels(klass).inc()
         try:
             return await handler(request)
         finally:

@@ -680,6 +680,19 @@ async def startup(config: RouterConfig, serve: bool = True) -> AppContext:
             await start_discovery(ctx.worker_registry, source, circuit_breaker_config=config.circuit_breaker)
         )
     await ctx.start_background()
+    # runtime self-observability: asyncio loop-lag canary (reference
+    # observability/runtime_metrics.rs — tokio event-loop canary + sampler)
+    if not ctx.metrics._null:
+
+        async def _loop_canary(interval: float = 0.5):
+            loop = asyncio.get_event_loop()
+            while True:
+                t0 = loop.time()
+                await asyncio.sleep(interval)
+                lag = max(0.0, loop.time() - t0 - interval)
+                ctx.metrics.event_loop_lag.observe(lag)
+
+        ctx._background.append(asyncio.ensure_future(_loop_canary()))
     if serve and config.health_check_port:
         probe_app = web.Application()
         probe_app[CTX_KEY] = ctx
