@@ -75,3 +75,47 @@ def test_engine_metrics_endpoint(runner):
             await stop_all(client, engines)
 
     runner(run())
+
+
+def test_extended_metric_families_registered():
+    from smg_amd.observability.metrics import GatewayMetrics
+
+    m = GatewayMetrics()
+    if m._null:
+        return  # prometheus_client absent (not this image)
+    text = m.export().decode()
+    for family in (
+        "smg_router_tpot_seconds",
+        "smg_pd_prefill_duration_seconds",
+        "smg_worker_circuit_breaker_state",
+        "smg_scheduler_admitted_total",
+        "smg_mesh_gossip_rounds_total",
+        "smg_tokenizer_l0_hits_total",
+        "smg_kv_events_applied_total",
+        "smg_event_loop_lag_seconds",
+        "smg_inflight_request_age_bucket",
+    ):
+        assert family in text, family
+
+
+def test_event_loop_canary_observes(runner):
+    from smg_amd.config import PolicyConfig, RouterConfig
+    from smg_amd.server.app import startup
+
+    async def run():
+        cfg = RouterConfig(policy=PolicyConfig(name="round_robin", gpu_tree=False))
+        cfg.health_check.disable = True
+        ctx = await startup(cfg, serve=False)
+        try:
+            if ctx.metrics._null:
+                return
+            await asyncio.sleep(1.2)  # > 2 canary intervals
+            text = ctx.metrics.export().decode()
+            import re
+
+            m = re.search(r"smg_event_loop_lag_seconds_count (\d+)", text)
+            assert m and int(m.group(1)) >= 1
+        finally:
+            await ctx.shutdown()
+
+    runner(run())
