@@ -51,6 +51,8 @@ def parse_args():
     p.add_argument("--suffix-len", type=int, default=64)
     p.add_argument("--max-new", type=int, default=32)
     p.add_argument("--tiny", action="store_true", help="tiny model (CPU smoke / CI)")
+    p.add_argument("--kv-fp8", action="store_true",
+                   help="opt-in fp8 (e4m3) KV cache; headline default stays bf16")
     p.add_argument("--no-graphs", action="store_true", help="disable hipGraph decode capture")
     p.add_argument("--decode-burst", type=int, default=2, help="decode iterations per tick")
     p.add_argument("--seed", type=int, default=1234)
@@ -85,6 +87,7 @@ def engine_config(args) -> TorchEngineConfig:
         cfg.max_seq = 256
         return cfg
     cfg = TorchEngineConfig.bench_1b()
+    cfg.kv_fp8 = bool(getattr(args, "kv_fp8", False))
     cfg.max_slots = args.concurrency + 8
     cfg.max_seq = args.prefix_len + args.suffix_len + args.max_new + 16
     return cfg
@@ -267,7 +270,7 @@ def gateway_main(rank: int, world: int, args, device: str, backend: str):
         "higher_is_better": True,
         "scaling": "weak",
         "vs_baseline": round(req_s / BASELINE_REQ_S, 2),
-        "dtype": "bf16" if use_gpu else "fp32",
+        "dtype": ("bf16 (fp8 kv-cache)" if getattr(args, "kv_fp8", False) else "bf16") if use_gpu else "fp32",
         "data": "synthetic",
         "config": {
             "model": "smg-bench-1b (16L d2048 h16, bf16, random-init)" if not args.tiny else "tiny",

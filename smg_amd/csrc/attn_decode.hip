@@ -35,10 +35,23 @@ __device__ __forceinline__ float bf16_to_f32(unsigned short u) {
     return c.f;
 }
 
-extern "C" __global__ void __launch_bounds__(WAVE) smg_attn_decode(
+__device__ __forceinline__ void fp8x4_to_f32(unsigned int w, float* o) {
+    // byte-select must be a literal for v_cvt_f32_fp8
+    o[0] = __builtin_amdgcn_cvt_f32_fp8(w, 0);
+    o[1] = __builtin_amdgcn_cvt_f32_fp8(w, 1);
+    o[2] = __builtin_amdgcn_cvt_f32_fp8(w, 2);
+    o[3] = __builtin_amdgcn_cvt_f32_fp8(w, 3);
+}
+
+// fp8 (OCP e4m3) KV-cache variant support: gfx950 has native fp8<->f32
+// converts (v_cvt_f32_fp8); the cache is stored quantized by the fused
+// rope/KV-store kernels and dequantized inline here — halving the HBM bytes
+// the decode-attention streams per step.
+template <bool KV8>
+__global__ void __launch_bounds__(WAVE) smg_attn_decode_t(
     const __hip_bfloat16* __restrict__ q,
-    const __hip_bfloat16* __restrict__ k,
-    const __hip_bfloat16* __restrict__ v,
+    const void* __restrict__ k,
+    const void* __restrict__ v,
     const int* __restrict__ pos,
     __hip_bfloat16* __restrict__ out,
     int n_slots, int n_heads, int max_seq, int head_dim, float scale) {
@@ -51,8 +64,10 @@ extern "C" __global__ void __launch_bounds__(WAVE) smg_attn_decode(
     if (T > max_seq) T = max_seq;
 
     const size_t head_base = ((size_t)slot * n_heads + head) * (size_t)max_seq * head_dim;
-    const __hip_bfloat16* kh = k + head_base;
-    const __hip_bfloat16* vh = v + head_base;
+    const __hip_bfloat16* kh = (const __hip_bfloat16*)k + head_base;
+    const __hip_bfloat16* vh = (const __hip_bfloat16*)v + head_base;
+    const unsigned char* kh8 = (const unsigned char*)k + head_base;
+    const unsigned char* vh8 = (const unsigned char*)v + head_base;
     const __hip_bfloat16* qh = q + ((size_t)slot * n_heads + head) * head_dim;
 
     // q staged once in LDS as f32 (read by every lane's dot)
@@ -82,15 +97,32 @@ extern "C" __global__ void __launch_bounds__(WAVE) smg_attn_decode(
         int t = base + lane;
         float score = -1e30f;
         if (t < T) {
-            const uint4* row = (const uint4*)(kh + (size_t)t * head_dim);
             float d = 0.f;
+            if constexpr (KV8) {
+                const uint4* row = (const uint4*)(kh8 + (size_t)t * head_dim);
 #pragma unroll 4
-            for (int c = 0; c < vec_n; ++c) {
-                uint4 w = row[c];  // 8 bf16
-                const unsigned short* hsp = (const unsigned short*)&w;
-                int ib = c * 8;
+                for (int c = 0; c < vec_n / 2; ++c) {
+                    uint4 w = row[c];  // 16 fp8
+                    const unsigned int* wp = (const unsigned int*)&w;
+                    int ib = c * 16;
+                    float dec[4];
 #pragma unroll
-                for (int j = 0; j < 8; ++j) d += s_q[ib + j] * bf16_to_f32(hsp[j]);
+                    for (int wi = 0; wi < 4; ++wi) {
+                        fp8x4_to_f32(wp[wi], dec);
+#pragma unroll
+                        for (int s = 0; s < 4; ++s) d += s_q[ib + wi * 4 + s] * dec[s];
+                    }
+                }
+            } else {
+                const uint4* row = (const uint4*)(kh + (size_t)t * head_dim);
+#pragma unroll 4
+                for (int c = 0; c < vec_n; ++c) {
+                    uint4 w = row[c];  // 8 bf16
+                    const unsigned short* hsp = (const unsigned short*)&w;
+                    int ib = c * 8;
+#pragma unroll
+                    for (int j = 0; j < 8; ++j) d += s_q[ib + j] * bf16_to_f32(hsp[j]);
+                }
             }
             score = d * scale;
         }
@@ -113,24 +145,55 @@ extern "C" __global__ void __launch_bounds__(WAVE) smg_attn_decode(
         // row.  Full tiles take the unrolled fast path (compile-time trip
         // count exposes all 16 loads to the scheduler); only the final
         // partial tile pays the runtime loop.
-        const __hip_bfloat16* vtile = vh + (size_t)base * head_dim + chunk * 8;
-        if (lim == WAVE && head_dim == 128) {
+        if constexpr (KV8) {
+            const unsigned char* vtile8 = vh8 + (size_t)base * head_dim + chunk * 8;
+            if (lim == WAVE && head_dim == 128) {
 #pragma unroll
-            for (int it = 0; it < 16; ++it) {
-                const int j = rgrp + it * 4;  // rows_per == 4 when chunks == 16
-                const uint4 w = *(const uint4*)(vtile + (size_t)j * head_dim);
-                const unsigned short* hsp = (const unsigned short*)&w;
-                float pj = s_p[j];
+                for (int it = 0; it < 16; ++it) {
+                    const int j = rgrp + it * 4;
+                    const uint2 w = *(const uint2*)(vtile8 + (size_t)j * head_dim);
+                    float pj = s_p[j];
+                    float dec[4];
+                    fp8x4_to_f32(w.x, dec);
 #pragma unroll
-                for (int jj = 0; jj < 8; ++jj) accv[jj] += pj * bf16_to_f32(hsp[jj]);
+                    for (int s = 0; s < 4; ++s) accv[s] += pj * dec[s];
+                    fp8x4_to_f32(w.y, dec);
+#pragma unroll
+                    for (int s = 0; s < 4; ++s) accv[4 + s] += pj * dec[s];
+                }
+            } else {
+                for (int j = rgrp; j < lim; j += rows_per) {
+                    const uint2 w = *(const uint2*)(vtile8 + (size_t)j * head_dim);
+                    float pj = s_p[j];
+                    float dec[4];
+                    fp8x4_to_f32(w.x, dec);
+#pragma unroll
+                    for (int s = 0; s < 4; ++s) accv[s] += pj * dec[s];
+                    fp8x4_to_f32(w.y, dec);
+#pragma unroll
+                    for (int s = 0; s < 4; ++s) accv[4 + s] += pj * dec[s];
+                }
             }
         } else {
-            for (int j = rgrp; j < lim; j += rows_per) {
-                const uint4 w = *(const uint4*)(vtile + (size_t)j * head_dim);
-                const unsigned short* hsp = (const unsigned short*)&w;
-                float pj = s_p[j];
+            const __hip_bfloat16* vtile = vh + (size_t)base * head_dim + chunk * 8;
+            if (lim == WAVE && head_dim == 128) {
 #pragma unroll
-                for (int jj = 0; jj < 8; ++jj) accv[jj] += pj * bf16_to_f32(hsp[jj]);
+                for (int it = 0; it < 16; ++it) {
+                    const int j = rgrp + it * 4;  // rows_per == 4 when chunks == 16
+                    const uint4 w = *(const uint4*)(vtile + (size_t)j * head_dim);
+                    const unsigned short* hsp = (const unsigned short*)&w;
+                    float pj = s_p[j];
+#pragma unroll
+                    for (int jj = 0; jj < 8; ++jj) accv[jj] += pj * bf16_to_f32(hsp[jj]);
+                }
+            } else {
+                for (int j = rgrp; j < lim; j += rows_per) {
+                    const uint4 w = *(const uint4*)(vtile + (size_t)j * head_dim);
+                    const unsigned short* hsp = (const unsigned short*)&w;
+                    float pj = s_p[j];
+#pragma unroll
+                    for (int jj = 0; jj < 8; ++jj) accv[jj] += pj * bf16_to_f32(hsp[jj]);
+                }
             }
         }
         __syncthreads();
@@ -149,16 +212,30 @@ extern "C" __global__ void __launch_bounds__(WAVE) smg_attn_decode(
     }
 }
 
-extern "C" int smg_attn_decode_launch(const void* q, const void* k, const void* v,
-                                      const void* pos, void* out, int n_slots, int n_heads,
-                                      int max_seq, int head_dim, float scale, void* stream) {
+extern "C" int smg_attn_decode_launch_ex(const void* q, const void* k, const void* v,
+                                         const void* pos, void* out, int n_slots, int n_heads,
+                                         int max_seq, int head_dim, float scale, void* stream,
+                                         int kv_fp8) {
     if (head_dim > 128 || (head_dim & 7)) return -1;
     int chunks = head_dim / 8;  // v6 P·V row-group mapping needs 2^k chunks
     if (chunks & (chunks - 1)) return -1;
+    if (kv_fp8 && (head_dim & 15)) return -1;  // fp8 K streaming is 16-wide
     dim3 grid(n_slots * n_heads);
-    hipLaunchKernelGGL(smg_attn_decode, grid, dim3(WAVE), 0, (hipStream_t)stream,
-                       (const __hip_bfloat16*)q, (const __hip_bfloat16*)k,
-                       (const __hip_bfloat16*)v, (const int*)pos, (__hip_bfloat16*)out,
-                       n_slots, n_heads, max_seq, head_dim, scale);
+    if (kv_fp8) {
+        hipLaunchKernelGGL(smg_attn_decode_t<true>, grid, dim3(WAVE), 0, (hipStream_t)stream,
+                           (const __hip_bfloat16*)q, k, v, (const int*)pos,
+                           (__hip_bfloat16*)out, n_slots, n_heads, max_seq, head_dim, scale);
+    } else {
+        hipLaunchKernelGGL(smg_attn_decode_t<false>, grid, dim3(WAVE), 0, (hipStream_t)stream,
+                           (const __hip_bfloat16*)q, k, v, (const int*)pos,
+                           (__hip_bfloat16*)out, n_slots, n_heads, max_seq, head_dim, scale);
+    }
     return hipGetLastError() == hipSuccess ? 0 : -2;
+}
+
+extern "C" int smg_attn_decode_launch(const void* q, const void* k, const void* v,
+                                      const void* pos, void* out, int n_slots, int n_heads,
+                                      int max_seq, int head_dim, float scale, void* stream) {
+    return smg_attn_decode_launch_ex(q, k, v, pos, out, n_slots, n_heads, max_seq, head_dim,
+                                     scale, stream, 0);
 }

@@ -29,12 +29,21 @@
 
 #define WAVE 64
 
-extern "C" __global__ void __launch_bounds__(WAVE) smg_rope_kv_store(
+// KV8: store the cache as OCP e4m3 fp8 (native v_cvt_pk_fp8_f32 pack of each
+// rotated pair) — the decode-attention kernel then streams half the bytes.
+__device__ __forceinline__ void store_pair_fp8(unsigned char* dst, float a, float b) {
+    unsigned int packed = 0;
+    packed = __builtin_amdgcn_cvt_pk_fp8_f32(a, b, packed, false);
+    *(unsigned short*)dst = (unsigned short)(packed & 0xFFFFu);
+}
+
+template <bool KV8>
+__global__ void __launch_bounds__(WAVE) smg_rope_kv_store_t(
     const __hip_bfloat16* __restrict__ qkv,
     const float2* __restrict__ freqs,
     const int* __restrict__ pos,
-    __hip_bfloat16* __restrict__ k_cache,
-    __hip_bfloat16* __restrict__ v_cache,
+    void* __restrict__ k_cache,
+    void* __restrict__ v_cache,
     __hip_bfloat16* __restrict__ q_out,
     int n_slots, int n_heads, int max_seq, int head_dim) {
     const int sh = blockIdx.x;
@@ -51,8 +60,11 @@ extern "C" __global__ void __launch_bounds__(WAVE) smg_rope_kv_store(
     const __hip_bfloat16* vrow = qrow + 2 * D;
     const float2* f = freqs + (size_t)p * pairs;
 
-    __hip_bfloat16* kdst = k_cache + (((size_t)slot * n_heads + head) * max_seq + p) * head_dim;
-    __hip_bfloat16* vdst = v_cache + (((size_t)slot * n_heads + head) * max_seq + p) * head_dim;
+    const size_t row_off = ((size_t)slot * n_heads + head) * max_seq + p;
+    __hip_bfloat16* kdst = (__hip_bfloat16*)k_cache + row_off * head_dim;
+    __hip_bfloat16* vdst = (__hip_bfloat16*)v_cache + row_off * head_dim;
+    unsigned char* kdst8 = (unsigned char*)k_cache + row_off * head_dim;
+    unsigned char* vdst8 = (unsigned char*)v_cache + row_off * head_dim;
     __hip_bfloat16* qdst = q_out + ((size_t)slot * n_heads + head) * head_dim;
 
     for (int i = lane; i < pairs; i += WAVE) {
@@ -64,11 +76,19 @@ extern "C" __global__ void __launch_bounds__(WAVE) smg_rope_kv_store(
         qdst[e + 1] = __float2bfloat16(q0 * cs.y + q1 * cs.x);
         const float k0 = __bfloat162float(krow[e]);
         const float k1 = __bfloat162float(krow[e + 1]);
-        kdst[e] = __float2bfloat16(k0 * cs.x - k1 * cs.y);
-        kdst[e + 1] = __float2bfloat16(k0 * cs.y + k1 * cs.x);
-        // v passes through untouched (pair copy keeps the access pattern)
-        vdst[e] = vrow[e];
-        vdst[e + 1] = vrow[e + 1];
+        const float kr0 = k0 * cs.x - k1 * cs.y;
+        const float kr1 = k0 * cs.y + k1 * cs.x;
+        const float v0 = __bfloat162float(vrow[e]);
+        const float v1 = __bfloat162float(vrow[e + 1]);
+        if constexpr (KV8) {
+            store_pair_fp8(kdst8 + e, kr0, kr1);
+            store_pair_fp8(vdst8 + e, v0, v1);
+        } else {
+            kdst[e] = __float2bfloat16(kr0);
+            kdst[e + 1] = __float2bfloat16(kr1);
+            vdst[e] = vrow[e];
+            vdst[e + 1] = vrow[e + 1];
+        }
     }
 }
 
@@ -94,14 +114,15 @@ extern "C" __global__ void __launch_bounds__(256) smg_silu_mul(
 // layout sdpa wants (so the flash prefill consumes them with zero transpose
 // copies).  Replaces the per-layer torch chain: 2x rope complex-mul chains,
 // 2x advanced-index KV scatter, and the implicit transpose-contiguous copies.
-extern "C" __global__ void __launch_bounds__(WAVE) smg_rope_prefill(
+template <bool KV8>
+__global__ void __launch_bounds__(WAVE) smg_rope_prefill_t(
     const __hip_bfloat16* __restrict__ qkv,   // [B, L, 3*D]
     const float2* __restrict__ freqs,         // [max_seq, hd/2]
     const int* __restrict__ slots,            // [B]
     const int* __restrict__ starts,           // [B]
-    __hip_bfloat16* __restrict__ k_cache,     // [n_slots, H, max_seq, hd]
-    __hip_bfloat16* __restrict__ v_cache,
-    __hip_bfloat16* __restrict__ q_out,       // [B, H, L, hd]
+    void* __restrict__ k_cache,               // [n_slots, H, max_seq, hd]
+    void* __restrict__ v_cache,
+    __hip_bfloat16* __restrict__ q_out,       // [B, H, L, hd] (always bf16)
     __hip_bfloat16* __restrict__ k_out,
     __hip_bfloat16* __restrict__ v_out,
     int B, int L, int n_heads, int max_seq, int head_dim) {
@@ -122,8 +143,11 @@ extern "C" __global__ void __launch_bounds__(WAVE) smg_rope_prefill(
     const __hip_bfloat16* vrow = qrow + 2 * D;
     const float2* f = freqs + (size_t)p * pairs;
 
-    __hip_bfloat16* kc = k_cache + (((size_t)slot * n_heads + h) * max_seq + p) * head_dim;
-    __hip_bfloat16* vc = v_cache + (((size_t)slot * n_heads + h) * max_seq + p) * head_dim;
+    const size_t crow = ((size_t)slot * n_heads + h) * max_seq + p;
+    __hip_bfloat16* kc = (__hip_bfloat16*)k_cache + crow * head_dim;
+    __hip_bfloat16* vc = (__hip_bfloat16*)v_cache + crow * head_dim;
+    unsigned char* kc8 = (unsigned char*)k_cache + crow * head_dim;
+    unsigned char* vc8 = (unsigned char*)v_cache + crow * head_dim;
     const size_t out_off = (((size_t)b * n_heads + h) * L + t) * head_dim;
     __hip_bfloat16* qo = q_out + out_off;
     __hip_bfloat16* ko = k_out + out_off;
@@ -138,32 +162,75 @@ extern "C" __global__ void __launch_bounds__(WAVE) smg_rope_prefill(
         qo[e + 1] = __float2bfloat16(q0 * cs.y + q1 * cs.x);
         const float k0 = __bfloat162float(krow[e]);
         const float k1 = __bfloat162float(krow[e + 1]);
-        const __hip_bfloat16 kr0 = __float2bfloat16(k0 * cs.x - k1 * cs.y);
-        const __hip_bfloat16 kr1 = __float2bfloat16(k0 * cs.y + k1 * cs.x);
+        const float krf0 = k0 * cs.x - k1 * cs.y;
+        const float krf1 = k0 * cs.y + k1 * cs.x;
+        const __hip_bfloat16 kr0 = __float2bfloat16(krf0);
+        const __hip_bfloat16 kr1 = __float2bfloat16(krf1);
         ko[e] = kr0;
         ko[e + 1] = kr1;
-        kc[e] = kr0;
-        kc[e + 1] = kr1;
         const __hip_bfloat16 v0 = vrow[e];
         const __hip_bfloat16 v1 = vrow[e + 1];
         vo[e] = v0;
         vo[e + 1] = v1;
-        vc[e] = v0;
-        vc[e + 1] = v1;
+        if constexpr (KV8) {
+            store_pair_fp8(kc8 + e, krf0, krf1);
+            store_pair_fp8(vc8 + e, __bfloat162float(v0), __bfloat162float(v1));
+        } else {
+            kc[e] = kr0;
+            kc[e + 1] = kr1;
+            vc[e] = v0;
+            vc[e + 1] = v1;
+        }
     }
+}
+
+extern "C" int smg_rope_prefill_launch_ex(
+    const void* qkv, const void* freqs, const void* slots, const void* starts,
+    void* k_cache, void* v_cache, void* q_out, void* k_out, void* v_out,
+    int B, int L, int n_heads, int max_seq, int head_dim, void* stream, int kv_fp8) {
+    if (head_dim % 2 != 0 || head_dim > 256) return 1;
+    dim3 grid((unsigned)B * L * n_heads);
+    if (kv_fp8) {
+        hipLaunchKernelGGL(smg_rope_prefill_t<true>, grid, dim3(WAVE), 0, (hipStream_t)stream,
+                           (const __hip_bfloat16*)qkv, (const float2*)freqs, (const int*)slots,
+                           (const int*)starts, k_cache, v_cache, (__hip_bfloat16*)q_out,
+                           (__hip_bfloat16*)k_out, (__hip_bfloat16*)v_out,
+                           B, L, n_heads, max_seq, head_dim);
+    } else {
+        hipLaunchKernelGGL(smg_rope_prefill_t<false>, grid, dim3(WAVE), 0, (hipStream_t)stream,
+                           (const __hip_bfloat16*)qkv, (const float2*)freqs, (const int*)slots,
+                           (const int*)starts, k_cache, v_cache, (__hip_bfloat16*)q_out,
+                           (__hip_bfloat16*)k_out, (__hip_bfloat16*)v_out,
+                           B, L, n_heads, max_seq, head_dim);
+    }
+    return (int)hipGetLastError();
 }
 
 extern "C" int smg_rope_prefill_launch(
     const void* qkv, const void* freqs, const void* slots, const void* starts,
     void* k_cache, void* v_cache, void* q_out, void* k_out, void* v_out,
     int B, int L, int n_heads, int max_seq, int head_dim, void* stream) {
+    return smg_rope_prefill_launch_ex(qkv, freqs, slots, starts, k_cache, v_cache, q_out,
+                                      k_out, v_out, B, L, n_heads, max_seq, head_dim, stream, 0);
+}
+
+extern "C" int smg_rope_kv_store_launch_ex(
+    const void* qkv, const void* freqs, const void* pos,
+    void* k_cache, void* v_cache, void* q_out,
+    int n_slots, int n_heads, int max_seq, int head_dim, void* stream, int kv_fp8) {
     if (head_dim % 2 != 0 || head_dim > 256) return 1;
-    dim3 grid((unsigned)B * L * n_heads);
-    hipLaunchKernelGGL(smg_rope_prefill, grid, dim3(WAVE), 0, (hipStream_t)stream,
-                       (const __hip_bfloat16*)qkv, (const float2*)freqs, (const int*)slots,
-                       (const int*)starts, (__hip_bfloat16*)k_cache, (__hip_bfloat16*)v_cache,
-                       (__hip_bfloat16*)q_out, (__hip_bfloat16*)k_out, (__hip_bfloat16*)v_out,
-                       B, L, n_heads, max_seq, head_dim);
+    dim3 grid(n_slots * n_heads);
+    if (kv_fp8) {
+        hipLaunchKernelGGL(smg_rope_kv_store_t<true>, grid, dim3(WAVE), 0, (hipStream_t)stream,
+                           (const __hip_bfloat16*)qkv, (const float2*)freqs, (const int*)pos,
+                           k_cache, v_cache, (__hip_bfloat16*)q_out,
+                           n_slots, n_heads, max_seq, head_dim);
+    } else {
+        hipLaunchKernelGGL(smg_rope_kv_store_t<false>, grid, dim3(WAVE), 0, (hipStream_t)stream,
+                           (const __hip_bfloat16*)qkv, (const float2*)freqs, (const int*)pos,
+                           k_cache, v_cache, (__hip_bfloat16*)q_out,
+                           n_slots, n_heads, max_seq, head_dim);
+    }
     return (int)hipGetLastError();
 }
 
@@ -171,13 +238,8 @@ extern "C" int smg_rope_kv_store_launch(
     const void* qkv, const void* freqs, const void* pos,
     void* k_cache, void* v_cache, void* q_out,
     int n_slots, int n_heads, int max_seq, int head_dim, void* stream) {
-    if (head_dim % 2 != 0 || head_dim > 256) return 1;
-    dim3 grid(n_slots * n_heads);
-    hipLaunchKernelGGL(smg_rope_kv_store, grid, dim3(WAVE), 0, (hipStream_t)stream,
-                       (const __hip_bfloat16*)qkv, (const float2*)freqs, (const int*)pos,
-                       (__hip_bfloat16*)k_cache, (__hip_bfloat16*)v_cache,
-                       (__hip_bfloat16*)q_out, n_slots, n_heads, max_seq, head_dim);
-    return (int)hipGetLastError();
+    return smg_rope_kv_store_launch_ex(qkv, freqs, pos, k_cache, v_cache, q_out,
+                                       n_slots, n_heads, max_seq, head_dim, stream, 0);
 }
 
 extern "C" int smg_silu_mul_launch(const void* gu, void* out, long long rows,

@@ -40,6 +40,11 @@ class TorchEngineConfig:
     prefix_cache_slots: int = 16
     prefix_cache_page: int = 64
     prefix_cache_max: int = 1024  # max cached prefix tokens
+    # opt-in fp8 (OCP e4m3) KV cache: halves the HBM bytes decode attention
+    # streams; compute stays bf16 (only the cache storage is quantized).
+    # Requires the gfx950 fused kernels.  Default OFF — the bench headline
+    # is measured with the bf16 cache.
+    kv_fp8: bool = False
 
     @classmethod
     def tiny(cls) -> "TorchEngineConfig":
@@ -109,9 +114,12 @@ class TorchEngine:
             self.layers = [_Layer(c, self.device, self.dtype, gen) for _ in range(c.n_layers)]
             self.ln_f = torch.ones(c.d_model, device=self.device, dtype=self.dtype)
             # KV cache: [layers, 2, slots, heads, max_seq, head_dim]
+            self.kv_dtype = (
+                torch.float8_e4m3fn if (c.kv_fp8 and self.device.type == "cuda") else self.dtype
+            )
             self.kv = torch.zeros(
                 c.n_layers, 2, c.max_slots, c.n_heads, c.max_seq, c.head_dim,
-                device=self.device, dtype=self.dtype,
+                device=self.device, dtype=self.kv_dtype,
             )
             # rotary tables (complex form: one complex mul applies the rotation)
             inv = 1.0 / (10000.0 ** (torch.arange(0, c.head_dim, 2, device=self.device).float() / c.head_dim))
@@ -138,7 +146,7 @@ class TorchEngine:
             self._pc_arena = torch.zeros(
                 pc.n_layers, 2, pc.prefix_cache_slots, pc.n_heads,
                 min(pc.prefix_cache_max, pc.max_seq), pc.head_dim,
-                device=self.device, dtype=self.dtype,
+                device=self.device, dtype=self.kv.dtype,
             )
         else:
             self._pc_arena = None
@@ -177,6 +185,12 @@ class TorchEngine:
             except ImportError:
                 pass
         self._flash_lse = self._probe_flash_lse() if self.device.type == "cuda" else None
+        if c.kv_fp8 and (self._hip_fused is None or self._hip_attn is None or
+                         getattr(self, "_hip_rope_prefill", None) is None):
+            raise RuntimeError(
+                "kv_fp8=True requires the gfx950 fused kernels (smg_amd._core with"
+                " rope_kv_store/rope_prefill/attn_decode) on a CUDA/HIP device"
+            )
 
     # ---- API -------------------------------------------------------------
     def submit(self, tokens: List[int], max_new_tokens: int, rid: Optional[str] = None) -> str:
@@ -293,6 +307,14 @@ class TorchEngine:
         return produced
 
     # ---- forwards ----------------------------------------------------------
+    def _kv_hist(self, li: int, which: int, slots):
+        """Per-request KV history gather for the prefill attention; an fp8
+        cache is dequantized to bf16 here (sdpa/flash consume bf16)."""
+        sel = self.kv[li, which].index_select(0, slots)
+        if sel.dtype == torch.float8_e4m3fn:
+            sel = sel.to(self.dtype)
+        return sel
+
     def _probe_flash_lse(self):
         """(q,k,v,is_causal) -> (out, logsumexp) via the flash kernel, or None
         when the aten op is unavailable.  The lse output lets the suffix-
@@ -432,6 +454,7 @@ class TorchEngine:
                     self.kv[li, 0].data_ptr(), self.kv[li, 1].data_ptr(),
                     qb.data_ptr(), kb.data_ptr(), vb.data_ptr(),
                     B, L, c.n_heads, c.max_seq, c.head_dim, stream,
+                    1 if self.kv.dtype == torch.float8_e4m3fn else 0,
                 )
                 q, k, v = qb, kb, vb
             else:
@@ -449,16 +472,16 @@ class TorchEngine:
                 # Replaces the masked-sdpa math path (bmm + 37 MB mask add +
                 # softmax per layer).
                 start0 = int(starts[0].item())
-                kk = self.kv[li, 0].index_select(0, slots)[:, :, :start0]
-                vv = self.kv[li, 1].index_select(0, slots)[:, :, :start0]
+                kk = self._kv_hist(li, 0, slots)[:, :, :start0]
+                vv = self._kv_hist(li, 1, slots)[:, :, :start0]
                 o1, lse1 = self._flash_lse(q, kk, vv, False)
                 o2, lse2 = self._flash_lse(q, k, v, True)
                 lse_tot = torch.logaddexp(lse1, lse2)
                 attn = o1 * (lse1 - lse_tot).exp().unsqueeze(-1) + o2 * (lse2 - lse_tot).exp().unsqueeze(-1)
                 attn = attn.to(q.dtype)
             else:
-                kk = self.kv[li, 0].index_select(0, slots)[:, :, :t_max]
-                vv = self.kv[li, 1].index_select(0, slots)[:, :, :t_max]
+                kk = self._kv_hist(li, 0, slots)[:, :, :t_max]
+                vv = self._kv_hist(li, 1, slots)[:, :, :t_max]
                 attn = F.scaled_dot_product_attention(q, kk, vv, attn_mask=mask)
             attn2 = attn.transpose(1, 2).reshape(B * L, c.d_model)
             h = torch.addmm(h.view(B * L, c.d_model), attn2, layer.wo).view(B, L, c.d_model)
@@ -526,18 +549,19 @@ class TorchEngine:
         stream = torch.cuda.current_stream().cuda_stream
         scale = 1.0 / math.sqrt(c.head_dim)
         h = self.embed[self._last_tok]  # [S, D]
+        kv8 = 1 if self.kv.dtype == torch.float8_e4m3fn else 0
         freqs_ptr = self.freqs_cis.data_ptr()  # complex64 [T, hd/2] == float2
         for li, layer in enumerate(self.layers):
             qkv = _rms(h, layer.ln1) @ layer.wqkv  # [S, 3D]
             self._hip_fused(
                 qkv.data_ptr(), freqs_ptr, self._pos_i32.data_ptr(),
                 self.kv[li, 0].data_ptr(), self.kv[li, 1].data_ptr(), self._q_buf.data_ptr(),
-                S, c.n_heads, c.max_seq, c.head_dim, stream,
+                S, c.n_heads, c.max_seq, c.head_dim, stream, kv8,
             )
             self._hip_attn(
                 self._q_buf.data_ptr(), self.kv[li, 0].data_ptr(), self.kv[li, 1].data_ptr(),
                 self._pos_i32.data_ptr(), self._attn_out.data_ptr(),
-                S, c.n_heads, c.max_seq, c.head_dim, scale, stream,
+                S, c.n_heads, c.max_seq, c.head_dim, scale, stream, kv8,
             )
             h = torch.addmm(h, self._attn_out.view(S, c.d_model), layer.wo)
             gu = _rms(h, layer.ln2) @ layer.w13  # [S, 2F]
