@@ -9,7 +9,6 @@ fallback on a GPU box.
 """
 from __future__ import annotations
 
-import time
 from typing import Dict, List, Optional, Sequence
 
 import numpy as np

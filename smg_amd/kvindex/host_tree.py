@@ -3,7 +3,7 @@ smg_amd._core.HostTokenTree.  Tenants are strings at this layer; the C++ core
 works in worker slots 0..63, mapped here."""
 from __future__ import annotations
 
-from typing import Dict, List, Optional, Sequence
+from typing import Dict, Sequence
 
 import numpy as np
 

@@ -14,7 +14,7 @@ import itertools
 import json
 import logging
 from dataclasses import dataclass, field
-from typing import Any, Callable, Dict, List, Optional
+from typing import Callable, Dict, List, Optional
 
 log = logging.getLogger("smg.mcp")
 

@@ -9,7 +9,7 @@ from __future__ import annotations
 import hashlib
 import random as _random
 import time
-from typing import Dict, List, Optional, Sequence
+from typing import Dict, Optional, Sequence
 
 from ..config import PolicyConfig
 from ..workers.worker import Worker

@@ -10,7 +10,6 @@ content_block_delta* -> content_block_stop -> message_delta -> message_stop.
 from __future__ import annotations
 
 import json
-import time
 import uuid
 from typing import Any, AsyncIterator, Dict, List
 

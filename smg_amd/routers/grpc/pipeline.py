@@ -10,12 +10,10 @@ SSE encode (reference regular/streaming.rs).
 """
 from __future__ import annotations
 
-import asyncio
-import json
 import time
 import uuid
 from dataclasses import dataclass, field
-from typing import Any, AsyncIterator, Dict, List, Optional
+from typing import Any, Dict, List, Optional
 
 from ...grpc import api
 from ...grpc.client import ClientPool
@@ -23,7 +21,7 @@ from ...parsers.reasoning import get_reasoning_parser
 from ...parsers.tool import get_parser
 from ...parsers.tool.stream import StreamingToolParser
 from ...policies import SelectWorkerInfo
-from ...protocols.openai import ProtocolError, error_body
+from ...protocols.openai import error_body
 from ...tokenizer.chat_template import ChatTemplate
 from ...tokenizer.stop import DecodeStream, StopSequenceDecoder
 from ..base import RouteRequest, RouteResponse

@@ -15,10 +15,8 @@ import json
 import logging
 import random
 import uuid
-from typing import Optional
 from urllib.parse import urlparse
 
-from ..config import RouterConfig
 from ..policies import SelectWorkerInfo
 from ..protocols.openai import ProtocolError, error_body, parse_request
 from ..workers.worker import WorkerType

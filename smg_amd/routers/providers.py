@@ -17,7 +17,6 @@ from typing import Any, Dict, Optional
 import aiohttp
 
 from ..workers.worker import Worker
-from .anthropic import chat_to_message, messages_to_chat
 
 
 class ProviderAdapter:

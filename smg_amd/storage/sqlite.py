@@ -10,7 +10,6 @@ concurrent readers.  `make_storage("sqlite")` uses an in-memory database;
 """
 from __future__ import annotations
 
-import asyncio
 import json
 import sqlite3
 import threading
