@@ -157,13 +157,18 @@ def build_parser() -> argparse.ArgumentParser:
     g.add_argument("--control-plane-api-keys", nargs="*", default=[])
     g.add_argument("--jwt-issuer", default=None)
     g.add_argument("--jwt-audience", default=None)
-    g.add_argument("--jwt-jwks-uri", default=None)
+    g.add_argument("--jwt-jwks-uri", default=None,
+                   help="http(s) JWKS URI or a local JWKS json file path")
+    g.add_argument("--jwt-role-claim", default="roles")
+    g.add_argument("--jwt-leeway-secs", type=float, default=60.0)
+    g.add_argument("--jwt-enable-jti-check", action="store_true",
+                   help="replay protection via a JTI LRU cache")
     g.add_argument("--disable-audit-logging", action="store_true")
     g.add_argument("--trust-tenant-header", action="store_true")
     g.add_argument("--tenant-header-name", default="x-smg-tenant")
 
     g = p.add_argument_group("Storage")
-    g.add_argument("--backend", default="memory", choices=["memory", "none", "postgres", "redis", "oracle"])
+    g.add_argument("--backend", default="memory", choices=["memory", "none", "sqlite", "postgres", "redis", "oracle"])
     g.add_argument("--history-backend", default=None)
     g.add_argument("--postgres-db-url", default=None)
     g.add_argument("--redis-url", default=None)
@@ -329,6 +334,9 @@ def to_router_config(argv: Optional[List[str]] = None) -> RouterConfig:
     cfg.auth.jwt_issuer = args.jwt_issuer
     cfg.auth.jwt_audience = args.jwt_audience
     cfg.auth.jwt_jwks_uri = args.jwt_jwks_uri
+    cfg.auth.jwt_role_claim = args.jwt_role_claim
+    cfg.auth.jwt_leeway_secs = args.jwt_leeway_secs
+    cfg.auth.jwt_enable_jti_check = args.jwt_enable_jti_check
     cfg.auth.disable_audit_logging = args.disable_audit_logging
     cfg.storage.backend = args.backend
     cfg.storage.history_backend = args.history_backend
