@@ -64,6 +64,7 @@ class HttpRouter(Router):
         self._session = session
         self.metrics = metrics or GatewayMetrics.null()
         self.retry_executor = RetryExecutor(self.config.retry)
+        self._model_retry_cache: dict = {}
 
     async def session(self) -> aiohttp.ClientSession:
         if self._session is None or self._session.closed:
@@ -127,8 +128,22 @@ class HttpRouter(Router):
         def on_retry(attempt_no: int, resp) -> None:
             self.metrics.count_retry(req.path)
 
-        resp = await self.retry_executor.execute(attempt, should_retry, on_retry)
+        resp = await self._retry_for(model_id).execute(attempt, should_retry, on_retry)
         return resp
+
+    def _retry_for(self, model_id):
+        """Per-model retry override (reference registry.rs model_retry_configs
+        consulted by route_typed_request)."""
+        if not model_id:
+            return self.retry_executor
+        cfg = self.registry.get_model_retry_config(model_id)
+        if cfg is None:
+            return self.retry_executor
+        ex = self._model_retry_cache.get(model_id)
+        if ex is None or ex.config is not cfg:
+            ex = RetryExecutor(cfg)
+            self._model_retry_cache[model_id] = ex
+        return ex
 
     # ---- dispatch --------------------------------------------------------
     async def _dispatch(self, worker: Worker, req: RouteRequest, info: SelectWorkerInfo) -> RouteResponse:

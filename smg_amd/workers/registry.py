@@ -28,6 +28,15 @@ class WorkerRegistry:
         self._dirty = False
         self._listeners: List[Callable[[str, Worker], None]] = []
         self._change_event: Optional[asyncio.Event] = None
+        # per-model retry overrides (reference registry.rs model_retry_configs)
+        self._model_retry: Dict[str, object] = {}
+
+    # ---- per-model retry configs ------------------------------------------
+    def set_model_retry_config(self, model_id: str, cfg) -> None:
+        self._model_retry[model_id] = cfg
+
+    def get_model_retry_config(self, model_id):
+        return self._model_retry.get(model_id)
 
     # ---- mutation --------------------------------------------------------
     def register(self, worker: Worker) -> Worker:

@@ -271,3 +271,34 @@ def test_metrics_exported(runner):
             await stop_all(client, engines)
 
     runner(run())
+
+
+def test_per_model_retry_override(runner):
+    """Per-model retry configs (reference registry.rs model_retry_configs):
+    a model with retries disabled fails fast while the default keeps
+    retrying to another worker."""
+    async def run():
+        from smg_amd.config import RetryConfig
+
+        ctx, engines = make_ctx(n_workers=2)
+        client = await start_client(ctx, engines)
+        try:
+            # disable retries for this model: first worker failure surfaces
+            ctx.worker_registry.set_model_retry_config(
+                "mock-model", RetryConfig(max_retries=0)
+            )
+            engines[0].sim.config.failure_rate = 1.0
+            engines[1].sim.config.failure_rate = 1.0
+            resp = await client.post(
+                "/v1/chat/completions", json=CHAT_BODY
+            )
+            assert resp.status >= 500  # no retry -> error escapes
+            # restore default (retries on): one healthy worker now recovers
+            ctx.worker_registry._model_retry.pop("mock-model", None)
+            engines[1].sim.config.failure_rate = 0.0
+            resp = await client.post("/v1/chat/completions", json=CHAT_BODY)
+            assert resp.status == 200
+        finally:
+            await stop_all(client, engines)
+
+    runner(run())
