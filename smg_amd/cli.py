@@ -203,6 +203,8 @@ def build_parser() -> argparse.ArgumentParser:
     g.add_argument("--rccl-tick-us", type=int, default=200)
 
     g = p.add_argument_group("Serve (worker spawn; reference bindings/python serve.py)")
+    g.add_argument("--kv-fp8", action="store_true",
+                   help="local GPU engines store their KV cache as fp8 e4m3 (compute stays bf16)")
     g.add_argument("--local-workers", type=int, default=0,
                    help="serve mode: spawn N in-process workers (GPU TorchEngine when available, mock simulator otherwise)")
     g.add_argument("--local-worker-model", default="local-model")
@@ -400,7 +402,9 @@ def main(argv: Optional[List[str]] = None) -> None:
                     from .engine.torch_engine import TorchEngine, TorchEngineConfig
                     from .grpc.servicer import EngineAdapter
 
-                    eng = TorchEngine(TorchEngineConfig.bench_1b(), device=f"cuda:{i % torch.cuda.device_count()}")
+                    ecfg = TorchEngineConfig.bench_1b()
+                    ecfg.kv_fp8 = bool(getattr(args, "kv_fp8", False))
+                    eng = TorchEngine(ecfg, device=f"cuda:{i % torch.cuda.device_count()}")
                     adapter = EngineAdapter(eng)
                     await adapter.start()
                     engines.append(adapter)
