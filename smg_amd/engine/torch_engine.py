@@ -304,6 +304,14 @@ class TorchEngine:
                 del self.running[s]
                 self._free_slots.append(s)
                 self._seq_len_host[s] = 0
+            if finished:
+                # reset the DEVICE positions too: the full-arena decode reads
+                # seq_len[slot] as each slot's live window — a freed slot left
+                # at its final position would keep streaming its whole stale
+                # KV window every step (in steady state most of the arena is
+                # free, so this is the difference between O(active) and
+                # O(capacity) attention work)
+                self.seq_len[torch.tensor(finished, device=self.device)] = 0
         return produced
 
     # ---- forwards ----------------------------------------------------------
