@@ -381,9 +381,26 @@ class TorchEngine:
         top_key = hash(tuple(tokens[:plen]))
         if top_key in self._pc_keys:
             return  # already cached
+        # block-level dedup (vLLM-style): if the cache already covers all but
+        # the last page or so of this prompt, a new entry would spend a
+        # multi-MB arena copy + an eviction on a one-off suffix — skip it.
+        # (Unchecked, the 500-stores/s churn evicted the hot shared prefixes
+        # and the hit rate decayed over long runs.)
+        existing = 0
+        for p in range(plen, page - 1, -page):
+            if hash(tuple(tokens[:p])) in self._pc_keys:
+                existing = p
+                break
+        if plen - existing < 2 * page:
+            return
         victim = self._pc_lru.pop(0)
         for k in self._pc_slot_keys.pop(victim, []):
-            self._pc_keys.pop(k, None)
+            # only drop keys the victim still owns: a later store of a longer
+            # prompt sharing this prefix re-points the page key at its own
+            # slot, and popping it here would orphan that live entry (the
+            # decaying-hit-rate bug: shared-prefix keys vanished over time)
+            if self._pc_keys.get(k, (None, 0))[0] == victim:
+                del self._pc_keys[k]
         self._pc_arena[:, :, victim, :, :plen] = self.kv[:, :, kv_slot, :, :plen]
         keys = []
         for p in range(page, plen + 1, page):
