@@ -386,13 +386,16 @@ class TorchEngine:
         # multi-MB arena copy + an eviction on a one-off suffix — skip it.
         # (Unchecked, the 500-stores/s churn evicted the hot shared prefixes
         # and the hit rate decayed over long runs.)
-        existing = 0
-        for p in range(plen, page - 1, -page):
-            if hash(tuple(tokens[:p])) in self._pc_keys:
-                existing = p
-                break
-        if plen - existing < 2 * page:
-            return
+        import os as _os
+
+        if not _os.environ.get("SMG_PC_NODEDUP"):
+            existing = 0
+            for p in range(plen, page - 1, -page):
+                if hash(tuple(tokens[:p])) in self._pc_keys:
+                    existing = p
+                    break
+            if plen - existing < 2 * page:
+                return
         victim = self._pc_lru.pop(0)
         for k in self._pc_slot_keys.pop(victim, []):
             # only drop keys the victim still owns: a later store of a longer
