@@ -67,6 +67,8 @@ int smg_rope_prefill_launch_ex(const void* qkv, const void* freqs, const void* s
                                void* k_out, void* v_out, int B, int L, int n_heads, int max_seq,
                                int head_dim, void* stream, int kv_fp8);
 int smg_silu_mul_launch(const void* gu, void* out, long long rows, long long inner, void* stream);
+int smg_lse_merge_launch(const void* o1, const void* o2, const void* lse1, const void* lse2,
+                         void* out, long long rows, int head_dim, void* stream);
 int smg_rope_prefill_launch(const void* qkv, const void* freqs, const void* slots,
                             const void* starts, void* k_cache, void* v_cache, void* q_out,
                             void* k_out, void* v_out, int B, int L, int n_heads, int max_seq,
@@ -373,6 +375,16 @@ PYBIND11_MODULE(_core, m) {
           py::arg("v_cache"), py::arg("q_out"), py::arg("k_out"), py::arg("v_out"), py::arg("B"),
           py::arg("L"), py::arg("n_heads"), py::arg("max_seq"), py::arg("head_dim"),
           py::arg("stream"), py::arg("kv_fp8") = 0);
+    m.def("lse_merge",
+          [](uintptr_t o1, uintptr_t o2, uintptr_t lse1, uintptr_t lse2, uintptr_t out,
+             long long rows, int head_dim, uintptr_t stream) {
+              int rc = smg_lse_merge_launch((const void*)o1, (const void*)o2, (const void*)lse1,
+                                            (const void*)lse2, (void*)out, rows, head_dim,
+                                            (void*)stream);
+              if (rc != 0) throw std::runtime_error("lse_merge launch failed rc=" + std::to_string(rc));
+          },
+          py::arg("o1"), py::arg("o2"), py::arg("lse1"), py::arg("lse2"), py::arg("out"),
+          py::arg("rows"), py::arg("head_dim"), py::arg("stream"));
     m.def("silu_mul",
           [](uintptr_t gu, uintptr_t out, long long rows, long long inner, uintptr_t stream) {
               int rc = smg_silu_mul_launch((const void*)gu, (void*)out, rows, inner, (void*)stream);

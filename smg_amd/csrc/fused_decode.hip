@@ -242,6 +242,42 @@ extern "C" int smg_rope_kv_store_launch(
                                        n_slots, n_heads, max_seq, head_dim, stream, 0);
 }
 
+// Suffix-prefill flash-LSE merge: o = o1*exp(lse1-lse) + o2*exp(lse2-lse),
+// lse = logaddexp(lse1, lse2) — one kernel instead of the ~6 torch
+// elementwise launches per layer the two-pass flash combination costs.
+extern "C" __global__ void __launch_bounds__(WAVE) smg_lse_merge(
+    const __hip_bfloat16* __restrict__ o1,
+    const __hip_bfloat16* __restrict__ o2,
+    const float* __restrict__ lse1,
+    const float* __restrict__ lse2,
+    __hip_bfloat16* __restrict__ out,
+    long long rows, int head_dim) {
+    const long long r = blockIdx.x;
+    if (r >= rows) return;
+    const float a = lse1[r];
+    const float b = lse2[r];
+    const float m = fmaxf(a, b);
+    const float lse = m + __logf(__expf(a - m) + __expf(b - m));
+    const float w1 = __expf(a - lse);
+    const float w2 = __expf(b - lse);
+    const __hip_bfloat16* r1 = o1 + r * head_dim;
+    const __hip_bfloat16* r2 = o2 + r * head_dim;
+    __hip_bfloat16* ro = out + r * head_dim;
+    for (int i = threadIdx.x; i < head_dim; i += WAVE) {
+        ro[i] = __float2bfloat16(w1 * __bfloat162float(r1[i]) + w2 * __bfloat162float(r2[i]));
+    }
+}
+
+extern "C" int smg_lse_merge_launch(const void* o1, const void* o2, const void* lse1,
+                                    const void* lse2, void* out, long long rows, int head_dim,
+                                    void* stream) {
+    if (rows < 1) return 0;
+    hipLaunchKernelGGL(smg_lse_merge, dim3((unsigned)rows), dim3(WAVE), 0, (hipStream_t)stream,
+                       (const __hip_bfloat16*)o1, (const __hip_bfloat16*)o2, (const float*)lse1,
+                       (const float*)lse2, (__hip_bfloat16*)out, rows, head_dim);
+    return (int)hipGetLastError();
+}
+
 extern "C" int smg_silu_mul_launch(const void* gu, void* out, long long rows,
                                    long long inner, void* stream) {
     long long total = rows * inner;

@@ -176,6 +176,7 @@ class TorchEngine:
                     self._hip_fused = _core.rope_kv_store
                     self._hip_silu_mul = _core.silu_mul
                     self._hip_rope_prefill = getattr(_core, "rope_prefill", None)
+                    self._hip_lse_merge = getattr(_core, "lse_merge", None)
                     self._q_buf = torch.zeros(
                         c.max_slots, c.d_model, device=self.device, dtype=self.dtype
                     )
@@ -504,9 +505,21 @@ class TorchEngine:
                 vv = self._kv_hist(li, 1, slots)[:, :, :start0]
                 o1, lse1 = self._flash_lse(q, kk, vv, False)
                 o2, lse2 = self._flash_lse(q, k, v, True)
-                lse_tot = torch.logaddexp(lse1, lse2)
-                attn = o1 * (lse1 - lse_tot).exp().unsqueeze(-1) + o2 * (lse2 - lse_tot).exp().unsqueeze(-1)
-                attn = attn.to(q.dtype)
+                merge = getattr(self, "_hip_lse_merge", None)
+                if merge is not None and o1.is_contiguous() and o2.is_contiguous():
+                    rows = B * c.n_heads * L
+                    attn = torch.empty_like(o2)
+                    merge(
+                        o1.data_ptr(), o2.data_ptr(),
+                        # aotriton may pad lse's last dim — slice to L first
+                        lse1[..., :L].contiguous().data_ptr(),
+                        lse2[..., :L].contiguous().data_ptr(),
+                        attn.data_ptr(), rows, c.head_dim, stream,
+                    )
+                else:
+                    lse_tot = torch.logaddexp(lse1, lse2)
+                    attn = o1 * (lse1 - lse_tot).exp().unsqueeze(-1) + o2 * (lse2 - lse_tot).exp().unsqueeze(-1)
+                    attn = attn.to(q.dtype)
             else:
                 kk = self._kv_hist(li, 0, slots)[:, :, :t_max]
                 vv = self._kv_hist(li, 1, slots)[:, :, :t_max]
