@@ -161,6 +161,8 @@ class TorchEngine:
         self._hip_fused = None
         self._hip_silu_mul = None
         if self.device.type == "cuda" and self.dtype == torch.bfloat16 and c.head_dim <= 128:
+            import os as _os
+
             try:
                 from .. import _core
 
@@ -183,8 +185,17 @@ class TorchEngine:
                     self._smul_buf = torch.zeros(
                         c.max_slots, c.d_ffn, device=self.device, dtype=self.dtype
                     )
-            except ImportError:
-                pass
+            except ImportError as exc:
+                # on a GPU box the gfx950 kernels ARE the engine; a silent
+                # torch-eager fallback would hide a broken build.  CPU logic
+                # tests never reach here (device=cpu); set SMG_ALLOW_EAGER=1
+                # to debug eager on a GPU explicitly.
+                if not _os.environ.get("SMG_ALLOW_EAGER"):
+                    raise RuntimeError(
+                        "smg_amd._core (gfx950 kernels) failed to import on a CUDA/HIP "
+                        "device — run `python -m smg_amd.csrc.build` (or set "
+                        "SMG_ALLOW_EAGER=1 to force the torch-eager fallback)"
+                    ) from exc
         self._flash_lse = self._probe_flash_lse() if self.device.type == "cuda" else None
         if c.kv_fp8 and (self._hip_fused is None or self._hip_attn is None or
                          getattr(self, "_hip_rope_prefill", None) is None):
