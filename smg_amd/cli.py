@@ -163,6 +163,12 @@ def build_parser() -> argparse.ArgumentParser:
     g.add_argument("--jwt-leeway-secs", type=float, default=60.0)
     g.add_argument("--jwt-enable-jti-check", action="store_true",
                    help="replay protection via a JTI LRU cache")
+    g.add_argument("--jwt-allow-missing-exp", action="store_true",
+                   help="accept tokens without an exp claim (default: reject, as the reference does)")
+    g.add_argument("--admin-role", default="admin",
+                   help="JWT role allowed to hit control-plane mutation endpoints")
+    g.add_argument("--plugin-dir", default=None,
+                   help="only plugin modules inside this directory may be loaded via POST /wasm")
     g.add_argument("--disable-audit-logging", action="store_true")
     g.add_argument("--trust-tenant-header", action="store_true")
     g.add_argument("--tenant-header-name", default="x-smg-tenant")
@@ -339,6 +345,9 @@ def to_router_config(argv: Optional[List[str]] = None) -> RouterConfig:
     cfg.auth.jwt_role_claim = args.jwt_role_claim
     cfg.auth.jwt_leeway_secs = args.jwt_leeway_secs
     cfg.auth.jwt_enable_jti_check = args.jwt_enable_jti_check
+    cfg.auth.jwt_require_exp = not args.jwt_allow_missing_exp
+    cfg.auth.admin_role = args.admin_role
+    cfg.plugin_dir = args.plugin_dir
     cfg.auth.disable_audit_logging = args.disable_audit_logging
     cfg.storage.backend = args.backend
     cfg.storage.history_backend = args.history_backend

@@ -191,6 +191,8 @@ class AuthConfig:
     jwt_enable_jti_check: bool = False
     jwt_role_claim: str = "roles"
     jwt_role_mapping: Dict[str, str] = field(default_factory=dict)
+    jwt_require_exp: bool = True  # reference jsonwebtoken Validation::default required_spec_claims={"exp"}
+    admin_role: str = "admin"  # JWT role allowed to hit control-plane mutations (Role::is_admin)
     control_plane_api_keys: List[str] = field(default_factory=list)
     disable_audit_logging: bool = False
 
@@ -302,6 +304,9 @@ class RouterConfig:
     mcp_config_path: Optional[str] = None
     tls_cert_path: Optional[str] = None
     tls_key_path: Optional[str] = None
+    # plugin modules may only be loaded from inside this directory; when unset,
+    # POST /wasm is refused (plugins load native code — operator opt-in only)
+    plugin_dir: Optional[str] = None
 
     def validate(self) -> None:
         from . import validation

@@ -119,9 +119,32 @@ def _get_jwt_validator(ctx: "AppContext"):
             role_claim=auth.jwt_role_claim,
             role_mapping=auth.jwt_role_mapping,
             enable_jti_check=auth.jwt_enable_jti_check,
+            require_exp=auth.jwt_require_exp,
         )
         ctx.jwt_validator = v
     return v
+
+
+def _is_admin_mutation(request: web.Request) -> bool:
+    """Control-plane mutations that require an admin identity once auth is
+    configured (reference ControlPlaneAuthState + Role::is_admin gating)."""
+    if request.method in ("GET", "HEAD", "OPTIONS"):
+        return False
+    p = request.path
+    return (
+        p in ("/flush_cache", "/start_profile", "/stop_profile", "/workers", "/wasm")
+        or p.startswith("/workers/")
+        or p.startswith("/wasm/")
+        or p.startswith("/tokenizers")
+        or p.startswith("/mcp/servers")
+    )
+
+
+def _admin_forbidden() -> web.Response:
+    return web.Response(
+        status=403, body=error_body("admin role required", 403, "permission_error"),
+        content_type="application/json",
+    )
 
 
 @web.middleware
@@ -129,16 +152,24 @@ async def auth_middleware(request: web.Request, handler):
     ctx: AppContext = request.app[CTX_KEY]
     auth = ctx.config.auth
     if request.path in PUBLIC_PATHS or (
-        not auth.api_key and not auth.tenant_api_keys and not _jwt_enabled(auth)
+        not auth.api_key and not auth.tenant_api_keys
+        and not auth.control_plane_api_keys and not _jwt_enabled(auth)
     ):
         return await handler(request)
     header = request.headers.get("authorization", "")
     token = header[7:] if header.lower().startswith("bearer ") else request.headers.get("x-api-key", "")
     if auth.api_key and _ct_eq(token, auth.api_key):
+        request["auth_role"] = "admin"  # the master key is the operator key
+        return await handler(request)
+    if any(_ct_eq(token, k) for k in auth.control_plane_api_keys):
+        request["auth_role"] = "admin"
         return await handler(request)
     tenant = auth.tenant_api_keys.get(token)
     if tenant is not None:
         request["tenant_id"] = tenant
+        request["auth_role"] = "user"
+        if _is_admin_mutation(request):
+            return _admin_forbidden()
         return await handler(request)
     if _jwt_enabled(auth) and token.count(".") == 2:
         from .jwt_auth import JwtError
@@ -154,6 +185,8 @@ async def auth_middleware(request: web.Request, handler):
         request["auth_role"] = validated.role
         if validated.claims.get("tenant_id"):
             request["tenant_id"] = validated.claims["tenant_id"]
+        if _is_admin_mutation(request) and validated.role != auth.admin_role:
+            return _admin_forbidden()
         return await handler(request)
     return web.Response(
         status=401, body=error_body("invalid API key", 401, "authentication_error"),
@@ -579,8 +612,25 @@ def build_app(ctx: AppContext) -> web.Application:
         body, _ = await _read_json(request)
         if not body or "path" not in body:
             return web.Response(status=400, body=error_body("'path' is required"), content_type="application/json")
+        # plugins execute native Python: only load from the operator-configured
+        # directory, never from an arbitrary request-supplied path
+        plugin_dir = ctx.config.plugin_dir
+        if not plugin_dir:
+            return web.Response(
+                status=403,
+                body=error_body("plugin loading disabled: no --plugin-dir configured", 403, "permission_error"),
+                content_type="application/json")
+        import os as _os
+
+        resolved = _os.path.realpath(str(body["path"]))
+        root = _os.path.realpath(plugin_dir)
+        if not (resolved == root or resolved.startswith(root + _os.sep)):
+            return web.Response(
+                status=403,
+                body=error_body(f"plugin path must be inside {plugin_dir}", 403, "permission_error"),
+                content_type="application/json")
         try:
-            mod_id = ctx.plugins.add_module(body["path"], body.get("name"))
+            mod_id = ctx.plugins.add_module(resolved, body.get("name"))
         except Exception as exc:
             return web.Response(status=400, body=error_body(str(exc)), content_type="application/json")
         return web.json_response({"module_uuid": mod_id}, status=201)

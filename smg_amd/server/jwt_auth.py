@@ -298,6 +298,7 @@ class JwtValidator:
         role_mapping: Optional[Dict[str, str]] = None,
         enable_jti_check: bool = False,
         jti_cache_size: int = 10_000,
+        require_exp: bool = True,
     ):
         self.jwks = jwks
         self.issuer = issuer
@@ -306,6 +307,7 @@ class JwtValidator:
         self.role_claim = role_claim
         self.role_mapping = role_mapping or {}
         self.enable_jti_check = enable_jti_check
+        self.require_exp = require_exp
         self._jti: OrderedDict[str, float] = OrderedDict()
         self._jti_cap = jti_cache_size
 
@@ -343,7 +345,12 @@ class JwtValidator:
     def _check_claims(self, claims: Dict[str, Any]) -> None:
         now = time.time()
         exp = claims.get("exp")
-        if exp is not None and now > float(exp) + self.leeway:
+        if exp is None:
+            # jsonwebtoken Validation::default() has required_spec_claims={"exp"}:
+            # a token minted without exp must not validate forever
+            if self.require_exp:
+                raise JwtError("token has no exp claim")
+        elif now > float(exp) + self.leeway:
             raise JwtError("token expired")
         nbf = claims.get("nbf")
         if nbf is not None and now < float(nbf) - self.leeway:

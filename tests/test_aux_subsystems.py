@@ -110,8 +110,13 @@ class TestPlugins:
         async def run():
             plug = self.write_plugin(tmp_path, "def on_response(ctx):\n    return {'extra_headers': {'x-plugged': '1'}}\n")
             ctx, engines = make_ctx()
+            ctx.config.plugin_dir = str(tmp_path)
             client = await start_client(ctx, engines)
             try:
+                # paths outside --plugin-dir are refused (never exec
+                # request-supplied native code)
+                resp = await client.post("/wasm", json={"path": "/etc/passwd", "name": "evil"})
+                assert resp.status == 403
                 resp = await client.post("/wasm", json={"path": plug, "name": "test"})
                 assert resp.status == 201
                 mod_id = (await resp.json())["module_uuid"]

@@ -155,6 +155,18 @@ def test_expired_and_audience_and_issuer():
         _run(_validator().validate(tok))
 
 
+def test_missing_exp_rejected():
+    # jsonwebtoken Validation::default() requires exp; a token minted without
+    # exp must not validate forever
+    c = _claims()
+    del c["exp"]
+    tok = encode_jwt(c, b"supersecret-hmac-key", "HS256", kid="hskey")
+    with pytest.raises(JwtError, match="exp"):
+        _run(_validator().validate(tok))
+    v = _run(_validator(require_exp=False).validate(tok))
+    assert v.subject == "user-1"
+
+
 def test_jti_replay():
     val = _validator(enable_jti_check=True)
     tok = encode_jwt(_claims(jti="once"), b"supersecret-hmac-key", "HS256", kid="hskey")
@@ -192,6 +204,51 @@ def test_middleware_jwt_e2e():
             assert r.status == 401
             r = await client.get("/health")  # public path bypasses auth
             assert r.status == 200
+            # control-plane mutations require the admin role (Role::is_admin)
+            hdr = {"Authorization": f"Bearer {tok}"}
+            r = await client.post("/flush_cache", headers=hdr)
+            assert r.status == 403
+            r = await client.post("/workers", json={"url": "http://w:1"}, headers=hdr)
+            assert r.status == 403
+            admin_tok = encode_jwt(_claims(roles=["admin"]), b"supersecret-hmac-key", "HS256", kid="hskey")
+            r = await client.post("/flush_cache", headers={"Authorization": f"Bearer {admin_tok}"})
+            assert r.status == 200
+        finally:
+            await client.close()
+
+    asyncio.new_event_loop().run_until_complete(run())
+
+
+def test_tenant_key_cannot_mutate_control_plane():
+    from aiohttp.test_utils import TestClient, TestServer
+
+    from smg_amd.config import PolicyConfig, RouterConfig
+    from smg_amd.server.app import build_app
+    from smg_amd.server.app_context import AppContext
+
+    async def run():
+        cfg = RouterConfig(policy=PolicyConfig(name="round_robin", gpu_tree=False))
+        cfg.health_check.disable = True
+        cfg.auth.api_key = "operator-key"
+        cfg.auth.tenant_api_keys = {"tenant-key": "acme"}
+        ctx = AppContext(cfg)
+        from smg_amd.routers.factory import RouterManager
+
+        ctx.router_manager = RouterManager(ctx, cfg)
+        client = TestClient(TestServer(build_app(ctx)))
+        await client.start_server()
+        try:
+            # tenant keys are data-plane identities: 403 on mutations
+            r = await client.post("/flush_cache", headers={"Authorization": "Bearer tenant-key"})
+            assert r.status == 403
+            r = await client.post("/wasm", json={"path": "/tmp/x.py"}, headers={"Authorization": "Bearer tenant-key"})
+            assert r.status == 403
+            # the operator (master) key is admin
+            r = await client.post("/flush_cache", headers={"Authorization": "Bearer operator-key"})
+            assert r.status == 200
+            # .. but /wasm still refuses without a configured plugin dir
+            r = await client.post("/wasm", json={"path": "/tmp/x.py"}, headers={"Authorization": "Bearer operator-key"})
+            assert r.status == 403
         finally:
             await client.close()
 
