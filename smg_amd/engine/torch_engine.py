@@ -150,9 +150,10 @@ class TorchEngine:
             )
         else:
             self._pc_arena = None
-        self._pc_keys: Dict[int, tuple] = {}  # hash(prefix pages) -> (slot, plen)
+        self._pc_keys: Dict[int, tuple] = {}  # chain-key -> (slot, plen)
         self._pc_lru: List[int] = list(range(pc.prefix_cache_slots))  # front = LRU victim
-        self._pc_slot_keys: Dict[int, List[int]] = {}  # slot -> its hash keys
+        self._pc_slot_keys: Dict[int, List[int]] = {}  # slot -> its chain keys
+        self._pc_slot_tokens: Dict[int, List[int]] = {}  # slot -> cached prefix (hit verification)
         self.prefix_cache_hits = 0
         self.prefix_cache_miss = 0
         # fused gfx950 decode-attention kernel (csrc/attn_decode.hip): reads
@@ -291,11 +292,30 @@ class TorchEngine:
             items = []
             for slot, req in group:
                 items.append((slot, req.prefilled, req.tokens[req.prefilled: req.prefilled + L]))
-            self._prefill_batch(items)
-            for slot, req in group:
+            first_toks = self._prefill_batch(items)
+            finished_pf = []
+            for (slot, req), tok in zip(group, first_toks):
                 req.prefilled += L
                 if req.prefilled >= len(req.tokens):
                     self._prefix_store(slot, req.tokens)
+                    # the final prefill chunk's logits sample the first
+                    # generated token — decode then starts from the SAMPLED
+                    # token at position n, instead of re-feeding the last
+                    # prompt token (which wasted a KV slot and conditioned
+                    # the first token on [.., t_{n-1}, t_{n-1}])
+                    req.generated.append(tok)
+                    produced += 1
+                    self.total_generated += 1
+                    if len(req.generated) >= req.max_new or self._seq_len_host[slot] >= c.max_seq - 2:
+                        req.done = True
+                        finished_pf.append(slot)
+                    self._step_events.append((req.rid, tok, 1 if req.done else 0))
+            for s in finished_pf:
+                del self.running[s]
+                self._free_slots.append(s)
+                self._seq_len_host[s] = 0
+            if finished_pf:
+                self.seq_len[torch.tensor(finished_pf, device=self.device)] = 0
 
         for _ in range(max(1, decode_burst)):
             decode_slots = [s for s, r in self.running.items() if r.prefilled >= len(r.tokens)]
@@ -363,16 +383,32 @@ class TorchEngine:
         return torch.view_as_real(xc * freqs).flatten(-2).to(x.dtype)
 
     # ---- prefix KV cache ---------------------------------------------------
+    # Keys are the kvindex chain-hash schedule (kvindex/chain_keys.py — the
+    # same content-addressed rolling hash the GPU tree uses): ONE O(n) pass
+    # yields the key at every page boundary, vs the old
+    # hash(tuple(tokens[:p]))-per-boundary which re-hashed the whole prefix
+    # at each depth (O(n·pages), on the host, per admission).  Content
+    # addressing also removes Python's per-process hash randomization, and a
+    # token compare on hit makes a 64-bit collision restore impossible
+    # (vLLM's content-addressed-hashing fix for cross-request KV leakage).
     def _prefix_lookup(self, tokens: List[int]):
-        """Longest page-aligned cached prefix of `tokens` -> (arena_slot, plen)."""
+        """Longest page-aligned cached prefix of `tokens` -> (arena_slot, plen).
+        Capped at len(tokens)-1 so prefill always runs at least one token —
+        the final-chunk logits sample the first generated token."""
         c = self.cfg
         page = c.prefix_cache_page
-        pmax = min((len(tokens) // page) * page, c.prefix_cache_max, c.max_seq - 1)
-        for p in range(pmax, page - 1, -page):
-            entry = self._pc_keys.get(hash(tuple(tokens[:p])))
+        pmax = min(((len(tokens) - 1) // page) * page, c.prefix_cache_max, c.max_seq - 1)
+        if pmax < page:
+            return None
+        from ..kvindex.chain_keys import chain_keys
+
+        keys = chain_keys(tokens[:pmax], page)
+        for i in range(len(keys) - 1, -1, -1):
+            p = (i + 1) * page
+            entry = self._pc_keys.get(keys[i])
             if entry is not None:
                 slot, stored = entry
-                if stored >= p:
+                if stored >= p and self._pc_slot_tokens.get(slot, [])[:p] == tokens[:p]:
                     # LRU touch
                     try:
                         self._pc_lru.remove(slot)
@@ -390,8 +426,10 @@ class TorchEngine:
         plen = min((len(tokens) // page) * page, c.prefix_cache_max, c.max_seq - 1)
         if plen < page:
             return
-        top_key = hash(tuple(tokens[:plen]))
-        if top_key in self._pc_keys:
+        from ..kvindex.chain_keys import chain_keys
+
+        keys = chain_keys(tokens[:plen], page)
+        if keys[-1] in self._pc_keys:
             return  # already cached
         # block-level dedup (vLLM-style): if the cache already covers all but
         # the last page or so of this prompt, a new entry would spend a
@@ -402,9 +440,9 @@ class TorchEngine:
 
         if not _os.environ.get("SMG_PC_NODEDUP"):
             existing = 0
-            for p in range(plen, page - 1, -page):
-                if hash(tuple(tokens[:p])) in self._pc_keys:
-                    existing = p
+            for i in range(len(keys) - 1, -1, -1):
+                if keys[i] in self._pc_keys:
+                    existing = (i + 1) * page
                     break
             if plen - existing < 2 * page:
                 return
@@ -417,12 +455,10 @@ class TorchEngine:
             if self._pc_keys.get(k, (None, 0))[0] == victim:
                 del self._pc_keys[k]
         self._pc_arena[:, :, victim, :, :plen] = self.kv[:, :, kv_slot, :, :plen]
-        keys = []
-        for p in range(page, plen + 1, page):
-            k = hash(tuple(tokens[:p]))
-            self._pc_keys[k] = (victim, p)
-            keys.append(k)
-        self._pc_slot_keys[victim] = keys
+        for i, k in enumerate(keys):
+            self._pc_keys[k] = (victim, (i + 1) * page)
+        self._pc_slot_keys[victim] = list(keys)
+        self._pc_slot_tokens[victim] = list(tokens[:plen])
         self._pc_lru.append(victim)
 
     def _mlp(self, h, layer):
@@ -538,10 +574,16 @@ class TorchEngine:
             attn2 = attn.transpose(1, 2).reshape(B * L, c.d_model)
             h = torch.addmm(h.view(B * L, c.d_model), attn2, layer.wo).view(B, L, c.d_model)
             h = self._mlp(h, layer)
+        # final-position logits: one [B,D]@[D,V] GEMM samples the next token.
+        # For requests whose prefill completes with this chunk this IS the
+        # first generated token; for the rest _last_tok is overwritten by
+        # their final chunk before decode ever reads it.
+        nxt = (_rms(h[:, -1], self.ln_f) @ self.embed.t()).argmax(-1)  # [B]
+        self._last_tok[slots] = nxt
         for slot, start, toks in items:
             self._seq_len_host[slot] = start + L
-            self._last_tok[slot] = int(toks[-1])
         self.seq_len[slots] = starts + L
+        return nxt.tolist()
 
     def _decode_core(self, maxlen: int) -> torch.Tensor:
         """Full-arena decode forward: every slot participates with a static
