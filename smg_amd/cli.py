@@ -376,12 +376,54 @@ def to_router_config(argv: Optional[List[str]] = None) -> RouterConfig:
     return cfg
 
 
+def rccl_worker_main(cfg) -> None:
+    """Worker-rank entry for `smg launch --connection-mode rccl` under
+    torchrun: ranks >= 1 run the lockstep plane <-> engine loop instead of
+    the HTTP server (routers/rccl_router.py run_worker_loop)."""
+    import os
+
+    import torch
+    import torch.distributed as dist
+
+    from .comm.plane import PlaneConfig, WorkerPlane
+    from .engine.torch_engine import TorchEngine, TorchEngineConfig
+    from .routers.rccl_router import run_worker_loop
+
+    local_rank = int(os.environ.get("LOCAL_RANK", os.environ.get("RANK", "0")))
+    use_gpu = torch.cuda.is_available()
+    device = f"cuda:{local_rank}" if use_gpu else "cpu"
+    if use_gpu:
+        torch.cuda.set_device(local_rank)
+    if not dist.is_initialized():
+        dist.init_process_group(backend="nccl" if use_gpu else "gloo")
+    eng = TorchEngine(
+        TorchEngineConfig() if use_gpu else TorchEngineConfig.tiny(),
+        device=device, graphs=use_gpu,
+    )
+    plane = WorkerPlane(
+        PlaneConfig(
+            max_reqs_per_tick=cfg.rccl.max_batch_requests,
+            max_prompt=cfg.rccl.max_tokens_per_msg,
+            device=device if use_gpu else "cpu",
+        )
+    )
+    run_worker_loop(eng, plane)
+    dist.destroy_process_group()
+
+
 def main(argv: Optional[List[str]] = None) -> None:
     raw_argv = list(sys.argv[1:] if argv is None else argv)
     cfg = to_router_config(argv)
     from .observability.logging import setup_logging
 
     setup_logging(cfg.log_level, cfg.log_json, cfg.log_dir)
+    import os as _os
+
+    from .config import ConnectionMode
+
+    if cfg.connection_mode == ConnectionMode.RCCL and int(_os.environ.get("RANK", "0")) > 0:
+        rccl_worker_main(cfg)
+        return
     # `smg serve`: spawn local workers next to the router (reference
     # bindings/python serve.py spawns engine workers + router together)
     n_local = 0

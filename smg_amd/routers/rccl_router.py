@@ -1,0 +1,480 @@
+"""RCCL serving router — the MI355X-native `connection_mode=rccl` transport.
+
+Carries /v1/chat/completions (and /v1/completions, /generate) end-to-end over
+the RCCL-over-xGMI lockstep plane (comm/plane.py) to N worker ranks running
+TorchEngine, one rank per GPU.  This is the serving integration of the plane:
+the same TickGateway core drives bench.py, so the measured benchmark number is
+the product's number.
+
+Reference equivalent of this integration: the collapsed same-host engine path
+wired in as a first-class router transport —
+crates/engine_zmq_client/src/connector.rs:235 (EngineCoreClient submit ->
+RequestStream) + model_gateway/src/routers/grpc/zmq_client.rs:1-12 (the ZMQ
+adapter presenting the engine surface to the router).  Here the "collapsed
+hop" is RCCL p2p over xGMI instead of ZMQ ipc://.
+
+Topology (torchrun, one process per GPU):
+  rank 0   — the gateway: HTTP server + policy + plane + its OWN engine
+  rank 1.. — worker loop (run_worker_loop): plane tick <-> TorchEngine step
+
+Single-rank mode (world=1) serves from the local engine with no plane.
+"""
+from __future__ import annotations
+
+import asyncio
+import itertools
+import json
+import statistics
+import threading
+import time
+from collections import deque
+from typing import Any, Callable, Dict, List, Optional, Sequence, Tuple
+
+from ..comm.plane import DONE, GatewayPlane, PlaneConfig, WorkerPlane
+from ..policies import SelectWorkerInfo
+from .base import RouteRequest, RouteResponse, Router
+from ..workers.worker import Worker
+
+
+class TickGateway:
+    """Synchronous rank-0 serving core: one lockstep tick routes pending
+    requests (batched policy select — ONE GPU tree kernel per tick), ships
+    them over the plane, steps the local engine while remote workers compute,
+    then collects token events.
+
+    Thread-contract: submit() may be called from any thread (deque appends
+    are atomic); tick() runs on exactly one thread; on_event fires on the
+    tick thread.
+    """
+
+    def __init__(
+        self,
+        workers: Sequence[Worker],
+        policy,
+        plane: Optional[GatewayPlane] = None,
+        local_engine=None,
+        decode_burst: int = 2,
+        model_id: str = "default",
+        max_new_arrivals_per_tick: int = 128,
+        on_event: Optional[Callable[[int, int, int], None]] = None,
+    ):
+        self.workers = list(workers)
+        self.policy = policy
+        self.plane = plane
+        self.local_engine = local_engine
+        self.decode_burst = decode_burst
+        self.model_id = model_id
+        self.max_new_arrivals = max_new_arrivals_per_tick
+        self.on_event = on_event
+        self._pending: deque = deque()  # (rid, tokens, max_new)
+        self._local_pending: List[Tuple[int, int, List[int]]] = []
+        self.inflight: Dict[int, int] = {}  # rid -> worker index
+        self._rid = itertools.count(1)
+        self.completed_total = 0
+        self.routing_lat: List[float] = []
+        self.phase_t = {"route": 0.0, "local": 0.0, "plane": 0.0, "events": 0.0, "ticks": 0}
+
+    # ---- submission (any thread) -----------------------------------------
+    def submit(self, tokens: List[int], max_new: int, rid: Optional[int] = None) -> int:
+        if rid is None:
+            rid = next(self._rid) & 0x7FFFFFFF
+        self._pending.append((rid, list(tokens), max_new))
+        return rid
+
+    @property
+    def pending_count(self) -> int:
+        return len(self._pending)
+
+    # ---- the tick (one thread) -------------------------------------------
+    def tick(self) -> int:
+        """One lockstep exchange; returns completions this tick."""
+        self.phase_t["ticks"] += 1
+        # 1) route this tick's arrivals in one batch (one GPU kernel launch)
+        new_reqs: List[Tuple[int, List[int], int]] = []
+        while self._pending and len(new_reqs) < self.max_new_arrivals:
+            new_reqs.append(self._pending.popleft())
+        if new_reqs:
+            infos = [
+                SelectWorkerInfo(
+                    request_id=str(rid), model_id=self.model_id, tokens=toks, est_tokens=len(toks)
+                )
+                for rid, toks, _ in new_reqs
+            ]
+            t0 = time.perf_counter()
+            if hasattr(self.policy, "select_worker_batch"):
+                sels = self.policy.select_worker_batch(self.workers, infos)
+            else:
+                sels = [self.policy.select_worker(self.workers, i) for i in infos]
+            dt = time.perf_counter() - t0
+            self.routing_lat.extend([dt / len(new_reqs)] * len(new_reqs))
+            self.phase_t["route"] += dt
+            for (rid, toks, max_new), sel in zip(new_reqs, sels):
+                sel = 0 if sel is None else sel
+                self.workers[sel].incr_load()
+                self.inflight[rid] = sel
+                rank = self.workers[sel].rccl_rank
+                if rank in (None, 0):
+                    self._local_pending.append((rid, max_new, toks))
+                else:
+                    self.plane.enqueue(rank, rid, max_new, toks)
+        # 2) ship remote work first so workers overlap the local engine step
+        if self.plane is not None:
+            tp = time.perf_counter()
+            self.plane.tick_send()
+            self.phase_t["plane"] += time.perf_counter() - tp
+        events: List[Tuple[int, int, int]] = []
+        if self.local_engine is not None:
+            tl = time.perf_counter()
+            for rid, max_new, toks in self._local_pending:
+                self.local_engine.submit(toks, max_new, rid=rid)
+            self._local_pending.clear()
+            self.local_engine.step(decode_burst=self.decode_burst)
+            events.extend(self.local_engine.drain_events())
+            self.phase_t["local"] += time.perf_counter() - tl
+        if self.plane is not None:
+            tp = time.perf_counter()
+            for _w, evs in self.plane.tick_recv().items():
+                events.extend(evs)
+            self.phase_t["plane"] += time.perf_counter() - tp
+        # 3) completions + event fan-out
+        tev = time.perf_counter()
+        done_now = 0
+        cb = self.on_event
+        for rid, token, flags in events:
+            if cb is not None:
+                cb(rid, token, flags)
+            if flags & DONE:
+                wrk = self.inflight.pop(rid, None)
+                if wrk is not None:
+                    self.workers[wrk].decr_load()  # also bumps processed_requests
+                    done_now += 1
+        self.completed_total += done_now
+        self.phase_t["events"] += time.perf_counter() - tev
+        return done_now
+
+    # ---- timing / shutdown ------------------------------------------------
+    def barrier_sync(self) -> None:
+        """Plane barrier tick + dist barrier (bench timing bracket)."""
+        if self.plane is not None:
+            import torch.distributed as dist
+
+            self.plane.tick(barrier=True)
+            dist.barrier()
+
+    def stop_workers(self) -> None:
+        if self.plane is not None:
+            self.plane.tick_send(stop=True)
+            self.plane.tick_recv()  # drain the workers' final event sends
+
+    def p50_routing_ms(self) -> Optional[float]:
+        return statistics.median(self.routing_lat) * 1e3 if self.routing_lat else None
+
+
+def run_worker_loop(engine, plane: WorkerPlane, decode_burst: int = 2) -> float:
+    """Worker-rank loop (ranks >= 1): lockstep plane ticks against the local
+    engine until the gateway sends STOP.  Returns this rank's timed-region
+    elapsed seconds when the gateway bracketed the run with barrier ticks
+    (bench), else 0."""
+    import torch
+    import torch.distributed as dist
+
+    events: List[tuple] = []
+    t0 = t1 = None
+    while True:
+        reqs, stop = plane.tick(events)
+        if plane.barrier_requested:
+            dist.barrier()
+            if hasattr(torch.cuda, "is_available") and torch.cuda.is_available():
+                torch.cuda.synchronize()
+            if t0 is None:
+                t0 = time.perf_counter()
+            else:
+                t1 = time.perf_counter()
+        if stop:
+            break
+        for rid, max_new, prompt in reqs:
+            engine.submit(prompt, max_new, rid=rid)
+        engine.step(decode_burst=decode_burst)
+        events = engine.drain_events()
+    return (t1 - t0) if (t0 is not None and t1 is not None) else 0.0
+
+
+# ---------------------------------------------------------------------------
+# asyncio serving adapter
+# ---------------------------------------------------------------------------
+class _ReqState:
+    __slots__ = ("queue", "tokens", "created")
+
+    def __init__(self):
+        self.queue: asyncio.Queue = asyncio.Queue()
+        self.tokens: List[int] = []
+        self.created = time.time()
+
+
+class RcclRouter(Router):
+    """`connection_mode=rccl` router: OpenAI chat/completions served over the
+    lockstep plane.  A dedicated tick thread drives TickGateway; per-request
+    events are forwarded onto the asyncio loop for SSE streaming."""
+
+    router_id = "rccl-regular"
+
+    def __init__(self, ctx, config, *, engine=None, plane=None, world: Optional[int] = None):
+        import os
+
+        self.ctx = ctx
+        self.config = config
+        self.model_id = "default"
+        world = world if world is not None else int(os.environ.get("WORLD_SIZE", "1"))
+        self.world = world
+        # fleet: one Worker per rank; rank 0 is the gateway's own engine
+        self.workers = []
+        for r in range(world):
+            w = Worker(f"rccl://rank-{r}", model_id=self.model_id, rccl_rank=r)
+            self.workers.append(w)
+            try:
+                if ctx.worker_registry.get_by_url(w.url) is None:
+                    ctx.worker_registry.register(w)
+            except Exception:
+                pass
+        if engine is None:
+            engine = self._build_local_engine(config)
+        self.engine = engine
+        if plane is None and world > 1:
+            import torch
+            import torch.distributed as dist
+
+            use_gpu = torch.cuda.is_available()
+            if not dist.is_initialized():
+                dist.init_process_group(backend="nccl" if use_gpu else "gloo")
+            rcfg = config.rccl
+            plane = GatewayPlane(
+                PlaneConfig(
+                    max_reqs_per_tick=rcfg.max_batch_requests,
+                    max_prompt=rcfg.max_tokens_per_msg,
+                    device="cuda:0" if use_gpu else "cpu",
+                ),
+                list(range(1, world)),
+            )
+        try:
+            self.loop = asyncio.get_running_loop()
+        except RuntimeError:
+            self.loop = asyncio.get_event_loop()
+        self._states: Dict[int, _ReqState] = {}
+        policy = ctx.policy_registry.get(self.model_id)
+        self.gw = TickGateway(
+            self.workers,
+            policy,
+            plane=plane,
+            local_engine=engine,
+            model_id=self.model_id,
+            on_event=self._on_event_tick_thread,
+        )
+        self._stop = False
+        self.idle_sleep_s = max(1, config.rccl.tick_interval_us) / 1e6
+        self._thread = threading.Thread(target=self._tick_loop, name="rccl-tick", daemon=True)
+        self._thread.start()
+
+    @staticmethod
+    def _build_local_engine(config):
+        import torch
+
+        from ..engine.torch_engine import TorchEngine, TorchEngineConfig
+
+        if torch.cuda.is_available():
+            return TorchEngine(TorchEngineConfig(), device="cuda:0", graphs=True)
+        return TorchEngine(TorchEngineConfig.tiny(), device="cpu")
+
+    # ---- tick thread ------------------------------------------------------
+    def _tick_loop(self):
+        gw = self.gw
+        while not self._stop:
+            busy = gw.inflight or gw.pending_count
+            gw.tick()
+            if not busy:
+                time.sleep(self.idle_sleep_s)
+        gw.stop_workers()
+
+    def _on_event_tick_thread(self, rid: int, token: int, flags: int):
+        self.loop.call_soon_threadsafe(self._on_event, rid, token, flags)
+
+    def _on_event(self, rid: int, token: int, flags: int):
+        st = self._states.get(rid)
+        if st is not None:
+            st.queue.put_nowait((token, flags))
+
+    # ---- request helpers ---------------------------------------------------
+    def _tokenizer(self):
+        reg = getattr(self.ctx, "tokenizer_registry", None)
+        tok = reg.get(self.model_id) if reg else None
+        if tok is None:
+            from ..tokenizer.registry import MockTokenizer
+
+            tok = MockTokenizer()
+        return tok
+
+    def _encode_chat(self, body: Dict[str, Any], tok) -> List[int]:
+        msgs = body.get("messages") or []
+        tpl = getattr(self.ctx, "chat_template", None)
+        if tpl is not None:
+            try:
+                text = tpl.render(msgs, add_generation_prompt=True, tools=body.get("tools"))
+            except Exception:
+                text = None
+        else:
+            text = None
+        if text is None:
+            text = "\n".join(f"{m.get('role', 'user')}: {_content_text(m.get('content'))}" for m in msgs)
+        return tok.encode(text)
+
+    async def route(self, req: RouteRequest) -> RouteResponse:
+        if req.path.endswith("/chat/completions"):
+            return await self.route_chat(req)
+        if req.path.endswith("/completions"):
+            return await self.route_completion(req)
+        if req.path.endswith("/generate"):
+            return await self.route_generate(req)
+        return RouteResponse(status=404, body=b'{"error":"unsupported path for rccl router"}')
+
+    async def route_chat(self, req: RouteRequest) -> RouteResponse:
+        body = req.body or {}
+        tok = self._tokenizer()
+        input_ids = self._encode_chat(body, tok)
+        return await self._serve(req, body, tok, input_ids, chat=True)
+
+    async def route_completion(self, req: RouteRequest) -> RouteResponse:
+        body = req.body or {}
+        tok = self._tokenizer()
+        prompt = body.get("prompt") or ""
+        if isinstance(prompt, list):
+            prompt = prompt[0] if prompt else ""
+        input_ids = prompt if isinstance(prompt, list) else tok.encode(str(prompt))
+        return await self._serve(req, body, tok, input_ids, chat=False)
+
+    async def route_generate(self, req: RouteRequest) -> RouteResponse:
+        body = req.body or {}
+        tok = self._tokenizer()
+        ids = body.get("input_ids") or tok.encode(str(body.get("text") or ""))
+        body2 = {"max_tokens": (body.get("sampling_params") or {}).get("max_new_tokens", 16)}
+        return await self._serve(req, body2, tok, ids, chat=False, generate=True)
+
+    async def _serve(self, req, body, tok, input_ids, chat: bool, generate: bool = False) -> RouteResponse:
+        max_new = int(
+            body.get("max_completion_tokens") or body.get("max_tokens") or 16
+        )
+        st = _ReqState()
+        rid = self.gw.submit(input_ids, max_new)
+        self._states[rid] = st
+        model = body.get("model") or self.model_id
+        stream = bool(body.get("stream"))
+        if stream:
+            return RouteResponse(
+                status=200,
+                headers={"Content-Type": "text/event-stream"},
+                stream=self._sse(rid, st, tok, model, chat),
+            )
+        try:
+            timeout = self.config.request_timeout_secs
+            deadline = time.monotonic() + timeout
+            while True:
+                tk, flags = await asyncio.wait_for(st.queue.get(), timeout=max(0.1, deadline - time.monotonic()))
+                st.tokens.append(tk)
+                if flags & DONE:
+                    break
+        except asyncio.TimeoutError:
+            return RouteResponse(status=504, body=b'{"error":"generation timed out"}')
+        finally:
+            self._states.pop(rid, None)
+        text = tok.decode(st.tokens)
+        if generate:
+            out = {"text": text, "output_ids": st.tokens, "meta_info": {"completion_tokens": len(st.tokens)}}
+        elif chat:
+            out = _chat_response(rid, model, text, len(input_ids), len(st.tokens))
+        else:
+            out = _completion_response(rid, model, text, len(input_ids), len(st.tokens))
+        return RouteResponse(status=200, body=json.dumps(out).encode())
+
+    async def _sse(self, rid: int, st: _ReqState, tok, model: str, chat: bool):
+        obj = "chat.completion.chunk" if chat else "text_completion"
+        created = int(time.time())
+        n_out = 0
+        try:
+            if chat:
+                first = {
+                    "id": f"chatcmpl-{rid}", "object": obj, "created": created, "model": model,
+                    "choices": [{"index": 0, "delta": {"role": "assistant", "content": ""}, "finish_reason": None}],
+                }
+                yield f"data: {json.dumps(first)}\n\n".encode()
+            while True:
+                tk, flags = await asyncio.wait_for(st.queue.get(), timeout=self.config.request_timeout_secs)
+                st.tokens.append(tk)
+                piece = tok.decode_incremental(st.tokens, len(st.tokens) - 1)
+                n_out += 1
+                done = bool(flags & DONE)
+                if chat:
+                    chunk = {
+                        "id": f"chatcmpl-{rid}", "object": obj, "created": created, "model": model,
+                        "choices": [{
+                            "index": 0,
+                            "delta": {"content": piece},
+                            "finish_reason": "stop" if done else None,
+                        }],
+                    }
+                else:
+                    chunk = {
+                        "id": f"cmpl-{rid}", "object": obj, "created": created, "model": model,
+                        "choices": [{"index": 0, "text": piece, "finish_reason": "stop" if done else None}],
+                    }
+                yield f"data: {json.dumps(chunk)}\n\n".encode()
+                if done:
+                    break
+            yield b"data: [DONE]\n\n"
+        finally:
+            self._states.pop(rid, None)
+
+    async def get_loads(self) -> Dict[str, Any]:
+        return {
+            "workers": [
+                {"url": w.url, "active": w.active_requests, "processed": w.processed_requests}
+                for w in self.workers
+            ],
+            "p50_routing_ms": self.gw.p50_routing_ms(),
+            "completed": self.gw.completed_total,
+        }
+
+    async def shutdown(self) -> None:
+        self._stop = True
+        self._thread.join(timeout=10)
+
+
+def _content_text(content) -> str:
+    if isinstance(content, str):
+        return content
+    if isinstance(content, list):
+        return " ".join(p.get("text", "") for p in content if isinstance(p, dict))
+    return ""
+
+
+def _chat_response(rid, model, text, n_in, n_out):
+    return {
+        "id": f"chatcmpl-{rid}",
+        "object": "chat.completion",
+        "created": int(time.time()),
+        "model": model,
+        "choices": [{
+            "index": 0,
+            "message": {"role": "assistant", "content": text},
+            "finish_reason": "stop",
+        }],
+        "usage": {"prompt_tokens": n_in, "completion_tokens": n_out, "total_tokens": n_in + n_out},
+    }
+
+
+def _completion_response(rid, model, text, n_in, n_out):
+    return {
+        "id": f"cmpl-{rid}",
+        "object": "text_completion",
+        "created": int(time.time()),
+        "model": model,
+        "choices": [{"index": 0, "text": text, "finish_reason": "stop"}],
+        "usage": {"prompt_tokens": n_in, "completion_tokens": n_out, "total_tokens": n_in + n_out},
+    }
