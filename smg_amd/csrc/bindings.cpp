@@ -27,6 +27,7 @@ void smg_gpu_tree_staging(void* p, uint32_t** tokens, uint32_t** offsets);
 int smg_gpu_tree_remove_tenant(void* p, int slot);
 int smg_gpu_tree_clear_entries(void* p, const unsigned long long* keys, int n, int slot);
 int smg_gpu_tree_evict_older(void* p, uint32_t cutoff);
+long long smg_gpu_tree_reclaim(void* p);
 int smg_gpu_tree_stats(void* p, unsigned long long* out);
 int smg_gpu_tree_clear(void* p);
 int smg_hip_device_count();
@@ -156,8 +157,13 @@ class PyGpuTree {
     void evict_older(uint32_t cutoff) {
         if (smg_gpu_tree_evict_older(h_, cutoff)) throw std::runtime_error("evict failed");
     }
+    long long reclaim() {
+        long long n = smg_gpu_tree_reclaim(h_);
+        if (n < 0) throw std::runtime_error("reclaim failed");
+        return n;
+    }
     py::dict stats() {
-        unsigned long long out[67];
+        unsigned long long out[68];
         if (smg_gpu_tree_stats(h_, out)) throw std::runtime_error("stats failed");
         py::dict d;
         py::list per_tenant;
@@ -166,6 +172,7 @@ class PyGpuTree {
         d["live_nodes"] = out[64];
         d["allocated_nodes"] = out[65];
         d["clock"] = out[66];
+        d["free_nodes"] = out[67];
         return d;
     }
     void clear() {
@@ -415,6 +422,7 @@ PYBIND11_MODULE(_core, m) {
         .def("remove_tenant", &PyGpuTree::remove_tenant)
         .def("clear_entries", &PyGpuTree::clear_entries)
         .def("evict_older", &PyGpuTree::evict_older)
+        .def("reclaim", &PyGpuTree::reclaim)
         .def("stats", &PyGpuTree::stats)
         .def("clear", &PyGpuTree::clear);
 

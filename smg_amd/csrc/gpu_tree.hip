@@ -45,15 +45,42 @@ struct GpuTreeDev {
     unsigned long long* node_tenants;  // bitmask of worker slots
     uint32_t* node_ts;                 // [node_cap * 64] per-tenant LRU stamp
     uint32_t* node_parent;             // parent node id (eviction bookkeeping)
-    uint32_t* node_slot;               // this node's table slot (for removal)
+    uint32_t* node_slot;               // this node's table slot (0xffffffff = unpublished)
+    // node reclamation: LIFO free list refilled by smg_tree_reclaim so a
+    // long-lived gateway keeps learning after the bump allocator hits
+    // node_cap (reference: token_tree.rs LRU eviction to --max-tree-size)
+    uint32_t* free_list;  // [node_cap] reclaimed node ids
+    int* free_top;        // stack top (count of free ids)
     // counters
-    uint32_t* next_node;  // node allocator (node 0 = root, never allocated)
+    uint32_t* next_node;  // bump allocator (node 0 = root, never allocated)
     uint32_t* clock_;     // logical LRU clock
     uint32_t node_cap;
     uint32_t table_mask;  // table_size - 1
     uint32_t page_size;
     uint32_t max_pages;   // walk depth cap
 };
+
+#define NODE_SLOT_NONE 0xffffffffu
+
+// Allocate a node id: pop the free list first, else bump-allocate.  All tree
+// kernels run on ONE stream (GpuTreeHost::stream), so reclamation is never
+// concurrent with match/insert — the free list needs no ABA protection.
+__device__ __forceinline__ uint32_t tree_alloc_node(const GpuTreeDev& T) {
+    int top = atomicSub(T.free_top, 1);
+    if (top > 0) return T.free_list[top - 1];
+    atomicAdd(T.free_top, 1);  // underflow repair
+    uint32_t fresh = atomicAdd(T.next_node, 1u);
+    if (fresh >= T.node_cap) {
+        atomicMin(T.next_node, T.node_cap + 1024u);  // keep the counter bounded
+        return NODE_SLOT_NONE;
+    }
+    return fresh;
+}
+
+__device__ __forceinline__ void tree_free_node(const GpuTreeDev& T, uint32_t id) {
+    int top = atomicAdd(T.free_top, 1);
+    if (top >= 0 && top < (int)T.node_cap) T.free_list[top] = id;
+}
 
 __device__ __forceinline__ unsigned long long mix64(unsigned long long x) {
     // splitmix64 finalizer
@@ -368,11 +395,18 @@ smg_tree_match_insert(GpuTreeDev T, BatchArgs A) {
         if (page >= n_pages) continue;
         int id = s_ids[page];
         if (id < 0) {
-            uint32_t fresh = atomicAdd(T.next_node, 1u);
-            if (fresh >= T.node_cap) continue;  // pool exhausted
+            uint32_t fresh = tree_alloc_node(T);
+            if (fresh == NODE_SLOT_NONE) continue;  // pool exhausted until next reclaim
             uint32_t oslot;
             id = probe_insert(T, s_keys[page], fresh, &oslot);
-            if (id < 0) continue;
+            if (id < 0) {
+                tree_free_node(T, fresh);  // table neighborhood full
+                continue;
+            }
+            if (id == (int)fresh)
+                T.node_slot[fresh] = oslot;  // we published it: remember the slot for reclaim
+            else
+                tree_free_node(T, fresh);  // same-key entry won the race; recycle ours
         }
         atomicOr(&T.node_tenants[id], tbit);
         atomicMax(&T.node_ts[(size_t)id * WAVE + selected], now);
@@ -383,7 +417,7 @@ smg_tree_match_insert(GpuTreeDev T, BatchArgs A) {
 // maintenance kernels
 // ---------------------------------------------------------------------------
 extern "C" __global__ void smg_tree_remove_tenant(GpuTreeDev T, int slot_idx) {
-    uint32_t n = *T.next_node;
+    uint32_t n = min(*T.next_node, T.node_cap);
     unsigned long long mask = ~(1ull << slot_idx);
     for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n; i += gridDim.x * blockDim.x) {
         atomicAnd(&T.node_tenants[i], mask);
@@ -394,7 +428,7 @@ extern "C" __global__ void smg_tree_remove_tenant(GpuTreeDev T, int slot_idx) {
 // Clear tenant bits whose stamp is older than `cutoff`; nodes left with no
 // tenants block matches through them (eviction semantics of token_tree.rs).
 extern "C" __global__ void smg_tree_evict_older(GpuTreeDev T, uint32_t cutoff) {
-    uint32_t n = *T.next_node;
+    uint32_t n = min(*T.next_node, T.node_cap);
     uint32_t node = blockIdx.x;  // one wave per node, lane per tenant slot
     int lane = threadIdx.x;
     for (uint32_t i = node; i < n; i += gridDim.x) {
@@ -403,6 +437,35 @@ extern "C" __global__ void smg_tree_evict_older(GpuTreeDev T, uint32_t cutoff) {
         if (!tenants) continue;
         if (((tenants >> lane) & 1ull) && T.node_ts[(size_t)i * WAVE + lane] < cutoff) {
             atomicAnd(&T.node_tenants[i], ~(1ull << lane));
+        }
+    }
+}
+
+// Reclaim nodes whose tenant mask went to zero (evict sweep / tenant removal
+// left them unreachable): tombstone the table slot, clear the LRU row, and
+// push the node id onto the free list for reuse.  One wave per node; runs
+// stream-ordered with match/insert so no extra synchronization is needed.
+// Reference behavior: token_tree.rs eviction frees nodes back to the pool so
+// the tree keeps learning at --max-tree-size.
+extern "C" __global__ void __launch_bounds__(WAVE)
+smg_tree_reclaim(GpuTreeDev T, unsigned long long* out_reclaimed) {
+    uint32_t n = min(*T.next_node, T.node_cap);
+    int lane = threadIdx.x;
+    for (uint32_t i = blockIdx.x; i < n; i += gridDim.x) {
+        if (i == 0) continue;  // root
+        unsigned long long tenants =
+            __hip_atomic_load(&T.node_tenants[i], __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        uint32_t slot = T.node_slot[i];
+        if (tenants != 0ull || slot == NODE_SLOT_NONE) continue;
+        // clear the per-tenant LRU row so a reused node starts cold
+        T.node_ts[(size_t)i * WAVE + lane] = 0;
+        if (lane == 0) {
+            if (__hip_atomic_load(&T.table_vals[slot], __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT) == i)
+                __hip_atomic_store(&T.table_keys[slot], TOMBSTONE_KEY, __ATOMIC_RELEASE,
+                                   __HIP_MEMORY_SCOPE_AGENT);
+            T.node_slot[i] = NODE_SLOT_NONE;
+            tree_free_node(T, i);
+            if (out_reclaimed) atomicAdd(out_reclaimed, 1ull);
         }
     }
 }
@@ -422,7 +485,7 @@ extern "C" __global__ void smg_tree_clear_entries(GpuTreeDev T, const unsigned l
 
 // Count live nodes / per-tenant attributed nodes (stats + tie-breaks).
 extern "C" __global__ void smg_tree_stats(GpuTreeDev T, unsigned long long* out_counts /*65*/) {
-    uint32_t n = *T.next_node;
+    uint32_t n = min(*T.next_node, T.node_cap);
     int lane = threadIdx.x;
     unsigned long long live = 0, mine = 0;
     // grid-stride over nodes; one wave per node, lane = tenant slot
@@ -492,11 +555,15 @@ extern "C" void* smg_gpu_tree_create(int device, uint32_t node_cap, uint32_t tab
     HIP_CHECK(hipMalloc(&D.node_ts, sizeof(uint32_t) * (size_t)node_cap * WAVE));
     HIP_CHECK(hipMalloc(&D.node_parent, sizeof(uint32_t) * node_cap));
     HIP_CHECK(hipMalloc(&D.node_slot, sizeof(uint32_t) * node_cap));
+    HIP_CHECK(hipMalloc(&D.free_list, sizeof(uint32_t) * node_cap));
+    HIP_CHECK(hipMalloc(&D.free_top, sizeof(int)));
     HIP_CHECK(hipMalloc(&D.next_node, sizeof(uint32_t) * 2));
     D.clock_ = D.next_node + 1;
     HIP_CHECK(hipMemset(D.table_keys, 0, sizeof(unsigned long long) * table_size));
     HIP_CHECK(hipMemset(D.node_tenants, 0, sizeof(unsigned long long) * node_cap));
     HIP_CHECK(hipMemset(D.node_ts, 0, sizeof(uint32_t) * (size_t)node_cap * WAVE));
+    HIP_CHECK(hipMemset(D.node_slot, 0xff, sizeof(uint32_t) * node_cap));  // NODE_SLOT_NONE
+    HIP_CHECK(hipMemset(D.free_top, 0, sizeof(int)));
     uint32_t init[2] = {1u, 0u};  // node 0 = root
     HIP_CHECK(hipMemcpy(D.next_node, init, sizeof(init), hipMemcpyHostToDevice));
     // polynomial powers: C^j for the in-page hash, W^p + W^64 for the chain
@@ -533,6 +600,7 @@ extern "C" void smg_gpu_tree_destroy(void* p) {
     hipFree(t->dev.table_keys); hipFree(t->dev.table_vals);
     hipFree(t->dev.node_tenants); hipFree(t->dev.node_ts);
     hipFree(t->dev.node_parent); hipFree(t->dev.node_slot);
+    hipFree(t->dev.free_list); hipFree(t->dev.free_top);
     hipFree(t->dev.next_node);
     hipHostFree(t->h_tokens); hipHostFree(t->h_offsets); hipHostFree(t->h_selected);
     hipHostFree(t->h_matched); hipHostFree(t->h_tenant);
@@ -624,17 +692,32 @@ extern "C" int smg_gpu_tree_evict_older(void* p, uint32_t cutoff) {
     return hipStreamSynchronize(t->stream) == hipSuccess ? 0 : -1;
 }
 
-// out[0..63] per-tenant node counts, out[64] live nodes, out[65] allocated, out[66] clock
-extern "C" int smg_gpu_tree_stats(void* p, unsigned long long* out /*67*/) {
+// Sweep tenant-less nodes back onto the free list; returns the count (or -1).
+extern "C" long long smg_gpu_tree_reclaim(void* p) {
+    GpuTreeHost* t = (GpuTreeHost*)p;
+    hipMemsetAsync(t->d_counts, 0, sizeof(unsigned long long), t->stream);
+    hipLaunchKernelGGL(smg_tree_reclaim, dim3(2048), dim3(WAVE), 0, t->stream, t->dev, t->d_counts);
+    unsigned long long n = 0;
+    hipMemcpyAsync(&n, t->d_counts, sizeof(n), hipMemcpyDeviceToHost, t->stream);
+    if (hipStreamSynchronize(t->stream) != hipSuccess) return -1;
+    return (long long)n;
+}
+
+// out[0..63] per-tenant node counts, out[64] live nodes, out[65] allocated,
+// out[66] clock, out[67] free-list depth
+extern "C" int smg_gpu_tree_stats(void* p, unsigned long long* out /*68*/) {
     GpuTreeHost* t = (GpuTreeHost*)p;
     hipMemsetAsync(t->d_counts, 0, sizeof(unsigned long long) * 65, t->stream);
     hipLaunchKernelGGL(smg_tree_stats, dim3(1024), dim3(WAVE), 0, t->stream, t->dev, t->d_counts);
     hipMemcpyAsync(out, t->d_counts, sizeof(unsigned long long) * 65, hipMemcpyDeviceToHost, t->stream);
     uint32_t counters[2];
+    int free_top = 0;
     hipMemcpyAsync(counters, t->dev.next_node, sizeof(counters), hipMemcpyDeviceToHost, t->stream);
+    hipMemcpyAsync(&free_top, t->dev.free_top, sizeof(int), hipMemcpyDeviceToHost, t->stream);
     if (hipStreamSynchronize(t->stream) != hipSuccess) return -1;
     out[65] = counters[0];
     out[66] = counters[1];
+    out[67] = (unsigned long long)(free_top > 0 ? free_top : 0);
     return 0;
 }
 
@@ -643,6 +726,8 @@ extern "C" int smg_gpu_tree_clear(void* p) {
     hipMemsetAsync(t->dev.table_keys, 0, sizeof(unsigned long long) * t->table_size, t->stream);
     hipMemsetAsync(t->dev.node_tenants, 0, sizeof(unsigned long long) * t->dev.node_cap, t->stream);
     hipMemsetAsync(t->dev.node_ts, 0, sizeof(uint32_t) * (size_t)t->dev.node_cap * WAVE, t->stream);
+    hipMemsetAsync(t->dev.node_slot, 0xff, sizeof(uint32_t) * t->dev.node_cap, t->stream);
+    hipMemsetAsync(t->dev.free_top, 0, sizeof(int), t->stream);
     uint32_t init[2] = {1u, 0u};
     hipMemcpyAsync(t->dev.next_node, init, sizeof(init), hipMemcpyHostToDevice, t->stream);
     return hipStreamSynchronize(t->stream) == hipSuccess ? 0 : -1;

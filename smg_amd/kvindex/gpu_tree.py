@@ -176,19 +176,34 @@ class GpuTokenTree:
             self.slots.release(tenant)
 
     def evict(self, max_nodes: int) -> int:
+        """LRU eviction to `max_nodes` that actually FREES pool memory
+        (reference token_tree.rs eviction to --max-tree-size): progressively
+        older->newer tenant-bit sweeps, each followed by a reclaim pass that
+        tombstones table slots and refills the node free list, so a long-lived
+        gateway keeps learning instead of silently saturating."""
         stats = self._tree.stats()
         live = int(stats["live_nodes"])
-        if live <= max_nodes:
+        # also reclaim when the bump allocator nears the pool cap even if the
+        # live count is modest (nodes stranded by tenant removal / KV events)
+        allocated = int(stats["allocated_nodes"])
+        free = int(stats.get("free_nodes", 0))
+        near_cap = allocated - free >= self.capacity - max(1024, self.capacity // 16)
+        if live <= max_nodes and not near_cap:
             return 0
-        # age-based sweep: drop the oldest half of the clock range
         clock = int(stats["clock"])
-        self._tree.evict_older(max(1, clock // 2))
-        if int(stats["allocated_nodes"]) >= self.capacity - 1024:
-            # pool exhausted: hard reset (reference clears the tree at cap too,
-            # cache_aware.rs hash-index clear)
-            self._tree.clear()
-            self._approx_nodes = 0
-        return live - int(self._tree.stats()["live_nodes"])
+        reclaimed = int(self._tree.reclaim())  # whatever earlier sweeps freed
+        for num in (2, 3, 7):  # clear oldest 1/4, then 1/2, then 7/8 of the clock range
+            self._tree.evict_older(max(1, clock * num // 8))
+            reclaimed += int(self._tree.reclaim())
+            live = int(self._tree.stats()["live_nodes"])
+            if live <= max_nodes:
+                break
+        else:
+            if live > max_nodes:
+                # pathological (all stamps current): last-resort full reset
+                self._tree.clear()
+                self._approx_nodes = 0
+        return reclaimed
 
     def clear(self) -> None:
         self._tree.clear()
