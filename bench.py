@@ -57,6 +57,9 @@ def parse_args():
     p.add_argument("--suffix-len", type=int, default=64)
     p.add_argument("--max-new", type=int, default=32)
     p.add_argument("--tiny", action="store_true", help="tiny model (CPU smoke / CI)")
+    p.add_argument("--model", choices=["mha", "gqa"], default="mha",
+                   help="bench model variant: mha = 16 q/kv heads; gqa = 16 q / 4 kv heads "
+                        "(same ~1.1B total params, Llama3-style group 4)")
     p.add_argument("--kv-fp8", action="store_true",
                    help="opt-in fp8 (e4m3) KV cache; headline default stays bf16")
     p.add_argument("--no-graphs", action="store_true", help="disable hipGraph decode capture")
@@ -92,11 +95,20 @@ def engine_config(args) -> TorchEngineConfig:
         cfg.max_slots = max(8, args.concurrency)
         cfg.max_seq = 256
         return cfg
-    cfg = TorchEngineConfig.bench_1b()
+    gqa = getattr(args, "model", "mha") == "gqa"
+    cfg = TorchEngineConfig.bench_1b_gqa() if gqa else TorchEngineConfig.bench_1b()
     cfg.kv_fp8 = bool(getattr(args, "kv_fp8", False))
     cfg.max_slots = args.concurrency + 8
     cfg.max_seq = args.prefix_len + args.suffix_len + args.max_new + 16
     return cfg
+
+
+def model_label(args) -> str:
+    if args.tiny:
+        return "tiny"
+    if getattr(args, "model", "mha") == "gqa":
+        return "smg-bench-1b-gqa (16L d2048 16q/4kv heads, ffn 6528, bf16, random-init)"
+    return "smg-bench-1b (16L d2048 h16, bf16, random-init)"
 
 
 def worker_main(rank: int, world: int, args, device: str, backend: str):
@@ -196,7 +208,7 @@ def gateway_main(rank: int, world: int, args, device: str, backend: str):
         "dtype": ("bf16 (fp8 kv-cache)" if getattr(args, "kv_fp8", False) else "bf16") if use_gpu else "fp32",
         "data": "synthetic",
         "config": {
-            "model": "smg-bench-1b (16L d2048 h16, bf16, random-init)" if not args.tiny else "tiny",
+            "model": model_label(args),
             "global_batch": args.concurrency * world,
             "seq_len": args.prefix_len + args.suffix_len,
             "parallelism": f"gateway+{world}workers (dp{world}, cache_aware, rccl-xgmi)",

@@ -1,24 +1,33 @@
 // Fused single-token (decode) attention for the GPU worker engine — gfx950.
 //
-// One wave per (slot, head): online-softmax over the slot's live KV window
+// One wave per (slot, KV head): online-softmax over the slot's live KV window
 // (kv[slot][:pos+1]); idle/short slots cost nothing, unlike sdpa's
 // rectangular [slots, maxlen] window.
 //
-// v5 layout notes (measured on MI355X):
+// v7 adds GQA: the wave carries all G = n_heads/n_kv_heads query heads of the
+// group, so each K/V row is streamed from HBM ONCE and dotted/FMAed against G
+// q vectors — decode attention is bandwidth-bound (~80% of achievable HBM BW
+// at MHA), so GQA cuts its device time by ~G.  G in {1,2,4,8} is a template
+// parameter (full unrolling); G=1 reproduces the v6 MHA kernel exactly.
+//
+// v5/v6 layout notes (measured on MI355X):
 //   * K dot phase: each lane streams ITS OWN timestep's K row with dwordx4
 //     loads — a row is 256 B = 4 consecutive cache lines, so per-lane
 //     streaming is line-efficient, and 64 lanes x 8 wide loads keep >500
 //     lines in flight with no LDS round-trip or barrier;
-//   * P·V phase: probabilities via LDS broadcast, V rows loaded 4 B/lane
-//     coalesced, no branch between shfl and load so loads pipeline;
-//   * zero LDS tiles -> no occupancy cap from shared memory.
+//   * P·V phase: probabilities via LDS broadcast, V rows loaded dwordx4 with
+//     a lane->(row-group, chunk) mapping so one wave instruction covers 4
+//     rows (1 KB); unrolled full-tile path; shfl_xor row-group fold;
+//   * q staged in LDS (wave-uniform broadcast reads), no LDS K/V tiles ->
+//     no occupancy cap from shared memory.
 //
 // Layout contract (the engine's KV arena, one layer):
-//   K, V:  [n_slots, n_heads, max_seq, head_dim]  bf16, contiguous
-//   q:     [n_slots, n_heads, head_dim]           bf16, contiguous
+//   K, V:  [n_slots, n_kv_heads, max_seq, head_dim]  bf16 or fp8-e4m3
+//   q:     [n_slots, n_heads, head_dim]              bf16, contiguous
 //   pos:   [n_slots] int32 — attend kpos <= pos[slot]
-//   out:   [n_slots, n_heads, head_dim]           bf16
-// head_dim <= 128 and divisible by 8.
+//   out:   [n_slots, n_heads, head_dim]              bf16
+// head_dim <= 128, divisible by 8, power-of-two chunk count; q heads
+// kvh*G .. kvh*G+G-1 share KV head kvh (repeat_interleave convention).
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
 
@@ -46,58 +55,59 @@ __device__ __forceinline__ void fp8x4_to_f32(unsigned int w, float* o) {
 // fp8 (OCP e4m3) KV-cache variant support: gfx950 has native fp8<->f32
 // converts (v_cvt_f32_fp8); the cache is stored quantized by the fused
 // rope/KV-store kernels and dequantized inline here — halving the HBM bytes
-// the decode-attention streams per step.
-template <bool KV8>
+// the decode-attention streams per step (composes with GQA's 1/G).
+template <bool KV8, int G>
 __global__ void __launch_bounds__(WAVE) smg_attn_decode_t(
     const __hip_bfloat16* __restrict__ q,
     const void* __restrict__ k,
     const void* __restrict__ v,
     const int* __restrict__ pos,
     __hip_bfloat16* __restrict__ out,
-    int n_slots, int n_heads, int max_seq, int head_dim, float scale) {
+    int n_slots, int n_heads, int n_kv_heads, int max_seq, int head_dim, float scale) {
     int sh = blockIdx.x;
-    int slot = sh / n_heads;
-    int head = sh % n_heads;
+    int slot = sh / n_kv_heads;
+    int kvh = sh % n_kv_heads;
     if (slot >= n_slots) return;
     int lane = threadIdx.x;
     int T = pos[slot] + 1;  // inclusive current position
     if (T > max_seq) T = max_seq;
 
-    const size_t head_base = ((size_t)slot * n_heads + head) * (size_t)max_seq * head_dim;
+    const size_t head_base = ((size_t)slot * n_kv_heads + kvh) * (size_t)max_seq * head_dim;
     const __hip_bfloat16* kh = (const __hip_bfloat16*)k + head_base;
     const __hip_bfloat16* vh = (const __hip_bfloat16*)v + head_base;
     const unsigned char* kh8 = (const unsigned char*)k + head_base;
     const unsigned char* vh8 = (const unsigned char*)v + head_base;
-    const __hip_bfloat16* qh = q + ((size_t)slot * n_heads + head) * head_dim;
+    // the group's G query rows are contiguous: heads kvh*G .. kvh*G+G-1
+    const __hip_bfloat16* qh = q + ((size_t)slot * n_heads + (size_t)kvh * G) * head_dim;
 
-    // q staged once in LDS as f32 (read by every lane's dot)
-    __shared__ float s_q[128];
-    __shared__ float s_p[WAVE];
-    for (int i = lane; i < head_dim; i += WAVE) s_q[i] = (float)qh[i];
+    // q staged once in LDS as f32 (wave-uniform broadcast reads in the dot)
+    __shared__ float s_q[G * 128];
+    __shared__ float s_p[G][WAVE];
+    for (int i = lane; i < G * head_dim; i += WAVE) s_q[i] = (float)qh[i];
     __syncthreads();
 
     const int vec_n = head_dim / 8;  // dwordx4 chunks per row (<=16)
-    // v6 P·V mapping: lane -> (row-group rgrp = lane/chunks, chunk = lane%chunks)
-    // so one wave instruction loads dwordx4 from `rows_per` different V rows —
-    // 1 KB per instruction (vs 256 B with the old 4 B/lane scheme), matching
-    // the K phase's streaming width.  Each lane accumulates 8 dims of its
-    // chunk over rows rgrp, rgrp+rows_per, ...; a log2(rows_per)-step
-    // shfl_xor tree folds the row groups at the end.
     const int chunks = vec_n;             // dwordx4 chunks per row (8 or 16)
     const int rows_per = WAVE / chunks;   // rows covered per instruction
     const int chunk = lane % chunks;
     const int rgrp = lane / chunks;
-    float accv[8];
+    float accv[G][8];
 #pragma unroll
-    for (int j = 0; j < 8; ++j) accv[j] = 0.f;
-    float m = -1e30f, l = 0.f;
+    for (int g = 0; g < G; ++g)
+#pragma unroll
+        for (int j = 0; j < 8; ++j) accv[g][j] = 0.f;
+    float m[G], l[G];
+#pragma unroll
+    for (int g = 0; g < G; ++g) { m[g] = -1e30f; l[g] = 0.f; }
 
     for (int base = 0; base < T; base += WAVE) {
         int lim = min(WAVE, T - base);
         int t = base + lane;
-        float score = -1e30f;
+        float d[G];
+#pragma unroll
+        for (int g = 0; g < G; ++g) d[g] = 0.f;
         if (t < T) {
-            float d = 0.f;
+            // stream this lane's K row ONCE; dot against all G q vectors
             if constexpr (KV8) {
                 const uint4* row = (const uint4*)(kh8 + (size_t)t * head_dim);
 #pragma unroll 4
@@ -110,7 +120,12 @@ __global__ void __launch_bounds__(WAVE) smg_attn_decode_t(
                     for (int wi = 0; wi < 4; ++wi) {
                         fp8x4_to_f32(wp[wi], dec);
 #pragma unroll
-                        for (int s = 0; s < 4; ++s) d += s_q[ib + wi * 4 + s] * dec[s];
+                        for (int s = 0; s < 4; ++s) {
+                            const float kv = dec[s];
+                            const int di = ib + wi * 4 + s;
+#pragma unroll
+                            for (int g = 0; g < G; ++g) d[g] += s_q[g * head_dim + di] * kv;
+                        }
                     }
                 }
             } else {
@@ -121,30 +136,38 @@ __global__ void __launch_bounds__(WAVE) smg_attn_decode_t(
                     const unsigned short* hsp = (const unsigned short*)&w;
                     int ib = c * 8;
 #pragma unroll
-                    for (int j = 0; j < 8; ++j) d += s_q[ib + j] * bf16_to_f32(hsp[j]);
+                    for (int j = 0; j < 8; ++j) {
+                        const float kv = bf16_to_f32(hsp[j]);
+#pragma unroll
+                        for (int g = 0; g < G; ++g) d[g] += s_q[g * head_dim + ib + j] * kv;
+                    }
                 }
             }
-            score = d * scale;
         }
-        // online softmax across the wave's scores
-        float mr = score;
+        // online softmax per query head across the wave's scores
+        float alpha[G];
 #pragma unroll
-        for (int off = 32; off > 0; off >>= 1) mr = fmaxf(mr, __shfl_xor(mr, off, WAVE));
-        float m_new = fmaxf(m, mr);
-        float alpha = __expf(m - m_new);
-        float p = (t < T) ? __expf(score - m_new) : 0.f;
-        s_p[lane] = p;
-        float pr = p;
+        for (int g = 0; g < G; ++g) {
+            float score = (t < T) ? d[g] * scale : -1e30f;
+            float mr = score;
 #pragma unroll
-        for (int off = 32; off > 0; off >>= 1) pr += __shfl_xor(pr, off, WAVE);
-        l = l * alpha + pr;
+            for (int off = 32; off > 0; off >>= 1) mr = fmaxf(mr, __shfl_xor(mr, off, WAVE));
+            float m_new = fmaxf(m[g], mr);
+            alpha[g] = __expf(m[g] - m_new);
+            float p = (t < T) ? __expf(score - m_new) : 0.f;
+            s_p[g][lane] = p;
+            float pr = p;
 #pragma unroll
-        for (int j = 0; j < 8; ++j) accv[j] *= alpha;
+            for (int off = 32; off > 0; off >>= 1) pr += __shfl_xor(pr, off, WAVE);
+            l[g] = l[g] * alpha[g] + pr;
+            m[g] = m_new;
+#pragma unroll
+            for (int j = 0; j < 8; ++j) accv[g][j] *= alpha[g];
+        }
         __syncthreads();
         // P·V: each lane streams dwordx4 of its chunk from every rows_per-th
-        // row.  Full tiles take the unrolled fast path (compile-time trip
-        // count exposes all 16 loads to the scheduler); only the final
-        // partial tile pays the runtime loop.
+        // row — ONE V load feeds all G accumulators.  Full tiles take the
+        // unrolled fast path; only the final partial tile pays the runtime loop.
         if constexpr (KV8) {
             const unsigned char* vtile8 = vh8 + (size_t)base * head_dim + chunk * 8;
             if (lim == WAVE && head_dim == 128) {
@@ -152,26 +175,28 @@ __global__ void __launch_bounds__(WAVE) smg_attn_decode_t(
                 for (int it = 0; it < 16; ++it) {
                     const int j = rgrp + it * 4;
                     const uint2 w = *(const uint2*)(vtile8 + (size_t)j * head_dim);
-                    float pj = s_p[j];
-                    float dec[4];
+                    float dec[8];
                     fp8x4_to_f32(w.x, dec);
+                    fp8x4_to_f32(w.y, dec + 4);
 #pragma unroll
-                    for (int s = 0; s < 4; ++s) accv[s] += pj * dec[s];
-                    fp8x4_to_f32(w.y, dec);
+                    for (int g = 0; g < G; ++g) {
+                        const float pj = s_p[g][j];
 #pragma unroll
-                    for (int s = 0; s < 4; ++s) accv[4 + s] += pj * dec[s];
+                        for (int s = 0; s < 8; ++s) accv[g][s] += pj * dec[s];
+                    }
                 }
             } else {
                 for (int j = rgrp; j < lim; j += rows_per) {
                     const uint2 w = *(const uint2*)(vtile8 + (size_t)j * head_dim);
-                    float pj = s_p[j];
-                    float dec[4];
+                    float dec[8];
                     fp8x4_to_f32(w.x, dec);
+                    fp8x4_to_f32(w.y, dec + 4);
 #pragma unroll
-                    for (int s = 0; s < 4; ++s) accv[s] += pj * dec[s];
-                    fp8x4_to_f32(w.y, dec);
+                    for (int g = 0; g < G; ++g) {
+                        const float pj = s_p[g][j];
 #pragma unroll
-                    for (int s = 0; s < 4; ++s) accv[4 + s] += pj * dec[s];
+                        for (int s = 0; s < 8; ++s) accv[g][s] += pj * dec[s];
+                    }
                 }
             }
         } else {
@@ -182,55 +207,95 @@ __global__ void __launch_bounds__(WAVE) smg_attn_decode_t(
                     const int j = rgrp + it * 4;  // rows_per == 4 when chunks == 16
                     const uint4 w = *(const uint4*)(vtile + (size_t)j * head_dim);
                     const unsigned short* hsp = (const unsigned short*)&w;
-                    float pj = s_p[j];
 #pragma unroll
-                    for (int jj = 0; jj < 8; ++jj) accv[jj] += pj * bf16_to_f32(hsp[jj]);
+                    for (int g = 0; g < G; ++g) {
+                        const float pj = s_p[g][j];
+#pragma unroll
+                        for (int jj = 0; jj < 8; ++jj) accv[g][jj] += pj * bf16_to_f32(hsp[jj]);
+                    }
                 }
             } else {
                 for (int j = rgrp; j < lim; j += rows_per) {
                     const uint4 w = *(const uint4*)(vtile + (size_t)j * head_dim);
                     const unsigned short* hsp = (const unsigned short*)&w;
-                    float pj = s_p[j];
 #pragma unroll
-                    for (int jj = 0; jj < 8; ++jj) accv[jj] += pj * bf16_to_f32(hsp[jj]);
+                    for (int g = 0; g < G; ++g) {
+                        const float pj = s_p[g][j];
+#pragma unroll
+                        for (int jj = 0; jj < 8; ++jj) accv[g][jj] += pj * bf16_to_f32(hsp[jj]);
+                    }
                 }
             }
         }
         __syncthreads();
-        m = m_new;
     }
     // fold the row groups: lanes sharing `chunk` differ in bits >= log2(chunks)
-    for (int off = chunks; off < WAVE; off <<= 1) {
 #pragma unroll
-        for (int jj = 0; jj < 8; ++jj) accv[jj] += __shfl_xor(accv[jj], off, WAVE);
+    for (int g = 0; g < G; ++g) {
+        for (int off = chunks; off < WAVE; off <<= 1) {
+#pragma unroll
+            for (int jj = 0; jj < 8; ++jj) accv[g][jj] += __shfl_xor(accv[g][jj], off, WAVE);
+        }
     }
-    float inv = l > 0.f ? 1.f / l : 0.f;
-    __hip_bfloat16* orow = out + ((size_t)slot * n_heads + head) * head_dim;
     if (rgrp == 0) {
 #pragma unroll
-        for (int jj = 0; jj < 8; ++jj) orow[chunk * 8 + jj] = (__hip_bfloat16)(accv[jj] * inv);
+        for (int g = 0; g < G; ++g) {
+            float inv = l[g] > 0.f ? 1.f / l[g] : 0.f;
+            __hip_bfloat16* orow =
+                out + ((size_t)slot * n_heads + (size_t)kvh * G + g) * head_dim;
+#pragma unroll
+            for (int jj = 0; jj < 8; ++jj) orow[chunk * 8 + jj] = (__hip_bfloat16)(accv[g][jj] * inv);
+        }
     }
+}
+
+template <bool KV8>
+static int launch_g(const void* q, const void* k, const void* v, const void* pos, void* out,
+                    int n_slots, int n_heads, int n_kv_heads, int max_seq, int head_dim,
+                    float scale, hipStream_t stream) {
+    dim3 grid(n_slots * n_kv_heads);
+    const int G = n_heads / n_kv_heads;
+#define LAUNCH_CASE(GG)                                                                        \
+    case GG:                                                                                   \
+        hipLaunchKernelGGL((smg_attn_decode_t<KV8, GG>), grid, dim3(WAVE), 0, stream,          \
+                           (const __hip_bfloat16*)q, k, v, (const int*)pos,                    \
+                           (__hip_bfloat16*)out, n_slots, n_heads, n_kv_heads, max_seq,        \
+                           head_dim, scale);                                                   \
+        break;
+    switch (G) {
+        LAUNCH_CASE(1)
+        LAUNCH_CASE(2)
+        LAUNCH_CASE(4)
+        LAUNCH_CASE(8)
+        default:
+            return -4;
+    }
+#undef LAUNCH_CASE
+    return hipGetLastError() == hipSuccess ? 0 : -2;
+}
+
+extern "C" int smg_attn_decode_launch_gqa(const void* q, const void* k, const void* v,
+                                          const void* pos, void* out, int n_slots, int n_heads,
+                                          int n_kv_heads, int max_seq, int head_dim, float scale,
+                                          void* stream, int kv_fp8) {
+    if (head_dim > 128 || (head_dim & 7)) return -1;
+    int chunks = head_dim / 8;  // v6 P·V row-group mapping needs 2^k chunks
+    if (chunks & (chunks - 1)) return -1;
+    if (kv_fp8 && (head_dim & 15)) return -1;  // fp8 K streaming is 16-wide
+    if (n_kv_heads <= 0 || n_heads % n_kv_heads) return -4;
+    if (kv_fp8)
+        return launch_g<true>(q, k, v, pos, out, n_slots, n_heads, n_kv_heads, max_seq, head_dim,
+                              scale, (hipStream_t)stream);
+    return launch_g<false>(q, k, v, pos, out, n_slots, n_heads, n_kv_heads, max_seq, head_dim,
+                           scale, (hipStream_t)stream);
 }
 
 extern "C" int smg_attn_decode_launch_ex(const void* q, const void* k, const void* v,
                                          const void* pos, void* out, int n_slots, int n_heads,
                                          int max_seq, int head_dim, float scale, void* stream,
                                          int kv_fp8) {
-    if (head_dim > 128 || (head_dim & 7)) return -1;
-    int chunks = head_dim / 8;  // v6 P·V row-group mapping needs 2^k chunks
-    if (chunks & (chunks - 1)) return -1;
-    if (kv_fp8 && (head_dim & 15)) return -1;  // fp8 K streaming is 16-wide
-    dim3 grid(n_slots * n_heads);
-    if (kv_fp8) {
-        hipLaunchKernelGGL(smg_attn_decode_t<true>, grid, dim3(WAVE), 0, (hipStream_t)stream,
-                           (const __hip_bfloat16*)q, k, v, (const int*)pos,
-                           (__hip_bfloat16*)out, n_slots, n_heads, max_seq, head_dim, scale);
-    } else {
-        hipLaunchKernelGGL(smg_attn_decode_t<false>, grid, dim3(WAVE), 0, (hipStream_t)stream,
-                           (const __hip_bfloat16*)q, k, v, (const int*)pos,
-                           (__hip_bfloat16*)out, n_slots, n_heads, max_seq, head_dim, scale);
-    }
-    return hipGetLastError() == hipSuccess ? 0 : -2;
+    return smg_attn_decode_launch_gqa(q, k, v, pos, out, n_slots, n_heads, n_heads, max_seq,
+                                      head_dim, scale, stream, kv_fp8);
 }
 
 extern "C" int smg_attn_decode_launch(const void* q, const void* k, const void* v,

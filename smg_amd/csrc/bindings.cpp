@@ -52,6 +52,18 @@ int smg_img_resize_normalize(void* p, const uint8_t* in, int in_w, int in_h, int
 int smg_attn_decode_launch(const void* q, const void* k, const void* v, const void* pos,
                            void* out, int n_slots, int n_heads, int max_seq, int head_dim,
                            float scale, void* stream);
+int smg_attn_decode_launch_gqa(const void* q, const void* k, const void* v, const void* pos,
+                               void* out, int n_slots, int n_heads, int n_kv_heads, int max_seq,
+                               int head_dim, float scale, void* stream, int kv_fp8);
+int smg_rope_kv_store_launch_gqa(const void* qkv, const void* freqs, const void* pos,
+                                 void* k_cache, void* v_cache, void* q_out, int n_slots,
+                                 int n_heads, int n_kv_heads, int max_seq, int head_dim,
+                                 void* stream, int kv_fp8);
+int smg_rope_prefill_launch_gqa(const void* qkv, const void* freqs, const void* slots,
+                                const void* starts, void* k_cache, void* v_cache, void* q_out,
+                                void* k_out, void* v_out, int B, int L, int n_heads,
+                                int n_kv_heads, int max_seq, int head_dim, void* stream,
+                                int kv_fp8);
 int smg_attn_decode_launch_ex(const void* q, const void* k, const void* v, const void* pos,
                               void* out, int n_slots, int n_heads, int max_seq, int head_dim,
                               float scale, void* stream, int kv_fp8);
@@ -345,43 +357,48 @@ PYBIND11_MODULE(_core, m) {
     // decode attention: raw device pointers + stream (ints from torch)
     m.def("attn_decode",
           [](uintptr_t q, uintptr_t k, uintptr_t v, uintptr_t pos, uintptr_t out, int n_slots,
-             int n_heads, int max_seq, int head_dim, float scale, uintptr_t stream, int kv_fp8) {
-              int rc = smg_attn_decode_launch_ex((const void*)q, (const void*)k, (const void*)v,
-                                                 (const void*)pos, (void*)out, n_slots, n_heads,
-                                                 max_seq, head_dim, scale, (void*)stream, kv_fp8);
+             int n_heads, int max_seq, int head_dim, float scale, uintptr_t stream, int kv_fp8,
+             int n_kv_heads) {
+              if (n_kv_heads <= 0) n_kv_heads = n_heads;
+              int rc = smg_attn_decode_launch_gqa((const void*)q, (const void*)k, (const void*)v,
+                                                  (const void*)pos, (void*)out, n_slots, n_heads,
+                                                  n_kv_heads, max_seq, head_dim, scale,
+                                                  (void*)stream, kv_fp8);
               if (rc != 0) throw std::runtime_error("attn_decode launch failed rc=" + std::to_string(rc));
           },
           py::arg("q"), py::arg("k"), py::arg("v"), py::arg("pos"), py::arg("out"),
           py::arg("n_slots"), py::arg("n_heads"), py::arg("max_seq"), py::arg("head_dim"),
-          py::arg("scale"), py::arg("stream"), py::arg("kv_fp8") = 0);
+          py::arg("scale"), py::arg("stream"), py::arg("kv_fp8") = 0, py::arg("n_kv_heads") = 0);
     // fused rope + KV-store + q-pack (fused_decode.hip)
     m.def("rope_kv_store",
           [](uintptr_t qkv, uintptr_t freqs, uintptr_t pos, uintptr_t k_cache, uintptr_t v_cache,
              uintptr_t q_out, int n_slots, int n_heads, int max_seq, int head_dim, uintptr_t stream,
-             int kv_fp8) {
-              int rc = smg_rope_kv_store_launch_ex((const void*)qkv, (const void*)freqs,
-                                                   (const void*)pos, (void*)k_cache, (void*)v_cache,
-                                                   (void*)q_out, n_slots, n_heads, max_seq, head_dim,
-                                                   (void*)stream, kv_fp8);
+             int kv_fp8, int n_kv_heads) {
+              if (n_kv_heads <= 0) n_kv_heads = n_heads;
+              int rc = smg_rope_kv_store_launch_gqa((const void*)qkv, (const void*)freqs,
+                                                    (const void*)pos, (void*)k_cache, (void*)v_cache,
+                                                    (void*)q_out, n_slots, n_heads, n_kv_heads,
+                                                    max_seq, head_dim, (void*)stream, kv_fp8);
               if (rc != 0) throw std::runtime_error("rope_kv_store launch failed rc=" + std::to_string(rc));
           },
           py::arg("qkv"), py::arg("freqs"), py::arg("pos"), py::arg("k_cache"), py::arg("v_cache"),
           py::arg("q_out"), py::arg("n_slots"), py::arg("n_heads"), py::arg("max_seq"),
-          py::arg("head_dim"), py::arg("stream"), py::arg("kv_fp8") = 0);
+          py::arg("head_dim"), py::arg("stream"), py::arg("kv_fp8") = 0, py::arg("n_kv_heads") = 0);
     m.def("rope_prefill",
           [](uintptr_t qkv, uintptr_t freqs, uintptr_t slots, uintptr_t starts, uintptr_t k_cache,
              uintptr_t v_cache, uintptr_t q_out, uintptr_t k_out, uintptr_t v_out, int B, int L,
-             int n_heads, int max_seq, int head_dim, uintptr_t stream, int kv_fp8) {
-              int rc = smg_rope_prefill_launch_ex(
+             int n_heads, int max_seq, int head_dim, uintptr_t stream, int kv_fp8, int n_kv_heads) {
+              if (n_kv_heads <= 0) n_kv_heads = n_heads;
+              int rc = smg_rope_prefill_launch_gqa(
                   (const void*)qkv, (const void*)freqs, (const void*)slots, (const void*)starts,
                   (void*)k_cache, (void*)v_cache, (void*)q_out, (void*)k_out, (void*)v_out, B, L,
-                  n_heads, max_seq, head_dim, (void*)stream, kv_fp8);
+                  n_heads, n_kv_heads, max_seq, head_dim, (void*)stream, kv_fp8);
               if (rc != 0) throw std::runtime_error("rope_prefill launch failed rc=" + std::to_string(rc));
           },
           py::arg("qkv"), py::arg("freqs"), py::arg("slots"), py::arg("starts"), py::arg("k_cache"),
           py::arg("v_cache"), py::arg("q_out"), py::arg("k_out"), py::arg("v_out"), py::arg("B"),
           py::arg("L"), py::arg("n_heads"), py::arg("max_seq"), py::arg("head_dim"),
-          py::arg("stream"), py::arg("kv_fp8") = 0);
+          py::arg("stream"), py::arg("kv_fp8") = 0, py::arg("n_kv_heads") = 0);
     m.def("lse_merge",
           [](uintptr_t o1, uintptr_t o2, uintptr_t lse1, uintptr_t lse2, uintptr_t out,
              long long rows, int head_dim, uintptr_t stream) {

@@ -37,6 +37,10 @@ __device__ __forceinline__ void store_pair_fp8(unsigned char* dst, float a, floa
     *(unsigned short*)dst = (unsigned short)(packed & 0xFFFFu);
 }
 
+// GQA: qkv rows are [q(D) | k(KD) | v(KD)] with D = n_heads*hd and
+// KD = n_kv_heads*hd.  One wave per (slot, q head); the wave whose head id is
+// < n_kv_heads also ropes+stores that KV head's k/v (q heads kvh*G..kvh*G+G-1
+// share KV head kvh — repeat_interleave convention, matched by attn_decode).
 template <bool KV8>
 __global__ void __launch_bounds__(WAVE) smg_rope_kv_store_t(
     const __hip_bfloat16* __restrict__ qkv,
@@ -45,7 +49,7 @@ __global__ void __launch_bounds__(WAVE) smg_rope_kv_store_t(
     void* __restrict__ k_cache,
     void* __restrict__ v_cache,
     __hip_bfloat16* __restrict__ q_out,
-    int n_slots, int n_heads, int max_seq, int head_dim) {
+    int n_slots, int n_heads, int n_kv_heads, int max_seq, int head_dim) {
     const int sh = blockIdx.x;
     const int slot = sh / n_heads;
     const int head = sh % n_heads;
@@ -54,13 +58,15 @@ __global__ void __launch_bounds__(WAVE) smg_rope_kv_store_t(
     const int p = pos[slot];
     const int pairs = head_dim >> 1;
     const int D = n_heads * head_dim;
+    const int KD = n_kv_heads * head_dim;
+    const bool do_kv = head < n_kv_heads;
 
-    const __hip_bfloat16* qrow = qkv + (size_t)slot * 3 * D + (size_t)head * head_dim;
-    const __hip_bfloat16* krow = qrow + D;
-    const __hip_bfloat16* vrow = qrow + 2 * D;
+    const __hip_bfloat16* qrow = qkv + (size_t)slot * (D + 2 * KD) + (size_t)head * head_dim;
+    const __hip_bfloat16* krow = qkv + (size_t)slot * (D + 2 * KD) + D + (size_t)head * head_dim;
+    const __hip_bfloat16* vrow = krow + KD;
     const float2* f = freqs + (size_t)p * pairs;
 
-    const size_t row_off = ((size_t)slot * n_heads + head) * max_seq + p;
+    const size_t row_off = ((size_t)slot * n_kv_heads + head) * max_seq + p;
     __hip_bfloat16* kdst = (__hip_bfloat16*)k_cache + row_off * head_dim;
     __hip_bfloat16* vdst = (__hip_bfloat16*)v_cache + row_off * head_dim;
     unsigned char* kdst8 = (unsigned char*)k_cache + row_off * head_dim;
@@ -74,6 +80,7 @@ __global__ void __launch_bounds__(WAVE) smg_rope_kv_store_t(
         const float q1 = __bfloat162float(qrow[e + 1]);
         qdst[e] = __float2bfloat16(q0 * cs.x - q1 * cs.y);
         qdst[e + 1] = __float2bfloat16(q0 * cs.y + q1 * cs.x);
+        if (!do_kv) continue;
         const float k0 = __bfloat162float(krow[e]);
         const float k1 = __bfloat162float(krow[e + 1]);
         const float kr0 = k0 * cs.x - k1 * cs.y;
@@ -116,16 +123,16 @@ extern "C" __global__ void __launch_bounds__(256) smg_silu_mul(
 // 2x advanced-index KV scatter, and the implicit transpose-contiguous copies.
 template <bool KV8>
 __global__ void __launch_bounds__(WAVE) smg_rope_prefill_t(
-    const __hip_bfloat16* __restrict__ qkv,   // [B, L, 3*D]
+    const __hip_bfloat16* __restrict__ qkv,   // [B, L, D + 2*KD]
     const float2* __restrict__ freqs,         // [max_seq, hd/2]
     const int* __restrict__ slots,            // [B]
     const int* __restrict__ starts,           // [B]
-    void* __restrict__ k_cache,               // [n_slots, H, max_seq, hd]
+    void* __restrict__ k_cache,               // [n_slots, KVH, max_seq, hd]
     void* __restrict__ v_cache,
     __hip_bfloat16* __restrict__ q_out,       // [B, H, L, hd] (always bf16)
-    __hip_bfloat16* __restrict__ k_out,
-    __hip_bfloat16* __restrict__ v_out,
-    int B, int L, int n_heads, int max_seq, int head_dim) {
+    __hip_bfloat16* __restrict__ k_out,       // [B, KVH, L, hd]
+    __hip_bfloat16* __restrict__ v_out,       // [B, KVH, L, hd]
+    int B, int L, int n_heads, int n_kv_heads, int max_seq, int head_dim) {
     const int idx = blockIdx.x;
     const int h = idx % n_heads;
     const int bt = idx / n_heads;
@@ -135,23 +142,25 @@ __global__ void __launch_bounds__(WAVE) smg_rope_prefill_t(
     const int lane = threadIdx.x;
     const int pairs = head_dim >> 1;
     const int D = n_heads * head_dim;
+    const int KD = n_kv_heads * head_dim;
+    const bool do_kv = h < n_kv_heads;
     const int p = starts[b] + t;
     const int slot = slots[b];
 
-    const __hip_bfloat16* qrow = qkv + ((size_t)b * L + t) * 3 * D + (size_t)h * head_dim;
-    const __hip_bfloat16* krow = qrow + D;
-    const __hip_bfloat16* vrow = qrow + 2 * D;
+    const __hip_bfloat16* qrow = qkv + ((size_t)b * L + t) * (D + 2 * KD) + (size_t)h * head_dim;
+    const __hip_bfloat16* krow = qkv + ((size_t)b * L + t) * (D + 2 * KD) + D + (size_t)h * head_dim;
+    const __hip_bfloat16* vrow = krow + KD;
     const float2* f = freqs + (size_t)p * pairs;
 
-    const size_t crow = ((size_t)slot * n_heads + h) * max_seq + p;
+    const size_t crow = ((size_t)slot * n_kv_heads + h) * max_seq + p;
     __hip_bfloat16* kc = (__hip_bfloat16*)k_cache + crow * head_dim;
     __hip_bfloat16* vc = (__hip_bfloat16*)v_cache + crow * head_dim;
     unsigned char* kc8 = (unsigned char*)k_cache + crow * head_dim;
     unsigned char* vc8 = (unsigned char*)v_cache + crow * head_dim;
-    const size_t out_off = (((size_t)b * n_heads + h) * L + t) * head_dim;
-    __hip_bfloat16* qo = q_out + out_off;
-    __hip_bfloat16* ko = k_out + out_off;
-    __hip_bfloat16* vo = v_out + out_off;
+    __hip_bfloat16* qo = q_out + (((size_t)b * n_heads + h) * L + t) * head_dim;
+    const size_t kv_off = (((size_t)b * n_kv_heads + h) * L + t) * head_dim;
+    __hip_bfloat16* ko = k_out + kv_off;
+    __hip_bfloat16* vo = v_out + kv_off;
 
     for (int i = lane; i < pairs; i += WAVE) {
         const float2 cs = f[i];
@@ -160,6 +169,7 @@ __global__ void __launch_bounds__(WAVE) smg_rope_prefill_t(
         const float q1 = __bfloat162float(qrow[e + 1]);
         qo[e] = __float2bfloat16(q0 * cs.x - q1 * cs.y);
         qo[e + 1] = __float2bfloat16(q0 * cs.y + q1 * cs.x);
+        if (!do_kv) continue;
         const float k0 = __bfloat162float(krow[e]);
         const float k1 = __bfloat162float(krow[e + 1]);
         const float krf0 = k0 * cs.x - k1 * cs.y;
@@ -184,26 +194,37 @@ __global__ void __launch_bounds__(WAVE) smg_rope_prefill_t(
     }
 }
 
-extern "C" int smg_rope_prefill_launch_ex(
+extern "C" int smg_rope_prefill_launch_gqa(
     const void* qkv, const void* freqs, const void* slots, const void* starts,
     void* k_cache, void* v_cache, void* q_out, void* k_out, void* v_out,
-    int B, int L, int n_heads, int max_seq, int head_dim, void* stream, int kv_fp8) {
+    int B, int L, int n_heads, int n_kv_heads, int max_seq, int head_dim, void* stream,
+    int kv_fp8) {
     if (head_dim % 2 != 0 || head_dim > 256) return 1;
+    if (n_kv_heads <= 0 || n_heads % n_kv_heads) return 2;
     dim3 grid((unsigned)B * L * n_heads);
     if (kv_fp8) {
         hipLaunchKernelGGL(smg_rope_prefill_t<true>, grid, dim3(WAVE), 0, (hipStream_t)stream,
                            (const __hip_bfloat16*)qkv, (const float2*)freqs, (const int*)slots,
                            (const int*)starts, k_cache, v_cache, (__hip_bfloat16*)q_out,
                            (__hip_bfloat16*)k_out, (__hip_bfloat16*)v_out,
-                           B, L, n_heads, max_seq, head_dim);
+                           B, L, n_heads, n_kv_heads, max_seq, head_dim);
     } else {
         hipLaunchKernelGGL(smg_rope_prefill_t<false>, grid, dim3(WAVE), 0, (hipStream_t)stream,
                            (const __hip_bfloat16*)qkv, (const float2*)freqs, (const int*)slots,
                            (const int*)starts, k_cache, v_cache, (__hip_bfloat16*)q_out,
                            (__hip_bfloat16*)k_out, (__hip_bfloat16*)v_out,
-                           B, L, n_heads, max_seq, head_dim);
+                           B, L, n_heads, n_kv_heads, max_seq, head_dim);
     }
     return (int)hipGetLastError();
+}
+
+extern "C" int smg_rope_prefill_launch_ex(
+    const void* qkv, const void* freqs, const void* slots, const void* starts,
+    void* k_cache, void* v_cache, void* q_out, void* k_out, void* v_out,
+    int B, int L, int n_heads, int max_seq, int head_dim, void* stream, int kv_fp8) {
+    return smg_rope_prefill_launch_gqa(qkv, freqs, slots, starts, k_cache, v_cache, q_out,
+                                       k_out, v_out, B, L, n_heads, n_heads, max_seq, head_dim,
+                                       stream, kv_fp8);
 }
 
 extern "C" int smg_rope_prefill_launch(
@@ -214,24 +235,34 @@ extern "C" int smg_rope_prefill_launch(
                                       k_out, v_out, B, L, n_heads, max_seq, head_dim, stream, 0);
 }
 
-extern "C" int smg_rope_kv_store_launch_ex(
+extern "C" int smg_rope_kv_store_launch_gqa(
     const void* qkv, const void* freqs, const void* pos,
     void* k_cache, void* v_cache, void* q_out,
-    int n_slots, int n_heads, int max_seq, int head_dim, void* stream, int kv_fp8) {
+    int n_slots, int n_heads, int n_kv_heads, int max_seq, int head_dim, void* stream,
+    int kv_fp8) {
     if (head_dim % 2 != 0 || head_dim > 256) return 1;
+    if (n_kv_heads <= 0 || n_heads % n_kv_heads) return 2;
     dim3 grid(n_slots * n_heads);
     if (kv_fp8) {
         hipLaunchKernelGGL(smg_rope_kv_store_t<true>, grid, dim3(WAVE), 0, (hipStream_t)stream,
                            (const __hip_bfloat16*)qkv, (const float2*)freqs, (const int*)pos,
                            k_cache, v_cache, (__hip_bfloat16*)q_out,
-                           n_slots, n_heads, max_seq, head_dim);
+                           n_slots, n_heads, n_kv_heads, max_seq, head_dim);
     } else {
         hipLaunchKernelGGL(smg_rope_kv_store_t<false>, grid, dim3(WAVE), 0, (hipStream_t)stream,
                            (const __hip_bfloat16*)qkv, (const float2*)freqs, (const int*)pos,
                            k_cache, v_cache, (__hip_bfloat16*)q_out,
-                           n_slots, n_heads, max_seq, head_dim);
+                           n_slots, n_heads, n_kv_heads, max_seq, head_dim);
     }
     return (int)hipGetLastError();
+}
+
+extern "C" int smg_rope_kv_store_launch_ex(
+    const void* qkv, const void* freqs, const void* pos,
+    void* k_cache, void* v_cache, void* q_out,
+    int n_slots, int n_heads, int max_seq, int head_dim, void* stream, int kv_fp8) {
+    return smg_rope_kv_store_launch_gqa(qkv, freqs, pos, k_cache, v_cache, q_out, n_slots,
+                                        n_heads, n_heads, max_seq, head_dim, stream, kv_fp8);
 }
 
 extern "C" int smg_rope_kv_store_launch(

@@ -29,6 +29,11 @@ class TorchEngineConfig:
     n_layers: int = 16
     d_model: int = 2048
     n_heads: int = 16
+    # GQA (Llama3/Qwen-style): n_kv_heads < n_heads shares each KV head
+    # across a group of query heads — the decode-attention kernel streams
+    # each K/V row once per GROUP, cutting its HBM bytes by n_heads/n_kv_heads.
+    # None = MHA (n_kv_heads == n_heads).
+    n_kv_heads: Optional[int] = None
     ffn_mult: float = 2.6875  # 5504/2048, llama-style
     max_slots: int = 64
     max_seq: int = 2048
@@ -55,9 +60,29 @@ class TorchEngineConfig:
         """The flagship bench model: ~1.1B params bf16 (≈2.3 GB weights)."""
         return cls()
 
+    @classmethod
+    def bench_1b_gqa(cls) -> "TorchEngineConfig":
+        """GQA flagship variant: 16 q heads / 4 kv heads (group 4, the
+        Llama3-8B ratio), with d_ffn widened 5504->6528 so total parameter
+        count matches bench_1b (the q/k/v projection shrinks by 2*(D-KD)*D
+        per layer; 3*D*Δf recovers it)."""
+        return cls(n_kv_heads=4, ffn_mult=6528 / 2048)
+
     @property
     def head_dim(self) -> int:
         return self.d_model // self.n_heads
+
+    @property
+    def kv_heads(self) -> int:
+        return self.n_kv_heads if self.n_kv_heads else self.n_heads
+
+    @property
+    def kv_dim(self) -> int:
+        return self.kv_heads * self.head_dim
+
+    @property
+    def gqa_group(self) -> int:
+        return self.n_heads // self.kv_heads
 
     @property
     def d_ffn(self) -> int:
@@ -72,7 +97,7 @@ class _Layer:
         def w(*shape):
             return (torch.rand(*shape, generator=gen, device=device, dtype=torch.float32) * 2 - 1).to(dtype) * k
 
-        self.wqkv = w(d, 3 * d)
+        self.wqkv = w(d, d + 2 * cfg.kv_dim)  # q(D) | k(KD) | v(KD)
         self.wo = w(d, d)
         self.w13 = w(d, 2 * f)  # gate+up fused: one GEMM instead of two
         self.w2 = w(f, d)
@@ -118,7 +143,7 @@ class TorchEngine:
                 torch.float8_e4m3fn if (c.kv_fp8 and self.device.type == "cuda") else self.dtype
             )
             self.kv = torch.zeros(
-                c.n_layers, 2, c.max_slots, c.n_heads, c.max_seq, c.head_dim,
+                c.n_layers, 2, c.max_slots, c.kv_heads, c.max_seq, c.head_dim,
                 device=self.device, dtype=self.kv_dtype,
             )
             # rotary tables (complex form: one complex mul applies the rotation)
@@ -144,7 +169,7 @@ class TorchEngine:
         pc = self.cfg
         if pc.prefix_cache_slots > 0:
             self._pc_arena = torch.zeros(
-                pc.n_layers, 2, pc.prefix_cache_slots, pc.n_heads,
+                pc.n_layers, 2, pc.prefix_cache_slots, pc.kv_heads,
                 min(pc.prefix_cache_max, pc.max_seq), pc.head_dim,
                 device=self.device, dtype=self.kv.dtype,
             )
@@ -480,11 +505,17 @@ class TorchEngine:
         c = self.cfg
         B, T, _ = h.shape
         qkv = _rms(h, layer.ln1) @ layer.wqkv
-        q, k, v = qkv.split(c.d_model, dim=-1)
+        q, k, v = qkv.split([c.d_model, c.kv_dim, c.kv_dim], dim=-1)
         q = q.view(B, T, c.n_heads, c.head_dim).transpose(1, 2)
-        k = k.view(B, T, c.n_heads, c.head_dim).transpose(1, 2)
-        v = v.view(B, T, c.n_heads, c.head_dim).transpose(1, 2)
+        k = k.view(B, T, c.kv_heads, c.head_dim).transpose(1, 2)
+        v = v.view(B, T, c.kv_heads, c.head_dim).transpose(1, 2)
         return self._apply_rope(q, freqs), self._apply_rope(k, freqs), v
+
+    def _expand_kv(self, t):
+        """[.., KVH, T, hd] -> [.., H, T, hd] for torch attention paths (the
+        HIP decode kernel does the group mapping in-kernel instead)."""
+        g = self.cfg.gqa_group
+        return t if g == 1 else t.repeat_interleave(g, dim=-3)
 
     @torch.no_grad()
     def _prefill(self, slot: int, tokens: List[int], start: int) -> None:
@@ -518,19 +549,20 @@ class TorchEngine:
             slots_i32 = slots.to(torch.int32)
             starts_i32 = starts.to(torch.int32)
             qb = torch.empty(B, c.n_heads, L, c.head_dim, device=self.device, dtype=self.dtype)
-            kb = torch.empty_like(qb)
-            vb = torch.empty_like(qb)
+            kb = torch.empty(B, c.kv_heads, L, c.head_dim, device=self.device, dtype=self.dtype)
+            vb = torch.empty_like(kb)
             stream = torch.cuda.current_stream().cuda_stream
             freqs_ptr = self.freqs_cis.data_ptr()
         for li, layer in enumerate(self.layers):
             if use_fused:
-                qkv = (_rms(h, layer.ln1) @ layer.wqkv).view(B, L, 3 * c.d_model)
+                qkv = (_rms(h, layer.ln1) @ layer.wqkv).view(B, L, c.d_model + 2 * c.kv_dim)
                 self._hip_rope_prefill(
                     qkv.data_ptr(), freqs_ptr, slots_i32.data_ptr(), starts_i32.data_ptr(),
                     self.kv[li, 0].data_ptr(), self.kv[li, 1].data_ptr(),
                     qb.data_ptr(), kb.data_ptr(), vb.data_ptr(),
                     B, L, c.n_heads, c.max_seq, c.head_dim, stream,
                     1 if self.kv.dtype == torch.float8_e4m3fn else 0,
+                    c.kv_heads,
                 )
                 q, k, v = qb, kb, vb
             else:
@@ -540,7 +572,8 @@ class TorchEngine:
                 self.kv[li, 1][slots[:, None], :, pos] = v.permute(0, 2, 1, 3)
             if fresh:
                 # no history: attend within the chunk itself, flash kernel
-                attn = F.scaled_dot_product_attention(q, k, v, is_causal=True)
+                attn = F.scaled_dot_product_attention(
+                    q, self._expand_kv(k), self._expand_kv(v), is_causal=True)
             elif use_fused and uniform_start and self._flash_lse:
                 # uniform-start suffix chunk (the prefix-cache-hit fast path):
                 # two FLASH passes — history cross-attention (all keys valid,
@@ -548,10 +581,11 @@ class TorchEngine:
                 # Replaces the masked-sdpa math path (bmm + 37 MB mask add +
                 # softmax per layer).
                 start0 = int(starts[0].item())
-                kk = self._kv_hist(li, 0, slots)[:, :, :start0]
-                vv = self._kv_hist(li, 1, slots)[:, :, :start0]
+                kk = self._expand_kv(self._kv_hist(li, 0, slots)[:, :, :start0])
+                vv = self._expand_kv(self._kv_hist(li, 1, slots)[:, :, :start0])
                 o1, lse1 = self._flash_lse(q, kk, vv, False)
-                o2, lse2 = self._flash_lse(q, k, v, True)
+                o2, lse2 = self._flash_lse(q, self._expand_kv(k).contiguous(),
+                                           self._expand_kv(v).contiguous(), True)
                 merge = getattr(self, "_hip_lse_merge", None)
                 if merge is not None and o1.is_contiguous() and o2.is_contiguous():
                     rows = B * c.n_heads * L
@@ -568,8 +602,8 @@ class TorchEngine:
                     attn = o1 * (lse1 - lse_tot).exp().unsqueeze(-1) + o2 * (lse2 - lse_tot).exp().unsqueeze(-1)
                     attn = attn.to(q.dtype)
             else:
-                kk = self._kv_hist(li, 0, slots)[:, :, :t_max]
-                vv = self._kv_hist(li, 1, slots)[:, :, :t_max]
+                kk = self._expand_kv(self._kv_hist(li, 0, slots)[:, :, :t_max])
+                vv = self._expand_kv(self._kv_hist(li, 1, slots)[:, :, :t_max])
                 attn = F.scaled_dot_product_attention(q, kk, vv, attn_mask=mask)
             attn2 = attn.transpose(1, 2).reshape(B * L, c.d_model)
             h = torch.addmm(h.view(B * L, c.d_model), attn2, layer.wo).view(B, L, c.d_model)
@@ -617,11 +651,12 @@ class TorchEngine:
                     qc.data_ptr(), self.kv[li, 0].data_ptr(), self.kv[li, 1].data_ptr(),
                     self._pos_i32.data_ptr(), self._attn_out.data_ptr(),
                     S, c.n_heads, c.max_seq, c.head_dim, scale, stream,
+                    0, c.kv_heads,
                 )
                 attn_flat = self._attn_out.view(S, 1, c.d_model)
             else:
-                kk = self.kv[li, 0][:, :, :maxlen]
-                vv = self.kv[li, 1][:, :, :maxlen]
+                kk = self._expand_kv(self.kv[li, 0][:, :, :maxlen])
+                vv = self._expand_kv(self.kv[li, 1][:, :, :maxlen])
                 attn = F.scaled_dot_product_attention(q, kk, vv, attn_mask=mask)
                 attn_flat = attn.transpose(1, 2).reshape(S, 1, c.d_model)
             h = h + attn_flat @ layer.wo
@@ -646,16 +681,16 @@ class TorchEngine:
         kv8 = 1 if self.kv.dtype == torch.float8_e4m3fn else 0
         freqs_ptr = self.freqs_cis.data_ptr()  # complex64 [T, hd/2] == float2
         for li, layer in enumerate(self.layers):
-            qkv = _rms(h, layer.ln1) @ layer.wqkv  # [S, 3D]
+            qkv = _rms(h, layer.ln1) @ layer.wqkv  # [S, D + 2*KD]
             self._hip_fused(
                 qkv.data_ptr(), freqs_ptr, self._pos_i32.data_ptr(),
                 self.kv[li, 0].data_ptr(), self.kv[li, 1].data_ptr(), self._q_buf.data_ptr(),
-                S, c.n_heads, c.max_seq, c.head_dim, stream, kv8,
+                S, c.n_heads, c.max_seq, c.head_dim, stream, kv8, c.kv_heads,
             )
             self._hip_attn(
                 self._q_buf.data_ptr(), self.kv[li, 0].data_ptr(), self.kv[li, 1].data_ptr(),
                 self._pos_i32.data_ptr(), self._attn_out.data_ptr(),
-                S, c.n_heads, c.max_seq, c.head_dim, scale, stream, kv8,
+                S, c.n_heads, c.max_seq, c.head_dim, scale, stream, kv8, c.kv_heads,
             )
             h = torch.addmm(h, self._attn_out.view(S, c.d_model), layer.wo)
             gu = _rms(h, layer.ln2) @ layer.w13  # [S, 2F]

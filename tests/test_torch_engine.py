@@ -122,3 +122,55 @@ class TestEngineGpu:
             eng.step()
         torch.cuda.synchronize()
         assert all(len(eng.collect(r)) == 8 for r in rids)
+
+
+class TestGqaEngine:
+    """GQA variant (n_kv_heads < n_heads) on the CPU logic path."""
+
+    def cfg(self):
+        c = TorchEngineConfig.tiny()
+        c.n_kv_heads = 2  # 4 q heads, group 2
+        return c
+
+    def test_decodes_and_deterministic(self):
+        eng1 = TorchEngine(self.cfg(), device="cpu")
+        eng2 = TorchEngine(self.cfg(), device="cpu")
+        out1 = out2 = None
+        r1 = eng1.submit(list(range(24)), max_new_tokens=4)
+        r2 = eng2.submit(list(range(24)), max_new_tokens=4)
+        for _ in range(64):
+            if eng1.finished(r1) and eng2.finished(r2):
+                break
+            eng1.step()
+            eng2.step()
+        out1, out2 = eng1.collect(r1), eng2.collect(r2)
+        assert len(out1) == 4 and out1 == out2
+
+    def test_chunked_matches_unchunked(self):
+        c1 = self.cfg()
+        c1.prefill_chunk = 8
+        eng1 = TorchEngine(c1, device="cpu")
+        r1 = eng1.submit(list(range(20)), max_new_tokens=2)
+        while not eng1.finished(r1):
+            eng1.step()
+        eng2 = TorchEngine(self.cfg(), device="cpu")
+        r2 = eng2.submit(list(range(20)), max_new_tokens=2)
+        while not eng2.finished(r2):
+            eng2.step()
+        assert eng1.collect(r1) == eng2.collect(r2)
+
+    def test_kv_arena_uses_kv_heads(self):
+        eng = TorchEngine(self.cfg(), device="cpu")
+        assert eng.kv.shape[3] == 2  # kv_heads, not n_heads
+
+    def test_bench_gqa_param_parity(self):
+        """bench_1b_gqa widens the FFN so total params match bench_1b."""
+
+        def n_params(c):
+            d, f, kd = c.d_model, c.d_ffn, c.kv_dim
+            per_layer = d * (d + 2 * kd) + d * d + d * 2 * f + f * d + 2 * d
+            return c.vocab_size * d + c.n_layers * per_layer + d
+
+        mha = n_params(TorchEngineConfig.bench_1b())
+        gqa = n_params(TorchEngineConfig.bench_1b_gqa())
+        assert abs(mha - gqa) / mha < 0.01, (mha, gqa)
