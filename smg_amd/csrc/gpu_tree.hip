@@ -149,31 +149,59 @@ __device__ int probe_find_lane(const GpuTreeDev& T, unsigned long long key) {
 // the slot (ours, or an existing same-key entry's), -1 when the table
 // neighborhood is full.
 #define BUSY_KEY 2ull
+__device__ __forceinline__ bool try_claim(const GpuTreeDev& T, uint32_t slot,
+                                          unsigned long long expected,
+                                          unsigned long long key, uint32_t val) {
+    if (__hip_atomic_compare_exchange_strong(&T.table_keys[slot], &expected, BUSY_KEY,
+                                             __ATOMIC_ACQ_REL, __ATOMIC_RELAXED,
+                                             __HIP_MEMORY_SCOPE_AGENT)) {
+        __hip_atomic_store(&T.table_vals[slot], val, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        __hip_atomic_store(&T.table_keys[slot], key, __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_AGENT);
+        return true;
+    }
+    return false;
+}
+
 __device__ int probe_insert(const GpuTreeDev& T, unsigned long long key, uint32_t val, uint32_t* out_slot) {
     uint32_t base = (uint32_t)(key & T.table_mask);
-    for (uint32_t i = 0; i < 1024u; ++i) {
+    // Pass 1: find the key (or the first EMPTY), remembering the first
+    // tombstone.  A tombstone must NOT be claimed before the scan proves the
+    // key is absent further down the chain — claiming it eagerly would
+    // create a duplicate entry that shadows the live one (lookups stop at
+    // the first match, so the live entry's tenant state would go dark).
+    int first_ts = -1;
+    uint32_t i = 0;
+    for (; i < 1024u; ++i) {
         uint32_t slot = (base + i) & T.table_mask;
         // relaxed scan: acquire loads cost 2-3x per hop and invalidate the
         // CU's L1 chip-wide (MI355X_MICROARCH.md §polling); ordering comes
-        // from the CAS below and the val-before-key publication
+        // from the CAS in try_claim and the val-before-key publication
         unsigned long long k = atomic_load_key(&T.table_keys[slot]);
         if (k == key) {
             *out_slot = slot;
             return (int)__hip_atomic_load(&T.table_vals[slot], __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
         }
-        if (k == EMPTY_KEY || k == TOMBSTONE_KEY) {
-            unsigned long long expected = k;
-            if (__hip_atomic_compare_exchange_strong(&T.table_keys[slot], &expected, BUSY_KEY,
-                                                     __ATOMIC_ACQ_REL, __ATOMIC_RELAXED,
-                                                     __HIP_MEMORY_SCOPE_AGENT)) {
-                __hip_atomic_store(&T.table_vals[slot], val, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-                __hip_atomic_store(&T.table_keys[slot], key, __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_AGENT);
-                *out_slot = slot;
-                return (int)val;
-            }
-            // lost the reservation; fall through and keep probing
-        }
+        if (k == EMPTY_KEY) break;
+        if (k == TOMBSTONE_KEY && first_ts < 0) first_ts = (int)slot;
         // BUSY or other key: probe on
+    }
+    // Pass 2: key absent — reuse the remembered tombstone first
+    if (first_ts >= 0 && try_claim(T, (uint32_t)first_ts, TOMBSTONE_KEY, key, val)) {
+        *out_slot = (uint32_t)first_ts;
+        return (int)val;
+    }
+    // then fight for empty/tombstone slots from the scan frontier on
+    for (; i < 1024u; ++i) {
+        uint32_t slot = (base + i) & T.table_mask;
+        unsigned long long k = atomic_load_key(&T.table_keys[slot]);
+        if (k == key) {
+            *out_slot = slot;
+            return (int)__hip_atomic_load(&T.table_vals[slot], __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        }
+        if ((k == EMPTY_KEY || k == TOMBSTONE_KEY) && try_claim(T, slot, k, key, val)) {
+            *out_slot = slot;
+            return (int)val;
+        }
     }
     return -1;
 }
@@ -336,11 +364,19 @@ smg_tree_match_insert(GpuTreeDev T, BatchArgs A) {
             if (other > packed) packed = other;
         }
         if ((packed & 0xffffffffu) != 0) matched_tenant = (int)(packed & 0xffffffffu) - 1;
-        // touch the matched tenant's stamp along the deepest node (mirrors
-        // token_tree.rs:693 touch_tenant on match)
-        if (lane == 0 && matched_tenant >= 0) {
-            uint32_t now = atomicAdd(T.clock_, 1u) + 1u;
-            atomicMax(&T.node_ts[(size_t)deepest_tenanted * WAVE + matched_tenant], now);
+        // touch the matched tenant's stamp along the WHOLE matched path
+        // (mirrors token_tree.rs:693 touch_tenant on match): refreshing only
+        // the deepest node would let the eviction sweep clear the shallow
+        // nodes of a hot path out from under it, breaking its match run
+        if (matched_tenant >= 0) {
+            uint32_t now = 0;
+            if (lane == 0) now = atomicAdd(T.clock_, 1u) + 1u;
+            now = __shfl(now, 0, WAVE);
+            for (uint32_t page = lane; page < matched_pages; page += WAVE) {
+                int pid = s_ids[page];
+                if (pid >= 0)
+                    atomicMax(&T.node_ts[(size_t)pid * WAVE + matched_tenant], now);
+            }
         }
     }
 
