@@ -307,6 +307,185 @@ class SchedulerServicer(grpc.GenericRpcHandler):
         self._event_loop = loop
 
 
+class SglangProtoServicer(grpc.GenericRpcHandler):
+    """Wire-compatible `sglang.grpc.scheduler.SglangScheduler` service
+    (reference sglang_scheduler.proto:11-59) over the same EngineAdapter —
+    protobuf messages from grpc/proto_wire.py, so a reference-protocol router
+    (tonic) can front THIS repo's engines directly.  Registered alongside the
+    msgpack SchedulerServicer; the method path picks the dialect."""
+
+    def __init__(self, adapter: EngineAdapter, model_id: str = "mock-model"):
+        from . import proto_wire as pw
+
+        self.pw = pw
+        self.adapter = adapter
+        self.model_id = model_id
+        self._event_loop = None
+
+    def bind_loop(self, loop):
+        self._event_loop = loop
+
+    def service(self, handler_call_details):
+        pw = self.pw
+        prefix = f"/{pw.SERVICE_NAME}/"
+        if not handler_call_details.method.startswith(prefix):
+            return None
+        name = handler_call_details.method[len(prefix):]
+        spec = pw.METHODS.get(name)
+        fn = getattr(self, f"_p_{name}", None)
+        if spec is None or fn is None:
+            return None
+        req_cls, resp_cls, streaming = spec
+        if streaming:
+            return grpc.unary_stream_rpc_method_handler(
+                fn, request_deserializer=req_cls.FromString,
+                response_serializer=lambda m: m.SerializeToString())
+        return grpc.unary_unary_rpc_method_handler(
+            fn, request_deserializer=req_cls.FromString,
+            response_serializer=lambda m: m.SerializeToString())
+
+    # ---- handlers ---------------------------------------------------------
+    def _p_Generate(self, req, context):
+        pw = self.pw
+        sp = req.sampling_params
+        internal = api.GenerateRequest(
+            request_id=req.request_id,
+            input_ids=list(req.tokenized.input_ids),
+            text=req.tokenized.original_text or None,
+            sampling=api.SamplingParams(
+                max_new_tokens=sp.max_new_tokens if sp.HasField("max_new_tokens") else 128,
+                temperature=sp.temperature or 1.0,
+                top_p=sp.top_p or 1.0,
+                top_k=sp.top_k or -1,
+                stop=list(sp.stop),
+                stop_token_ids=list(sp.stop_token_ids),
+                ignore_eos=sp.ignore_eos,
+                skip_special_tokens=sp.skip_special_tokens,
+            ),
+            stream=True,
+            lora_id=req.lora_id or None,
+            dp_rank=req.data_parallel_rank or None,
+        )
+        if req.HasField("disaggregated_params"):
+            internal.bootstrap_host = req.disaggregated_params.bootstrap_host or None
+            internal.bootstrap_port = req.disaggregated_params.bootstrap_port or None
+            internal.bootstrap_room = req.disaggregated_params.bootstrap_room or None
+        loop = self._event_loop
+        agen = self.adapter.generate(internal)
+        all_ids = []
+        try:
+            while True:
+                chunk = asyncio.run_coroutine_threadsafe(agen.__anext__(), loop).result()
+                all_ids.extend(chunk.token_ids)
+                resp = pw.GenerateResponse()
+                resp.request_id = req.request_id
+                if chunk.finished:
+                    resp.complete.output_ids.extend(all_ids)
+                    resp.complete.finish_reason = chunk.finish_reason or "stop"
+                    resp.complete.prompt_tokens = chunk.prompt_tokens
+                    resp.complete.completion_tokens = chunk.completion_tokens
+                    resp.complete.cached_tokens = chunk.cached_tokens
+                    yield resp
+                    break
+                resp.chunk.token_ids.extend(chunk.token_ids)
+                resp.chunk.prompt_tokens = chunk.prompt_tokens
+                resp.chunk.completion_tokens = chunk.completion_tokens
+                resp.chunk.cached_tokens = chunk.cached_tokens
+                yield resp
+        except StopAsyncIteration:
+            pass
+        except Exception as exc:
+            log.warning("proto generate stream error: %s", exc)
+            context.abort(grpc.StatusCode.INTERNAL, str(exc))
+
+    def _p_HealthCheck(self, req, context):
+        r = self.pw.HealthCheckResponse()
+        r.healthy = True
+        return r
+
+    def _p_Abort(self, req, context):
+        self.adapter.abort(req.request_id)
+        r = self.pw.AbortResponse()
+        r.success = True
+        return r
+
+    def _p_GetModelInfo(self, req, context):
+        r = self.pw.GetModelInfoResponse()
+        r.model_path = self.model_id
+        r.served_model_name = self.model_id
+        r.is_generation = True
+        r.max_context_length = 131072
+        return r
+
+    def _p_GetServerInfo(self, req, context):
+        r = self.pw.GetServerInfoResponse()
+        r.server_type = "grpc"
+        r.sglang_version = "smg-amd-0.1.0"
+        return r
+
+    def _p_GetLoads(self, req, context):
+        snap = self.adapter.loads()
+        r = self.pw.GetLoadsResponse()
+        r.dp_rank_count = 1
+        ld = r.loads.add()
+        ld.num_running_reqs = int(snap.get("num_running_reqs") or 0)
+        ld.num_waiting_reqs = int(snap.get("num_queue_reqs") or 0)
+        ld.num_used_tokens = int(snap.get("num_inflight_tokens") or 0)
+        ld.num_waiting_uncached_tokens = int(snap.get("num_queue_tokens") or 0)
+        ld.token_usage = float(snap.get("token_usage") or 0.0)
+        ld.gen_throughput = float(snap.get("gen_throughput") or 0.0)
+        r.aggregate.total_running_reqs = ld.num_running_reqs
+        r.aggregate.total_waiting_reqs = ld.num_waiting_reqs
+        return r
+
+    def _p_FlushCache(self, req, context):
+        if self.adapter.is_mock:
+            self.adapter.engine.sim.prefix_cache.clear()
+        r = self.pw.FlushCacheResponse()
+        r.success = True
+        return r
+
+    def _p_Embed(self, req, context):
+        import hashlib
+
+        ids = list(req.tokenized.input_ids)
+        vec = [int.from_bytes(hashlib.blake2b(bytes(str(i), "utf8"), digest_size=4).digest(),
+                              "little") % 1000 / 1000.0 for i in ids[:16]]
+        vec += [0.0] * (16 - len(vec))
+        r = self.pw.EmbedResponse()
+        r.embedding_dim = len(vec)
+        r.embedding.extend(vec)
+        r.prompt_tokens = len(ids)
+        return r
+
+    def _p_SubscribeKvEvents(self, req, context):
+        import time
+
+        seq = int(req.start_sequence_number)
+        while context.is_active():
+            events = self.adapter.drain_kv_events()
+            if events:
+                seq += 1
+                b = self.pw.KvEventBatch()
+                b.sequence_number = seq
+                b.timestamp = time.time()
+                for ev in events:
+                    e = b.events.add()
+                    e.event_id = seq
+                    if ev.get("type") == "stored":
+                        for blk in ev.get("blocks", []):
+                            kb = e.stored.blocks.add()
+                            kb.block_hash = int(blk.get("block_hash", 0))
+                            kb.token_ids.extend(blk.get("token_ids", []))
+                            kb.block_size = int(blk.get("block_size", 16))
+                    elif ev.get("type") == "removed":
+                        e.removed.block_hashes.extend(ev.get("block_hashes", []))
+                    else:
+                        e.cleared.SetInParent()
+                yield b
+            time.sleep(0.05)
+
+
 async def serve_grpc_worker(
     host: str = "127.0.0.1",
     port: int = 50051,
@@ -314,7 +493,10 @@ async def serve_grpc_worker(
     model_id: str = "mock-model",
     sim_config: Optional[SimConfig] = None,
 ):
-    """Start a gRPC engine worker; returns (server, adapter, bound_port)."""
+    """Start a gRPC engine worker; returns (server, adapter, bound_port).
+    Serves BOTH dialects on one port: the in-repo msgpack service
+    (smg.Scheduler) and the reference-wire proto service
+    (sglang.grpc.scheduler.SglangScheduler)."""
     from concurrent.futures import ThreadPoolExecutor
 
     if engine is None:
@@ -323,8 +505,10 @@ async def serve_grpc_worker(
     await adapter.start()
     servicer = SchedulerServicer(adapter, model_id)
     servicer.bind_loop(asyncio.get_event_loop())
+    proto_servicer = SglangProtoServicer(adapter, model_id)
+    proto_servicer.bind_loop(asyncio.get_event_loop())
     server = grpc.server(ThreadPoolExecutor(max_workers=32))
-    server.add_generic_rpc_handlers((servicer,))
+    server.add_generic_rpc_handlers((servicer, proto_servicer))
     bound = server.add_insecure_port(f"{host}:{port}")
     server.start()
     return server, adapter, bound
