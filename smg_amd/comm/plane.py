@@ -12,8 +12,14 @@ peer (no tags), so the plane is a fixed-cadence LOCKSTEP TICK:
         w -> gateway  : one event tensor    (token/done events, padded, int32)
 
 Exactly one send and one recv per peer per tick, always in the same order —
-deadlock-free by construction.  isend/irecv across peers overlap on the tick.
-Tensors live on the GPU: token ids move over xGMI without host staging.
+deadlock-free by construction.  isend/irecv across peers overlap on the tick,
+and the gateway posts BOTH directions at tick_send time so the whole exchange
+overlaps its local engine step (the world-8 rehearsal showed the serialized
+recv+Python-parse path capping scaling at 44% efficiency; see
+scripts/world8_rehearsal.py).
+
+Host staging is numpy-vectorized into persistent (pinned on GPU) buffers —
+no per-request torch tensor construction, no per-event Python tuple parsing.
 
 Wire layout (int32):
   request tensor [1 + MAX_REQS*(3 + MAX_PROMPT)]:
@@ -26,8 +32,9 @@ Wire layout (int32):
 from __future__ import annotations
 
 from dataclasses import dataclass
-from typing import Dict, List, Tuple
+from typing import Dict, List, Optional, Tuple
 
+import numpy as np
 import torch
 import torch.distributed as dist
 
@@ -57,6 +64,11 @@ class PlaneConfig:
         return 1 + self.max_events_per_tick * 3
 
 
+def _host_staging(n: int, gpu: bool) -> Tuple[torch.Tensor, np.ndarray]:
+    t = torch.zeros(n, dtype=torch.int32, pin_memory=gpu)
+    return t, t.numpy()
+
+
 class GatewayPlane:
     """Rank-0 side: one staging pair per worker rank."""
 
@@ -64,61 +76,82 @@ class GatewayPlane:
         self.cfg = cfg
         self.worker_ranks = worker_ranks
         dev = torch.device(cfg.device)
+        self._gpu = dev.type == "cuda"
         self._send = {w: torch.zeros(cfg.req_len, dtype=torch.int32, device=dev) for w in worker_ranks}
         self._recv = {w: torch.zeros(cfg.ev_len, dtype=torch.int32, device=dev) for w in worker_ranks}
+        # persistent host staging (pinned when the wire tensors are on-device)
+        self._send_h: Dict[int, Tuple[torch.Tensor, np.ndarray]] = {}
+        self._recv_h: Dict[int, Tuple[torch.Tensor, np.ndarray]] = {}
+        for w in worker_ranks:
+            if self._gpu:
+                self._send_h[w] = _host_staging(cfg.req_len, True)
+                self._recv_h[w] = _host_staging(cfg.ev_len, True)
+            else:
+                self._send_h[w] = (self._send[w], self._send[w].numpy())
+                self._recv_h[w] = (self._recv[w], self._recv[w].numpy())
         self._pending: Dict[int, List[Tuple[int, int, List[int]]]] = {w: [] for w in worker_ranks}
+        self._works = []  # outstanding isend/irecv handles for this tick
 
     def enqueue(self, worker_rank: int, rid: int, max_new: int, prompt: List[int]) -> None:
         self._pending[worker_rank].append((rid, max_new, prompt[-self.cfg.max_prompt:]))
 
     def tick_send(self, stop: bool = False, barrier: bool = False) -> None:
-        """Phase 1: ship this tick's requests to every worker (the workers
-        start their engine step as soon as the send lands, overlapping the
-        gateway's own local work)."""
+        """Phase 1: ship this tick's requests to every worker AND post the
+        event recvs, so the whole exchange progresses while the gateway runs
+        its local engine step."""
         cfg = self.cfg
+        # previous tick's handles must be drained before buffers are reused
+        for work in self._works:
+            work.wait()
+        self._works = []
         for w in self.worker_ranks:
-            buf = self._send[w]
             pend = self._pending[w][: cfg.max_reqs_per_tick]
             self._pending[w] = self._pending[w][len(pend):]
             hdr = len(pend) | (STOP_FLAG if stop else 0) | (BARRIER_FLAG if barrier else 0)
-            rows = torch.zeros(cfg.req_len, dtype=torch.int32)
+            _, rows = self._send_h[w]
             rows[0] = hdr
             for i, (rid, max_new, prompt) in enumerate(pend):
                 base = 1 + i * cfg.req_stride
                 rows[base] = rid
                 rows[base + 1] = max_new
                 rows[base + 2] = len(prompt)
-                rows[base + 3: base + 3 + len(prompt)] = torch.tensor(prompt, dtype=torch.int32)
-            buf.copy_(rows.to(buf.device))
-        ops = [dist.P2POp(dist.isend, self._send[w], w) for w in self.worker_ranks]
-        if ops:
-            for work in dist.batch_isend_irecv(ops):
-                work.wait()
-
-    def tick_recv(self) -> Dict[int, List[Tuple[int, int, int]]]:
-        """Phase 2: collect every worker's event tensor."""
-        cfg = self.cfg
-        ops = [dist.P2POp(dist.irecv, self._recv[w], w) for w in self.worker_ranks]
-        if ops:
-            for work in dist.batch_isend_irecv(ops):
-                work.wait()
-        out: Dict[int, List[Tuple[int, int, int]]] = {}
+                rows[base + 3: base + 3 + len(prompt)] = prompt  # numpy list assign, one C pass
+            if self._gpu:
+                self._send[w].copy_(self._send_h[w][0], non_blocking=True)
+        ops = []
         for w in self.worker_ranks:
-            ev = self._recv[w].cpu()
-            n = int(ev[0])
-            events = []
-            for i in range(min(n, cfg.max_events_per_tick)):
-                base = 1 + i * 3
-                events.append((int(ev[base]), int(ev[base + 1]), int(ev[base + 2])))
-            out[w] = events
+            ops.append(dist.P2POp(dist.isend, self._send[w], w))
+            ops.append(dist.P2POp(dist.irecv, self._recv[w], w))
+        if ops:
+            self._works = dist.batch_isend_irecv(ops)
+
+    def tick_recv(self) -> Dict[int, np.ndarray]:
+        """Phase 2: collect every worker's event tensor.  Returns
+        {worker_rank: int32 array [n_events, 3] of (rid, token, flags)}."""
+        cfg = self.cfg
+        for work in self._works:
+            work.wait()
+        self._works = []
+        if self._gpu:
+            for w in self.worker_ranks:
+                self._recv_h[w][0].copy_(self._recv[w], non_blocking=True)
+            torch.cuda.synchronize()
+        out: Dict[int, np.ndarray] = {}
+        for w in self.worker_ranks:
+            ev = self._recv_h[w][1]
+            n = min(int(ev[0]), cfg.max_events_per_tick)
+            out[w] = ev[1: 1 + n * 3].reshape(n, 3).copy()
         return out
 
-    def tick(self, stop: bool = False, barrier: bool = False) -> Dict[int, List[Tuple[int, int, int]]]:
-        """One lockstep exchange (send + recv back-to-back).  The bench's tick
+    def tick(self, stop: bool = False, barrier: bool = False) -> Dict[int, np.ndarray]:
+        """One lockstep exchange (send + recv back-to-back).  The serving tick
         loop uses tick_send()/tick_recv() split around the gateway's local
         engine step so remote workers compute concurrently."""
         self.tick_send(stop=stop, barrier=barrier)
         if stop:
+            for work in self._works:
+                work.wait()
+            self._works = []
             return {}
         return self.tick_recv()
 
@@ -130,29 +163,37 @@ class WorkerPlane:
         self.cfg = cfg
         self.gateway_rank = gateway_rank
         dev = torch.device(cfg.device)
+        self._gpu = dev.type == "cuda"
         self._recv = torch.zeros(cfg.req_len, dtype=torch.int32, device=dev)
         self._send = torch.zeros(cfg.ev_len, dtype=torch.int32, device=dev)
+        if self._gpu:
+            self._recv_h = _host_staging(cfg.req_len, True)
+            self._send_h = _host_staging(cfg.ev_len, True)
+        else:
+            self._recv_h = (self._recv, self._recv.numpy())
+            self._send_h = (self._send, self._send.numpy())
         self._send_work = None
+        self.barrier_requested = False
 
     def tick(self, events: List[Tuple[int, int, int]]) -> Tuple[List[Tuple[int, int, List[int]]], bool]:
         """One lockstep exchange: sends `events` [(rid, token, flags)], receives
         new requests.  Returns (new_requests, stop).
 
         Only the request RECV is awaited here: the event send drains while the
-        gateway runs its own local engine step (its recv is posted after), so
-        worker and gateway compute concurrently.  The send handle is awaited
-        at the next tick before the buffer is reused."""
+        gateway runs its own local engine step (its recv is posted alongside
+        its send), so worker and gateway compute concurrently.  The send
+        handle is awaited at the next tick before the buffer is reused."""
         cfg = self.cfg
         if self._send_work is not None:
             self._send_work.wait()
             self._send_work = None
-        ev = torch.zeros(cfg.ev_len, dtype=torch.int32)
+        _, ev = self._send_h
         n = min(len(events), cfg.max_events_per_tick)
         ev[0] = n
-        for i, (rid, token, flags) in enumerate(events[:n]):
-            base = 1 + i * 3
-            ev[base], ev[base + 1], ev[base + 2] = rid, token, flags
-        self._send.copy_(ev.to(self._send.device))
+        if n:
+            ev[1: 1 + n * 3] = np.asarray(events[:n], dtype=np.int32).reshape(-1)
+        if self._gpu:
+            self._send.copy_(self._send_h[0], non_blocking=True)
         ops = [
             dist.P2POp(dist.irecv, self._recv, self.gateway_rank),
             dist.P2POp(dist.isend, self._send, self.gateway_rank),
@@ -160,7 +201,10 @@ class WorkerPlane:
         works = dist.batch_isend_irecv(ops)
         works[0].wait()  # requests arrived; step can start
         self._send_work = works[1]
-        req = self._recv.cpu()
+        if self._gpu:
+            self._recv_h[0].copy_(self._recv, non_blocking=True)
+            torch.cuda.synchronize()
+        req = self._recv_h[1]
         hdr = int(req[0])
         stop = bool(hdr & STOP_FLAG)
         self.barrier_requested = bool(hdr & BARRIER_FLAG)

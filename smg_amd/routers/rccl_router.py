@@ -117,40 +117,57 @@ class TickGateway:
                     self._local_pending.append((rid, max_new, toks))
                 else:
                     self.plane.enqueue(rank, rid, max_new, toks)
-        # 2) ship remote work first so workers overlap the local engine step
+        # 2) ship remote work first (send AND recv posted together) so the
+        # whole plane exchange overlaps the local engine step
         if self.plane is not None:
             tp = time.perf_counter()
             self.plane.tick_send()
             self.phase_t["plane"] += time.perf_counter() - tp
-        events: List[Tuple[int, int, int]] = []
+        local_events: List[Tuple[int, int, int]] = []
         if self.local_engine is not None:
             tl = time.perf_counter()
             for rid, max_new, toks in self._local_pending:
                 self.local_engine.submit(toks, max_new, rid=rid)
             self._local_pending.clear()
             self.local_engine.step(decode_burst=self.decode_burst)
-            events.extend(self.local_engine.drain_events())
+            local_events = self.local_engine.drain_events()
             self.phase_t["local"] += time.perf_counter() - tl
+        remote_arrays = []
         if self.plane is not None:
             tp = time.perf_counter()
-            for _w, evs in self.plane.tick_recv().items():
-                events.extend(evs)
+            remote_arrays = list(self.plane.tick_recv().values())
             self.phase_t["plane"] += time.perf_counter() - tp
         # 3) completions + event fan-out
         tev = time.perf_counter()
         done_now = 0
         cb = self.on_event
-        for rid, token, flags in events:
-            if cb is not None:
+        if cb is not None:
+            rows = list(local_events)
+            for arr in remote_arrays:
+                rows.extend(arr.tolist())
+            for rid, token, flags in rows:
                 cb(rid, token, flags)
-            if flags & DONE:
-                wrk = self.inflight.pop(rid, None)
-                if wrk is not None:
-                    self.workers[wrk].decr_load()  # also bumps processed_requests
-                    done_now += 1
+                if flags & DONE:
+                    done_now += self._complete(rid)
+        else:
+            # bench fast path: vectorized DONE extraction, no per-token work
+            for rid, _tok, flags in local_events:
+                if flags & DONE:
+                    done_now += self._complete(rid)
+            for arr in remote_arrays:
+                if len(arr):
+                    for rid in arr[(arr[:, 2] & DONE) != 0, 0].tolist():
+                        done_now += self._complete(rid)
         self.completed_total += done_now
         self.phase_t["events"] += time.perf_counter() - tev
         return done_now
+
+    def _complete(self, rid: int) -> int:
+        wrk = self.inflight.pop(rid, None)
+        if wrk is None:
+            return 0
+        self.workers[wrk].decr_load()  # also bumps processed_requests
+        return 1
 
     # ---- timing / shutdown ------------------------------------------------
     def barrier_sync(self) -> None:
