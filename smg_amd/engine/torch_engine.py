@@ -124,6 +124,9 @@ class _Request:
 
 
 class TorchEngine:
+    _flash_gqa = False
+    _sdpa_gqa = False
+
     def __init__(self, cfg: Optional[TorchEngineConfig] = None, device: str = "cuda:0", graphs: bool = False):
         if device.startswith("cuda") and not torch.cuda.is_available():
             raise RuntimeError("TorchEngine requires a GPU (use mock/engine.py on CPU)")
@@ -384,20 +387,41 @@ class TorchEngine:
         """(q,k,v,is_causal) -> (out, logsumexp) via the flash kernel, or None
         when the aten op is unavailable.  The lse output lets the suffix-
         prefill path merge a no-mask history pass with a square causal chunk
-        pass instead of taking the masked-sdpa math path."""
+        pass instead of taking the masked-sdpa math path.  Also probes
+        native GQA support (mismatched q/kv head counts) — when the flash
+        kernel handles it, prefill skips the repeat_interleave expansion
+        (measured 11% of device time + 5% copyBuffer at group 4)."""
 
         def call(q, k, v, causal):
             r = torch.ops.aten._scaled_dot_product_flash_attention(q, k, v, 0.0, causal, False)
             return r[0], r[1]
 
+        self._flash_gqa = False
+        self._sdpa_gqa = False
         try:
             probe = torch.randn(1, 2, 8, 64, device=self.device, dtype=self.dtype)
             out, lse = call(probe, probe, probe, True)
             if out.shape != probe.shape or lse.shape[-1] != 8:
                 return None
-            return call
         except Exception:
             return None
+        if self.cfg.gqa_group > 1:
+            try:
+                q4 = torch.randn(1, 4, 8, 64, device=self.device, dtype=self.dtype)
+                kv = torch.randn(1, 2, 8, 64, device=self.device, dtype=self.dtype)
+                out, lse = call(q4, kv, kv, False)
+                ref = F.scaled_dot_product_attention(
+                    q4.float(), kv.float().repeat_interleave(2, 1), kv.float().repeat_interleave(2, 1))
+                self._flash_gqa = (out.shape == q4.shape
+                                   and (out.float() - ref).abs().max().item() < 0.05)
+            except Exception:
+                self._flash_gqa = False
+            try:
+                out = F.scaled_dot_product_attention(q4, kv, kv, is_causal=True, enable_gqa=True)
+                self._sdpa_gqa = out.shape == q4.shape
+            except Exception:
+                self._sdpa_gqa = False
+        return call
 
     @staticmethod
     def _apply_rope(x, freqs):
@@ -571,9 +595,14 @@ class TorchEngine:
                 self.kv[li, 0][slots[:, None], :, pos] = k.permute(0, 2, 1, 3)
                 self.kv[li, 1][slots[:, None], :, pos] = v.permute(0, 2, 1, 3)
             if fresh:
-                # no history: attend within the chunk itself, flash kernel
-                attn = F.scaled_dot_product_attention(
-                    q, self._expand_kv(k), self._expand_kv(v), is_causal=True)
+                # no history: attend within the chunk itself, flash kernel.
+                # GQA stays native when the backend supports it (no
+                # repeat_interleave expansion copies)
+                if self.cfg.gqa_group > 1 and self._sdpa_gqa:
+                    attn = F.scaled_dot_product_attention(q, k, v, is_causal=True, enable_gqa=True)
+                else:
+                    attn = F.scaled_dot_product_attention(
+                        q, self._expand_kv(k), self._expand_kv(v), is_causal=True)
             elif use_fused and uniform_start and self._flash_lse:
                 # uniform-start suffix chunk (the prefix-cache-hit fast path):
                 # two FLASH passes — history cross-attention (all keys valid,
@@ -581,11 +610,17 @@ class TorchEngine:
                 # Replaces the masked-sdpa math path (bmm + 37 MB mask add +
                 # softmax per layer).
                 start0 = int(starts[0].item())
-                kk = self._expand_kv(self._kv_hist(li, 0, slots)[:, :, :start0])
-                vv = self._expand_kv(self._kv_hist(li, 1, slots)[:, :, :start0])
+                if self.cfg.gqa_group > 1 and not self._flash_gqa:
+                    kk = self._expand_kv(self._kv_hist(li, 0, slots)[:, :, :start0])
+                    vv = self._expand_kv(self._kv_hist(li, 1, slots)[:, :, :start0])
+                    kc = self._expand_kv(k).contiguous()
+                    vc = self._expand_kv(v).contiguous()
+                else:
+                    kk = self._kv_hist(li, 0, slots)[:, :, :start0]
+                    vv = self._kv_hist(li, 1, slots)[:, :, :start0]
+                    kc, vc = k, v
                 o1, lse1 = self._flash_lse(q, kk, vv, False)
-                o2, lse2 = self._flash_lse(q, self._expand_kv(k).contiguous(),
-                                           self._expand_kv(v).contiguous(), True)
+                o2, lse2 = self._flash_lse(q, kc, vc, True)
                 merge = getattr(self, "_hip_lse_merge", None)
                 if merge is not None and o1.is_contiguous() and o2.is_contiguous():
                     rows = B * c.n_heads * L
@@ -602,9 +637,14 @@ class TorchEngine:
                     attn = o1 * (lse1 - lse_tot).exp().unsqueeze(-1) + o2 * (lse2 - lse_tot).exp().unsqueeze(-1)
                     attn = attn.to(q.dtype)
             else:
-                kk = self._expand_kv(self._kv_hist(li, 0, slots)[:, :, :t_max])
-                vv = self._expand_kv(self._kv_hist(li, 1, slots)[:, :, :t_max])
-                attn = F.scaled_dot_product_attention(q, kk, vv, attn_mask=mask)
+                if self.cfg.gqa_group > 1 and self._sdpa_gqa:
+                    kk = self._kv_hist(li, 0, slots)[:, :, :t_max]
+                    vv = self._kv_hist(li, 1, slots)[:, :, :t_max]
+                    attn = F.scaled_dot_product_attention(q, kk, vv, attn_mask=mask, enable_gqa=True)
+                else:
+                    kk = self._expand_kv(self._kv_hist(li, 0, slots)[:, :, :t_max])
+                    vv = self._expand_kv(self._kv_hist(li, 1, slots)[:, :, :t_max])
+                    attn = F.scaled_dot_product_attention(q, kk, vv, attn_mask=mask)
             attn2 = attn.transpose(1, 2).reshape(B * L, c.d_model)
             h = torch.addmm(h.view(B * L, c.d_model), attn2, layer.wo).view(B, L, c.d_model)
             h = self._mlp(h, layer)
