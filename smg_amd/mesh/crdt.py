@@ -135,6 +135,34 @@ class MeshKV:
             out.extend(ns_map.values())
         return out
 
+    def namespaces(self) -> List[str]:
+        return list(self._mergers.keys())
+
+    def repair_page(self, ns: str, cursor: str = "", max_bytes: int = 2 << 20) -> dict:
+        """One incremental repair page (reference tree_sync.rs:38-67
+        `tree:req:`/`tree:page:` protocol, page cap mesh-v2 §3.1): current
+        state-ops for keys > `cursor` in key order, filled until the next
+        entry would push the page over `max_bytes`.  Returns
+        {entries, next_cursor, done}; applying entries through apply_remote
+        is idempotent (HLC-merged), so repair converges without a full join
+        snapshot."""
+        import json as _json
+
+        ns_map = self._data.get(ns, {})
+        keys = sorted(k for k in ns_map if k > cursor)
+        entries: List[dict] = []
+        used = 0
+        last = cursor
+        for k in keys:
+            d = ns_map[k].to_dict()
+            sz = len(_json.dumps(d))
+            if entries and used + sz > max_bytes:
+                return {"entries": entries, "next_cursor": last, "done": False}
+            entries.append(d)
+            used += sz
+            last = k
+        return {"entries": entries, "next_cursor": last, "done": True}
+
     def compact(self, keep_last: int = 10_000) -> None:
         if len(self._log) > keep_last:
             self._log = self._log[-keep_last:]

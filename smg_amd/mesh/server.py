@@ -28,11 +28,19 @@ def add_mesh_routes(app: web.Application, mesh: MeshNode) -> None:
     async def members(request: web.Request):
         return web.json_response({"node_id": mesh.node_id, "members": mesh._member_dicts()})
 
+    async def repair(request: web.Request):
+        return web.json_response(mesh.handle_repair(await request.json()))
+
+    async def partition(request: web.Request):
+        return web.json_response(mesh.partition_state())
+
     app.router.add_post("/mesh/ping", ping)
     app.router.add_post("/mesh/ping_req", ping_req)
     app.router.add_post("/mesh/sync", sync)
     app.router.add_post("/mesh/join", join)
+    app.router.add_post("/mesh/repair", repair)
     app.router.add_get("/mesh/members", members)
+    app.router.add_get("/mesh/partition", partition)
 
 
 async def start_mesh_server(mesh: MeshNode, host: str, port: int):

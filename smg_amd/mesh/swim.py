@@ -59,6 +59,22 @@ class MeshNode:
         self._stopped = asyncio.Event()
         self._sent_watermarks: Dict[str, int] = {}  # peer -> last local seq sent
         self.incarnation = 0
+        # partition detection (reference mesh/src/partition.rs): when a
+        # majority of known members are unreachable at once, this side is
+        # (likely) the isolated minority — flagged for metrics/readiness and
+        # used to trigger anti-entropy repair on heal
+        self.partitioned = False
+        self._partition_since: Optional[float] = None
+        self.partition_heals = 0
+        # incremental page-repair sessions (reference tree_sync.rs:38-67)
+        self.max_repair_page_bytes = 2 << 20
+        self.repair_max_retries = 3
+        self.repairs_completed = 0
+        self._repair_tasks: Dict[str, asyncio.Task] = {}
+        # chunked sync (reference mesh transport/chunking.rs): cap each sync
+        # POST so a post-partition backlog drains in bounded batches
+        self.max_sync_ops_per_post = 500
+        self.max_sync_posts_per_round = 8
 
     # ---- lifecycle ---------------------------------------------------------
     async def start(self, peer_urls: List[str]) -> None:
@@ -76,6 +92,9 @@ class MeshNode:
                 await self._task
             except (asyncio.CancelledError, Exception):
                 pass
+        for t in self._repair_tasks.values():
+            if not t.done():
+                t.cancel()
         if self._session:
             await self._session.close()
 
@@ -125,6 +144,7 @@ class MeshNode:
         # SWIM probe: one random member per round
         if live:
             target = random.choice(live)
+            was = target.state
             ok = await self._ping(target)
             if not ok:
                 ok = await self._indirect_ping(target, live)
@@ -132,6 +152,9 @@ class MeshNode:
                 target.state = ALIVE
                 target.last_ack = now
                 target.suspect_since = None
+                if was != ALIVE:
+                    # rejoin after suspicion: reconverge via page repair
+                    self._schedule_repair(target)
             else:
                 if target.state == ALIVE:
                     target.state = SUSPECT
@@ -139,8 +162,99 @@ class MeshNode:
                 elif target.state == SUSPECT and now - (target.suspect_since or now) > self.suspect_timeout:
                     target.state = DEAD
                     log.info("mesh member %s declared dead", target.node_id)
+        # also probe one dead member occasionally so partitions can HEAL
+        dead = [m for m in self.members.values() if m.state == DEAD]
+        if dead:
+            target = random.choice(dead)
+            if await self._ping(target):
+                target.state = ALIVE
+                target.last_ack = now
+                target.suspect_since = None
+                log.info("mesh member %s returned from the dead", target.node_id)
+                self._schedule_repair(target)
+        self._update_partition_state()
         # sync round: ship new ops to every live peer
         await asyncio.gather(*(self._sync_peer(m) for m in self.members.values() if m.state != DEAD))
+
+    # ---- partition detection / heal (reference partition.rs) ---------------
+    def _update_partition_state(self) -> None:
+        total = len(self.members)
+        if total == 0:
+            self.partitioned = False
+            return
+        unreachable = sum(1 for m in self.members.values() if m.state != ALIVE)
+        now_partitioned = unreachable * 2 > total
+        if now_partitioned and not self.partitioned:
+            self.partitioned = True
+            self._partition_since = time.monotonic()
+            log.warning("mesh partition detected: %d/%d members unreachable", unreachable, total)
+        elif not now_partitioned and self.partitioned:
+            self.partitioned = False
+            dur = time.monotonic() - (self._partition_since or time.monotonic())
+            self._partition_since = None
+            self.partition_heals += 1
+            log.info("mesh partition healed after %.1fs; starting anti-entropy repair", dur)
+            for m in self.members.values():
+                if m.state == ALIVE:
+                    self._schedule_repair(m)
+
+    def partition_state(self) -> dict:
+        total = len(self.members)
+        unreachable = sum(1 for m in self.members.values() if m.state != ALIVE)
+        return {
+            "partitioned": self.partitioned,
+            "members": total,
+            "unreachable": unreachable,
+            "since": self._partition_since,
+            "heals": self.partition_heals,
+        }
+
+    # ---- incremental page repair (reference tree_sync.rs:38-67) -------------
+    def _schedule_repair(self, m: Member) -> None:
+        existing = self._repair_tasks.get(m.node_id)
+        if existing is not None and not existing.done():
+            return
+        self._repair_tasks[m.node_id] = asyncio.ensure_future(self._repair_peer(m))
+
+    async def _repair_peer(self, m: Member) -> None:
+        """Pull the peer's state in byte-capped pages per namespace and merge
+        through the CRDT (idempotent) — incremental reconvergence instead of
+        a full join snapshot."""
+        for ns in self.kv.namespaces():
+            cursor = ""
+            retries = 0
+            while True:
+                try:
+                    async with self._session.post(
+                        m.url + "/mesh/repair",
+                        json={"from": self.node_id, "ns": ns, "cursor": cursor,
+                              "max_bytes": self.max_repair_page_bytes},
+                        timeout=aiohttp.ClientTimeout(total=5),
+                    ) as resp:
+                        if resp.status != 200:
+                            raise RuntimeError(f"repair HTTP {resp.status}")
+                        page = await resp.json()
+                except Exception as exc:
+                    retries += 1
+                    if retries > self.repair_max_retries:
+                        log.warning("repair of %s ns=%s abandoned: %s", m.node_id, ns, exc)
+                        return
+                    await asyncio.sleep(0.2 * retries)
+                    continue
+                retries = 0
+                for opd in page.get("entries", []):
+                    op = Op.from_dict(opd)
+                    self.kv.apply_remote(op, op.hlc[2])
+                if page.get("done", True):
+                    break
+                cursor = page.get("next_cursor", "")
+        self.repairs_completed += 1
+
+    def handle_repair(self, payload: dict) -> dict:
+        return self.kv.repair_page(
+            payload.get("ns", ""), payload.get("cursor", ""),
+            int(payload.get("max_bytes") or self.max_repair_page_bytes),
+        )
 
     async def _ping(self, m: Member) -> bool:
         try:
@@ -174,19 +288,26 @@ class MeshNode:
         return False
 
     async def _sync_peer(self, m: Member) -> None:
-        since = self._sent_watermarks.get(m.node_id, 0)
-        ops = self.kv.ops_since(since)
-        if not ops:
-            return
-        try:
-            async with self._session.post(
-                m.url + "/mesh/sync",
-                json={"from": self.node_id, "ops": [op.to_dict() for op in ops]},
-            ) as resp:
-                if resp.status == 200:
+        """Ship new ops in bounded chunks (reference transport/chunking.rs):
+        a post-partition backlog drains over a few capped POSTs per round
+        instead of one unbounded body."""
+        for _ in range(self.max_sync_posts_per_round):
+            since = self._sent_watermarks.get(m.node_id, 0)
+            ops = self.kv.ops_since(since)[: self.max_sync_ops_per_post]
+            if not ops:
+                return
+            try:
+                async with self._session.post(
+                    m.url + "/mesh/sync",
+                    json={"from": self.node_id, "ops": [op.to_dict() for op in ops]},
+                ) as resp:
+                    if resp.status != 200:
+                        return
                     self._sent_watermarks[m.node_id] = ops[-1].seq
-        except Exception:
-            pass
+            except Exception:
+                return
+            if len(ops) < self.max_sync_ops_per_post:
+                return
 
     # ---- membership helpers -------------------------------------------------
     def _member_dicts(self) -> List[dict]:

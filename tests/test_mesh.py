@@ -218,3 +218,132 @@ def test_tree_delta_sync_prewarms_replica(runner):
             await sb.close()
 
     runner(run())
+
+
+class TestPartitionAndRepair:
+    """Mesh hardening (reference mesh/src/partition.rs + tree_sync.rs:38-67
+    repair protocol): partition detection, incremental page repair replacing
+    join-snapshot healing, chunked sync batches."""
+
+    def test_repair_page_paging(self):
+        kv = MeshKV("a")
+        kv.register_namespace("td")
+        for i in range(50):
+            kv.put("td", f"k{i:04d}", {"tokens": list(range(20)), "tenant": f"w{i}"})
+        pages, cursor, total = 0, "", 0
+        while True:
+            page = kv.repair_page("td", cursor, max_bytes=800)
+            total += len(page["entries"])
+            pages += 1
+            if page["done"]:
+                break
+            cursor = page["next_cursor"]
+        assert total == 50
+        assert pages > 3  # byte cap forced multiple pages
+        # idempotent application on a fresh node
+        kv2 = MeshKV("b")
+        kv2.register_namespace("td")
+        cursor = ""
+        while True:
+            page = kv.repair_page("td", cursor, max_bytes=800)
+            for opd in page["entries"]:
+                op = Op.from_dict(opd)
+                kv2.apply_remote(op, op.hlc[2])
+            if page["done"]:
+                break
+            cursor = page["next_cursor"]
+        assert kv2.items("td") == kv.items("td")
+
+    def test_partition_detection_and_heal_counter(self, runner):
+        async def run():
+            a, sa = await make_node("a")
+            b, sb = await make_node("b", peers=[a])
+            c, sc = await make_node("c", peers=[a])
+            try:
+                await asyncio.sleep(0.4)
+                assert len(a.members) == 2
+                # kill both peers' inbound servers -> a sees a majority gone
+                await b.stop()
+                await sb.close()
+                await c.stop()
+                await sc.close()
+                for _ in range(80):
+                    if a.partitioned:
+                        break
+                    await asyncio.sleep(0.1)
+                assert a.partitioned
+                st = a.partition_state()
+                assert st["unreachable"] == 2 and st["members"] == 2
+            finally:
+                await a.stop()
+                await sa.close()
+
+        runner(run())
+
+    def test_kill_and_heal_reconverges_without_snapshot(self, runner):
+        """A peer that missed ops it can never receive via the op-log
+        (compacted during the outage) reconverges through DEAD->ALIVE page
+        repair — the reference's tree:req:/tree:page: behavior."""
+
+        async def run():
+            a, sa = await make_node("a")
+            b, sb = await make_node("b", peers=[a])
+            a.kv.register_namespace("td")
+            b.kv.register_namespace("td")
+            try:
+                await asyncio.sleep(0.3)
+                a.kv.put("td", "seed", {"v": 1})
+                for _ in range(30):
+                    if b.kv.get("td", "seed"):
+                        break
+                    await asyncio.sleep(0.1)
+                assert b.kv.get("td", "seed") == {"v": 1}
+                # ---- outage window: a writes 80 ops b will never see via
+                # the log (compacted away + watermark already advanced)
+                for i in range(80):
+                    a.kv.put("td", f"missed{i:03d}", {"v": i})
+                a.kv._log.clear()  # compaction during the outage
+                a._sent_watermarks["b"] = a.kv.local_seq()
+                await asyncio.sleep(0.4)  # sync rounds run; nothing can ship
+                assert b.kv.get("td", "missed000") is None
+                # ---- heal: b observed a as DEAD, then the dead-probe
+                # succeeds -> DEAD->ALIVE transition schedules page repair
+                b.members["a"].state = DEAD
+                for _ in range(100):
+                    if b.kv.get("td", "missed079") is not None:
+                        break
+                    await asyncio.sleep(0.1)
+                assert b.kv.get("td", "missed079") == {"v": 79}
+                assert all(b.kv.get("td", f"missed{i:03d}") == {"v": i} for i in range(80))
+                assert b.repairs_completed >= 1
+            finally:
+                await a.stop()
+                await b.stop()
+                await sa.close()
+                await sb.close()
+
+        runner(run())
+
+    def test_chunked_sync_drains_backlog(self, runner):
+        async def run():
+            a, sa = await make_node("a")
+            a.max_sync_ops_per_post = 20
+            b, sb = await make_node("b", peers=[a])
+            a.kv.register_namespace("x")
+            b.kv.register_namespace("x")
+            try:
+                await asyncio.sleep(0.2)
+                for i in range(150):  # backlog >> per-POST cap
+                    a.kv.put("x", f"bk{i:04d}", i)
+                for _ in range(80):
+                    if b.kv.get("x", "bk0149") is not None:
+                        break
+                    await asyncio.sleep(0.1)
+                assert b.kv.get("x", "bk0149") == 149
+            finally:
+                await a.stop()
+                await b.stop()
+                await sa.close()
+                await sb.close()
+
+        runner(run())
