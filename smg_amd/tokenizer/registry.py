@@ -122,7 +122,11 @@ class TokenizerRegistry:
         if path == "mock":
             tok = MockTokenizer(name)
         else:
-            tok = HFTokenizer(path, name)
+            # hub-aware family dispatch: HF cache resolution + tiktoken-format
+            # detection (incl. Kimi-K2), else the tokenizer.json loader
+            from .hub import load_tokenizer
+
+            tok = load_tokenizer(path, name)
         if chat_template:
             tok.chat_template = chat_template
         self._tokenizers[name] = tok
