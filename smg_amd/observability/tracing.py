@@ -95,6 +95,101 @@ class Tracer:
 GLOBAL_TRACER = Tracer(enabled=False)
 
 
+class OtlpHttpExporter:
+    """Wire-level OTLP/HTTP JSON span exporter (reference otel_trace.rs's
+    batch exporter; endpoint = `--otlp-traces-endpoint`, path /v1/traces).
+    Spans batch until `max_batch` or `flush_interval_s`, then POST as the
+    OTLP JSON mapping (resourceSpans -> scopeSpans -> spans with hex ids and
+    unix-nano timestamps) — consumable by any OTLP collector, no otel SDK
+    needed."""
+
+    def __init__(self, endpoint: str, service_name: str = "smg-amd",
+                 max_batch: int = 64, flush_interval_s: float = 5.0):
+        self.endpoint = endpoint.rstrip("/")
+        if not self.endpoint.endswith("/v1/traces"):
+            self.endpoint += "/v1/traces"
+        self.service_name = service_name
+        self.max_batch = max_batch
+        self.flush_interval_s = flush_interval_s
+        self._buf: List[Span] = []
+        self._last_flush = time.time()
+        self._task = None
+        self.exported = 0
+        self.export_errors = 0
+
+    # sync hook called from Tracer.end_span
+    def __call__(self, span: Span) -> None:
+        self._buf.append(span)
+        if len(self._buf) >= self.max_batch or (
+            time.time() - self._last_flush > self.flush_interval_s and self._buf
+        ):
+            self.schedule_flush()
+
+    def schedule_flush(self) -> None:
+        import asyncio
+
+        try:
+            loop = asyncio.get_running_loop()
+        except RuntimeError:
+            return  # no loop (sync test context); flush() can be awaited manually
+        if self._task is None or self._task.done():
+            self._task = loop.create_task(self.flush())
+
+    @staticmethod
+    def _span_json(s: Span) -> dict:
+        to_nano = lambda t: str(int(t * 1e9))
+        return {
+            "traceId": s.trace_id,
+            "spanId": s.span_id,
+            "parentSpanId": s.parent_id or "",
+            "name": s.name,
+            "kind": 2,  # SERVER
+            "startTimeUnixNano": to_nano(s.start),
+            "endTimeUnixNano": to_nano(s.end or s.start),
+            "attributes": [
+                {"key": k, "value": {"stringValue": str(v)}} for k, v in s.attributes.items()
+            ],
+            "events": [
+                {"timeUnixNano": to_nano(e.get("ts", s.start)), "name": e.get("name", "")}
+                for e in s.events
+            ],
+        }
+
+    def payload(self, spans: List[Span]) -> dict:
+        return {
+            "resourceSpans": [{
+                "resource": {"attributes": [
+                    {"key": "service.name", "value": {"stringValue": self.service_name}}
+                ]},
+                "scopeSpans": [{
+                    "scope": {"name": "smg_amd.tracing"},
+                    "spans": [self._span_json(s) for s in spans],
+                }],
+            }]
+        }
+
+    async def flush(self) -> int:
+        if not self._buf:
+            return 0
+        batch, self._buf = self._buf, []
+        self._last_flush = time.time()
+        try:
+            import aiohttp
+
+            async with aiohttp.ClientSession() as session:
+                async with session.post(
+                    self.endpoint, json=self.payload(batch),
+                    timeout=aiohttp.ClientTimeout(total=10),
+                ) as resp:
+                    if resp.status // 100 == 2:
+                        self.exported += len(batch)
+                        return len(batch)
+                    self.export_errors += 1
+        except Exception:
+            self.export_errors += 1
+        return 0
+
+
 def inject_trace_context(headers: Dict[str, str]) -> None:
     """Stamp the current span's W3C traceparent onto outbound worker headers
     (reference inject_trace_context_http)."""

@@ -214,6 +214,10 @@ class HttpRouter(Router):
         headers["content-type"] = "application/json"
         if req.request_id:
             headers["x-request-id"] = req.request_id
+        # W3C trace propagation to the worker (reference inject_trace_context_http)
+        from ..observability.tracing import inject_trace_context
+
+        inject_trace_context(headers)
         if worker.api_key:
             headers["authorization"] = f"Bearer {worker.api_key}"
         url = worker.url + req.path

@@ -57,7 +57,9 @@ class TickGateway:
         model_id: str = "default",
         max_new_arrivals_per_tick: int = 128,
         on_event: Optional[Callable[[int, int, int], None]] = None,
+        metrics=None,
     ):
+        self.metrics = metrics
         self.workers = list(workers)
         self.policy = policy
         self.plane = plane
@@ -160,6 +162,12 @@ class TickGateway:
                         done_now += self._complete(rid)
         self.completed_total += done_now
         self.phase_t["events"] += time.perf_counter() - tev
+        m = self.metrics
+        if m is not None and not m._null:
+            m.plane_ticks.inc()
+            n_ev = len(local_events) + sum(len(a) for a in remote_arrays)
+            if n_ev:
+                m.plane_events_received.inc(n_ev)
         return done_now
 
     def _complete(self, rid: int) -> int:
@@ -285,6 +293,7 @@ class RcclRouter(Router):
             local_engine=engine,
             model_id=self.model_id,
             on_event=self._on_event_tick_thread,
+            metrics=getattr(ctx, "metrics", None),
         )
         self._stop = False
         self.idle_sleep_s = max(1, config.rccl.tick_interval_us) / 1e6

@@ -201,6 +201,27 @@ def _ct_eq(a: str, b: str) -> bool:
 
 
 @web.middleware
+async def tracing_middleware(request: web.Request, handler):
+    """Span per request, continuing an incoming W3C traceparent; the
+    contextvar makes the span visible to routers for outbound propagation
+    (reference otel_trace.rs + http/router.rs:466)."""
+    ctx: AppContext = request.app[CTX_KEY]
+    tracer = getattr(ctx, "tracer", None)
+    if tracer is None or not tracer.enabled:
+        return await handler(request)
+    with tracer.span(
+        f"http {request.path}",
+        traceparent=request.headers.get("traceparent"),
+        method=request.method,
+        path=request.path,
+        request_id=request.get("request_id", ""),
+    ) as span:
+        resp = await handler(request)
+        span.set("status", getattr(resp, "status", 0))
+        return resp
+
+
+@web.middleware
 async def tenant_middleware(request: web.Request, handler):
     ctx: AppContext = request.app[CTX_KEY]
     if "tenant_id" not in request and ctx.config.trust_tenant_header:
@@ -359,7 +380,49 @@ async def get_server_info(request):
 
 async def metrics_endpoint(request):
     ctx: AppContext = request.app[CTX_KEY]
+    _refresh_scrape_gauges(ctx)
     return web.Response(body=ctx.metrics.export(), content_type="text/plain")
+
+
+def _refresh_scrape_gauges(ctx: "AppContext") -> None:
+    """Just-in-time gauges computed at scrape (mesh partition state, worker
+    pool sizes, GPU-tree occupancy, tokenizer registry) — state the exporter
+    reads rather than events that push."""
+    m = ctx.metrics
+    if m._null:
+        return
+    try:
+        by_model = {}
+        for w in ctx.worker_registry.all():
+            by_model[w.model_id] = by_model.get(w.model_id, 0) + 1
+            m.worker_requests_active.labels(w.url).set(w.active_requests)
+        for model, n in by_model.items():
+            m.worker_pool_size.labels(model).set(n)
+        if ctx.mesh is not None:
+            st = ctx.mesh.partition_state()
+            m.mesh_partitioned.set(1 if st["partitioned"] else 0)
+            m.mesh_peers.set(len(ctx.mesh.live_members()))
+        if ctx.tokenizer_registry is not None:
+            m.tokenizers_registered.set(len(ctx.tokenizer_registry.list()))
+        for policy in ctx.policy_registry.all_policies():
+            trees = getattr(policy, "token_trees", None)
+            if not trees:
+                continue
+            for model_id, tree in list(trees.items())[:16]:
+                stats = getattr(tree, "stats", None)
+                if callable(stats):
+                    try:
+                        st = stats()
+                        if isinstance(st, dict) and "live_nodes" in st:
+                            m.gpu_tree_nodes_live.labels(model_id).set(int(st["live_nodes"]))
+                            m.gpu_tree_nodes_allocated.labels(model_id).set(
+                                int(st.get("allocated_nodes", 0)))
+                    except Exception:
+                        pass
+                elif hasattr(tree, "__len__"):
+                    m.tree_size.labels(model_id).set(len(tree))
+    except Exception:  # scrape must never fail the endpoint
+        pass
 
 
 async def engine_metrics(request):
@@ -473,6 +536,7 @@ def build_app(ctx: AppContext) -> web.Application:
     app = web.Application(
         middlewares=[
             request_id_middleware,
+            tracing_middleware,
             metrics_middleware,
             auth_middleware,
             tenant_middleware,

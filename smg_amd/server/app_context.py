@@ -50,6 +50,15 @@ class AppContext:
             policy_registry=self.policy_registry,
         )
         self.router_manager = None  # wired by startup()
+        # distributed tracing (reference otel_trace.rs): spans + optional
+        # wire-level OTLP/HTTP export when --otlp-traces-endpoint is set
+        from ..observability.tracing import OtlpHttpExporter, Tracer
+
+        exporter = (
+            OtlpHttpExporter(config.trace.otlp_endpoint)
+            if (config.trace.enabled and config.trace.otlp_endpoint) else None
+        )
+        self.tracer = Tracer(enabled=config.trace.enabled, otlp_exporter=exporter)
         self.tokenizer_registry = None
         self.rate_limiter = None
         self.scheduler = None
