@@ -729,6 +729,7 @@ async def startup(config: RouterConfig, serve: bool = True) -> AppContext:
         sched_cfg = SchedulerConfig.from_yaml(config.priority_scheduler.config_path)
         sched_cfg.default_class = config.priority_scheduler.default_max_class
         ctx.scheduler = PriorityScheduler(sched_cfg, ctx.worker_registry, ctx.metrics)
+        ctx.scheduler.start_sampler()
     if config.tenant_rate_limit.enabled:
         from ..rate_limit.tenant import RateLimitManager, TenantRateLimitSettings
 

@@ -123,6 +123,8 @@ class AppContext:
 
     async def shutdown(self) -> None:
         await self.worker_monitor.stop()
+        if self.scheduler is not None and hasattr(self.scheduler, "stop"):
+            await self.scheduler.stop()
         if self.kv_event_monitor is not None:
             await self.kv_event_monitor.stop()
         if self.mesh is not None:

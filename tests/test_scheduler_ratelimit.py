@@ -9,8 +9,11 @@ from smg_amd.scheduler.engine import CLASSES, PriorityScheduler, SchedulerConfig
 
 
 def sched(**kw):
-    cfg = SchedulerConfig(min_slots=kw.pop("slots", 4), queue_size=kw.pop("queue_size", 8),
+    queue_size = kw.pop("queue_size", 8)
+    cfg = SchedulerConfig(min_slots=kw.pop("slots", 4),
                           queue_timeout_secs=kw.pop("timeout", 0.5), **kw)
+    for c in cfg.classes.values():
+        c.queue_size = queue_size
     return PriorityScheduler(cfg)
 
 
@@ -69,21 +72,96 @@ class TestScheduler:
 
         runner(run())
 
-    def test_preemption_rejects_queued_bulk(self, runner):
+    def test_class_queue_overflow_rejects(self, runner):
         async def run():
             s = sched(slots=1, queue_size=1, timeout=0.5)
-            assert await s.acquire("system")
-            bulk_w = asyncio.ensure_future(s.acquire("bulk"))
-            await asyncio.sleep(0.05)  # bulk now fills the queue
-            sys_w = asyncio.ensure_future(s.acquire("interactive"))
-            await asyncio.sleep(0.05)
-            assert s.preempted == 1  # bulk waiter evicted for the interactive one
-            s.release("system")
-            assert await asyncio.wait_for(sys_w, 2.0)
-            with pytest.raises((asyncio.CancelledError, asyncio.TimeoutError)):
-                await bulk_w
+            assert await s.acquire("bulk")
+            w1 = asyncio.ensure_future(s.acquire("bulk"))  # fills bulk's queue
+            await asyncio.sleep(0.02)
+            assert not await s.acquire("bulk")  # fixed FIFO depth: reject
+            assert s.rejected == 1
+            s.release("bulk")
+            assert await asyncio.wait_for(w1, 2.0)
 
         runner(run())
+
+    def test_inflight_preemption_50ms_budget(self, runner):
+        """engine.rs:193-232: a saturated higher-class arrival cancels the
+        newest LOWER-class inflight request and takes its slot within the
+        50 ms wait budget; the victim's admit() answers 429 preempted."""
+        from aiohttp import web
+        from aiohttp.test_utils import TestClient, TestServer
+
+        async def run():
+            s = sched(slots=1, timeout=2.0)
+
+            async def slow_handler(request):
+                await asyncio.sleep(30)
+                return web.Response(text="done")
+
+            async def admitted(request):
+                return await s.admit(request, slow_handler)
+
+            app = web.Application()
+            app.router.add_get("/work", admitted)
+            client = TestClient(TestServer(app))
+            await client.start_server()
+            try:
+                victim = asyncio.ensure_future(
+                    client.get("/work", headers={"x-smg-priority": "bulk"}))
+                for _ in range(100):
+                    if s.total_in_use() == 1:
+                        break
+                    await asyncio.sleep(0.01)
+                assert s.stats()["inflight"] == 1
+                t0 = asyncio.get_event_loop().time()
+                assert await s.acquire("interactive")  # preempts the bulk body
+                took = asyncio.get_event_loop().time() - t0
+                assert took < 1.0  # well within budget+scheduling slack
+                assert s.preempted_inflight == 1
+                resp = await asyncio.wait_for(victim, 5.0)
+                assert resp.status == 429
+                assert resp.headers.get("x-smg-preempted") == "1"
+                s.release("interactive")
+            finally:
+                await client.close()
+
+        runner(run())
+
+    def test_bulk_never_preempts(self, runner):
+        async def run():
+            s = sched(slots=1, timeout=0.1)
+            assert await s.acquire("default")
+            assert not await s.acquire("bulk")  # queues then times out
+            assert s.preempted_inflight == 0
+
+        runner(run())
+
+    def test_capacity_shrink_clamps_priority_ordered(self):
+        s = sched(slots=8)
+        for c, cc in s.config.classes.items():
+            cc.floor = {"system": 4, "interactive": 4, "default": 4, "bulk": 4}[c]
+        s.apply_new_capacity(10)  # desired 16 > 10: clamp fills system->bulk
+        assert s.reserved("system") == 4
+        assert s.reserved("interactive") == 4
+        assert s.reserved("default") == 2
+        assert s.reserved("bulk") == 0
+        s.apply_new_capacity(100)  # grow restores the full shares
+        assert s.reserved("default") >= 4
+
+    def test_yaml_class_config(self, tmp_path):
+        p = tmp_path / "sched.yaml"
+        p.write_text(
+            "per_worker_concurrency: 8\n"
+            "classes:\n  interactive: {floor: 3, share: 0.4, queue_size: 9}\n"
+            "reservations:\n  bulk: 0.05\n"
+            "tenants:\n  vip: interactive\n"
+        )
+        cfg = SchedulerConfig.from_yaml(str(p))
+        assert cfg.classes["interactive"].floor == 3
+        assert cfg.classes["interactive"].queue_size == 9
+        assert cfg.classes["bulk"].share == 0.05
+        assert cfg.tenant_classes["vip"] == "interactive"
 
     def test_classify(self):
         cfg = SchedulerConfig(tenant_classes={"vip": "interactive"})
