@@ -105,17 +105,22 @@ class StdioTransport:
 
 
 class HttpTransport:
-    """Streamable-HTTP MCP transport: POST JSON-RPC to the server URL."""
+    """Streamable-HTTP MCP transport: POST JSON-RPC to the server URL, over
+    a pooled connector (reference core/ connection pool)."""
 
-    def __init__(self, url: str):
+    def __init__(self, url: str, pool_size: int = 8):
         self.url = url
+        self.pool_size = pool_size
         self._session = None
         self._ids = itertools.count(1)
 
     async def start(self) -> None:
         import aiohttp
 
-        self._session = aiohttp.ClientSession(timeout=aiohttp.ClientTimeout(total=30))
+        self._session = aiohttp.ClientSession(
+            connector=aiohttp.TCPConnector(limit=self.pool_size),
+            timeout=aiohttp.ClientTimeout(total=30),
+        )
 
     async def request(self, method: str, params: Optional[dict] = None, timeout: float = 30.0) -> dict:
         msg = {"jsonrpc": "2.0", "id": next(self._ids), "method": method, "params": params or {}}
@@ -130,6 +135,154 @@ class HttpTransport:
     async def close(self) -> None:
         if self._session:
             await self._session.close()
+
+
+class SseTransport:
+    """MCP HTTP+SSE transport (reference core/transports SSE): a long-lived
+    GET to the SSE endpoint carries server->client messages; the server's
+    initial `endpoint` event names the URL to POST client->server JSON-RPC.
+    Responses are matched to requests by id off the event stream."""
+
+    def __init__(self, url: str):
+        self.url = url
+        self._session = None
+        self._post_url: Optional[str] = None
+        self._ids = itertools.count(1)
+        self._pending: Dict[object, asyncio.Future] = {}
+        self._reader_task: Optional[asyncio.Task] = None
+        self._endpoint_ready: Optional[asyncio.Event] = None
+
+    async def start(self) -> None:
+        import aiohttp
+
+        self._session = aiohttp.ClientSession(timeout=aiohttp.ClientTimeout(total=None, sock_read=None))
+        self._endpoint_ready = asyncio.Event()
+        self._reader_task = asyncio.ensure_future(self._read_stream())
+        await asyncio.wait_for(self._endpoint_ready.wait(), 10.0)
+
+    async def _read_stream(self) -> None:
+        from urllib.parse import urljoin
+
+        try:
+            async with self._session.get(self.url, headers={"Accept": "text/event-stream"}) as resp:
+                if resp.status != 200:
+                    raise McpError(f"mcp sse {resp.status}")
+                event, data_lines = None, []
+                async for raw in resp.content:
+                    line = raw.decode("utf-8", "replace").rstrip("\n\r")
+                    if line.startswith("event:"):
+                        event = line[6:].strip()
+                    elif line.startswith("data:"):
+                        data_lines.append(line[5:].strip())
+                    elif line == "":
+                        data = "\n".join(data_lines)
+                        data_lines = []
+                        if event == "endpoint":
+                            self._post_url = urljoin(self.url, data)
+                            self._endpoint_ready.set()
+                        elif data:
+                            try:
+                                msg = json.loads(data)
+                            except json.JSONDecodeError:
+                                continue
+                            fut = self._pending.pop(msg.get("id"), None)
+                            if fut is not None and not fut.done():
+                                fut.set_result(msg)
+                        event = None
+        except asyncio.CancelledError:
+            pass
+        except Exception as exc:
+            log.debug("mcp sse stream ended: %s", exc)
+        for fut in self._pending.values():
+            if not fut.done():
+                fut.set_exception(McpError("mcp sse stream closed"))
+        self._pending.clear()
+
+    async def request(self, method: str, params: Optional[dict] = None, timeout: float = 30.0) -> dict:
+        if self._post_url is None or (self._reader_task and self._reader_task.done()):
+            raise McpError("sse transport not running")
+        mid = next(self._ids)
+        msg = {"jsonrpc": "2.0", "id": mid, "method": method, "params": params or {}}
+        fut = asyncio.get_event_loop().create_future()
+        self._pending[mid] = fut
+        async with self._session.post(self._post_url, json=msg) as resp:
+            if resp.status not in (200, 202):
+                self._pending.pop(mid, None)
+                raise McpError(f"mcp sse post {resp.status}")
+        resp_msg = await asyncio.wait_for(fut, timeout)
+        if "error" in resp_msg:
+            raise McpError(str(resp_msg["error"]))
+        return resp_msg.get("result", {})
+
+    async def close(self) -> None:
+        if self._reader_task:
+            self._reader_task.cancel()
+            try:
+                await self._reader_task
+            except (asyncio.CancelledError, Exception):
+                pass
+        if self._session:
+            await self._session.close()
+
+
+class ResilientTransport:
+    """Reconnect-with-backoff wrapper (reference core/ session reconnect;
+    backoff ladder mirrors kv_event_monitor.rs 100ms -> 30s).  A failed
+    request tears the inner transport down, reconnects, replays the MCP
+    `initialize` handshake, and retries the request once per attempt."""
+
+    BACKOFF_S = (0.1, 0.5, 2.0, 10.0, 30.0)
+
+    def __init__(self, make_transport, max_attempts: int = 4, on_reconnect=None):
+        self._make = make_transport
+        self._inner = None
+        self.max_attempts = max_attempts
+        self.reconnects = 0
+        self._on_reconnect = on_reconnect  # async hook(transport) after handshake
+
+    async def start(self) -> None:
+        self._inner = self._make()
+        await self._inner.start()
+
+    async def _handshake(self) -> None:
+        await self._inner.request(
+            "initialize",
+            {"protocolVersion": "2024-11-05", "capabilities": {},
+             "clientInfo": {"name": "smg", "version": "0.1"}},
+        )
+        if self._on_reconnect is not None:
+            await self._on_reconnect(self._inner)
+
+    async def request(self, method: str, params: Optional[dict] = None, timeout: float = 30.0) -> dict:
+        last: Optional[Exception] = None
+        for attempt in range(self.max_attempts):
+            if self._inner is None:
+                await self.start()
+            try:
+                return await self._inner.request(method, params, timeout)
+            except (McpError, asyncio.TimeoutError, ConnectionError, OSError) as exc:
+                last = exc
+                try:
+                    await self._inner.close()
+                except Exception:
+                    pass
+                self._inner = None
+                if attempt + 1 >= self.max_attempts:
+                    break
+                await asyncio.sleep(self.BACKOFF_S[min(attempt, len(self.BACKOFF_S) - 1)])
+                try:
+                    await self.start()
+                    self.reconnects += 1
+                    await self._handshake()
+                except Exception as exc2:
+                    last = exc2
+                    self._inner = None
+        raise McpError(f"mcp request failed after {self.max_attempts} attempts: {last}")
+
+    async def close(self) -> None:
+        if self._inner is not None:
+            await self._inner.close()
+            self._inner = None
 
 
 # ---------------------------------------------------------------------------
@@ -200,27 +353,33 @@ class McpOrchestrator:
         return orch
 
     async def register_server(self, cfg: McpServerConfig) -> int:
-        """Connect + list tools; returns number of tools discovered."""
-        if cfg.transport == "stdio":
-            t = StdioTransport(cfg.command or [], cfg.env)
-        elif cfg.transport == "http":
-            t = HttpTransport(cfg.url or "")
-        else:
+        """Connect + list tools; returns number of tools discovered.  Every
+        transport is wrapped in ResilientTransport: a dead session reconnects
+        with backoff, replays initialize, and refreshes this server's
+        inventory (reference core/ session + reconnect)."""
+
+        def make():
+            if cfg.transport == "stdio":
+                return StdioTransport(cfg.command or [], cfg.env)
+            if cfg.transport == "http":
+                return HttpTransport(cfg.url or "")
+            if cfg.transport == "sse":
+                return SseTransport(cfg.url or "")
             raise McpError(f"unknown transport {cfg.transport}")
+
+        async def refresh_inventory(inner) -> None:
+            result = await inner.request("tools/list")
+            for q in [q for q in self.inventory if q.startswith(cfg.name + ".")]:
+                del self.inventory[q]
+            for tool in result.get("tools", []):
+                self.inventory[f"{cfg.name}.{tool['name']}"] = tool
+
+        t = ResilientTransport(make, on_reconnect=refresh_inventory)
         await t.start()
-        await t.request(
-            "initialize",
-            {"protocolVersion": "2024-11-05", "capabilities": {}, "clientInfo": {"name": "smg", "version": "0.1"}},
-        )
-        result = await t.request("tools/list")
+        await t._handshake()  # initialize + initial tools/list
         self.servers[cfg.name] = t
         self.configs[cfg.name] = cfg
-        n = 0
-        for tool in result.get("tools", []):
-            qname = f"{cfg.name}.{tool['name']}"
-            self.inventory[qname] = tool
-            n += 1
-        return n
+        return sum(1 for q in self.inventory if q.startswith(cfg.name + "."))
 
     async def start_all(self) -> None:
         for cfg in list(self.configs.values()):

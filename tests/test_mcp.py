@@ -109,3 +109,94 @@ def test_remove_server(runner):
         assert orch.inventory == {}
 
     runner(run())
+
+
+class TestSseTransport:
+    """MCP HTTP+SSE transport (reference core/transports SSE): endpoint
+    event handshake + id-matched responses off the event stream."""
+
+    @staticmethod
+    async def make_sse_server():
+        import asyncio
+
+        from aiohttp import web
+        from aiohttp.test_utils import TestServer
+
+        queues = {}
+
+        async def sse(request):
+            resp = web.StreamResponse(headers={"Content-Type": "text/event-stream"})
+            await resp.prepare(request)
+            q = asyncio.Queue()
+            queues["q"] = q
+            await resp.write(b"event: endpoint\ndata: /messages\n\n")
+            try:
+                while True:
+                    msg = await q.get()
+                    await resp.write(b"event: message\ndata: " + json.dumps(msg).encode() + b"\n\n")
+            except (ConnectionResetError, asyncio.CancelledError):
+                pass
+            return resp
+
+        async def messages(request):
+            msg = await request.json()
+            method = msg.get("method")
+            if method == "initialize":
+                result = {"serverInfo": {"name": "sse-mock"}}
+            elif method == "tools/list":
+                result = {"tools": [{"name": "ping", "description": "pong",
+                                     "inputSchema": {"type": "object"}}]}
+            elif method == "tools/call":
+                result = {"content": [{"type": "text",
+                                       "text": msg["params"]["arguments"].get("x", "") + "!"}]}
+            else:
+                result = {}
+            await queues["q"].put({"jsonrpc": "2.0", "id": msg["id"], "result": result})
+            return web.json_response({}, status=202)
+
+        app = web.Application()
+        app.router.add_get("/sse", sse)
+        app.router.add_post("/messages", messages)
+        server = TestServer(app)
+        await server.start_server()
+        return server
+
+    def test_sse_register_and_call(self, runner):
+        async def run():
+            server = await self.make_sse_server()
+            orch = McpOrchestrator()
+            try:
+                cfg = McpServerConfig(name="ssemock", transport="sse",
+                                      url=f"http://127.0.0.1:{server.port}/sse")
+                n = await orch.register_server(cfg)
+                assert n == 1 and "ssemock.ping" in orch.inventory
+                res = await orch.call_tool("ssemock.ping", {"x": "hello"})
+                assert res["content"][0]["text"] == "hello!"
+            finally:
+                await orch.shutdown()
+                await server.close()
+
+        runner(run())
+
+
+def test_reconnect_after_server_death(runner):
+    """ResilientTransport: kill the stdio server mid-session; the next
+    request reconnects with backoff, replays initialize, refreshes the
+    inventory, and succeeds."""
+
+    async def run():
+        orch = McpOrchestrator()
+        await orch.register_server(make_cfg())
+        res = await orch.call_tool("mock.add", {"a": 1, "b": 1})
+        assert res
+        # kill the inner subprocess
+        transport = orch.servers["mock"]
+        transport._inner._proc.kill()
+        await transport._inner._proc.wait()
+        res = await orch.call_tool("mock.add", {"a": 2, "b": 5})
+        assert res["content"][0]["text"] == "7"
+        assert transport.reconnects >= 1
+        assert "mock.add" in orch.inventory  # inventory refreshed on reconnect
+        await orch.shutdown()
+
+    runner(run())
