@@ -32,10 +32,12 @@ def build(verbose: bool = True, force: bool = False) -> Path:
     if not force and _newer(out, srcs):
         return out
     py_inc = sysconfig.get_paths()["include"]
+    mode = os.environ.get("SMG_BUILD_MODE", "release")
+    opt_flags = ["-O3"] if mode != "debug" else ["-O1", "-g", "-UNDEBUG"]
     cmd = [
         "hipcc",
         f"--offload-arch={ARCH}",
-        "-O3",
+        *opt_flags,
         "-std=c++17",
         "-fPIC",
         "-shared",
@@ -58,5 +60,37 @@ def build(verbose: bool = True, force: bool = False) -> Path:
     return out
 
 
+def build_sanitized(verbose: bool = False, force: bool = False) -> Path:
+    """ASan+UBSan build of the host C++ self-test (SURVEY §5.2 sanitizer
+    discipline): pure host code, so plain g++ compiles it — no GPU needed.
+    Run by tests/test_sanitizer.py in CPU CI."""
+    out = CSRC / "host_tree_selftest"
+    srcs = [CSRC / "host_tree_selftest.cpp", CSRC / "host_tree.cpp"]
+    if not force and _newer(out, srcs):
+        return out
+    cmd = [
+        "g++", "-std=c++17", "-g", "-O1",
+        "-fsanitize=address,undefined", "-fno-omit-frame-pointer",
+        "-fno-sanitize-recover=all",
+        str(CSRC / "host_tree_selftest.cpp"),
+        "-o", str(out),
+    ]
+    if verbose:
+        print("+", " ".join(cmd), flush=True)
+    subprocess.run(cmd, check=True)
+    return out
+
+
+def build_debug(verbose: bool = True, force: bool = False) -> Path:
+    """Device-debug variant of the extension (-g -O1, assertions on) for
+    kernel debugging sessions; select with SMG_BUILD_MODE=debug."""
+    os.environ["SMG_BUILD_MODE"] = "debug"
+    return build(verbose=verbose, force=True)
+
+
 if __name__ == "__main__":
-    build(force="--force" in sys.argv)
+    if "--sanitize" in sys.argv:
+        path = build_sanitized(verbose=True, force="--force" in sys.argv)
+        subprocess.run([str(path)], check=True)
+    else:
+        build(force="--force" in sys.argv)
