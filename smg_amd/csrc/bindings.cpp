@@ -64,6 +64,10 @@ int smg_rope_prefill_launch_gqa(const void* qkv, const void* freqs, const void* 
                                 void* k_out, void* v_out, int B, int L, int n_heads,
                                 int n_kv_heads, int max_seq, int head_dim, void* stream,
                                 int kv_fp8);
+int smg_attn_decode_launch_split(const void* q, const void* k, const void* v, const void* pos,
+                                 void* out, void* part_out, void* part_ml, int n_slots,
+                                 int n_heads, int n_kv_heads, int n_split, int max_seq,
+                                 int head_dim, float scale, void* stream, int kv_fp8);
 int smg_attn_decode_launch_ex(const void* q, const void* k, const void* v, const void* pos,
                               void* out, int n_slots, int n_heads, int max_seq, int head_dim,
                               float scale, void* stream, int kv_fp8);
@@ -369,6 +373,24 @@ PYBIND11_MODULE(_core, m) {
           py::arg("q"), py::arg("k"), py::arg("v"), py::arg("pos"), py::arg("out"),
           py::arg("n_slots"), py::arg("n_heads"), py::arg("max_seq"), py::arg("head_dim"),
           py::arg("scale"), py::arg("stream"), py::arg("kv_fp8") = 0, py::arg("n_kv_heads") = 0);
+    // T-split (flash-decoding) decode attention + LSE merge (v8): restores
+    // chip occupancy when n_slots*n_kv_heads alone under-fills the CUs
+    m.def("attn_decode_split",
+          [](uintptr_t q, uintptr_t k, uintptr_t v, uintptr_t pos, uintptr_t out,
+             uintptr_t part_out, uintptr_t part_ml, int n_slots, int n_heads, int n_kv_heads,
+             int n_split, int max_seq, int head_dim, float scale, uintptr_t stream, int kv_fp8) {
+              if (n_kv_heads <= 0) n_kv_heads = n_heads;
+              int rc = smg_attn_decode_launch_split(
+                  (const void*)q, (const void*)k, (const void*)v, (const void*)pos, (void*)out,
+                  (void*)part_out, (void*)part_ml, n_slots, n_heads, n_kv_heads, n_split,
+                  max_seq, head_dim, scale, (void*)stream, kv_fp8);
+              if (rc != 0)
+                  throw std::runtime_error("attn_decode_split launch failed rc=" + std::to_string(rc));
+          },
+          py::arg("q"), py::arg("k"), py::arg("v"), py::arg("pos"), py::arg("out"),
+          py::arg("part_out"), py::arg("part_ml"), py::arg("n_slots"), py::arg("n_heads"),
+          py::arg("n_kv_heads"), py::arg("n_split"), py::arg("max_seq"), py::arg("head_dim"),
+          py::arg("scale"), py::arg("stream"), py::arg("kv_fp8") = 0);
     // fused rope + KV-store + q-pack (fused_decode.hip)
     m.def("rope_kv_store",
           [](uintptr_t qkv, uintptr_t freqs, uintptr_t pos, uintptr_t k_cache, uintptr_t v_cache,

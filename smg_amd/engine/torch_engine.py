@@ -126,6 +126,8 @@ class _Request:
 class TorchEngine:
     _flash_gqa = False
     _sdpa_gqa = False
+    _attn_split = 1
+    _hip_attn_split = None
 
     def __init__(self, cfg: Optional[TorchEngineConfig] = None, device: str = "cuda:0", graphs: bool = False):
         if device.startswith("cuda") and not torch.cuda.is_available():
@@ -201,6 +203,28 @@ class TorchEngine:
                     self._attn_out = torch.zeros(
                         c.max_slots, c.n_heads, c.head_dim, device=self.device, dtype=self.dtype
                     )
+                    # v8 T-split (flash-decoding): when slots*kv_heads alone
+                    # under-fills the chip (GQA quarters the wave count),
+                    # split each KV window over n_split waves + LSE merge.
+                    # Target >=8192 waves (~1.6x the resident wave slots).
+                    waves = c.max_slots * c.kv_heads
+                    split = 1
+                    while split < 8 and waves * split < 8192:
+                        split *= 2
+                    split = int(_os.environ.get("SMG_ATTN_SPLIT", split) or split)
+                    self._attn_split = max(1, min(16, split))
+                    if self._attn_split > 1 and hasattr(_core, "attn_decode_split"):
+                        self._hip_attn_split = _core.attn_decode_split
+                        G = c.gqa_group
+                        self._attn_part = torch.zeros(
+                            c.max_slots, c.kv_heads, self._attn_split, G, c.head_dim,
+                            device=self.device, dtype=torch.float32)
+                        self._attn_ml = torch.zeros(
+                            c.max_slots, c.kv_heads, self._attn_split, G, 2,
+                            device=self.device, dtype=torch.float32)
+                    else:
+                        self._attn_split = 1
+                        self._hip_attn_split = None
                 # fused rope+KV-store+q-pack and silu*mul (csrc/fused_decode.hip):
                 # removes ~10 elementwise launches per layer from the decode loop
                 if hasattr(_core, "rope_kv_store"):
@@ -727,11 +751,20 @@ class TorchEngine:
                 self.kv[li, 0].data_ptr(), self.kv[li, 1].data_ptr(), self._q_buf.data_ptr(),
                 S, c.n_heads, c.max_seq, c.head_dim, stream, kv8, c.kv_heads,
             )
-            self._hip_attn(
-                self._q_buf.data_ptr(), self.kv[li, 0].data_ptr(), self.kv[li, 1].data_ptr(),
-                self._pos_i32.data_ptr(), self._attn_out.data_ptr(),
-                S, c.n_heads, c.max_seq, c.head_dim, scale, stream, kv8, c.kv_heads,
-            )
+            if self._attn_split > 1:
+                self._hip_attn_split(
+                    self._q_buf.data_ptr(), self.kv[li, 0].data_ptr(), self.kv[li, 1].data_ptr(),
+                    self._pos_i32.data_ptr(), self._attn_out.data_ptr(),
+                    self._attn_part.data_ptr(), self._attn_ml.data_ptr(),
+                    S, c.n_heads, c.kv_heads, self._attn_split, c.max_seq, c.head_dim,
+                    scale, stream, kv8,
+                )
+            else:
+                self._hip_attn(
+                    self._q_buf.data_ptr(), self.kv[li, 0].data_ptr(), self.kv[li, 1].data_ptr(),
+                    self._pos_i32.data_ptr(), self._attn_out.data_ptr(),
+                    S, c.n_heads, c.max_seq, c.head_dim, scale, stream, kv8, c.kv_heads,
+                )
             h = torch.addmm(h, self._attn_out.view(S, c.d_model), layer.wo)
             gu = _rms(h, layer.ln2) @ layer.w13  # [S, 2F]
             self._hip_silu_mul(gu.data_ptr(), self._smul_buf.data_ptr(), S, c.d_ffn, stream)

@@ -121,3 +121,46 @@ class TestAttnDecodeKernel:
             eng_ref.step()
         out1, out2 = eng_hip.collect(r1), eng_ref.collect(r2)
         assert out1[:2] == out2[:2]
+
+    @pytest.mark.parametrize("S,H,KVH,T,D,maxseq,fp8,nsplit", [
+        (8, 16, 4, 600, 128, 704, 0, 4),
+        (8, 16, 4, 600, 128, 704, 0, 8),
+        (4, 16, 4, 63, 128, 704, 0, 4),   # T smaller than one split chunk
+        (8, 16, 4, 600, 128, 704, 1, 4),  # fp8 cache
+        (6, 8, 8, 200, 64, 256, 0, 2),    # MHA via split path
+    ])
+    def test_split_matches_sdpa(self, S, H, KVH, T, D, maxseq, fp8, nsplit):
+        """v8 T-split (flash-decoding) + LSE merge vs fp32 sdpa."""
+        dev = "cuda:0"
+        g = torch.Generator(device=dev).manual_seed(2)
+        q = torch.randn(S, H, D, generator=g, device=dev, dtype=torch.float32).to(torch.bfloat16)
+        kf32 = torch.randn(S, KVH, maxseq, D, generator=g, device=dev, dtype=torch.float32)
+        vf32 = torch.randn(S, KVH, maxseq, D, generator=g, device=dev, dtype=torch.float32)
+        if fp8:
+            k = kf32.to(torch.float8_e4m3fn); v = vf32.to(torch.float8_e4m3fn)
+            kref, vref = k.to(torch.float32), v.to(torch.float32)
+            tol = 0.25
+        else:
+            k = kf32.to(torch.bfloat16); v = vf32.to(torch.bfloat16)
+            kref, vref = k.float(), v.float()
+            tol = 0.05
+        pos = torch.randint(0, T, (S,), generator=g, device=dev, dtype=torch.int32)
+        out = torch.zeros(S, H, D, device=dev, dtype=torch.bfloat16)
+        G = H // KVH
+        part = torch.zeros(S, KVH, nsplit, G, D, device=dev, dtype=torch.float32)
+        ml = torch.zeros(S, KVH, nsplit, G, 2, device=dev, dtype=torch.float32)
+        core.attn_decode_split(
+            q.contiguous().data_ptr(), k.contiguous().data_ptr(), v.contiguous().data_ptr(),
+            pos.data_ptr(), out.data_ptr(), part.data_ptr(), ml.data_ptr(),
+            S, H, KVH, nsplit, maxseq, D, 1.0 / math.sqrt(D),
+            torch.cuda.current_stream().cuda_stream, fp8,
+        )
+        torch.cuda.synchronize()
+        kx = kref.repeat_interleave(G, dim=1)
+        vx = vref.repeat_interleave(G, dim=1)
+        kpos = torch.arange(maxseq, device=dev)
+        mask = (kpos.unsqueeze(0) <= pos.unsqueeze(1).long()).unsqueeze(1).unsqueeze(1)
+        ref = torch.nn.functional.scaled_dot_product_attention(
+            q.float().unsqueeze(2), kx, vx, attn_mask=mask).squeeze(2)
+        err = (out.float() - ref).abs().max().item()
+        assert err < tol, f"max err {err}"

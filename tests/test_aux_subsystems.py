@@ -47,11 +47,15 @@ class TestDiscovery:
 
         runner(run())
 
-    def test_k8s_requires_package(self):
+    def test_k8s_outside_cluster_raises(self, monkeypatch):
+        # the raw-HTTP watch needs no client package; outside a cluster the
+        # api_base resolution fails loudly
         from smg_amd.discovery.source import KubernetesDiscovery
 
-        with pytest.raises(RuntimeError, match="kubernetes"):
-            KubernetesDiscovery({"app": "x"}, 8000)
+        monkeypatch.delenv("KUBERNETES_SERVICE_HOST", raising=False)
+        src = KubernetesDiscovery({"app": "x"}, 8000)
+        with pytest.raises(RuntimeError, match="KUBERNETES_SERVICE_HOST"):
+            src.api_base()
 
 
 class TestTracing:
