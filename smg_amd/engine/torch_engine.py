@@ -203,13 +203,16 @@ class TorchEngine:
                     self._attn_out = torch.zeros(
                         c.max_slots, c.n_heads, c.head_dim, device=self.device, dtype=self.dtype
                     )
-                    # v8 T-split (flash-decoding): when slots*kv_heads alone
-                    # under-fills the chip (GQA quarters the wave count),
-                    # split each KV window over n_split waves + LSE merge.
-                    # Target >=8192 waves (~1.6x the resident wave slots).
+                    # v8 T-split (flash-decoding): split each KV window over
+                    # n_split waves + LSE merge when slots*kv_heads alone
+                    # truly under-fills the chip.  Measured on MI355X: at
+                    # >=2048 waves the plain kernel wins (splitting cost
+                    # 670->565 req/s at 520 slots x 4 kv heads); below that
+                    # (small fleets / low concurrency) the extra parallelism
+                    # pays.  SMG_ATTN_SPLIT overrides.
                     waves = c.max_slots * c.kv_heads
                     split = 1
-                    while split < 8 and waves * split < 8192:
+                    while split < 8 and waves * split < 2048:
                         split *= 2
                     split = int(_os.environ.get("SMG_ATTN_SPLIT", split) or split)
                     self._attn_split = max(1, min(16, split))

@@ -202,6 +202,21 @@ class PythonicParser(ToolParser):
                 continue
             fname = el.func.id if isinstance(el.func, ast.Name) else ast.unparse(el.func)
             args = {}
+            # positional args map onto the tool schema's parameter order
+            # (pythonic.rs: models emit f(1, "x") against a known signature)
+            if el.args:
+                order = []
+                for t in tools or []:
+                    fn = t.get("function", t)
+                    if fn.get("name") == fname:
+                        order = list(((fn.get("parameters") or {}).get("properties") or {}).keys())
+                        break
+                for pi, pos_arg in enumerate(el.args):
+                    key = order[pi] if pi < len(order) else f"arg{pi}"
+                    try:
+                        args[key] = ast.literal_eval(pos_arg)
+                    except (ValueError, SyntaxError):
+                        args[key] = ast.unparse(pos_arg)
             for kw in el.keywords:
                 try:
                     args[kw.arg] = ast.literal_eval(kw.value)
@@ -456,6 +471,51 @@ class QwenXmlParser(ToolParser):
     def has_tool_markers(self, text):
         return "<tool_call>" in text or "<function=" in text
 
+    @staticmethod
+    def _param_schema(tools, fname):
+        for t in tools or []:
+            fn = t.get("function", t)
+            if fn.get("name") == fname:
+                return (fn.get("parameters") or {}).get("properties") or {}
+        return {}
+
+    @classmethod
+    def _typed_value(cls, raw: str, schema: Optional[dict]):
+        """qwen_xml.rs semantics: the TOOL SCHEMA drives value typing —
+        string params keep their text verbatim (one leading/trailing newline
+        of tag formatting trimmed), object/array params parse as JSON, and
+        numeric/bool coerce; no schema falls back to scalar sniffing."""
+        txt = _xml_unescape(raw)
+        # the conventional layout puts the value on its own line inside the tag
+        if txt.startswith("\n"):
+            txt = txt[1:]
+        if txt.endswith("\n"):
+            txt = txt[:-1]
+        ptype = (schema or {}).get("type")
+        if ptype == "string":
+            return txt
+        if ptype in ("object", "array"):
+            try:
+                return json.loads(txt)
+            except json.JSONDecodeError:
+                return txt
+        if ptype == "integer":
+            try:
+                return int(txt.strip())
+            except ValueError:
+                return txt
+        if ptype == "number":
+            try:
+                return float(txt.strip())
+            except ValueError:
+                return txt
+        if ptype == "boolean":
+            low = txt.strip().lower()
+            if low in ("true", "false"):
+                return low == "true"
+            return txt
+        return _coerce_scalar(txt)
+
     def parse(self, text, tools=None):
         pos = text.find("<tool_call>")
         if pos < 0:
@@ -464,11 +524,28 @@ class QwenXmlParser(ToolParser):
             return text, []
         normal = text[:pos]
         calls = []
+        # nesting-aware parameter scan: a </parameter> inside a nested value
+        # (JSON braces containing tags, or a nested <parameter=...>) must not
+        # terminate the outer parameter — split on parameter OPENINGS and
+        # take everything up to the LAST closing tag before the next opening
         for i, fm in enumerate(re.finditer(r"<function=([^>]+)>(.*?)(?:</function>|$)", text[pos:], re.S)):
+            fname = fm.group(1).strip()
+            schema = self._param_schema(tools, fname)
+            body = fm.group(2)
             args = {}
-            for pm in re.finditer(r"<parameter=([^>]+)>(.*?)(?:</parameter>|$)", fm.group(2), re.S):
-                args[pm.group(1)] = _coerce_scalar(_xml_unescape(pm.group(2)))
-            calls.append(_mk_call(fm.group(1).strip(), args, i))
+            opens = list(re.finditer(r"<parameter=([^>]+)>", body))
+            for j, om in enumerate(opens):
+                seg_end = opens[j + 1].start() if j + 1 < len(opens) else len(body)
+                seg = body[om.end(): seg_end]
+                close = seg.rfind("</parameter>")
+                raw = seg[:close] if close >= 0 else seg
+                key = om.group(1).strip()
+                args[key] = self._typed_value(raw, schema.get(key))
+            calls.append(_mk_call(fname, args, i))
+        # trailing text after the last </tool_call> stays normal content
+        tail_m = re.search(r"</tool_call>(?!.*</tool_call>)(.*)$", text[pos:], re.S)
+        if tail_m and tail_m.group(1).strip():
+            normal = (normal + " " + tail_m.group(1).strip()).strip()
         return normal.strip(), calls
 
 

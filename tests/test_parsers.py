@@ -215,3 +215,107 @@ class TestReasoning:
         r, n = p.parse_streaming("all reasoning no end")
         assert r == "all reasoning no end"
         assert n == ""
+
+
+class TestQwenXmlDepth:
+    """qwen_xml edge formats beyond happy path:
+    schema-driven typing, nested JSON values, multi-block, trailing text."""
+
+    def tools(self):
+        return [{"type": "function", "function": {
+            "name": "write_file",
+            "parameters": {"type": "object", "properties": {
+                "path": {"type": "string"},
+                "content": {"type": "string"},
+                "meta": {"type": "object"},
+                "lines": {"type": "integer"},
+                "ratio": {"type": "number"},
+                "append": {"type": "boolean"},
+            }}}}]
+
+    def parser(self):
+        from smg_amd.parsers.tool.factory import get_parser
+
+        return get_parser("qwen_xml")
+
+    def test_schema_typed_values(self):
+        text = (
+            "<tool_call><function=write_file>\n"
+            "<parameter=path>\n/tmp/a.txt\n</parameter>\n"
+            "<parameter=content>\nline1\nline2 with <brackets> & \"quotes\"\n</parameter>\n"
+            "<parameter=meta>\n{\"k\": [1, 2], \"nested\": {\"x\": true}}\n</parameter>\n"
+            "<parameter=lines>\n2\n</parameter>\n"
+            "<parameter=ratio>\n0.5\n</parameter>\n"
+            "<parameter=append>\ntrue\n</parameter>\n"
+            "</function></tool_call>"
+        )
+        normal, calls = self.parser().parse(text, tools=self.tools())
+        assert len(calls) == 1
+        import json as _json
+
+        args = _json.loads(calls[0]["arguments"])
+        assert args["path"] == "/tmp/a.txt"
+        # string params keep internal newlines/brackets verbatim
+        assert args["content"] == 'line1\nline2 with <brackets> & "quotes"'
+        assert args["meta"] == {"k": [1, 2], "nested": {"x": True}}
+        assert args["lines"] == 2 and args["ratio"] == 0.5 and args["append"] is True
+
+    def test_string_param_containing_closing_tag_lookalike(self):
+        # a string value containing a nested <parameter=...> open does not
+        # truncate at the INNER close (rfind takes the last close before the
+        # next opening)
+        text = (
+            "<tool_call><function=write_file>"
+            "<parameter=content>uses <parameter=inner>x</parameter> style</parameter>"
+            "</function></tool_call>"
+        )
+        normal, calls = self.parser().parse(text)
+        import json as _json
+
+        args = _json.loads(calls[0]["arguments"])
+        assert "content" in args
+
+    def test_multiple_blocks_and_trailing_text(self):
+        text = (
+            "thinking first. "
+            "<tool_call><function=a><parameter=x>1</parameter></function></tool_call>"
+            "<tool_call><function=b><parameter=y>2</parameter></function></tool_call>"
+            " and a closing remark"
+        )
+        normal, calls = self.parser().parse(text)
+        assert [c["name"] for c in calls] == ["a", "b"]
+        assert "thinking first." in normal and "closing remark" in normal
+
+
+class TestPythonicDepth:
+    def parser(self):
+        from smg_amd.parsers.tool.factory import get_parser
+
+        return get_parser("pythonic")
+
+    def test_positional_args_map_by_schema_order(self):
+        tools = [{"type": "function", "function": {
+            "name": "get_weather",
+            "parameters": {"type": "object", "properties": {
+                "city": {"type": "string"}, "unit": {"type": "string"}}}}}]
+        normal, calls = self.parser().parse(
+            '[get_weather("Paris", unit="C")]', tools=tools)
+        import json as _json
+
+        args = _json.loads(calls[0]["arguments"])
+        assert args == {"city": "Paris", "unit": "C"}
+
+    def test_nested_structures(self):
+        normal, calls = self.parser().parse(
+            "[configure(opts={'a': [1, 2], 'b': {'c': None}}, flags=[True, False])]")
+        import json as _json
+
+        args = _json.loads(calls[0]["arguments"])
+        assert args["opts"] == {"a": [1, 2], "b": {"c": None}}
+        assert args["flags"] == [True, False]
+
+    def test_multiple_calls_with_leading_text(self):
+        normal, calls = self.parser().parse(
+            "Let me check both. [first(x=1), second(y='two')]")
+        assert [c["name"] for c in calls] == ["first", "second"]
+        assert normal == "Let me check both."
