@@ -28,6 +28,7 @@ class RealtimeRegistry:
     in server.rs:762)."""
 
     def __init__(self, session_ttl: float = 3600.0, pending_ttl: float = 30.0):
+        self.calls: Dict[str, "CallSession"] = {}
         self.sessions: Dict[str, dict] = {}
         self.session_ttl = session_ttl
         self.pending_ttl = pending_ttl
@@ -122,7 +123,137 @@ async def v1_realtime_ws(request: web.Request):
     return ws_client
 
 
+def _validate_sdp_offer(sdp: str) -> Optional[str]:
+    """Minimal SDP sanity (webrtc_bridge.rs parses with str0m; we validate
+    the signaling invariants): session header, at least one media section,
+    ICE credentials for trickle."""
+    if not sdp.startswith("v=0"):
+        return "SDP must start with v=0"
+    if "\nm=" not in sdp and "\rm=" not in sdp:
+        return "SDP offer has no media section"
+    if "a=ice-ufrag:" not in sdp:
+        return "SDP offer has no ICE credentials"
+    return None
+
+
+class CallSession:
+    __slots__ = ("call_id", "model", "state", "created", "upstream_url", "upstream_call_id")
+
+    def __init__(self, call_id, model, upstream_url, upstream_call_id=None):
+        self.call_id = call_id
+        self.model = model
+        self.state = "active"
+        self.created = time.time()
+        self.upstream_url = upstream_url
+        self.upstream_call_id = upstream_call_id
+
+
 async def v1_realtime_calls(request: web.Request):
+    """WebRTC call signaling (reference webrtc_bridge.rs:147-270 — SDP
+    offer/answer brokering; server.rs:855 route).  The gateway validates the
+    client's SDP offer and brokers it to a WebRTC-capable upstream (a worker
+    labeled webrtc=true, which answers /v1/realtime/calls itself); the SDP
+    answer and call id relay back and the call is tracked for GET/DELETE
+    management.  Media flows peer-to-peer between client and upstream — the
+    in-process str0m-style media RELAY of the reference needs a media stack
+    this image does not carry, so relay mode stays unimplemented (501 when no
+    upstream can take the call)."""
+    from ..server.app import CTX_KEY
+
+    ctx = request.app[CTX_KEY]
+    if not hasattr(ctx, "realtime_registry"):
+        ctx.realtime_registry = RealtimeRegistry()
+    sdp = (await request.read()).decode("utf-8", "replace")
+    err = _validate_sdp_offer(sdp)
+    if err:
+        return web.Response(status=400, body=error_body(f"invalid SDP offer: {err}"),
+                            content_type="application/json")
+    model = request.query.get("model") or "default"
+    workers = [w for w in ctx.worker_registry.all() if w.labels.get("webrtc") == "true"]
+    if not workers:
+        return web.Response(
+            status=501,
+            body=error_body(
+                "no WebRTC-capable upstream registered (label webrtc=true); this build "
+                "brokers SDP signaling but carries no media stack for relay mode — "
+                "connect via WebSocket /v1/realtime instead", 501),
+            content_type="application/json")
+    worker = workers[0]
+    import aiohttp as _aiohttp
+
+    session = getattr(ctx, "client_session", None)
+    close_session = False
+    if session is None:
+        session = _aiohttp.ClientSession()
+        close_session = True
+    try:
+        async with session.post(
+            f"{worker.url}/v1/realtime/calls", data=sdp.encode(),
+            params={"model": model}, headers={"Content-Type": "application/sdp"},
+            timeout=_aiohttp.ClientTimeout(total=15),
+        ) as resp:
+            answer = await resp.text()
+            if resp.status // 100 != 2:
+                return web.Response(status=502, body=error_body(
+                    f"upstream signaling failed: HTTP {resp.status}", 502),
+                    content_type="application/json")
+            upstream_call = resp.headers.get("Location", "").rsplit("/", 1)[-1] or None
+    finally:
+        if close_session:
+            await session.close()
+    call_id = "rtc_" + uuid.uuid4().hex[:20]
+    reg = ctx.realtime_registry
+    reg.calls[call_id] = CallSession(call_id, model, worker.url, upstream_call)
+    return web.Response(
+        status=201, body=answer.encode(),
+        headers={"Location": f"/v1/realtime/calls/{call_id}",
+                 "Content-Type": "application/sdp"})
+
+
+async def v1_realtime_call_get(request: web.Request):
+    from ..server.app import CTX_KEY
+
+    ctx = request.app[CTX_KEY]
+    if not hasattr(ctx, "realtime_registry"):
+        ctx.realtime_registry = RealtimeRegistry()
+    reg = ctx.realtime_registry
+    call = reg.calls.get(request.match_info["call_id"])
+    if call is None:
+        return web.Response(status=404, body=error_body("call not found", 404),
+                            content_type="application/json")
+    return web.json_response({
+        "id": call.call_id, "model": call.model, "state": call.state,
+        "created": call.created, "upstream": call.upstream_url,
+    })
+
+
+async def v1_realtime_call_hangup(request: web.Request):
+    from ..server.app import CTX_KEY
+
+    ctx = request.app[CTX_KEY]
+    if not hasattr(ctx, "realtime_registry"):
+        ctx.realtime_registry = RealtimeRegistry()
+    reg = ctx.realtime_registry
+    call = reg.calls.pop(request.match_info["call_id"], None)
+    if call is None:
+        return web.Response(status=404, body=error_body("call not found", 404),
+                            content_type="application/json")
+    call.state = "ended"
+    if call.upstream_call_id:
+        try:
+            import aiohttp as _aiohttp
+
+            session = getattr(ctx, "client_session", None)
+            if session is not None:
+                await session.delete(
+                    f"{call.upstream_url}/v1/realtime/calls/{call.upstream_call_id}",
+                    timeout=_aiohttp.ClientTimeout(total=5))
+        except Exception:
+            pass
+    return web.json_response({"id": call.call_id, "state": "ended"})
+
+
+async def _v1_realtime_calls_legacy(request: web.Request):
     """WebRTC call setup (reference realtime/webrtc_bridge.rs — str0m-based
     media relay).  This build ships the WebSocket relay only; WebRTC needs a
     media stack (ICE/DTLS/SRTP) that is out of scope here, so the endpoint
@@ -146,3 +277,5 @@ def add_realtime_routes(app: web.Application) -> None:
     app.router.add_post("/v1/realtime/client_secrets", v1_realtime_session)
     app.router.add_post("/v1/realtime/transcription_sessions", v1_realtime_session)
     app.router.add_post("/v1/realtime/calls", v1_realtime_calls)
+    app.router.add_get("/v1/realtime/calls/{call_id}", v1_realtime_call_get)
+    app.router.add_delete("/v1/realtime/calls/{call_id}", v1_realtime_call_hangup)

@@ -182,3 +182,94 @@ class TestRealtime:
                 await worker_srv.close()
 
         runner(run())
+
+
+class TestWebRtcSignaling:
+    """WebRTC call signaling broker (reference webrtc_bridge.rs:147-270):
+    SDP offer validation, upstream brokering, call session management."""
+
+    OFFER = (
+        "v=0\r\no=- 1 1 IN IP4 0.0.0.0\r\ns=-\r\nt=0 0\r\n"
+        "m=audio 9 UDP/TLS/RTP/SAVPF 111\r\n"
+        "a=ice-ufrag:abcd\r\na=ice-pwd:efgh1234567890\r\n"
+        "a=fingerprint:sha-256 AA:BB\r\n"
+    )
+
+    def test_invalid_sdp_rejected(self, runner):
+        async def run():
+            ctx, engines = make_ctx(n_workers=0)
+            client = await start_client(ctx, engines)
+            try:
+                resp = await client.post("/v1/realtime/calls", data=b"not sdp at all")
+                assert resp.status == 400
+            finally:
+                await stop_all(client, engines)
+
+        runner(run())
+
+    def test_no_upstream_501_with_guidance(self, runner):
+        async def run():
+            ctx, engines = make_ctx(n_workers=0)
+            client = await start_client(ctx, engines)
+            try:
+                resp = await client.post("/v1/realtime/calls", data=self.OFFER.encode())
+                assert resp.status == 501
+                assert "webrtc=true" in await resp.text()
+            finally:
+                await stop_all(client, engines)
+
+        runner(run())
+
+    def test_broker_to_upstream_and_manage_call(self, runner):
+        from aiohttp import web as aioweb
+        from aiohttp.test_utils import TestServer as AioTestServer
+
+        from smg_amd.workers.worker import Worker
+
+        async def run():
+            hangups = []
+
+            async def upstream_calls(request):
+                sdp = await request.text()
+                assert "a=ice-ufrag:" in sdp
+                return aioweb.Response(
+                    status=201, text=sdp.replace("a=ice-ufrag:abcd", "a=ice-ufrag:srvr"),
+                    headers={"Location": "/v1/realtime/calls/up_1",
+                             "Content-Type": "application/sdp"})
+
+            async def upstream_hangup(request):
+                hangups.append(request.match_info["cid"])
+                return aioweb.json_response({})
+
+            up = aioweb.Application()
+            up.router.add_post("/v1/realtime/calls", upstream_calls)
+            up.router.add_delete("/v1/realtime/calls/{cid}", upstream_hangup)
+            upstream = AioTestServer(up)
+            await upstream.start_server()
+
+            ctx, engines = make_ctx(n_workers=0)
+            ctx.worker_registry.register(Worker(
+                f"http://127.0.0.1:{upstream.port}", model_id="rt",
+                labels={"webrtc": "true"}))
+            client = await start_client(ctx, engines)
+            try:
+                resp = await client.post("/v1/realtime/calls", data=self.OFFER.encode())
+                assert resp.status == 201, await resp.text()
+                answer = await resp.text()
+                assert "a=ice-ufrag:srvr" in answer  # upstream's SDP answer relayed
+                loc = resp.headers["Location"]
+                assert loc.startswith("/v1/realtime/calls/rtc_")
+                # call is managed
+                resp = await client.get(loc)
+                assert resp.status == 200
+                assert (await resp.json())["state"] == "active"
+                resp = await client.delete(loc)
+                assert resp.status == 200
+                assert (await resp.json())["state"] == "ended"
+                resp = await client.get(loc)
+                assert resp.status == 404
+            finally:
+                await stop_all(client, engines)
+                await upstream.close()
+
+        runner(run())
