@@ -64,6 +64,9 @@ int smg_rope_prefill_launch_gqa(const void* qkv, const void* freqs, const void* 
                                 void* k_out, void* v_out, int B, int L, int n_heads,
                                 int n_kv_heads, int max_seq, int head_dim, void* stream,
                                 int kv_fp8);
+int smg_rms_gemm_launch(const void* a, const void* wt, const void* invrms, void* c, int M,
+                        int K, int N, void* stream);
+int smg_row_invrms_launch(const void* a, void* out, int m, int k, float eps, void* stream);
 int smg_attn_decode_launch_split(const void* q, const void* k, const void* v, const void* pos,
                                  void* out, void* part_out, void* part_ml, int n_slots,
                                  int n_heads, int n_kv_heads, int n_split, int max_seq,
@@ -373,6 +376,23 @@ PYBIND11_MODULE(_core, m) {
           py::arg("q"), py::arg("k"), py::arg("v"), py::arg("pos"), py::arg("out"),
           py::arg("n_slots"), py::arg("n_heads"), py::arg("max_seq"), py::arg("head_dim"),
           py::arg("scale"), py::arg("stream"), py::arg("kv_fp8") = 0, py::arg("n_kv_heads") = 0);
+    // fused rms_norm + MFMA GEMM (rms_gemm.hip): C = invrms ⊙ (A @ Wt^T)
+    m.def("row_invrms",
+          [](uintptr_t a, uintptr_t out, int m, int k, float eps, uintptr_t stream) {
+              int rc = smg_row_invrms_launch((const void*)a, (void*)out, m, k, eps, (void*)stream);
+              if (rc != 0) throw std::runtime_error("row_invrms launch failed rc=" + std::to_string(rc));
+          },
+          py::arg("a"), py::arg("out"), py::arg("m"), py::arg("k"), py::arg("eps"),
+          py::arg("stream"));
+    m.def("rms_gemm",
+          [](uintptr_t a, uintptr_t wt, uintptr_t invrms, uintptr_t c, int M, int K, int N,
+             uintptr_t stream) {
+              int rc = smg_rms_gemm_launch((const void*)a, (const void*)wt, (const void*)invrms,
+                                           (void*)c, M, K, N, (void*)stream);
+              if (rc != 0) throw std::runtime_error("rms_gemm launch failed rc=" + std::to_string(rc));
+          },
+          py::arg("a"), py::arg("wt"), py::arg("invrms"), py::arg("c"), py::arg("M"),
+          py::arg("K"), py::arg("N"), py::arg("stream"));
     // T-split (flash-decoding) decode attention + LSE merge (v8): restores
     // chip occupancy when n_slots*n_kv_heads alone under-fills the CUs
     m.def("attn_decode_split",

@@ -28,7 +28,7 @@ def build(verbose: bool = True, force: bool = False) -> Path:
     import pybind11
 
     out = PKG / "_core.so"
-    srcs = [CSRC / "gpu_tree.hip", CSRC / "bpe.hip", CSRC / "image.hip", CSRC / "attn_decode.hip", CSRC / "fused_decode.hip", CSRC / "bindings.cpp", CSRC / "host_tree.cpp"]
+    srcs = [CSRC / "gpu_tree.hip", CSRC / "bpe.hip", CSRC / "image.hip", CSRC / "attn_decode.hip", CSRC / "fused_decode.hip", CSRC / "rms_gemm.hip", CSRC / "bindings.cpp", CSRC / "host_tree.cpp"]
     if not force and _newer(out, srcs):
         return out
     py_inc = sysconfig.get_paths()["include"]
@@ -50,6 +50,7 @@ def build(verbose: bool = True, force: bool = False) -> Path:
         str(CSRC / "image.hip"),
         str(CSRC / "attn_decode.hip"),
         str(CSRC / "fused_decode.hip"),
+        str(CSRC / "rms_gemm.hip"),
         str(CSRC / "bindings.cpp"),
         "-o",
         str(out),

@@ -128,6 +128,7 @@ class TorchEngine:
     _sdpa_gqa = False
     _attn_split = 1
     _hip_attn_split = None
+    _use_mfma = False
 
     def __init__(self, cfg: Optional[TorchEngineConfig] = None, device: str = "cuda:0", graphs: bool = False):
         if device.startswith("cuda") and not torch.cuda.is_available():
@@ -228,6 +229,31 @@ class TorchEngine:
                     else:
                         self._attn_split = 1
                         self._hip_attn_split = None
+                # fused rms_norm + MFMA GEMM (csrc/rms_gemm.hip): the decode
+                # QKV/W13 projections run as hand-written 16x16x32 bf16 MFMA
+                # tiles with the rms fold (g into the weight, 1/rms as an
+                # output row scale) — removes a layer_norm launch + an HBM
+                # round trip of h per projection.  SMG_MFMA=0 disables (A/B).
+                self._use_mfma = False
+                if (hasattr(_core, "rms_gemm") and c.d_model % 512 == 0
+                        and (c.d_model + 2 * c.kv_dim) % 128 == 0
+                        and (2 * c.d_ffn) % 128 == 0
+                        and _os.environ.get("SMG_MFMA", "1") != "0"):
+                    self._hip_rms_gemm = _core.rms_gemm
+                    self._hip_row_invrms = _core.row_invrms
+                    self._use_mfma = True
+                    self._invrms = torch.zeros(c.max_slots, dtype=torch.float32, device=self.device)
+                    self._qkv_buf = torch.zeros(
+                        c.max_slots, c.d_model + 2 * c.kv_dim, device=self.device, dtype=self.dtype)
+                    self._gu_buf = torch.zeros(
+                        c.max_slots, 2 * c.d_ffn, device=self.device, dtype=self.dtype)
+                    for layer in self.layers:
+                        layer.wqkv_tg = (
+                            (layer.wqkv.float() * layer.ln1.float().unsqueeze(1))
+                            .t().contiguous().to(self.dtype))
+                        layer.w13_tg = (
+                            (layer.w13.float() * layer.ln2.float().unsqueeze(1))
+                            .t().contiguous().to(self.dtype))
                 # fused rope+KV-store+q-pack and silu*mul (csrc/fused_decode.hip):
                 # removes ~10 elementwise launches per layer from the decode loop
                 if hasattr(_core, "rope_kv_store"):
@@ -747,8 +773,18 @@ class TorchEngine:
         h = self.embed[self._last_tok]  # [S, D]
         kv8 = 1 if self.kv.dtype == torch.float8_e4m3fn else 0
         freqs_ptr = self.freqs_cis.data_ptr()  # complex64 [T, hd/2] == float2
+        use_mfma = self._use_mfma
         for li, layer in enumerate(self.layers):
-            qkv = _rms(h, layer.ln1) @ layer.wqkv  # [S, D + 2*KD]
+            if use_mfma:
+                # hand-written MFMA path: invrms(h) ⊙ (h @ (ln1 ⊙ wqkv)^T)
+                self._hip_row_invrms(h.data_ptr(), self._invrms.data_ptr(),
+                                     S, c.d_model, 1e-5, stream)
+                self._hip_rms_gemm(h.data_ptr(), layer.wqkv_tg.data_ptr(),
+                                   self._invrms.data_ptr(), self._qkv_buf.data_ptr(),
+                                   S, c.d_model, c.d_model + 2 * c.kv_dim, stream)
+                qkv = self._qkv_buf
+            else:
+                qkv = _rms(h, layer.ln1) @ layer.wqkv  # [S, D + 2*KD]
             self._hip_fused(
                 qkv.data_ptr(), freqs_ptr, self._pos_i32.data_ptr(),
                 self.kv[li, 0].data_ptr(), self.kv[li, 1].data_ptr(), self._q_buf.data_ptr(),
@@ -769,7 +805,15 @@ class TorchEngine:
                     S, c.n_heads, c.max_seq, c.head_dim, scale, stream, kv8, c.kv_heads,
                 )
             h = torch.addmm(h, self._attn_out.view(S, c.d_model), layer.wo)
-            gu = _rms(h, layer.ln2) @ layer.w13  # [S, 2F]
+            if use_mfma:
+                self._hip_row_invrms(h.data_ptr(), self._invrms.data_ptr(),
+                                     S, c.d_model, 1e-5, stream)
+                self._hip_rms_gemm(h.data_ptr(), layer.w13_tg.data_ptr(),
+                                   self._invrms.data_ptr(), self._gu_buf.data_ptr(),
+                                   S, c.d_model, 2 * c.d_ffn, stream)
+                gu = self._gu_buf
+            else:
+                gu = _rms(h, layer.ln2) @ layer.w13  # [S, 2F]
             self._hip_silu_mul(gu.data_ptr(), self._smul_buf.data_ptr(), S, c.d_ffn, stream)
             h = torch.addmm(h, self._smul_buf, layer.w2)
         h = _rms(h, self.ln_f)
