@@ -454,7 +454,7 @@ def main(argv: Optional[List[str]] = None) -> None:
                     from .grpc.servicer import EngineAdapter
 
                     ecfg = TorchEngineConfig.bench_1b()
-                    ecfg.kv_fp8 = bool(getattr(args, "kv_fp8", False))
+                    ecfg.kv_fp8 = "--kv-fp8" in raw_argv
                     eng = TorchEngine(ecfg, device=f"cuda:{i % torch.cuda.device_count()}")
                     adapter = EngineAdapter(eng)
                     await adapter.start()
