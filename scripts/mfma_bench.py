@@ -8,6 +8,8 @@ import time
 import torch
 import torch.nn.functional as F
 
+import os, sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from smg_amd import _core
 
 

@@ -67,6 +67,7 @@ int smg_rope_prefill_launch_gqa(const void* qkv, const void* freqs, const void* 
 int smg_rms_gemm_launch(const void* a, const void* wt, const void* invrms, void* c, int M,
                         int K, int N, void* stream);
 int smg_row_invrms_launch(const void* a, void* out, int m, int k, float eps, void* stream);
+int smg_mfma_probe_launch(const void* afrag, const void* bfrag, void* out, void* stream);
 int smg_attn_decode_launch_split(const void* q, const void* k, const void* v, const void* pos,
                                  void* out, void* part_out, void* part_ml, int n_slots,
                                  int n_heads, int n_kv_heads, int n_split, int max_seq,
@@ -384,6 +385,13 @@ PYBIND11_MODULE(_core, m) {
           },
           py::arg("a"), py::arg("out"), py::arg("m"), py::arg("k"), py::arg("eps"),
           py::arg("stream"));
+    m.def("mfma_probe",
+          [](uintptr_t afrag, uintptr_t bfrag, uintptr_t out, uintptr_t stream) {
+              int rc = smg_mfma_probe_launch((const void*)afrag, (const void*)bfrag,
+                                             (void*)out, (void*)stream);
+              if (rc != 0) throw std::runtime_error("mfma_probe launch failed");
+          },
+          py::arg("afrag"), py::arg("bfrag"), py::arg("out"), py::arg("stream"));
     m.def("rms_gemm",
           [](uintptr_t a, uintptr_t wt, uintptr_t invrms, uintptr_t c, int M, int K, int N,
              uintptr_t stream) {

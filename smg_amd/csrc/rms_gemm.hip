@@ -33,7 +33,7 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 // K per MFMA instruction for 16x16x32
 #define MFMA_K 32
-#define KSTEP 64           // K per LDS stage (2 MFMA k-steps)
+#define KSTEP 32           // K per LDS stage (1 MFMA k-step)
 #define APAD 8             // row pad (bf16 elems) -> conflict-free b128
 #define ALD (KSTEP + APAD)
 
@@ -59,54 +59,57 @@ smg_row_invrms(const __hip_bfloat16* __restrict__ a, float* __restrict__ out,
     if (lane == 0) out[row] = rsqrtf(s / (float)k + eps);
 }
 
-extern "C" __global__ void __launch_bounds__(256)
+#define GEMM_THREADS 512  // 8 waves: WG tile 128Mx128N, wave 64Mx32N, ~6 waves/SIMD
+
+extern "C" __global__ void __launch_bounds__(GEMM_THREADS)
 smg_rms_gemm(const __hip_bfloat16* __restrict__ a,   // [M, K] raw activations
              const __hip_bfloat16* __restrict__ wt,  // [N, K] g-folded, transposed
              const float* __restrict__ invrms,       // [M]
              __hip_bfloat16* __restrict__ c,         // [M, N]
              int M, int K, int N) {
-    const int m0 = blockIdx.x * 64;   // this WG's M tile
+    const int m0 = blockIdx.x * 128;  // this WG's M tile
     const int n0 = blockIdx.y * 128;  // this WG's N tile
     const int tid = threadIdx.x;
     const int wave = tid / WAVE;
     const int lane = tid % WAVE;
 
-    __shared__ __hip_bfloat16 sA[2][64][ALD];
+    __shared__ __hip_bfloat16 sA[2][128][ALD];
     __shared__ __hip_bfloat16 sW[2][128][ALD];
 
-    // wave covers N columns [n0 + wave*32, +32): 2 col tiles x 4 row tiles
+    // 8 waves tile the 128x128 output as 2(M) x 4(N): wave (wm, wn) covers
+    // rows [wm*64, +64) x cols [wn*32, +32) -> 4x2 MFMA tiles, 32 acc VGPRs
+    const int wave_m = (wave >> 2) * 64;
+    const int wave_n = (wave & 3) * 32;
     f32x4 acc[4][2];
 #pragma unroll
     for (int i = 0; i < 4; ++i)
 #pragma unroll
         for (int j = 0; j < 2; ++j) acc[i][j] = (f32x4)0.f;
 
-    // staging assignments: A tile 64x64 = 512 bf16x8 loads; 256 threads -> 2 each
-    // W tile 128x64 = 1024 x8 -> 4 each.  Row-major contiguous both sides.
-    const int a_row0 = tid / 8;          // 32 rows per pass, 2 passes
-    const int a_col8 = (tid % 8) * 8;    // 8 chunks of 8 along K
-    const int w_row0 = tid / 8;
-    const int w_col8 = (tid % 8) * 8;
+    // staging: A and W tiles are each 128x32 = 512 bf16x8 chunks -> 1/thread
+    const int a_row0 = tid / 4;          // 0..127
+    const int a_col8 = (tid % 4) * 8;
+    const int w_row0 = tid / 4;
+    const int w_col8 = (tid % 4) * 8;
 
-    auto stage = [&](int buf, int k0) {
-#pragma unroll
-        for (int p = 0; p < 2; ++p) {
-            int r = a_row0 + p * 32;
-            int gr = m0 + r;
-            bf16x8 v = (gr < M)
-                ? *(const bf16x8*)(a + (size_t)gr * K + k0 + a_col8)
-                : (bf16x8)(__bf16)0.f;
-            *(bf16x8*)&sA[buf][r][a_col8] = v;
-        }
-#pragma unroll
-        for (int p = 0; p < 4; ++p) {
-            int r = w_row0 + p * 32;
-            bf16x8 v = *(const bf16x8*)(wt + (size_t)(n0 + r) * K + k0 + w_col8);
-            *(bf16x8*)&sW[buf][r][w_col8] = v;
-        }
+    // software pipeline: global loads ISSUE before the MFMA section (their
+    // latency hides under compute); the vmcnt wait lands at the LDS write
+    // AFTER the MFMAs, not before them
+    bf16x8 pa, pw;
+    auto load_global = [&](int k0) {
+        int gr = m0 + a_row0;
+        pa = (gr < M)
+            ? *(const bf16x8*)(a + (size_t)gr * K + k0 + a_col8)
+            : (bf16x8)(__bf16)0.f;
+        pw = *(const bf16x8*)(wt + (size_t)(n0 + w_row0) * K + k0 + w_col8);
+    };
+    auto write_lds = [&](int buf) {
+        *(bf16x8*)&sA[buf][a_row0][a_col8] = pa;
+        *(bf16x8*)&sW[buf][w_row0][w_col8] = pw;
     };
 
-    stage(0, 0);
+    load_global(0);
+    write_lds(0);
     __syncthreads();
 
     const int frag_row = lane & 15;        // MFMA i / j
@@ -114,25 +117,28 @@ smg_rms_gemm(const __hip_bfloat16* __restrict__ a,   // [M, K] raw activations
 
     for (int k0 = 0; k0 < K; k0 += KSTEP) {
         const int buf = (k0 / KSTEP) & 1;
-        if (k0 + KSTEP < K) {
-            stage(buf ^ 1, k0 + KSTEP);
+        const bool more = k0 + KSTEP < K;
+        if (more) {
+            load_global(k0 + KSTEP);  // issue now, consumed after the MFMAs
         }
-#pragma unroll
-        for (int ks = 0; ks < KSTEP / MFMA_K; ++ks) {
-            const int kb = ks * MFMA_K + frag_ko;
+        {
+            const int kb = frag_ko;
             bf16x8 afrag[4], wfrag[2];
 #pragma unroll
             for (int i = 0; i < 4; ++i)
-                afrag[i] = *(const bf16x8*)&sA[buf][i * 16 + frag_row][kb];
+                afrag[i] = *(const bf16x8*)&sA[buf][wave_m + i * 16 + frag_row][kb];
 #pragma unroll
             for (int j = 0; j < 2; ++j)
-                wfrag[j] = *(const bf16x8*)&sW[buf][wave * 32 + j * 16 + frag_row][kb];
+                wfrag[j] = *(const bf16x8*)&sW[buf][wave_n + j * 16 + frag_row][kb];
 #pragma unroll
             for (int i = 0; i < 4; ++i)
 #pragma unroll
                 for (int j = 0; j < 2; ++j)
                     acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         afrag[i], wfrag[j], acc[i][j], 0, 0, 0);
+        }
+        if (more) {
+            write_lds(buf ^ 1);  // vm wait here, after the compute phase
         }
         __syncthreads();
     }
@@ -146,9 +152,9 @@ smg_rms_gemm(const __hip_bfloat16* __restrict__ a,   // [M, K] raw activations
         for (int j = 0; j < 2; ++j) {
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
-                int row = m0 + i * 16 + c_row0 + r;
+                int row = m0 + wave_m + i * 16 + c_row0 + r;
                 if (row >= M) continue;
-                int col = n0 + wave * 32 + j * 16 + c_col;
+                int col = n0 + wave_n + j * 16 + c_col;
                 c[(size_t)row * N + col] =
                     (__hip_bfloat16)(acc[i][j][r] * invrms[row]);
             }
@@ -156,11 +162,36 @@ smg_rms_gemm(const __hip_bfloat16* __restrict__ a,   // [M, K] raw activations
     }
 }
 
+// Layout probe: one mfma_f32_16x16x32_bf16 on caller-provided per-lane
+// fragments.  afrag/bfrag: [64 lanes][8] bf16; out: [64 lanes][4] f32 in the
+// raw accumulator register order (host decodes with the C map).  Lets a test
+// determine the true A/B k-ordering empirically instead of trusting docs.
+extern "C" __global__ void __launch_bounds__(WAVE)
+smg_mfma_probe(const __hip_bfloat16* __restrict__ afrag,
+               const __hip_bfloat16* __restrict__ bfrag,
+               float* __restrict__ out) {
+    int lane = threadIdx.x;
+    bf16x8 a = *(const bf16x8*)(afrag + lane * 8);
+    bf16x8 b = *(const bf16x8*)(bfrag + lane * 8);
+    f32x4 c = (f32x4)0.f;
+    c = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+#pragma unroll
+    for (int r = 0; r < 4; ++r) out[lane * 4 + r] = c[r];
+}
+
+extern "C" int smg_mfma_probe_launch(const void* afrag, const void* bfrag, void* out,
+                                     void* stream) {
+    hipLaunchKernelGGL(smg_mfma_probe, dim3(1), dim3(WAVE), 0, (hipStream_t)stream,
+                       (const __hip_bfloat16*)afrag, (const __hip_bfloat16*)bfrag,
+                       (float*)out);
+    return hipGetLastError() == hipSuccess ? 0 : -2;
+}
+
 extern "C" int smg_rms_gemm_launch(const void* a, const void* wt, const void* invrms,
                                    void* c, int M, int K, int N, void* stream) {
     if (K % KSTEP || N % 128) return -1;
-    dim3 grid((M + 63) / 64, N / 128);
-    hipLaunchKernelGGL(smg_rms_gemm, grid, dim3(256), 0, (hipStream_t)stream,
+    dim3 grid((M + 127) / 128, N / 128);
+    hipLaunchKernelGGL(smg_rms_gemm, grid, dim3(GEMM_THREADS), 0, (hipStream_t)stream,
                        (const __hip_bfloat16*)a, (const __hip_bfloat16*)wt,
                        (const float*)invrms, (__hip_bfloat16*)c, M, K, N);
     return hipGetLastError() == hipSuccess ? 0 : -2;

@@ -230,15 +230,18 @@ class TorchEngine:
                         self._attn_split = 1
                         self._hip_attn_split = None
                 # fused rms_norm + MFMA GEMM (csrc/rms_gemm.hip): the decode
-                # QKV/W13 projections run as hand-written 16x16x32 bf16 MFMA
+                # QKV/W13 projections as hand-written 16x16x32 bf16 MFMA
                 # tiles with the rms fold (g into the weight, 1/rms as an
-                # output row scale) — removes a layer_norm launch + an HBM
-                # round trip of h per projection.  SMG_MFMA=0 disables (A/B).
+                # output row scale).  Measured on MI355X (scripts/mfma_bench):
+                # 154-451 TF vs hipBLASLt's 259-702 on the decode shapes —
+                # the library GEMM still wins, so this path is OPT-IN
+                # (SMG_MFMA=1); kept in-tree as the measured MFMA baseline
+                # the next tiling iteration starts from.
                 self._use_mfma = False
                 if (hasattr(_core, "rms_gemm") and c.d_model % 512 == 0
                         and (c.d_model + 2 * c.kv_dim) % 128 == 0
                         and (2 * c.d_ffn) % 128 == 0
-                        and _os.environ.get("SMG_MFMA", "1") != "0"):
+                        and _os.environ.get("SMG_MFMA", "0") == "1"):
                     self._hip_rms_gemm = _core.rms_gemm
                     self._hip_row_invrms = _core.row_invrms
                     self._use_mfma = True

@@ -42,11 +42,17 @@ class TestRmsGemm:
         ref_inv = torch.rsqrt(a.float().pow(2).mean(-1) + 1e-5)
         assert torch.allclose(invrms, ref_inv, rtol=1e-3, atol=1e-5)
         ref = ref_rms_gemm(a.float(), gln.float(), w.float())
-        # bf16 inputs + fp32 accumulate: compare against the fp32 reference
-        # with bf16-grade tolerance (K up to 2048 accumulations)
+        # rms_norm makes rows unit-RMS, so outputs are ~N(0,1).  bf16 inputs
+        # with fp32 accumulate deviate from the all-fp32 reference by
+        # ~sqrt(K)·2^-9 in the worst element — measured: torch's own bf16
+        # hipBLASLt GEMM shows max|err| 0.027 on the K=2048 shapes, the MFMA
+        # kernel 0.015 (closer).  Gate on that scale, plus a mean bound.
         err = (out.float() - ref).abs()
-        rel = err / (ref.abs() + 1e-2)
-        assert rel.max().item() < 0.05, (rel.max().item(), err.max().item())
+        assert err.max().item() < 0.08, err.max().item()
+        assert err.mean().item() < 0.01, err.mean().item()
+        # and the kernel must be in the same rounding class as torch bf16
+        tb = ((a @ wt.t().contiguous()).float() * invrms[:, None])
+        assert (out.float() - tb).abs().max().item() < 0.08
 
     def test_asymmetric_b_orientation(self):
         """Guide rule: an asymmetric B catches row/col-swapped C writes."""
@@ -77,13 +83,13 @@ class TestRmsGemm:
 
         cfg = TorchEngineConfig(vocab_size=4096, n_layers=2, d_model=512, n_heads=8,
                                 n_kv_heads=2, max_slots=8, max_seq=256, prefill_chunk=256)
-        eng_mfma = TorchEngine(cfg, device="cuda:0")
-        assert eng_mfma._use_mfma, "rms_gemm path must be active for this config"
-        os.environ["SMG_MFMA"] = "0"
+        os.environ["SMG_MFMA"] = "1"
         try:
-            eng_ref = TorchEngine(cfg, device="cuda:0")
+            eng_mfma = TorchEngine(cfg, device="cuda:0")
         finally:
             os.environ.pop("SMG_MFMA", None)
+        assert eng_mfma._use_mfma, "rms_gemm path must be active for this config"
+        eng_ref = TorchEngine(cfg, device="cuda:0")
         assert not eng_ref._use_mfma
         prompt = list(range(40))
         r1 = eng_mfma.submit(prompt, max_new_tokens=8)
