@@ -102,9 +102,25 @@ async def v1_realtime_ws(request: web.Request):
             url += f"?model={model}"
         async with aiohttp.ClientSession() as session:
             async with session.ws_connect(url) as ws_worker:
-                async def pump(src, dst):
+                from ..protocols.realtime_events import RealtimeEventError, parse_event
+
+                async def pump(src, dst, validate_client=False):
                     async for msg in src:
                         if msg.type == WSMsgType.TEXT:
+                            if validate_client:
+                                # typed event validation (reference
+                                # realtime_events.rs ClientEvent): malformed
+                                # events answer an `error` event, not a relay
+                                try:
+                                    ev = json.loads(msg.data)
+                                    parse_event(ev, "client")
+                                except (json.JSONDecodeError, RealtimeEventError) as exc:
+                                    await src.send_str(json.dumps({
+                                        "type": "error",
+                                        "error": {"type": "invalid_request_error",
+                                                  "message": str(exc)},
+                                    }))
+                                    continue
                             await dst.send_str(msg.data)
                         elif msg.type == WSMsgType.BINARY:
                             await dst.send_bytes(msg.data)
@@ -112,7 +128,8 @@ async def v1_realtime_ws(request: web.Request):
                             break
                     await dst.close()
 
-                await asyncio.gather(pump(ws_client, ws_worker), pump(ws_worker, ws_client))
+                await asyncio.gather(pump(ws_client, ws_worker, validate_client=True),
+                                     pump(ws_worker, ws_client))
         worker.record_outcome(True)
     except Exception:
         worker.record_outcome(False)

@@ -490,6 +490,13 @@ async def create_worker(request):
     body, _ = await _read_json(request)
     if not body or "url" not in body:
         return web.Response(status=400, body=error_body("'url' is required"), content_type="application/json")
+    # typed validation (reference WorkerSpec worker.rs:604 via serde)
+    from ..protocols.worker_spec import WorkerSpec, WorkerSpecError
+
+    try:
+        WorkerSpec.from_dict(body)
+    except WorkerSpecError as exc:
+        return web.Response(status=400, body=error_body(str(exc)), content_type="application/json")
     from .jobs import JobKind
 
     try:
