@@ -170,12 +170,14 @@ class TestRealtime:
                 assert sess["client_secret"]["value"].startswith("ek_")
 
                 ws = await client.ws_connect("/v1/realtime?model=mock-model")
-                await ws.send_str("hello")
+                # the relay validates client events (realtime_events.rs), so
+                # frames must be well-formed realtime JSON
+                await ws.send_str(json.dumps({"type": "input_audio_buffer.append", "audio": "QQ=="}))
                 msg = await ws.receive(timeout=5)
-                assert msg.data == "echo:hello"
-                await ws.send_str("bye")
+                assert msg.data.startswith("echo:") and "input_audio_buffer.append" in msg.data
+                await ws.send_str(json.dumps({"type": "response.create"}))
                 msg = await ws.receive(timeout=5)
-                assert msg.data == "echo:bye"
+                assert "response.create" in msg.data
                 await ws.close()
             finally:
                 await stop_all(client, engines)
