@@ -63,6 +63,9 @@ def parse_args():
                         "(round-1 headline model, kept for comparability)")
     p.add_argument("--kv-fp8", action="store_true",
                    help="opt-in fp8 (e4m3) KV cache; headline default stays bf16")
+    p.add_argument("--pd", action="store_true",
+                   help="PD disaggregation over the plane: odd ranks prefill, even ranks "
+                        "(incl. the gateway) decode; KV hands off over xGMI p2p")
     p.add_argument("--no-graphs", action="store_true", help="disable hipGraph decode capture")
     p.add_argument("--decode-burst", type=int, default=2, help="decode iterations per tick")
     p.add_argument("--seed", type=int, default=1234)
@@ -112,10 +115,16 @@ def model_label(args) -> str:
     return "smg-bench-1b (16L d2048 h16, bf16, random-init)"
 
 
+def pd_role(rank: int, args) -> str:
+    if not args.pd:
+        return "regular"
+    return "prefill" if rank % 2 == 1 else "decode"
+
+
 def worker_main(rank: int, world: int, args, device: str, backend: str):
     eng = TorchEngine(engine_config(args), device=device, graphs=device.startswith("cuda") and not args.no_graphs)
     plane = WorkerPlane(PlaneConfig(max_prompt=args.prefix_len + args.suffix_len + 8, device=device if backend == "nccl" else "cpu"))
-    elapsed = run_worker_loop(eng, plane, decode_burst=args.decode_burst)
+    elapsed = run_worker_loop(eng, plane, decode_burst=args.decode_burst, role=pd_role(rank, args))
     el = torch.tensor([elapsed], device=device if backend == "nccl" else "cpu")
     torch.distributed.all_reduce(el, op=torch.distributed.ReduceOp.MAX)
 
@@ -143,9 +152,13 @@ def gateway_main(rank: int, world: int, args, device: str, backend: str):
         workers.append(w)
     pol_cfg = PolicyConfig(name="cache_aware", block_size=16, gpu_tree=use_gpu, gpu_tree_device=0)
     policy = CacheAwarePolicy(pol_cfg)
+    pd_roles = {r: pd_role(r, args) for r in range(world)} if args.pd else None
+    if args.pd and world < 2:
+        raise SystemExit("--pd needs world >= 2 (at least one prefill rank)")
     gw = TickGateway(
         workers, policy, plane=plane, local_engine=eng,
         decode_burst=args.decode_burst, model_id="bench-1b",
+        pd_roles=pd_roles,
     )
     vocab = engine_config(args).vocab_size
     gen = LoadGen(args, vocab)
@@ -212,7 +225,13 @@ def gateway_main(rank: int, world: int, args, device: str, backend: str):
             "model": model_label(args),
             "global_batch": args.concurrency * world,
             "seq_len": args.prefix_len + args.suffix_len,
-            "parallelism": f"gateway+{world}workers (dp{world}, cache_aware, rccl-xgmi)",
+            "parallelism": (
+                f"gateway+{world}workers (pd over rccl-xgmi: "
+                f"{sum(1 for r in range(world) if r % 2 == 1)}p+"
+                f"{sum(1 for r in range(world) if r % 2 == 0)}d)"
+                if args.pd else
+                f"gateway+{world}workers (dp{world}, cache_aware, rccl-xgmi)"
+            ),
             "policy": "cache_aware",
             "gpu_tree": use_gpu,
             "p50_routing_latency_ms": round(p50_route, 4) if p50_route is not None else None,

@@ -42,6 +42,14 @@ STOP_FLAG = 1 << 16
 BARRIER_FLAG = 1 << 17  # worker joins a dist.barrier() + device sync after this tick
 DONE = 1
 ABORT = 2
+PREFILLED = 4  # PD prefill leg finished; KV parked for transfer
+
+# PD KV-handoff instructions carried in the request tensor (both endpoint
+# ranks receive the same ordered list, so their send/recv pairs match by
+# construction — deadlock-free like the main lockstep exchange)
+KV_SEND = 1
+KV_RECV = 2
+TR_INTS = 6  # rid, peer_rank, direction, n_tokens, first_tok, max_new
 
 
 @dataclass
@@ -49,6 +57,7 @@ class PlaneConfig:
     max_reqs_per_tick: int = 64
     max_prompt: int = 1024
     max_events_per_tick: int = 4096
+    max_transfers_per_tick: int = 32
     device: str = "cpu"  # "cuda:<rank>" on GPU
 
     @property
@@ -56,8 +65,12 @@ class PlaneConfig:
         return 3 + self.max_prompt
 
     @property
-    def req_len(self) -> int:
+    def tr_off(self) -> int:
         return 1 + self.max_reqs_per_tick * self.req_stride
+
+    @property
+    def req_len(self) -> int:
+        return self.tr_off + 1 + self.max_transfers_per_tick * TR_INTS
 
     @property
     def ev_len(self) -> int:
@@ -90,10 +103,18 @@ class GatewayPlane:
                 self._send_h[w] = (self._send[w], self._send[w].numpy())
                 self._recv_h[w] = (self._recv[w], self._recv[w].numpy())
         self._pending: Dict[int, List[Tuple[int, int, List[int]]]] = {w: [] for w in worker_ranks}
+        self._pending_tr: Dict[int, List[Tuple[int, ...]]] = {w: [] for w in worker_ranks}
         self._works = []  # outstanding isend/irecv handles for this tick
 
     def enqueue(self, worker_rank: int, rid: int, max_new: int, prompt: List[int]) -> None:
         self._pending[worker_rank].append((rid, max_new, prompt[-self.cfg.max_prompt:]))
+
+    def enqueue_transfer(self, worker_rank: int, rid: int, peer: int, direction: int,
+                         n_tokens: int, first_tok: int, max_new: int) -> None:
+        """Instruct `worker_rank` to send (KV_SEND) or receive (KV_RECV) the
+        KV of `rid` with `peer` next tick (PD handoff).  The gateway enqueues
+        the matching instruction on both endpoints in the same tick."""
+        self._pending_tr[worker_rank].append((rid, peer, direction, n_tokens, first_tok, max_new))
 
     def tick_send(self, stop: bool = False, barrier: bool = False) -> None:
         """Phase 1: ship this tick's requests to every worker AND post the
@@ -116,6 +137,12 @@ class GatewayPlane:
                 rows[base + 1] = max_new
                 rows[base + 2] = len(prompt)
                 rows[base + 3: base + 3 + len(prompt)] = prompt  # numpy list assign, one C pass
+            trs = self._pending_tr[w][: cfg.max_transfers_per_tick]
+            self._pending_tr[w] = self._pending_tr[w][len(trs):]
+            rows[cfg.tr_off] = len(trs)
+            for i, tr in enumerate(trs):
+                base = cfg.tr_off + 1 + i * TR_INTS
+                rows[base: base + TR_INTS] = tr
             if self._gpu:
                 self._send[w].copy_(self._send_h[w][0], non_blocking=True)
         ops = []
@@ -156,6 +183,34 @@ class GatewayPlane:
         return self.tick_recv()
 
 
+KV_TAG = 7  # gloo matches p2p by tag-slot: the KV stream must not share the
+# tick exchange's slot 0 (observed: a KV send matched the event irecv and
+# aborted the pair).  NCCL ignores tags and orders per pair, where our
+# program order already pairs correctly.
+
+
+def execute_transfers(engine, transfers) -> int:
+    """Run one tick's ordered PD KV handoffs against the local engine.
+    Entries: (rid, peer, direction, n_tokens, first_tok, max_new).  Both
+    endpoints iterate the SAME list, so sends and recvs pair in order; fp8
+    caches travel as uint8 views (NCCL has no fp8 dtype)."""
+    tag = KV_TAG if dist.get_backend() == "gloo" else 0
+    done = 0
+    for rid, peer, direction, n_tokens, first_tok, max_new in transfers:
+        if direction == KV_SEND:
+            t, plen, _ft = engine.export_kv(rid)
+            payload = t.view(torch.uint8) if t.dtype == torch.float8_e4m3fn else t
+            dist.send(payload.contiguous(), dst=peer, tag=tag)
+        else:
+            shape = engine.kv_transfer_shape(n_tokens)
+            buf = torch.empty(shape, dtype=engine.kv.dtype, device=engine.kv.device)
+            payload = buf.view(torch.uint8) if buf.dtype == torch.float8_e4m3fn else buf
+            dist.recv(payload, src=peer, tag=tag)
+            engine.import_kv(rid, buf, n_tokens, first_tok, max_new)
+        done += 1
+    return done
+
+
 class WorkerPlane:
     """Worker-rank side: mirror of the gateway's per-tick exchange."""
 
@@ -174,6 +229,7 @@ class WorkerPlane:
             self._send_h = (self._send, self._send.numpy())
         self._send_work = None
         self.barrier_requested = False
+        self.transfers: List[Tuple[int, ...]] = []
 
     def tick(self, events: List[Tuple[int, int, int]]) -> Tuple[List[Tuple[int, int, List[int]]], bool]:
         """One lockstep exchange: sends `events` [(rid, token, flags)], receives
@@ -217,4 +273,12 @@ class WorkerPlane:
             plen = int(req[base + 2])
             prompt = req[base + 3: base + 3 + plen].tolist()
             out.append((rid, max_new, prompt))
+        # PD KV-handoff instructions for this tick (ordered; both endpoints
+        # hold the same list so their p2p calls pair up)
+        n_tr = int(req[cfg.tr_off])
+        self.transfers = [
+            tuple(int(x) for x in req[cfg.tr_off + 1 + i * TR_INTS:
+                                      cfg.tr_off + 1 + (i + 1) * TR_INTS])
+            for i in range(min(n_tr, cfg.max_transfers_per_tick))
+        ]
         return out, stop

@@ -111,9 +111,10 @@ def _rms(x, weight):
 
 
 class _Request:
-    __slots__ = ("rid", "tokens", "max_new", "slot", "prefilled", "generated", "done")
+    __slots__ = ("rid", "tokens", "max_new", "slot", "prefilled", "generated", "done",
+                 "prefill_only")
 
-    def __init__(self, rid, tokens, max_new):
+    def __init__(self, rid, tokens, max_new, prefill_only=False):
         self.rid = rid
         self.tokens = tokens
         self.max_new = max_new
@@ -121,6 +122,7 @@ class _Request:
         self.prefilled = 0
         self.generated: List[int] = []
         self.done = False
+        self.prefill_only = prefill_only
 
 
 class TorchEngine:
@@ -166,6 +168,7 @@ class TorchEngine:
         self.running: Dict[int, _Request] = {}  # slot -> request
         self._requests: Dict[str, _Request] = {}
         self._rid_counter = 0
+        self._parked: Dict[str, tuple] = {}  # rid -> (slot, req, first_tok) awaiting export_kv
         self.total_generated = 0
         self._step_events: List[tuple] = []  # (rid, token, done) since last drain
         self.graphs = graphs
@@ -290,7 +293,14 @@ class TorchEngine:
             )
 
     # ---- API -------------------------------------------------------------
-    def submit(self, tokens: List[int], max_new_tokens: int, rid: Optional[str] = None) -> str:
+    def submit(self, tokens: List[int], max_new_tokens: int, rid: Optional[str] = None,
+               prefill_only: bool = False) -> str:
+        """`prefill_only` is the PD prefill leg: the request prefills, samples
+        its first token (emitted with the PREFILLED flag, value 4) and PARKS —
+        the slot stays allocated until export_kv() hands its KV off to the
+        decode rank (PD over the rccl plane; reference PD delegates this
+        transfer to engine-side Mooncake/NIXL, here the engine is ours and
+        the handoff is an xGMI p2p send)."""
         if rid is None:
             self._rid_counter += 1
             rid = f"req-{self._rid_counter}"
@@ -299,10 +309,52 @@ class TorchEngine:
         if len(toks) > c.max_seq - 2:
             toks = toks[-(c.max_seq - 2):]  # keep the prompt tail
         max_new_tokens = max(1, min(max_new_tokens, c.max_seq - 1 - len(toks)))
-        req = _Request(rid, toks, max_new_tokens)
+        req = _Request(rid, toks, max_new_tokens, prefill_only=prefill_only)
         self._requests[rid] = req
         self.waiting.append(req)
         return rid
+
+    # ---- PD disaggregation: KV handoff -----------------------------------
+    PREFILLED = 4  # event flag: prefill leg done, KV parked for export
+    PLEN_INFO = 8  # event flag: token field carries the prefilled length
+
+    def export_kv(self, rid: str):
+        """Parked prefill -> (kv tensor [L, 2, KVH, plen, hd], plen,
+        first_token); frees the slot."""
+        entry = self._parked.pop(rid, None)
+        if entry is None:
+            raise KeyError(f"no parked prefill for {rid}")
+        slot, req, first_tok = entry
+        plen = req.prefilled
+        t = self.kv[:, :, slot, :, :plen].clone()
+        self._free_slots.append(slot)
+        self._seq_len_host[slot] = 0
+        self.seq_len[slot] = 0
+        self._requests.pop(rid, None)
+        return t, plen, first_tok
+
+    def import_kv(self, rid, kv_tensor, plen: int, first_tok: int, max_new: int) -> bool:
+        """Decode-leg intake: place the transferred KV into a free slot and
+        continue decoding from the prefill-sampled first token."""
+        if not self._free_slots:
+            return False
+        c = self.cfg
+        slot = self._free_slots.pop()
+        self.kv[:, :, slot, :, :plen] = kv_tensor.to(self.kv.dtype)
+        self._seq_len_host[slot] = plen
+        self.seq_len[slot] = plen
+        self._last_tok[slot] = int(first_tok)
+        req = _Request(rid, [0] * plen, max_new)  # token VALUES unused post-prefill
+        req.prefilled = plen
+        req.slot = slot
+        req.generated = [int(first_tok)]
+        self._requests[rid] = req
+        self.running[slot] = req
+        return True
+
+    def kv_transfer_shape(self, plen: int):
+        c = self.cfg
+        return (c.n_layers, 2, c.kv_heads, plen, c.head_dim)
 
     def finished(self, rid: str) -> bool:
         r = self._requests.get(rid)
@@ -382,6 +434,15 @@ class TorchEngine:
                 req.prefilled += L
                 if req.prefilled >= len(req.tokens):
                     self._prefix_store(slot, req.tokens)
+                    if req.prefill_only:
+                        # PD prefill leg: announce the prefilled length, then
+                        # the sampled first token with the PREFILLED flag, and
+                        # PARK the slot for export_kv
+                        del self.running[slot]
+                        self._parked[req.rid] = (slot, req, int(tok))
+                        self._step_events.append((req.rid, req.prefilled, self.PLEN_INFO))
+                        self._step_events.append((req.rid, tok, self.PREFILLED))
+                        continue
                     # the final prefill chunk's logits sample the first
                     # generated token — decode then starts from the SAMPLED
                     # token at position n, instead of re-feeding the last

@@ -30,7 +30,18 @@ import time
 from collections import deque
 from typing import Any, Callable, Dict, List, Optional, Sequence, Tuple
 
-from ..comm.plane import DONE, GatewayPlane, PlaneConfig, WorkerPlane
+from ..comm.plane import (
+    DONE,
+    KV_RECV,
+    KV_SEND,
+    PREFILLED,
+    GatewayPlane,
+    PlaneConfig,
+    WorkerPlane,
+    execute_transfers,
+)
+
+PLEN_INFO = 8
 from ..policies import SelectWorkerInfo
 from .base import RouteRequest, RouteResponse, Router
 from ..workers.worker import Worker
@@ -58,8 +69,21 @@ class TickGateway:
         max_new_arrivals_per_tick: int = 128,
         on_event: Optional[Callable[[int, int, int], None]] = None,
         metrics=None,
+        pd_roles: Optional[Dict[int, str]] = None,
     ):
         self.metrics = metrics
+        # PD over the plane: rank -> "prefill" | "decode".  Arrivals route
+        # to PREFILL ranks (prefill_only role on the worker side); PREFILLED
+        # events trigger an xGMI KV handoff to a min-load DECODE rank.
+        self.pd_roles = pd_roles
+        if pd_roles:
+            self.prefill_idxs = [i for i, w in enumerate(workers)
+                                 if pd_roles.get(w.rccl_rank) == "prefill"]
+            self.decode_idxs = [i for i, w in enumerate(workers)
+                                if pd_roles.get(w.rccl_rank, "decode") == "decode"]
+        self._req_meta: Dict[int, Tuple[int, int]] = {}  # rid -> (n_tokens, max_new)
+        self._pd_plen: Dict[int, int] = {}  # rid -> prefilled length (PLEN_INFO)
+        self._local_transfers: List[Tuple[int, ...]] = []
         self.workers = list(workers)
         self.policy = policy
         self.plane = plane
@@ -80,6 +104,8 @@ class TickGateway:
     def submit(self, tokens: List[int], max_new: int, rid: Optional[int] = None) -> int:
         if rid is None:
             rid = next(self._rid) & 0x7FFFFFFF
+        if self.pd_roles:
+            self._req_meta[rid] = (len(tokens), max_new)
         self._pending.append((rid, list(tokens), max_new))
         return rid
 
@@ -96,6 +122,9 @@ class TickGateway:
         while self._pending and len(new_reqs) < self.max_new_arrivals:
             new_reqs.append(self._pending.popleft())
         if new_reqs:
+            # PD: arrivals go to the PREFILL fleet only
+            route_pool = ([self.workers[i] for i in self.prefill_idxs]
+                          if self.pd_roles else self.workers)
             infos = [
                 SelectWorkerInfo(
                     request_id=str(rid), model_id=self.model_id, tokens=toks, est_tokens=len(toks)
@@ -104,14 +133,16 @@ class TickGateway:
             ]
             t0 = time.perf_counter()
             if hasattr(self.policy, "select_worker_batch"):
-                sels = self.policy.select_worker_batch(self.workers, infos)
+                sels = self.policy.select_worker_batch(route_pool, infos)
             else:
-                sels = [self.policy.select_worker(self.workers, i) for i in infos]
+                sels = [self.policy.select_worker(route_pool, i) for i in infos]
             dt = time.perf_counter() - t0
             self.routing_lat.extend([dt / len(new_reqs)] * len(new_reqs))
             self.phase_t["route"] += dt
             for (rid, toks, max_new), sel in zip(new_reqs, sels):
                 sel = 0 if sel is None else sel
+                if self.pd_roles:
+                    sel = self.prefill_idxs[sel]
                 self.workers[sel].incr_load()
                 self.inflight[rid] = sel
                 rank = self.workers[sel].rccl_rank
@@ -124,6 +155,10 @@ class TickGateway:
         if self.plane is not None:
             tp = time.perf_counter()
             self.plane.tick_send()
+            # the gateway's side of this tick's PD handoffs: executed right
+            # after tick_send so the peers' instruction lists (sent in that
+            # same tensor) pair with ours in order
+            self._flush_local_transfers()
             self.phase_t["plane"] += time.perf_counter() - tp
         local_events: List[Tuple[int, int, int]] = []
         if self.local_engine is not None:
@@ -142,6 +177,8 @@ class TickGateway:
         # 3) completions + event fan-out
         tev = time.perf_counter()
         done_now = 0
+        if self.pd_roles:
+            local_events, remote_arrays = self._handle_pd_events(local_events, remote_arrays)
         cb = self.on_event
         if cb is not None:
             rows = list(local_events)
@@ -170,6 +207,64 @@ class TickGateway:
                 m.plane_events_received.inc(n_ev)
         return done_now
 
+    def _handle_pd_events(self, local_events, remote_arrays):
+        """Strip PD control events (PLEN_INFO, PREFILLED) out of the streams
+        and schedule the KV handoffs.  The PREFILLED event's token is the
+        request's FIRST generated token — it re-enters the stream as a plain
+        token event so SSE consumers see it."""
+        out_local: List[Tuple[int, int, int]] = []
+
+        def handle(rid: int, token: int, flags: int) -> Optional[Tuple[int, int, int]]:
+            if flags == PLEN_INFO:
+                self._pd_plen[rid] = token
+                return None
+            if flags == PREFILLED:
+                self._schedule_handoff(rid, first_tok=token)
+                return (rid, token, 0)  # first token streams normally
+            return (rid, token, flags)
+
+        for rid, token, flags in local_events:
+            ev = handle(rid, token, flags)
+            if ev is not None:
+                out_local.append(ev)
+        out_remote = []
+        for arr in remote_arrays:
+            if len(arr) == 0 or not (arr[:, 2] >= PREFILLED).any():
+                out_remote.append(arr)
+                continue
+            kept = []
+            for rid, token, flags in arr.tolist():
+                ev = handle(rid, token, flags)
+                if ev is not None:
+                    kept.append(ev)
+            out_local.extend(kept)
+        return out_local, out_remote
+
+    def _schedule_handoff(self, rid: int, first_tok: int) -> None:
+        plen = self._pd_plen.pop(rid, None)
+        meta = self._req_meta.pop(rid, None)
+        p_idx = self.inflight.get(rid)
+        if plen is None or meta is None or p_idx is None:
+            return
+        max_new = meta[1]
+        # min-load decode rank
+        d_idx = min(self.decode_idxs, key=lambda i: self.workers[i].active_requests)
+        p_rank = self.workers[p_idx].rccl_rank or 0
+        d_rank = self.workers[d_idx].rccl_rank or 0
+        self.workers[p_idx].decr_load()
+        self.workers[d_idx].incr_load()
+        self.inflight[rid] = d_idx
+        send_instr = (rid, d_rank, KV_SEND, plen, first_tok, max_new)
+        recv_instr = (rid, p_rank, KV_RECV, plen, first_tok, max_new)
+        if p_rank == 0:
+            self._local_transfers.append(send_instr)
+        else:
+            self.plane.enqueue_transfer(p_rank, *send_instr)
+        if d_rank == 0:
+            self._local_transfers.append(recv_instr)
+        else:
+            self.plane.enqueue_transfer(d_rank, *recv_instr)
+
     def _complete(self, rid: int) -> int:
         wrk = self.inflight.pop(rid, None)
         if wrk is None:
@@ -178,35 +273,55 @@ class TickGateway:
         return 1
 
     # ---- timing / shutdown ------------------------------------------------
+    def _flush_local_transfers(self) -> None:
+        if self._local_transfers and self.local_engine is not None:
+            trs, self._local_transfers = self._local_transfers, []
+            execute_transfers(self.local_engine, trs)
+
     def barrier_sync(self) -> None:
-        """Plane barrier tick + dist barrier (bench timing bracket)."""
+        """Plane barrier tick + dist barrier (bench timing bracket).  Any
+        handoffs scheduled in the final event pass still ship with this tick,
+        so the gateway's local side must execute too — otherwise a remote
+        peer blocks in its matching send/recv (shutdown deadlock)."""
         if self.plane is not None:
             import torch.distributed as dist
 
-            self.plane.tick(barrier=True)
+            self.plane.tick_send(barrier=True)
+            self._flush_local_transfers()
+            self.plane.tick_recv()
             dist.barrier()
 
     def stop_workers(self) -> None:
         if self.plane is not None:
             self.plane.tick_send(stop=True)
+            self._flush_local_transfers()
             self.plane.tick_recv()  # drain the workers' final event sends
 
     def p50_routing_ms(self) -> Optional[float]:
         return statistics.median(self.routing_lat) * 1e3 if self.routing_lat else None
 
 
-def run_worker_loop(engine, plane: WorkerPlane, decode_burst: int = 2) -> float:
+def run_worker_loop(engine, plane: WorkerPlane, decode_burst: int = 2,
+                    role: str = "regular") -> float:
     """Worker-rank loop (ranks >= 1): lockstep plane ticks against the local
-    engine until the gateway sends STOP.  Returns this rank's timed-region
+    engine until the gateway sends STOP.  `role="prefill"` makes every
+    submission a PD prefill leg (first token sampled, KV parked for the
+    handoff the gateway schedules).  Returns this rank's timed-region
     elapsed seconds when the gateway bracketed the run with barrier ticks
     (bench), else 0."""
     import torch
     import torch.distributed as dist
 
+    prefill_kw = {"prefill_only": True} if role == "prefill" else {}
     events: List[tuple] = []
     t0 = t1 = None
     while True:
         reqs, stop = plane.tick(events)
+        # PD handoffs pair with their peers FIRST: the gateway (and peer
+        # workers) execute their side right after this tick's send, before
+        # any barrier — a barrier before the transfer deadlocks the pair
+        if plane.transfers:
+            execute_transfers(engine, plane.transfers)
         if plane.barrier_requested:
             dist.barrier()
             if hasattr(torch.cuda, "is_available") and torch.cuda.is_available():
@@ -218,7 +333,7 @@ def run_worker_loop(engine, plane: WorkerPlane, decode_burst: int = 2) -> float:
         if stop:
             break
         for rid, max_new, prompt in reqs:
-            engine.submit(prompt, max_new, rid=rid)
+            engine.submit(prompt, max_new, rid=rid, **prefill_kw)
         engine.step(decode_burst=decode_burst)
         events = engine.drain_events()
     return (t1 - t0) if (t0 is not None and t1 is not None) else 0.0
