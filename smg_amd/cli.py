@@ -407,7 +407,14 @@ def rccl_worker_main(cfg) -> None:
             device=device if use_gpu else "cpu",
         )
     )
-    run_worker_loop(eng, plane)
+    from .config import RoutingMode
+    from .routers.rccl_router import pd_rank_roles
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    role = (pd_rank_roles(world).get(rank, "regular")
+            if cfg.mode == RoutingMode.PREFILL_DECODE else "regular")
+    run_worker_loop(eng, plane, role=role)
     dist.destroy_process_group()
 
 

@@ -22,7 +22,9 @@ def validate_config(cfg: RouterConfig) -> None:
         if sub is not None:
             sub.validate()
 
-    if cfg.mode == RoutingMode.PREFILL_DECODE:
+    if cfg.mode == RoutingMode.PREFILL_DECODE and cfg.connection_mode != ConnectionMode.RCCL:
+        # rccl PD derives its prefill/decode fleets from the torchrun ranks
+        # (odd = prefill), not from URLs
         if not cfg.prefill_urls and not cfg.discovery.enabled:
             raise ConfigError("prefill_decode mode requires --prefill URLs or service discovery")
         if not cfg.decode_urls and not cfg.discovery.enabled:

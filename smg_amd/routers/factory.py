@@ -19,6 +19,7 @@ ROUTER_IDS = {
     (ConnectionMode.GRPC, RoutingMode.PREFILL_DECODE): "grpc-pd",
     (ConnectionMode.GRPC, RoutingMode.ENCODE_PREFILL_DECODE): "grpc-epd",
     (ConnectionMode.RCCL, RoutingMode.REGULAR): "rccl-regular",
+    (ConnectionMode.RCCL, RoutingMode.PREFILL_DECODE): "rccl-pd",
 }
 
 
@@ -36,7 +37,9 @@ def create_router(ctx, config: Optional[RouterConfig] = None) -> Router:
         from .grpc.router import GrpcRouter
 
         return GrpcRouter(ctx, config)
-    if router_id == "rccl-regular":
+    if router_id in ("rccl-regular", "rccl-pd"):
+        # RcclRouter reads config.mode itself: PREFILL_DECODE turns on the
+        # plane PD topology (odd ranks prefill, KV handoff over xGMI)
         from .rccl_router import RcclRouter
 
         return RcclRouter(ctx, config)

@@ -301,6 +301,11 @@ class TickGateway:
         return statistics.median(self.routing_lat) * 1e3 if self.routing_lat else None
 
 
+def pd_rank_roles(world: int) -> Dict[int, str]:
+    """The plane PD topology rule: odd ranks prefill, even ranks decode."""
+    return {r: ("prefill" if r % 2 == 1 else "decode") for r in range(world)}
+
+
 def run_worker_loop(engine, plane: WorkerPlane, decode_burst: int = 2,
                     role: str = "regular") -> float:
     """Worker-rank loop (ranks >= 1): lockstep plane ticks against the local
@@ -401,6 +406,16 @@ class RcclRouter(Router):
             self.loop = asyncio.get_event_loop()
         self._states: Dict[int, _ReqState] = {}
         policy = ctx.policy_registry.get(self.model_id)
+        # PD serving over the plane: in PREFILL_DECODE mode, odd ranks run
+        # the prefill leg and even ranks (incl. the gateway's local engine)
+        # decode, with the xGMI KV handoff in between (pd_rank_roles mirrors
+        # bench --pd and cli.rccl_worker_main's parity rule)
+        from ..config import RoutingMode
+
+        pd_roles = (
+            pd_rank_roles(world)
+            if config.mode == RoutingMode.PREFILL_DECODE and world > 1 else None
+        )
         self.gw = TickGateway(
             self.workers,
             policy,
@@ -409,6 +424,7 @@ class RcclRouter(Router):
             model_id=self.model_id,
             on_event=self._on_event_tick_thread,
             metrics=getattr(ctx, "metrics", None),
+            pd_roles=pd_roles,
         )
         self._stop = False
         self.idle_sleep_s = max(1, config.rccl.tick_interval_us) / 1e6

@@ -14,12 +14,15 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 def build_cfg(port: int):
     from smg_amd.cli import to_router_config
 
-    cfg = to_router_config([
+    argv = [
         "launch",
         "--host", "127.0.0.1", "--port", str(port),
         "--connection-mode", "rccl",
         "--policy", "round_robin",
-    ])
+    ]
+    if os.environ.get("SMG_TEST_PD"):
+        argv += ["--pd-disaggregation"]
+    cfg = to_router_config(argv)
     cfg.health_check.disable = True
     cfg.prometheus_port = None
     return cfg

@@ -102,3 +102,26 @@ def test_bench_pd_three_rank_gloo():
     assert result["value"] > 0
     assert "pd over rccl-xgmi" in result["config"]["parallelism"]
     assert "1p+2d" in result["config"]["parallelism"]
+
+
+def test_rccl_pd_serving_three_rank_gloo():
+    """`smg launch --connection-mode rccl --pd-disaggregation` world 3:
+    chat served with the prefill leg on rank 1 and KV handoffs to decode
+    ranks 0/2."""
+    env = dict(os.environ)
+    env["MASTER_ADDR"] = "127.0.0.1"
+    env["SMG_TEST_PORT"] = "31899"
+    env["SMG_TEST_PD"] = "1"
+    cmd = [
+        sys.executable, "-m", "torch.distributed.run",
+        "--nnodes=1", "--nproc-per-node", "3",
+        "--master-addr", "127.0.0.1", "--master-port", "29589",
+        os.path.join(REPO, "tests", "rccl_serve_helper.py"),
+    ]
+    out = subprocess.run(cmd, capture_output=True, text=True, timeout=300, cwd=REPO, env=env)
+    assert out.returncode == 0, (out.stdout[-1500:], out.stderr[-1500:])
+    line = [l for l in out.stdout.splitlines() if l.startswith("RESULT ")][-1]
+    results = json.loads(line[7:])
+    assert results["completions"] == [4] * 6
+    # the prefill rank handled every prefill leg; decode ranks produced tokens
+    assert results["worker_processed"].get("rccl://rank-1", 0) >= 6
