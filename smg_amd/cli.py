@@ -197,6 +197,9 @@ def build_parser() -> argparse.ArgumentParser:
     g.add_argument("--mesh-advertise-host", default=None)
     g.add_argument("--mesh-port", type=int, default=32300)
     g.add_argument("--mesh-peer-urls", nargs="*", default=[])
+    g.add_argument("--mesh-mtls-cert", default=None, help="mesh mTLS: this node's certificate (PEM)")
+    g.add_argument("--mesh-mtls-key", default=None)
+    g.add_argument("--mesh-mtls-ca", default=None, help="mesh CA every replica chains to")
 
     g = p.add_argument_group("Service Discovery (Kubernetes)")
     g.add_argument("--service-discovery", action="store_true")
@@ -359,6 +362,9 @@ def to_router_config(argv: Optional[List[str]] = None) -> RouterConfig:
     cfg.mesh.advertise_host = args.mesh_advertise_host
     cfg.mesh.port = args.mesh_port
     cfg.mesh.peer_urls = args.mesh_peer_urls
+    cfg.mesh.mtls_cert = args.mesh_mtls_cert
+    cfg.mesh.mtls_key = args.mesh_mtls_key
+    cfg.mesh.mtls_ca = args.mesh_mtls_ca
     cfg.discovery.enabled = args.service_discovery
     cfg.discovery.selector = dict(s.split("=", 1) for s in args.selector if "=" in s)
     cfg.discovery.port = args.service_discovery_port

@@ -155,6 +155,12 @@ class MeshConfig:
     advertise_host: Optional[str] = None
     port: int = 32300
     peer_urls: List[str] = field(default_factory=list)
+    # mutual TLS between gateway replicas (reference crates/mesh/src/mtls.rs):
+    # when cert+key+ca are set, the mesh listener REQUIRES client certs from
+    # the same CA and outbound gossip presents this node's cert
+    mtls_cert: Optional[str] = None
+    mtls_key: Optional[str] = None
+    mtls_ca: Optional[str] = None
 
 
 @dataclass

@@ -48,6 +48,7 @@ async def start_mesh_server(mesh: MeshNode, host: str, port: int):
     add_mesh_routes(app, mesh)
     runner = web.AppRunner(app, access_log=None)
     await runner.setup()
-    site = web.TCPSite(runner, host, port)
+    # mTLS listener (reference mtls.rs): requires a peer cert from the mesh CA
+    site = web.TCPSite(runner, host, port, ssl_context=mesh.server_ssl_context())
     await site.start()
     return runner

@@ -45,6 +45,9 @@ class MeshNode:
         probe_timeout: float = 0.5,
         suspect_timeout: float = 3.0,
         indirect_k: int = 2,
+        mtls_cert: Optional[str] = None,
+        mtls_key: Optional[str] = None,
+        mtls_ca: Optional[str] = None,
     ):
         self.node_id = node_id
         self.advertise_url = advertise_url.rstrip("/")
@@ -75,11 +78,44 @@ class MeshNode:
         # POST so a post-partition backlog drains in bounded batches
         self.max_sync_ops_per_post = 500
         self.max_sync_posts_per_round = 8
+        # mutual TLS (reference mesh/src/mtls.rs): the SAME CA signs every
+        # replica; outbound sessions present our cert, the listener requires
+        # a peer cert chained to the CA
+        self._mtls = (mtls_cert, mtls_key, mtls_ca)
+
+    def client_ssl_context(self):
+        """SSLContext for OUTBOUND gossip (client cert + CA pinning), or None
+        when mTLS is off."""
+        cert, key, ca = self._mtls
+        if not (cert and key and ca):
+            return None
+        import ssl
+
+        ctx = ssl.create_default_context(ssl.Purpose.SERVER_AUTH, cafile=ca)
+        ctx.load_cert_chain(cert, key)
+        ctx.check_hostname = False  # peers are addressed by IP inside the mesh
+        return ctx
+
+    def server_ssl_context(self):
+        """SSLContext for the mesh LISTENER: requires a client cert from the
+        mesh CA, or None when mTLS is off."""
+        cert, key, ca = self._mtls
+        if not (cert and key and ca):
+            return None
+        import ssl
+
+        ctx = ssl.create_default_context(ssl.Purpose.CLIENT_AUTH, cafile=ca)
+        ctx.load_cert_chain(cert, key)
+        ctx.verify_mode = ssl.CERT_REQUIRED
+        return ctx
 
     # ---- lifecycle ---------------------------------------------------------
     async def start(self, peer_urls: List[str]) -> None:
         self._stopped.clear()
-        self._session = aiohttp.ClientSession(timeout=aiohttp.ClientTimeout(total=5))
+        ssl_ctx = self.client_ssl_context()
+        connector = aiohttp.TCPConnector(ssl=ssl_ctx) if ssl_ctx else None
+        self._session = aiohttp.ClientSession(
+            timeout=aiohttp.ClientTimeout(total=5), connector=connector)
         for url in peer_urls:
             await self.join(url)
         self._task = asyncio.ensure_future(self._event_loop())

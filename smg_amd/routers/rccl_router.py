@@ -64,7 +64,7 @@ class TickGateway:
         policy,
         plane: Optional[GatewayPlane] = None,
         local_engine=None,
-        decode_burst: int = 2,
+        decode_burst: int = 1,
         model_id: str = "default",
         max_new_arrivals_per_tick: int = 128,
         on_event: Optional[Callable[[int, int, int], None]] = None,
@@ -306,7 +306,7 @@ def pd_rank_roles(world: int) -> Dict[int, str]:
     return {r: ("prefill" if r % 2 == 1 else "decode") for r in range(world)}
 
 
-def run_worker_loop(engine, plane: WorkerPlane, decode_burst: int = 2,
+def run_worker_loop(engine, plane: WorkerPlane, decode_burst: int = 1,
                     role: str = "regular") -> float:
     """Worker-rank loop (ranks >= 1): lockstep plane ticks against the local
     engine until the gateway sends STOP.  `role="prefill"` makes every

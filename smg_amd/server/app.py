@@ -781,8 +781,11 @@ async def startup(config: RouterConfig, serve: bool = True) -> AppContext:
         from ..mesh.swim import MeshNode
 
         node_id = config.mesh.server_name or f"{config.host}:{config.port}"
-        advertise = f"http://{config.mesh.advertise_host or '127.0.0.1'}:{config.mesh.port}"
-        ctx.mesh = MeshNode(node_id, advertise)
+        m = config.mesh
+        scheme = "https" if (m.mtls_cert and m.mtls_key and m.mtls_ca) else "http"
+        advertise = f"{scheme}://{m.advertise_host or '127.0.0.1'}:{m.port}"
+        ctx.mesh = MeshNode(node_id, advertise,
+                            mtls_cert=m.mtls_cert, mtls_key=m.mtls_key, mtls_ca=m.mtls_ca)
         ctx.mesh_adapters = MeshAdapters(ctx.mesh, ctx)
         if serve:
             ctx._mesh_runner = await start_mesh_server(ctx.mesh, config.mesh.host, config.mesh.port)

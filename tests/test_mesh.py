@@ -347,3 +347,97 @@ class TestPartitionAndRepair:
                 await sb.close()
 
         runner(run())
+
+
+class TestMeshMtls:
+    """Mutual TLS between replicas (reference crates/mesh/src/mtls.rs): the
+    listener requires a client cert from the mesh CA; outbound gossip
+    presents this node's cert; a cert from a DIFFERENT CA is rejected."""
+
+    @staticmethod
+    def make_ca_and_cert(tmp_path, name, ca_name="mesh-ca"):
+        import subprocess
+
+        ca_key = tmp_path / f"{ca_name}.key"
+        ca_crt = tmp_path / f"{ca_name}.crt"
+        if not ca_crt.exists():
+            subprocess.run(["openssl", "req", "-x509", "-newkey", "rsa:2048", "-nodes",
+                            "-keyout", str(ca_key), "-out", str(ca_crt), "-days", "1",
+                            "-subj", f"/CN={ca_name}"], check=True, capture_output=True)
+        key = tmp_path / f"{name}.key"
+        csr = tmp_path / f"{name}.csr"
+        crt = tmp_path / f"{name}.crt"
+        ext = tmp_path / f"{name}.ext"
+        ext.write_text("subjectAltName=IP:127.0.0.1,DNS:localhost\n")
+        subprocess.run(["openssl", "req", "-newkey", "rsa:2048", "-nodes",
+                        "-keyout", str(key), "-out", str(csr), "-subj", f"/CN={name}"],
+                       check=True, capture_output=True)
+        subprocess.run(["openssl", "x509", "-req", "-in", str(csr), "-CA", str(ca_crt),
+                        "-CAkey", str(ca_key), "-CAcreateserial", "-out", str(crt),
+                        "-days", "1", "-extfile", str(ext)], check=True, capture_output=True)
+        return str(crt), str(key), str(ca_crt)
+
+    def test_mtls_gossip_and_rejection(self, runner, tmp_path):
+        from smg_amd.mesh.server import start_mesh_server
+
+        async def run():
+            a_crt, a_key, ca = self.make_ca_and_cert(tmp_path, "node-a")
+            b_crt, b_key, _ = self.make_ca_and_cert(tmp_path, "node-b")
+            import socket
+
+            s = socket.socket(); s.bind(("127.0.0.1", 0)); port_a = s.getsockname()[1]; s.close()
+            s = socket.socket(); s.bind(("127.0.0.1", 0)); port_b = s.getsockname()[1]; s.close()
+            a = MeshNode("a", f"https://127.0.0.1:{port_a}", probe_interval=0.1,
+                         probe_timeout=0.5, suspect_timeout=1.0,
+                         mtls_cert=a_crt, mtls_key=a_key, mtls_ca=ca)
+            b = MeshNode("b", f"https://127.0.0.1:{port_b}", probe_interval=0.1,
+                         probe_timeout=0.5, suspect_timeout=1.0,
+                         mtls_cert=b_crt, mtls_key=b_key, mtls_ca=ca)
+            ra = await start_mesh_server(a, "127.0.0.1", port_a)
+            rb = await start_mesh_server(b, "127.0.0.1", port_b)
+            try:
+                await a.start([])
+                await b.start([a.advertise_url])
+                assert "a" in b.members
+                a.kv.register_namespace("x"); b.kv.register_namespace("x")
+                a.kv.put("x", "secure", 1)
+                for _ in range(40):
+                    if b.kv.get("x", "secure"):
+                        break
+                    await asyncio.sleep(0.1)
+                assert b.kv.get("x", "secure") == 1  # gossip over mTLS works
+
+                # a client WITHOUT a cert is rejected by the handshake
+                import aiohttp
+                import ssl as _ssl
+
+                anon_ctx = _ssl.create_default_context(cafile=ca)
+                anon_ctx.check_hostname = False
+                async with aiohttp.ClientSession(
+                        connector=aiohttp.TCPConnector(ssl=anon_ctx)) as sess:
+                    with pytest.raises(Exception):
+                        async with sess.post(a.advertise_url + "/mesh/ping",
+                                             json={"from": "evil"},
+                                             timeout=aiohttp.ClientTimeout(total=3)) as r:
+                            await r.read()
+
+                # a cert from a DIFFERENT CA is rejected too
+                evil_crt, evil_key, _evil_ca = self.make_ca_and_cert(
+                    tmp_path, "evil", ca_name="other-ca")
+                evil_ctx = _ssl.create_default_context(cafile=ca)
+                evil_ctx.check_hostname = False
+                evil_ctx.load_cert_chain(evil_crt, evil_key)
+                async with aiohttp.ClientSession(
+                        connector=aiohttp.TCPConnector(ssl=evil_ctx)) as sess:
+                    with pytest.raises(Exception):
+                        async with sess.post(a.advertise_url + "/mesh/ping",
+                                             json={"from": "evil"},
+                                             timeout=aiohttp.ClientTimeout(total=3)) as r:
+                            await r.read()
+            finally:
+                await a.stop()
+                await b.stop()
+                await ra.cleanup()
+                await rb.cleanup()
+
+        runner(run())
