@@ -67,6 +67,10 @@ def parse_args():
                    help="PD disaggregation over the plane: odd ranks prefill, even ranks "
                         "(incl. the gateway) decode; KV hands off over xGMI p2p")
     p.add_argument("--no-graphs", action="store_true", help="disable hipGraph decode capture")
+    p.add_argument("--prefill-group", type=int, default=16,
+                   help="requests sharing one batched prefill forward")
+    p.add_argument("--arrival-cap", type=int, default=128,
+                   help="max new arrivals routed per tick")
     p.add_argument("--decode-burst", type=int, default=1, help="decode iterations per tick "
                    "(1 measured fastest with the GQA model: 859 vs 682 req/s at burst 2)")
     p.add_argument("--seed", type=int, default=1234)
@@ -103,6 +107,7 @@ def engine_config(args) -> TorchEngineConfig:
     gqa = getattr(args, "model", "mha") == "gqa"
     cfg = TorchEngineConfig.bench_1b_gqa() if gqa else TorchEngineConfig.bench_1b()
     cfg.kv_fp8 = bool(getattr(args, "kv_fp8", False))
+    cfg.prefill_group = getattr(args, "prefill_group", 16)
     cfg.max_slots = args.concurrency + 8
     cfg.max_seq = args.prefix_len + args.suffix_len + args.max_new + 16
     return cfg
@@ -159,7 +164,7 @@ def gateway_main(rank: int, world: int, args, device: str, backend: str):
     gw = TickGateway(
         workers, policy, plane=plane, local_engine=eng,
         decode_burst=args.decode_burst, model_id="bench-1b",
-        pd_roles=pd_roles,
+        pd_roles=pd_roles, max_new_arrivals_per_tick=args.arrival_cap,
     )
     vocab = engine_config(args).vocab_size
     gen = LoadGen(args, vocab)

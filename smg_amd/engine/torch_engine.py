@@ -38,6 +38,10 @@ class TorchEngineConfig:
     max_slots: int = 64
     max_seq: int = 2048
     prefill_chunk: int = 2048
+    # batched-prefill shaping: how many equal-remaining requests share one
+    # forward, and the total-token budget multiple that caps the group
+    prefill_group: int = 16
+    prefill_budget_mult: float = 2.0
     dtype: str = "bfloat16"
     seed: int = 1234
     # prefix KV cache (vLLM/SGLang-style): shared prompt prefixes skip
@@ -421,9 +425,9 @@ class TorchEngine:
             pending.sort(key=lambda sr: len(sr[1].tokens) - sr[1].prefilled)
             min_rem = len(pending[0][1].tokens) - pending[0][1].prefilled
             L = min(min_rem, c.prefill_chunk)
-            group = [sr for sr in pending if len(sr[1].tokens) - sr[1].prefilled >= L][:16]
-            # keep the total under ~2x budget
-            while len(group) > 1 and len(group) * L > 2 * c.prefill_chunk:
+            group = [sr for sr in pending if len(sr[1].tokens) - sr[1].prefilled >= L][: c.prefill_group]
+            # keep the total under the budget multiple
+            while len(group) > 1 and len(group) * L > c.prefill_budget_mult * c.prefill_chunk:
                 group.pop()
             items = []
             for slot, req in group:
