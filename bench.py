@@ -67,7 +67,7 @@ def parse_args():
                    help="PD disaggregation over the plane: odd ranks prefill, even ranks "
                         "(incl. the gateway) decode; KV hands off over xGMI p2p")
     p.add_argument("--no-graphs", action="store_true", help="disable hipGraph decode capture")
-    p.add_argument("--prefill-group", type=int, default=16,
+    p.add_argument("--prefill-group", type=int, default=32,
                    help="requests sharing one batched prefill forward")
     p.add_argument("--arrival-cap", type=int, default=128,
                    help="max new arrivals routed per tick")
@@ -107,7 +107,7 @@ def engine_config(args) -> TorchEngineConfig:
     gqa = getattr(args, "model", "mha") == "gqa"
     cfg = TorchEngineConfig.bench_1b_gqa() if gqa else TorchEngineConfig.bench_1b()
     cfg.kv_fp8 = bool(getattr(args, "kv_fp8", False))
-    cfg.prefill_group = getattr(args, "prefill_group", 16)
+    cfg.prefill_group = getattr(args, "prefill_group", 32)
     cfg.max_slots = args.concurrency + 8
     cfg.max_seq = args.prefix_len + args.suffix_len + args.max_new + 16
     return cfg

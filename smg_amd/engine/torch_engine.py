@@ -40,7 +40,7 @@ class TorchEngineConfig:
     prefill_chunk: int = 2048
     # batched-prefill shaping: how many equal-remaining requests share one
     # forward, and the total-token budget multiple that caps the group
-    prefill_group: int = 16
+    prefill_group: int = 32
     prefill_budget_mult: float = 2.0
     dtype: str = "bfloat16"
     seed: int = 1234
@@ -109,6 +109,22 @@ class _Layer:
         self.ln2 = torch.ones(d, device=device, dtype=dtype)
 
 
+class _StepHandle:
+    """Deferred-read result of TorchEngine.step_launch: a CUDA event gating a
+    pinned slab of sampled token values, plus the (container, key, slab_idx)
+    patch list and the event rows awaiting those values."""
+
+    __slots__ = ("ev", "slab", "n", "fills", "events", "produced")
+
+    def __init__(self):
+        self.ev = None
+        self.slab: Optional[torch.Tensor] = None
+        self.n = 0
+        self.fills: List[tuple] = []
+        self.events: List[list] = []
+        self.produced = 0
+
+
 def _rms(x, weight):
     # fused on ROCm (single kernel); computes in fp32 internally
     return F.rms_norm(x, (x.shape[-1],), weight=weight, eps=1e-5)
@@ -175,6 +191,8 @@ class TorchEngine:
         self._parked: Dict[str, tuple] = {}  # rid -> (slot, req, first_tok) awaiting export_kv
         self.total_generated = 0
         self._step_events: List[tuple] = []  # (rid, token, done) since last drain
+        self._slabs: List[Optional[torch.Tensor]] = [None] * 4  # pinned D2H slabs
+        self._slab_i = 0
         self.graphs = graphs
         self._graph_cache: Dict[int, tuple] = {}
         # prefix KV cache arena: same [L, 2, slot, H, T, D] layout as self.kv
@@ -390,11 +408,17 @@ class TorchEngine:
         }
 
     # ---- engine step -----------------------------------------------------
-    def step(self, decode_burst: int = 1) -> int:
-        """One engine iteration: admit + one prefill chunk + up to
-        `decode_burst` decode tokens for every running slot (bursting
-        amortizes the host-side tick/event overhead; the serving tick loop
-        uses burst 4).  Returns tokens produced."""
+    def step_launch(self, decode_burst: int = 1) -> "_StepHandle":
+        """Launch one engine iteration WITHOUT reading results back.
+
+        All admission/completion control flow here is host-deterministic —
+        "done" is a token-COUNT condition and sampling is argmax payload — so
+        the sampled token VALUES are only event payload.  They are copied
+        device→pinned-slab asynchronously and read in step_finish().  The
+        pipelined TickGateway resolves tick N-1's handle while tick N's
+        kernels run, which keeps the GPU fed through the gateway's
+        routing/event CPU phase (measured: 56% GPU-busy in the timed bench
+        region before this split, with a ~5 ms CPU gap every tick)."""
         c = self.cfg
         # admission
         while self.waiting and self._free_slots:
@@ -403,7 +427,9 @@ class TorchEngine:
             self.running[req.slot] = req
             self._seq_len_host[req.slot] = 0
 
-        produced = 0
+        h = _StepHandle()
+        samples: List[tuple] = []  # (device tensor, slab base offset)
+        off = 0
         # prefill: prefix-cache restore for fresh slots, then BATCHED chunked
         # prefill — equal-remaining requests share one forward (the common
         # case: cache-hit suffixes of identical length)
@@ -432,9 +458,12 @@ class TorchEngine:
             items = []
             for slot, req in group:
                 items.append((slot, req.prefilled, req.tokens[req.prefilled: req.prefilled + L]))
-            first_toks = self._prefill_batch(items)
+            nxt_pf = self._prefill_batch(items)  # device [len(group)]
+            pf_off = off
+            samples.append((nxt_pf, pf_off))
+            off += len(items)
             finished_pf = []
-            for (slot, req), tok in zip(group, first_toks):
+            for i, (slot, req) in enumerate(group):
                 req.prefilled += L
                 if req.prefilled >= len(req.tokens):
                     self._prefix_store(slot, req.tokens)
@@ -443,22 +472,29 @@ class TorchEngine:
                         # the sampled first token with the PREFILLED flag, and
                         # PARK the slot for export_kv
                         del self.running[slot]
-                        self._parked[req.rid] = (slot, req, int(tok))
-                        self._step_events.append((req.rid, req.prefilled, self.PLEN_INFO))
-                        self._step_events.append((req.rid, tok, self.PREFILLED))
+                        parked = [slot, req, None]
+                        self._parked[req.rid] = parked
+                        h.fills.append((parked, 2, pf_off + i))
+                        h.events.append([req.rid, req.prefilled, self.PLEN_INFO])
+                        row = [req.rid, None, self.PREFILLED]
+                        h.fills.append((row, 1, pf_off + i))
+                        h.events.append(row)
                         continue
                     # the final prefill chunk's logits sample the first
                     # generated token — decode then starts from the SAMPLED
                     # token at position n, instead of re-feeding the last
                     # prompt token (which wasted a KV slot and conditioned
                     # the first token on [.., t_{n-1}, t_{n-1}])
-                    req.generated.append(tok)
-                    produced += 1
+                    req.generated.append(-1)  # value patched in step_finish
+                    h.fills.append((req.generated, len(req.generated) - 1, pf_off + i))
+                    h.produced += 1
                     self.total_generated += 1
                     if len(req.generated) >= req.max_new or self._seq_len_host[slot] >= c.max_seq - 2:
                         req.done = True
                         finished_pf.append(slot)
-                    self._step_events.append((req.rid, tok, 1 if req.done else 0))
+                    row = [req.rid, None, 1 if req.done else 0]
+                    h.fills.append((row, 1, pf_off + i))
+                    h.events.append(row)
             for s in finished_pf:
                 del self.running[s]
                 self._free_slots.append(s)
@@ -470,17 +506,23 @@ class TorchEngine:
             decode_slots = [s for s, r in self.running.items() if r.prefilled >= len(r.tokens)]
             if not decode_slots:
                 break
-            new_tokens = self._decode(decode_slots)
+            nxt_full = self._decode_launch(decode_slots)  # device [max_slots]
+            base = off
+            samples.append((nxt_full, base))
+            off += nxt_full.numel()
             finished = []
-            for s, tok in zip(decode_slots, new_tokens):
+            for s in decode_slots:
                 r = self.running[s]
-                r.generated.append(int(tok))
-                produced += 1
+                r.generated.append(-1)  # value patched in step_finish
+                h.fills.append((r.generated, len(r.generated) - 1, base + s))
+                h.produced += 1
                 self.total_generated += 1
                 if len(r.generated) >= r.max_new or self._seq_len_host[s] >= c.max_seq - 2:
                     r.done = True
                     finished.append(s)
-                self._step_events.append((r.rid, int(tok), 1 if r.done else 0))
+                row = [r.rid, None, 1 if r.done else 0]
+                h.fills.append((row, 1, base + s))
+                h.events.append(row)
             for s in finished:
                 del self.running[s]
                 self._free_slots.append(s)
@@ -493,7 +535,50 @@ class TorchEngine:
                 # free, so this is the difference between O(active) and
                 # O(capacity) attention work)
                 self.seq_len[torch.tensor(finished, device=self.device)] = 0
-        return produced
+
+        if off:
+            slab = self._get_slab(off)
+            for t, base in samples:
+                slab[base: base + t.numel()].copy_(t, non_blocking=True)
+            h.slab, h.n = slab, off
+            if self.device.type == "cuda":
+                h.ev = torch.cuda.Event()
+                h.ev.record()
+        return h
+
+    def step_finish(self, handle: "_StepHandle") -> int:
+        """Resolve a step_launch handle: wait for its D2H copies, patch the
+        token values into events/generated/parked, and publish the events.
+        Waits only on the handle's own CUDA event — later launches keep
+        running."""
+        if handle is None:
+            return 0
+        if handle.ev is not None:
+            handle.ev.synchronize()
+        if handle.fills:
+            vals = handle.slab[: handle.n].tolist()
+            for container, key, idx in handle.fills:
+                container[key] = int(vals[idx])
+        self._step_events.extend(tuple(r) for r in handle.events)
+        return handle.produced
+
+    def step(self, decode_burst: int = 1) -> int:
+        """One engine iteration: admit + one prefill chunk + up to
+        `decode_burst` decode tokens for every running slot.  Synchronous
+        form of step_launch + step_finish; returns tokens produced."""
+        return self.step_finish(self.step_launch(decode_burst))
+
+    def _get_slab(self, n: int) -> torch.Tensor:
+        """Rotating pinned host slabs for the deferred token reads (4 deep:
+        the pipelined gateway holds at most one unresolved handle while the
+        next launch writes a different slab)."""
+        self._slab_i = (self._slab_i + 1) % len(self._slabs)
+        s = self._slabs[self._slab_i]
+        if s is None or s.numel() < n:
+            s = torch.empty(max(n, 1024), dtype=torch.long,
+                            pin_memory=self.device.type == "cuda")
+            self._slabs[self._slab_i] = s
+        return s
 
     # ---- forwards ----------------------------------------------------------
     def _kv_hist(self, li: int, which: int, slots):
@@ -680,8 +765,11 @@ class TorchEngine:
         h = self.embed[t]  # [B, L, D]
         pos = starts.unsqueeze(1) + torch.arange(L, device=self.device)  # [B, L]
         freqs = self.freqs_cis[pos].unsqueeze(1)  # [B, 1, L, D/2]
-        t_max = int(starts.max().item()) + L
-        fresh = int(starts.max().item()) == 0  # cold prefill: flash causal path
+        # host-derived (starts is built from host ints): no device sync here,
+        # so the whole prefill launches without stalling the pipeline
+        max_start = max(st for _, st, _ in items)
+        t_max = max_start + L
+        fresh = max_start == 0  # cold prefill: flash causal path
         if not fresh:
             kpos = torch.arange(t_max, device=self.device)
             mask = kpos.view(1, 1, 1, -1) <= pos.view(B, 1, L, 1)
@@ -730,7 +818,7 @@ class TorchEngine:
                 # no mask) + square causal chunk — merged by logsumexp.
                 # Replaces the masked-sdpa math path (bmm + 37 MB mask add +
                 # softmax per layer).
-                start0 = int(starts[0].item())
+                start0 = int(items[0][1])
                 if self.cfg.gqa_group > 1 and not self._flash_gqa:
                     kk = self._expand_kv(self._kv_hist(li, 0, slots)[:, :, :start0])
                     vv = self._expand_kv(self._kv_hist(li, 1, slots)[:, :, :start0])
@@ -778,7 +866,7 @@ class TorchEngine:
         for slot, start, toks in items:
             self._seq_len_host[slot] = start + L
         self.seq_len[slots] = starts + L
-        return nxt.tolist()
+        return nxt  # device tensor: the caller defers (or performs) the D2H read
 
     def _decode_core(self, maxlen: int) -> torch.Tensor:
         """Full-arena decode forward: every slot participates with a static
@@ -918,7 +1006,11 @@ class TorchEngine:
         return out
 
     @torch.no_grad()
-    def _decode(self, slots: List[int]) -> List[int]:
+    def _decode_launch(self, slots: List[int]) -> torch.Tensor:
+        """One decode pass for every slot in `slots`; returns the FULL
+        [max_slots] sampled-token tensor on device without any host sync —
+        the caller reads the values later (step_finish) so the host keeps
+        launching while the GPU works."""
         maxlen = max(self._seq_len_host) + 1
         if self.graphs and self.device.type == "cuda":
             nxt = self._decode_graphed(maxlen)
@@ -927,7 +1019,12 @@ class TorchEngine:
         act = torch.tensor(slots, device=self.device)
         self._last_tok.index_copy_(0, act, nxt.index_select(0, act))
         self.seq_len.index_add_(0, act, torch.ones_like(act))
-        toks = nxt.index_select(0, act).tolist()
         for s in slots:
             self._seq_len_host[s] += 1
-        return toks
+        return nxt
+
+    @torch.no_grad()
+    def _decode(self, slots: List[int]) -> List[int]:
+        nxt = self._decode_launch(slots)
+        act = torch.tensor(slots, device=self.device)
+        return nxt.index_select(0, act).tolist()
