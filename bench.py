@@ -81,10 +81,14 @@ class LoadGen:
     """Shared-prefix synthetic chat traffic (reference scripts/sim_load.py)."""
 
     def __init__(self, args, vocab: int):
+        import numpy as np
+
+        self.np = np
         self.rng = random.Random(args.seed)
         self.vocab = vocab
         self.prefixes = [
-            [self.rng.randrange(vocab) for _ in range(args.prefix_len)] for _ in range(args.prefix_pool)
+            np.array([self.rng.randrange(vocab) for _ in range(args.prefix_len)], dtype=np.int64)
+            for _ in range(args.prefix_pool)
         ]
         self.suffix_len = args.suffix_len
         self.max_new = args.max_new
@@ -92,9 +96,9 @@ class LoadGen:
 
     def make(self):
         self._rid += 1
-        prompt = list(self.rng.choice(self.prefixes)) + [
-            self.rng.randrange(self.vocab) for _ in range(self.suffix_len)
-        ]
+        suffix = self.np.array(
+            [self.rng.randrange(self.vocab) for _ in range(self.suffix_len)], dtype=self.np.int64)
+        prompt = self.np.concatenate([self.rng.choice(self.prefixes), suffix])
         return self._rid, prompt, self.max_new
 
 

@@ -19,6 +19,7 @@ import math
 from dataclasses import dataclass
 from typing import Dict, List, Optional
 
+import numpy as np
 import torch
 import torch.nn.functional as F
 
@@ -193,6 +194,8 @@ class TorchEngine:
         self._step_events: List[tuple] = []  # (rid, token, done) since last drain
         self._slabs: List[Optional[torch.Tensor]] = [None] * 4  # pinned D2H slabs
         self._slab_i = 0
+        self._pf_stages: List[Optional[torch.Tensor]] = [None] * 3  # pinned H2D staging
+        self._pf_stage_i = 0
         self.graphs = graphs
         self._graph_cache: Dict[int, tuple] = {}
         # prefix KV cache arena: same [L, 2, slot, H, T, D] layout as self.kv
@@ -327,7 +330,11 @@ class TorchEngine:
             self._rid_counter += 1
             rid = f"req-{self._rid_counter}"
         c = self.cfg
-        toks = [t % c.vocab_size for t in tokens]
+        # prompts stay numpy int64 end-to-end: vectorized clamp here, zero-copy
+        # page-slices in the batched prefill, vectorized chain-key hashing in
+        # the prefix cache (the per-token Python list work was milliseconds per
+        # serving tick at 32 admissions x ~1.7K tokens)
+        toks = np.asarray(tokens, dtype=np.int64) % c.vocab_size
         if len(toks) > c.max_seq - 2:
             toks = toks[-(c.max_seq - 2):]  # keep the prompt tail
         max_new_tokens = max(1, min(max_new_tokens, c.max_seq - 1 - len(toks)))
@@ -663,7 +670,9 @@ class TorchEngine:
             entry = self._pc_keys.get(keys[i])
             if entry is not None:
                 slot, stored = entry
-                if stored >= p and self._pc_slot_tokens.get(slot, [])[:p] == tokens[:p]:
+                st = self._pc_slot_tokens.get(slot)
+                if (stored >= p and st is not None and len(st) >= p
+                        and np.array_equal(st[:p], np.asarray(tokens[:p], dtype=np.int64))):
                     # LRU touch
                     try:
                         self._pc_lru.remove(slot)
@@ -713,7 +722,7 @@ class TorchEngine:
         for i, k in enumerate(keys):
             self._pc_keys[k] = (victim, (i + 1) * page)
         self._pc_slot_keys[victim] = list(keys)
-        self._pc_slot_tokens[victim] = list(tokens[:plen])
+        self._pc_slot_tokens[victim] = np.array(tokens[:plen], dtype=np.int64)
         self._pc_lru.append(victim)
 
     def _mlp(self, h, layer):
@@ -747,9 +756,30 @@ class TorchEngine:
         g = self.cfg.gqa_group
         return t if g == 1 else t.repeat_interleave(g, dim=-3)
 
+    def _stage_tokens(self, rows: List, L: int) -> torch.Tensor:
+        """[B, L] token tensor on device via rotating pinned staging buffers
+        (3 deep: with the pipelined tick, tick N's H2D may still be in flight
+        when tick N+1 builds its batch).  torch.tensor(list-of-lists) was the
+        single most expensive host op in the launch phase."""
+        B = len(rows)
+        if self.device.type != "cuda":
+            return torch.from_numpy(
+                np.stack([np.asarray(r, dtype=np.int64) for r in rows]))
+        self._pf_stage_i = (self._pf_stage_i + 1) % len(self._pf_stages)
+        stage = self._pf_stages[self._pf_stage_i]
+        if stage is None or stage.shape[0] < B or stage.shape[1] < L:
+            stage = torch.empty(max(B, self.cfg.prefill_group),
+                                max(L, self.cfg.prefill_chunk),
+                                dtype=torch.long, pin_memory=True)
+            self._pf_stages[self._pf_stage_i] = stage
+        sn = stage.numpy()
+        for i, r in enumerate(rows):
+            sn[i, :L] = r
+        return stage[:B, :L].to(self.device, non_blocking=True)
+
     @torch.no_grad()
     def _prefill(self, slot: int, tokens: List[int], start: int) -> None:
-        self._prefill_batch([(slot, start, list(tokens))])
+        self._prefill_batch([(slot, start, np.asarray(tokens, dtype=np.int64))])
 
     @torch.no_grad()
     def _prefill_batch(self, items: List[tuple]) -> None:
@@ -761,7 +791,7 @@ class TorchEngine:
         L = len(items[0][2])
         slots = torch.tensor([s for s, _, _ in items], device=self.device)
         starts = torch.tensor([st for _, st, _ in items], device=self.device)
-        t = torch.tensor([toks for _, _, toks in items], device=self.device, dtype=torch.long)
+        t = self._stage_tokens([toks for _, _, toks in items], L)
         h = self.embed[t]  # [B, L, D]
         pos = starts.unsqueeze(1) + torch.arange(L, device=self.device)  # [B, L]
         freqs = self.freqs_cis[pos].unsqueeze(1)  # [B, 1, L, D/2]

@@ -184,7 +184,7 @@ class CacheAwarePolicy(LoadBalancingPolicy):
             min_idx = _min_load_idx(workers, candidates)
             imbalanced = self.is_imbalanced(workers, candidates)
             results = tree.match_and_insert_batch(
-                [list(info.tokens) for _, info in token_batches],
+                [info.tokens for _, info in token_batches],  # ndarray passes through
                 urls=[w.url for w in workers],
                 candidates=candidates,
                 loads=[w.active_requests for w in workers],

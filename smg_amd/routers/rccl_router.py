@@ -27,6 +27,8 @@ import json
 import statistics
 import threading
 import time
+
+import numpy as np
 from collections import deque
 from typing import Any, Callable, Dict, List, Optional, Sequence, Tuple
 
@@ -111,7 +113,9 @@ class TickGateway:
             rid = next(self._rid) & 0x7FFFFFFF
         if self.pd_roles:
             self._req_meta[rid] = (len(tokens), max_new)
-        self._pending.append((rid, list(tokens), max_new))
+        # numpy end-to-end: no-copy for ndarray submitters (bench/engine), one
+        # conversion for list submitters (HTTP path)
+        self._pending.append((rid, np.asarray(tokens, dtype=np.int64), max_new))
         return rid
 
     @property
