@@ -575,6 +575,189 @@ extern "C" int smg_attn_decode_launch_gqa(const void* q, const void* k, const vo
                            scale, (hipStream_t)stream);
 }
 
+// ---------------------------------------------------------------------------
+// v9 (bf16 only): packed v_dot2c_f32_bf16 K phase.
+// Measured motivation: the v7 kernel runs at 2.8 TB/s (35% of HBM peak) at
+// bench shapes and is latency/VALU-bound, not bandwidth-bound — the K dot
+// spends 5 VALU ops per bf16 pair (2 converts + 2 FMA + addressing).  Here
+// the K dot is one v_dot2c_f32_bf16 per pair against q pre-packed as bf16x2
+// in LDS (2.5x fewer K-phase VALU ops).  A two-tile (128-timestep) softmax
+// round was also tried: 256 VGPRs + 56 spilled — worse than the barrier
+// cost it saves, so rounds stay 64 timesteps.
+// ---------------------------------------------------------------------------
+typedef short __bf16x2 __attribute__((__vector_size__(2 * sizeof(short))));
+
+__device__ __forceinline__ float dot2_bf16(unsigned int a, unsigned int b, float acc) {
+    union {
+        unsigned int u;
+        __bf16x2 v;
+    } ca{a}, cb{b};
+    return __builtin_amdgcn_fdot2_f32_bf16(ca.v, cb.v, acc, false);
+}
+
+template <int G>
+__global__ void __launch_bounds__(WAVE, 2) smg_attn_decode2_t(
+    const __hip_bfloat16* __restrict__ q,
+    const void* __restrict__ k,
+    const void* __restrict__ v,
+    const int* __restrict__ pos,
+    __hip_bfloat16* __restrict__ out,
+    int n_slots, int n_heads, int n_kv_heads, int max_seq, int head_dim, float scale) {
+    int sh = blockIdx.x;
+    int slot = sh / n_kv_heads;
+    int kvh = sh % n_kv_heads;
+    if (slot >= n_slots) return;
+    int lane = threadIdx.x;
+    int T = pos[slot] + 1;
+    if (T > max_seq) T = max_seq;
+
+    const size_t head_base = ((size_t)slot * n_kv_heads + kvh) * (size_t)max_seq * head_dim;
+    const __hip_bfloat16* kh = (const __hip_bfloat16*)k + head_base;
+    const __hip_bfloat16* vh = (const __hip_bfloat16*)v + head_base;
+    const __hip_bfloat16* qh = q + ((size_t)slot * n_heads + (size_t)kvh * G) * head_dim;
+
+    // q staged as packed bf16 pairs (dot2 operands)
+    __shared__ unsigned int s_q2[G * 64];
+    __shared__ float s_p[G][WAVE];
+    {
+        const unsigned int* q32 = (const unsigned int*)qh;  // 256 B-aligned rows
+        for (int i = lane; i < G * (head_dim / 2); i += WAVE) s_q2[i] = q32[i];
+    }
+    __syncthreads();
+
+    const int vec_n = head_dim / 8;
+    const int chunks = vec_n;
+    const int rows_per = WAVE / chunks;
+    const int chunk = lane % chunks;
+    const int rgrp = lane / chunks;
+    float accv[G][8];
+#pragma unroll
+    for (int g = 0; g < G; ++g)
+#pragma unroll
+        for (int j = 0; j < 8; ++j) accv[g][j] = 0.f;
+    float m[G], l[G];
+#pragma unroll
+    for (int g = 0; g < G; ++g) { m[g] = -1e30f; l[g] = 0.f; }
+
+    for (int base = 0; base < T; base += WAVE) {
+        const int lim = min(WAVE, T - base);
+        const int t = base + lane;
+        float d[G];
+#pragma unroll
+        for (int g = 0; g < G; ++g) d[g] = 0.f;
+        if (t < T) {
+            const uint4* row = (const uint4*)(kh + (size_t)t * head_dim);
+#pragma unroll 4
+            for (int c = 0; c < vec_n; ++c) {
+                uint4 w = row[c];
+                const unsigned int* wp = (const unsigned int*)&w;
+#pragma unroll
+                for (int wi = 0; wi < 4; ++wi)
+#pragma unroll
+                    for (int g = 0; g < G; ++g)
+                        d[g] = dot2_bf16(wp[wi], s_q2[g * (head_dim / 2) + c * 4 + wi], d[g]);
+            }
+        }
+        float alpha[G];
+#pragma unroll
+        for (int g = 0; g < G; ++g) {
+            float score = (t < T) ? d[g] * scale : -1e30f;
+            float mr = score;
+#pragma unroll
+            for (int off = 32; off > 0; off >>= 1) mr = fmaxf(mr, __shfl_xor(mr, off, WAVE));
+            float m_new = fmaxf(m[g], mr);
+            alpha[g] = __expf(m[g] - m_new);
+            float p = (t < T) ? __expf(score - m_new) : 0.f;
+            s_p[g][lane] = p;
+            float pr = p;
+#pragma unroll
+            for (int off = 32; off > 0; off >>= 1) pr += __shfl_xor(pr, off, WAVE);
+            l[g] = l[g] * alpha[g] + pr;
+            m[g] = m_new;
+#pragma unroll
+            for (int j = 0; j < 8; ++j) accv[g][j] *= alpha[g];
+        }
+        __syncthreads();
+        const __hip_bfloat16* vtile = vh + (size_t)base * head_dim + chunk * 8;
+        if (lim == WAVE && head_dim == 128) {
+#pragma unroll
+            for (int it = 0; it < 16; ++it) {
+                const int j = rgrp + it * 4;  // rows_per == 4 at hd128
+                const uint4 w = *(const uint4*)(vtile + (size_t)j * head_dim);
+                const unsigned short* hsp = (const unsigned short*)&w;
+#pragma unroll
+                for (int g = 0; g < G; ++g) {
+                    const float pj = s_p[g][j];
+#pragma unroll
+                    for (int jj = 0; jj < 8; ++jj) accv[g][jj] += pj * bf16_to_f32(hsp[jj]);
+                }
+            }
+        } else {
+            for (int j = rgrp; j < lim; j += rows_per) {
+                const uint4 w = *(const uint4*)(vtile + (size_t)j * head_dim);
+                const unsigned short* hsp = (const unsigned short*)&w;
+#pragma unroll
+                for (int g = 0; g < G; ++g) {
+                    const float pj = s_p[g][j];
+#pragma unroll
+                    for (int jj = 0; jj < 8; ++jj) accv[g][jj] += pj * bf16_to_f32(hsp[jj]);
+                }
+            }
+        }
+        __syncthreads();
+    }
+#pragma unroll
+    for (int g = 0; g < G; ++g) {
+        for (int off = chunks; off < WAVE; off <<= 1) {
+#pragma unroll
+            for (int jj = 0; jj < 8; ++jj) accv[g][jj] += __shfl_xor(accv[g][jj], off, WAVE);
+        }
+    }
+    if (rgrp == 0) {
+#pragma unroll
+        for (int g = 0; g < G; ++g) {
+            float inv = l[g] > 0.f ? 1.f / l[g] : 0.f;
+            __hip_bfloat16* orow =
+                out + ((size_t)slot * n_heads + (size_t)kvh * G + g) * head_dim;
+#pragma unroll
+            for (int jj = 0; jj < 8; ++jj) orow[chunk * 8 + jj] = (__hip_bfloat16)(accv[g][jj] * inv);
+        }
+    }
+}
+
+extern "C" int smg_attn_decode_launch_gqa2(const void* q, const void* k, const void* v,
+                                           const void* pos, void* out, int n_slots, int n_heads,
+                                           int n_kv_heads, int max_seq, int head_dim, float scale,
+                                           void* stream, int kv_fp8) {
+    if (kv_fp8) // fp8 is convert-bound, not helped by dot2 — keep the v7 path
+        return smg_attn_decode_launch_gqa(q, k, v, pos, out, n_slots, n_heads, n_kv_heads,
+                                          max_seq, head_dim, scale, stream, kv_fp8);
+    if (head_dim > 128 || (head_dim & 7)) return -1;
+    int chunks = head_dim / 8;
+    if (chunks & (chunks - 1)) return -1;
+    if (n_kv_heads <= 0 || n_heads % n_kv_heads) return -4;
+    dim3 grid(n_slots * n_kv_heads);
+    const int G = n_heads / n_kv_heads;
+    hipStream_t s = (hipStream_t)stream;
+#define LAUNCH2_CASE(GG)                                                                       \
+    case GG:                                                                                   \
+        hipLaunchKernelGGL((smg_attn_decode2_t<GG>), grid, dim3(WAVE), 0, s,                   \
+                           (const __hip_bfloat16*)q, k, v, (const int*)pos,                    \
+                           (__hip_bfloat16*)out, n_slots, n_heads, n_kv_heads, max_seq,        \
+                           head_dim, scale);                                                   \
+        break;
+    switch (G) {
+        LAUNCH2_CASE(1)
+        LAUNCH2_CASE(2)
+        LAUNCH2_CASE(4)
+        LAUNCH2_CASE(8)
+        default:
+            return -4;
+    }
+#undef LAUNCH2_CASE
+    return hipGetLastError() == hipSuccess ? 0 : -2;
+}
+
 extern "C" int smg_attn_decode_launch_ex(const void* q, const void* k, const void* v,
                                          const void* pos, void* out, int n_slots, int n_heads,
                                          int max_seq, int head_dim, float scale, void* stream,

@@ -55,6 +55,9 @@ int smg_attn_decode_launch(const void* q, const void* k, const void* v, const vo
 int smg_attn_decode_launch_gqa(const void* q, const void* k, const void* v, const void* pos,
                                void* out, int n_slots, int n_heads, int n_kv_heads, int max_seq,
                                int head_dim, float scale, void* stream, int kv_fp8);
+int smg_attn_decode_launch_gqa2(const void* q, const void* k, const void* v, const void* pos,
+                                void* out, int n_slots, int n_heads, int n_kv_heads, int max_seq,
+                                int head_dim, float scale, void* stream, int kv_fp8);
 int smg_rope_kv_store_launch_gqa(const void* qkv, const void* freqs, const void* pos,
                                  void* k_cache, void* v_cache, void* q_out, int n_slots,
                                  int n_heads, int n_kv_heads, int max_seq, int head_dim,
@@ -373,6 +376,21 @@ PYBIND11_MODULE(_core, m) {
                                                   n_kv_heads, max_seq, head_dim, scale,
                                                   (void*)stream, kv_fp8);
               if (rc != 0) throw std::runtime_error("attn_decode launch failed rc=" + std::to_string(rc));
+          },
+          py::arg("q"), py::arg("k"), py::arg("v"), py::arg("pos"), py::arg("out"),
+          py::arg("n_slots"), py::arg("n_heads"), py::arg("max_seq"), py::arg("head_dim"),
+          py::arg("scale"), py::arg("stream"), py::arg("kv_fp8") = 0, py::arg("n_kv_heads") = 0);
+    // v9 decode attention: two-tile softmax + v_dot2c_f32_bf16 K phase
+    m.def("attn_decode2",
+          [](uintptr_t q, uintptr_t k, uintptr_t v, uintptr_t pos, uintptr_t out, int n_slots,
+             int n_heads, int max_seq, int head_dim, float scale, uintptr_t stream, int kv_fp8,
+             int n_kv_heads) {
+              if (n_kv_heads <= 0) n_kv_heads = n_heads;
+              int rc = smg_attn_decode_launch_gqa2((const void*)q, (const void*)k, (const void*)v,
+                                                   (const void*)pos, (void*)out, n_slots, n_heads,
+                                                   n_kv_heads, max_seq, head_dim, scale,
+                                                   (void*)stream, kv_fp8);
+              if (rc != 0) throw std::runtime_error("attn_decode2 launch failed rc=" + std::to_string(rc));
           },
           py::arg("q"), py::arg("k"), py::arg("v"), py::arg("pos"), py::arg("out"),
           py::arg("n_slots"), py::arg("n_heads"), py::arg("max_seq"), py::arg("head_dim"),

@@ -227,7 +227,12 @@ class TorchEngine:
                 from .. import _core
 
                 if hasattr(_core, "attn_decode"):
+                    # v9 (attn_decode2): v_dot2c_f32_bf16 K phase, +14% at bench
+                    # shapes; its launcher routes fp8 back to the v7 kernel
+                    # (fp8 is convert-bound).  SMG_ATTN_V1=1 pins v7.
                     self._hip_attn = _core.attn_decode
+                    if hasattr(_core, "attn_decode2") and not _os.environ.get("SMG_ATTN_V1"):
+                        self._hip_attn = _core.attn_decode2
                     self._pos_i32 = torch.zeros(c.max_slots, dtype=torch.int32, device=self.device)
                     self._attn_out = torch.zeros(
                         c.max_slots, c.n_heads, c.head_dim, device=self.device, dtype=self.dtype

@@ -100,6 +100,54 @@ class TestAttnDecodeKernel:
         err = (out.float() - ref).abs().max().item()
         assert err < tol, f"max err {err}"
 
+    @pytest.mark.parametrize("S,H,KVH,T,D,maxseq,fp8", [
+        (8, 16, 4, 600, 128, 704, 0),
+        (4, 8, 2, 77, 64, 128, 0),
+        (6, 16, 2, 200, 128, 256, 0),
+        (8, 16, 4, 600, 128, 704, 1),  # fp8 routes back to the v7 kernel
+        (3, 8, 8, 50, 128, 64, 0),
+        (5, 8, 4, 129, 128, 192, 0),   # odd T exercises the partial tile
+    ])
+    def test_v9_dot2_matches_sdpa(self, S, H, KVH, T, D, maxseq, fp8):
+        """v9 kernel (packed v_dot2c_f32_bf16 K phase, csrc/attn_decode.hip
+        smg_attn_decode2_t) vs fp32 sdpa — same contract as the v7 tests."""
+        if not hasattr(core, "attn_decode2"):
+            pytest.skip("attn_decode2 not in this build")
+        dev = "cuda:0"
+        g = torch.Generator(device=dev).manual_seed(2)
+        q = torch.randn(S, H, D, generator=g, device=dev, dtype=torch.float32).to(torch.bfloat16)
+        kf32 = torch.randn(S, KVH, maxseq, D, generator=g, device=dev, dtype=torch.float32)
+        vf32 = torch.randn(S, KVH, maxseq, D, generator=g, device=dev, dtype=torch.float32)
+        if fp8:
+            k = kf32.to(torch.float8_e4m3fn)
+            v = vf32.to(torch.float8_e4m3fn)
+            kref, vref = k.to(torch.float32), v.to(torch.float32)
+            tol = 0.25
+        else:
+            k = kf32.to(torch.bfloat16)
+            v = vf32.to(torch.bfloat16)
+            kref, vref = k.float(), v.float()
+            tol = 0.05
+        pos = torch.randint(0, T, (S,), generator=g, device=dev, dtype=torch.int32)
+        out = torch.zeros(S, H, D, device=dev, dtype=torch.bfloat16)
+        core.attn_decode2(
+            q.contiguous().data_ptr(), k.contiguous().data_ptr(), v.contiguous().data_ptr(),
+            pos.data_ptr(), out.data_ptr(), S, H, maxseq, D,
+            1.0 / math.sqrt(D), torch.cuda.current_stream().cuda_stream,
+            fp8, KVH,
+        )
+        torch.cuda.synchronize()
+        G = H // KVH
+        kx = kref.repeat_interleave(G, dim=1)
+        vx = vref.repeat_interleave(G, dim=1)
+        kpos = torch.arange(maxseq, device=dev)
+        mask = (kpos.unsqueeze(0) <= pos.unsqueeze(1).long()).unsqueeze(1).unsqueeze(1)
+        ref = torch.nn.functional.scaled_dot_product_attention(
+            q.float().unsqueeze(2), kx, vx, attn_mask=mask
+        ).squeeze(2)
+        err = (out.float() - ref).abs().max().item()
+        assert err < tol, f"max err {err}"
+
     def test_gqa_engine_decode_matches_eager(self):
         """Full GQA engine with the HIP kernels vs the torch-eager GQA path."""
         from smg_amd.engine.torch_engine import TorchEngine, TorchEngineConfig
