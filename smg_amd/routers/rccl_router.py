@@ -105,7 +105,8 @@ class TickGateway:
         self._rid = itertools.count(1)
         self.completed_total = 0
         self.routing_lat: List[float] = []
-        self.phase_t = {"route": 0.0, "local": 0.0, "plane": 0.0, "events": 0.0, "ticks": 0}
+        self.phase_t = {"route": 0.0, "local": 0.0, "plane": 0.0, "events": 0.0,
+                        "launch": 0.0, "finish": 0.0, "ticks": 0}
 
     # ---- submission (any thread) -----------------------------------------
     def submit(self, tokens: List[int], max_new: int, rid: Optional[int] = None) -> int:
@@ -219,7 +220,9 @@ class TickGateway:
             eng.submit(toks, max_new, rid=rid)
         self._local_pending.clear()
         handle = eng.step_launch(self.decode_burst)
-        self.phase_t["local"] += time.perf_counter() - tl
+        dt = time.perf_counter() - tl
+        self.phase_t["local"] += dt
+        self.phase_t["launch"] += dt
         # 2) plane exchange posted now (remote enqueues from last tick's
         # routing ride out with it, overlapping the local step)
         if self.plane is not None:
@@ -237,7 +240,9 @@ class TickGateway:
         if prev is not None:
             eng.step_finish(prev)
             local_events = eng.drain_events()
-        self.phase_t["local"] += time.perf_counter() - tl
+        dt = time.perf_counter() - tl
+        self.phase_t["local"] += dt
+        self.phase_t["finish"] += dt
         remote_arrays = []
         if self.plane is not None:
             tp = time.perf_counter()
