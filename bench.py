@@ -256,7 +256,12 @@ def gateway_main(rank: int, world: int, args, device: str, backend: str):
     phase_t["gen"] = phase_gen[0]
     ticks = max(1, phase_t.pop("ticks"))
     print("# tick breakdown ms/tick: " + " ".join(f"{k}={v*1e3/ticks:.3f}" for k, v in phase_t.items())
-          + f" ticks={ticks}", file=_sys.stderr)
+          + f" ticks={ticks} pfg_hits={getattr(eng, 'pfg_hits', 0)}"
+          + f" pfg_eager={getattr(eng, 'pfg_eager', 0)}", file=_sys.stderr)
+    if getattr(eng, "launch_t", None):
+        print("# launch breakdown ms/tick: "
+              + " ".join(f"{k}={v*1e3/ticks:.3f}" for k, v in eng.launch_t.items()),
+              file=_sys.stderr)
     print(json.dumps(result), flush=True)
 
 

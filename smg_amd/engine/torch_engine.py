@@ -16,6 +16,7 @@ graphs can capture it (enable with graphs=True).
 from __future__ import annotations
 
 import math
+import time
 from dataclasses import dataclass
 from typing import Dict, List, Optional
 
@@ -196,6 +197,18 @@ class TorchEngine:
         self._slab_i = 0
         self._pf_stages: List[Optional[torch.Tensor]] = [None] * 3  # pinned H2D staging
         self._pf_stage_i = 0
+        self._pfg_cache: Dict[tuple, tuple] = {}  # (B,L,start0,fresh) -> captured graph
+        self._pfg_bad: set = set()  # shapes whose capture failed (stay eager)
+        self.pfg_hits = 0
+        self.pfg_eager = 0
+        # host-time breakdown of step_launch (pf_book/dec_book include pf/decode)
+        self.launch_t = {"restore": 0.0, "pf": 0.0, "pf_book": 0.0,
+                         "decode": 0.0, "dec_book": 0.0}
+        # small-transfer staging: torch.tensor(list, device=cuda) is a
+        # BLOCKING H2D — the host waits for the copy, which is stream-ordered
+        # behind the whole queued tick, so each one drains the pipeline
+        self._h2d_bufs: List[Optional[tuple]] = [None] * 12
+        self._h2d_i = 0
         self.graphs = graphs
         self._graph_cache: Dict[int, tuple] = {}
         # prefix KV cache arena: same [L, 2, slot, H, T, D] layout as self.kv
@@ -432,6 +445,9 @@ class TorchEngine:
         routing/event CPU phase (measured: 56% GPU-busy in the timed bench
         region before this split, with a ~5 ms CPU gap every tick)."""
         c = self.cfg
+        lt = self.launch_t
+        _t = time.perf_counter
+        t0 = _t()
         # admission
         while self.waiting and self._free_slots:
             req = self.waiting.pop(0)
@@ -444,20 +460,36 @@ class TorchEngine:
         off = 0
         # prefill: prefix-cache restore for fresh slots, then BATCHED chunked
         # prefill — equal-remaining requests share one forward (the common
-        # case: cache-hit suffixes of identical length)
+        # case: cache-hit suffixes of identical length).  The arena copies
+        # are grouped by prefix length into ONE indexed device copy per
+        # length: per-slot 6-D slice assigns were ~150 us of host dispatch
+        # each, x32 hits/tick in the bench.
+        restore: Dict[int, list] = {}
         for slot, req in self.running.items():
             if req.prefilled == 0 and self._pc_arena is not None:
                 hit = self._prefix_lookup(req.tokens)
                 if hit is not None:
                     pslot, plen = hit
-                    self.kv[:, :, slot, :, :plen] = self._pc_arena[:, :, pslot, :, :plen]
+                    restore.setdefault(plen, []).append((slot, pslot, int(req.tokens[plen - 1])))
                     self._seq_len_host[slot] = plen
-                    self.seq_len[slot] = plen
-                    self._last_tok[slot] = int(req.tokens[plen - 1])
                     req.prefilled = plen
                     self.prefix_cache_hits += 1
                 else:
                     self.prefix_cache_miss += 1
+        for plen, group in restore.items():
+            if len(group) == 1:
+                slot, pslot, last = group[0]
+                self.kv[:, :, slot, :, :plen] = self._pc_arena[:, :, pslot, :, :plen]
+                self.seq_len[slot] = plen
+                self._last_tok[slot] = last
+            else:
+                meta = self._h2d_i64(np.asarray(group, dtype=np.int64).ravel()).view(-1, 3)
+                slots_t, pslots_t, lasts_t = meta[:, 0], meta[:, 1], meta[:, 2]
+                self.kv[:, :, slots_t, :, :plen] = self._pc_arena[:, :, pslots_t, :, :plen]
+                self.seq_len[slots_t] = plen
+                self._last_tok[slots_t] = lasts_t
+        t1 = _t()
+        lt["restore"] += t1 - t0
         pending = [(s, r) for s, r in self.running.items() if r.prefilled < len(r.tokens)]
         if pending:
             pending.sort(key=lambda sr: len(sr[1].tokens) - sr[1].prefilled)
@@ -470,7 +502,9 @@ class TorchEngine:
             items = []
             for slot, req in group:
                 items.append((slot, req.prefilled, req.tokens[req.prefilled: req.prefilled + L]))
+            t2 = _t()
             nxt_pf = self._prefill_batch(items)  # device [len(group)]
+            lt["pf"] += _t() - t2
             pf_off = off
             samples.append((nxt_pf, pf_off))
             off += len(items)
@@ -512,13 +546,17 @@ class TorchEngine:
                 self._free_slots.append(s)
                 self._seq_len_host[s] = 0
             if finished_pf:
-                self.seq_len[torch.tensor(finished_pf, device=self.device)] = 0
+                self.seq_len[self._h2d_i64(np.fromiter(finished_pf, dtype=np.int64))] = 0
+            lt["pf_book"] += _t() - t2
 
+        t3 = _t()
         for _ in range(max(1, decode_burst)):
             decode_slots = [s for s, r in self.running.items() if r.prefilled >= len(r.tokens)]
             if not decode_slots:
                 break
+            t4 = _t()
             nxt_full = self._decode_launch(decode_slots)  # device [max_slots]
+            lt["decode"] += _t() - t4
             base = off
             samples.append((nxt_full, base))
             off += nxt_full.numel()
@@ -546,8 +584,9 @@ class TorchEngine:
                 # KV window every step (in steady state most of the arena is
                 # free, so this is the difference between O(active) and
                 # O(capacity) attention work)
-                self.seq_len[torch.tensor(finished, device=self.device)] = 0
+                self.seq_len[self._h2d_i64(np.fromiter(finished, dtype=np.int64))] = 0
 
+        lt["dec_book"] += _t() - t3
         if off:
             slab = self._get_slab(off)
             for t, base in samples:
@@ -579,6 +618,27 @@ class TorchEngine:
         `decode_burst` decode tokens for every running slot.  Synchronous
         form of step_launch + step_finish; returns tokens produced."""
         return self.step_finish(self.step_launch(decode_burst))
+
+    def _h2d_i64(self, arr) -> torch.Tensor:
+        """Small int64 host->device transfer that does NOT block the host:
+        pinned staging + non_blocking copy into a persistent device buffer
+        (rotating 12-deep so in-flight DMAs are never overwritten within the
+        pipeline's one-tick lag)."""
+        arr = np.ascontiguousarray(arr, dtype=np.int64)
+        n = arr.shape[0]
+        if self.device.type != "cuda":
+            return torch.from_numpy(arr)
+        self._h2d_i = (self._h2d_i + 1) % len(self._h2d_bufs)
+        entry = self._h2d_bufs[self._h2d_i]
+        if entry is None or entry[0].numel() < n:
+            cap = max(n, 3 * (self.cfg.max_slots + 8))
+            entry = (torch.empty(cap, dtype=torch.long, pin_memory=True),
+                     torch.empty(cap, dtype=torch.long, device=self.device))
+            self._h2d_bufs[self._h2d_i] = entry
+        pin, dev = entry
+        pin[:n].numpy()[:] = arr
+        dev[:n].copy_(pin[:n], non_blocking=True)
+        return dev[:n]
 
     def _get_slab(self, n: int) -> torch.Tensor:
         """Rotating pinned host slabs for the deferred token reads (4 deep:
@@ -766,6 +826,15 @@ class TorchEngine:
         (3 deep: with the pipelined tick, tick N's H2D may still be in flight
         when tick N+1 builds its batch).  torch.tensor(list-of-lists) was the
         single most expensive host op in the launch phase."""
+        stage = self._fill_stage(rows, L)
+        if self.device.type != "cuda":
+            return stage
+        return stage.to(self.device, non_blocking=True)
+
+    def _fill_stage(self, rows: List, L: int) -> torch.Tensor:
+        """Fill a rotating pinned staging buffer with [B, L] token rows and
+        return the (host) slice — callers H2D it into a fresh tensor (eager)
+        or an existing graph input buffer (graphed)."""
         B = len(rows)
         if self.device.type != "cuda":
             return torch.from_numpy(
@@ -780,7 +849,7 @@ class TorchEngine:
         sn = stage.numpy()
         for i, r in enumerate(rows):
             sn[i, :L] = r
-        return stage[:B, :L].to(self.device, non_blocking=True)
+        return stage[:B, :L]
 
     @torch.no_grad()
     def _prefill(self, slot: int, tokens: List[int], start: int) -> None:
@@ -790,26 +859,133 @@ class TorchEngine:
     def _prefill_batch(self, items: List[tuple]) -> None:
         """Batched chunked prefill: `items` = [(slot, start, tokens)] with
         equal chunk lengths.  Per-request absolute positions drive rope and a
-        per-request causal mask over each slot's own KV window."""
+        per-request causal mask over each slot's own KV window.
+
+        The device work lives in _prefill_forward; on GPU with graphs on,
+        recurring (B, L, start0) shapes replay a captured hipGraph — the
+        eager prefill was ~260 Python-dispatched launches per tick and the
+        measured launch phase (18.2 ms/tick) was pure host time."""
         c = self.cfg
         B = len(items)
         L = len(items[0][2])
-        slots = torch.tensor([s for s, _, _ in items], device=self.device)
-        starts = torch.tensor([st for _, st, _ in items], device=self.device)
-        t = self._stage_tokens([toks for _, _, toks in items], L)
+        starts_host = [int(st) for _, st, _ in items]
+        max_start = max(starts_host)
+        t_max = max_start + L
+        fresh = max_start == 0  # cold prefill: flash causal path
+        uniform_start = not fresh and all(st == starts_host[0] for st in starts_host)
+        start0 = starts_host[0]
+        nxt = None
+        if self._pfg_enabled(B, fresh, uniform_start):
+            nxt = self._prefill_graphed(items, B, L, t_max, fresh, uniform_start, start0)
+        if nxt is None:
+            self.pfg_eager += 1
+        else:
+            self.pfg_hits += 1
+        if nxt is None:
+            slots = self._h2d_i64(np.fromiter((s for s, _, _ in items), dtype=np.int64, count=B))
+            starts = self._h2d_i64(np.asarray(starts_host, dtype=np.int64))
+            t = self._stage_tokens([toks for _, _, toks in items], L)
+            nxt = self._prefill_forward(slots, starts, t, B, L, t_max, fresh, uniform_start, start0)
+        for slot, start, toks in items:
+            self._seq_len_host[slot] = int(start) + L
+        return nxt  # device tensor: the caller defers (or performs) the D2H read
+
+    # ---- prefill hipGraph cache -------------------------------------------
+    PFG_CAP = 8  # distinct (B, L, start0, fresh) shapes kept captured
+
+    def _pfg_enabled(self, B: int, fresh: bool, uniform_start: bool) -> bool:
+        import os as _os
+
+        if not (self.graphs and self.device.type == "cuda") or _os.environ.get("SMG_NO_PREFILL_GRAPH"):
+            return False
+        if B < 4:  # singles/tiny groups: capture overhead isn't worth it
+            return False
+        # graphable paths: cold flash-causal, or uniform-start suffix merge
+        if fresh:
+            return True
+        use_fused = self._hip_fused is not None and getattr(self, "_hip_rope_prefill", None) is not None
+        return bool(uniform_start and use_fused and self._flash_lse)
+
+    def _prefill_graphed(self, items, B, L, t_max, fresh, uniform_start, start0):
+        """Replay (or capture, first time per shape) the prefill graph.
+        Returns the graph's output tensor, or None to fall back to eager.
+
+        Capture protocol: warmup runs execute the REAL batch (prefill is
+        idempotent — same inputs rewrite the same KV rows / seq_len /
+        _last_tok values), then capture records without executing, then one
+        replay performs this tick's work.  Input buffers (tokens / slots /
+        starts) are content-dynamic; B, L and the history depth start0 are
+        shape-defining and key the cache."""
+        key = (B, L, 0 if fresh else start0, fresh)
+        entry = self._pfg_cache.get(key)
+        if entry is None:
+            if key in self._pfg_bad or len(self._pfg_cache) >= self.PFG_CAP:
+                return None
+            try:
+                entry = self._pfg_capture(key, items, B, L, t_max, fresh, uniform_start, start0)
+            except Exception as exc:
+                import os as _os
+                import sys as _sys
+
+                self._pfg_bad.add(key)
+                if _os.environ.get("SMG_PFG_DEBUG"):
+                    print(f"[pfg] capture failed for {key}: {exc!r}", file=_sys.stderr)
+                # the failed attempt may have half-written state; the eager
+                # fallback below rewrites it (idempotent)
+                return None
+            self._pfg_cache[key] = entry
+        g, t_buf, slots_buf, starts_buf, nxt_out, slots_stage, starts_stage = entry
+        stage = self._fill_stage([toks for _, _, toks in items], L)
+        t_buf.copy_(stage, non_blocking=True)
+        slots_stage.copy_(torch.from_numpy(np.fromiter((s for s, _, _ in items), dtype=np.int64, count=B)))
+        starts_stage.copy_(torch.from_numpy(np.fromiter((st for _, st, _ in items), dtype=np.int64, count=B)))
+        slots_buf.copy_(slots_stage, non_blocking=True)
+        starts_buf.copy_(starts_stage, non_blocking=True)
+        g.replay()
+        return nxt_out
+
+    def _pfg_capture(self, key, items, B, L, t_max, fresh, uniform_start, start0):
+        dev = self.device
+        t_buf = torch.zeros(B, L, dtype=torch.long, device=dev)
+        slots_buf = torch.zeros(B, dtype=torch.long, device=dev)
+        starts_buf = torch.zeros(B, dtype=torch.long, device=dev)
+        slots_stage = torch.empty(B, dtype=torch.long, pin_memory=True)
+        starts_stage = torch.empty(B, dtype=torch.long, pin_memory=True)
+        # seed the buffers with the REAL batch so warmup executes this tick's
+        # actual (idempotent) work
+        stage = self._fill_stage([toks for _, _, toks in items], L)
+        t_buf.copy_(stage)
+        slots_buf.copy_(torch.tensor([s for s, _, _ in items], dtype=torch.long))
+        starts_buf.copy_(torch.tensor([st for _, st, _ in items], dtype=torch.long))
+        torch.cuda.synchronize()
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):  # warmup allocations outside capture
+                self._prefill_forward(slots_buf, starts_buf, t_buf, B, L, t_max,
+                                      fresh, uniform_start, start0)
+        torch.cuda.current_stream().wait_stream(s)
+        torch.cuda.synchronize()
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            nxt_out = self._prefill_forward(slots_buf, starts_buf, t_buf, B, L, t_max,
+                                            fresh, uniform_start, start0)
+        return (g, t_buf, slots_buf, starts_buf, nxt_out, slots_stage, starts_stage)
+
+    @torch.no_grad()
+    def _prefill_forward(self, slots, starts, t, B: int, L: int, t_max: int,
+                         fresh: bool, uniform_start: bool, start0: int):
+        """Device-only prefill forward (hipGraph-capturable: every input is a
+        tensor whose CONTENT may change between replays; B/L/start0 are
+        shape-defining and key the graph cache)."""
+        c = self.cfg
         h = self.embed[t]  # [B, L, D]
         pos = starts.unsqueeze(1) + torch.arange(L, device=self.device)  # [B, L]
         freqs = self.freqs_cis[pos].unsqueeze(1)  # [B, 1, L, D/2]
-        # host-derived (starts is built from host ints): no device sync here,
-        # so the whole prefill launches without stalling the pipeline
-        max_start = max(st for _, st, _ in items)
-        t_max = max_start + L
-        fresh = max_start == 0  # cold prefill: flash causal path
         if not fresh:
             kpos = torch.arange(t_max, device=self.device)
             mask = kpos.view(1, 1, 1, -1) <= pos.view(B, 1, L, 1)
         use_fused = self._hip_fused is not None and getattr(self, "_hip_rope_prefill", None) is not None
-        uniform_start = not fresh and all(int(st) == int(items[0][1]) for _, st, _ in items)
         if use_fused:
             # fused rope + KV scatter + [B,H,L,hd] emit (csrc/fused_decode.hip
             # smg_rope_prefill): replaces the rope complex-mul chains, the two
@@ -853,7 +1029,6 @@ class TorchEngine:
                 # no mask) + square causal chunk — merged by logsumexp.
                 # Replaces the masked-sdpa math path (bmm + 37 MB mask add +
                 # softmax per layer).
-                start0 = int(items[0][1])
                 if self.cfg.gqa_group > 1 and not self._flash_gqa:
                     kk = self._expand_kv(self._kv_hist(li, 0, slots)[:, :, :start0])
                     vv = self._expand_kv(self._kv_hist(li, 1, slots)[:, :, :start0])
@@ -898,8 +1073,6 @@ class TorchEngine:
         # their final chunk before decode ever reads it.
         nxt = (_rms(h[:, -1], self.ln_f) @ self.embed.t()).argmax(-1)  # [B]
         self._last_tok[slots] = nxt
-        for slot, start, toks in items:
-            self._seq_len_host[slot] = start + L
         self.seq_len[slots] = starts + L
         return nxt  # device tensor: the caller defers (or performs) the D2H read
 
@@ -1051,7 +1224,7 @@ class TorchEngine:
             nxt = self._decode_graphed(maxlen)
         else:
             nxt = self._decode_core(maxlen)
-        act = torch.tensor(slots, device=self.device)
+        act = self._h2d_i64(np.fromiter(slots, dtype=np.int64, count=len(slots)))
         self._last_tok.index_copy_(0, act, nxt.index_select(0, act))
         self.seq_len.index_add_(0, act, torch.ones_like(act))
         for s in slots:
