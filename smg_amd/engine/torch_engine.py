@@ -116,7 +116,7 @@ class _StepHandle:
     pinned slab of sampled token values, plus the (container, key, slab_idx)
     patch list and the event rows awaiting those values."""
 
-    __slots__ = ("ev", "slab", "n", "fills", "events", "produced")
+    __slots__ = ("ev", "slab", "n", "fills", "events", "produced", "dec_batches")
 
     def __init__(self):
         self.ev = None
@@ -125,6 +125,10 @@ class _StepHandle:
         self.fills: List[tuple] = []
         self.events: List[list] = []
         self.produced = 0
+        # decode bursts as arrays: (reqs, act_np, done_mask, slab_base) — the
+        # per-token tuple building happens in step_finish, overlapped with the
+        # NEXT tick's GPU work in the pipelined gateway
+        self.dec_batches: List[tuple] = []
 
 
 def _rms(x, weight):
@@ -134,7 +138,7 @@ def _rms(x, weight):
 
 class _Request:
     __slots__ = ("rid", "tokens", "max_new", "slot", "prefilled", "generated", "done",
-                 "prefill_only")
+                 "prefill_only", "pc_keys")
 
     def __init__(self, rid, tokens, max_new, prefill_only=False):
         self.rid = rid
@@ -145,6 +149,7 @@ class _Request:
         self.generated: List[int] = []
         self.done = False
         self.prefill_only = prefill_only
+        self.pc_keys = None  # prefix-cache chain keys, precomputed at submit
 
 
 class TorchEngine:
@@ -182,7 +187,9 @@ class TorchEngine:
             freqs = torch.outer(t, inv)
             self.freqs_cis = torch.polar(torch.ones_like(freqs), freqs)  # complex64 [T, D/2]
         self.seq_len = torch.zeros(c.max_slots, dtype=torch.long, device=self.device)
-        self._seq_len_host = [0] * c.max_slots
+        self._seq_len_host = np.zeros(c.max_slots, dtype=np.int64)
+        self._gen_cnt = np.zeros(c.max_slots, dtype=np.int64)  # len(generated) per slot
+        self._maxnew_arr = np.zeros(c.max_slots, dtype=np.int64)
         self._last_tok = torch.zeros(c.max_slots, dtype=torch.long, device=self.device)
         self._arange_slots = torch.arange(c.max_slots, device=self.device)
         self._free_slots = list(range(c.max_slots - 1, -1, -1))
@@ -357,6 +364,16 @@ class TorchEngine:
             toks = toks[-(c.max_seq - 2):]  # keep the prompt tail
         max_new_tokens = max(1, min(max_new_tokens, c.max_seq - 1 - len(toks)))
         req = _Request(rid, toks, max_new_tokens, prefill_only=prefill_only)
+        if self._pc_arena is not None:
+            # hash here: in the pipelined gateway, submit() runs in the
+            # routing phase (overlapped with the GPU tick) while the prefix
+            # lookup runs in the launch critical path
+            page = c.prefix_cache_page
+            pmax = min(((len(toks) - 1) // page) * page, c.prefix_cache_max, c.max_seq - 1)
+            if pmax >= page:
+                from ..kvindex.chain_keys import chain_keys
+
+                req.pc_keys = chain_keys(toks[:pmax], page)
         self._requests[rid] = req
         self.waiting.append(req)
         return rid
@@ -395,6 +412,8 @@ class TorchEngine:
         req.prefilled = plen
         req.slot = slot
         req.generated = [int(first_tok)]
+        self._gen_cnt[slot] = 1
+        self._maxnew_arr[slot] = max_new
         self._requests[rid] = req
         self.running[slot] = req
         return True
@@ -422,7 +441,7 @@ class TorchEngine:
 
     def load_snapshot(self) -> Dict:
         c = self.cfg
-        used = sum(self._seq_len_host[s] for s in self.running)
+        used = int(sum(self._seq_len_host[s] for s in self.running))
         return {
             "num_queue_tokens": sum(len(r.tokens) for r in self.waiting),
             "num_inflight_tokens": used,
@@ -454,6 +473,8 @@ class TorchEngine:
             req.slot = self._free_slots.pop()
             self.running[req.slot] = req
             self._seq_len_host[req.slot] = 0
+            self._gen_cnt[req.slot] = len(req.generated)
+            self._maxnew_arr[req.slot] = req.max_new
 
         h = _StepHandle()
         samples: List[tuple] = []  # (device tensor, slab base offset)
@@ -467,7 +488,7 @@ class TorchEngine:
         restore: Dict[int, list] = {}
         for slot, req in self.running.items():
             if req.prefilled == 0 and self._pc_arena is not None:
-                hit = self._prefix_lookup(req.tokens)
+                hit = self._prefix_lookup(req.tokens, keys=req.pc_keys)
                 if hit is not None:
                     pslot, plen = hit
                     restore.setdefault(plen, []).append((slot, pslot, int(req.tokens[plen - 1])))
@@ -533,6 +554,7 @@ class TorchEngine:
                     # the first token on [.., t_{n-1}, t_{n-1}])
                     req.generated.append(-1)  # value patched in step_finish
                     h.fills.append((req.generated, len(req.generated) - 1, pf_off + i))
+                    self._gen_cnt[slot] = len(req.generated)
                     h.produced += 1
                     self.total_generated += 1
                     if len(req.generated) >= req.max_new or self._seq_len_host[slot] >= c.max_seq - 2:
@@ -554,37 +576,37 @@ class TorchEngine:
             decode_slots = [s for s, r in self.running.items() if r.prefilled >= len(r.tokens)]
             if not decode_slots:
                 break
+            act_np = np.fromiter(decode_slots, dtype=np.int64, count=len(decode_slots))
+            reqs = [self.running[s] for s in decode_slots]
             t4 = _t()
-            nxt_full = self._decode_launch(decode_slots)  # device [max_slots]
+            nxt_full = self._decode_launch_np(act_np)  # device [max_slots]
             lt["decode"] += _t() - t4
             base = off
             samples.append((nxt_full, base))
             off += nxt_full.numel()
-            finished = []
-            for s in decode_slots:
-                r = self.running[s]
-                r.generated.append(-1)  # value patched in step_finish
-                h.fills.append((r.generated, len(r.generated) - 1, base + s))
-                h.produced += 1
-                self.total_generated += 1
-                if len(r.generated) >= r.max_new or self._seq_len_host[s] >= c.max_seq - 2:
-                    r.done = True
-                    finished.append(s)
-                row = [r.rid, None, 1 if r.done else 0]
-                h.fills.append((row, 1, base + s))
-                h.events.append(row)
-            for s in finished:
-                del self.running[s]
-                self._free_slots.append(s)
-                self._seq_len_host[s] = 0
-            if finished:
+            # vectorized lifecycle: counts are authoritative in _gen_cnt, the
+            # per-token event tuples are built in step_finish (overlapped
+            # with the next tick's GPU work in the pipelined gateway)
+            self._gen_cnt[act_np] += 1
+            done_mask = ((self._gen_cnt[act_np] >= self._maxnew_arr[act_np])
+                         | (self._seq_len_host[act_np] >= c.max_seq - 2))
+            h.produced += len(decode_slots)
+            self.total_generated += len(decode_slots)
+            h.dec_batches.append((reqs, act_np, done_mask, base))
+            if done_mask.any():
+                finished = act_np[done_mask]
+                for s in finished.tolist():
+                    self.running[s].done = True
+                    del self.running[s]
+                    self._free_slots.append(s)
+                    self._seq_len_host[s] = 0
                 # reset the DEVICE positions too: the full-arena decode reads
                 # seq_len[slot] as each slot's live window — a freed slot left
                 # at its final position would keep streaming its whole stale
                 # KV window every step (in steady state most of the arena is
                 # free, so this is the difference between O(active) and
                 # O(capacity) attention work)
-                self.seq_len[self._h2d_i64(np.fromiter(finished, dtype=np.int64))] = 0
+                self.seq_len[self._h2d_i64(finished)] = 0
 
         lt["dec_book"] += _t() - t3
         if off:
@@ -606,11 +628,19 @@ class TorchEngine:
             return 0
         if handle.ev is not None:
             handle.ev.synchronize()
+        vals = None
+        if handle.n:
+            vals = handle.slab[: handle.n].numpy()  # pinned host view, no copy
         if handle.fills:
-            vals = handle.slab[: handle.n].tolist()
             for container, key, idx in handle.fills:
                 container[key] = int(vals[idx])
         self._step_events.extend(tuple(r) for r in handle.events)
+        ev = self._step_events
+        for reqs, act_np, done_mask, base in handle.dec_batches:
+            toks = vals[base + act_np].tolist()
+            for r, t, d in zip(reqs, toks, done_mask.tolist()):
+                r.generated.append(t)
+                ev.append((r.rid, t, 1 if d else 0))
         return handle.produced
 
     def step(self, decode_burst: int = 1) -> int:
@@ -718,18 +748,22 @@ class TorchEngine:
     # addressing also removes Python's per-process hash randomization, and a
     # token compare on hit makes a 64-bit collision restore impossible
     # (vLLM's content-addressed-hashing fix for cross-request KV leakage).
-    def _prefix_lookup(self, tokens: List[int]):
+    def _prefix_lookup(self, tokens: List[int], keys: Optional[List[int]] = None):
         """Longest page-aligned cached prefix of `tokens` -> (arena_slot, plen).
         Capped at len(tokens)-1 so prefill always runs at least one token —
-        the final-chunk logits sample the first generated token."""
+        the final-chunk logits sample the first generated token.  `keys`
+        skips the O(n) hash when submit() precomputed it."""
         c = self.cfg
         page = c.prefix_cache_page
-        pmax = min(((len(tokens) - 1) // page) * page, c.prefix_cache_max, c.max_seq - 1)
-        if pmax < page:
-            return None
-        from ..kvindex.chain_keys import chain_keys
+        if keys is None:
+            pmax = min(((len(tokens) - 1) // page) * page, c.prefix_cache_max, c.max_seq - 1)
+            if pmax < page:
+                return None
+            from ..kvindex.chain_keys import chain_keys
 
-        keys = chain_keys(tokens[:pmax], page)
+            keys = chain_keys(tokens[:pmax], page)
+        if not keys:
+            return None
         for i in range(len(keys) - 1, -1, -1):
             p = (i + 1) * page
             entry = self._pc_keys.get(keys[i])
@@ -1219,16 +1253,19 @@ class TorchEngine:
         [max_slots] sampled-token tensor on device without any host sync —
         the caller reads the values later (step_finish) so the host keeps
         launching while the GPU works."""
-        maxlen = max(self._seq_len_host) + 1
+        return self._decode_launch_np(np.fromiter(slots, dtype=np.int64, count=len(slots)))
+
+    @torch.no_grad()
+    def _decode_launch_np(self, act_np: "np.ndarray") -> torch.Tensor:
+        maxlen = int(self._seq_len_host.max()) + 1
         if self.graphs and self.device.type == "cuda":
             nxt = self._decode_graphed(maxlen)
         else:
             nxt = self._decode_core(maxlen)
-        act = self._h2d_i64(np.fromiter(slots, dtype=np.int64, count=len(slots)))
+        act = self._h2d_i64(act_np)
         self._last_tok.index_copy_(0, act, nxt.index_select(0, act))
         self.seq_len.index_add_(0, act, torch.ones_like(act))
-        for s in slots:
-            self._seq_len_host[s] += 1
+        self._seq_len_host[act_np] += 1
         return nxt
 
     @torch.no_grad()

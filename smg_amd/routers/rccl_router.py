@@ -79,6 +79,7 @@ class TickGateway:
         # inside the tick (handoff instructions), so it stays synchronous
         self.pipeline = bool(pipeline) and not pd_roles
         self._prev_handle = None
+        self._submit_direct = False  # set per-tick by the pipelined path
         # PD over the plane: rank -> "prefill" | "decode".  Arrivals route
         # to PREFILL ranks (prefill_only role on the worker side); PREFILLED
         # events trigger an xGMI KV handoff to a min-load DECODE rank.
@@ -171,7 +172,13 @@ class TickGateway:
                 self.inflight[rid] = sel
                 rank = self.workers[sel].rccl_rank
                 if rank in (None, 0):
-                    self._local_pending.append((rid, max_new, toks))
+                    if self._submit_direct:
+                        # pipelined mode: routing runs while the GPU executes
+                        # this tick, so submit (incl. prefix-key hashing)
+                        # happens here, off the launch critical path
+                        self.local_engine.submit(toks, max_new, rid=rid)
+                    else:
+                        self._local_pending.append((rid, max_new, toks))
                 else:
                     self.plane.enqueue(rank, rid, max_new, toks)
 
@@ -231,8 +238,13 @@ class TickGateway:
             self._flush_local_transfers()
             self.phase_t["plane"] += time.perf_counter() - tp
         # 3) route this tick's arrivals while the GPU runs (the tree kernel
-        # uses its own HIP stream, so it doesn't queue behind the engine)
-        self._route_arrivals()
+        # uses its own HIP stream, so it doesn't queue behind the engine);
+        # local selections submit straight into the engine here
+        self._submit_direct = True
+        try:
+            self._route_arrivals()
+        finally:
+            self._submit_direct = False
         # 4) resolve tick N-1
         tl = time.perf_counter()
         prev, self._prev_handle = self._prev_handle, handle
