@@ -36,40 +36,63 @@ from smg_amd.workers.worker import Worker
 
 
 class DelayEngine:
-    """Engine stand-in with the measured per-tick cost of the GPU engine:
-    step(decode_burst) busy-costs `step_ms` regardless of batch (the GPU
-    decode is one fused full-arena pass), emits decode_burst tokens per
-    active request per step."""
+    """Engine stand-in with the measured per-tick cost of the GPU engine.
 
-    def __init__(self, step_ms: float, max_new_default: int = 32):
+    Mirrors the pipelined TorchEngine API: step_launch busy-costs
+    `launch_ms` of HOST time (the measured launch phase: graph replays +
+    submits + bookkeeping) and schedules the "GPU" to be done step_ms after
+    the device frees up; step_finish busy-waits to that ready time (the
+    CUDA-event sync).  The pipelined gateway therefore overlaps its routing
+    phase with the simulated device exactly as on hardware."""
+
+    def __init__(self, step_ms: float, launch_ms: float = 0.0, max_new_default: int = 32):
         self.step_ms = step_ms
+        self.launch_ms = launch_ms
         self.active = {}  # rid -> [produced, max_new]
         self.events = []
         self.busy_s = 0.0
+        self._gpu_free = 0.0
 
     def submit(self, prompt, max_new, rid=None):
         self.active[rid] = [0, max_new]
         return rid
 
-    def step(self, decode_burst: int = 2):
-        t0 = time.perf_counter()
-        # simulate the fused decode pass cost
-        target = t0 + self.step_ms / 1e3
-        while time.perf_counter() < target:
-            pass  # busy-wait: sleep() oversleeps at ms scale and hides CPU contention
-        done = []
+    def _advance(self, decode_burst: int):
+        evs, done = [], []
         for rid, st in self.active.items():
             for _ in range(decode_burst):
                 st[0] += 1
                 fin = st[0] >= st[1]
-                self.events.append((rid, 1000 + st[0], 1 if fin else 0))
+                evs.append((rid, 1000 + st[0], 1 if fin else 0))
                 if fin:
                     done.append(rid)
                     break
         for rid in done:
             del self.active[rid]
+        return evs
+
+    def step_launch(self, decode_burst: int = 1):
+        t0 = time.perf_counter()
+        target = t0 + self.launch_ms / 1e3
+        while time.perf_counter() < target:
+            pass  # busy-wait: sleep() oversleeps at ms scale and hides CPU contention
+        ready = max(self._gpu_free, t0) + self.step_ms / 1e3
+        self._gpu_free = ready
+        evs = self._advance(decode_burst)
         self.busy_s += time.perf_counter() - t0
-        return len(self.events)
+        return (ready, evs)
+
+    def step_finish(self, handle):
+        ready, evs = handle
+        t0 = time.perf_counter()
+        while time.perf_counter() < ready:
+            pass  # the CUDA-event sync: waits out the remaining device time
+        self.events.extend(evs)
+        self.busy_s += time.perf_counter() - t0
+        return len(evs)
+
+    def step(self, decode_burst: int = 2):
+        return self.step_finish(self.step_launch(decode_burst))
 
     def drain_events(self):
         out, self.events = self.events, []
@@ -79,10 +102,13 @@ class DelayEngine:
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--seconds", type=float, default=6.0)
-    ap.add_argument("--step-ms", type=float, default=8.0)
+    ap.add_argument("--step-ms", type=float, default=13.0,
+                    help="simulated GPU device time per tick (r02 profile: ~13 ms)")
+    ap.add_argument("--launch-ms", type=float, default=4.0,
+                    help="host launch-phase cost per tick (r02 steady state)")
     ap.add_argument("--concurrency", type=int, default=64, help="per-rank in-flight")
     ap.add_argument("--max-new", type=int, default=32)
-    ap.add_argument("--decode-burst", type=int, default=2)
+    ap.add_argument("--decode-burst", type=int, default=1)
     ap.add_argument("--prompt-len", type=int, default=576)
     args = ap.parse_args()
 
@@ -90,7 +116,7 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", "1"))
     dist.init_process_group("gloo")
     pcfg = PlaneConfig(max_prompt=args.prompt_len + 8, device="cpu")
-    eng = DelayEngine(args.step_ms)
+    eng = DelayEngine(args.step_ms, launch_ms=args.launch_ms)
 
     if rank == 0:
         plane = GatewayPlane(pcfg, list(range(1, world))) if world > 1 else None
