@@ -230,16 +230,20 @@ class MistralParser(ToolParser):
     """`[TOOL_CALLS] [{...}, ...]` (mistral.rs:108)."""
 
     name = "mistral"
-    BOT = "[TOOL_CALLS] ["
+    BOT = "[TOOL_CALLS]"
 
     def has_tool_markers(self, text):
         return "[TOOL_CALLS]" in text
 
     def parse(self, text, tools=None):
+        # the wire format is `[TOOL_CALLS][{...}]`; some templates emit a
+        # space before the array — accept both (mistral.rs strips it too)
         pos = text.find(self.BOT)
         if pos < 0:
             return text, []
-        arr_start = pos + len(self.BOT) - 1
+        arr_start = text.find("[", pos + len(self.BOT))
+        if arr_start < 0:
+            return text, []
         depth = 0
         in_str = False
         esc = False

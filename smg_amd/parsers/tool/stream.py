@@ -255,6 +255,27 @@ class StreamingToolParser:
                     break
         return best
 
+    END_MARKERS = {
+        "kimik2": ["<|tool_call_end|>", "<|tool_calls_section_end|>"],
+    }
+
+    def _open_span_holdback(self) -> int:
+        """Chars to withhold from an OPEN raw-argument span (ve=None) so a
+        partially-arrived end marker never leaks into the streamed argument
+        deltas (e.g. kimik2's `<|tool_call_end|` prefix)."""
+        ends = list(self.END_MARKERS.get(self.parser.name, []))
+        end_tag = getattr(self.parser, "end", None)
+        if end_tag:
+            ends.append(end_tag)
+        best = 0
+        for m in ends:
+            lim = min(len(m) - 1, len(self.buffer))
+            for k in range(lim, 0, -1):
+                if self.buffer.endswith(m[:k]):
+                    best = max(best, k)
+                    break
+        return best
+
     def _ensure(self, i: int) -> None:
         while len(self.emitted_args) <= i:
             self.emitted_args.append(0)
@@ -287,7 +308,8 @@ class StreamingToolParser:
             if not self.named[i] and name:
                 self.named[i] = True
                 events.append(("tool_name", i, name))
-            avail = (ve if ve is not None else len(self.buffer)) - vs
+            avail = (ve if ve is not None
+                     else len(self.buffer) - self._open_span_holdback()) - vs
             prev = self.emitted_args[i]
             if self.named[i] and avail > prev:
                 events.append(("tool_args", i, self.buffer[vs + prev: vs + avail]))
