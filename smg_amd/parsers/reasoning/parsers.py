@@ -66,12 +66,18 @@ class ReasoningParser:
 
     # ---- streaming -------------------------------------------------------
     def _partial_marker_len(self, s: str) -> int:
-        """Length of a suffix of `s` that is a proper prefix of either marker."""
+        """Length of the LONGEST suffix of `s` that is a proper prefix of
+        either marker.  Must take the max over both markers: when they share
+        a prefix (kimi_k3's `<|open|>…` / `<|close|>…`), returning the
+        start-marker's short match let end-marker bytes leak into the
+        streamed reasoning and the full end marker was then never found."""
+        best = 0
         for marker in (self.think_start, self.think_end):
             for k in range(min(len(marker) - 1, len(s)), 0, -1):
                 if s.endswith(marker[:k]):
-                    return k
-        return 0
+                    best = max(best, k)
+                    break
+        return best
 
     def parse_streaming(self, delta: str) -> Tuple[str, str]:
         self._buffer += delta

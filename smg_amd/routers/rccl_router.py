@@ -556,7 +556,11 @@ class RcclRouter(Router):
         from ..engine.torch_engine import TorchEngine, TorchEngineConfig
 
         if torch.cuda.is_available():
-            return TorchEngine(TorchEngineConfig(), device="cuda:0", graphs=True)
+            cfg = TorchEngineConfig()
+            # serving default sized for 288 GB HBM3E: 256 concurrent slots
+            # (the 64-slot dataclass default is the CI/tiny-memory shape)
+            cfg.max_slots = 256
+            return TorchEngine(cfg, device="cuda:0", graphs=True)
         return TorchEngine(TorchEngineConfig.tiny(), device="cpu")
 
     # ---- tick thread ------------------------------------------------------
