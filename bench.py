@@ -66,7 +66,12 @@ def parse_args():
     p.add_argument("--pd", action="store_true",
                    help="PD disaggregation over the plane: odd ranks prefill, even ranks "
                         "(incl. the gateway) decode; KV hands off over xGMI p2p")
-    p.add_argument("--no-graphs", action="store_true", help="disable hipGraph decode capture")
+    p.add_argument("--no-graphs", action="store_true", help="disable hipGraph capture")
+    p.add_argument("--graphs", action="store_true",
+                   help="enable hipGraph capture (default OFF since the pipelined tick: "
+                        "same-box steps-60 A/B 1464.5 no-graphs vs 1395.9 graphs — eager "
+                        "launches fully overlap and capture/replay overhead nets negative; "
+                        "PD mode still defaults ON, its synchronous tick benefits)")
     p.add_argument("--prefill-group", type=int, default=32,
                    help="requests sharing one batched prefill forward")
     p.add_argument("--arrival-cap", type=int, default=128,
@@ -132,7 +137,8 @@ def pd_role(rank: int, args) -> str:
 
 
 def worker_main(rank: int, world: int, args, device: str, backend: str):
-    eng = TorchEngine(engine_config(args), device=device, graphs=device.startswith("cuda") and not args.no_graphs)
+    use_graphs = device.startswith("cuda") and not args.no_graphs and (args.graphs or args.pd)
+    eng = TorchEngine(engine_config(args), device=device, graphs=use_graphs)
     plane = WorkerPlane(PlaneConfig(max_prompt=args.prefix_len + args.suffix_len + 8, device=device if backend == "nccl" else "cpu"))
     elapsed = run_worker_loop(eng, plane, decode_burst=args.decode_burst, role=pd_role(rank, args))
     el = torch.tensor([elapsed], device=device if backend == "nccl" else "cpu")
@@ -144,7 +150,8 @@ def gateway_main(rank: int, world: int, args, device: str, backend: str):
     TickGateway) under synthetic load — the same tick loop `smg launch
     --connection-mode rccl` serves HTTP from."""
     use_gpu = device.startswith("cuda")
-    eng = TorchEngine(engine_config(args), device=device, graphs=use_gpu and not args.no_graphs)
+    use_graphs = use_gpu and not args.no_graphs and (args.graphs or args.pd)
+    eng = TorchEngine(engine_config(args), device=device, graphs=use_graphs)
     remote_ranks = list(range(1, world))
     plane = (
         GatewayPlane(

@@ -206,6 +206,7 @@ class TorchEngine:
         self._pf_stage_i = 0
         self._pfg_cache: Dict[tuple, tuple] = {}  # (B,L,start0,fresh) -> captured graph
         self._pfg_bad: set = set()  # shapes whose capture failed (stay eager)
+        self._pfg_seen: Dict[tuple, int] = {}  # shape repeat counts (capture gate)
         self.pfg_hits = 0
         self.pfg_eager = 0
         # host-time breakdown of step_launch (pf_book/dec_book include pf/decode)
@@ -925,7 +926,12 @@ class TorchEngine:
         return nxt  # device tensor: the caller defers (or performs) the D2H read
 
     # ---- prefill hipGraph cache -------------------------------------------
-    PFG_CAP = 8  # distinct (B, L, start0, fresh) shapes kept captured
+    PFG_CAP = 16      # distinct (B, L, start0, fresh) shapes kept captured
+    PFG_MIN_SEEN = 3  # capture a shape only once it repeats: a one-off group
+    #                   shape captured mid-run costs ~60 ms of full-stream
+    #                   syncs inside the timed region (measured: steps-12
+    #                   bench 1189 with eager capture vs 1274 graphs-off);
+    #                   hot shapes repeat during warmup and capture there
 
     def _pfg_enabled(self, B: int, fresh: bool, uniform_start: bool) -> bool:
         import os as _os
@@ -954,6 +960,10 @@ class TorchEngine:
         entry = self._pfg_cache.get(key)
         if entry is None:
             if key in self._pfg_bad or len(self._pfg_cache) >= self.PFG_CAP:
+                return None
+            seen = self._pfg_seen.get(key, 0) + 1
+            self._pfg_seen[key] = seen
+            if seen < self.PFG_MIN_SEEN:
                 return None
             try:
                 entry = self._pfg_capture(key, items, B, L, t_max, fresh, uniform_start, start0)
@@ -1257,8 +1267,11 @@ class TorchEngine:
 
     @torch.no_grad()
     def _decode_launch_np(self, act_np: "np.ndarray") -> torch.Tensor:
+        import os as _os
+
         maxlen = int(self._seq_len_host.max()) + 1
-        if self.graphs and self.device.type == "cuda":
+        if (self.graphs and self.device.type == "cuda"
+                and not _os.environ.get("SMG_NO_DECODE_GRAPH")):
             nxt = self._decode_graphed(maxlen)
         else:
             nxt = self._decode_core(maxlen)
