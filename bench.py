@@ -48,8 +48,8 @@ BASELINE_REQ_S = None
 def parse_args():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
-    p.add_argument("--steps", type=int, default=8)
-    p.add_argument("--warmup", type=int, default=6)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=10)
     p.add_argument("--reqs-per-step", type=int, default=32)
     p.add_argument("--concurrency", type=int, default=512, help="in-flight requests per worker")
     p.add_argument("--prefix-pool", type=int, default=8)
