@@ -105,6 +105,10 @@ class GatewayPlane:
         self._pending: Dict[int, List[Tuple[int, int, List[int]]]] = {w: [] for w in worker_ranks}
         self._pending_tr: Dict[int, List[Tuple[int, ...]]] = {w: [] for w in worker_ranks}
         self._works = []  # outstanding isend/irecv handles for this tick
+        # NCCL Work.wait() only adds a STREAM dependency — before the host
+        # rewrites the pinned staging, the previous tick's H2D wire copy must
+        # have actually completed (event host-sync; gloo/cpu path skips it)
+        self._h2d_ev = torch.cuda.Event() if self._gpu else None
 
     def enqueue(self, worker_rank: int, rid: int, max_new: int, prompt: List[int]) -> None:
         self._pending[worker_rank].append((rid, max_new, prompt[-self.cfg.max_prompt:]))
@@ -125,6 +129,8 @@ class GatewayPlane:
         for work in self._works:
             work.wait()
         self._works = []
+        if self._h2d_ev is not None:
+            self._h2d_ev.synchronize()  # staging safe to rewrite (see __init__)
         for w in self.worker_ranks:
             pend = self._pending[w][: cfg.max_reqs_per_tick]
             self._pending[w] = self._pending[w][len(pend):]
@@ -145,6 +151,8 @@ class GatewayPlane:
                 rows[base: base + TR_INTS] = tr
             if self._gpu:
                 self._send[w].copy_(self._send_h[w][0], non_blocking=True)
+        if self._h2d_ev is not None:
+            self._h2d_ev.record()
         ops = []
         for w in self.worker_ranks:
             ops.append(dist.P2POp(dist.isend, self._send[w], w))
@@ -230,6 +238,7 @@ class WorkerPlane:
         self._send_work = None
         self.barrier_requested = False
         self.transfers: List[Tuple[int, ...]] = []
+        self._h2d_ev = torch.cuda.Event() if self._gpu else None  # see GatewayPlane
 
     def tick(self, events: List[Tuple[int, int, int]]) -> Tuple[List[Tuple[int, int, List[int]]], bool]:
         """One lockstep exchange: sends `events` [(rid, token, flags)], receives
@@ -243,6 +252,8 @@ class WorkerPlane:
         if self._send_work is not None:
             self._send_work.wait()
             self._send_work = None
+        if self._h2d_ev is not None:
+            self._h2d_ev.synchronize()  # last tick's H2D truly done: staging reusable
         _, ev = self._send_h
         n = min(len(events), cfg.max_events_per_tick)
         ev[0] = n
@@ -250,6 +261,8 @@ class WorkerPlane:
             ev[1: 1 + n * 3] = np.asarray(events[:n], dtype=np.int32).reshape(-1)
         if self._gpu:
             self._send.copy_(self._send_h[0], non_blocking=True)
+        if self._h2d_ev is not None:
+            self._h2d_ev.record()
         ops = [
             dist.P2POp(dist.irecv, self._recv, self.gateway_rank),
             dist.P2POp(dist.isend, self._send, self.gateway_rank),
