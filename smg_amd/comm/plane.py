@@ -44,11 +44,15 @@ DONE = 1
 ABORT = 2
 PREFILLED = 4  # PD prefill leg finished; KV parked for transfer
 
-# PD KV-handoff instructions carried in the request tensor (both endpoint
+# PD/EPD handoff instructions carried in the request tensor (both endpoint
 # ranks receive the same ordered list, so their send/recv pairs match by
 # construction — deadlock-free like the main lockstep exchange)
-KV_SEND = 1
+KV_SEND = 1   # PD: prefill rank ships a parked KV window
 KV_RECV = 2
+EMB_SEND = 3  # EPD: encode rank ships [n_tokens, d_model] vision embeddings
+EMB_RECV = 4
+PIX_SEND = 5  # EPD: gateway ships [C, H, W] u8 pixels to the encode rank
+PIX_RECV = 6  # (fields: n_tokens=C, first_tok=H, max_new=W)
 TR_INTS = 6  # rid, peer_rank, direction, n_tokens, first_tok, max_new
 
 
@@ -209,12 +213,28 @@ def execute_transfers(engine, transfers) -> int:
             t, plen, _ft = engine.export_kv(rid)
             payload = t.view(torch.uint8) if t.dtype == torch.float8_e4m3fn else t
             dist.send(payload.contiguous(), dst=peer, tag=tag)
-        else:
+        elif direction == KV_RECV:
             shape = engine.kv_transfer_shape(n_tokens)
             buf = torch.empty(shape, dtype=engine.kv.dtype, device=engine.kv.device)
             payload = buf.view(torch.uint8) if buf.dtype == torch.float8_e4m3fn else buf
             dist.recv(payload, src=peer, tag=tag)
             engine.import_kv(rid, buf, n_tokens, first_tok, max_new)
+        elif direction == EMB_SEND:
+            emb = engine.export_embed(rid)  # [n_tokens, d_model]
+            dist.send(emb.contiguous(), dst=peer, tag=tag)
+        elif direction == EMB_RECV:
+            buf = torch.empty(n_tokens, engine.cfg.d_model, dtype=engine.dtype,
+                              device=engine.device)
+            dist.recv(buf, src=peer, tag=tag)
+            engine.accept_embed(rid, buf)
+        elif direction == PIX_SEND:
+            px = engine.export_pixels(rid)  # [C, H, W] u8
+            dist.send(px.contiguous(), dst=peer, tag=tag)
+        elif direction == PIX_RECV:
+            buf = torch.empty(n_tokens, first_tok, max_new, dtype=torch.uint8,
+                              device=getattr(engine, "device", "cpu"))
+            dist.recv(buf, src=peer, tag=tag)
+            engine.accept_pixels(rid, buf)
         done += 1
     return done
 

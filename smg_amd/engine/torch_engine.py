@@ -138,7 +138,7 @@ def _rms(x, weight):
 
 class _Request:
     __slots__ = ("rid", "tokens", "max_new", "slot", "prefilled", "generated", "done",
-                 "prefill_only", "pc_keys")
+                 "prefill_only", "pc_keys", "mm_embed")
 
     def __init__(self, rid, tokens, max_new, prefill_only=False):
         self.rid = rid
@@ -150,6 +150,7 @@ class _Request:
         self.done = False
         self.prefill_only = prefill_only
         self.pc_keys = None  # prefix-cache chain keys, precomputed at submit
+        self.mm_embed = None  # [E, d_model] EPD vision embeddings (seq prefix)
 
 
 class TorchEngine:
@@ -198,6 +199,7 @@ class TorchEngine:
         self._requests: Dict[str, _Request] = {}
         self._rid_counter = 0
         self._parked: Dict[str, tuple] = {}  # rid -> (slot, req, first_tok) awaiting export_kv
+        self._pending_embeds: Dict = {}  # rid -> [E, d_model] EPD embeddings (accept_embed)
         self.total_generated = 0
         self._step_events: List[tuple] = []  # (rid, token, done) since last drain
         self._slabs: List[Optional[torch.Tensor]] = [None] * 4  # pinned D2H slabs
@@ -345,13 +347,19 @@ class TorchEngine:
 
     # ---- API -------------------------------------------------------------
     def submit(self, tokens: List[int], max_new_tokens: int, rid: Optional[str] = None,
-               prefill_only: bool = False) -> str:
+               prefill_only: bool = False, mm_embed=None) -> str:
         """`prefill_only` is the PD prefill leg: the request prefills, samples
         its first token (emitted with the PREFILLED flag, value 4) and PARKS —
         the slot stays allocated until export_kv() hands its KV off to the
         decode rank (PD over the rccl plane; reference PD delegates this
         transfer to engine-side Mooncake/NIXL, here the engine is ours and
-        the handoff is an xGMI p2p send)."""
+        the handoff is an xGMI p2p send).
+
+        `mm_embed` is the EPD embedding leg: an [E, d_model] tensor of
+        vision-tower embeddings that occupy sequence positions 0..E-1 in
+        front of the prompt tokens (the reference ships these engine-side
+        via Mooncake after the encode fleet runs; here they arrive over the
+        xGMI plane — comm/plane.py EMB transfer directions)."""
         if rid is None:
             self._rid_counter += 1
             rid = f"req-{self._rid_counter}"
@@ -361,11 +369,23 @@ class TorchEngine:
         # the prefix cache (the per-token Python list work was milliseconds per
         # serving tick at 32 admissions x ~1.7K tokens)
         toks = np.asarray(tokens, dtype=np.int64) % c.vocab_size
-        if len(toks) > c.max_seq - 2:
-            toks = toks[-(c.max_seq - 2):]  # keep the prompt tail
-        max_new_tokens = max(1, min(max_new_tokens, c.max_seq - 1 - len(toks)))
+        n_emb = 0
+        if mm_embed is not None:
+            mm_embed = torch.as_tensor(mm_embed).to(self.device, self.dtype)
+            if mm_embed.dim() != 2 or mm_embed.shape[1] != c.d_model:
+                raise ValueError(f"mm_embed must be [E, {c.d_model}]")
+            n_emb = mm_embed.shape[0]
+        if len(toks) + n_emb > c.max_seq - 2:
+            keep = max(1, c.max_seq - 2 - n_emb)
+            toks = toks[-keep:]  # keep the prompt tail
+        max_new_tokens = max(1, min(max_new_tokens, c.max_seq - 1 - len(toks) - n_emb))
         req = _Request(rid, toks, max_new_tokens, prefill_only=prefill_only)
-        if self._pc_arena is not None:
+        if mm_embed is None:
+            # EPD decode leg: embeddings that arrived over the plane
+            # (accept_embed) pair with the request by rid
+            mm_embed = self._pending_embeds.pop(rid, None)
+        req.mm_embed = mm_embed
+        if mm_embed is None and self._pc_arena is not None:
             # hash here: in the pipelined gateway, submit() runs in the
             # routing phase (overlapped with the GPU tick) while the prefix
             # lookup runs in the launch critical path
@@ -422,6 +442,16 @@ class TorchEngine:
     def kv_transfer_shape(self, plen: int):
         c = self.cfg
         return (c.n_layers, 2, c.kv_heads, plen, c.head_dim)
+
+    # ---- EPD: vision embeddings over the plane ----------------------------
+    def accept_embed(self, rid, emb: torch.Tensor) -> None:
+        """EMB_RECV landing: [E, d_model] embeddings stash until the matching
+        request submit()s (plane transfers execute before submits each tick)."""
+        self._pending_embeds[rid] = emb.to(self.device, self.dtype)
+        # a request that arrived first picks the embedding up here instead
+        req = self._requests.get(rid)
+        if req is not None and req.prefilled == 0 and req.mm_embed is None:
+            req.mm_embed = self._pending_embeds.pop(rid)
 
     def finished(self, rid: str) -> bool:
         r = self._requests.get(rid)
@@ -488,7 +518,7 @@ class TorchEngine:
         # each, x32 hits/tick in the bench.
         restore: Dict[int, list] = {}
         for slot, req in self.running.items():
-            if req.prefilled == 0 and self._pc_arena is not None:
+            if req.prefilled == 0 and self._pc_arena is not None and req.mm_embed is None:
                 hit = self._prefix_lookup(req.tokens, keys=req.pc_keys)
                 if hit is not None:
                     pslot, plen = hit
@@ -512,7 +542,32 @@ class TorchEngine:
                 self._last_tok[slots_t] = lasts_t
         t1 = _t()
         lt["restore"] += t1 - t0
-        pending = [(s, r) for s, r in self.running.items() if r.prefilled < len(r.tokens)]
+        # EPD embedding-prefixed requests prefill one at a time (exception
+        # path: [E, D] vision embeddings occupy positions 0..E-1, so they
+        # can't share the token-batched forward)
+        emb_pending = [(s, r) for s, r in self.running.items()
+                       if r.prefilled == 0 and r.mm_embed is not None]
+        for slot, req in emb_pending[:4]:
+            nxt_e = self._prefill_embedded(slot, req)
+            samples.append((nxt_e, off))
+            req.prefilled = len(req.tokens)
+            req.generated.append(-1)
+            h.fills.append((req.generated, 0, off))
+            self._gen_cnt[slot] = 1
+            h.produced += 1
+            self.total_generated += 1
+            if len(req.generated) >= req.max_new or self._seq_len_host[slot] >= c.max_seq - 2:
+                req.done = True
+                del self.running[slot]
+                self._free_slots.append(slot)
+                self._seq_len_host[slot] = 0
+                self.seq_len[slot] = 0
+            row = [req.rid, None, 1 if req.done else 0]
+            h.fills.append((row, 1, off))
+            h.events.append(row)
+            off += 1
+        pending = [(s, r) for s, r in self.running.items()
+                   if r.prefilled < len(r.tokens) and r.mm_embed is None]
         if pending:
             pending.sort(key=lambda sr: len(sr[1].tokens) - sr[1].prefilled)
             min_rem = len(pending[0][1].tokens) - pending[0][1].prefilled
@@ -1017,13 +1072,29 @@ class TorchEngine:
         return (g, t_buf, slots_buf, starts_buf, nxt_out, slots_stage, starts_stage)
 
     @torch.no_grad()
+    def _prefill_embedded(self, slot: int, req) -> torch.Tensor:
+        """EPD prefill: [E, D] vision embeddings at positions 0..E-1 followed
+        by the prompt tokens; returns the sampled first token ([1], device)."""
+        t = torch.from_numpy(np.ascontiguousarray(req.tokens, dtype=np.int64))
+        if self.device.type == "cuda":
+            t = t.to(self.device, non_blocking=True)
+        h0 = torch.cat([req.mm_embed, self.embed[t]], 0).unsqueeze(0)  # [1, E+n, D]
+        L = h0.shape[1]
+        slots = self._h2d_i64(np.asarray([slot], dtype=np.int64))
+        starts = self._h2d_i64(np.asarray([0], dtype=np.int64))
+        nxt = self._prefill_forward(slots, starts, None, 1, L, L, True, False, 0, h0=h0)
+        self._seq_len_host[slot] = L
+        return nxt
+
+    @torch.no_grad()
     def _prefill_forward(self, slots, starts, t, B: int, L: int, t_max: int,
-                         fresh: bool, uniform_start: bool, start0: int):
+                         fresh: bool, uniform_start: bool, start0: int, h0=None):
         """Device-only prefill forward (hipGraph-capturable: every input is a
         tensor whose CONTENT may change between replays; B/L/start0 are
-        shape-defining and key the graph cache)."""
+        shape-defining and key the graph cache).  `h0` bypasses the token
+        embedding lookup (EPD embedding-prefixed requests)."""
         c = self.cfg
-        h = self.embed[t]  # [B, L, D]
+        h = h0 if h0 is not None else self.embed[t]  # [B, L, D]
         pos = starts.unsqueeze(1) + torch.arange(L, device=self.device)  # [B, L]
         freqs = self.freqs_cis[pos].unsqueeze(1)  # [B, 1, L, D/2]
         if not fresh:
