@@ -138,7 +138,7 @@ def _rms(x, weight):
 
 class _Request:
     __slots__ = ("rid", "tokens", "max_new", "slot", "prefilled", "generated", "done",
-                 "prefill_only", "pc_keys", "mm_embed")
+                 "prefill_only", "pc_keys", "mm_embed", "await_embed")
 
     def __init__(self, rid, tokens, max_new, prefill_only=False):
         self.rid = rid
@@ -151,6 +151,7 @@ class _Request:
         self.prefill_only = prefill_only
         self.pc_keys = None  # prefix-cache chain keys, precomputed at submit
         self.mm_embed = None  # [E, d_model] EPD vision embeddings (seq prefix)
+        self.await_embed = False  # EPD: hold prefill until accept_embed pairs
 
 
 class TorchEngine:
@@ -347,7 +348,7 @@ class TorchEngine:
 
     # ---- API -------------------------------------------------------------
     def submit(self, tokens: List[int], max_new_tokens: int, rid: Optional[str] = None,
-               prefill_only: bool = False, mm_embed=None) -> str:
+               prefill_only: bool = False, mm_embed=None, await_embed: bool = False) -> str:
         """`prefill_only` is the PD prefill leg: the request prefills, samples
         its first token (emitted with the PREFILLED flag, value 4) and PARKS —
         the slot stays allocated until export_kv() hands its KV off to the
@@ -385,7 +386,10 @@ class TorchEngine:
             # (accept_embed) pair with the request by rid
             mm_embed = self._pending_embeds.pop(rid, None)
         req.mm_embed = mm_embed
-        if mm_embed is None and self._pc_arena is not None:
+        # hold prefill until the embedding transfer lands (pipelined gateway:
+        # the local EMB_RECV may flush after this submit)
+        req.await_embed = bool(await_embed) and mm_embed is None
+        if mm_embed is None and not req.await_embed and self._pc_arena is not None:
             # hash here: in the pipelined gateway, submit() runs in the
             # routing phase (overlapped with the GPU tick) while the prefix
             # lookup runs in the launch critical path
@@ -452,6 +456,7 @@ class TorchEngine:
         req = self._requests.get(rid)
         if req is not None and req.prefilled == 0 and req.mm_embed is None:
             req.mm_embed = self._pending_embeds.pop(rid)
+            req.await_embed = False
 
     def finished(self, rid: str) -> bool:
         r = self._requests.get(rid)
@@ -518,7 +523,8 @@ class TorchEngine:
         # each, x32 hits/tick in the bench.
         restore: Dict[int, list] = {}
         for slot, req in self.running.items():
-            if req.prefilled == 0 and self._pc_arena is not None and req.mm_embed is None:
+            if (req.prefilled == 0 and self._pc_arena is not None
+                    and req.mm_embed is None and not req.await_embed):
                 hit = self._prefix_lookup(req.tokens, keys=req.pc_keys)
                 if hit is not None:
                     pslot, plen = hit
@@ -567,7 +573,8 @@ class TorchEngine:
             h.events.append(row)
             off += 1
         pending = [(s, r) for s, r in self.running.items()
-                   if r.prefilled < len(r.tokens) and r.mm_embed is None]
+                   if r.prefilled < len(r.tokens) and r.mm_embed is None
+                   and not r.await_embed]
         if pending:
             pending.sort(key=lambda sr: len(sr[1].tokens) - sr[1].prefilled)
             min_rem = len(pending[0][1].tokens) - pending[0][1].prefilled

@@ -65,6 +65,10 @@ class EncodeWorker:
     def export_embed(self, rid) -> torch.Tensor:
         return self._embeds.pop(rid)
 
+    def export_pixels(self, rid) -> torch.Tensor:
+        """Gateway-side use: outgoing pixel stash (PIX_SEND)."""
+        return self._pixels.pop(rid)
+
     def embed_len(self) -> int:
         return self.encoder.n_embed
 

@@ -34,8 +34,12 @@ from typing import Any, Callable, Dict, List, Optional, Sequence, Tuple
 
 from ..comm.plane import (
     DONE,
+    EMB_RECV,
+    EMB_SEND,
     KV_RECV,
     KV_SEND,
+    PIX_RECV,
+    PIX_SEND,
     PREFILLED,
     GatewayPlane,
     PlaneConfig,
@@ -76,19 +80,32 @@ class TickGateway:
     ):
         self.metrics = metrics
         # pipelined ticks defer token reads one tick; PD needs token values
-        # inside the tick (handoff instructions), so it stays synchronous
-        self.pipeline = bool(pipeline) and not pd_roles
+        # inside the tick (handoff instructions), so it stays synchronous.
+        # EPD (encode roles only, no prefill split) pipelines fine: its
+        # transfer scheduling is tick-phased, not token-event-driven.
+        has_prefill = bool(pd_roles) and any(r == "prefill" for r in pd_roles.values())
+        self.pipeline = bool(pipeline) and not has_prefill
         self._prev_handle = None
         self._submit_direct = False  # set per-tick by the pipelined path
-        # PD over the plane: rank -> "prefill" | "decode".  Arrivals route
-        # to PREFILL ranks (prefill_only role on the worker side); PREFILLED
-        # events trigger an xGMI KV handoff to a min-load DECODE rank.
-        self.pd_roles = pd_roles
-        if pd_roles:
+        # PD/EPD over the plane: rank -> "prefill" | "decode" | "encode".
+        # Arrivals route to PREFILL ranks when a prefill fleet exists
+        # (prefill_only role on the worker side); PREFILLED events trigger an
+        # xGMI KV handoff to a min-load DECODE rank.  ENCODE ranks run the
+        # vision tower: pixel-carrying arrivals ship pixels there first, the
+        # embeddings hop rank-to-rank next tick, then the request dispatches.
+        self.pd_roles = pd_roles if has_prefill else None
+        self.epd_roles = pd_roles if (pd_roles and not has_prefill) else None
+        roles = pd_roles or {}
+        self.encode_idxs = [i for i, w in enumerate(workers)
+                            if roles.get(w.rccl_rank) == "encode"]
+        if self.pd_roles:
             self.prefill_idxs = [i for i, w in enumerate(workers)
-                                 if pd_roles.get(w.rccl_rank) == "prefill"]
-            self.decode_idxs = [i for i, w in enumerate(workers)
-                                if pd_roles.get(w.rccl_rank, "decode") == "decode"]
+                                 if roles.get(w.rccl_rank) == "prefill"]
+        self.decode_idxs = [i for i, w in enumerate(workers)
+                            if roles.get(w.rccl_rank, "decode") == "decode"]
+        self._pending_mm: deque = deque()  # (rid, tokens, max_new, pixels)
+        self._epd_phase2: List[tuple] = []  # (rid, toks, max_new, e_idx, d_idx)
+        self._local_encoder = None  # lazy EncodeWorker (world-1 / local encode)
         self._req_meta: Dict[int, Tuple[int, int]] = {}  # rid -> (n_tokens, max_new)
         self._pd_plen: Dict[int, int] = {}  # rid -> prefilled length (PLEN_INFO)
         self._local_transfers: List[Tuple[int, ...]] = []
@@ -110,15 +127,131 @@ class TickGateway:
                         "launch": 0.0, "finish": 0.0, "ticks": 0}
 
     # ---- submission (any thread) -----------------------------------------
-    def submit(self, tokens: List[int], max_new: int, rid: Optional[int] = None) -> int:
+    def submit(self, tokens: List[int], max_new: int, rid: Optional[int] = None,
+               pixels=None) -> int:
+        """`pixels` ([3, H, W] uint8) marks an EPD multimodal request: the
+        image ships to an encode rank, its [E, d_model] embeddings hop
+        rank-to-rank into the chosen decode rank's prefill, then the text
+        request dispatches (reference: grpc EncodeStage + engine-side
+        Mooncake; here every hop is the xGMI plane)."""
         if rid is None:
             rid = next(self._rid) & 0x7FFFFFFF
         if self.pd_roles:
             self._req_meta[rid] = (len(tokens), max_new)
+        toks = np.asarray(tokens, dtype=np.int64)
+        if pixels is not None:
+            self._pending_mm.append((rid, toks, max_new, pixels))
+            return rid
         # numpy end-to-end: no-copy for ndarray submitters (bench/engine), one
         # conversion for list submitters (HTTP path)
-        self._pending.append((rid, np.asarray(tokens, dtype=np.int64), max_new))
+        self._pending.append((rid, toks, max_new))
         return rid
+
+    class _LocalMMEndpoint:
+        """engine + encoder composite so the gateway's ORDERED local transfer
+        list (KV, EMB and PIX entries interleaved) executes as one sequence —
+        splitting it per-target could reorder against a peer's single list
+        and deadlock the pairing."""
+
+        def __init__(self, engine, encoder):
+            self.engine, self.encoder = engine, encoder
+            self.cfg = engine.cfg
+            self.dtype = engine.dtype
+            self.device = engine.device
+            self.kv = engine.kv
+
+        def export_kv(self, rid):
+            return self.engine.export_kv(rid)
+
+        def import_kv(self, *a, **k):
+            return self.engine.import_kv(*a, **k)
+
+        def kv_transfer_shape(self, n):
+            return self.engine.kv_transfer_shape(n)
+
+        def accept_embed(self, rid, t):
+            self.engine.accept_embed(rid, t)
+
+        def export_embed(self, rid):
+            return self.encoder.export_embed(rid)
+
+        def export_pixels(self, rid):
+            return self.encoder.export_pixels(rid)
+
+        def accept_pixels(self, rid, t):
+            self.encoder.accept_pixels(rid, t)
+
+    def _encoder(self):
+        if self._local_encoder is None:
+            import torch
+
+            from ..multimodal.encoder import EncodeWorker, ToyVisionEncoder
+
+            eng = self.local_engine
+            d_model = eng.cfg.d_model
+            self._local_encoder = EncodeWorker(ToyVisionEncoder(
+                d_model, image_size=64, patch=16,
+                device=str(getattr(eng, "device", "cpu")),
+                dtype=getattr(eng, "dtype", torch.float32)))
+        return self._local_encoder
+
+    def _process_mm_arrivals(self, cap: int = 8) -> None:
+        """Phase 1 of the EPD hop: ship pixels to the encode rank (or encode
+        locally when none exists); the embedding transfer + text dispatch
+        happen next tick (_process_epd_phase2)."""
+        n = 0
+        while self._pending_mm and n < cap:
+            rid, toks, max_new, pixels = self._pending_mm.popleft()
+            n += 1
+            # decode target via min-load over the decode pool
+            pool = self.decode_idxs or list(range(len(self.workers)))
+            d_idx = min(pool, key=lambda i: self.workers[i].active_requests)
+            if self.encode_idxs:
+                e_idx = min(self.encode_idxs, key=lambda i: self.workers[i].processed_requests)
+                e_rank = self.workers[e_idx].rccl_rank
+                C, H, W = pixels.shape
+                self._encoder()._pixels[rid] = pixels  # outgoing stash
+                self._local_transfers.append((rid, e_rank, PIX_SEND, C, H, W))
+                self.plane.enqueue_transfer(e_rank, rid, 0, PIX_RECV, C, H, W)
+            else:
+                e_idx = None
+                self._encoder().accept_pixels(rid, pixels)
+            self._epd_phase2.append((rid, toks, max_new, e_idx, d_idx))
+
+    def _process_epd_phase2(self) -> None:
+        """Phase 2: embeddings hop encode->decode, then the request itself
+        dispatches (worker loops execute transfers BEFORE submits, so the
+        embedding is present when the request admits)."""
+        if not self._epd_phase2:
+            return
+        batch, self._epd_phase2 = self._epd_phase2, []
+        enc = self._encoder()
+        E = enc.embed_len()
+        for rid, toks, max_new, e_idx, d_idx in batch:
+            d_rank = self.workers[d_idx].rccl_rank or 0
+            e_rank = self.workers[e_idx].rccl_rank if e_idx is not None else 0
+            if e_idx is not None:
+                self.plane.enqueue_transfer(e_rank, rid, d_rank, EMB_SEND, E, 0, 0)
+                if d_rank == 0:
+                    self._local_transfers.append((rid, e_rank, EMB_RECV, E, 0, 0))
+                else:
+                    self.plane.enqueue_transfer(d_rank, rid, e_rank, EMB_RECV, E, 0, 0)
+            else:
+                # locally-encoded embedding
+                if d_rank == 0:
+                    self.local_engine.accept_embed(rid, enc.export_embed(rid))
+                else:
+                    self._local_transfers.append((rid, d_rank, EMB_SEND, E, 0, 0))
+                    self.plane.enqueue_transfer(d_rank, rid, 0, EMB_RECV, E, 0, 0)
+            self.workers[d_idx].incr_load()
+            self.inflight[rid] = d_idx
+            if d_rank == 0:
+                # await_embed holds prefill until the local EMB_RECV flush
+                # (remote encode); a locally-encoded embedding already pairs
+                self.local_engine.submit(toks, max_new, rid=rid,
+                                         await_embed=e_idx is not None)
+            else:
+                self.plane.enqueue(d_rank, rid, max_new, toks)
 
     @property
     def pending_count(self) -> int:
@@ -147,9 +280,16 @@ class TickGateway:
         while self._pending and len(new_reqs) < self.max_new_arrivals:
             new_reqs.append(self._pending.popleft())
         if new_reqs:
-            # PD: arrivals go to the PREFILL fleet only
-            route_pool = ([self.workers[i] for i in self.prefill_idxs]
-                          if self.pd_roles else self.workers)
+            # PD: arrivals go to the PREFILL fleet only; EPD: encode ranks
+            # never take text requests
+            if self.pd_roles:
+                route_pool = [self.workers[i] for i in self.prefill_idxs]
+                remap = self.prefill_idxs
+            elif self.encode_idxs:
+                route_pool = [self.workers[i] for i in self.decode_idxs]
+                remap = self.decode_idxs
+            else:
+                route_pool, remap = self.workers, None
             infos = [
                 SelectWorkerInfo(
                     request_id=str(rid), model_id=self.model_id, tokens=toks, est_tokens=len(toks)
@@ -166,8 +306,8 @@ class TickGateway:
             self.phase_t["route"] += dt
             for (rid, toks, max_new), sel in zip(new_reqs, sels):
                 sel = 0 if sel is None else sel
-                if self.pd_roles:
-                    sel = self.prefill_idxs[sel]
+                if remap is not None:
+                    sel = remap[sel]
                 self.workers[sel].incr_load()
                 self.inflight[rid] = sel
                 rank = self.workers[sel].rccl_rank
@@ -184,7 +324,11 @@ class TickGateway:
 
     def _tick_sync(self) -> int:
         self.phase_t["ticks"] += 1
-        # 1) route this tick's arrivals in one batch (one GPU kernel launch)
+        # 1) EPD hops (phase2 before phase1: both sides append per-rid
+        # entries in the same order, keeping the pairing deadlock-free),
+        # then route this tick's arrivals in one batch (one GPU kernel)
+        self._process_epd_phase2()
+        self._process_mm_arrivals()
         self._route_arrivals()
         # 2) ship remote work first (send AND recv posted together) so the
         # whole plane exchange overlaps the local engine step
@@ -239,7 +383,11 @@ class TickGateway:
             self.phase_t["plane"] += time.perf_counter() - tp
         # 3) route this tick's arrivals while the GPU runs (the tree kernel
         # uses its own HIP stream, so it doesn't queue behind the engine);
-        # local selections submit straight into the engine here
+        # local selections submit straight into the engine here.  EPD hops
+        # scheduled here ride the NEXT tick_send, paired with the local
+        # flush that follows it.
+        self._process_epd_phase2()
+        self._process_mm_arrivals()
         self._submit_direct = True
         try:
             self._route_arrivals()
@@ -373,7 +521,9 @@ class TickGateway:
     def _flush_local_transfers(self) -> None:
         if self._local_transfers and self.local_engine is not None:
             trs, self._local_transfers = self._local_transfers, []
-            execute_transfers(self.local_engine, trs)
+            target = (self._LocalMMEndpoint(self.local_engine, self._local_encoder)
+                      if self._local_encoder is not None else self.local_engine)
+            execute_transfers(target, trs)
 
     def barrier_sync(self) -> None:
         """Plane barrier tick + dist barrier (bench timing bracket).  Any
@@ -405,6 +555,12 @@ def pd_rank_roles(world: int) -> Dict[int, str]:
     return {r: ("prefill" if r % 2 == 1 else "decode") for r in range(world)}
 
 
+def epd_rank_roles(world: int) -> Dict[int, str]:
+    """The plane EPD topology rule: rank 1 runs the vision encoder, every
+    other rank decodes (world 1: the gateway encodes locally)."""
+    return {r: ("encode" if r == 1 and world > 1 else "decode") for r in range(world)}
+
+
 def run_worker_loop(engine, plane: WorkerPlane, decode_burst: int = 1,
                     role: str = "regular") -> float:
     """Worker-rank loop (ranks >= 1): lockstep plane ticks against the local
@@ -415,6 +571,21 @@ def run_worker_loop(engine, plane: WorkerPlane, decode_burst: int = 1,
     (bench), else 0."""
     import torch
     import torch.distributed as dist
+
+    if role == "encode":
+        # EPD encode rank: no text engine — lockstep ticks + PIX/EMB
+        # transfers against the EncodeWorker passed as `engine`
+        while True:
+            reqs, stop = plane.tick([])
+            if plane.transfers:
+                execute_transfers(engine, plane.transfers)
+            if plane.barrier_requested:
+                dist.barrier()
+                if hasattr(torch.cuda, "is_available") and torch.cuda.is_available():
+                    torch.cuda.synchronize()
+            if stop:
+                break
+        return 0.0
 
     prefill_kw = {"prefill_only": True} if role == "prefill" else {}
     # Pipelined stepping (regular role): launch this tick's GPU work, resolve
