@@ -414,12 +414,23 @@ def rccl_worker_main(cfg) -> None:
         )
     )
     from .config import RoutingMode
-    from .routers.rccl_router import pd_rank_roles
+    from .routers.rccl_router import epd_rank_roles, pd_rank_roles
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
-    role = (pd_rank_roles(world).get(rank, "regular")
-            if cfg.mode == RoutingMode.PREFILL_DECODE else "regular")
+    if cfg.mode == RoutingMode.PREFILL_DECODE:
+        role = pd_rank_roles(world).get(rank, "regular")
+    elif cfg.mode == RoutingMode.ENCODE_PREFILL_DECODE:
+        role = epd_rank_roles(world).get(rank, "regular")
+    else:
+        role = "regular"
+    if role == "encode":
+        # EPD encode rank: the vision tower instead of a text engine
+        from .multimodal.encoder import EncodeWorker, ToyVisionEncoder
+
+        eng = EncodeWorker(ToyVisionEncoder(
+            eng.cfg.d_model, image_size=64, patch=16,
+            device=device, dtype=eng.dtype))
     run_worker_loop(eng, plane, role=role)
     dist.destroy_process_group()
 

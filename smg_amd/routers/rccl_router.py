@@ -701,10 +701,12 @@ class RcclRouter(Router):
         # bench --pd and cli.rccl_worker_main's parity rule)
         from ..config import RoutingMode
 
-        pd_roles = (
-            pd_rank_roles(world)
-            if config.mode == RoutingMode.PREFILL_DECODE and world > 1 else None
-        )
+        if config.mode == RoutingMode.PREFILL_DECODE and world > 1:
+            pd_roles = pd_rank_roles(world)
+        elif config.mode == RoutingMode.ENCODE_PREFILL_DECODE and world > 1:
+            pd_roles = epd_rank_roles(world)
+        else:
+            pd_roles = None
         self.gw = TickGateway(
             self.workers,
             policy,
