@@ -1023,6 +1023,8 @@ class TorchEngine:
         if entry is None:
             if key in self._pfg_bad or len(self._pfg_cache) >= self.PFG_CAP:
                 return None
+            if len(self._pfg_seen) > 512:  # arbitrary-L traffic: bound the counter map
+                self._pfg_seen.clear()
             seen = self._pfg_seen.get(key, 0) + 1
             self._pfg_seen[key] = seen
             if seen < self.PFG_MIN_SEEN:

@@ -303,6 +303,8 @@ class TickGateway:
                 sels = [self.policy.select_worker(route_pool, i) for i in infos]
             dt = time.perf_counter() - t0
             self.routing_lat.extend([dt / len(new_reqs)] * len(new_reqs))
+            if len(self.routing_lat) > 8192:  # long-lived server: cap the window
+                del self.routing_lat[:-4096]
             self.phase_t["route"] += dt
             for (rid, toks, max_new), sel in zip(new_reqs, sels):
                 sel = 0 if sel is None else sel
