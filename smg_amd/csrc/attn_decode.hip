@@ -601,7 +601,19 @@ __device__ __forceinline__ float dot2_bf16(unsigned int a, unsigned int b, float
     return __builtin_amdgcn_fdot2_f32_bf16(ca.v, cb.v, acc, false);
 }
 
-template <int G>
+__device__ __forceinline__ void fp8x4_to_bf16x2(unsigned int w, unsigned int* p) {
+    // 4 fp8 -> 2 packed bf16 pairs: v_cvt_pk_f32_fp8 + v_cvt_pk_bf16_f32,
+    // EXACT (e4m3 mantissa/exponent embed in bf16), 4 ops per 4 values vs 4
+    // converts + widening — and the result feeds v_dot2c directly
+    __f32x2 lo = __builtin_amdgcn_cvt_pk_f32_fp8(w, false);
+    __f32x2 hi = __builtin_amdgcn_cvt_pk_f32_fp8(w, true);
+    __hip_bfloat162 a = __float22bfloat162_rn(float2{lo[0], lo[1]});
+    __hip_bfloat162 b = __float22bfloat162_rn(float2{hi[0], hi[1]});
+    p[0] = *(const unsigned int*)&a;
+    p[1] = *(const unsigned int*)&b;
+}
+
+template <bool KV8, int G>
 __global__ void __launch_bounds__(WAVE, 2) smg_attn_decode2_t(
     const __hip_bfloat16* __restrict__ q,
     const void* __restrict__ k,
@@ -620,6 +632,8 @@ __global__ void __launch_bounds__(WAVE, 2) smg_attn_decode2_t(
     const size_t head_base = ((size_t)slot * n_kv_heads + kvh) * (size_t)max_seq * head_dim;
     const __hip_bfloat16* kh = (const __hip_bfloat16*)k + head_base;
     const __hip_bfloat16* vh = (const __hip_bfloat16*)v + head_base;
+    const unsigned char* kh8 = (const unsigned char*)k + head_base;
+    const unsigned char* vh8 = (const unsigned char*)v + head_base;
     const __hip_bfloat16* qh = q + ((size_t)slot * n_heads + (size_t)kvh * G) * head_dim;
 
     // q staged as packed bf16 pairs (dot2 operands)
@@ -652,16 +666,39 @@ __global__ void __launch_bounds__(WAVE, 2) smg_attn_decode2_t(
 #pragma unroll
         for (int g = 0; g < G; ++g) d[g] = 0.f;
         if (t < T) {
-            const uint4* row = (const uint4*)(kh + (size_t)t * head_dim);
+            if constexpr (KV8) {
+                // fp8 K row: packed fp8->f32->bf16x2 repack (exact), then the
+                // same dot2 stream as bf16 — 6 ops per 4 elems vs 9 for the
+                // per-element convert+FMA form
+                const uint4* row = (const uint4*)(kh8 + (size_t)t * head_dim);
 #pragma unroll 4
-            for (int c = 0; c < vec_n; ++c) {
-                uint4 w = row[c];
-                const unsigned int* wp = (const unsigned int*)&w;
+                for (int c = 0; c < vec_n / 2; ++c) {
+                    uint4 w = row[c];  // 16 fp8
+                    const unsigned int* wp = (const unsigned int*)&w;
 #pragma unroll
-                for (int wi = 0; wi < 4; ++wi)
+                    for (int wi = 0; wi < 4; ++wi) {
+                        unsigned int pk[2];
+                        fp8x4_to_bf16x2(wp[wi], pk);
+                        const int qb = c * 8 + wi * 2;  // bf16-pair index of elem c*16+wi*4
 #pragma unroll
-                    for (int g = 0; g < G; ++g)
-                        d[g] = dot2_bf16(wp[wi], s_q2[g * (head_dim / 2) + c * 4 + wi], d[g]);
+                        for (int g = 0; g < G; ++g) {
+                            d[g] = dot2_bf16(pk[0], s_q2[g * (head_dim / 2) + qb], d[g]);
+                            d[g] = dot2_bf16(pk[1], s_q2[g * (head_dim / 2) + qb + 1], d[g]);
+                        }
+                    }
+                }
+            } else {
+                const uint4* row = (const uint4*)(kh + (size_t)t * head_dim);
+#pragma unroll 4
+                for (int c = 0; c < vec_n; ++c) {
+                    uint4 w = row[c];
+                    const unsigned int* wp = (const unsigned int*)&w;
+#pragma unroll
+                    for (int wi = 0; wi < 4; ++wi)
+#pragma unroll
+                        for (int g = 0; g < G; ++g)
+                            d[g] = dot2_bf16(wp[wi], s_q2[g * (head_dim / 2) + c * 4 + wi], d[g]);
+                }
             }
         }
         float alpha[G];
@@ -684,29 +721,45 @@ __global__ void __launch_bounds__(WAVE, 2) smg_attn_decode2_t(
             for (int j = 0; j < 8; ++j) accv[g][j] *= alpha[g];
         }
         __syncthreads();
-        const __hip_bfloat16* vtile = vh + (size_t)base * head_dim + chunk * 8;
-        if (lim == WAVE && head_dim == 128) {
-#pragma unroll
-            for (int it = 0; it < 16; ++it) {
-                const int j = rgrp + it * 4;  // rows_per == 4 at hd128
-                const uint4 w = *(const uint4*)(vtile + (size_t)j * head_dim);
-                const unsigned short* hsp = (const unsigned short*)&w;
+        if constexpr (KV8) {
+            const unsigned char* vtile8 = vh8 + (size_t)base * head_dim + chunk * 8;
+            for (int j = rgrp; j < lim; j += rows_per) {
+                const uint2 w = *(const uint2*)(vtile8 + (size_t)j * head_dim);
+                float dec[8];
+                fp8x4_to_f32(w.x, dec);
+                fp8x4_to_f32(w.y, dec + 4);
 #pragma unroll
                 for (int g = 0; g < G; ++g) {
                     const float pj = s_p[g][j];
 #pragma unroll
-                    for (int jj = 0; jj < 8; ++jj) accv[g][jj] += pj * bf16_to_f32(hsp[jj]);
+                    for (int s = 0; s < 8; ++s) accv[g][s] += pj * dec[s];
                 }
             }
         } else {
-            for (int j = rgrp; j < lim; j += rows_per) {
-                const uint4 w = *(const uint4*)(vtile + (size_t)j * head_dim);
-                const unsigned short* hsp = (const unsigned short*)&w;
+            const __hip_bfloat16* vtile = vh + (size_t)base * head_dim + chunk * 8;
+            if (lim == WAVE && head_dim == 128) {
 #pragma unroll
-                for (int g = 0; g < G; ++g) {
-                    const float pj = s_p[g][j];
+                for (int it = 0; it < 16; ++it) {
+                    const int j = rgrp + it * 4;  // rows_per == 4 at hd128
+                    const uint4 w = *(const uint4*)(vtile + (size_t)j * head_dim);
+                    const unsigned short* hsp = (const unsigned short*)&w;
 #pragma unroll
-                    for (int jj = 0; jj < 8; ++jj) accv[g][jj] += pj * bf16_to_f32(hsp[jj]);
+                    for (int g = 0; g < G; ++g) {
+                        const float pj = s_p[g][j];
+#pragma unroll
+                        for (int jj = 0; jj < 8; ++jj) accv[g][jj] += pj * bf16_to_f32(hsp[jj]);
+                    }
+                }
+            } else {
+                for (int j = rgrp; j < lim; j += rows_per) {
+                    const uint4 w = *(const uint4*)(vtile + (size_t)j * head_dim);
+                    const unsigned short* hsp = (const unsigned short*)&w;
+#pragma unroll
+                    for (int g = 0; g < G; ++g) {
+                        const float pj = s_p[g][j];
+#pragma unroll
+                        for (int jj = 0; jj < 8; ++jj) accv[g][jj] += pj * bf16_to_f32(hsp[jj]);
+                    }
                 }
             }
         }
@@ -735,30 +788,36 @@ extern "C" int smg_attn_decode_launch_gqa2(const void* q, const void* k, const v
                                            const void* pos, void* out, int n_slots, int n_heads,
                                            int n_kv_heads, int max_seq, int head_dim, float scale,
                                            void* stream, int kv_fp8) {
-    if (kv_fp8) // fp8 is convert-bound, not helped by dot2 — keep the v7 path
-        return smg_attn_decode_launch_gqa(q, k, v, pos, out, n_slots, n_heads, n_kv_heads,
-                                          max_seq, head_dim, scale, stream, kv_fp8);
     if (head_dim > 128 || (head_dim & 7)) return -1;
     int chunks = head_dim / 8;
     if (chunks & (chunks - 1)) return -1;
+    if (kv_fp8 && (head_dim & 15)) return -1;  // fp8 K streaming is 16-wide
     if (n_kv_heads <= 0 || n_heads % n_kv_heads) return -4;
     dim3 grid(n_slots * n_kv_heads);
     const int G = n_heads / n_kv_heads;
     hipStream_t s = (hipStream_t)stream;
-#define LAUNCH2_CASE(GG)                                                                       \
-    case GG:                                                                                   \
-        hipLaunchKernelGGL((smg_attn_decode2_t<GG>), grid, dim3(WAVE), 0, s,                   \
-                           (const __hip_bfloat16*)q, k, v, (const int*)pos,                    \
-                           (__hip_bfloat16*)out, n_slots, n_heads, n_kv_heads, max_seq,        \
-                           head_dim, scale);                                                   \
-        break;
-    switch (G) {
-        LAUNCH2_CASE(1)
-        LAUNCH2_CASE(2)
-        LAUNCH2_CASE(4)
-        LAUNCH2_CASE(8)
-        default:
-            return -4;
+#define LAUNCH2_CASE(K8, GG)                                                                   \
+    hipLaunchKernelGGL((smg_attn_decode2_t<K8, GG>), grid, dim3(WAVE), 0, s,                   \
+                       (const __hip_bfloat16*)q, k, v, (const int*)pos,                        \
+                       (__hip_bfloat16*)out, n_slots, n_heads, n_kv_heads, max_seq,            \
+                       head_dim, scale);                                                       \
+    break;
+    if (kv_fp8) {
+        switch (G) {
+            case 1: LAUNCH2_CASE(true, 1)
+            case 2: LAUNCH2_CASE(true, 2)
+            case 4: LAUNCH2_CASE(true, 4)
+            case 8: LAUNCH2_CASE(true, 8)
+            default: return -4;
+        }
+    } else {
+        switch (G) {
+            case 1: LAUNCH2_CASE(false, 1)
+            case 2: LAUNCH2_CASE(false, 2)
+            case 4: LAUNCH2_CASE(false, 4)
+            case 8: LAUNCH2_CASE(false, 8)
+            default: return -4;
+        }
     }
 #undef LAUNCH2_CASE
     return hipGetLastError() == hipSuccess ? 0 : -2;
