@@ -65,6 +65,42 @@ class TestStreamingEquivalence:
         assert r_s.strip() == "hmm"
         assert n_s.strip() == "done"
 
+    def test_harmony_channels_char_at_a_time(self):
+        """gpt-oss Harmony: analysis -> reasoning, final -> normal, streamed
+        one char at a time must equal the complete parse."""
+        text = ("<|start|>assistant<|channel|>analysis<|message|>let me think<|end|>"
+                "<|start|>assistant<|channel|>final<|message|>The answer is 4.<|end|>")
+        r_c, n_c = parse_reasoning_complete("harmony", text)
+        assert "let me think" in r_c and "answer is 4" in n_c
+        for chunk in (1, 5):
+            r_s, n_s = drip("harmony", text, chunk)
+            assert r_s.strip() == r_c
+            assert n_s.strip() == n_c
+
+    def test_harmony_tool_call_reemitted_for_tool_parser(self):
+        """A commentary tool-call segment must survive the reasoning stage in
+        raw Harmony framing so the downstream harmony TOOL parser extracts
+        it (pipeline order: reasoning -> tool)."""
+        from smg_amd.parsers.tool import parse_complete as tool_complete
+
+        text = ("<|start|>assistant<|channel|>analysis<|message|>need weather<|end|>"
+                "<|start|>assistant<|channel|>commentary to=functions.get_weather"
+                "<|message|>{\"city\": \"Paris\"}<|call|>")
+        r_s, n_s = drip("harmony", text, 3)
+        assert "need weather" in r_s
+        _, calls = tool_complete("harmony", n_s)
+        assert calls and calls[0]["name"] == "get_weather"
+        assert "Paris" in calls[0]["arguments"]
+
+    def test_harmony_truncated_no_crash(self):
+        text = "<|start|>assistant<|channel|>analysis<|message|>thinking ab"
+        r_s, n_s = drip("harmony", text)
+        p = get_reasoning_parser("harmony")
+        # truncation at every prefix must not raise
+        for cut in range(0, len(text), 7):
+            q = get_reasoning_parser("harmony")
+            q.parse_streaming(text[:cut])
+
     def test_model_pattern_dispatch(self):
         assert get_reasoning_parser("deepseek-r1-distill").name == "deepseek_r1"
         assert get_reasoning_parser("Qwen3-32B").name == "qwen3"
