@@ -792,11 +792,38 @@ class RcclRouter(Router):
             return await self.route_generate(req)
         return RouteResponse(status=404, body=b'{"error":"unsupported path for rccl router"}')
 
+    async def _extract_pixels(self, body: Dict[str, Any]):
+        """First image_url content part (data:/file URL) -> [3, H, W] u8
+        tensor for the EPD pixel path, or None for text-only requests."""
+        for m in body.get("messages") or []:
+            content = m.get("content")
+            if not isinstance(content, list):
+                continue
+            for part in content:
+                if isinstance(part, dict) and part.get("type") == "image_url":
+                    url = (part.get("image_url") or {}).get("url") or ""
+                    import numpy as np_
+                    import torch
+
+                    from ..multimodal.media import decode_image, fetch_image_bytes
+
+                    raw = await fetch_image_bytes(url)
+                    arr = decode_image(raw)  # [H, W, 3] u8
+                    return torch.from_numpy(np_.ascontiguousarray(
+                        arr.transpose(2, 0, 1)))
+        return None
+
     async def route_chat(self, req: RouteRequest) -> RouteResponse:
         body = req.body or {}
         tok = self._tokenizer()
         input_ids = self._encode_chat(body, tok)
-        return await self._serve(req, body, tok, input_ids, chat=True)
+        try:
+            pixels = await self._extract_pixels(body)
+        except Exception as exc:
+            return RouteResponse(status=400, body=json.dumps(
+                {"error": {"message": f"image fetch/decode failed: {exc}",
+                           "type": "invalid_request_error"}}).encode())
+        return await self._serve(req, body, tok, input_ids, chat=True, pixels=pixels)
 
     async def route_completion(self, req: RouteRequest) -> RouteResponse:
         body = req.body or {}
@@ -814,12 +841,13 @@ class RcclRouter(Router):
         body2 = {"max_tokens": (body.get("sampling_params") or {}).get("max_new_tokens", 16)}
         return await self._serve(req, body2, tok, ids, chat=False, generate=True)
 
-    async def _serve(self, req, body, tok, input_ids, chat: bool, generate: bool = False) -> RouteResponse:
+    async def _serve(self, req, body, tok, input_ids, chat: bool, generate: bool = False,
+                     pixels=None) -> RouteResponse:
         max_new = int(
             body.get("max_completion_tokens") or body.get("max_tokens") or 16
         )
         st = _ReqState()
-        rid = self.gw.submit(input_ids, max_new)
+        rid = self.gw.submit(input_ids, max_new, pixels=pixels)
         self._states[rid] = st
         model = body.get("model") or self.model_id
         stream = bool(body.get("stream"))

@@ -71,6 +71,52 @@ class TestRcclRouterSingleRank:
 
         asyncio.new_event_loop().run_until_complete(run())
 
+    def test_multimodal_chat_epd_local_encode(self):
+        """world=1 EPD intake: an image_url data: URI in the chat body rides
+        the pixel path (local ToyVisionEncoder) and the embedded request
+        serves through the same tick loop."""
+
+        async def run():
+            import base64
+            import io
+
+            from aiohttp.test_utils import TestClient, TestServer
+            from PIL import Image
+
+            from smg_amd.routers.factory import RouterManager
+            from smg_amd.server.app import build_app
+
+            buf = io.BytesIO()
+            Image.new("RGB", (40, 30), (200, 40, 90)).save(buf, format="PNG")
+            uri = "data:image/png;base64," + base64.b64encode(buf.getvalue()).decode()
+
+            ctx, cfg = make_rccl_ctx()
+            ctx.router_manager = RouterManager(ctx, cfg)
+            client = TestClient(TestServer(build_app(ctx)))
+            await client.start_server()
+            try:
+                body = {
+                    "model": "default",
+                    "messages": [{"role": "user", "content": [
+                        {"type": "text", "text": "what is in this image?"},
+                        {"type": "image_url", "image_url": {"url": uri}},
+                    ]}],
+                    "max_tokens": 4,
+                }
+                r = await client.post("/v1/chat/completions", json=body)
+                assert r.status == 200, await r.text()
+                j = await r.json()
+                assert j["usage"]["completion_tokens"] == 4
+                # a corrupt image is a 400, not a hang
+                body["messages"][0]["content"][1]["image_url"]["url"] = "data:image/png;base64,AAAA"
+                r = await client.post("/v1/chat/completions", json=body)
+                assert r.status == 400
+            finally:
+                await ctx.router_manager.shutdown()
+                await client.close()
+
+        asyncio.new_event_loop().run_until_complete(run())
+
     def test_streaming_sse(self):
         async def run():
             from aiohttp.test_utils import TestClient, TestServer
