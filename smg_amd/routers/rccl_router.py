@@ -255,7 +255,7 @@ class TickGateway:
 
     @property
     def pending_count(self) -> int:
-        return len(self._pending)
+        return len(self._pending) + len(self._pending_mm)
 
     # ---- the tick (one thread) -------------------------------------------
     def tick(self) -> int:
