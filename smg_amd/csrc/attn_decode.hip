@@ -587,9 +587,11 @@ extern "C" int smg_attn_decode_launch_gqa(const void* q, const void* k, const vo
 // bench shapes and is latency/VALU-bound, not bandwidth-bound — the K dot
 // spends 5 VALU ops per bf16 pair (2 converts + 2 FMA + addressing).  Here
 // the K dot is one v_dot2c_f32_bf16 per pair against q pre-packed as bf16x2
-// in LDS (2.5x fewer K-phase VALU ops).  A two-tile (128-timestep) softmax
-// round was also tried: 256 VGPRs + 56 spilled — worse than the barrier
-// cost it saves, so rounds stay 64 timesteps.
+// in LDS (2.5x fewer K-phase VALU ops).  Also tried and MEASURED WORSE:
+// a two-tile (128-timestep) softmax round (256 VGPRs + 56 spilled) and
+// prefetching 6 V rows across the softmax phase (242 VGPRs, no spill, but
+// 198 -> 205 us — the register pressure costs more scheduling freedom than
+// the overlap buys), so rounds stay 64 timesteps with inline V loads.
 // ---------------------------------------------------------------------------
 typedef short __bf16x2 __attribute__((__vector_size__(2 * sizeof(short))));
 
