@@ -451,6 +451,8 @@ class TorchEngine:
     def accept_embed(self, rid, emb: torch.Tensor) -> None:
         """EMB_RECV landing: [E, d_model] embeddings stash until the matching
         request submit()s (plane transfers execute before submits each tick)."""
+        if len(self._pending_embeds) > 4096:  # orphaned rids (aborted requests)
+            self._pending_embeds.clear()
         self._pending_embeds[rid] = emb.to(self.device, self.dtype)
         # a request that arrived first picks the embedding up here instead
         req = self._requests.get(rid)
