@@ -110,6 +110,40 @@ def test_embedded_prefill_gpu_matches_eager():
 
 
 class TestGatewayEpd:
+    def _world1_roundtrip(self, device: str):
+        from smg_amd.config import PolicyConfig
+        from smg_amd.policies import create_policy
+        from smg_amd.routers.rccl_router import TickGateway
+        from smg_amd.workers.worker import Worker
+
+        cfg = TorchEngineConfig.tiny()
+        if device != "cpu":
+            cfg.dtype = "bfloat16"
+        eng = TorchEngine(cfg, device=device)
+        gw = TickGateway([Worker("rccl://rank-0", rccl_rank=0)],
+                         create_policy(PolicyConfig(name="round_robin")),
+                         local_engine=eng)
+        g = torch.Generator().manual_seed(4)
+        px = (torch.rand(3, 36, 44, generator=g) * 255).to(torch.uint8)
+        seen = []
+        gw.on_event = lambda rid, tok, fl: seen.append((rid, tok, fl))
+        gw.submit(list(range(16)), 4, rid=9, pixels=px)
+        for _ in range(60):
+            gw.tick()
+            if gw.completed_total:
+                break
+        gw._drain_pipeline()
+        assert gw.completed_total == 1
+        toks = [t for rid, t, _ in seen if rid == 9]
+        assert len(toks) == 4 and all(t >= 0 for t in toks)
+        return toks
+
+    @pytest.mark.gpu
+    def test_world1_epd_serving_gpu(self):
+        """The full pixels->encode->embedded-prefill->decode serving loop on
+        cuda:0 (hardware coverage of the gateway EPD wrapper)."""
+        self._world1_roundtrip("cuda:0")
+
     def test_world1_local_encode_serving(self):
         """TickGateway EPD at world 1: pixels encode locally, the embedded
         request decodes on the local engine through the normal tick loop."""
